@@ -1,0 +1,240 @@
+"""SELECT finalization: GROUP BY aggregation, ORDER BY, DISTINCT,
+LIMIT/OFFSET, projection.
+
+Ref parity: execute_query.rs:404-475 (aggregate_rows: COUNT/SUM/AVG/MIN/MAX),
+:477 (numeric-aware ORDER BY), engine.rs:785-905 (subquery finalize).
+
+MI355X-native: aggregation runs on the device value column (float64 parsed
+once at encode time — K4 class segmented reduce) instead of per-row string
+parsing; only final aggregate *results* are encoded back into the host
+dictionary.
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Optional
+
+import torch
+
+from ..parsing.ast import Projection, SelectQuery
+from ..storage.terms import UNBOUND
+from .bindings import Bindings
+from .tensor_utils import group_index, lexsort, unique_rows
+
+
+def _fmt_num(v: float) -> str:
+    if v != v:  # NaN
+        return "0"
+    if v == int(v) and abs(v) < 1e15:
+        return str(int(v))
+    return repr(v)
+
+
+def _encode_i32(db, s: str) -> int:
+    x = db.dictionary.encode(s) & 0xFFFFFFFF
+    return x - 0x1_0000_0000 if x >= 0x8000_0000 else x
+
+
+def finalize_select_bindings(select: SelectQuery, rows: Bindings, db
+                             ) -> Bindings:
+    """ID-level finalize (used for subqueries and as the core of the
+    top-level path)."""
+    dev = rows.device
+    has_agg = any(p.aggregate for p in select.variables)
+
+    if has_agg or select.group_by:
+        rows = _aggregate(select, rows, db)
+    # projection
+    if select.select_star or not select.variables:
+        proj_names = rows.variables
+    else:
+        proj_names = [p.output_name() for p in select.variables]
+        rows = rows.project(proj_names)
+    # ORDER BY (numeric-aware, ref execute_query.rs:477)
+    if select.order_by and rows.n > 1:
+        perm = _order_perm(select, rows, db)
+        rows = rows.gather(perm)
+    if select.distinct and rows.n > 1:
+        cols = [rows.col(v) for v in proj_names]
+        if select.order_by:
+            # stable distinct preserving order: keep first occurrence
+            gid, _ = group_index(cols)
+            first = torch.zeros(rows.n, dtype=torch.bool, device=dev)
+            seen: Dict[int, bool] = {}
+            gl = gid.cpu().tolist()
+            keep_idx = []
+            for i, g in enumerate(gl):
+                if g not in seen:
+                    seen[g] = True
+                    keep_idx.append(i)
+            rows = rows.gather(torch.tensor(keep_idx, dtype=torch.long, device=dev))
+        else:
+            uc = unique_rows(cols)
+            rows = Bindings(dict(zip(proj_names, uc)),
+                            uc[0].numel() if uc else 0, dev)
+    # OFFSET / LIMIT
+    off = select.offset or 0
+    if off or select.limit is not None:
+        end = rows.n if select.limit is None else min(rows.n, off + select.limit)
+        idx = torch.arange(off, max(off, end), dtype=torch.long, device=dev)
+        rows = rows.gather(idx)
+    return rows
+
+
+def _order_perm(select: SelectQuery, rows: Bindings, db) -> torch.Tensor:
+    dev = rows.device
+    perm = torch.arange(rows.n, dtype=torch.long, device=dev)
+    for cond in reversed(select.order_by):
+        if not rows.has(cond.var):
+            continue
+        ids = rows.col(cond.var)[perm]
+        ids_u = ids.to(torch.int64) & 0xFFFFFFFF
+        strs = [db.dictionary.decode(int(x)) or "" for x in ids_u.cpu().tolist()]
+        all_numeric = all(_is_num(s) for s in strs) and len(strs) > 0
+        if all_numeric:
+            vc = db.value_column()
+            vals = vc[torch.clamp(ids_u, max=vc.numel() - 1)]
+            key = torch.argsort(vals, stable=True, descending=cond.descending)
+        else:
+            import numpy as np
+            order = sorted(range(len(strs)), key=lambda i: strs[i],
+                           reverse=cond.descending)
+            key = torch.tensor(order, dtype=torch.long, device=dev)
+        perm = perm[key]
+    return perm
+
+
+def _is_num(s: str) -> bool:
+    try:
+        float(s)
+        return True
+    except ValueError:
+        return False
+
+
+def _aggregate(select: SelectQuery, rows: Bindings, db) -> Bindings:
+    """GROUP BY + aggregates on the device value column (K4 class)."""
+    dev = rows.device
+    group_vars = [v for v in select.group_by if rows.has(v)]
+    n = rows.n
+    if n == 0:
+        out_cols = {}
+        for v in group_vars:
+            out_cols[v] = torch.empty(0, dtype=torch.int32, device=dev)
+        for p in select.variables:
+            if p.aggregate:
+                out_cols[p.output_name()] = torch.empty(0, dtype=torch.int32, device=dev)
+        if not group_vars:
+            # aggregates over empty input: COUNT=0, others empty-string
+            vals = {}
+            for p in select.variables:
+                if p.aggregate == "COUNT":
+                    vals[p.output_name()] = _encode_i32(db, "0")
+                elif p.aggregate:
+                    vals[p.output_name()] = _encode_i32(db, "")
+            return Bindings(
+                {k: torch.tensor([v], dtype=torch.int32, device=dev)
+                 for k, v in vals.items()},
+                1, dev)
+        return Bindings(out_cols, 0, dev)
+
+    if group_vars:
+        gid, ng = group_index([rows.col(v) for v in group_vars])
+    else:
+        gid = torch.zeros(n, dtype=torch.long, device=dev)
+        ng = 1
+    # representative row per group (first occurrence)
+    rep = torch.full((ng,), -1, dtype=torch.long, device=dev)
+    rev = torch.arange(n - 1, -1, -1, dtype=torch.long, device=dev)
+    rep.scatter_(0, gid.flip(0), rev)  # last write wins => first row index
+    out_cols: Dict[str, torch.Tensor] = {}
+    for v in group_vars:
+        out_cols[v] = rows.col(v)[rep]
+    vc = db.value_column()
+    for p in select.variables:
+        if not p.aggregate:
+            if p.var and p.var not in out_cols and rows.has(p.var):
+                out_cols[p.var] = rows.col(p.var)[rep]
+            continue
+        name = p.output_name()
+        if p.aggregate == "COUNT":
+            if p.agg_arg is None:
+                cnt = torch.zeros(ng, dtype=torch.int64, device=dev)
+                cnt.scatter_add_(0, gid, torch.ones(n, dtype=torch.int64, device=dev))
+            else:
+                col = rows.col(p.agg_arg) if rows.has(p.agg_arg) else None
+                if col is None:
+                    cnt = torch.zeros(ng, dtype=torch.int64, device=dev)
+                elif p.distinct:
+                    pair = unique_rows([gid.to(torch.int32), col])
+                    bound = pair[1] != UNBOUND
+                    cnt = torch.zeros(ng, dtype=torch.int64, device=dev)
+                    cnt.scatter_add_(0, pair[0][bound].to(torch.int64),
+                                     torch.ones(int(bound.sum()), dtype=torch.int64, device=dev))
+                else:
+                    bound = (col != UNBOUND).to(torch.int64)
+                    cnt = torch.zeros(ng, dtype=torch.int64, device=dev)
+                    cnt.scatter_add_(0, gid, bound)
+            out_cols[name] = _encode_numbers(db, cnt.to(torch.float64), dev, integral=True)
+            continue
+        col = rows.col(p.agg_arg) if p.agg_arg and rows.has(p.agg_arg) else None
+        if col is None:
+            out_cols[name] = torch.full((ng,), _encode_i32(db, ""),
+                                        dtype=torch.int32, device=dev)
+            continue
+        ids_u = col.to(torch.int64) & 0xFFFFFFFF
+        vals = vc[torch.clamp(ids_u, max=vc.numel() - 1)]
+        bound = col != UNBOUND
+        if p.aggregate in ("SUM", "AVG"):
+            acc = torch.zeros(ng, dtype=torch.float64, device=dev)
+            acc.scatter_add_(0, gid[bound], vals[bound])
+            if p.aggregate == "AVG":
+                cnt = torch.zeros(ng, dtype=torch.float64, device=dev)
+                cnt.scatter_add_(0, gid[bound],
+                                 torch.ones(int(bound.sum()), dtype=torch.float64, device=dev))
+                acc = torch.where(cnt > 0, acc / torch.clamp(cnt, min=1), acc)
+            out_cols[name] = _encode_numbers(db, acc, dev)
+        elif p.aggregate in ("MIN", "MAX"):
+            init = float("inf") if p.aggregate == "MIN" else float("-inf")
+            acc = torch.full((ng,), init, dtype=torch.float64, device=dev)
+            acc.scatter_reduce_(0, gid[bound], vals[bound],
+                                reduce="amin" if p.aggregate == "MIN" else "amax")
+            acc = torch.where(torch.isinf(acc), torch.zeros_like(acc), acc)
+            out_cols[name] = _encode_numbers(db, acc, dev)
+        else:
+            raise ValueError(f"unknown aggregate {p.aggregate}")
+    return Bindings(out_cols, ng, dev)
+
+
+def _encode_numbers(db, vals: torch.Tensor, dev, integral: bool = False
+                    ) -> torch.Tensor:
+    """Host-encode aggregate results back into the dictionary."""
+    out = []
+    for v in vals.cpu().tolist():
+        s = str(int(v)) if integral else _fmt_num(v)
+        out.append(_encode_i32(db, s))
+    return torch.tensor(out, dtype=torch.int32, device=dev)
+
+
+def decode_rows(select: SelectQuery, rows: Bindings, db) -> List[List[str]]:
+    """Final string decode (only at the top, ref engine.rs:347-373)."""
+    if select.select_star or not select.variables:
+        names = rows.variables
+    else:
+        names = [p.output_name() for p in select.variables]
+    host = {}
+    for v in names:
+        if rows.has(v):
+            host[v] = (rows.col(v).to(torch.int64) & 0xFFFFFFFF).cpu().tolist()
+        else:
+            host[v] = [None] * rows.n
+    out: List[List[str]] = []
+    for i in range(rows.n):
+        row = []
+        for v in names:
+            x = host[v][i]
+            if x is None or x == 0xFFFFFFFF:
+                row.append("")
+            else:
+                row.append(db.decode_term(x) or "")
+        out.append(row)
+    return out

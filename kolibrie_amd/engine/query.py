@@ -1,0 +1,98 @@
+"""Unified query entry point (ref: kolibrie/src/execute_query.rs — parse ->
+prepare extensions -> dataset view -> lower -> optimize -> execute ->
+decode -> finalize_select; update path with deletes-before-inserts).
+"""
+from __future__ import annotations
+
+from typing import Dict, List, Optional
+
+from ..parsing.ast import CombinedQuery, SelectQuery
+from ..parsing.sparql import ParseError, parse_combined_query
+from ..storage.dataset import DEFAULT_GRAPH
+from .bindings import Bindings
+from .executor import DatasetView, ExecutionContext, ExecutionEngine
+from .finalize import decode_rows, finalize_select_bindings
+from .update import execute_update
+
+
+def _build_view(select: SelectQuery, db, prefixes) -> DatasetView:
+    """FROM / FROM NAMED -> DatasetView (ref execute_query.rs:228)."""
+    view = DatasetView()
+    if select.from_graphs:
+        view.default_graphs = [
+            db.dictionary.encode(db.resolve_lexical(g, prefixes))
+            for g in select.from_graphs
+        ]
+    if select.from_named:
+        view.named_graphs = [
+            db.dictionary.encode(db.resolve_lexical(g, prefixes))
+            for g in select.from_named
+        ]
+    elif select.from_graphs:
+        # explicit dataset without FROM NAMED: no named graphs visible
+        view.named_graphs = []
+    return view
+
+
+def execute_select(select: SelectQuery, db, prefixes: Dict[str, str]
+                   ) -> List[List[str]]:
+    from ..plan.lower import build_logical_plan
+    from ..plan.optimizer import Streamertail
+    view = _build_view(select, db, prefixes)
+    stats = db.get_or_build_stats()
+    logical = build_logical_plan(select.where, db, prefixes)
+    physical = Streamertail(stats).find_best_plan(logical)
+    ctx = ExecutionContext(db, view)
+    rows = ExecutionEngine(ctx).execute(physical, Bindings.unit(db.device))
+    final = finalize_select_bindings(select, rows, db)
+    return decode_rows(select, final, db)
+
+
+def prepare_extensions(cq: CombinedQuery, db, prefixes: Dict[str, str]):
+    """Register MODEL / NEURAL RELATION / TRAIN decls and RULEs into the DB
+    (ref execute_query.rs:164 prepare_extensions, parser.rs:3607
+    process_rule_definition)."""
+    for m in cq.models:
+        db.neural_models[m.name] = {"decl": m}
+    for nr in cq.neural_relations:
+        db.neural_relations[nr.name] = {"decl": nr}
+    for td in cq.train_decls:
+        db.train_decls.append({"decl": td})
+        from ..ml.train import execute_train_decl
+        execute_train_decl(td, db, prefixes)
+    for r in cq.rules:
+        db.rule_map[r.name] = r
+        from ..reasoning.rule import convert_combined_rule
+        rule = convert_combined_rule(r, db, prefixes)
+        db.rules.append(rule)
+
+
+def execute_query(sparql: str, db) -> List[List[str]]:
+    """Full request entry (ref execute_query_rayon_parallel2_volcano,
+    execute_query.rs:52)."""
+    cq = parse_combined_query(sparql)
+    prefixes = dict(db.prefixes)
+    prefixes.update(cq.prefixes)
+    prepare_extensions(cq, db, prefixes)
+    if cq.updates:
+        for op in cq.updates:
+            execute_update(op, db, prefixes)
+        if cq.select is None:
+            return []
+    if cq.select is not None:
+        return execute_select(cq.select, db, prefixes)
+    return []
+
+
+def execute_sparql_query(sparql: str, db) -> List[List[str]]:
+    """Query-only entry: rejects updates (ref execute_query.rs:71-89,
+    used by the HTTP /query endpoint)."""
+    cq = parse_combined_query(sparql)
+    if cq.updates:
+        raise ValueError("update operations are not allowed on the query endpoint")
+    prefixes = dict(db.prefixes)
+    prefixes.update(cq.prefixes)
+    prepare_extensions(cq, db, prefixes)
+    if cq.select is None:
+        return []
+    return execute_select(cq.select, db, prefixes)

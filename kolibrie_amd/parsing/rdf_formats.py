@@ -1,0 +1,283 @@
+"""RDF data-format readers: N-Triples(-star), N-Quads, Turtle(-star), RDF/XML.
+
+Parity surface: kolibrie/src/sparql_database.rs:630-1463 (RDF/XML pull
+parser, Turtle-star tokenizer with `<< >>` depth and `{| p o |}` annotation
+syntax, N3/N-Triples chunked parsers, N-Quads with optional graph term).
+
+Host-side only: parsing produces u32 ID columns which bulk-insert into the
+device store.  The tokenizer is shared across Turtle/N-Triples/N-Quads.
+"""
+from __future__ import annotations
+
+import re
+import xml.etree.ElementTree as ET
+from typing import List, Optional, Tuple
+
+RDF_NS = "http://www.w3.org/1999/02/22-rdf-syntax-ns#"
+RDF_TYPE = RDF_NS + "type"
+
+
+# --------------------------------------------------------------------- tokens
+_TOKEN_RE = re.compile(
+    r"""
+      (?P<ws>\s+|\#[^\n]*)
+    | (?P<qopen><<)
+    | (?P<qclose>>>)
+    | (?P<aopen>\{\|)
+    | (?P<aclose>\|\})
+    | (?P<iri><[^<>\s]*>)
+    | (?P<literal>
+        ("(?:[^"\\]|\\.)*"|'(?:[^'\\]|\\.)*')
+        (?:\^\^<[^<>\s]*>|\^\^[A-Za-z_][\w.-]*:[\w.-]*|@[A-Za-z][A-Za-z0-9-]*)?
+      )
+    | (?P<punct>[;,.\[\]\(\)])
+    | (?P<bnode>_:[A-Za-z0-9_.-]+)
+    | (?P<name>[^\s;,.\[\]\(\)]+)
+    """,
+    re.VERBOSE | re.DOTALL,
+)
+
+
+def tokenize(text: str) -> List[str]:
+    out = []
+    pos = 0
+    n = len(text)
+    while pos < n:
+        m = _TOKEN_RE.match(text, pos)
+        if m is None:
+            raise ValueError(f"tokenizer stuck at: {text[pos:pos+40]!r}")
+        pos = m.end()
+        if m.lastgroup == "ws":
+            continue
+        out.append(m.group())
+    return out
+
+
+class _TokenStream:
+    def __init__(self, tokens: List[str]):
+        self.toks = tokens
+        self.i = 0
+
+    def peek(self) -> Optional[str]:
+        return self.toks[self.i] if self.i < len(self.toks) else None
+
+    def next(self) -> str:
+        t = self.toks[self.i]
+        self.i += 1
+        return t
+
+    def expect(self, tok: str):
+        t = self.next()
+        if t != tok:
+            raise ValueError(f"expected {tok!r}, got {t!r}")
+
+    def eof(self) -> bool:
+        return self.i >= len(self.toks)
+
+
+def _read_term(ts: _TokenStream) -> str:
+    """Read one term (possibly a quoted triple) back into its surface string."""
+    t = ts.peek()
+    if t == "<<":
+        ts.next()
+        s = _read_term(ts)
+        p = _read_term(ts)
+        o = _read_term(ts)
+        ts.expect(">>")
+        return f"<< {s} {p} {o} >>"
+    return ts.next()
+
+
+# ------------------------------------------------------------------ N-Triples
+def parse_ntriples_into(db, text: str):
+    """Line-oriented N-Triples-star (ref sparql_database.rs:1345-1463)."""
+    for line in text.split("\n"):
+        line = line.strip()
+        if not line or line.startswith("#"):
+            continue
+        ts = _TokenStream(tokenize(line))
+        s = _read_term(ts)
+        p = _read_term(ts)
+        o = _read_term(ts)
+        if not ts.eof() and ts.peek() == ".":
+            ts.next()
+        db.store.insert_quad(
+            0,
+            db.encode_term_star(s),
+            db.encode_term_star(p),
+            db.encode_term_star(o),
+        )
+
+
+def parse_nquads_into(db, text: str):
+    """N-Quads: optional 4th graph term (ref sparql_database.rs:1411-1463)."""
+    for line in text.split("\n"):
+        line = line.strip()
+        if not line or line.startswith("#"):
+            continue
+        ts = _TokenStream(tokenize(line))
+        s = _read_term(ts)
+        p = _read_term(ts)
+        o = _read_term(ts)
+        g = None
+        if not ts.eof() and ts.peek() != ".":
+            g = _read_term(ts)
+        gid = 0 if g is None else db.dictionary.encode(db.resolve_lexical(g))
+        db.store.insert_quad(
+            gid,
+            db.encode_term_star(s),
+            db.encode_term_star(p),
+            db.encode_term_star(o),
+        )
+
+
+# --------------------------------------------------------------------- Turtle
+def parse_turtle_into(db, text: str):
+    """Turtle-star subset: @prefix/PREFIX, `a`, `;` and `,` groups, quoted
+    triples, `{| p o |}` annotations (ref sparql_database.rs:965-1247)."""
+    ts = _TokenStream(tokenize(text))
+    local_prefixes = dict(db.prefixes)
+
+    def resolve(term: str) -> str:
+        return term
+
+    while not ts.eof():
+        t = ts.peek()
+        if t in ("@prefix", "@PREFIX", "PREFIX", "prefix"):
+            ts.next()
+            name = ts.next()
+            if name.endswith(":"):
+                name = name[:-1]
+            iri = ts.next()
+            if iri.startswith("<") and iri.endswith(">"):
+                iri = iri[1:-1]
+            local_prefixes[name] = iri
+            db.prefixes.setdefault(name, iri)
+            if ts.peek() == ".":
+                ts.next()
+            continue
+        if t in ("@base", "BASE", "base"):
+            ts.next()
+            ts.next()
+            if ts.peek() == ".":
+                ts.next()
+            continue
+        subj = _read_term(ts)
+        while True:
+            pred = _read_term(ts)
+            while True:
+                obj = _read_term(ts)
+                s_id = db.encode_term_star(subj, local_prefixes)
+                p_id = db.encode_term_star(pred, local_prefixes)
+                o_id = db.encode_term_star(obj, local_prefixes)
+                db.store.insert_quad(0, s_id, p_id, o_id)
+                # annotation syntax: <s> <p> <o> {| <ap> <ao> |} — asserts
+                # << s p o >> ap ao (ref tokenize_turtle_star_line:1144)
+                while ts.peek() == "{|":
+                    ts.next()
+                    qt_id = db.quoted_triples.encode(s_id, p_id, o_id)
+                    while ts.peek() not in ("|}", None):
+                        ap = _read_term(ts)
+                        ao = _read_term(ts)
+                        db.store.insert_quad(
+                            0,
+                            qt_id,
+                            db.encode_term_star(ap, local_prefixes),
+                            db.encode_term_star(ao, local_prefixes),
+                        )
+                        if ts.peek() == ";":
+                            ts.next()
+                    ts.expect("|}")
+                if ts.peek() == ",":
+                    ts.next()
+                    continue
+                break
+            if ts.peek() == ";":
+                ts.next()
+                if ts.peek() in (".", ";", None):
+                    continue
+                continue
+            break
+        if ts.peek() == ".":
+            ts.next()
+
+
+# --------------------------------------------------------------------- RDF/XML
+def _expand_qname(tag: str) -> str:
+    """ElementTree gives tags as {namespace}local -> namespace+local."""
+    if tag.startswith("{"):
+        ns, local = tag[1:].split("}", 1)
+        return ns + local
+    return tag
+
+
+def parse_rdf_xml_into(db, xml_text: str):
+    """RDF/XML reader (ref sparql_database.rs:630-963).  Supports
+    rdf:Description/typed nodes, rdf:about/ID/nodeID, property elements with
+    rdf:resource, nested nodes and text content.  Bulk-encodes into columns.
+    """
+    root = ET.fromstring(xml_text)
+    triples: List[Tuple[str, str, str]] = []
+    bnode_counter = [0]
+
+    def fresh_bnode() -> str:
+        bnode_counter[0] += 1
+        return f"_:genid{bnode_counter[0]}"
+
+    def node_id(el) -> str:
+        about = el.get(f"{{{RDF_NS}}}about")
+        if about is not None:
+            return about
+        rid = el.get(f"{{{RDF_NS}}}ID")
+        if rid is not None:
+            return "#" + rid
+        nid = el.get(f"{{{RDF_NS}}}nodeID")
+        if nid is not None:
+            return "_:" + nid
+        return fresh_bnode()
+
+    def walk_node(el) -> str:
+        subj = node_id(el)
+        tag = _expand_qname(el.tag)
+        if tag != RDF_NS + "Description":
+            triples.append((subj, RDF_TYPE, tag))
+        for k, v in el.attrib.items():
+            ek = _expand_qname(k)
+            if ek.startswith(RDF_NS):
+                continue
+            triples.append((subj, ek, v))
+        for prop in el:
+            pred = _expand_qname(prop.tag)
+            res = prop.get(f"{{{RDF_NS}}}resource")
+            nid = prop.get(f"{{{RDF_NS}}}nodeID")
+            children = list(prop)
+            if res is not None:
+                triples.append((subj, pred, res))
+            elif nid is not None:
+                triples.append((subj, pred, "_:" + nid))
+            elif children:
+                for child in children:
+                    triples.append((subj, pred, walk_node(child)))
+            else:
+                text = (prop.text or "").strip()
+                triples.append((subj, pred, text))
+        return subj
+
+    root_tag = _expand_qname(root.tag)
+    nodes = list(root) if root_tag == RDF_NS + "RDF" else [root]
+    for el in nodes:
+        walk_node(el)
+
+    # bulk encode -> columnar insert (GPU-side dedup/sort on commit)
+    import numpy as np
+    n = len(triples)
+    s_ids = np.empty(n, dtype=np.uint32)
+    p_ids = np.empty(n, dtype=np.uint32)
+    o_ids = np.empty(n, dtype=np.uint32)
+    enc = db.dictionary.encode
+    for i, (s, p, o) in enumerate(triples):
+        s_ids[i] = enc(s)
+        p_ids[i] = enc(p)
+        o_ids[i] = enc(o)
+    if n:
+        db.store.insert_bulk(0, s_ids, p_ids, o_ids)

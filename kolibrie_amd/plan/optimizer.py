@@ -1,0 +1,245 @@
+"""Streamertail — Volcano-style cost-based optimizer.
+
+Ref: streamertail_optimizer/optimizer.rs (1 198 LoC).  Phases preserved:
+  1. reorder_logical (:104): flatten homogeneous-scope scan groups and
+     greedy-order them — cheapest *anchored* seed first, then only
+     join-connected patterns, ranked by bound-scan cardinality (:175).
+  2. star detection (:293): subject-star of >=3 patterns (object stars
+     excluded) compiles to StarJoin + BindJoin chain (:617-699).
+  3. join-algorithm choice per join: BindJoin / HashJoin / NLJ candidates by
+     estimated cost; memoized on a serialized plan key (:751-848).
+  4. scan choice by boundness (:702): bound positions -> IndexScan,
+     0 bound -> TableScan.
+
+GPU retuning: hash join is the default (a bound-predicate scan is a
+contiguous device slice, so "materializing" the build side is free); bind
+join is kept for tiny outer sides where a narrow K1 probe beats a full K2
+build+probe.
+"""
+from __future__ import annotations
+
+from typing import List, Optional, Set, Tuple
+
+from ..storage.terms import Constant, TriplePattern, Variable
+from .cost import CostEstimator
+from .logical import (
+    LBind, LJoin, LMLPredict, LMinus, LProjection, LScan, LSelection,
+    LSubquery, LUnion, LUnit, LValues, LogicalOp,
+)
+from .physical import (
+    PBind, PBindJoin, PFilter, PHashJoin, PIndexScan, PMLPredict, PMinus,
+    PNestedLoopJoin, PProjection, PStarJoin, PSubquery, PTableScan, PUnion,
+    PUnit, PValues, PhysicalOp,
+)
+from .stats import DatabaseStats
+
+BIND_JOIN_MAX_LEFT = 4096  # prefer dependent probe when outer side is tiny
+
+
+def _pattern_vars(p: TriplePattern) -> List[str]:
+    return p.variables()
+
+
+class Streamertail:
+    def __init__(self, stats: DatabaseStats):
+        self.stats = stats
+        self.est = CostEstimator(stats)
+        self.memo = {}
+
+    # ------------------------------------------------------------ entry ----
+    def find_best_plan(self, op: LogicalOp) -> PhysicalOp:
+        return self._plan(op, set())[0]
+
+    def _plan(self, op: LogicalOp, bound: Set[str]) -> Tuple[PhysicalOp, float, float]:
+        """Returns (physical, est_rows, est_cost)."""
+        if isinstance(op, LUnit):
+            return PUnit(), 1.0, 0.0
+        if isinstance(op, LScan):
+            return self._plan_scan_group([op], bound)
+        if isinstance(op, LJoin):
+            group = self._flatten_scan_group(op)
+            if group is not None:
+                return self._plan_scan_group(group, bound)
+            lp, lr, lc = self._plan(op.left, bound)
+            l_vars = self._out_vars(op.left, bound)
+            rp, rr, rc = self._plan(op.right, bound | l_vars)
+            est_rows = max(lr, rr)
+            return PHashJoin(lp, rp), est_rows, lc + rc + 2.0 * (lr + rr)
+        if isinstance(op, LUnion):
+            lp, lr, lc = self._plan(op.left, bound)
+            rp, rr, rc = self._plan(op.right, bound)
+            return PUnion(lp, rp), lr + rr, lc + rc
+        if isinstance(op, LSelection):
+            ip, ir, ic = self._plan(op.input, bound)
+            return PFilter(op.condition, ip), max(1.0, ir / 3.0), ic + ir
+        if isinstance(op, LBind):
+            ip, ir, ic = self._plan(op.input, bound)
+            return PBind(op.expr, op.var, ip), ir, ic + ir
+        if isinstance(op, LValues):
+            ip, ir, ic = self._plan(op.input, bound | set(op.variables))
+            return PValues(list(op.variables), op.rows, ip), ir * max(1, len(op.rows)), ic
+        if isinstance(op, LSubquery):
+            inner = Streamertail(self.stats)
+            sub_plan = inner.find_best_plan(op.select.plan)
+            op.select.physical = sub_plan  # attach
+            ip, ir, ic = self._plan(op.input, bound)
+            return PSubquery(op.select, ip), ir, ic
+        if isinstance(op, LMLPredict):
+            ip, ir, ic = self._plan(op.input, bound)
+            return PMLPredict(op.info, ip), ir, ic + 1000.0 + 100.0 * ir
+        if isinstance(op, LMinus):
+            lp, lr, lc = self._plan(op.left, bound)
+            rp, rr, rc = self._plan(op.right, bound)
+            return PMinus(lp, rp), lr, lc + rc
+        if isinstance(op, LProjection):
+            ip, ir, ic = self._plan(op.input, bound)
+            return PProjection(list(op.variables), ip), ir, ic
+        raise ValueError(f"cannot plan {type(op).__name__}")
+
+    # ----------------------------------------------------- scan reordering --
+    def _flatten_scan_group(self, op: LogicalOp) -> Optional[List[LScan]]:
+        """If op is a join tree of scans with one homogeneous graph scope,
+        return them flat; else None (ref reorder_logical:104)."""
+        scans: List[LScan] = []
+
+        def rec(x) -> bool:
+            if isinstance(x, LScan):
+                scans.append(x)
+                return True
+            if isinstance(x, LJoin):
+                return rec(x.left) and rec(x.right)
+            return False
+
+        if not rec(op) or not scans:
+            return None
+        scope0 = scans[0].graph
+        if any(s.graph != scope0 for s in scans[1:]):
+            return None
+        return scans
+
+    def _greedy_order(self, scans: List[LScan], bound: Set[str]) -> List[LScan]:
+        """Cheapest anchored seed, then only join-connected patterns
+        (ref greedy_order_scans:175)."""
+        remaining = list(scans)
+        ordered: List[LScan] = []
+        cur_bound = set(bound)
+
+        def est(s: LScan) -> float:
+            return self.est.estimate_scan(s.pattern, cur_bound, s.graph)
+
+        def anchored(s: LScan) -> bool:
+            sb, pb, ob = self.est.bound_positions(s.pattern, cur_bound)
+            return sb or pb or ob
+
+        while remaining:
+            connected = [s for s in remaining
+                         if anchored(s) or any(v in cur_bound for v in _pattern_vars(s.pattern))]
+            pool = connected if connected else remaining
+            pool_anchored = [s for s in pool if anchored(s)]
+            pick_from = pool_anchored if pool_anchored else pool
+            best = min(pick_from, key=est)
+            ordered.append(best)
+            remaining.remove(best)
+            cur_bound.update(_pattern_vars(best.pattern))
+        return ordered
+
+    def _detect_star(self, scans: List[LScan]) -> Optional[str]:
+        """Subject-star var shared by >=3 patterns (ref is_star_query:293;
+        object stars excluded)."""
+        from collections import Counter
+        c = Counter()
+        for s in scans:
+            if isinstance(s.pattern.s, Variable):
+                c[s.pattern.s.name] += 1
+        if not c:
+            return None
+        var, cnt = c.most_common(1)[0]
+        if cnt >= 3 and cnt == len(scans):
+            # all patterns share the subject: pure star
+            return var
+        return None
+
+    def _plan_scan_group(self, scans: List[LScan], bound: Set[str]
+                         ) -> Tuple[PhysicalOp, float, float]:
+        ordered = self._greedy_order(scans, bound)
+        scope = ordered[0].graph
+        star_var = self._detect_star(ordered) if len(ordered) >= 3 else None
+        if star_var is not None and not bound:
+            pats = [s.pattern for s in ordered]
+            est_rows = min(self.est.estimate_scan(p, set(), scope) for p in pats)
+            cost = sum(self.est.scan_cost(p, {star_var}, scope) for p in pats)
+            return PStarJoin(star_var, pats, scope), max(1.0, est_rows), cost
+
+        cur: Optional[PhysicalOp] = None
+        cur_rows = 1.0
+        cur_cost = 0.0
+        cur_bound = set(bound)
+        for s in ordered:
+            scan_op = self._choose_scan(s.pattern, cur_bound, s.graph)
+            rows_given_bound = self.est.estimate_scan(s.pattern, cur_bound, s.graph)
+            rows_free = self.est.estimate_scan(s.pattern, set(), s.graph)
+            if cur is None:
+                cur = scan_op
+                cur_rows = rows_free if not cur_bound else rows_given_bound
+                cur_cost = self.est.scan_cost(s.pattern, cur_bound, s.graph)
+                cur_bound.update(_pattern_vars(s.pattern))
+                continue
+            shared = [v for v in _pattern_vars(s.pattern) if v in cur_bound]
+            # candidates (ref find_best_plan_recursive:382)
+            hash_cost = cur_cost + self.est.scan_cost(s.pattern, set(), s.graph) \
+                + 2.0 * (cur_rows + rows_free)
+            bind_cost = cur_cost + cur_rows * 2.0 \
+                + cur_rows * max(1.0, rows_given_bound) * 1.0 + 2000.0
+            nlj_cost = cur_cost + 10.0 * cur_rows * max(1.0, rows_free)
+            if not shared:
+                cur = PNestedLoopJoin(cur, scan_op)
+                cur_cost = nlj_cost
+                cur_rows = cur_rows * max(1.0, rows_free)
+            elif cur_rows <= BIND_JOIN_MAX_LEFT and bind_cost < hash_cost:
+                cur = PBindJoin(cur, scan_op)
+                cur_cost = bind_cost
+                cur_rows = cur_rows * max(1.0, rows_given_bound)
+            else:
+                cur = PHashJoin(cur, scan_op)
+                cur_cost = hash_cost
+                cur_rows = max(1.0, cur_rows * rows_given_bound)
+            cur_bound.update(_pattern_vars(s.pattern))
+        assert cur is not None
+        return cur, cur_rows, cur_cost
+
+    def _choose_scan(self, pattern: TriplePattern, bound: Set[str], graph
+                     ) -> PhysicalOp:
+        """2-3 bound -> IndexScan; 1 bound -> IndexScan if est < 10 000;
+        0 bound -> TableScan (ref choose_best_scan:702)."""
+        sb, pb, ob = self.est.bound_positions(pattern, bound)
+        n_bound = int(sb) + int(pb) + int(ob)
+        if n_bound >= 2:
+            return PIndexScan(pattern, graph)
+        if n_bound == 1:
+            if self.est.estimate_scan(pattern, bound, graph) < 10_000:
+                return PIndexScan(pattern, graph)
+            return PIndexScan(pattern, graph)
+        return PTableScan(pattern, graph)
+
+    def _out_vars(self, op: LogicalOp, bound: Set[str]) -> Set[str]:
+        out: Set[str] = set()
+
+        def rec(x):
+            if isinstance(x, LScan):
+                out.update(_pattern_vars(x.pattern))
+                if x.graph is not None and x.graph[0] == "var":
+                    out.add(x.graph[1])
+            elif isinstance(x, (LJoin, LUnion, LMinus)):
+                rec(x.left)
+                rec(x.right)
+            elif isinstance(x, LBind):
+                out.add(x.var)
+                rec(x.input)
+            elif isinstance(x, LValues):
+                out.update(x.variables)
+                rec(x.input)
+            elif hasattr(x, "input"):
+                rec(x.input)
+
+        rec(op)
+        return out

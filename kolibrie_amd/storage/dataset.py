@@ -1,0 +1,392 @@
+"""The quad store: columnar int32 triples with sorted index permutations.
+
+MI355X-native redesign of the reference's DatasetIndex
+(shared/src/dataset_index.rs: nested HashMap gspo/gpos/gosp/spog indexes).
+Instead of pointer-chasing hash maps we keep, per named graph, three sorted
+column permutations:
+
+    SPO  — rows sorted by (s,p,o)   — serves (s??), (sp?), (spo)
+    POS  — rows sorted by (p,o,s)   — serves (?p?), (?po)
+    OSP  — rows sorted by (o,s,p)   — serves (??o), (s?o)
+
+Each order stores a packed int64 key of its two leading columns plus the
+trailing column as int32 — 12 B/row/order.  Pattern lookup is a binary-search
+range (torch.searchsorted on host/CPU tensors; the K1 scan-probe HIP kernel
+does the same per-row on device).  Mutations buffer on the host and commit
+with a sort+dedup pass (K9 class).
+
+All tensors live on `device` ("cpu" or "cuda:N"); on an MI355X the committed
+columns are HBM-resident and queries never leave the device until final
+string decode.
+"""
+from __future__ import annotations
+
+from typing import Dict, Iterable, List, Optional, Sequence, Tuple
+
+import numpy as np
+import torch
+
+from ..engine.tensor_utils import (
+    lexsort,
+    membership_mask,
+    pack2,
+    rows_equal_prev,
+    unpack2,
+)
+from .terms import NULL_ID
+
+DEFAULT_GRAPH = NULL_ID  # graph id 0 == default graph
+
+# order codes
+SPO, POS, OSP = 0, 1, 2
+_ORDER_COLS = {SPO: (0, 1, 2), POS: (1, 2, 0), OSP: (2, 0, 1)}
+
+
+def _as_i32(x) -> torch.Tensor:
+    if isinstance(x, torch.Tensor):
+        return x.to(torch.int32)
+    a = np.asarray(x)
+    if a.dtype == np.uint32:
+        a = a.view(np.int32)
+    return torch.from_numpy(np.ascontiguousarray(a)).to(torch.int32)
+
+
+def _u32_tensor_to_i64(c: torch.Tensor) -> torch.Tensor:
+    return c.to(torch.int64) & 0xFFFFFFFF
+
+
+class GraphIndex:
+    """Immutable sorted index of one graph's triples (or a merged view)."""
+
+    __slots__ = ("device", "n", "orders")
+
+    def __init__(self, device: torch.device, n: int, orders):
+        self.device = device
+        self.n = n
+        # orders[code] = (key12 int64 [n] sorted, z int32 [n])
+        self.orders = orders
+
+    @staticmethod
+    def empty(device) -> "GraphIndex":
+        device = torch.device(device)
+        e64 = torch.empty(0, dtype=torch.int64, device=device)
+        e32 = torch.empty(0, dtype=torch.int32, device=device)
+        return GraphIndex(device, 0, {k: (e64, e32.clone()) for k in (SPO, POS, OSP)})
+
+    @staticmethod
+    def from_columns(s, p, o, device="cpu", dedup: bool = True) -> "GraphIndex":
+        device = torch.device(device)
+        cols = [_as_i32(s).to(device), _as_i32(p).to(device), _as_i32(o).to(device)]
+        n = cols[0].numel()
+        if n == 0:
+            return GraphIndex.empty(device)
+        if dedup:
+            # canonical SPO sort + dedup once, then derive other orders
+            key12 = pack2(cols[0], cols[1])
+            perm = lexsort([key12, cols[2]])
+            key12 = key12[perm]
+            z = cols[2][perm]
+            dup = rows_equal_prev([key12, z])
+            keep = ~dup
+            key12, z = key12[keep], z[keep]
+            s_c, p_c = unpack2(key12)
+            cols = [s_c, p_c, z]
+            n = key12.numel()
+            orders = {SPO: (key12, z)}
+            start = 1
+        else:
+            orders = {}
+            start = 0
+        for code in (SPO, POS, OSP):
+            if code in orders:
+                continue
+            a, b, c = _ORDER_COLS[code]
+            k = pack2(cols[a], cols[b])
+            perm = lexsort([k, cols[c]])
+            orders[code] = (k[perm], cols[c][perm])
+        _ = start
+        return GraphIndex(device, n, orders)
+
+    # -- access -------------------------------------------------------------
+    def columns(self) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+        """All triples as (s, p, o) int32 columns (SPO-sorted)."""
+        key12, z = self.orders[SPO]
+        s, p = unpack2(key12)
+        return s, p, z
+
+    def _range_key1(self, code: int, v: int) -> Tuple[int, int]:
+        key12, _ = self.orders[code]
+        lo_key = pack2(
+            torch.tensor([v], dtype=torch.int32),
+            torch.tensor([0], dtype=torch.int32),
+        ).to(self.device)
+        hi_key = pack2(
+            torch.tensor([v], dtype=torch.int32),
+            torch.tensor([-1], dtype=torch.int32),  # 0xFFFFFFFF
+        ).to(self.device)
+        lo = int(torch.searchsorted(key12, lo_key, side="left").item())
+        hi = int(torch.searchsorted(key12, hi_key, side="right").item())
+        return lo, hi
+
+    def _range_key12(self, code: int, v1: int, v2: int) -> Tuple[int, int]:
+        key12, _ = self.orders[code]
+        k = pack2(
+            torch.tensor([v1], dtype=torch.int32),
+            torch.tensor([v2], dtype=torch.int32),
+        ).to(self.device)
+        lo = int(torch.searchsorted(key12, k, side="left").item())
+        hi = int(torch.searchsorted(key12, k, side="right").item())
+        return lo, hi
+
+    def lookup(
+        self, s: Optional[int], p: Optional[int], o: Optional[int]
+    ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+        """Return matching triples as (s,p,o) int32 columns.
+
+        Bound values are u32 ints; None = wildcard.  8-way dispatch mirrors
+        DatasetIndex::query_graph (dataset_index.rs:223-344).
+        """
+        def i32(v):
+            return v - 0x1_0000_0000 if v >= 0x8000_0000 else v
+
+        if s is None and p is None and o is None:
+            return self.columns()
+        if s is not None and p is not None:
+            lo, hi = self._range_key12(SPO, i32(s), i32(p))
+            key12, z = self.orders[SPO]
+            zz = z[lo:hi]
+            if o is not None:
+                zlo = int(torch.searchsorted(zz, torch.tensor(i32(o), dtype=torch.int32, device=self.device), side="left").item())
+                zhi = int(torch.searchsorted(zz, torch.tensor(i32(o), dtype=torch.int32, device=self.device), side="right").item())
+                zz = zz[zlo:zhi]
+            n = zz.numel()
+            return (
+                torch.full((n,), i32(s), dtype=torch.int32, device=self.device),
+                torch.full((n,), i32(p), dtype=torch.int32, device=self.device),
+                zz,
+            )
+        if s is not None and o is not None:  # (s ? o) -> OSP with (o,s) bound
+            lo, hi = self._range_key12(OSP, i32(o), i32(s))
+            _, z = self.orders[OSP]
+            zz = z[lo:hi]
+            n = zz.numel()
+            return (
+                torch.full((n,), i32(s), dtype=torch.int32, device=self.device),
+                zz,
+                torch.full((n,), i32(o), dtype=torch.int32, device=self.device),
+            )
+        if s is not None:  # (s ? ?)
+            lo, hi = self._range_key1(SPO, i32(s))
+            key12, z = self.orders[SPO]
+            _, p_c = unpack2(key12[lo:hi])
+            n = p_c.numel()
+            return (
+                torch.full((n,), i32(s), dtype=torch.int32, device=self.device),
+                p_c,
+                z[lo:hi],
+            )
+        if p is not None and o is not None:  # (? p o) -> POS
+            lo, hi = self._range_key12(POS, i32(p), i32(o))
+            _, z = self.orders[POS]
+            zz = z[lo:hi]
+            n = zz.numel()
+            return (
+                zz,
+                torch.full((n,), i32(p), dtype=torch.int32, device=self.device),
+                torch.full((n,), i32(o), dtype=torch.int32, device=self.device),
+            )
+        if p is not None:  # (? p ?)
+            lo, hi = self._range_key1(POS, i32(p))
+            key12, z = self.orders[POS]
+            _, o_c = unpack2(key12[lo:hi])
+            n = o_c.numel()
+            return (
+                z[lo:hi],
+                torch.full((n,), i32(p), dtype=torch.int32, device=self.device),
+                o_c,
+            )
+        # (? ? o) -> OSP
+        lo, hi = self._range_key1(OSP, i32(o))
+        key12, z = self.orders[OSP]
+        _, s_c = unpack2(key12[lo:hi])
+        n = s_c.numel()
+        return (
+            s_c,
+            z[lo:hi],
+            torch.full((n,), i32(o), dtype=torch.int32, device=self.device),
+        )
+
+    def contains(self, s: int, p: int, o: int) -> bool:
+        ss, _, _ = self.lookup(s, p, o)
+        return ss.numel() > 0
+
+    @staticmethod
+    def merge(indexes: Sequence["GraphIndex"], device) -> "GraphIndex":
+        """Union with dedup (FROM-merge semantics, dataset_index.rs:207)."""
+        cols = [[], [], []]
+        for gi in indexes:
+            s, p, o = gi.columns()
+            cols[0].append(s)
+            cols[1].append(p)
+            cols[2].append(o)
+        if not cols[0]:
+            return GraphIndex.empty(device)
+        return GraphIndex.from_columns(
+            torch.cat(cols[0]), torch.cat(cols[1]), torch.cat(cols[2]), device=device
+        )
+
+
+class _GraphBuffer:
+    """Mutable per-graph state: committed index + host-side pending edits."""
+
+    __slots__ = ("index", "pend_add", "pend_del")
+
+    def __init__(self, device):
+        self.index = GraphIndex.empty(device)
+        self.pend_add: List[Tuple[int, int, int]] = []
+        self.pend_del: List[Tuple[int, int, int]] = []
+
+
+class QuadStore:
+    """Named-graph quad store (ref: shared/src/dataset_index.rs DatasetIndex).
+
+    Graph id 0 is the default graph.  The named-graph *catalog* tracks graphs
+    that exist even when empty (CREATE GRAPH), mirroring the reference's
+    named_graphs set (dataset_index.rs:69-72).
+    """
+
+    def __init__(self, device="cpu"):
+        self.device = torch.device(device)
+        self.graphs: Dict[int, _GraphBuffer] = {DEFAULT_GRAPH: _GraphBuffer(self.device)}
+        self.catalog: set = set()  # named graphs explicitly created or written
+        self.version = 0
+        self._merged_cache: Dict[Tuple[int, ...], Tuple[int, GraphIndex]] = {}
+
+    # -- mutation ------------------------------------------------------------
+    def _buf(self, g: int) -> _GraphBuffer:
+        b = self.graphs.get(g)
+        if b is None:
+            b = _GraphBuffer(self.device)
+            self.graphs[g] = b
+        return b
+
+    def insert_quad(self, g: int, s: int, p: int, o: int):
+        b = self._buf(g)
+        b.pend_add.append((s, p, o))
+        self.version += 1
+
+    def delete_quad(self, g: int, s: int, p: int, o: int):
+        if g not in self.graphs:
+            return
+        self.graphs[g].pend_del.append((s, p, o))
+        self.version += 1
+
+    def insert_bulk(self, g: int, s, p, o):
+        """Bulk columnar insert (parser fast path).  s/p/o: arrays of u32."""
+        b = self._buf(g)
+        self._commit(g)
+        new_idx = GraphIndex.from_columns(s, p, o, device=self.device)
+        b.index = GraphIndex.merge([b.index, new_idx], self.device)
+        self.version += 1
+
+    def create_graph(self, g: int):
+        self._buf(g)
+        self.catalog.add(g)
+        self.version += 1
+
+    def drop_graph(self, g: int) -> bool:
+        existed = g in self.graphs and (
+            g in self.catalog or self.graphs[g].index.n > 0 or self.graphs[g].pend_add
+        )
+        self.graphs.pop(g, None)
+        self.catalog.discard(g)
+        self.version += 1
+        return existed
+
+    def clear_graph(self, g: int):
+        if g in self.graphs:
+            b = self.graphs[g]
+            b.index = GraphIndex.empty(self.device)
+            b.pend_add.clear()
+            b.pend_del.clear()
+            self.version += 1
+
+    # -- commit --------------------------------------------------------------
+    def _commit(self, g: int):
+        b = self.graphs.get(g)
+        if b is None or (not b.pend_add and not b.pend_del):
+            return
+        s0, p0, o0 = b.index.columns()
+        if b.pend_add:
+            arr = np.asarray(b.pend_add, dtype=np.uint32).reshape(-1, 3).view(np.int32)
+            t = torch.from_numpy(arr.copy()).to(self.device)
+            s0 = torch.cat([s0, t[:, 0]])
+            p0 = torch.cat([p0, t[:, 1]])
+            o0 = torch.cat([o0, t[:, 2]])
+        idx = GraphIndex.from_columns(s0, p0, o0, device=self.device)
+        if b.pend_del:
+            arr = np.asarray(b.pend_del, dtype=np.uint32).reshape(-1, 3).view(np.int32)
+            t = torch.from_numpy(arr.copy()).to(self.device)
+            s, p, o = idx.columns()
+            from ..engine.tensor_utils import unique_rows
+            dset = unique_rows([t[:, 0], t[:, 1], t[:, 2]])
+            mask = ~membership_mask([s, p, o], dset)
+            idx = GraphIndex.from_columns(s[mask], p[mask], o[mask], device=self.device, dedup=False)
+        b.index = idx
+        b.pend_add.clear()
+        b.pend_del.clear()
+
+    def commit_all(self):
+        for g in list(self.graphs):
+            self._commit(g)
+
+    # -- query ---------------------------------------------------------------
+    def graph_index(self, g: int) -> GraphIndex:
+        self._commit(g)
+        b = self.graphs.get(g)
+        return b.index if b is not None else GraphIndex.empty(self.device)
+
+    def merged_index(self, graph_ids: Sequence[int]) -> GraphIndex:
+        """Dedup'd union view over several graphs (FROM merge)."""
+        key = tuple(sorted(set(graph_ids)))
+        cached = self._merged_cache.get(key)
+        if cached is not None and cached[0] == self.version:
+            return cached[1]
+        idx = GraphIndex.merge([self.graph_index(g) for g in key], self.device)
+        self._merged_cache[key] = (self.version, idx)
+        return idx
+
+    def named_graph_ids(self) -> List[int]:
+        self.commit_all()
+        out = set(self.catalog)
+        for g, b in self.graphs.items():
+            if g != DEFAULT_GRAPH and b.index.n > 0:
+                out.add(g)
+        return sorted(out)
+
+    def query_graph(
+        self, g: int, s: Optional[int], p: Optional[int], o: Optional[int]
+    ):
+        return self.graph_index(g).lookup(s, p, o)
+
+    def contains(self, g: int, s: int, p: int, o: int) -> bool:
+        return self.graph_index(g).contains(s, p, o)
+
+    def all_quads(self) -> List[Tuple[int, int, int, int]]:
+        """(g,s,p,o) u32 tuples across all graphs — host-side debugging /
+        serialization path (ref dataset_index.rs:485)."""
+        out = []
+        self.commit_all()
+        for g in sorted(self.graphs):
+            idx = self.graphs[g].index
+            s, p, o = idx.columns()
+            su = (s.to(torch.int64) & 0xFFFFFFFF).cpu().numpy()
+            pu = (p.to(torch.int64) & 0xFFFFFFFF).cpu().numpy()
+            ou = (o.to(torch.int64) & 0xFFFFFFFF).cpu().numpy()
+            for i in range(len(su)):
+                out.append((g, int(su[i]), int(pu[i]), int(ou[i])))
+        return out
+
+    def triple_count(self) -> int:
+        self.commit_all()
+        return sum(b.index.n for b in self.graphs.values())

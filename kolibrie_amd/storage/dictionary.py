@@ -1,0 +1,157 @@
+"""Host-side bidirectional string <-> u32 dictionary.
+
+Strings never touch the GPU: kernels see only int32 IDs.  Alongside the
+string table we maintain a *numeric value column* (float64 per ID, parsed at
+encode time, non-numeric => 0.0) that is uploaded to the device so FILTER
+ordering comparisons and aggregates never decode strings (the reference
+decodes and `parse::<f64>().unwrap_or(0.0)` per row —
+streamertail_optimizer/execution/types.rs:349-359; we pre-parse once).
+
+RDF-star quoted triples are interned with bit 31 set
+(ref: shared/src/quoted_triple_store.rs:17-80) and recurse on encode/decode
+(ref: shared/src/dictionary.rs:62 decode_term).
+
+Parity surface: shared/src/dictionary.rs:17-91 (encode/decode/merge).
+"""
+from __future__ import annotations
+
+import math
+from typing import Dict, List, Optional, Tuple
+
+from .terms import QUOTED_TRIPLE_ID_BIT, UNBOUND_U32, is_quoted_id
+
+
+def _try_parse_float(s: str) -> float:
+    try:
+        v = float(s)
+        if math.isnan(v) or math.isinf(v):
+            return 0.0
+        return v
+    except (ValueError, TypeError):
+        return 0.0
+
+
+class Dictionary:
+    """Bidirectional string<->id map.  ID 0 is reserved for "" (and doubles
+    as the default-graph ID)."""
+
+    __slots__ = ("str_to_id", "id_to_str", "values", "_values_dirty")
+
+    def __init__(self):
+        self.str_to_id: Dict[str, int] = {"": 0}
+        self.id_to_str: List[str] = [""]
+        # values[i] = float64 numeric interpretation of id_to_str[i] (0.0 if
+        # non-numeric) — the device value column source.
+        self.values: List[float] = [0.0]
+        self._values_dirty = True
+
+    def __len__(self):
+        return len(self.id_to_str)
+
+    def encode(self, s: str) -> int:
+        """Intern a string, returning its u32 ID (ref dictionary.rs:32)."""
+        i = self.str_to_id.get(s)
+        if i is not None:
+            return i
+        i = len(self.id_to_str)
+        if i >= QUOTED_TRIPLE_ID_BIT:
+            raise OverflowError("dictionary ID space exhausted (2^31 terms)")
+        self.str_to_id[s] = i
+        self.id_to_str.append(s)
+        self.values.append(_try_parse_float(s))
+        self._values_dirty = True
+        return i
+
+    def lookup(self, s: str) -> Optional[int]:
+        return self.str_to_id.get(s)
+
+    def decode(self, i: int) -> Optional[str]:
+        """ID -> string (plain terms only; ref dictionary.rs:49)."""
+        i &= 0xFFFFFFFF
+        if i < len(self.id_to_str):
+            return self.id_to_str[i]
+        return None
+
+    def contains(self, s: str) -> bool:
+        return s in self.str_to_id
+
+    def numeric_value(self, i: int) -> float:
+        i &= 0xFFFFFFFF
+        if i < len(self.values):
+            return self.values[i]
+        return 0.0
+
+    def merge(self, other: "Dictionary") -> Dict[int, int]:
+        """Merge `other` into self, returning an old-id -> new-id remap
+        (ref dictionary.rs:82 — used by parallel parse shards)."""
+        remap: Dict[int, int] = {}
+        for old_id, s in enumerate(other.id_to_str):
+            remap[old_id] = self.encode(s)
+        return remap
+
+
+class QuotedTripleStore:
+    """Interns RDF-star quoted triples as IDs with bit 31 set.
+
+    Ref: shared/src/quoted_triple_store.rs:17-80.  A quoted triple is a
+    (s,p,o) of u32 IDs (any of which may itself be quoted).  IDs are
+    QUOTED_TRIPLE_ID_BIT | index; the allocator stays clear of UNBOUND_U32.
+    """
+
+    __slots__ = ("triple_to_id", "id_to_triple")
+
+    def __init__(self):
+        self.triple_to_id: Dict[Tuple[int, int, int], int] = {}
+        self.id_to_triple: List[Tuple[int, int, int]] = []
+
+    def __len__(self):
+        return len(self.id_to_triple)
+
+    def encode(self, s: int, p: int, o: int) -> int:
+        key = (s & 0xFFFFFFFF, p & 0xFFFFFFFF, o & 0xFFFFFFFF)
+        i = self.triple_to_id.get(key)
+        if i is not None:
+            return i
+        idx = len(self.id_to_triple)
+        qid = QUOTED_TRIPLE_ID_BIT | idx
+        if qid >= UNBOUND_U32:
+            raise OverflowError("quoted-triple ID space exhausted")
+        self.triple_to_id[key] = qid
+        self.id_to_triple.append(key)
+        return qid
+
+    def decode(self, qid: int) -> Optional[Tuple[int, int, int]]:
+        qid &= 0xFFFFFFFF
+        if not (qid & QUOTED_TRIPLE_ID_BIT):
+            return None
+        idx = qid & ~QUOTED_TRIPLE_ID_BIT
+        if idx < len(self.id_to_triple):
+            return self.id_to_triple[idx]
+        return None
+
+    def lookup(self, s: int, p: int, o: int) -> Optional[int]:
+        return self.triple_to_id.get((s & 0xFFFFFFFF, p & 0xFFFFFFFF, o & 0xFFFFFFFF))
+
+    def merge(self, other: "QuotedTripleStore", id_remap: Dict[int, int]) -> Dict[int, int]:
+        """Merge quoted triples from `other` given a plain-ID remap; returns
+        quoted-id remap (ref quoted_triple_store.rs:73)."""
+        qremap: Dict[int, int] = {}
+
+        def remap_one(old_qid: int) -> int:
+            if old_qid in qremap:
+                return qremap[old_qid]
+            t = other.decode(old_qid)
+            assert t is not None
+            parts = []
+            for x in t:
+                if is_quoted_id(x):
+                    parts.append(remap_one(x))
+                else:
+                    parts.append(id_remap.get(x, x))
+            new_qid = self.encode(*parts)
+            qremap[old_qid] = new_qid
+            return new_qid
+
+        for idx in range(len(other.id_to_triple)):
+            remap_one(QUOTED_TRIPLE_ID_BIT | idx)
+        return qremap

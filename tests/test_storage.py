@@ -1,0 +1,145 @@
+"""Storage layer: dictionary, quoted triples, quad store, serializers.
+
+Mirrors reference unit tests in shared/src/dataset_index.rs:593-825 and
+dictionary/quoted-store behavior.
+"""
+import pytest
+import torch
+
+from kolibrie_amd.storage.dictionary import Dictionary, QuotedTripleStore
+from kolibrie_amd.storage.dataset import DEFAULT_GRAPH, GraphIndex, QuadStore
+from kolibrie_amd import SparqlDatabase
+
+
+def test_dictionary_roundtrip():
+    d = Dictionary()
+    a = d.encode("http://example.org/a")
+    b = d.encode("hello world")
+    assert d.encode("http://example.org/a") == a
+    assert d.decode(a) == "http://example.org/a"
+    assert d.decode(b) == "hello world"
+    assert d.lookup("missing") is None
+
+
+def test_dictionary_numeric_values():
+    d = Dictionary()
+    x = d.encode("42.5")
+    y = d.encode("not a number")
+    assert d.numeric_value(x) == 42.5
+    assert d.numeric_value(y) == 0.0
+
+
+def test_dictionary_merge():
+    d1 = Dictionary()
+    d2 = Dictionary()
+    a2 = d2.encode("alpha")
+    b2 = d2.encode("beta")
+    d1.encode("beta")
+    remap = d1.merge(d2)
+    assert d1.decode(remap[a2]) == "alpha"
+    assert d1.decode(remap[b2]) == "beta"
+
+
+def test_quoted_triple_store():
+    q = QuotedTripleStore()
+    qid = q.encode(1, 2, 3)
+    assert qid & 0x8000_0000
+    assert q.encode(1, 2, 3) == qid
+    assert q.decode(qid) == (1, 2, 3)
+    nested = q.encode(qid, 5, 6)
+    assert q.decode(nested) == (qid, 5, 6)
+
+
+def test_graph_index_lookup():
+    s = [1, 1, 2, 3, 3, 3]
+    p = [10, 11, 10, 10, 11, 11]
+    o = [100, 101, 100, 102, 103, 104]
+    gi = GraphIndex.from_columns(s, p, o, device="cpu")
+    assert gi.n == 6
+    ss, pp, oo = gi.lookup(1, None, None)
+    assert ss.numel() == 2
+    ss, pp, oo = gi.lookup(None, 11, None)
+    assert sorted(oo.tolist()) == [101, 103, 104]
+    ss, pp, oo = gi.lookup(None, None, 100)
+    assert sorted(ss.tolist()) == [1, 2]
+    ss, pp, oo = gi.lookup(3, 11, None)
+    assert sorted(oo.tolist()) == [103, 104]
+    assert gi.contains(3, 11, 104)
+    assert not gi.contains(3, 11, 999)
+
+
+def test_graph_index_dedup():
+    gi = GraphIndex.from_columns([1, 1, 1], [2, 2, 2], [3, 3, 4], device="cpu")
+    assert gi.n == 2
+
+
+def test_quad_store_graphs():
+    qs = QuadStore()
+    qs.insert_quad(DEFAULT_GRAPH, 1, 2, 3)
+    qs.insert_quad(7, 1, 2, 4)
+    qs.insert_quad(7, 1, 2, 4)  # duplicate
+    assert qs.triple_count() == 2
+    assert qs.named_graph_ids() == [7]
+    s, p, o = qs.query_graph(7, 1, None, None)
+    assert o.tolist() == [4]
+    qs.delete_quad(7, 1, 2, 4)
+    assert qs.triple_count() == 1
+    assert qs.named_graph_ids() == []
+
+
+def test_quad_store_merged_dedup():
+    qs = QuadStore()
+    qs.insert_quad(1, 5, 6, 7)
+    qs.insert_quad(2, 5, 6, 7)  # same triple, different graph
+    qs.insert_quad(2, 5, 6, 8)
+    merged = qs.merged_index([1, 2])
+    assert merged.n == 2  # FROM-merge dedups
+
+
+def test_graph_management():
+    qs = QuadStore()
+    qs.create_graph(9)
+    assert 9 in qs.named_graph_ids()
+    qs.insert_quad(9, 1, 1, 1)
+    qs.clear_graph(9)
+    assert qs.triple_count() == 0
+    assert 9 in qs.named_graph_ids()  # catalog survives CLEAR
+    assert qs.drop_graph(9)
+    assert 9 not in qs.named_graph_ids()
+
+
+def test_db_add_and_decode():
+    db = SparqlDatabase()
+    db.add_triple("<http://ex.org/alice>", "<http://ex.org/knows>", "<http://ex.org/bob>")
+    db.add_triple("<http://ex.org/alice>", "<http://ex.org/name>", '"Alice"')
+    trips = db.triples_as_strings()
+    assert ("http://ex.org/alice", "http://ex.org/name", "Alice") in trips
+
+
+def test_db_rdf_star_encode():
+    db = SparqlDatabase()
+    db.add_triple("<< <http://e/s> <http://e/p> <http://e/o> >>",
+                  "<http://e/certainty>", '"0.9"')
+    trips = db.triples_as_strings()
+    assert len(trips) == 1
+    assert trips[0][0].startswith("<<")
+    assert "certainty" in trips[0][1]
+
+
+def test_nquads_roundtrip():
+    db = SparqlDatabase()
+    db.add_quad_parts("<http://e/s>", "<http://e/p>", '"val"', "<http://e/g>")
+    db.add_triple("<http://e/a>", "<http://e/b>", "<http://e/c>")
+    text = db.generate_nquads()
+    db2 = SparqlDatabase()
+    db2.parse_nquads(text)
+    assert db2.generate_nquads() == text
+
+
+def test_union_reencoding():
+    db1 = SparqlDatabase()
+    db1.add_triple("<http://e/a>", "<http://e/p>", '"1"')
+    db2 = SparqlDatabase()
+    db2.add_triple("<http://e/b>", "<http://e/p>", '"2"')
+    u = db1.union(db2)
+    assert u.triple_count() == 2
