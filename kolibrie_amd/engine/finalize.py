@@ -43,16 +43,17 @@ def finalize_select_bindings(select: SelectQuery, rows: Bindings, db
 
     if has_agg or select.group_by:
         rows = _aggregate(select, rows, db)
+    # ORDER BY first — it may sort on non-projected variables
+    # (numeric-aware, ref execute_query.rs:477)
+    if select.order_by and rows.n > 1:
+        perm = _order_perm(select, rows, db)
+        rows = rows.gather(perm)
     # projection
     if select.select_star or not select.variables:
         proj_names = rows.variables
     else:
         proj_names = [p.output_name() for p in select.variables]
         rows = rows.project(proj_names)
-    # ORDER BY (numeric-aware, ref execute_query.rs:477)
-    if select.order_by and rows.n > 1:
-        perm = _order_perm(select, rows, db)
-        rows = rows.gather(perm)
     if select.distinct and rows.n > 1:
         cols = [rows.col(v) for v in proj_names]
         if select.order_by:

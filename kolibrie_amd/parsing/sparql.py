@@ -542,8 +542,7 @@ class Parser:
             return ELit(t.text, is_number=True)
         if t.kind == "string":
             self.next()
-            from ..storage.database import literal_lexical_value
-            return ELit(literal_lexical_value(t.text))
+            return ELit(t.text)  # raw surface; resolved at compile time
         if t.kind == "iri":
             self.next()
             return ELit(t.text)

@@ -1,0 +1,283 @@
+"""End-to-end SELECT/UPDATE queries (mirrors kolibrie/tests/
+sparql_unified_test.rs, sparql_graph_test.rs, integration_test.rs)."""
+import pytest
+
+from kolibrie_amd import SparqlDatabase
+
+EX = "http://example.org/"
+
+
+def _people_db():
+    db = SparqlDatabase()
+    data = [
+        ("alice", "name", '"Alice"'),
+        ("alice", "age", '"30"'),
+        ("alice", "worksFor", "<http://example.org/acme>"),
+        ("bob", "name", '"Bob"'),
+        ("bob", "age", '"45"'),
+        ("bob", "worksFor", "<http://example.org/acme>"),
+        ("carol", "name", '"Carol"'),
+        ("carol", "age", '"22"'),
+        ("carol", "worksFor", "<http://example.org/initech>"),
+    ]
+    for s, p, o in data:
+        db.add_triple(f"<{EX}{s}>", f"<{EX}{p}>", o)
+    return db
+
+
+def test_single_pattern(db):
+    db.add_triple(f"<{EX}a>", f"<{EX}p>", '"v1"')
+    db.add_triple(f"<{EX}b>", f"<{EX}p>", '"v2"')
+    rows = db.query(f"SELECT ?s ?o WHERE {{ ?s <{EX}p> ?o }}")
+    assert sorted(rows) == [[f"{EX}a", "v1"], [f"{EX}b", "v2"]]
+
+
+def test_two_pattern_join():
+    db = _people_db()
+    rows = db.query(f"""
+        SELECT ?n ?a WHERE {{
+            ?x <{EX}name> ?n .
+            ?x <{EX}age> ?a .
+        }}""")
+    assert sorted(rows) == [["Alice", "30"], ["Bob", "45"], ["Carol", "22"]]
+
+
+def test_three_pattern_star_join():
+    db = _people_db()
+    rows = db.query(f"""
+        SELECT ?n ?a ?w WHERE {{
+            ?x <{EX}name> ?n .
+            ?x <{EX}age> ?a .
+            ?x <{EX}worksFor> ?w .
+        }}""")
+    assert len(rows) == 3
+    assert ["Alice", "30", f"{EX}acme"] in rows
+
+
+def test_filter_numeric():
+    db = _people_db()
+    rows = db.query(f"""
+        SELECT ?n WHERE {{
+            ?x <{EX}name> ?n . ?x <{EX}age> ?a .
+            FILTER(?a > 25 && ?a <= 45)
+        }}""")
+    assert sorted(r[0] for r in rows) == ["Alice", "Bob"]
+
+
+def test_filter_string_equality():
+    db = _people_db()
+    rows = db.query(f"""
+        SELECT ?x WHERE {{ ?x <{EX}name> ?n . FILTER(?n = "Alice") }}""")
+    assert rows == [[f"{EX}alice"]]
+
+
+def test_filter_var_var_id_equality():
+    db = SparqlDatabase()
+    db.add_triple(f"<{EX}x>", f"<{EX}a>", '"same"')
+    db.add_triple(f"<{EX}x>", f"<{EX}b>", '"same"')
+    db.add_triple(f"<{EX}y>", f"<{EX}a>", '"one"')
+    db.add_triple(f"<{EX}y>", f"<{EX}b>", '"two"')
+    rows = db.query(f"""
+        SELECT ?s WHERE {{ ?s <{EX}a> ?v1 . ?s <{EX}b> ?v2 . FILTER(?v1 = ?v2) }}""")
+    assert rows == [[f"{EX}x"]]
+
+
+def test_union_multiplicity():
+    # UNION preserves duplicates (multiset semantics)
+    db = SparqlDatabase()
+    db.add_triple(f"<{EX}a>", f"<{EX}p>", '"v"')
+    rows = db.query(f"""
+        SELECT ?s WHERE {{
+            {{ ?s <{EX}p> ?o }} UNION {{ ?s <{EX}p> ?o }}
+        }}""")
+    assert len(rows) == 2
+
+
+def test_bind_concat():
+    db = _people_db()
+    rows = db.query(f"""
+        SELECT ?g WHERE {{
+            ?x <{EX}name> ?n . BIND(CONCAT("Hello ", ?n) AS ?g)
+        }}""")
+    assert ["Hello Alice"] in rows
+
+
+def test_values_inline():
+    db = _people_db()
+    rows = db.query(f"""
+        SELECT ?n WHERE {{
+            VALUES ?x {{ <{EX}alice> <{EX}bob> }}
+            ?x <{EX}name> ?n .
+        }}""")
+    assert sorted(r[0] for r in rows) == ["Alice", "Bob"]
+
+
+def test_values_undef_wildcard():
+    db = SparqlDatabase()
+    db.add_triple(f"<{EX}a>", f"<{EX}p>", '"1"')
+    db.add_triple(f"<{EX}b>", f"<{EX}p>", '"2"')
+    rows = db.query(f"""
+        SELECT ?x ?o WHERE {{
+            VALUES (?x ?o) {{ (<{EX}a> UNDEF) }}
+            ?x <{EX}p> ?o .
+        }}""")
+    assert rows == [[f"{EX}a", "1"]]
+
+
+def test_distinct_order_limit():
+    db = _people_db()
+    rows = db.query(f"""
+        SELECT DISTINCT ?w WHERE {{ ?x <{EX}worksFor> ?w }} ORDER BY ?w""")
+    assert rows == [[f"{EX}acme"], [f"{EX}initech"]]
+    rows = db.query(f"""
+        SELECT ?a WHERE {{ ?x <{EX}age> ?a }} ORDER BY DESC(?a) LIMIT 2""")
+    assert rows == [["45"], ["30"]]
+
+
+def test_aggregates_group_by():
+    db = _people_db()
+    rows = db.query(f"""
+        SELECT ?w (COUNT(?x) AS ?c) WHERE {{
+            ?x <{EX}worksFor> ?w .
+        }} GROUP BY ?w ORDER BY ?w""")
+    assert rows == [[f"{EX}acme", "2"], [f"{EX}initech", "1"]]
+
+
+def test_aggregate_sum_avg():
+    db = _people_db()
+    rows = db.query(f"""
+        SELECT (SUM(?a) AS ?s) (AVG(?a) AS ?m) (MIN(?a) AS ?lo) (MAX(?a) AS ?hi)
+        WHERE {{ ?x <{EX}age> ?a }}""")
+    assert rows == [["97", repr(97 / 3), "22", "45"]]
+
+
+def test_subquery_with_limit():
+    db = _people_db()
+    rows = db.query(f"""
+        SELECT ?n WHERE {{
+            {{ SELECT ?x WHERE {{ ?x <{EX}age> ?a }} ORDER BY DESC(?a) LIMIT 1 }}
+            ?x <{EX}name> ?n .
+        }}""")
+    assert rows == [["Bob"]]
+
+
+def test_named_graph_query():
+    db = SparqlDatabase()
+    db.add_quad_parts(f"<{EX}a>", f"<{EX}p>", '"in-g1"', f"<{EX}g1>")
+    db.add_quad_parts(f"<{EX}b>", f"<{EX}p>", '"in-g2"', f"<{EX}g2>")
+    db.add_triple(f"<{EX}c>", f"<{EX}p>", '"default"')
+    rows = db.query(f"SELECT ?o WHERE {{ GRAPH <{EX}g1> {{ ?s <{EX}p> ?o }} }}")
+    assert rows == [["in-g1"]]
+    # GRAPH ?g never matches the default graph
+    rows = db.query(f"SELECT ?g ?o WHERE {{ GRAPH ?g {{ ?s <{EX}p> ?o }} }}")
+    assert sorted(rows) == [[f"{EX}g1", "in-g1"], [f"{EX}g2", "in-g2"]]
+
+
+def test_graph_var_join_consistency():
+    db = SparqlDatabase()
+    db.add_quad_parts(f"<{EX}a>", f"<{EX}p>", f"<{EX}b>", f"<{EX}g1>")
+    db.add_quad_parts(f"<{EX}b>", f"<{EX}q>", '"x"', f"<{EX}g1>")
+    db.add_quad_parts(f"<{EX}b>", f"<{EX}q>", '"y"', f"<{EX}g2>")
+    rows = db.query(f"""
+        SELECT ?o WHERE {{
+            GRAPH ?g {{ ?a <{EX}p> ?b . ?b <{EX}q> ?o }}
+        }}""")
+    assert rows == [["x"]]
+
+
+def test_from_merged_dedup():
+    db = SparqlDatabase()
+    db.add_quad_parts(f"<{EX}a>", f"<{EX}p>", '"v"', f"<{EX}g1>")
+    db.add_quad_parts(f"<{EX}a>", f"<{EX}p>", '"v"', f"<{EX}g2>")
+    rows = db.query(f"""
+        SELECT ?s FROM <{EX}g1> FROM <{EX}g2> WHERE {{ ?s <{EX}p> "v" }}""")
+    assert len(rows) == 1  # merged-FROM default dedups
+
+
+def test_insert_data_and_delete():
+    db = SparqlDatabase()
+    db.query(f'INSERT DATA {{ <{EX}a> <{EX}p> "v1" . <{EX}b> <{EX}p> "v2" }}')
+    assert db.triple_count() == 2
+    db.query(f'DELETE DATA {{ <{EX}a> <{EX}p> "v1" }}')
+    assert db.triple_count() == 1
+
+
+def test_modify_update():
+    db = _people_db()
+    db.query(f"""
+        DELETE {{ ?x <{EX}worksFor> <{EX}acme> }}
+        INSERT {{ ?x <{EX}worksFor> <{EX}megacorp> }}
+        WHERE {{ ?x <{EX}worksFor> <{EX}acme> }}""")
+    rows = db.query(f"SELECT ?x WHERE {{ ?x <{EX}worksFor> <{EX}megacorp> }}")
+    assert len(rows) == 2
+    rows = db.query(f"SELECT ?x WHERE {{ ?x <{EX}worksFor> <{EX}acme> }}")
+    assert rows == []
+
+
+def test_update_in_named_graph():
+    db = SparqlDatabase()
+    db.query(f'INSERT DATA {{ GRAPH <{EX}g> {{ <{EX}a> <{EX}p> "v" }} }}')
+    rows = db.query(f"SELECT ?o WHERE {{ GRAPH <{EX}g> {{ ?s ?p ?o }} }}")
+    assert rows == [["v"]]
+    db.query(f"CLEAR GRAPH <{EX}g>")
+    rows = db.query(f"SELECT ?o WHERE {{ GRAPH <{EX}g> {{ ?s ?p ?o }} }}")
+    assert rows == []
+
+
+def test_rdf_star_query():
+    db = SparqlDatabase()
+    db.add_triple(f"<< <{EX}s> <{EX}p> <{EX}o> >>", f"<{EX}certainty>", '"0.9"')
+    rows = db.query(f"""
+        SELECT ?c WHERE {{ << <{EX}s> <{EX}p> <{EX}o> >> <{EX}certainty> ?c }}""")
+    assert rows == [["0.9"]]
+
+
+def test_rdf_star_variable_inside_quoted():
+    db = SparqlDatabase()
+    db.add_triple(f"<< <{EX}s1> <{EX}p> <{EX}o1> >>", f"<{EX}cert>", '"0.9"')
+    db.add_triple(f"<< <{EX}s2> <{EX}p> <{EX}o2> >>", f"<{EX}cert>", '"0.5"')
+    rows = db.query(f"""
+        SELECT ?s ?c WHERE {{ << ?s <{EX}p> ?o >> <{EX}cert> ?c }}""")
+    assert sorted(rows) == [[f"{EX}s1", "0.9"], [f"{EX}s2", "0.5"]]
+
+
+def test_istriple_builtin():
+    db = SparqlDatabase()
+    db.add_triple(f"<< <{EX}s> <{EX}p> <{EX}o> >>", f"<{EX}cert>", '"0.9"')
+    db.add_triple(f"<{EX}plain>", f"<{EX}cert>", '"1.0"')
+    rows = db.query(f"""
+        SELECT ?c WHERE {{ ?t <{EX}cert> ?c . FILTER(isTRIPLE(?t)) }}""")
+    assert rows == [["0.9"]]
+
+
+def test_udf_in_bind():
+    db = SparqlDatabase()
+    db.add_triple(f"<{EX}a>", f"<{EX}v>", '"5"')
+    db.register_udf("DOUBLE", lambda x: str(int(x) * 2))
+    rows = db.query(f"""
+        SELECT ?d WHERE {{ ?s <{EX}v> ?v . BIND(DOUBLE(?v) AS ?d) }}""")
+    assert rows == [["10"]]
+
+
+def test_repeated_variable_in_pattern():
+    db = SparqlDatabase()
+    db.add_triple(f"<{EX}a>", f"<{EX}p>", f"<{EX}a>")
+    db.add_triple(f"<{EX}a>", f"<{EX}p>", f"<{EX}b>")
+    rows = db.query(f"SELECT ?x WHERE {{ ?x <{EX}p> ?x }}")
+    assert rows == [[f"{EX}a"]]
+
+
+def test_query_only_endpoint_rejects_update():
+    from kolibrie_amd.engine.query import execute_sparql_query
+    db = SparqlDatabase()
+    with pytest.raises(ValueError):
+        execute_sparql_query(f'INSERT DATA {{ <{EX}a> <{EX}p> "v" }}', db)
+
+
+def test_prefixed_query():
+    db = SparqlDatabase()
+    db.add_triple(f"<{EX}alice>", "<http://xmlns.com/foaf/0.1/name>", '"Alice"')
+    rows = db.query("""
+        PREFIX foaf: <http://xmlns.com/foaf/0.1/>
+        SELECT ?n WHERE { ?x foaf:name ?n }""")
+    assert rows == [["Alice"]]

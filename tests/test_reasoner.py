@@ -1,0 +1,135 @@
+"""Datalog reasoning (mirrors datalog/tests/reasoning_tests.rs: naive vs
+semi-naive equality, NAF, transitive closure, constraints/repairs)."""
+import pytest
+
+from kolibrie_amd import Reasoner
+from kolibrie_amd.reasoning.rule import Rule
+from kolibrie_amd.storage.terms import Constant, TriplePattern, Variable
+
+
+def _c(r, s):
+    x = r.dictionary.encode(s) & 0xFFFFFFFF
+    return Constant(x - 0x1_0000_0000 if x >= 0x8000_0000 else x)
+
+
+def _tp(r, s, p, o):
+    def t(x):
+        if isinstance(x, str) and x.startswith("?"):
+            return Variable(x[1:])
+        return _c(r, x)
+    return TriplePattern(t(s), t(p), t(o))
+
+
+def test_transitive_closure_semi_naive():
+    r = Reasoner()
+    for i in range(10):
+        r.add_abox_triple(f"n{i}", "edge", f"n{i+1}")
+    r.add_rule(Rule(
+        premise=[_tp(r, "?x", "edge", "?y"), _tp(r, "?y", "reach", "?z")],
+        conclusion=[_tp(r, "?x", "reach", "?z")],
+    ))
+    r.add_rule(Rule(
+        premise=[_tp(r, "?x", "edge", "?y")],
+        conclusion=[_tp(r, "?x", "reach", "?y")],
+    ))
+    n = r.infer_new_facts_semi_naive()
+    # reach = all pairs i<j over 11 nodes: 55; 10 are direct edges
+    assert n == 55
+    assert r.contains_fact("n0", "reach", "n10")
+
+
+def test_naive_equals_semi_naive():
+    def build():
+        r = Reasoner()
+        r.add_abox_triple("a", "p", "b")
+        r.add_abox_triple("b", "p", "c")
+        r.add_abox_triple("c", "p", "d")
+        r.add_rule(Rule(
+            premise=[_tp(r, "?x", "p", "?y"), _tp(r, "?y", "p", "?z")],
+            conclusion=[_tp(r, "?x", "q", "?z")],
+        ))
+        return r
+    r1, r2 = build(), build()
+    r1.infer_new_facts()
+    r2.infer_new_facts_semi_naive()
+    assert r1.all_fact_tuples() == r2.all_fact_tuples()
+
+
+def test_rule_text_parsing():
+    r = Reasoner()
+    r.add_abox_triple("http://e/alice", "http://e/parent", "http://e/bob")
+    r.add_abox_triple("http://e/bob", "http://e/parent", "http://e/carol")
+    r.add_rule_text("""
+        RULE :Grandparent :- CONSTRUCT { ?x <http://e/grandparent> ?z }
+        WHERE { ?x <http://e/parent> ?y . ?y <http://e/parent> ?z } .
+    """)
+    r.infer_new_facts_semi_naive()
+    assert r.contains_fact("http://e/alice", "http://e/grandparent", "http://e/carol")
+
+
+def test_naf_negation():
+    r = Reasoner()
+    r.add_abox_triple("a", "item", "x")
+    r.add_abox_triple("b", "item", "y")
+    r.add_abox_triple("a", "blocked", "x")
+    r.add_rule(Rule(
+        premise=[_tp(r, "?s", "item", "?v")],
+        negative_premise=[_tp(r, "?s", "blocked", "?v")],
+        conclusion=[_tp(r, "?s", "allowed", "?v")],
+    ))
+    r.infer_new_facts_semi_naive()
+    assert not r.contains_fact("a", "allowed", "x")
+    assert r.contains_fact("b", "allowed", "y")
+
+
+def test_rule_filters():
+    r = Reasoner()
+    r.add_abox_triple("m1", "temp", "95")
+    r.add_abox_triple("m2", "temp", "50")
+    r.add_rule_text("""
+        RULE :Hot :- CONSTRUCT { ?m <alert> "hot" }
+        WHERE { ?m <temp> ?t . FILTER(?t > 90) } .
+    """)
+    r.infer_new_facts_semi_naive()
+    assert r.contains_fact("m1", "alert", "hot")
+    assert not r.contains_fact("m2", "alert", "hot")
+
+
+def test_query_abox():
+    r = Reasoner()
+    r.add_abox_triple("a", "p", "b")
+    r.add_abox_triple("a", "q", "c")
+    out = r.query_abox("a", None, None)
+    assert out == [("a", "p", "b"), ("a", "q", "c")]
+    assert r.query_abox(None, "q", None) == [("a", "q", "c")]
+
+
+def test_constraints_and_repairs():
+    r = Reasoner()
+    r.add_abox_triple("x", "status", "on")
+    r.add_abox_triple("x", "status", "off")
+    # constraint: nothing may be both on and off
+    r.add_constraint(Rule(
+        premise=[_tp(r, "?s", "status", "on"), _tp(r, "?s", "status", "off")],
+        conclusion=[],
+    ))
+    assert r.violates_constraints()
+    repairs = r.compute_repairs()
+    assert repairs
+    assert all(len(rep) == 1 for rep in repairs)
+
+
+def test_deep_taxonomy_scaling():
+    """Deep-taxonomy shape: subclass chain depth 100."""
+    r = Reasoner()
+    depth = 100
+    for i in range(depth):
+        r.add_abox_triple(f"C{i}", "subClassOf", f"C{i+1}")
+    r.add_abox_triple("i0", "type", "C0")
+    r.add_rule(Rule(
+        premise=[_tp(r, "?x", "type", "?c"), _tp(r, "?c", "subClassOf", "?d")],
+        conclusion=[_tp(r, "?x", "type", "?d")],
+    ))
+    n = r.infer_new_facts_semi_naive()
+    assert n == depth
+    assert r.contains_fact("i0", "type", f"C{depth}")
