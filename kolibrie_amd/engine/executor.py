@@ -353,10 +353,19 @@ def join_bindings(left: Bindings, right: Bindings) -> Bindings:
     l_keyed, l_unkeyed = left.select(lb), left.select(~lb)
     r_keyed, r_unkeyed = right.select(rb), right.select(~rb)
     if l_keyed.n and r_keyed.n:
-        key_cols = [torch.cat([l_keyed.col(v), r_keyed.col(v)]) for v in shared]
-        gid, _ = group_index(key_cols)
-        lkey, rkey = gid[:l_keyed.n], gid[l_keyed.n:]
-        li, ri = merge_join_indices(lkey, rkey)
+        from ..ops import native_for
+        native = native_for(l_keyed.col(shared[0])) if len(shared) <= 4 else None
+        if native is not None:
+            # K2: chained open-addressing hash join on device
+            li, ri = native.hash_join(
+                [l_keyed.col(v).contiguous() for v in shared],
+                [r_keyed.col(v).contiguous() for v in shared],
+            )
+        else:
+            key_cols = [torch.cat([l_keyed.col(v), r_keyed.col(v)]) for v in shared]
+            gid, _ = group_index(key_cols)
+            lkey, rkey = gid[:l_keyed.n], gid[l_keyed.n:]
+            li, ri = merge_join_indices(lkey, rkey)
         parts.append(_merge_pairs(l_keyed, r_keyed, li, ri, shared))
     if l_unkeyed.n and right.n:
         parts.append(_compat_nlj(l_unkeyed, right, shared))

@@ -58,8 +58,19 @@ class CompiledExpr:
 
     # -------------------------------------------------------------- evaluate
     def eval_mask(self, b: Bindings, db) -> torch.Tensor:
-        m = _eval_bool(self.ast, b, db)
-        return m
+        if b.n > 0 and b.device.type == "cuda":
+            from ..ops import native_for
+            from ..ops.filter_bytecode import compile_filter
+            probe_col = next(iter(b.cols.values())) if b.cols else None
+            native = native_for(probe_col) if probe_col is not None else None
+            if native is not None:
+                prog = compile_filter(self.ast, b)
+                if prog is not None:
+                    ops_t, args_t, consts_t, col_list = prog
+                    return native.filter_bytecode(
+                        ops_t, args_t, consts_t, col_list,
+                        db.value_column(), b.n)
+        return _eval_bool(self.ast, b, db)
 
     def variables(self) -> List[str]:
         out: List[str] = []

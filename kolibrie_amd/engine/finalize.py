@@ -138,11 +138,52 @@ def _aggregate(select: SelectQuery, rows: Bindings, db) -> Bindings:
                 1, dev)
         return Bindings(out_cols, 0, dev)
 
-    if group_vars:
-        gid, ng = group_index([rows.col(v) for v in group_vars])
-    else:
-        gid = torch.zeros(n, dtype=torch.long, device=dev)
-        ng = 1
+    if not group_vars:
+        # single-group fast path: no scatter needed (COUNT/SUM/... over all)
+        out_cols: Dict[str, torch.Tensor] = {}
+        vc = db.value_column()
+        for p in select.variables:
+            if not p.aggregate:
+                continue
+            name = p.output_name()
+            if p.aggregate == "COUNT":
+                if p.agg_arg is None:
+                    c = n
+                else:
+                    col = rows.col(p.agg_arg) if rows.has(p.agg_arg) else None
+                    if col is None:
+                        c = 0
+                    elif p.distinct:
+                        bound = col[col != UNBOUND]
+                        c = int(torch.unique(bound).numel())
+                    else:
+                        c = int((col != UNBOUND).sum().item())
+                out_cols[name] = torch.tensor([_encode_i32(db, str(c))],
+                                              dtype=torch.int32, device=dev)
+                continue
+            col = rows.col(p.agg_arg) if p.agg_arg and rows.has(p.agg_arg) else None
+            if col is None:
+                out_cols[name] = torch.tensor([_encode_i32(db, "")],
+                                              dtype=torch.int32, device=dev)
+                continue
+            ids_u = col.to(torch.int64) & 0xFFFFFFFF
+            vals = vc[torch.clamp(ids_u, max=vc.numel() - 1)]
+            vals = vals[col != UNBOUND]
+            if p.aggregate == "SUM":
+                r = float(vals.sum().item()) if vals.numel() else 0.0
+            elif p.aggregate == "AVG":
+                r = float(vals.mean().item()) if vals.numel() else 0.0
+            elif p.aggregate == "MIN":
+                r = float(vals.min().item()) if vals.numel() else 0.0
+            elif p.aggregate == "MAX":
+                r = float(vals.max().item()) if vals.numel() else 0.0
+            else:
+                raise ValueError(f"unknown aggregate {p.aggregate}")
+            out_cols[name] = torch.tensor([_encode_i32(db, _fmt_num(r))],
+                                          dtype=torch.int32, device=dev)
+        return Bindings(out_cols, 1, dev)
+
+    gid, ng = group_index([rows.col(v) for v in group_vars])
     # representative row per group (first occurrence)
     rep = torch.full((ng,), -1, dtype=torch.long, device=dev)
     rev = torch.arange(n - 1, -1, -1, dtype=torch.long, device=dev)

@@ -126,6 +126,32 @@ def scan_probe(
             return probes[position]
         return torch.full((n_rows,), consts[position], dtype=torch.int32, device=dev)
 
+    # ---- native K1 path on device --------------------------------------
+    from ..ops import native_for
+    native = native_for(key12)
+    if native is not None and plen > 0:
+        if plen == 2:
+            keys = pack2(col_for(pos[0]).contiguous(), col_for(pos[1]).contiguous())
+            li, b_col, z_col = native.probe_exact(key12, z, keys.contiguous())
+        else:
+            li, b_col, z_col = native.probe_range(key12, z,
+                                                  col_for(pos[0]).contiguous())
+        out = [None, None, None]
+        out[pos[0]] = col_for(pos[0])[li]
+        out[pos[1]] = b_col
+        out[pos[2]] = z_col
+        s, p, o = out  # type: ignore[assignment]
+        covered = set(pos[:plen])
+        mask = None
+        for position in bound - covered:
+            col = (s, p, o)[position]
+            want = consts[position] if position in consts else probes[position][li]
+            m = col == want
+            mask = m if mask is None else (mask & m)
+        if mask is not None:
+            li, s, p, o = li[mask], s[mask], p[mask], o[mask]
+        return li, s, p, o
+
     if plen == 2:
         keys = pack2(col_for(pos[0]), col_for(pos[1]))
         lo = torch.searchsorted(key12, keys, side="left")
