@@ -16,8 +16,8 @@ _native = None
 _native_err: str | None = None
 
 try:
-    from . import _native as _native_mod  # type: ignore[attr-defined]
-    _native = _native_mod
+    import importlib
+    _native = importlib.import_module("kolibrie_amd.ops._native")
 except ImportError as e:  # extension not built
     _native_err = str(e)
 
