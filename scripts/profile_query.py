@@ -1,0 +1,101 @@
+#!/usr/bin/env python3
+"""Stage-level timing of the flagship query on a loaded 100M-triple shard.
+
+Usage: python scripts/profile_query.py [--triples N] [--device cuda:0]
+"""
+import argparse
+import sys
+import time
+
+import torch
+
+sys.path.insert(0, ".")
+
+from kolibrie_amd.engine.bindings import Bindings
+from kolibrie_amd.engine.executor import DatasetView, ExecutionContext, ExecutionEngine
+from kolibrie_amd.engine.finalize import decode_rows, finalize_select_bindings
+from kolibrie_amd.parallel.dist_engine import DistributedDatabase
+from kolibrie_amd.parallel.synthetic import FLAGSHIP_QUERY, generate_partition, plan_dataset
+from kolibrie_amd.parsing.sparql import parse_combined_query
+from kolibrie_amd.plan.lower import build_logical_plan
+from kolibrie_amd.plan.optimizer import Streamertail
+
+
+def main():
+    ap = argparse.ArgumentParser()
+    ap.add_argument("--triples", type=int, default=100_000_000)
+    ap.add_argument("--device", type=str,
+                    default="cuda:0" if torch.cuda.is_available() else "cpu")
+    ap.add_argument("--iters", type=int, default=10)
+    args = ap.parse_args()
+
+    dev = torch.device(args.device)
+    use_cuda = dev.type == "cuda"
+
+    ddb = DistributedDatabase(0, 1, dev)
+    ds = plan_dataset(ddb.db, args.triples)
+    t0 = time.perf_counter()
+    s, p, o = generate_partition(ds, 0, 1, 1234, dev)
+    print(f"gen: {time.perf_counter()-t0:.2f}s", flush=True)
+    t0 = time.perf_counter()
+    ddb.load_shard_columns(s, p, o)
+    if use_cuda:
+        torch.cuda.synchronize()
+    print(f"index build: {time.perf_counter()-t0:.2f}s", flush=True)
+    db = ddb.db
+
+    def sync():
+        if use_cuda:
+            torch.cuda.synchronize()
+
+    stages = {k: 0.0 for k in
+              ["parse", "stats", "lower", "optimize", "execute", "finalize", "decode"]}
+
+    # warmup
+    for _ in range(3):
+        db.query(FLAGSHIP_QUERY)
+    sync()
+
+    for _ in range(args.iters):
+        t = time.perf_counter()
+        cq = parse_combined_query(FLAGSHIP_QUERY)
+        prefixes = dict(db.prefixes)
+        prefixes.update(cq.prefixes)
+        sel = cq.select
+        sync(); stages["parse"] += time.perf_counter() - t
+
+        t = time.perf_counter()
+        stats = db.get_or_build_stats()
+        sync(); stages["stats"] += time.perf_counter() - t
+
+        t = time.perf_counter()
+        logical = build_logical_plan(sel.where, db, prefixes)
+        sync(); stages["lower"] += time.perf_counter() - t
+
+        t = time.perf_counter()
+        physical = Streamertail(stats).find_best_plan(logical)
+        sync(); stages["optimize"] += time.perf_counter() - t
+
+        t = time.perf_counter()
+        ctx = ExecutionContext(db, DatasetView())
+        rows = ExecutionEngine(ctx).execute(physical, Bindings.unit(db.device))
+        sync(); stages["execute"] += time.perf_counter() - t
+
+        t = time.perf_counter()
+        final = finalize_select_bindings(sel, rows, db)
+        sync(); stages["finalize"] += time.perf_counter() - t
+
+        t = time.perf_counter()
+        out = decode_rows(sel, final, db)
+        sync(); stages["decode"] += time.perf_counter() - t
+
+    print(f"physical plan: {physical}")
+    print(f"result: {out}")
+    total = sum(stages.values())
+    for k, v in stages.items():
+        print(f"{k:10s} {v/args.iters*1000:8.3f} ms  ({100*v/total:5.1f}%)")
+    print(f"{'TOTAL':10s} {total/args.iters*1000:8.3f} ms")
+
+
+if __name__ == "__main__":
+    main()
