@@ -356,11 +356,18 @@ def join_bindings(left: Bindings, right: Bindings) -> Bindings:
         from ..ops import native_for
         native = native_for(l_keyed.col(shared[0])) if len(shared) <= 4 else None
         if native is not None:
-            # K2: chained open-addressing hash join on device
-            li, ri = native.hash_join(
-                [l_keyed.col(v).contiguous() for v in shared],
-                [r_keyed.col(v).contiguous() for v in shared],
-            )
+            # K2: chained open-addressing hash join on device.
+            # Build on the smaller side, probe with the larger.
+            if l_keyed.n >= r_keyed.n:
+                li, ri = native.hash_join(
+                    [l_keyed.col(v).contiguous() for v in shared],
+                    [r_keyed.col(v).contiguous() for v in shared],
+                )
+            else:
+                ri, li = native.hash_join(
+                    [r_keyed.col(v).contiguous() for v in shared],
+                    [l_keyed.col(v).contiguous() for v in shared],
+                )
         else:
             key_cols = [torch.cat([l_keyed.col(v), r_keyed.col(v)]) for v in shared]
             gid, _ = group_index(key_cols)

@@ -33,7 +33,15 @@ from .physical import (
 )
 from .stats import DatabaseStats
 
-BIND_JOIN_MAX_LEFT = 4096  # prefer dependent probe when outer side is tiny
+BIND_JOIN_MAX_LEFT = 4096  # (legacy constant; kept for reference parity)
+PROBE_FACTOR = 8.0  # per-row cost of a K1 binary-search probe (calibrated on
+                    # MI355X: upper tree levels stay in L2/LLC)
+
+
+def _join_mode() -> str:
+    """A/B knob: KOLIBRIE_JOIN_MODE = auto|hash|bind."""
+    import os
+    return os.environ.get("KOLIBRIE_JOIN_MODE", "auto")
 
 
 def _pattern_vars(p: TriplePattern) -> List[str]:
@@ -185,20 +193,24 @@ class Streamertail:
                 cur_bound.update(_pattern_vars(s.pattern))
                 continue
             shared = [v for v in _pattern_vars(s.pattern) if v in cur_bound]
-            # candidates (ref find_best_plan_recursive:382)
-            hash_cost = cur_cost + self.est.scan_cost(s.pattern, set(), s.graph) \
-                + 2.0 * (cur_rows + rows_free)
-            bind_cost = cur_cost + cur_rows * 2.0 \
-                + cur_rows * max(1.0, rows_given_bound) * 1.0 + 2000.0
+            # candidates (ref find_best_plan_recursive:382).  GPU-era costs:
+            # bind join = K1 per-row index probe (~PROBE_FACTOR reads/row,
+            # upper index levels cache-resident); hash join = K2 build of the
+            # full right scan + probe; both emit the join output.
+            emit_rows = cur_rows * max(1.0, rows_given_bound)
+            bind_cost = cur_cost + cur_rows * PROBE_FACTOR + emit_rows + 2000.0
+            hash_cost = cur_cost + 3.0 * max(1.0, rows_free) \
+                + 2.0 * cur_rows + emit_rows + 6000.0
             nlj_cost = cur_cost + 10.0 * cur_rows * max(1.0, rows_free)
+            mode = _join_mode()
             if not shared:
                 cur = PNestedLoopJoin(cur, scan_op)
                 cur_cost = nlj_cost
                 cur_rows = cur_rows * max(1.0, rows_free)
-            elif cur_rows <= BIND_JOIN_MAX_LEFT and bind_cost < hash_cost:
+            elif mode == "bind" or (mode == "auto" and bind_cost <= hash_cost):
                 cur = PBindJoin(cur, scan_op)
                 cur_cost = bind_cost
-                cur_rows = cur_rows * max(1.0, rows_given_bound)
+                cur_rows = emit_rows
             else:
                 cur = PHashJoin(cur, scan_op)
                 cur_cost = hash_cost
