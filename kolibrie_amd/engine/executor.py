@@ -67,12 +67,19 @@ class ExecutionEngine:
     def execute(self, op: PhysicalOp, incoming: Bindings) -> Bindings:
         if isinstance(op, PUnit):
             return incoming
+        needed = getattr(op, "needed", None)
         if isinstance(op, (PTableScan, PIndexScan)):
-            return self._exec_scan(op.pattern, op.graph, incoming)
+            return self._exec_scan(op.pattern, op.graph, incoming, needed)
         if isinstance(op, PStarJoin):
             out = incoming
-            for pat in op.patterns:
-                out = self._exec_scan(pat, op.graph, out)
+            for i, pat in enumerate(op.patterns):
+                # keep vars required by later patterns of the chain
+                step_needed = None
+                if needed is not None:
+                    step_needed = set(needed) | {op.join_var}
+                    for later in op.patterns[i + 1:]:
+                        step_needed.update(later.variables())
+                out = self._exec_scan(pat, op.graph, out, step_needed)
                 if out.is_empty():
                     break
             return out
@@ -81,7 +88,7 @@ class ExecutionEngine:
             if left.is_empty():
                 return left
             right = self.execute(op.right, Bindings.unit(self.device))
-            return join_bindings(left, right)
+            return join_bindings(left, right, needed)
         if isinstance(op, PBindJoin):
             left = self.execute(op.left, incoming)
             if left.is_empty():
@@ -92,7 +99,7 @@ class ExecutionEngine:
             if left.is_empty():
                 return left
             right = self.execute(op.right, Bindings.unit(self.device))
-            return join_bindings(left, right)
+            return join_bindings(left, right, needed)
         if isinstance(op, PUnion):
             l = self.execute(op.left, incoming)
             r = self.execute(op.right, incoming)
@@ -102,17 +109,17 @@ class ExecutionEngine:
             if rows.is_empty():
                 return rows
             mask = op.condition.eval_mask(rows, self.db)
-            return rows.select(mask)
+            return _prune(rows.select(mask), needed)
         if isinstance(op, PBind):
             rows = self.execute(op.input, incoming)
             if rows.is_empty():
                 return rows.project(rows.variables + [op.var])
             col = op.expr.eval_ids(rows, self.db)
-            return rows.with_col(op.var, col)
+            return _prune(rows.with_col(op.var, col), needed)
         if isinstance(op, PValues):
             rows = self.execute(op.input, incoming)
             vals = self._values_bindings(op.variables, op.rows)
-            return join_bindings(rows, vals)
+            return join_bindings(rows, vals, needed)
         if isinstance(op, PSubquery):
             rows = self.execute(op.input, incoming)
             sub = self._exec_subquery(op.select)
@@ -143,21 +150,26 @@ class ExecutionEngine:
         return Bindings(cols, n, dev)
 
     # ----------------------------------------------------------------- scan --
-    def _exec_scan(self, pattern: TriplePattern, scope, incoming: Bindings
-                   ) -> Bindings:
+    def _exec_scan(self, pattern: TriplePattern, scope, incoming: Bindings,
+                   needed=None) -> Bindings:
         if scope is None:
             idx = self.ctx.default_index()
-            return self._scan_index(idx, pattern, incoming, extra=None)
+            return self._scan_index(idx, pattern, incoming, extra=None,
+                                    needed=needed)
         if scope[0] == "const":
             idx = self.db.store.graph_index(scope[1] & 0xFFFFFFFF)
-            return self._scan_index(idx, pattern, incoming, extra=None)
+            return self._scan_index(idx, pattern, incoming, extra=None,
+                                    needed=needed)
         # GRAPH ?g — iterate named graphs, bind the graph variable
         gvar = scope[1]
+        if needed is not None:
+            needed = set(needed) | {gvar}
         parts = []
         for gid in self.ctx.named_graph_ids():
             idx = self.db.store.graph_index(gid)
             gid_i32 = gid - 0x1_0000_0000 if gid >= 0x8000_0000 else gid
-            res = self._scan_index(idx, pattern, incoming, extra=(gvar, gid_i32))
+            res = self._scan_index(idx, pattern, incoming, extra=(gvar, gid_i32),
+                                   needed=needed)
             parts.append(res)
         if not parts:
             all_vars = pattern.variables() + [gvar]
@@ -165,8 +177,8 @@ class ExecutionEngine:
         return Bindings.concat(parts, self.device)
 
     def _scan_index(self, idx: GraphIndex, pattern: TriplePattern,
-                    incoming: Bindings, extra: Optional[Tuple[str, int]]
-                    ) -> Bindings:
+                    incoming: Bindings, extra: Optional[Tuple[str, int]],
+                    needed=None) -> Bindings:
         """Scan one index, extending each incoming row (engine.rs:1018)."""
         dev = self.device
         # graph-variable consistency: if ?g already bound, pre-filter rows
@@ -194,11 +206,11 @@ class ExecutionEngine:
         if not probe_vars:
             s, p, o = scan_unit(idx, consts)
             exec_stats.bump("QUADS_EXAMINED", s.numel())
-            cand = self._build_candidate(s, p, o, var_pos, qt_pos, None)
+            cand = self._build_candidate(s, p, o, var_pos, qt_pos, None, needed)
             if is_unit:
                 exec_stats.bump("ROWS_EMITTED", cand.n)
                 return self._add_extra(cand, extra)
-            out = join_bindings(inc, cand)
+            out = join_bindings(_prune(inc, needed), cand, needed)
             exec_stats.bump("ROWS_EMITTED", out.n)
             return self._add_extra(out, extra)
 
@@ -216,16 +228,18 @@ class ExecutionEngine:
                       if (sval >> k) & 1}
             if not active:
                 s, p, o = scan_unit(idx, consts)
-                cand = self._build_candidate(s, p, o, var_pos, qt_pos, None)
+                cand = self._build_candidate(s, p, o, var_pos, qt_pos, None, needed)
                 # cartesian: these rows have the vars unbound -> binder
                 sub2 = sub.drop_cols(list(probe_vars.values()))
-                parts.append(join_bindings(sub2, cand))
+                parts.append(join_bindings(sub2, cand, needed))
                 continue
             probes = {i: sub.col(v) for i, v in active.items()}
             li, s, p, o = scan_probe(idx, consts, probes)
             exec_stats.bump("QUADS_EXAMINED", s.numel())
-            base = sub.gather(li)
-            cand = self._build_candidate(s, p, o, var_pos, qt_pos, base)
+            # projection pushdown: gather only columns still needed
+            base_src = sub if needed is None else _prune(sub, set(needed) | set(active.values()))
+            base = base_src.gather(li)
+            cand = self._build_candidate(s, p, o, var_pos, qt_pos, base, needed)
             parts.append(cand)
         out = Bindings.concat(parts, dev) if len(parts) != 1 else parts[0]
         exec_stats.bump("ROWS_EMITTED", out.n)
@@ -246,7 +260,7 @@ class ExecutionEngine:
 
     def _build_candidate(self, s, p, o, var_pos: Dict[int, str],
                          qt_pos: Dict[int, QuotedTriplePattern],
-                         base: Optional[Bindings]) -> Bindings:
+                         base: Optional[Bindings], needed=None) -> Bindings:
         """Assemble bindings from scanned triple columns: bind variables,
         enforce repeated-variable equality (engine.rs:1223-1239) and match
         quoted-triple sub-patterns (engine.rs:1253)."""
@@ -274,9 +288,9 @@ class ExecutionEngine:
             mask &= m
             out_cols.update(qt_cols)
         res = Bindings(out_cols, n, dev)
-        if bool(mask.all()):
-            return res
-        return res.select(mask)
+        if not bool(mask.all()):
+            res = res.select(mask)
+        return _prune(res, needed)
 
     def _match_quoted(self, ids: torch.Tensor, qtp: QuotedTriplePattern,
                       bound_cols: Dict[str, torch.Tensor]
@@ -322,7 +336,17 @@ class ExecutionEngine:
 
 
 # ------------------------------------------------------------------- joins --
-def join_bindings(left: Bindings, right: Bindings) -> Bindings:
+def _prune(b: Bindings, needed) -> Bindings:
+    """Drop columns outside the needed set (projection pushdown)."""
+    if needed is None:
+        return b
+    cols = {v: c for v, c in b.cols.items() if v in needed}
+    if len(cols) == len(b.cols):
+        return b
+    return Bindings(cols, b.n, b.device)
+
+
+def join_bindings(left: Bindings, right: Bindings, needed=None) -> Bindings:
     """SPARQL-compatible natural join (multiset).
 
     Keyed rows (all shared vars bound) go through the sort-merge equi-join
@@ -331,9 +355,9 @@ def join_bindings(left: Bindings, right: Bindings) -> Bindings:
     """
     dev = left.device
     if left.n == 1 and not left.cols:
-        return right
+        return _prune(right, needed)
     if right.n == 1 and not right.cols:
-        return left
+        return _prune(left, needed)
     if left.is_empty() or right.is_empty():
         vars_ = list(dict.fromkeys(left.variables + right.variables))
         return Bindings.empty(dev, vars_)
@@ -342,7 +366,7 @@ def join_bindings(left: Bindings, right: Bindings) -> Bindings:
         # cartesian product
         li = torch.arange(left.n, dtype=torch.long, device=dev).repeat_interleave(right.n)
         ri = torch.arange(right.n, dtype=torch.long, device=dev).repeat(left.n)
-        return _merge_pairs(left, right, li, ri, shared)
+        return _merge_pairs(left, right, li, ri, shared, needed)
     lb = torch.ones(left.n, dtype=torch.bool, device=dev)
     for v in shared:
         lb &= left.col(v) != UNBOUND
@@ -373,11 +397,11 @@ def join_bindings(left: Bindings, right: Bindings) -> Bindings:
             gid, _ = group_index(key_cols)
             lkey, rkey = gid[:l_keyed.n], gid[l_keyed.n:]
             li, ri = merge_join_indices(lkey, rkey)
-        parts.append(_merge_pairs(l_keyed, r_keyed, li, ri, shared))
+        parts.append(_merge_pairs(l_keyed, r_keyed, li, ri, shared, needed))
     if l_unkeyed.n and right.n:
-        parts.append(_compat_nlj(l_unkeyed, right, shared))
+        parts.append(_compat_nlj(l_unkeyed, right, shared, needed))
     if l_keyed.n and r_unkeyed.n:
-        parts.append(_compat_nlj(l_keyed, r_unkeyed, shared))
+        parts.append(_compat_nlj(l_keyed, r_unkeyed, shared, needed))
     if not parts:
         vars_ = list(dict.fromkeys(left.variables + right.variables))
         return Bindings.empty(dev, vars_)
@@ -385,25 +409,33 @@ def join_bindings(left: Bindings, right: Bindings) -> Bindings:
 
 
 def _merge_pairs(left: Bindings, right: Bindings, li, ri,
-                 shared: Sequence[str]) -> Bindings:
+                 shared: Sequence[str], needed=None) -> Bindings:
     dev = left.device
     cols: Dict[str, torch.Tensor] = {}
     for v, c in left.cols.items():
+        if needed is not None and v not in needed:
+            continue
         cols[v] = c[li]
     for v, c in right.cols.items():
+        if needed is not None and v not in needed:
+            continue
         if v in cols:
             if v in shared:
                 # take bound value (left may be UNBOUND in compat path)
                 lvals = cols[v]
                 rvals = c[ri]
                 cols[v] = torch.where(lvals != UNBOUND, lvals, rvals)
+        elif v in shared and left.has(v):
+            lvals = left.col(v)[li]
+            rvals = c[ri]
+            cols[v] = torch.where(lvals != UNBOUND, lvals, rvals)
         else:
             cols[v] = c[ri]
     return Bindings(cols, li.numel(), dev)
 
 
-def _compat_nlj(left: Bindings, right: Bindings, shared: Sequence[str]
-                ) -> Bindings:
+def _compat_nlj(left: Bindings, right: Bindings, shared: Sequence[str],
+                needed=None) -> Bindings:
     """Cartesian + compatibility mask (UNBOUND matches anything)."""
     dev = left.device
     li = torch.arange(left.n, dtype=torch.long, device=dev).repeat_interleave(right.n)
@@ -414,7 +446,7 @@ def _compat_nlj(left: Bindings, right: Bindings, shared: Sequence[str]
         rv = right.col(v)[ri]
         mask &= (lv == rv) | (lv == UNBOUND) | (rv == UNBOUND)
     li, ri = li[mask], ri[mask]
-    return _merge_pairs(left, right, li, ri, shared)
+    return _merge_pairs(left, right, li, ri, shared, needed)
 
 
 def anti_join(left: Bindings, right: Bindings) -> Bindings:

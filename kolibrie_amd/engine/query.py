@@ -34,6 +34,21 @@ def _build_view(select: SelectQuery, db, prefixes) -> DatasetView:
     return view
 
 
+def _top_needed(select: SelectQuery):
+    """Variables the finalize stage consumes (projection pushdown root).
+    None = keep everything (SELECT *)."""
+    if select.select_star or not select.variables:
+        return None
+    needed = set(select.group_by)
+    needed.update(c.var for c in select.order_by)
+    for p in select.variables:
+        if p.var:
+            needed.add(p.var)
+        if p.agg_arg:
+            needed.add(p.agg_arg)
+    return needed
+
+
 def execute_select(select: SelectQuery, db, prefixes: Dict[str, str]
                    ) -> List[List[str]]:
     from ..plan.lower import build_logical_plan
@@ -42,6 +57,8 @@ def execute_select(select: SelectQuery, db, prefixes: Dict[str, str]
     stats = db.get_or_build_stats()
     logical = build_logical_plan(select.where, db, prefixes)
     physical = Streamertail(stats).find_best_plan(logical)
+    from ..plan.optimizer import annotate_needed
+    annotate_needed(physical, _top_needed(select))
     ctx = ExecutionContext(db, view)
     rows = ExecutionEngine(ctx).execute(physical, Bindings.unit(db.device))
     final = finalize_select_bindings(select, rows, db)

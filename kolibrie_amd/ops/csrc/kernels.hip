@@ -116,6 +116,33 @@ __global__ void probe_emit(const int64_t* __restrict__ key12,
   }
 }
 
+// balanced emit: parallel over OUTPUT elements (skew-proof — a probe row
+// with 10^4 matches no longer serializes one thread); each output finds its
+// probe row by binary search over the offset array.
+__global__ void probe_emit_balanced(const int64_t* __restrict__ key12,
+                                    const int32_t* __restrict__ z, int64_t m,
+                                    const int64_t* __restrict__ lo,
+                                    const int64_t* __restrict__ offs,  // [m+1]
+                                    int64_t total,
+                                    int64_t* __restrict__ li_out,
+                                    int32_t* __restrict__ b_out,
+                                    int32_t* __restrict__ z_out) {
+  for (int64_t idx = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       idx < total; idx += (int64_t)gridDim.x * blockDim.x) {
+    // find probe row i with offs[i] <= idx < offs[i+1]
+    int64_t a = 0, b = m;
+    while (a + 1 < b) {
+      int64_t mid = (a + b) >> 1;
+      if (offs[mid] <= idx) a = mid; else b = mid;
+    }
+    int64_t j = idx - offs[a];
+    int64_t src = lo[a] + j;
+    li_out[idx] = a;
+    b_out[idx] = static_cast<int32_t>(key12[src] & 0xFFFFFFFFLL);
+    z_out[idx] = z[src];
+  }
+}
+
 // ------------------------------------------------------------- K2: hash join
 constexpr int kMaxKeyCols = 4;
 
@@ -302,6 +329,39 @@ inline hipStream_t cur_stream() {
 
 // ------------------------------------------------------------ host launchers
 
+// shared emit phase for the K1 probes: per-probe-row emit for low fanout,
+// output-parallel balanced emit for skewed ranges.
+static std::vector<at::Tensor> emit_phase(at::Tensor key12, at::Tensor z,
+                                          at::Tensor lo, at::Tensor cnt,
+                                          int64_t m, hipStream_t stream) {
+  auto opts_long = key12.options();
+  auto offs = at::cumsum(cnt, 0, at::kLong);
+  int64_t total = m > 0 ? offs[-1].item<int64_t>() : 0;
+  auto li = at::empty({total}, opts_long);
+  auto b = at::empty({total}, z.options());
+  auto zz = at::empty({total}, z.options());
+  if (total == 0) return {li, b, zz};
+  if (total >= 8 * m) {
+    auto offs_full = at::zeros({m + 1}, opts_long);
+    offs_full.narrow(0, 1, m).copy_(offs);
+    hipLaunchKernelGGL(probe_emit_balanced, dim3(grid_for(total)),
+                       dim3(kBlock), 0, stream, key12.data_ptr<int64_t>(),
+                       z.data_ptr<int32_t>(), m, lo.data_ptr<int64_t>(),
+                       offs_full.data_ptr<int64_t>(), total,
+                       li.data_ptr<int64_t>(), b.data_ptr<int32_t>(),
+                       zz.data_ptr<int32_t>());
+  } else {
+    auto offs_excl = offs - cnt.to(at::kLong);
+    hipLaunchKernelGGL(probe_emit, dim3(grid_for(m)), dim3(kBlock), 0, stream,
+                       key12.data_ptr<int64_t>(), z.data_ptr<int32_t>(), m,
+                       lo.data_ptr<int64_t>(), cnt.data_ptr<int32_t>(),
+                       offs_excl.data_ptr<int64_t>(), li.data_ptr<int64_t>(),
+                       b.data_ptr<int32_t>(), zz.data_ptr<int32_t>());
+  }
+  HIP_OK(hipGetLastError());
+  return {li, b, zz};
+}
+
 // K1 probe — exact (packed 2-col key) mode.
 std::vector<at::Tensor> probe_exact(at::Tensor key12, at::Tensor z,
                                     at::Tensor keys) {
@@ -321,21 +381,7 @@ std::vector<at::Tensor> probe_exact(at::Tensor key12, at::Tensor z,
                        cnt.data_ptr<int32_t>());
     HIP_OK(hipGetLastError());
   }
-  auto offs = at::cumsum(cnt, 0, at::kLong);
-  int64_t total = m > 0 ? offs[-1].item<int64_t>() : 0;
-  auto offs_excl = offs - cnt.to(at::kLong);
-  auto li = at::empty({total}, keys.options());
-  auto b = at::empty({total}, z.options());
-  auto zz = at::empty({total}, z.options());
-  if (total > 0) {
-    hipLaunchKernelGGL(probe_emit, dim3(grid_for(m)), dim3(kBlock), 0, stream,
-                       key12.data_ptr<int64_t>(), z.data_ptr<int32_t>(), m,
-                       lo.data_ptr<int64_t>(), cnt.data_ptr<int32_t>(),
-                       offs_excl.data_ptr<int64_t>(), li.data_ptr<int64_t>(),
-                       b.data_ptr<int32_t>(), zz.data_ptr<int32_t>());
-    HIP_OK(hipGetLastError());
-  }
-  return {li, b, zz};
+  return emit_phase(key12, z, lo, cnt, m, stream);
 }
 
 // K1 probe — range (1-col prefix) mode.
@@ -355,21 +401,7 @@ std::vector<at::Tensor> probe_range(at::Tensor key12, at::Tensor z,
                        cnt.data_ptr<int32_t>());
     HIP_OK(hipGetLastError());
   }
-  auto offs = at::cumsum(cnt, 0, at::kLong);
-  int64_t total = m > 0 ? offs[-1].item<int64_t>() : 0;
-  auto offs_excl = offs - cnt.to(at::kLong);
-  auto li = at::empty({total}, key12.options());
-  auto b = at::empty({total}, z.options());
-  auto zz = at::empty({total}, z.options());
-  if (total > 0) {
-    hipLaunchKernelGGL(probe_emit, dim3(grid_for(m)), dim3(kBlock), 0, stream,
-                       key12.data_ptr<int64_t>(), z.data_ptr<int32_t>(), m,
-                       lo.data_ptr<int64_t>(), cnt.data_ptr<int32_t>(),
-                       offs_excl.data_ptr<int64_t>(), li.data_ptr<int64_t>(),
-                       b.data_ptr<int32_t>(), zz.data_ptr<int32_t>());
-    HIP_OK(hipGetLastError());
-  }
-  return {li, b, zz};
+  return emit_phase(key12, z, lo, cnt, m, stream);
 }
 
 // K2 hash join: returns (li, ri) index pairs; multiset semantics.

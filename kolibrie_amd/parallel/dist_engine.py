@@ -76,6 +76,9 @@ class DistributedDatabase:
         stats = db.get_or_build_stats()
         logical = build_logical_plan(sel.where, db, prefixes)
         physical = Streamertail(stats).find_best_plan(logical)
+        from ..engine.query import _top_needed
+        from ..plan.optimizer import annotate_needed
+        annotate_needed(physical, _top_needed(sel))
         ctx = ExecutionContext(db, DatasetView())
         rows = ExecutionEngine(ctx).execute(physical, Bindings.unit(db.device))
         # apply projection only (no aggregates here)

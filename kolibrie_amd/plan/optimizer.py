@@ -234,6 +234,10 @@ class Streamertail:
         return PTableScan(pattern, graph)
 
     def _out_vars(self, op: LogicalOp, bound: Set[str]) -> Set[str]:
+        return _logical_out_vars(op)
+
+
+def _logical_out_vars(op: LogicalOp) -> Set[str]:
         out: Set[str] = set()
 
         def rec(x):
@@ -255,3 +259,126 @@ class Streamertail:
 
         rec(op)
         return out
+
+
+# ----------------------------------------------------------- projection push
+def phys_out_vars(op: PhysicalOp) -> Set[str]:
+    """Static output-variable set of a physical subtree."""
+    from .physical import (
+        PBind, PBindJoin, PFilter, PHashJoin, PIndexScan, PInMemoryBuffer,
+        PMLPredict, PMinus, PNestedLoopJoin, PProjection, PStarJoin,
+        PSubquery, PTableScan, PUnion, PValues,
+    )
+    if isinstance(op, (PTableScan, PIndexScan)):
+        out = set(op.pattern.variables())
+        if op.graph is not None and op.graph[0] == "var":
+            out.add(op.graph[1])
+        return out
+    if isinstance(op, PStarJoin):
+        out = {op.join_var}
+        for p in op.patterns:
+            out.update(p.variables())
+        if op.graph is not None and op.graph[0] == "var":
+            out.add(op.graph[1])
+        return out
+    if isinstance(op, (PHashJoin, PBindJoin, PNestedLoopJoin, PUnion, PMinus)):
+        return phys_out_vars(op.left) | phys_out_vars(op.right)
+    if isinstance(op, PFilter):
+        return phys_out_vars(op.input)
+    if isinstance(op, PBind):
+        return phys_out_vars(op.input) | {op.var}
+    if isinstance(op, PValues):
+        return phys_out_vars(op.input) | set(op.variables)
+    if isinstance(op, PSubquery):
+        sub_vars = set()
+        sel = op.select.select
+        if sel.select_star or not sel.variables:
+            sub_vars = _logical_out_vars(op.select.plan)
+        else:
+            sub_vars = {p.output_name() for p in sel.variables}
+        return phys_out_vars(op.input) | sub_vars
+    if isinstance(op, PMLPredict):
+        out = phys_out_vars(op.input)
+        ov = op.info.get("output_var")
+        if ov:
+            out.add(ov)
+        return out
+    if isinstance(op, PInMemoryBuffer):
+        return set(op.bindings.variables) if op.bindings is not None else set()
+    return set()
+
+
+def annotate_needed(op: PhysicalOp, needed):
+    """Projection pushdown: attach the set of variables each node must emit
+    (None = all).  Executors drop/skip unneeded columns — on 100M-row joins
+    this eliminates whole gather passes over HBM."""
+    from .physical import (
+        PBind, PBindJoin, PFilter, PHashJoin, PMLPredict, PMinus,
+        PNestedLoopJoin, PStarJoin, PSubquery, PUnion, PValues,
+    )
+    op.needed = None if needed is None else frozenset(needed)
+    if isinstance(op, (PHashJoin, PNestedLoopJoin)):
+        lv = phys_out_vars(op.left)
+        rv = phys_out_vars(op.right)
+        shared = lv & rv
+        annotate_needed(op.left, None if needed is None else (set(needed) | shared) & lv)
+        annotate_needed(op.right, None if needed is None else (set(needed) | shared) & rv)
+        return
+    if isinstance(op, PBindJoin):
+        lv = phys_out_vars(op.left)
+        rv = phys_out_vars(op.right)
+        shared = lv & rv
+        annotate_needed(op.left, None if needed is None else (set(needed) | shared) & lv)
+        annotate_needed(op.right, needed)
+        return
+    if isinstance(op, PMinus):
+        lv = phys_out_vars(op.left)
+        rv = phys_out_vars(op.right)
+        shared = lv & rv
+        annotate_needed(op.left, None if needed is None else set(needed) | shared)
+        annotate_needed(op.right, None)
+        return
+    if isinstance(op, PUnion):
+        annotate_needed(op.left, needed)
+        annotate_needed(op.right, needed)
+        return
+    if isinstance(op, PFilter):
+        cond_vars = set(op.condition.variables())
+        annotate_needed(op.input, None if needed is None else set(needed) | cond_vars)
+        return
+    if isinstance(op, PBind):
+        from ..engine.filters import _collect_vars
+        expr_vars = []
+        _collect_vars(op.expr.ast, expr_vars)
+        child = None if needed is None else (set(needed) - {op.var}) | set(expr_vars)
+        annotate_needed(op.input, child)
+        return
+    if isinstance(op, PValues):
+        annotate_needed(op.input,
+                        None if needed is None else set(needed) | set(op.variables))
+        return
+    if isinstance(op, PSubquery):
+        sub_sel = op.select.select
+        sub_out = ({p.output_name() for p in sub_sel.variables}
+                   if sub_sel.variables and not sub_sel.select_star else None)
+        if op.select.physical is not None:
+            inner_needed = None
+            if sub_out is not None:
+                inner_needed = set()
+                for p in sub_sel.variables:
+                    if p.var:
+                        inner_needed.add(p.var)
+                    if p.agg_arg:
+                        inner_needed.add(p.agg_arg)
+                inner_needed.update(sub_sel.group_by)
+                inner_needed.update(c.var for c in sub_sel.order_by)
+            annotate_needed(op.select.physical, inner_needed)
+        annotate_needed(op.input,
+                        None if needed is None or sub_out is None
+                        else set(needed) | sub_out)
+        return
+    if isinstance(op, PMLPredict):
+        annotate_needed(op.input, None)
+        return
+    if hasattr(op, "input"):
+        annotate_needed(op.input, needed)
