@@ -379,7 +379,27 @@ def join_bindings(left: Bindings, right: Bindings, needed=None) -> Bindings:
     if l_keyed.n and r_keyed.n:
         from ..ops import native_for
         native = native_for(l_keyed.col(shared[0])) if len(shared) <= 4 else None
-        if native is not None:
+        merged = None
+        if len(shared) == 1:
+            # sortedness-aware merge join: a PSO-slice scan output is already
+            # sorted by subject — searchsorted beats a hash build (no table,
+            # sequential probe locality).  Detecting monotonicity is one
+            # cheap pass.
+            lk = l_keyed.col(shared[0])
+            rk = r_keyed.col(shared[0])
+            if _is_sorted(rk):
+                lo = torch.searchsorted(rk, lk, side="left")
+                hi = torch.searchsorted(rk, lk, side="right")
+                li, ri = _expand_ranges(lo, hi)
+                merged = _merge_pairs(l_keyed, r_keyed, li, ri, shared, needed)
+            elif _is_sorted(lk):
+                lo = torch.searchsorted(lk, rk, side="left")
+                hi = torch.searchsorted(lk, rk, side="right")
+                ri, li = _expand_ranges(lo, hi)
+                merged = _merge_pairs(l_keyed, r_keyed, li, ri, shared, needed)
+        if merged is not None:
+            parts.append(merged)
+        elif native is not None:
             # K2: chained open-addressing hash join on device.
             # Build on the smaller side, probe with the larger.
             if l_keyed.n >= r_keyed.n:
@@ -397,7 +417,8 @@ def join_bindings(left: Bindings, right: Bindings, needed=None) -> Bindings:
             gid, _ = group_index(key_cols)
             lkey, rkey = gid[:l_keyed.n], gid[l_keyed.n:]
             li, ri = merge_join_indices(lkey, rkey)
-        parts.append(_merge_pairs(l_keyed, r_keyed, li, ri, shared, needed))
+        if merged is None:
+            parts.append(_merge_pairs(l_keyed, r_keyed, li, ri, shared, needed))
     if l_unkeyed.n and right.n:
         parts.append(_compat_nlj(l_unkeyed, right, shared, needed))
     if l_keyed.n and r_unkeyed.n:
@@ -406,6 +427,27 @@ def join_bindings(left: Bindings, right: Bindings, needed=None) -> Bindings:
         vars_ = list(dict.fromkeys(left.variables + right.variables))
         return Bindings.empty(dev, vars_)
     return Bindings.concat(parts, dev) if len(parts) > 1 else parts[0]
+
+
+def _is_sorted(col: torch.Tensor) -> bool:
+    if col.numel() <= 1:
+        return True
+    return bool((col[1:] >= col[:-1]).all())
+
+
+def _expand_ranges(lo: torch.Tensor, hi: torch.Tensor):
+    """Expand per-row [lo, hi) ranges into (row_idx, range_pos) pairs."""
+    dev = lo.device
+    cnt = hi - lo
+    total = int(cnt.sum().item())
+    n = lo.numel()
+    if total == 0:
+        e = torch.empty(0, dtype=torch.long, device=dev)
+        return e, e.clone()
+    li = torch.repeat_interleave(torch.arange(n, dtype=torch.long, device=dev), cnt)
+    starts = torch.cumsum(cnt, 0) - cnt
+    pos = torch.arange(total, dtype=torch.long, device=dev) - starts[li]
+    return li, lo[li] + pos
 
 
 def _merge_pairs(left: Bindings, right: Bindings, li, ri,
