@@ -69,7 +69,8 @@ class ExecutionEngine:
             return incoming
         needed = getattr(op, "needed", None)
         if isinstance(op, (PTableScan, PIndexScan)):
-            return self._exec_scan(op.pattern, op.graph, incoming, needed)
+            return self._exec_scan(op.pattern, op.graph, incoming, needed,
+                                   sort_hint=getattr(op, "sort_hint", None))
         if isinstance(op, PStarJoin):
             out = incoming
             for i, pat in enumerate(op.patterns):
@@ -79,7 +80,8 @@ class ExecutionEngine:
                     step_needed = set(needed) | {op.join_var}
                     for later in op.patterns[i + 1:]:
                         step_needed.update(later.variables())
-                out = self._exec_scan(pat, op.graph, out, step_needed)
+                out = self._exec_scan(pat, op.graph, out, step_needed,
+                                      sort_hint=0 if i == 0 else None)
                 if out.is_empty():
                     break
             return out
@@ -151,15 +153,15 @@ class ExecutionEngine:
 
     # ----------------------------------------------------------------- scan --
     def _exec_scan(self, pattern: TriplePattern, scope, incoming: Bindings,
-                   needed=None) -> Bindings:
+                   needed=None, sort_hint=None) -> Bindings:
         if scope is None:
             idx = self.ctx.default_index()
             return self._scan_index(idx, pattern, incoming, extra=None,
-                                    needed=needed)
+                                    needed=needed, sort_hint=sort_hint)
         if scope[0] == "const":
             idx = self.db.store.graph_index(scope[1] & 0xFFFFFFFF)
             return self._scan_index(idx, pattern, incoming, extra=None,
-                                    needed=needed)
+                                    needed=needed, sort_hint=sort_hint)
         # GRAPH ?g — iterate named graphs, bind the graph variable
         gvar = scope[1]
         if needed is not None:
@@ -178,7 +180,7 @@ class ExecutionEngine:
 
     def _scan_index(self, idx: GraphIndex, pattern: TriplePattern,
                     incoming: Bindings, extra: Optional[Tuple[str, int]],
-                    needed=None) -> Bindings:
+                    needed=None, sort_hint=None) -> Bindings:
         """Scan one index, extending each incoming row (engine.rs:1018)."""
         dev = self.device
         # graph-variable consistency: if ?g already bound, pre-filter rows
@@ -204,7 +206,7 @@ class ExecutionEngine:
         exec_stats.bump("SCAN_PROBES", max(1, inc.n))
 
         if not probe_vars:
-            s, p, o = scan_unit(idx, consts)
+            s, p, o = scan_unit(idx, consts, sort_hint=sort_hint)
             exec_stats.bump("QUADS_EXAMINED", s.numel())
             cand = self._build_candidate(s, p, o, var_pos, qt_pos, None, needed)
             if is_unit:

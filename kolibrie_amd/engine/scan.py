@@ -24,7 +24,8 @@ from .tensor_utils import pack2, unpack2
 _ORDER_POS = {SPO: (0, 1, 2), POS: (1, 2, 0), OSP: (2, 0, 1), PSO: (1, 0, 2)}
 
 
-def choose_order(bound: set, consts: set = frozenset()) -> Tuple[int, int]:
+def choose_order(bound: set, consts: set = frozenset(),
+                 prefer_second: int = None) -> Tuple[int, int]:
     """Pick the order with the deepest bound prefix.
 
     Tie-break: prefer an order whose LEADING component is a constant — the
@@ -34,7 +35,7 @@ def choose_order(bound: set, consts: set = frozenset()) -> Tuple[int, int]:
 
     Returns (order_code, prefix_len in {0,1,2}).
     """
-    best, best_len, best_const = SPO, 0, False
+    best, best_len, best_const, best_pref = SPO, 0, False, False
     for code, (a, b, _c) in _ORDER_POS.items():
         ln = 0
         if a in bound:
@@ -42,8 +43,15 @@ def choose_order(bound: set, consts: set = frozenset()) -> Tuple[int, int]:
             if b in bound:
                 ln = 2
         lead_const = a in consts
-        if ln > best_len or (ln == best_len and lead_const and not best_const):
-            best, best_len, best_const = code, ln, lead_const
+        # a plen-1 slice is sorted by its SECOND component: honoring a
+        # sort hint makes downstream joins merge joins
+        pref = (prefer_second is not None and ln == 1 and b == prefer_second)
+        better = (ln > best_len
+                  or (ln == best_len and lead_const and not best_const)
+                  or (ln == best_len and lead_const == best_const
+                      and pref and not best_pref))
+        if better:
+            best, best_len, best_const, best_pref = code, ln, lead_const, pref
     return best, best_len
 
 
@@ -61,7 +69,8 @@ def _cols_from_order(idx: GraphIndex, code: int, sel: torch.Tensor
     return out[0], out[1], out[2]  # type: ignore
 
 
-def scan_unit(idx: GraphIndex, consts: Dict[int, int]
+def scan_unit(idx: GraphIndex, consts: Dict[int, int],
+              sort_hint: int = None
               ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
     """Scan with only constant bounds: one contiguous range slice.
 
@@ -71,7 +80,8 @@ def scan_unit(idx: GraphIndex, consts: Dict[int, int]
     if idx.n == 0:
         e = torch.empty(0, dtype=torch.int32, device=dev)
         return e, e.clone(), e.clone()
-    code, plen = choose_order(set(consts.keys()), set(consts.keys()))
+    code, plen = choose_order(set(consts.keys()), set(consts.keys()),
+                              prefer_second=sort_hint)
     pos = _ORDER_POS[code]
     key12, z = idx.orders[code]
     if plen == 0:

@@ -169,19 +169,55 @@ class Streamertail:
 
     def _plan_scan_group(self, scans: List[LScan], bound: Set[str]
                          ) -> Tuple[PhysicalOp, float, float]:
-        ordered = self._greedy_order(scans, bound)
-        scope = ordered[0].graph
-        star_var = self._detect_star(ordered) if len(ordered) >= 3 else None
-        if star_var is not None and not bound:
-            pats = [s.pattern for s in ordered]
-            est_rows = min(self.est.estimate_scan(p, set(), scope) for p in pats)
-            cost = sum(self.est.scan_cost(p, {star_var}, scope) for p in pats)
-            return PStarJoin(star_var, pats, scope), max(1.0, est_rows), cost
+        """Cost two candidate orders: selectivity-greedy (ref
+        greedy_order_scans:175) and star-first (subject-star merge chain on
+        PSO slices — the MI355X-native StarJoin: sorted probes measured 2.5x
+        faster than random).  Pick the cheaper chain."""
+        candidates = [self._greedy_order(scans, bound)]
+        star = self._star_subgroup(scans, bound)
+        if star is not None and star != candidates[0]:
+            candidates.append(star)
+        best = None
+        for ordered in candidates:
+            plan = self._build_chain(ordered, bound)
+            if best is None or plan[2] < best[2]:
+                best = plan
+        assert best is not None
+        return best
 
+    def _star_subgroup(self, scans: List[LScan], bound: Set[str]
+                       ) -> Optional[List[LScan]]:
+        """Order the largest subject-star subgroup (>=2 const-predicate
+        patterns sharing an unbound subject var) first, ascending by size."""
+        from collections import defaultdict
+        groups = defaultdict(list)
+        for s in scans:
+            if (isinstance(s.pattern.s, Variable)
+                    and isinstance(s.pattern.p, Constant)
+                    and s.pattern.s.name not in bound):
+                groups[s.pattern.s.name].append(s)
+        best_var = None
+        for var, ss in groups.items():
+            if len(ss) >= 2 and (best_var is None
+                                 or len(ss) > len(groups[best_var])):
+                best_var = var
+        if best_var is None:
+            return None
+        star = sorted(groups[best_var],
+                      key=lambda s: self.est.estimate_scan(s.pattern, set(), s.graph))
+        rest = [s for s in scans if s not in star]
+        cur_bound = set(bound)
+        for s in star:
+            cur_bound.update(_pattern_vars(s.pattern))
+        return star + (self._greedy_order(rest, cur_bound) if rest else [])
+
+    def _build_chain(self, ordered: List[LScan], bound: Set[str]
+                     ) -> Tuple[PhysicalOp, float, float]:
         cur: Optional[PhysicalOp] = None
         cur_rows = 1.0
         cur_cost = 0.0
         cur_bound = set(bound)
+        sorted_var: Optional[str] = None  # var the intermediate stays ordered by
         for s in ordered:
             scan_op = self._choose_scan(s.pattern, cur_bound, s.graph)
             rows_given_bound = self.est.estimate_scan(s.pattern, cur_bound, s.graph)
@@ -190,15 +226,30 @@ class Streamertail:
                 cur = scan_op
                 cur_rows = rows_free if not cur_bound else rows_given_bound
                 cur_cost = self.est.scan_cost(s.pattern, cur_bound, s.graph)
+                if (isinstance(s.pattern.s, Variable)
+                        and isinstance(s.pattern.p, Constant)
+                        and hasattr(scan_op, "sort_hint")):
+                    # PSO slice: output sorted by subject
+                    scan_op.sort_hint = 0
+                    sorted_var = s.pattern.s.name
                 cur_bound.update(_pattern_vars(s.pattern))
                 continue
             shared = [v for v in _pattern_vars(s.pattern) if v in cur_bound]
-            # candidates (ref find_best_plan_recursive:382).  GPU-era costs:
-            # bind join = K1 per-row index probe (~PROBE_FACTOR reads/row,
-            # upper index levels cache-resident); hash join = K2 build of the
-            # full right scan + probe; both emit the join output.
+            # candidates (ref find_best_plan_recursive:382).  GPU-calibrated:
+            # a K1 probe over sorted keys streams sequentially (~1.5/row);
+            # random probes into a small (cache-resident) predicate region
+            # ~2/row; random into a large region ~8/row; K2 hash join =
+            # build 3/row of the right scan + 2/row probe.
+            probed = [v for v in shared
+                      if isinstance(s.pattern.p, Constant) or True]
+            if sorted_var is not None and sorted_var in shared:
+                probe_factor = 1.5
+            elif rows_free <= 2_000_000:
+                probe_factor = 2.0
+            else:
+                probe_factor = PROBE_FACTOR
             emit_rows = cur_rows * max(1.0, rows_given_bound)
-            bind_cost = cur_cost + cur_rows * PROBE_FACTOR + emit_rows + 2000.0
+            bind_cost = cur_cost + cur_rows * probe_factor + emit_rows + 2000.0
             hash_cost = cur_cost + 3.0 * max(1.0, rows_free) \
                 + 2.0 * cur_rows + emit_rows + 6000.0
             nlj_cost = cur_cost + 10.0 * cur_rows * max(1.0, rows_free)
@@ -207,14 +258,17 @@ class Streamertail:
                 cur = PNestedLoopJoin(cur, scan_op)
                 cur_cost = nlj_cost
                 cur_rows = cur_rows * max(1.0, rows_free)
+                sorted_var = None
             elif mode == "bind" or (mode == "auto" and bind_cost <= hash_cost):
                 cur = PBindJoin(cur, scan_op)
                 cur_cost = bind_cost
                 cur_rows = emit_rows
+                # probe emit preserves the left chain's row order
             else:
                 cur = PHashJoin(cur, scan_op)
                 cur_cost = hash_cost
                 cur_rows = max(1.0, cur_rows * rows_given_bound)
+                sorted_var = None
             cur_bound.update(_pattern_vars(s.pattern))
         assert cur is not None
         return cur, cur_rows, cur_cost

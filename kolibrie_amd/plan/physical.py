@@ -28,12 +28,14 @@ class PUnit(PhysicalOp):
 class PTableScan(PhysicalOp):
     pattern: TriplePattern
     graph: GraphScope = None
+    sort_hint: Optional[int] = None  # position (0=s,1=p,2=o) to sort output by
 
 
 @dataclass
 class PIndexScan(PhysicalOp):
     pattern: TriplePattern
     graph: GraphScope = None
+    sort_hint: Optional[int] = None
 
 
 @dataclass
