@@ -64,58 +64,17 @@ __device__ __forceinline__ int64_t upper_bound_i64(const int64_t* __restrict__ a
 // exact mode: probe key = packed (a,b) per row; range mode: probe the full
 // range of leading component a (keys built from the int32 value).
 //
-// Block-narrowed search: each 256-thread tile first reduces its probe keys'
-// min/max (wave shuffles + LDS), binary-searches the index ONCE for that
-// window, then every thread searches only inside [win_lo, win_hi).  With
-// sorted probes (PSO-slice chains — the planner's star-merge order) the
-// window is a few hundred rows and stays L1-resident, turning the probe
-// into a merge join; with random probes the narrowing is a no-op costing
-// two extra searches per tile.
-__device__ __forceinline__ void tile_minmax(int64_t k, int64_t& mn, int64_t& mx,
-                                            int64_t* s_red) {
-  int64_t lo = k, hi = k;
-  for (int off = 32; off > 0; off >>= 1) {
-    lo = min(lo, __shfl_down(lo, off));
-    hi = max(hi, __shfl_down(hi, off));
-  }
-  int wave = threadIdx.x >> 6;
-  int lane = threadIdx.x & 63;
-  if (lane == 0) { s_red[wave * 2] = lo; s_red[wave * 2 + 1] = hi; }
-  __syncthreads();
-  int n_waves = blockDim.x >> 6;
-  mn = s_red[0]; mx = s_red[1];
-  for (int w = 1; w < n_waves; ++w) {
-    mn = min(mn, s_red[w * 2]);
-    mx = max(mx, s_red[w * 2 + 1]);
-  }
-  __syncthreads();  // self-contained: s_red reusable immediately after
-}
-
 __global__ void probe_count_exact(const int64_t* __restrict__ key12, int64_t n,
                                   const int64_t* __restrict__ keys, int64_t m,
                                   int64_t* __restrict__ lo_out,
                                   int32_t* __restrict__ cnt_out) {
-  __shared__ int64_t s_red[8 * 2];
-  __shared__ int64_t s_win[2];
-  for (int64_t base = (int64_t)blockIdx.x * blockDim.x; base < m;
-       base += (int64_t)gridDim.x * blockDim.x) {
-    int64_t i = base + threadIdx.x;
-    int64_t k = keys[i < m ? i : m - 1];
-    int64_t mn, mx;
-    tile_minmax(k, mn, mx, s_red);
-    if (threadIdx.x == 0) {
-      s_win[0] = lower_bound_i64(key12, n, mn);
-      s_win[1] = upper_bound_i64(key12 + s_win[0], n - s_win[0], mx) + s_win[0];
-    }
-    __syncthreads();
-    int64_t wlo = s_win[0], wspan = s_win[1] - s_win[0];
-    if (i < m) {
-      int64_t lo = wlo + lower_bound_i64(key12 + wlo, wspan, k);
-      int64_t hi = wlo + upper_bound_i64(key12 + wlo, wspan, k);
-      lo_out[i] = lo;
-      cnt_out[i] = static_cast<int32_t>(hi - lo);
-    }
-    __syncthreads();
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < m;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t k = keys[i];
+    int64_t lo = lower_bound_i64(key12, n, k);
+    int64_t hi = upper_bound_i64(key12, n, k);
+    lo_out[i] = lo;
+    cnt_out[i] = static_cast<int32_t>(hi - lo);
   }
 }
 
@@ -123,31 +82,56 @@ __global__ void probe_count_range(const int64_t* __restrict__ key12, int64_t n,
                                   const int32_t* __restrict__ vals, int64_t m,
                                   int64_t* __restrict__ lo_out,
                                   int32_t* __restrict__ cnt_out) {
-  __shared__ int64_t s_red[8 * 2];
-  __shared__ int64_t s_win[2];
-  for (int64_t base = (int64_t)blockIdx.x * blockDim.x; base < m;
-       base += (int64_t)gridDim.x * blockDim.x) {
-    int64_t i = base + threadIdx.x;
-    int64_t v = static_cast<int64_t>(vals[i < m ? i : m - 1]);
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < m;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t v = static_cast<int64_t>(vals[i]);
     int64_t klo = (v << 32);                       // (v, 0x00000000)
     int64_t khi = (v << 32) | 0xFFFFFFFFLL;        // (v, 0xFFFFFFFF)
-    int64_t mn, mx;
-    tile_minmax(klo, mn, mx, s_red);
-    int64_t mn2, mx2;
-    tile_minmax(khi, mn2, mx2, s_red);
-    if (threadIdx.x == 0) {
-      s_win[0] = lower_bound_i64(key12, n, mn);
-      s_win[1] = upper_bound_i64(key12 + s_win[0], n - s_win[0], mx2) + s_win[0];
-    }
-    __syncthreads();
-    int64_t wlo = s_win[0], wspan = s_win[1] - s_win[0];
-    if (i < m) {
-      int64_t lo = wlo + lower_bound_i64(key12 + wlo, wspan, klo);
-      int64_t hi = wlo + upper_bound_i64(key12 + wlo, wspan, khi);
-      lo_out[i] = lo;
-      cnt_out[i] = static_cast<int32_t>(hi - lo);
-    }
-    __syncthreads();
+    int64_t lo = lower_bound_i64(key12, n, klo);
+    int64_t hi = upper_bound_i64(key12, n, khi);
+    lo_out[i] = lo;
+    cnt_out[i] = static_cast<int32_t>(hi - lo);
+  }
+}
+
+// ---- merge-path variant for SORTED probes ---------------------------------
+// When the probe keys are themselves sorted (PSO-slice star/merge chains),
+// a fully-parallel boundary pass computes each 256-probe tile's index
+// window; the count pass then searches only inside its tile window, which
+// is small and cache-hot — the probe becomes a merge join.
+constexpr int kTile = 256;
+
+__global__ void tile_bounds(const int64_t* __restrict__ key12, int64_t n,
+                            const int64_t* __restrict__ keys, int64_t m,
+                            int64_t n_tiles,
+                            int64_t* __restrict__ win_lo,
+                            int64_t* __restrict__ win_hi) {
+  for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; t < n_tiles;
+       t += (int64_t)gridDim.x * blockDim.x) {
+    int64_t first = keys[t * kTile];
+    int64_t last = keys[min((t + 1) * kTile - 1, m - 1)];
+    win_lo[t] = lower_bound_i64(key12, n, first);
+    win_hi[t] = upper_bound_i64(key12, n, last);
+  }
+}
+
+__global__ void probe_count_exact_sorted(const int64_t* __restrict__ key12,
+                                         const int64_t* __restrict__ keys,
+                                         int64_t m,
+                                         const int64_t* __restrict__ win_lo,
+                                         const int64_t* __restrict__ win_hi,
+                                         int64_t* __restrict__ lo_out,
+                                         int32_t* __restrict__ cnt_out) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < m;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t t = i / kTile;
+    int64_t wlo = win_lo[t];
+    int64_t wspan = win_hi[t] - wlo;
+    int64_t k = keys[i];
+    int64_t lo = wlo + lower_bound_i64(key12 + wlo, wspan, k);
+    int64_t hi = wlo + upper_bound_i64(key12 + wlo, wspan, k);
+    lo_out[i] = lo;
+    cnt_out[i] = static_cast<int32_t>(hi - lo);
   }
 }
 
@@ -433,11 +417,34 @@ std::vector<at::Tensor> probe_exact(at::Tensor key12, at::Tensor z,
   auto cnt = at::empty({m}, keys.options().dtype(at::kInt));
   auto stream = cur_stream();
   if (m > 0) {
-    hipLaunchKernelGGL(probe_count_exact, dim3(grid_for(m)), dim3(kBlock), 0,
-                       stream, key12.data_ptr<int64_t>(), n,
-                       keys.data_ptr<int64_t>(), m, lo.data_ptr<int64_t>(),
-                       cnt.data_ptr<int32_t>());
-    HIP_OK(hipGetLastError());
+    bool sorted = false;
+    if (m >= 262144) {
+      // one cheap reduction decides the merge-path variant
+      sorted = at::all(keys.slice(0, 1, m) >= keys.slice(0, 0, m - 1))
+                   .item<bool>();
+    }
+    if (sorted) {
+      int64_t n_tiles = (m + kTile - 1) / kTile;
+      auto wlo = at::empty({n_tiles}, keys.options());
+      auto whi = at::empty({n_tiles}, keys.options());
+      hipLaunchKernelGGL(tile_bounds, dim3(grid_for(n_tiles)), dim3(kBlock), 0,
+                         stream, key12.data_ptr<int64_t>(), n,
+                         keys.data_ptr<int64_t>(), m, n_tiles,
+                         wlo.data_ptr<int64_t>(), whi.data_ptr<int64_t>());
+      HIP_OK(hipGetLastError());
+      hipLaunchKernelGGL(probe_count_exact_sorted, dim3(grid_for(m)),
+                         dim3(kBlock), 0, stream, key12.data_ptr<int64_t>(),
+                         keys.data_ptr<int64_t>(), m, wlo.data_ptr<int64_t>(),
+                         whi.data_ptr<int64_t>(), lo.data_ptr<int64_t>(),
+                         cnt.data_ptr<int32_t>());
+      HIP_OK(hipGetLastError());
+    } else {
+      hipLaunchKernelGGL(probe_count_exact, dim3(grid_for(m)), dim3(kBlock), 0,
+                         stream, key12.data_ptr<int64_t>(), n,
+                         keys.data_ptr<int64_t>(), m, lo.data_ptr<int64_t>(),
+                         cnt.data_ptr<int32_t>());
+      HIP_OK(hipGetLastError());
+    }
   }
   return emit_phase(key12, z, lo, cnt, m, stream);
 }
