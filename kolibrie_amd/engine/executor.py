@@ -205,10 +205,21 @@ class ExecutionEngine:
                       if not is_unit and inc.has(v)}
         exec_stats.bump("SCAN_PROBES", max(1, inc.n))
 
+        # lazy materialization: which (s,p,o) positions must the scan emit?
+        from collections import Counter
+        name_counts = Counter(var_pos.values())
+        scan_need = set(qt_pos.keys())
+        for pos_i, name in var_pos.items():
+            if name_counts[name] > 1 or needed is None or name in needed:
+                scan_need.add(pos_i)
+
         if not probe_vars:
-            s, p, o = scan_unit(idx, consts, sort_hint=sort_hint)
-            exec_stats.bump("QUADS_EXAMINED", s.numel())
-            cand = self._build_candidate(s, p, o, var_pos, qt_pos, None, needed)
+            s, p, o = scan_unit(idx, consts, sort_hint=sort_hint,
+                                need=scan_need)
+            n_sc = next((c.numel() for c in (s, p, o) if c is not None), 0)
+            exec_stats.bump("QUADS_EXAMINED", n_sc)
+            cand = self._build_candidate(s, p, o, var_pos, qt_pos, None,
+                                         needed, n_sc)
             if is_unit:
                 exec_stats.bump("ROWS_EMITTED", cand.n)
                 return self._add_extra(cand, extra)
@@ -236,12 +247,18 @@ class ExecutionEngine:
                 parts.append(join_bindings(sub2, cand, needed))
                 continue
             probes = {i: sub.col(v) for i, v in active.items()}
-            li, s, p, o = scan_probe(idx, consts, probes)
-            exec_stats.bump("QUADS_EXAMINED", s.numel())
-            # projection pushdown: gather only columns still needed
-            base_src = sub if needed is None else _prune(sub, set(needed) | set(active.values()))
+            li, s, p, o = scan_probe(idx, consts, probes, need=scan_need)
+            exec_stats.bump("QUADS_EXAMINED", li.numel())
+            # projection pushdown: gather only incoming columns the scan
+            # does not itself supply (pattern vars re-emerge from the index)
+            if needed is None:
+                base_src = sub
+            else:
+                pattern_names = set(var_pos.values())
+                base_src = _prune(sub, set(needed) - pattern_names)
             base = base_src.gather(li)
-            cand = self._build_candidate(s, p, o, var_pos, qt_pos, base, needed)
+            cand = self._build_candidate(s, p, o, var_pos, qt_pos, base,
+                                         needed, li.numel())
             parts.append(cand)
         out = Bindings.concat(parts, dev) if len(parts) != 1 else parts[0]
         exec_stats.bump("ROWS_EMITTED", out.n)
@@ -262,19 +279,24 @@ class ExecutionEngine:
 
     def _build_candidate(self, s, p, o, var_pos: Dict[int, str],
                          qt_pos: Dict[int, QuotedTriplePattern],
-                         base: Optional[Bindings], needed=None) -> Bindings:
+                         base: Optional[Bindings], needed=None,
+                         n: int = None) -> Bindings:
         """Assemble bindings from scanned triple columns: bind variables,
         enforce repeated-variable equality (engine.rs:1223-1239) and match
-        quoted-triple sub-patterns (engine.rs:1253)."""
+        quoted-triple sub-patterns (engine.rs:1253).  Columns may be None
+        when projection pushdown proved them unneeded."""
         dev = self.device
         cols_all = (s, p, o)
-        n = s.numel()
+        if n is None:
+            n = next((c.numel() for c in cols_all if c is not None), 0)
         out_cols: Dict[str, torch.Tensor] = {} if base is None else dict(base.cols)
         base_vars = set() if base is None else set(base.cols.keys())
         mask = torch.ones(n, dtype=torch.bool, device=dev)
         seen_here: set = set()
         for i, name in var_pos.items():
             col = cols_all[i]
+            if col is None:
+                continue  # pruned: not needed downstream
             if name in base_vars and name not in seen_here:
                 # probed position: equality enforced by scan_probe already
                 out_cols[name] = col

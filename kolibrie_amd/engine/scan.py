@@ -7,12 +7,16 @@ path vectorizes the probe as a batched searchsorted + repeat_interleave
 expansion — identical math to the kernel's per-wave binary search + two-pass
 emit, so it serves as the differential-test oracle.
 
+Column materialization is LAZY: `need` names the (s,p,o) positions the
+caller will actually read (projection pushdown reaches the kernels) — an
+unneeded position is returned as None and never unpacked from HBM.
+
 Ref parity: engine.rs:1018-1251 (execute_quad_scan_with_ids / match_quad),
 dataset_index.rs:223-344 (8-way bound-pattern dispatch).
 """
 from __future__ import annotations
 
-from typing import Dict, Optional, Tuple
+from typing import Dict, Optional, Sequence, Tuple
 
 import torch
 
@@ -23,15 +27,18 @@ from .tensor_utils import pack2, unpack2
 # order -> (leading col, second col, trailing col) position indices (0=s,1=p,2=o)
 _ORDER_POS = {SPO: (0, 1, 2), POS: (1, 2, 0), OSP: (2, 0, 1), PSO: (1, 0, 2)}
 
+ALL_POSITIONS = frozenset((0, 1, 2))
+
 
 def choose_order(bound: set, consts: set = frozenset(),
                  prefer_second: int = None) -> Tuple[int, int]:
     """Pick the order with the deepest bound prefix.
 
-    Tie-break: prefer an order whose LEADING component is a constant — the
-    binary search then converges into that predicate's contiguous region,
-    which stays cache-resident (measured 2.5x on MI355X vs probes spread
-    over the whole array).
+    Tie-breaks: (1) prefer an order whose LEADING component is a constant —
+    the binary search then converges into that predicate's contiguous
+    region, which stays cache-resident; (2) prefer the order whose SECOND
+    component matches `prefer_second` — a plen-1 slice is sorted by it, and
+    subject-sorted (PSO) slices make downstream joins merge joins.
 
     Returns (order_code, prefix_len in {0,1,2}).
     """
@@ -43,8 +50,6 @@ def choose_order(bound: set, consts: set = frozenset(),
             if b in bound:
                 ln = 2
         lead_const = a in consts
-        # a plen-1 slice is sorted by its SECOND component: honoring a
-        # sort hint makes downstream joins merge joins
         pref = (prefer_second is not None and ln == 1 and b == prefer_second)
         better = (ln > best_len
                   or (ln == best_len and lead_const and not best_const)
@@ -55,37 +60,46 @@ def choose_order(bound: set, consts: set = frozenset(),
     return best, best_len
 
 
-def _cols_from_order(idx: GraphIndex, code: int, sel: torch.Tensor
-                     ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
-    """Materialize (s,p,o) for selected row indices of an order."""
+def _cols_from_order(idx: GraphIndex, code: int, sel: Optional[torch.Tensor],
+                     need: frozenset = ALL_POSITIONS):
+    """Materialize the requested (s,p,o) positions for selected rows."""
     key12, z = idx.orders[code]
-    a, b = unpack2(key12[sel] if sel is not None else key12)
-    c = z[sel] if sel is not None else z
     pos = _ORDER_POS[code]
     out = [None, None, None]
-    out[pos[0]] = a
-    out[pos[1]] = b
-    out[pos[2]] = c
-    return out[0], out[1], out[2]  # type: ignore
+    k = key12[sel] if sel is not None else key12
+    if pos[0] in need and pos[1] in need:
+        a, b = unpack2(k)
+        out[pos[0]], out[pos[1]] = a, b
+    elif pos[0] in need:
+        out[pos[0]] = (k >> 32).to(torch.int32)
+    elif pos[1] in need:
+        out[pos[1]] = (k & 0xFFFFFFFF).to(torch.int32)
+    if pos[2] in need:
+        out[pos[2]] = z[sel] if sel is not None else z
+    return out[0], out[1], out[2]
 
 
 def scan_unit(idx: GraphIndex, consts: Dict[int, int],
-              sort_hint: int = None
-              ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+              sort_hint: int = None, need=None):
     """Scan with only constant bounds: one contiguous range slice.
 
-    consts: {position: i32 value}; returns (s,p,o) int32 columns.
+    consts: {position: i32 value}; returns (s,p,o) int32 columns (positions
+    outside `need` are None).
     """
     dev = idx.device
+    need_set = ALL_POSITIONS if need is None else frozenset(need)
     if idx.n == 0:
         e = torch.empty(0, dtype=torch.int32, device=dev)
-        return e, e.clone(), e.clone()
+        return tuple(e.clone() if i in need_set else None for i in range(3))
     code, plen = choose_order(set(consts.keys()), set(consts.keys()),
                               prefer_second=sort_hint)
     pos = _ORDER_POS[code]
     key12, z = idx.orders[code]
+    uncovered = set(consts.keys()) - set(pos[:plen])
+    mat_need = frozenset(need_set | uncovered)
     if plen == 0:
-        s, p, o = _cols_from_order(idx, SPO, None)
+        s, p, o = _cols_from_order(idx, SPO, None, mat_need)
+        n_rows = idx.n
     else:
         if plen == 2:
             k = pack2(
@@ -103,39 +117,51 @@ def scan_unit(idx: GraphIndex, consts: Dict[int, int],
             lo = int(torch.searchsorted(key12, klo, side="left").item())
             hi = int(torch.searchsorted(key12, khi, side="right").item())
         sel = torch.arange(lo, hi, dtype=torch.long, device=dev)
-        s, p, o = _cols_from_order(idx, code, sel)
-    # post-filter remaining constants not covered by the prefix
-    mask = None
-    for position, val in consts.items():
-        col = (s, p, o)[position]
-        m = col == val
-        mask = m if mask is None else (mask & m)
-    if mask is not None and plen < len(consts):
-        s, p, o = s[mask], p[mask], o[mask]
-    return s, p, o
+        s, p, o = _cols_from_order(idx, code, sel, mat_need)
+        n_rows = hi - lo
+    # post-filter constants not covered by the prefix
+    if uncovered:
+        mask = None
+        for position in uncovered:
+            col = (s, p, o)[position]
+            m = col == consts[position]
+            mask = m if mask is None else (mask & m)
+        cols = tuple(c[mask] if c is not None else None for c in (s, p, o))
+        s, p, o = cols
+    # drop post-filter-only columns the caller did not ask for
+    return tuple(
+        c if (i in need_set) else None for i, c in enumerate((s, p, o))
+    )
 
 
 def scan_probe(
     idx: GraphIndex,
     consts: Dict[int, int],
     probes: Dict[int, torch.Tensor],
-) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor, torch.Tensor]:
+    need=None,
+):
     """Per-row index probe (K1).  probes: {position: int32 value per row}.
 
     Returns (row_idx, s, p, o): row_idx indexes the probe rows; each probe
-    row expands to its matching triples.  Constants and probe positions not
-    covered by the chosen order's prefix are post-filtered.
+    row expands to its matching triples.  Positions outside `need` are None.
+    Constants and probe positions not covered by the chosen order's prefix
+    are post-filtered.
     """
     dev = idx.device
+    need_set = ALL_POSITIONS if need is None else frozenset(need)
     n_rows = next(iter(probes.values())).numel()
     e = torch.empty(0, dtype=torch.int32, device=dev)
     el = torch.empty(0, dtype=torch.long, device=dev)
     if idx.n == 0 or n_rows == 0:
-        return el, e, e.clone(), e.clone()
+        return (el,) + tuple(e.clone() if i in need_set else None
+                             for i in range(3))
     bound = set(consts.keys()) | set(probes.keys())
     code, plen = choose_order(bound, set(consts.keys()))
     pos = _ORDER_POS[code]
     key12, z = idx.orders[code]
+    covered = set(pos[:plen])
+    post_positions = bound - covered
+    mat_need = frozenset(need_set | post_positions)
 
     def col_for(position: int) -> torch.Tensor:
         if position in probes:
@@ -153,20 +179,24 @@ def scan_probe(
             li, b_col, z_col = native.probe_range(key12, z,
                                                   col_for(pos[0]).contiguous())
         out = [None, None, None]
-        out[pos[0]] = col_for(pos[0])[li]
-        out[pos[1]] = b_col
-        out[pos[2]] = z_col
-        s, p, o = out  # type: ignore[assignment]
-        covered = set(pos[:plen])
+        if pos[0] in mat_need:
+            out[pos[0]] = col_for(pos[0])[li]
+        if pos[1] in mat_need:
+            out[pos[1]] = b_col
+        if pos[2] in mat_need:
+            out[pos[2]] = z_col
+        s, p, o = out
         mask = None
-        for position in bound - covered:
+        for position in post_positions:
             col = (s, p, o)[position]
             want = consts[position] if position in consts else probes[position][li]
             m = col == want
             mask = m if mask is None else (mask & m)
         if mask is not None:
-            li, s, p, o = li[mask], s[mask], p[mask], o[mask]
-        return li, s, p, o
+            li = li[mask]
+            s, p, o = (c[mask] if c is not None else None for c in (s, p, o))
+        return (li,) + tuple(
+            c if i in need_set else None for i, c in enumerate((s, p, o)))
 
     if plen == 2:
         keys = pack2(col_for(pos[0]), col_for(pos[1]))
@@ -179,28 +209,27 @@ def scan_probe(
         lo = torch.searchsorted(key12, klo, side="left")
         hi = torch.searchsorted(key12, khi, side="right")
     else:
-        # no probe position matches any order prefix (cannot happen: probes
-        # non-empty means at least one position is bound)
         lo = torch.zeros(n_rows, dtype=torch.long, device=dev)
         hi = torch.full((n_rows,), idx.n, dtype=torch.long, device=dev)
     cnt = hi - lo
     total = int(cnt.sum().item())
     if total == 0:
-        return el, e, e.clone(), e.clone()
+        return (el,) + tuple(e.clone() if i in need_set else None
+                             for i in range(3))
     li = torch.repeat_interleave(
         torch.arange(n_rows, dtype=torch.long, device=dev), cnt)
     starts = torch.cumsum(cnt, 0) - cnt
     offs = torch.arange(total, dtype=torch.long, device=dev) - starts[li]
     sel = lo[li] + offs
-    s, p, o = _cols_from_order(idx, code, sel)
-    # post-filter positions not in the prefix
-    covered = set(pos[:plen])
+    s, p, o = _cols_from_order(idx, code, sel, mat_need)
     mask = None
-    for position in bound - covered:
+    for position in post_positions:
         col = (s, p, o)[position]
         want = consts[position] if position in consts else probes[position][li]
         m = col == want
         mask = m if mask is None else (mask & m)
     if mask is not None:
-        li, s, p, o = li[mask], s[mask], p[mask], o[mask]
-    return li, s, p, o
+        li = li[mask]
+        s, p, o = (c[mask] if c is not None else None for c in (s, p, o))
+    return (li,) + tuple(
+        c if i in need_set else None for i, c in enumerate((s, p, o)))

@@ -245,13 +245,17 @@ class Streamertail:
             if sorted_var is not None and sorted_var in shared:
                 probe_factor = 1.5
             elif rows_free <= 2_000_000:
-                probe_factor = 2.0
+                probe_factor = 4.0
             else:
                 probe_factor = PROBE_FACTOR
             emit_rows = cur_rows * max(1.0, rows_given_bound)
             bind_cost = cur_cost + cur_rows * probe_factor + emit_rows + 2000.0
+            # K2 hash join: build 3/row of the right scan, O(1) cache-hot
+            # probe 1/row of the left (or a sorted-slice merge join which
+            # costs about the same) — beats dependent binary searches when
+            # the build side is small
             hash_cost = cur_cost + 3.0 * max(1.0, rows_free) \
-                + 2.0 * cur_rows + emit_rows + 6000.0
+                + 1.0 * cur_rows + emit_rows + 6000.0
             nlj_cost = cur_cost + 10.0 * cur_rows * max(1.0, rows_free)
             mode = _join_mode()
             if not shared:
@@ -279,13 +283,14 @@ class Streamertail:
         0 bound -> TableScan (ref choose_best_scan:702)."""
         sb, pb, ob = self.est.bound_positions(pattern, bound)
         n_bound = int(sb) + int(pb) + int(ob)
-        if n_bound >= 2:
-            return PIndexScan(pattern, graph)
-        if n_bound == 1:
-            if self.est.estimate_scan(pattern, bound, graph) < 10_000:
-                return PIndexScan(pattern, graph)
-            return PIndexScan(pattern, graph)
-        return PTableScan(pattern, graph)
+        # subject-sorted (PSO) output whenever the predicate is constant and
+        # the subject free: costs nothing, enables downstream merge joins
+        hint = 0 if (isinstance(pattern.p, Constant)
+                     and isinstance(pattern.s, Variable)
+                     and pattern.s.name not in bound) else None
+        if n_bound >= 1:
+            return PIndexScan(pattern, graph, sort_hint=hint)
+        return PTableScan(pattern, graph, sort_hint=hint)
 
     def _out_vars(self, op: LogicalOp, bound: Set[str]) -> Set[str]:
         return _logical_out_vars(op)
