@@ -247,6 +247,14 @@ class ExecutionEngine:
                 parts.append(join_bindings(sub2, cand, needed))
                 continue
             probes = {i: sub.col(v) for i, v in active.items()}
+            if (needed is not None and not needed and not scan_need
+                    and not qt_pos and extra is None):
+                # COUNT(*) pushdown: no columns needed — count matches only
+                from .scan import scan_probe_count
+                cnt = scan_probe_count(idx, consts, probes)
+                if cnt is not None:
+                    parts.append(Bindings({}, cnt, dev))
+                    continue
             li, s, p, o = scan_probe(idx, consts, probes, need=scan_need)
             exec_stats.bump("QUADS_EXAMINED", li.numel())
             # projection pushdown: gather only incoming columns the scan
@@ -400,7 +408,12 @@ def join_bindings(left: Bindings, right: Bindings, needed=None) -> Bindings:
     parts: List[Bindings] = []
     l_keyed, l_unkeyed = left.select(lb), left.select(~lb)
     r_keyed, r_unkeyed = right.select(rb), right.select(~rb)
-    if l_keyed.n and r_keyed.n:
+    count_only = needed is not None and len(needed) == 0
+    if count_only and l_keyed.n and r_keyed.n:
+        # COUNT(*) pushdown: match counts without emitting pairs
+        n_match = _join_count(l_keyed, r_keyed, shared)
+        parts.append(Bindings({}, n_match, dev))
+    elif l_keyed.n and r_keyed.n:
         from ..ops import native_for
         native = native_for(l_keyed.col(shared[0])) if len(shared) <= 4 else None
         merged = None
@@ -451,6 +464,33 @@ def join_bindings(left: Bindings, right: Bindings, needed=None) -> Bindings:
         vars_ = list(dict.fromkeys(left.variables + right.variables))
         return Bindings.empty(dev, vars_)
     return Bindings.concat(parts, dev) if len(parts) > 1 else parts[0]
+
+
+def _join_count(l_keyed: Bindings, r_keyed: Bindings,
+                shared: Sequence[str]) -> int:
+    dev = l_keyed.device
+    if len(shared) == 1:
+        lk = l_keyed.col(shared[0])
+        rk = r_keyed.col(shared[0])
+        if _is_sorted(rk):
+            lo = torch.searchsorted(rk, lk, side="left")
+            hi = torch.searchsorted(rk, lk, side="right")
+            return int((hi - lo).sum().item())
+        if _is_sorted(lk):
+            lo = torch.searchsorted(lk, rk, side="left")
+            hi = torch.searchsorted(lk, rk, side="right")
+            return int((hi - lo).sum().item())
+    from ..ops import native_for
+    native = native_for(l_keyed.col(shared[0])) if len(shared) <= 4 else None
+    if native is not None:
+        cnt = native.hash_join_counts(
+            [l_keyed.col(v).contiguous() for v in shared],
+            [r_keyed.col(v).contiguous() for v in shared])
+        return int(cnt.to(torch.int64).sum().item())
+    key_cols = [torch.cat([l_keyed.col(v), r_keyed.col(v)]) for v in shared]
+    gid, _ = group_index(key_cols)
+    li, _ri = merge_join_indices(gid[:l_keyed.n], gid[l_keyed.n:])
+    return int(li.numel())
 
 
 def _is_sorted(col: torch.Tensor) -> bool:

@@ -39,6 +39,11 @@ def _top_needed(select: SelectQuery):
     None = keep everything (SELECT *)."""
     if select.select_star or not select.variables:
         return None
+    if (not select.distinct and not select.group_by and not select.order_by
+            and select.variables
+            and all(p.aggregate == "COUNT" and p.agg_arg is None
+                    and not p.distinct for p in select.variables)):
+        return set()  # COUNT(*) only: no columns needed at all
     needed = set(select.group_by)
     needed.update(c.var for c in select.order_by)
     for p in select.variables:

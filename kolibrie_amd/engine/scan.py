@@ -233,3 +233,46 @@ def scan_probe(
         s, p, o = (c[mask] if c is not None else None for c in (s, p, o))
     return (li,) + tuple(
         c if i in need_set else None for i, c in enumerate((s, p, o)))
+
+
+def scan_probe_count(idx: GraphIndex, consts: Dict[int, int],
+                     probes: Dict[int, torch.Tensor]) -> Optional[int]:
+    """COUNT(*)-only probe: total match count with no emit pass.
+
+    Returns None when a post-filter position would be required (caller
+    falls back to the emitting path)."""
+    n_rows = next(iter(probes.values())).numel()
+    if idx.n == 0 or n_rows == 0:
+        return 0
+    bound = set(consts.keys()) | set(probes.keys())
+    code, plen = choose_order(bound, set(consts.keys()))
+    pos = _ORDER_POS[code]
+    if plen == 0 or (bound - set(pos[:plen])):
+        return None
+    key12, _z = idx.orders[code]
+
+    def col_for(position: int) -> torch.Tensor:
+        if position in probes:
+            return probes[position]
+        return torch.full((n_rows,), consts[position], dtype=torch.int32,
+                          device=idx.device)
+
+    from ..ops import native_for
+    native = native_for(key12)
+    if native is not None:
+        if plen == 2:
+            keys = pack2(col_for(pos[0]).contiguous(),
+                         col_for(pos[1]).contiguous())
+            cnt = native.probe_exact_counts(key12, keys.contiguous())
+        else:
+            cnt = native.probe_range_counts(key12, col_for(pos[0]).contiguous())
+        return int(cnt.sum().item())
+    if plen == 2:
+        keys = pack2(col_for(pos[0]), col_for(pos[1]))
+        lo = torch.searchsorted(key12, keys, side="left")
+        hi = torch.searchsorted(key12, keys, side="right")
+    else:
+        v = col_for(pos[0])
+        lo = torch.searchsorted(key12, pack2(v, torch.zeros_like(v)), side="left")
+        hi = torch.searchsorted(key12, pack2(v, torch.full_like(v, -1)), side="right")
+    return int((hi - lo).sum().item())

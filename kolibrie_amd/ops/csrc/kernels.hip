@@ -469,6 +469,72 @@ std::vector<at::Tensor> probe_range(at::Tensor key12, at::Tensor z,
   return emit_phase(key12, z, lo, cnt, m, stream);
 }
 
+// K1 count-only variants: COUNT(*) queries skip the emit pass entirely.
+at::Tensor probe_exact_counts(at::Tensor key12, at::Tensor keys) {
+  TORCH_CHECK(key12.is_cuda() && keys.is_cuda());
+  auto m = keys.numel();
+  auto n = key12.numel();
+  auto lo = at::empty({m}, keys.options());
+  auto cnt = at::empty({m}, keys.options().dtype(at::kInt));
+  if (m > 0) {
+    hipLaunchKernelGGL(probe_count_exact, dim3(grid_for(m)), dim3(kBlock), 0,
+                       cur_stream(), key12.data_ptr<int64_t>(), n,
+                       keys.data_ptr<int64_t>(), m, lo.data_ptr<int64_t>(),
+                       cnt.data_ptr<int32_t>());
+    HIP_OK(hipGetLastError());
+  }
+  return cnt;
+}
+
+at::Tensor probe_range_counts(at::Tensor key12, at::Tensor vals) {
+  TORCH_CHECK(key12.is_cuda() && vals.is_cuda());
+  auto m = vals.numel();
+  auto n = key12.numel();
+  auto lo = at::empty({m}, key12.options());
+  auto cnt = at::empty({m}, vals.options());
+  if (m > 0) {
+    hipLaunchKernelGGL(probe_count_range, dim3(grid_for(m)), dim3(kBlock), 0,
+                       cur_stream(), key12.data_ptr<int64_t>(), n,
+                       vals.data_ptr<int32_t>(), m, lo.data_ptr<int64_t>(),
+                       cnt.data_ptr<int32_t>());
+    HIP_OK(hipGetLastError());
+  }
+  return cnt;
+}
+
+// K2 hash join, count-only: per-probe-row match counts (no emit).
+at::Tensor hash_join_counts(std::vector<at::Tensor> left_cols,
+                            std::vector<at::Tensor> right_cols) {
+  TORCH_CHECK(!left_cols.empty() && left_cols.size() == right_cols.size());
+  TORCH_CHECK(left_cols.size() <= kMaxKeyCols);
+  int64_t l = left_cols[0].numel();
+  int64_t r = right_cols[0].numel();
+  auto opts_int = left_cols[0].options().dtype(at::kInt);
+  if (l == 0 || r == 0) return at::zeros({l}, opts_int);
+  KeyCols probe{}, build{};
+  probe.k = build.k = static_cast<int>(left_cols.size());
+  for (size_t j = 0; j < left_cols.size(); ++j) {
+    probe.c[j] = left_cols[j].data_ptr<int32_t>();
+    build.c[j] = right_cols[j].data_ptr<int32_t>();
+  }
+  uint64_t h = 1;
+  while (h < static_cast<uint64_t>(2 * r)) h <<= 1;
+  uint32_t mask = static_cast<uint32_t>(h - 1);
+  auto heads = at::full({static_cast<int64_t>(h)}, -1, opts_int);
+  auto next = at::empty({r}, opts_int);
+  auto stream = cur_stream();
+  hipLaunchKernelGGL(hj_build, dim3(grid_for(r)), dim3(kBlock), 0, stream,
+                     build, r, mask, heads.data_ptr<int32_t>(),
+                     next.data_ptr<int32_t>());
+  HIP_OK(hipGetLastError());
+  auto cnt = at::empty({l}, opts_int);
+  hipLaunchKernelGGL(hj_count, dim3(grid_for(l)), dim3(kBlock), 0, stream,
+                     probe, l, build, heads.data_ptr<int32_t>(),
+                     next.data_ptr<int32_t>(), mask, cnt.data_ptr<int32_t>());
+  HIP_OK(hipGetLastError());
+  return cnt;
+}
+
 // K2 hash join: returns (li, ri) index pairs; multiset semantics.
 std::vector<at::Tensor> hash_join(std::vector<at::Tensor> left_cols,
                                   std::vector<at::Tensor> right_cols) {
@@ -558,6 +624,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "K1 scan-probe, leading-component range -> (li, b, z)");
   m.def("hash_join", &hash_join,
         "K2 chained hash join over int32 key columns -> (li, ri)");
+  m.def("probe_exact_counts", &probe_exact_counts,
+        "K1 count-only exact probe -> per-row match counts");
+  m.def("probe_range_counts", &probe_range_counts,
+        "K1 count-only range probe -> per-row match counts");
+  m.def("hash_join_counts", &hash_join_counts,
+        "K2 count-only hash join -> per-left-row match counts");
   m.def("filter_bytecode", &filter_bytecode,
         "K5 filter bytecode evaluation -> bool mask");
 }

@@ -122,7 +122,7 @@ def generate_partition(ds: EmployeeDataset, rank: int, world: int, seed: int,
 
 FLAGSHIP_QUERY = f"""
 PREFIX ds: <{DS}>
-SELECT (COUNT(?e) AS ?c) WHERE {{
+SELECT (COUNT(*) AS ?c) WHERE {{
     ?e ds:worksFor ?d .
     ?e ds:annual_salary ?sal .
     ?d ds:locatedIn ?city .
