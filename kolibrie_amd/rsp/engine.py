@@ -1,0 +1,343 @@
+"""RSPEngine — streaming query orchestration.
+
+Ref parity: kolibrie/src/rsp_engine.rs (1 266 LoC):
+  - per-window processor (create_window_processor:105-212): evict the
+    previous firing's triples from the R2R store, add the new content,
+    materialize(), run the per-window physical plan, hand results to the
+    cross-window joiner or directly to R2S;
+  - SingleThread = window callbacks, MultiThread = one consumer thread per
+    window + a cross-window coordinator (register_window!:215-236,
+    start_cross_window_coordinator:539-770);
+  - SyncPolicy Wait / Steal / Timeout coordination of multi-window joins
+    (process_single_thread_window_results:812-887);
+  - emission (emit_results:1012): natural join across window results
+    (join_window_results:1089) plus static-plan bindings (:1052-1086),
+    then the R2S operator;
+  - stream routing with IRI normalization, '?var' matches all (:773-810);
+  - static data shares the dictionary and never enters windows (:321-326,
+    add_static_ntriples:923);
+  - cross-window SDS+ mode over raw (triple, ts) per window (:1122-1213).
+"""
+from __future__ import annotations
+
+import threading
+import time as _time
+from dataclasses import dataclass, field
+from queue import Empty, Queue
+from typing import Callable, Dict, List, Optional, Tuple
+
+from ..parsing.ast import RegisterClause, SelectQuery, SyncPolicy
+from .r2r import SimpleR2R
+from .r2s import Relation2StreamOperator, StreamOperator
+from .s2r import CSPARQLWindow, ContentContainer, Report, ReportStrategy, Tick
+
+
+class OperationMode:
+    SINGLE_THREAD = "SingleThread"
+    MULTI_THREAD = "MultiThread"
+
+
+class QueryExecutionMode:
+    STANDARD = "Standard"
+    VOLCANO = "Volcano"
+
+
+class CrossWindowReasoningMode:
+    NAIVE = "Naive"
+    INCREMENTAL = "Incremental"
+
+
+def normalize_iri(iri: str) -> str:
+    iri = iri.strip()
+    if iri.startswith("<") and iri.endswith(">"):
+        iri = iri[1:-1]
+    return iri
+
+
+@dataclass
+class WindowResult:
+    window_iri: str
+    rows: List[Tuple[str, ...]]
+    variables: List[str]
+    ts: int
+
+
+@dataclass
+class _WindowEntry:
+    iri: str
+    stream_iri: str
+    window: CSPARQLWindow
+    plan: Optional[tuple]          # (SelectQuery, PhysicalOp)
+    plan_vars: List[str]
+    thread: Optional[threading.Thread] = None
+    queue: Optional[Queue] = None
+
+
+class RSPEngine:
+    def __init__(
+        self,
+        register: Optional[RegisterClause] = None,
+        device: str = "cpu",
+        operation_mode: str = OperationMode.SINGLE_THREAD,
+        sync_policy: Optional[SyncPolicy] = None,
+        cross_window_mode: Optional[str] = None,
+        hybrid_config=None,
+    ):
+        self.device = device
+        self.register_clause = register
+        self.operation_mode = operation_mode
+        self.sync_policy = sync_policy or SyncPolicy("Wait")
+        self.cross_window_mode = cross_window_mode
+        self.store = SimpleR2R(device=device, hybrid_config=hybrid_config)
+        self.static_db = None       # separate DB sharing the dictionary
+        self.windows: Dict[str, _WindowEntry] = {}
+        self.r2s = Relation2StreamOperator(
+            register.stream_type if register else StreamOperator.RSTREAM)
+        self.consumers: List[Callable[[List[Tuple]], None]] = []
+        self.projection: Optional[SelectQuery] = None
+        self._pending_results: Dict[str, WindowResult] = {}
+        self._raw_by_window: Dict[str, List[Tuple[Tuple[int, int, int], int]]] = {}
+        self._window_triples: Dict[str, List[Tuple[int, int, int]]] = {}
+        self._coord_lock = threading.Lock()
+        self._result_queue: Queue = Queue()
+        self._stopped = False
+        self.latest_hybrid_results: Dict = {}
+        self.cross_window_rules = []   # (Rule, window ctx) for SDS+
+        self._sds_window_widths: Dict[str, int] = {}
+
+    # --------------------------------------------------------------- setup
+    def add_window(self, window_iri: str, stream_iri: str, width: int,
+                   slide: int, report: Optional[str] = None,
+                   tick: Optional[str] = None, plan: Optional[tuple] = None,
+                   plan_vars: Optional[List[str]] = None):
+        rep = Report()
+        rep.add(report or ReportStrategy.ON_WINDOW_CLOSE)
+        w = CSPARQLWindow(width, slide, rep, tick or Tick.TIME_DRIVEN,
+                          normalize_iri(window_iri))
+        entry = _WindowEntry(
+            iri=normalize_iri(window_iri), stream_iri=normalize_iri(stream_iri),
+            window=w, plan=plan, plan_vars=plan_vars or [])
+        self.windows[entry.iri] = entry
+        self._window_triples[entry.iri] = []
+        self._raw_by_window[entry.iri] = []
+        if self.operation_mode == OperationMode.MULTI_THREAD:
+            q = w.register()
+            entry.queue = q
+            t = threading.Thread(
+                target=self._window_worker, args=(entry,), daemon=True)
+            entry.thread = t
+            t.start()
+        else:
+            w.register_callback(
+                lambda content, e=entry: self._process_window(e, content))
+        return entry
+
+    def add_consumer(self, fn: Callable[[List[Tuple]], None]):
+        self.consumers.append(fn)
+
+    def load_static_ntriples(self, text: str):
+        """Static triples share the dictionary, never enter windows
+        (ref rsp_engine.rs:923 add_static_ntriples)."""
+        from ..storage.database import SparqlDatabase
+        if self.static_db is None:
+            self.static_db = SparqlDatabase(device=self.device)
+            self.static_db.dictionary = self.store.db.dictionary
+            self.static_db.quoted_triples = self.store.db.quoted_triples
+        self.static_db.parse_ntriples(text)
+
+    add_static_ntriples = load_static_ntriples
+
+    # ------------------------------------------------------------ ingestion
+    def add_to_stream(self, stream_iri: str, triple, ts: int):
+        """Route an event to all windows listening on the stream
+        (ref rsp_engine.rs:773-810; '?var' stream matches all)."""
+        stream_iri = normalize_iri(stream_iri)
+        item = self.store._encode(triple)
+        for entry in self.windows.values():
+            if (entry.stream_iri == stream_iri
+                    or entry.stream_iri.startswith("?")
+                    or stream_iri.startswith("?")):
+                self._raw_by_window[entry.iri].append((item, ts))
+                entry.window.add_to_window(item, ts)
+
+    def add_probabilistic_to_stream(self, stream_iri: str, triple, ts: int,
+                                    seed_id: str, probability: float):
+        """Probabilistic event: seeds the hybrid registry
+        (ref rsp_engine.rs:960)."""
+        from .s2r import EventKey, ProbabilisticOccurrence
+        stream_iri = normalize_iri(stream_iri)
+        item = self.store._encode(triple)
+        self.store.db.probability_seeds[item] = probability
+        occ = ProbabilisticOccurrence(
+            item=item, event=EventKey(stream_iri, ts), seed_id=seed_id)
+        for entry in self.windows.values():
+            if entry.stream_iri == stream_iri or entry.stream_iri.startswith("?"):
+                entry.window.add_probabilistic_to_window(occ)
+
+    def flush_windows(self):
+        for entry in self.windows.values():
+            entry.window.flush()
+        if self.operation_mode == OperationMode.MULTI_THREAD:
+            self._drain_multithread(deadline_ms=2000)
+
+    # ----------------------------------------------------- window processing
+    def _window_worker(self, entry: _WindowEntry):
+        while not self._stopped:
+            try:
+                content = entry.queue.get(timeout=0.1)
+            except Empty:
+                continue
+            self._process_window(entry, content)
+
+    def _process_window(self, entry: _WindowEntry, content: ContentContainer):
+        """Per-firing processor (ref create_window_processor:105-212)."""
+        with self._coord_lock:
+            # evict previous firing's triples, add the new content
+            for t in self._window_triples[entry.iri]:
+                self.store.remove(t)
+            triples = list(content.items())
+            self._window_triples[entry.iri] = triples
+            for t in triples:
+                self.store.add(t, content.last_timestamp_changed)
+            if self.cross_window_mode is not None:
+                self._emit_cross_window(content.last_timestamp_changed)
+                return
+            self.store.materialize()
+            self.latest_hybrid_results = self.store.latest_hybrid_results
+            if entry.plan is None:
+                rows = [tuple(self.store.db.decode_term(x) for x in t)
+                        for t in triples]
+                result = WindowResult(entry.iri, rows, ["s", "p", "o"],
+                                      content.last_timestamp_changed)
+            else:
+                rows = self.store.execute_query(entry.plan)
+                result = WindowResult(entry.iri, rows, entry.plan_vars,
+                                      content.last_timestamp_changed)
+            if len(self.windows) > 1:
+                self._pending_results[entry.iri] = result
+                self._try_emit_joined(result.ts)
+            else:
+                self._emit([result], result.ts)
+
+    # -------------------------------------------------- multi-window joining
+    def _try_emit_joined(self, ts: int):
+        """SyncPolicy coordination (ref :812-887; Timeout degrades to Wait
+        in SingleThread mode)."""
+        policy = self.sync_policy.kind
+        have = set(self._pending_results.keys())
+        want = set(self.windows.keys())
+        if policy in ("Wait", "Timeout"):
+            if have != want:
+                return
+        elif policy == "Steal":
+            if not have:
+                return
+            # use stale/absent results for missing windows: missing -> empty
+        results = [self._pending_results[w] for w in sorted(have)]
+        if policy in ("Wait", "Timeout"):
+            self._pending_results = {}
+        self._emit(results, ts)
+
+    def _drain_multithread(self, deadline_ms: int):
+        t0 = _time.time()
+        while any(e.queue is not None and not e.queue.empty()
+                  for e in self.windows.values()):
+            if (_time.time() - t0) * 1000 > deadline_ms:
+                break
+            _time.sleep(0.005)
+
+    # -------------------------------------------------------------- emission
+    def _emit(self, results: List[WindowResult], ts: int):
+        """Natural join across window results + static bindings -> R2S ->
+        consumers (ref emit_results:1012, join_window_results:1089)."""
+        joined_rows, joined_vars = _natural_join_results(results)
+        if self.static_db is not None and self.register_clause is not None:
+            joined_rows, joined_vars = self._join_static(joined_rows, joined_vars)
+        if self.projection is not None and self.projection.variables \
+                and not self.projection.select_star:
+            names = [p.output_name() for p in self.projection.variables]
+            idx = [joined_vars.index(v) if v in joined_vars else None
+                   for v in names]
+            joined_rows = [tuple("" if i is None else r[i] for i in idx)
+                           for r in joined_rows]
+            joined_vars = names
+        out = self.r2s.eval(joined_rows, ts)
+        for fn in self.consumers:
+            fn(out)
+        return out
+
+    def _join_static(self, rows, vars_):
+        """Join window results with the static-data plan (ref :1052-1086)."""
+        if self.register_clause is None or self.static_db is None:
+            return rows, vars_
+        static_patterns = getattr(self, "_static_select", None)
+        if static_patterns is None:
+            return rows, vars_
+        static_rows = self.static_db.query(static_patterns)
+        from ..parsing.sparql import parse_combined_query
+        cq = parse_combined_query(static_patterns)
+        svars = [p.output_name() for p in cq.select.variables] \
+            if cq.select and cq.select.variables else []
+        shared = [v for v in vars_ if v in svars]
+        if not shared:
+            return rows, vars_
+        out_vars = vars_ + [v for v in svars if v not in vars_]
+        out_rows = []
+        for r in rows:
+            rd = dict(zip(vars_, r))
+            for srow in static_rows:
+                sd = dict(zip(svars, srow))
+                if all(rd.get(v) == sd.get(v) for v in shared):
+                    merged = dict(sd)
+                    merged.update(rd)
+                    out_rows.append(tuple(merged.get(v, "") for v in out_vars))
+        return out_rows, out_vars
+
+    # ------------------------------------------------------ cross-window SDS
+    def _emit_cross_window(self, ts: int):
+        """SDS+ over latest raw (triple, ts) per window (ref :1122-1213)."""
+        from ..reasoning.sds import Sds, WindowedTriple, naive_sds_plus, incremental_sds_plus
+        sds = Sds()
+        for wiri, raws in self._raw_by_window.items():
+            width = self._sds_window_widths.get(
+                wiri, self.windows[wiri].window.width)
+            for (item, ets) in raws:
+                sds.add(WindowedTriple(wiri, item, ets), width)
+        if self.cross_window_mode == CrossWindowReasoningMode.NAIVE:
+            inferred = naive_sds_plus(sds, self.cross_window_rules,
+                                      self.store.db, ts)
+        else:
+            inferred = incremental_sds_plus(sds, self.cross_window_rules,
+                                            self.store.db, ts)
+        rows = [tuple(self.store.db.decode_term(x) for x in t)
+                for t in sorted(inferred)]
+        out = self.r2s.eval(rows, ts)
+        for fn in self.consumers:
+            fn(out)
+        return out
+
+    def stop(self):
+        self._stopped = True
+
+
+def _natural_join_results(results: List[WindowResult]
+                          ) -> Tuple[List[Tuple], List[str]]:
+    """Natural join of window result tables on shared variables
+    (ref rsp_engine.rs:1089 join_window_results + :1052 natural_join)."""
+    if not results:
+        return [], []
+    rows = [dict(zip(results[0].variables, r)) for r in results[0].rows]
+    vars_ = list(results[0].variables)
+    for res in results[1:]:
+        shared = [v for v in vars_ if v in res.variables]
+        new_rows = []
+        right = [dict(zip(res.variables, r)) for r in res.rows]
+        for lr in rows:
+            for rr in right:
+                if all(lr.get(v) == rr.get(v) for v in shared):
+                    merged = dict(rr)
+                    merged.update(lr)
+                    new_rows.append(merged)
+        rows = new_rows
+        vars_ = vars_ + [v for v in res.variables if v not in vars_]
+    return [tuple(r.get(v, "") for v in vars_) for r in rows], vars_

@@ -1,0 +1,232 @@
+"""RSP streaming tests (mirrors kolibrie/tests/rsp_engine_test.rs, 26 tests,
+and s2r.rs inline window-firing tests :422-497).  Streams are simulated by
+add_to_stream with synthetic timestamps — logical time, deterministic."""
+import pytest
+
+from kolibrie_amd.rsp import (
+    CSPARQLWindow, ContentContainer, Report, ReportStrategy, RSPBuilder,
+    Relation2StreamOperator, StreamOperator, Tick,
+)
+
+EX = "http://example.org/"
+
+
+# ------------------------------------------------------------------- windows
+def _mk_window(width, slide, strategy=ReportStrategy.ON_WINDOW_CLOSE):
+    r = Report()
+    r.add(strategy)
+    return CSPARQLWindow(width, slide, r, Tick.TIME_DRIVEN, "w")
+
+
+def test_window_firing_counts():
+    w = _mk_window(10, 5)
+    fired = []
+    w.register_callback(lambda c: fired.append(sorted(c.items())))
+    for ts in range(0, 30):
+        w.add_to_window(("s", "p", f"o{ts}"), ts)
+    # [0,5) fires at 5 (negative-origin saturation, ref s2r.rs `as usize`),
+    # then [0,10) at 10, [5,15) at 15, [10,20) at 20, [15,25) at 25
+    assert len(fired) == 5
+    assert fired[0] == [("s", "p", f"o{i}") for i in range(5)]
+    second = fired[1]
+    assert ("s", "p", "o0") in second and ("s", "p", "o9") in second
+    assert ("s", "p", "o10") not in second
+
+
+def test_window_content_multiset_last_ts():
+    c = ContentContainer()
+    c.add(("a",), 1)
+    c.add(("a",), 5)
+    assert len(c) == 1
+    assert dict(c.iter_with_timestamps())[("a",)] == 5
+
+
+def test_window_channel_consumer():
+    w = _mk_window(4, 4)
+    q = w.register()
+    for ts in range(0, 9):
+        w.add_to_window(("e", ts), ts)
+    contents = []
+    while not q.empty():
+        contents.append(q.get())
+    assert len(contents) == 2
+
+
+def test_window_flush():
+    w = _mk_window(100, 100)
+    fired = []
+    w.register_callback(lambda c: fired.append(c))
+    w.add_to_window(("x",), 1)
+    w.add_to_window(("y",), 2)
+    assert not fired
+    w.flush()
+    assert len(fired) == 1
+    assert len(fired[0]) == 2
+
+
+# ----------------------------------------------------------------------- r2s
+def test_rstream_istream_dstream():
+    r = Relation2StreamOperator(StreamOperator.ISTREAM)
+    assert r.eval([(1,), (2,)], 0) == [(1,), (2,)]
+    assert r.eval([(2,), (3,)], 1) == [(3,)]
+    d = Relation2StreamOperator(StreamOperator.DSTREAM)
+    assert d.eval([(1,), (2,)], 0) == []
+    assert d.eval([(2,)], 1) == [(1,)]
+    rs = Relation2StreamOperator(StreamOperator.RSTREAM)
+    assert rs.eval([(1,)], 0) == [(1,)]
+    assert rs.eval([(1,)], 1) == [(1,)]
+
+
+# -------------------------------------------------------------------- engine
+def _simple_engine(stream_type="RSTREAM", **kw):
+    q = f"""
+        PREFIX ex: <{EX}>
+        REGISTER {stream_type} <http://out> AS
+        SELECT ?s ?o
+        FROM NAMED WINDOW <http://w1> ON STREAM <http://s1> [RANGE 10 STEP 10]
+        WHERE {{ WINDOW <http://w1> {{ ?s ex:temp ?o }} }}
+    """
+    outputs = []
+    b = RSPBuilder().add_rsp_ql_query(q).add_consumer(outputs.append)
+    for k, v in kw.items():
+        getattr(b, k)(v)
+    return b.build(), outputs
+
+
+def test_engine_single_window_rstream():
+    eng, outputs = _simple_engine()
+    for ts in range(0, 25):
+        eng.add_to_stream("http://s1", (f"<{EX}m{ts % 3}>", f"<{EX}temp>", f'"{ts}"'), ts)
+    # windows [0,10) and [10,20) fired
+    assert len(outputs) == 2
+    flat = outputs[0]
+    assert all(len(r) == 2 for r in flat)
+    assert any(r[0] == f"{EX}m0" for r in flat)
+
+
+def test_engine_istream_only_new():
+    eng, outputs = _simple_engine(stream_type="ISTREAM")
+    # same triple in both windows: second firing emits nothing new
+    eng.add_to_stream("http://s1", (f"<{EX}a>", f"<{EX}temp>", '"1"'), 1)
+    eng.add_to_stream("http://s1", (f"<{EX}a>", f"<{EX}temp>", '"1"'), 11)
+    eng.add_to_stream("http://s1", (f"<{EX}a>", f"<{EX}temp>", '"1"'), 21)
+    assert len(outputs) == 2
+    assert len(outputs[0]) == 1   # first firing: new row
+    assert outputs[1] == []       # same content -> ISTREAM empty
+
+
+def test_engine_reasoning_in_window():
+    q = f"""
+        PREFIX ex: <{EX}>
+        REGISTER RSTREAM <http://out> AS
+        SELECT ?m
+        FROM NAMED WINDOW <http://w1> ON STREAM <http://s1> [RANGE 10 STEP 10]
+        WHERE {{ WINDOW <http://w1> {{ ?m ex:alert "hot" }} }}
+    """
+    rules = f"""
+        RULE :Hot :- CONSTRUCT {{ ?m <{EX}alert> "hot" }}
+        WHERE {{ ?m <{EX}temp> ?t . FILTER(?t > 90) }} .
+    """
+    outputs = []
+    eng = (RSPBuilder().add_rsp_ql_query(q).add_sparql_rules(rules)
+           .add_consumer(outputs.append).build())
+    eng.add_to_stream("http://s1", (f"<{EX}m1>", f"<{EX}temp>", '"95"'), 1)
+    eng.add_to_stream("http://s1", (f"<{EX}m2>", f"<{EX}temp>", '"50"'), 2)
+    eng.add_to_stream("http://s1", (f"<{EX}m3>", f"<{EX}temp>", '"99"'), 12)
+    assert len(outputs) == 1
+    assert sorted(r[0] for r in outputs[0]) == [f"{EX}m1"]
+
+
+def test_engine_multi_window_join_wait():
+    q = f"""
+        PREFIX ex: <{EX}>
+        REGISTER RSTREAM <http://out> AS
+        SELECT ?m ?t ?l
+        FROM NAMED WINDOW <http://w1> ON STREAM <http://s1> [RANGE 10 STEP 10]
+        FROM NAMED WINDOW <http://w2> ON STREAM <http://s2> [RANGE 10 STEP 10]
+        WHERE {{
+            WINDOW <http://w1> {{ ?m ex:temp ?t }}
+            WINDOW <http://w2> {{ ?m ex:loc ?l }}
+        }}
+    """
+    outputs = []
+    eng = (RSPBuilder().add_rsp_ql_query(q).set_sync_policy("Wait")
+           .add_consumer(outputs.append).build())
+    eng.add_to_stream("http://s1", (f"<{EX}m1>", f"<{EX}temp>", '"95"'), 1)
+    eng.add_to_stream("http://s1", (f"<{EX}m1>", f"<{EX}temp>", '"96"'), 11)
+    assert outputs == []  # w2 never fired: Wait blocks
+    eng.add_to_stream("http://s2", (f"<{EX}m1>", f"<{EX}loc>", '"lab"'), 5)
+    eng.add_to_stream("http://s2", (f"<{EX}m1>", f"<{EX}loc>", '"lab"'), 15)
+    assert len(outputs) == 1
+    assert (f"{EX}m1", "95", "lab") in {tuple(r) for r in outputs[0]}
+
+
+def test_engine_static_join():
+    q = f"""
+        PREFIX ex: <{EX}>
+        REGISTER RSTREAM <http://out> AS
+        SELECT ?m ?t
+        FROM NAMED WINDOW <http://w1> ON STREAM <http://s1> [RANGE 10 STEP 10]
+        WHERE {{ WINDOW <http://w1> {{ ?m ex:temp ?t }} }}
+    """
+    outputs = []
+    eng = (RSPBuilder().add_rsp_ql_query(q)
+           .add_consumer(outputs.append).build())
+    eng.add_static_ntriples(f'<{EX}m1> <{EX}type> "sensor" .')
+    eng.add_to_stream("http://s1", (f"<{EX}m1>", f"<{EX}temp>", '"20"'), 1)
+    eng.add_to_stream("http://s1", (f"<{EX}m1>", f"<{EX}temp>", '"21"'), 11)
+    assert len(outputs) == 1
+    assert outputs[0]
+
+
+def test_engine_multithread_mode():
+    from kolibrie_amd.rsp import OperationMode
+    q = f"""
+        PREFIX ex: <{EX}>
+        REGISTER RSTREAM <http://out> AS
+        SELECT ?s ?o
+        FROM NAMED WINDOW <http://w1> ON STREAM <http://s1> [RANGE 5 STEP 5]
+        WHERE {{ WINDOW <http://w1> {{ ?s ex:p ?o }} }}
+    """
+    outputs = []
+    eng = (RSPBuilder().add_rsp_ql_query(q)
+           .set_operation_mode(OperationMode.MULTI_THREAD)
+           .add_consumer(outputs.append).build())
+    for ts in range(0, 12):
+        eng.add_to_stream("http://s1", (f"<{EX}x>", f"<{EX}p>", f'"{ts}"'), ts)
+    import time
+    deadline = time.time() + 5
+    while len(outputs) < 2 and time.time() < deadline:
+        time.sleep(0.01)
+    eng.stop()
+    assert len(outputs) >= 2
+
+
+def test_cross_window_sds_naive_vs_incremental():
+    from kolibrie_amd.reasoning.sds import (
+        Sds, WindowedTriple, incremental_sds_plus, naive_sds_plus,
+    )
+    from kolibrie_amd.reasoning.rule import Rule
+    from kolibrie_amd.storage.terms import Constant, TriplePattern, Variable
+    from kolibrie_amd.storage.database import SparqlDatabase
+    db = SparqlDatabase()
+    p = db.dictionary.encode("p")
+    q = db.dictionary.encode("q")
+    a, b, c = (db.dictionary.encode(x) for x in "abc")
+    rule = Rule(
+        premise=[
+            TriplePattern(Variable("x"), Constant(p), Variable("y")),
+            TriplePattern(Variable("y"), Constant(p), Variable("z")),
+        ],
+        conclusion=[TriplePattern(Variable("x"), Constant(q), Variable("z"))],
+    )
+    sds = Sds()
+    sds.add(WindowedTriple("w1", (a, p, b), 0), 10)   # expires at 10
+    sds.add(WindowedTriple("w2", (b, p, c), 5), 10)   # expires at 15
+    for ts in (1, 7, 12, 20):
+        got_n = naive_sds_plus(sds, [rule], db, ts)
+        got_i = incremental_sds_plus(sds, [rule], db, ts)
+        assert got_n == got_i, ts
+    # derived (a,q,c) lives until min(10, 15) = 10
+    assert (a, q, c) in naive_sds_plus(sds, [rule], db, 7)
+    assert (a, q, c) not in naive_sds_plus(sds, [rule], db, 12)
