@@ -12,6 +12,13 @@ import pytest
 REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
 
 
+def _free_port() -> int:
+    import socket
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
+
+
 def _run_bench(nproc: int, port: int) -> dict:
     cmd = [
         sys.executable, "-m", "torch.distributed.run",
@@ -31,8 +38,8 @@ def _run_bench(nproc: int, port: int) -> dict:
 
 
 def test_distributed_count_matches_single_rank():
-    r1 = _run_bench(1, 29531)
-    r2 = _run_bench(2, 29532)
+    r1 = _run_bench(1, _free_port())
+    r2 = _run_bench(2, _free_port())
     assert r1["config"]["result_count"] == r2["config"]["result_count"]
     assert r1["config"]["result_count"] > 0
 
@@ -58,7 +65,7 @@ print("rank", rank, "ok")
     cmd = [
         sys.executable, "-m", "torch.distributed.run",
         "--nnodes=1", "--nproc-per-node=2",
-        "--master-addr", "127.0.0.1", "--master-port", "29533",
+        "--master-addr", "127.0.0.1", "--master-port", str(_free_port()),
         "-m", "no_module",
     ]
     # run via -c through a wrapper file instead
@@ -71,7 +78,7 @@ print("rank", rank, "ok")
         cmd = [
             sys.executable, "-m", "torch.distributed.run",
             "--nnodes=1", "--nproc-per-node=2",
-            "--master-addr", "127.0.0.1", "--master-port", "29533",
+            "--master-addr", "127.0.0.1", "--master-port", str(_free_port()),
             path,
         ]
         out = subprocess.run(cmd, cwd=REPO, capture_output=True, text=True,
