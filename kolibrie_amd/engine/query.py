@@ -58,6 +58,11 @@ def execute_select(select: SelectQuery, db, prefixes: Dict[str, str]
                    ) -> List[List[str]]:
     from ..plan.lower import build_logical_plan
     from ..plan.optimizer import Streamertail
+    # neural relations referenced by the query materialize first
+    # (ref execute_query.rs:201 materialize_neural_relations_for_patterns)
+    if db.neural_relations:
+        from ..ml.neural_relations import materialize_for_select
+        materialize_for_select(select, db, prefixes)
     view = _build_view(select, db, prefixes)
     stats = db.get_or_build_stats()
     logical = build_logical_plan(select.where, db, prefixes)

@@ -43,11 +43,6 @@ class MlpNeuralPredicate(nn.Module):
                     "labels": self.labels}, path)
 
 
-def train_from_patterns(entry: dict, decl, db, prefixes):
-    """Train an MLP from DATA graph-pattern feature rows."""
-    raise NotImplementedError("neurosymbolic training arrives with the provenance phase")
-
-
 def predict_rows(info: dict, rows, db):
     """Evaluate a registered model over binding rows; binds the output var."""
     from ..engine.bindings import Bindings

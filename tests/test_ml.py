@@ -1,0 +1,134 @@
+"""Neurosymbolic ML tests (mirrors ml crate behavior and
+kolibrie/tests/ml_predict_candle_runtime.rs shapes)."""
+import os
+import pickle
+import tempfile
+
+import pytest
+import torch
+
+from kolibrie_amd import SparqlDatabase
+
+EX = "http://example.org/"
+
+
+def test_mlp_neural_predicate_shapes():
+    from kolibrie_amd.ml.neural import MlpNeuralPredicate
+    m = MlpNeuralPredicate(3, [8], 1, "binary")
+    x = torch.randn(5, 3)
+    p = m.predict_proba(x)
+    assert p.shape == (5,)
+    assert bool(((p >= 0) & (p <= 1)).all())
+    mc = MlpNeuralPredicate(3, [8], 4, "categorical", labels=list("abcd"))
+    pc = mc.predict_proba(x)
+    assert pc.shape == (5, 4)
+    assert torch.allclose(pc.sum(-1), torch.ones(5), atol=1e-5)
+
+
+def test_feature_loader_from_bindings():
+    from kolibrie_amd.engine.bindings import Bindings
+    from kolibrie_amd.ml.feature_loader import features_from_bindings
+    db = SparqlDatabase()
+    v1 = db.dictionary.encode("1.5")
+    v2 = db.dictionary.encode("2.5")
+    rows = Bindings.from_dicts([{"a": v1}, {"a": v2}], "cpu")
+    x = features_from_bindings(rows, ["a"], db)
+    assert x.tolist() == [[1.5], [2.5]]
+
+
+def test_ml_handler_sklearn_pickle():
+    sklearn = pytest.importorskip("sklearn")
+    from sklearn.linear_model import LogisticRegression
+    from kolibrie_amd.ml.handler import MLHandler
+    from kolibrie_amd.ml.mlschema import emit_mlschema_ttl
+    X = [[0.0], [1.0], [2.0], [3.0]]
+    y = [0, 0, 1, 1]
+    clf = LogisticRegression().fit(X, y)
+    with tempfile.TemporaryDirectory() as d:
+        path = os.path.join(d, "fraud.pkl")
+        with open(path, "wb") as f:
+            pickle.dump(clf, f)
+        ttl = emit_mlschema_ttl("fraud", ["amount"], "accuracy", 0.95)
+        with open(os.path.join(d, "fraud.ttl"), "w") as f:
+            f.write(ttl)
+        h = MLHandler()
+        info = h.load_model(path)
+        assert info.measure == pytest.approx(0.95)
+        assert h.best_model().name == "fraud"
+        preds = h.predict("fraud", [[0.0], [3.0]])
+        assert preds[0] < 0.5 < preds[1]
+
+
+def test_train_decl_and_neural_relation_materialization():
+    db = SparqlDatabase()
+    # training data: salary -> fraud flag ("1"/"0" numeric labels)
+    for i in range(40):
+        sal = 1000 + i * 100
+        label = "1" if sal > 3000 else "0"
+        db.add_triple(f"<{EX}p{i}>", f"<{EX}salary>", f'"{sal}"')
+        db.add_triple(f"<{EX}p{i}>", f"<{EX}fraudLabel>", f'"{label}"')
+    q = f"""
+        MODEL "frauddet" {{ ARCH MLP {{ HIDDEN [8] }} OUTPUT BINARY {{"suspect"}} }}
+        NEURAL RELATION <{EX}suspicious> USING MODEL "frauddet" {{
+            INPUT {{ ?x <{EX}salary> ?s }}
+            FEATURES {{ ?s }}
+        }}
+        TRAIN NEURAL RELATION <{EX}suspicious> USING MODEL "frauddet" {{
+            DATA {{ ?x <{EX}salary> ?s . ?x <{EX}fraudLabel> ?y }}
+            label = ?y
+            epochs = 200
+        }}
+        SELECT ?x WHERE {{ ?x <{EX}suspicious> "suspect" }}
+    """
+    rows = db.query(q)
+    found = {r[0] for r in rows}
+    # high-salary people should be flagged; low-salary not
+    assert f"{EX}p39" in found
+    assert f"{EX}p0" not in found
+    # probability seeds recorded for hybrid use
+    assert db.probability_seeds
+
+
+def test_neurosymbolic_wmc_training_converges():
+    """The SDD-WMC gradient bridge: network outputs feed a 2-premise rule;
+    the target label is reachable only if both seeds fire."""
+    from kolibrie_amd.ml.neural import MlpNeuralPredicate
+    from kolibrie_amd.ml.train import train_neurosymbolic
+    from kolibrie_amd.reasoning.rule import Rule
+    from kolibrie_amd.storage.terms import Constant, TriplePattern, Variable
+    P, Q = 50, 51
+    a, b, c = 1, 2, 3
+    rule = Rule(
+        premise=[TriplePattern(Variable("x"), Constant(P), Variable("y")),
+                 TriplePattern(Variable("y"), Constant(P), Variable("z"))],
+        conclusion=[TriplePattern(Variable("x"), Constant(Q), Variable("z"))],
+    )
+    torch.manual_seed(0)
+    model = MlpNeuralPredicate(1, [8], 1, "binary")
+    samples = [{
+        "x": torch.tensor([[1.0], [2.0]]),
+        "seed_triples": [(a, P, b), (b, P, c)],
+        "target": (a, Q, c),
+        "label": 1.0,
+    }]
+    losses = train_neurosymbolic(model, samples, [rule], epochs=60)
+    assert losses[-1] < losses[0]
+    p = model.predict_proba(samples[0]["x"])
+    assert float(p.min()) > 0.7   # both premises pushed towards firing
+
+
+def test_ml_predict_in_query():
+    from kolibrie_amd.ml.neural import MlpNeuralPredicate
+    db = SparqlDatabase()
+    db.add_triple(f"<{EX}a>", f"<{EX}v>", '"4.0"')
+    model = MlpNeuralPredicate(1, [4], 1, "binary")
+    db.neural_models["m1"] = {"model": model, "features": ["v"],
+                              "labels": ["yes"]}
+    from kolibrie_amd.engine.bindings import Bindings
+    from kolibrie_amd.ml.predict import execute_ml_predict
+    v_id = db.dictionary.encode("4.0")
+    rows = Bindings.from_dicts([{"v": v_id}], "cpu")
+    out = execute_ml_predict({"model": "m1", "output_var": "p"}, rows, db)
+    assert out.has("p")
+    val = db.dictionary.decode(int(out.col("p")[0].item()) & 0xFFFFFFFF)
+    assert 0.0 <= float(val) <= 1.0
