@@ -50,6 +50,9 @@ def materialize_neural_relation(name: str, entry: dict, db,
     if rows.is_empty():
         return 0
     x = features_from_bindings(rows, feat_vars, db)
+    norm = (model_entry or {}).get("norm")
+    if norm is not None:
+        x = (x - norm[0]) / norm[1]
     proba = model.predict_proba(x)
     if proba.dim() > 1:
         proba = proba[:, -1]

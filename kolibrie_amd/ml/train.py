@@ -49,6 +49,11 @@ def _train_supervised(entry: dict, decl, db, prefixes,
     if feat_vars is None:
         feat_vars = [v for v in rows.variables if v != label_var]
     x = features_from_bindings(rows, feat_vars, db)
+    # standardize features: raw literal magnitudes (salaries etc.) saturate
+    # the MLP otherwise; the normalization ships with the model entry
+    mu = x.mean(dim=0, keepdim=True)
+    sigma = x.std(dim=0, keepdim=True).clamp(min=1e-6)
+    x = (x - mu) / sigma
     labels_ids = rows.col(label_var).to(torch.int64) & 0xFFFFFFFF
     vc = db.value_column()
     y = vc[torch.clamp(labels_ids, max=vc.numel() - 1)].to(torch.float32)
@@ -65,6 +70,7 @@ def _train_supervised(entry: dict, decl, db, prefixes,
         loss.backward()
         opt.step()
     entry["model"] = model
+    entry["norm"] = (mu, sigma)
     entry["features"] = feat_vars
     entry["labels"] = (entry.get("decl").options.get("labels", "true").split("\x1f")
                        if entry.get("decl") is not None else ["true"])

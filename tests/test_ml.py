@@ -60,6 +60,7 @@ def test_ml_handler_sklearn_pickle():
 
 
 def test_train_decl_and_neural_relation_materialization():
+    torch.manual_seed(7)
     db = SparqlDatabase()
     # training data: salary -> fraud flag ("1"/"0" numeric labels)
     for i in range(40):
