@@ -175,6 +175,20 @@ class Reasoner:
         self._flush()
         return self.facts.index().contains(*[i & 0xFFFFFFFF for i in ids])  # type: ignore
 
+    # -------------------------------------------------- backward chaining --
+    def backward_chaining(self, goal, max_depth: int = 24):
+        """SLD goal resolution (ref backward_chaining.rs:150).  `goal` is a
+        TriplePattern or an (s,p,o) tuple of strings/'?var's."""
+        from ..storage.terms import Constant, TriplePattern, Variable
+        if not isinstance(goal, TriplePattern):
+            def term(x):
+                if isinstance(x, str) and x.startswith("?"):
+                    return Variable(x[1:])
+                return Constant(self._i32(self.dictionary.encode(x)))
+            goal = TriplePattern(*(term(x) for x in goal))
+        from .backward import backward_chain
+        return backward_chain(goal, self, max_depth=max_depth)
+
     # ------------------------------------------------------- constraints --
     def violates_constraints(self) -> bool:
         """A constraint is a rule whose body matching any facts = violation
