@@ -29,23 +29,41 @@ from .rule import Rule
 
 
 class FactStore:
-    """Growing columnar fact set with a sorted index."""
+    """Growing columnar fact set with an amortized-rebuild sorted index.
+
+    The index covers a `base` prefix; facts added since the last rebuild
+    live in a small unsorted `tail` that joins brute-force.  A rebuild
+    (full sort) happens only when the tail outgrows 25% of the base, so a
+    fixpoint of R rounds costs O(log R) sorts instead of R."""
+
+    REBUILD_FRACTION = 0.25
+    REBUILD_MIN = 16_384
 
     def __init__(self, device):
         self.device = torch.device(device)
         e = torch.empty(0, dtype=torch.int32, device=self.device)
         self.s, self.p, self.o = e, e.clone(), e.clone()
         self._index: Optional[GraphIndex] = None
+        self._base_n = 0
 
     @property
     def n(self) -> int:
         return self.s.numel()
 
     def index(self) -> GraphIndex:
-        if self._index is None:
+        """Sorted index over the base prefix (pair with tail_columns())."""
+        if self._index is None or (
+                self.n - self._base_n
+                > max(self.REBUILD_MIN, self._base_n * self.REBUILD_FRACTION)):
             self._index = GraphIndex.from_columns(
                 self.s, self.p, self.o, device=self.device, dedup=False)
+            self._base_n = self.n
         return self._index
+
+    def tail_columns(self):
+        """Facts newer than the current index build."""
+        return (self.s[self._base_n:], self.p[self._base_n:],
+                self.o[self._base_n:])
 
     def add_columns(self, s, p, o) -> int:
         """Add new facts (must already be deduped against self); returns the
@@ -55,15 +73,48 @@ class FactStore:
         self.s = torch.cat([self.s, s])
         self.p = torch.cat([self.p, p])
         self.o = torch.cat([self.o, o])
-        self._index = None
         return s.numel()
 
     def set_columns(self, s, p, o):
         self.s, self.p, self.o = s, p, o
         self._index = None
+        self._base_n = 0
 
     def sorted_unique_rows(self):
         return unique_rows([self.s, self.p, self.o])
+
+
+def membership_in_index(idx: GraphIndex, s, p, o) -> torch.Tensor:
+    """Vectorized membership of (s,p,o) rows in a sorted index — exact
+    (s,p) searchsorted plus an in-range manual binary search on o.  No
+    per-round sort of the known set (the old membership_mask lexsorted
+    everything every round)."""
+    from ..engine.tensor_utils import pack2
+    from ..storage.dataset import SPO
+    key12, z = idx.orders[SPO]
+    n = s.numel()
+    dev = s.device
+    if n == 0 or idx.n == 0:
+        return torch.zeros(n, dtype=torch.bool, device=dev)
+    k = pack2(s, p)
+    lo = torch.searchsorted(key12, k, side="left")
+    hi = torch.searchsorted(key12, k, side="right")
+    l, h = lo.clone(), hi.clone()
+    zmax = z.numel() - 1
+    while True:
+        active = l < h
+        if not bool(active.any()):
+            break
+        mid = (l + h) >> 1
+        zm = z[torch.clamp(mid, max=zmax)]
+        less = zm < o
+        go_r = active & less
+        go_l = active & ~less
+        l = torch.where(go_r, mid + 1, l)
+        h = torch.where(go_l, mid, h)
+    in_range = l < hi
+    zl = z[torch.clamp(l, max=zmax)]
+    return in_range & (zl == o)
 
 
 def _match_premise_against_delta(
@@ -92,10 +143,27 @@ def _match_premise_against_delta(
 
 
 def _join_premise_all_facts(
-    b: Bindings, prem: TriplePattern, idx: GraphIndex, device
+    b: Bindings, prem: TriplePattern, idx: GraphIndex, device,
+    tail=None,
 ) -> Bindings:
     """Join current bindings with a premise matched against ALL facts
-    (ref rules.rs:167 join_premise_with_hash_join -> K1/K2)."""
+    (ref rules.rs:167 join_premise_with_hash_join -> K1/K2).  `tail` is
+    the store's unsorted post-rebuild suffix, joined brute-force."""
+    base_res = _join_premise_base(b, prem, idx, device)
+    if tail is None or tail[0].numel() == 0:
+        return base_res
+    tail_cand = _match_premise_against_delta(prem, tail[0], tail[1], tail[2],
+                                             device)
+    if tail_cand is None or tail_cand.is_empty():
+        return base_res
+    from ..engine.executor import join_bindings
+    tail_res = join_bindings(b, tail_cand)
+    return Bindings.concat([base_res, tail_res], device)
+
+
+def _join_premise_base(
+    b: Bindings, prem: TriplePattern, idx: GraphIndex, device
+) -> Bindings:
     consts: Dict[int, int] = {}
     var_pos: Dict[int, str] = {}
     for i, t in enumerate(prem.terms()):
@@ -143,9 +211,22 @@ def _join_premise_all_facts(
 
 
 def _apply_negative(b: Bindings, neg: List[TriplePattern], idx: GraphIndex,
-                    device) -> Bindings:
+                    device, tail=None) -> Bindings:
     """NAF: drop bindings for which a negative premise matches a fact."""
     for prem in neg:
+        if tail is not None and tail[0].numel() and not b.is_empty():
+            tail_cand = _match_premise_against_delta(
+                prem, tail[0], tail[1], tail[2], device)
+            if tail_cand is not None and not tail_cand.is_empty():
+                from ..engine.executor import join_bindings
+                marked = b.with_col(
+                    "__row", torch.arange(b.n, dtype=torch.int32,
+                                          device=device))
+                hit_rows = join_bindings(marked, tail_cand)
+                hit = torch.zeros(b.n, dtype=torch.bool, device=device)
+                if not hit_rows.is_empty() and hit_rows.has("__row"):
+                    hit[hit_rows.col("__row").to(torch.long)] = True
+                b = b.select(~hit)
         if b.is_empty():
             return b
         consts: Dict[int, int] = {}
@@ -230,6 +311,7 @@ def infer_round(
     (ref semi_naive.rs:17-86)."""
     device = facts.device
     idx = facts.index()
+    tail = facts.tail_columns()
     news = [[], [], []]
     for rule in rules:
         np_ = len(rule.premise)
@@ -242,7 +324,8 @@ def infer_round(
             for j in range(np_):
                 if j == i:
                     continue
-                b = _join_premise_all_facts(b, rule.premise[j], idx, device)
+                b = _join_premise_all_facts(b, rule.premise[j], idx, device,
+                                            tail)
                 if b.is_empty():
                     ok = False
                     break
@@ -251,7 +334,7 @@ def infer_round(
             b = _eval_filters(rule, b, db)
             if b.is_empty():
                 continue
-            b = _apply_negative(b, rule.negative_premise, idx, device)
+            b = _apply_negative(b, rule.negative_premise, idx, device, tail)
             if b.is_empty():
                 continue
             s, p, o = _instantiate(rule, b, device)
@@ -265,13 +348,14 @@ def infer_round(
     s = torch.cat(news[0])
     p = torch.cat(news[1])
     o = torch.cat(news[2])
-    # dedup within the round, then against known facts
+    # dedup within the round, then against known facts: searchsorted
+    # membership against the sorted base + mask against the small tail
     s, p, o = unique_rows([s, p, o])
-    known = facts.sorted_unique_rows()
-    if known[0].numel():
-        hit = membership_mask([s, p, o], known)
-        keep = ~hit
-        s, p, o = s[keep], p[keep], o[keep]
+    keep = ~membership_in_index(idx, s, p, o)
+    if tail[0].numel():
+        tail_sorted = unique_rows(list(tail))
+        keep &= ~membership_mask([s, p, o], tail_sorted)
+    s, p, o = s[keep], p[keep], o[keep]
     return s, p, o
 
 
@@ -280,6 +364,21 @@ def infer_fixpoint(rules: List[Rule], facts: FactStore, db,
     """Loop until no new facts (ref infer_generic.rs:27).  Returns the
     number of derived facts."""
     device = facts.device
+    # small working sets run the host hash path (deep-taxonomy-shaped
+    # workloads: thousands of tiny rounds where per-round device-op
+    # overhead dominates); large sets stay columnar on device
+    from .host_fixpoint import HOST_PATH_MAX_FACTS, infer_fixpoint_host
+    if semi_naive and facts.n <= HOST_PATH_MAX_FACTS:
+        tuples = list(zip(facts.s.cpu().tolist(), facts.p.cpu().tolist(),
+                          facts.o.cpu().tolist()))
+        derived = infer_fixpoint_host(rules, tuples, db)
+        if derived:
+            import numpy as np
+            arr = np.asarray(derived, dtype=np.int64).astype(np.int32)
+            t = torch.from_numpy(arr).to(device)
+            facts.add_columns(t[:, 0].contiguous(), t[:, 1].contiguous(),
+                              t[:, 2].contiguous())
+        return len(derived)
     total_new = 0
     if semi_naive:
         ds, dp, do_ = facts.s, facts.p, facts.o
