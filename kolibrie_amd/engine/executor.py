@@ -214,6 +214,14 @@ class ExecutionEngine:
                 scan_need.add(pos_i)
 
         if not probe_vars:
+            if (needed is not None and not scan_need and not qt_pos
+                    and extra is None and is_unit):
+                # COUNT(*) over a bare scan: range size only
+                from .scan import scan_unit_count
+                cnt = scan_unit_count(idx, consts)
+                if cnt is not None:
+                    exec_stats.bump("ROWS_EMITTED", cnt)
+                    return Bindings({}, cnt, dev)
             s, p, o = scan_unit(idx, consts, sort_hint=sort_hint,
                                 need=scan_need)
             n_sc = next((c.numel() for c in (s, p, o) if c is not None), 0)

@@ -276,3 +276,33 @@ def scan_probe_count(idx: GraphIndex, consts: Dict[int, int],
         lo = torch.searchsorted(key12, pack2(v, torch.zeros_like(v)), side="left")
         hi = torch.searchsorted(key12, pack2(v, torch.full_like(v, -1)), side="right")
     return int((hi - lo).sum().item())
+
+
+def scan_unit_count(idx: GraphIndex, consts: Dict[int, int]) -> Optional[int]:
+    """Row count of a constant-bound scan without materializing columns
+    (COUNT(*) pushdown).  Returns None when post-filtering would be needed
+    (caller falls back to the materializing path)."""
+    if idx.n == 0:
+        return 0
+    code, plen = choose_order(set(consts.keys()), set(consts.keys()))
+    pos = _ORDER_POS[code]
+    if set(consts.keys()) - set(pos[:plen]):
+        return None
+    if plen == 0:
+        return idx.n
+    dev = idx.device
+    key12, _z = idx.orders[code]
+    if plen == 2:
+        k = pack2(torch.tensor([consts[pos[0]]], dtype=torch.int32, device=dev),
+                  torch.tensor([consts[pos[1]]], dtype=torch.int32, device=dev))
+        lo = int(torch.searchsorted(key12, k, side="left").item())
+        hi = int(torch.searchsorted(key12, k, side="right").item())
+    else:
+        v = consts[pos[0]]
+        klo = pack2(torch.tensor([v], dtype=torch.int32, device=dev),
+                    torch.tensor([0], dtype=torch.int32, device=dev))
+        khi = pack2(torch.tensor([v], dtype=torch.int32, device=dev),
+                    torch.tensor([-1], dtype=torch.int32, device=dev))
+        lo = int(torch.searchsorted(key12, klo, side="left").item())
+        hi = int(torch.searchsorted(key12, khi, side="right").item())
+    return hi - lo

@@ -152,7 +152,12 @@ class RSPBuilder:
             plan = None
             plan_vars: List[str] = []
             if inner is not None:
-                plan, plan_vars = _build_window_plan(inner, db, prefixes)
+                agg_select = reg.select if (
+                    len(reg.windows) == 1
+                    and any(p.aggregate for p in reg.select.variables)
+                ) else None
+                plan, plan_vars = _build_window_plan(inner, db, prefixes,
+                                                     agg_select)
             engine.add_window(
                 wiri, wc.stream_iri, wc.spec.width, wc.spec.slide or wc.spec.width,
                 report=wc.spec.report, tick=wc.spec.tick,
@@ -183,16 +188,26 @@ def _collect_window_blocks(g: GGP) -> List[Tuple[str, GGP]]:
     return out
 
 
-def _build_window_plan(inner: GGP, db, prefixes):
-    """Build an optimized physical plan projecting all window variables
-    (ref builder.rs:279-324 create_rsp_query_plan)."""
+def _build_window_plan(inner: GGP, db, prefixes, agg_select=None):
+    """Build an optimized physical plan projecting all window variables;
+    for a single-window aggregate query the register's own SELECT (with
+    aggregates) finalizes on device (ref builder.rs:279-324)."""
     from ..plan.lower import build_logical_plan
     from ..plan.optimizer import Streamertail, annotate_needed
     logical = build_logical_plan(inner, db, prefixes)
     stats = db.get_or_build_stats()
     physical = Streamertail(stats).find_best_plan(logical)
-    annotate_needed(physical, None)
     from ..plan.optimizer import _logical_out_vars
+    if agg_select is not None:
+        from ..engine.query import _top_needed
+        sel = SelectQuery(
+            variables=list(agg_select.variables),
+            group_by=list(agg_select.group_by),
+            distinct=agg_select.distinct, where=inner)
+        annotate_needed(physical, _top_needed(sel))
+        plan_vars = [p.output_name() for p in sel.variables]
+        return (sel, physical), plan_vars
+    annotate_needed(physical, None)
     plan_vars = sorted(_logical_out_vars(logical))
     sel = SelectQuery(variables=[Projection(var=v) for v in plan_vars],
                       where=inner)

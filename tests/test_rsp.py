@@ -230,3 +230,38 @@ def test_cross_window_sds_naive_vs_incremental():
     # derived (a,q,c) lives until min(10, 15) = 10
     assert (a, q, c) in naive_sds_plus(sds, [rule], db, 7)
     assert (a, q, c) not in naive_sds_plus(sds, [rule], db, 12)
+
+
+def test_bulk_ingest_matches_per_event_path():
+    """K7 columnar ingest must produce the same window counts as the
+    per-event host path."""
+    import torch
+    q = f"""PREFIX ex: <{EX}>
+REGISTER RSTREAM <http://out> AS
+SELECT (COUNT(*) AS ?c)
+FROM NAMED WINDOW <http://w1> ON STREAM <http://s1> [RANGE 10 STEP 10]
+WHERE {{ WINDOW <http://w1> {{ ?m ex:temp ?v }} }}"""
+    # bulk path
+    out_bulk = []
+    eng = RSPBuilder().add_rsp_ql_query(q).add_consumer(out_bulk.append).build()
+    db = eng.store.db
+    temp = db.encode_term(f"<{EX}temp>")
+    n = 50
+    s = torch.arange(n, dtype=torch.int32) % 7 + 10_000
+    p = torch.full((n,), temp, dtype=torch.int32)
+    o = torch.arange(n, dtype=torch.int32) + 20_000
+    ts = (torch.arange(n, dtype=torch.int64) * 25) // n  # 0..24
+    eng.add_to_stream_bulk("http://s1", s, p, o, ts)
+    counts_bulk = [int(rows[0][0]) for rows in out_bulk if rows]
+    # per-event oracle: same events through CSPARQLWindow + host processor
+    out_host = []
+    eng2 = RSPBuilder().add_rsp_ql_query(q).add_consumer(out_host.append).build()
+    db2 = eng2.store.db
+    for i in range(n):
+        eng2.add_to_stream(
+            "http://s1",
+            (f"<http://sensor/{int(s[i])}>", f"<{EX}temp>", f'"{int(o[i])}"'),
+            int(ts[i]))
+    counts_host = [int(rows[0][0]) for rows in out_host if rows]
+    assert counts_bulk == counts_host
+    assert sum(counts_bulk) == sum(1 for t in ts.tolist() if t < 20)
