@@ -94,9 +94,53 @@ def prepare_extensions(cq: CombinedQuery, db, prefixes: Dict[str, str]):
         db.rules.append(rule)
 
 
+class PreparedQuery:
+    """A parsed+planned SELECT, reusable across executions (plan cache).
+
+    Invalidation: plans embed dictionary-encoded constants and cost-based
+    decisions from the stats snapshot, so entries key on the store version.
+    """
+
+    __slots__ = ("select", "physical", "view", "store_version")
+
+    def __init__(self, select, physical, view, store_version):
+        self.select = select
+        self.physical = physical
+        self.view = view
+        self.store_version = store_version
+
+
+def _prepare_select(select: SelectQuery, db, prefixes) -> "PreparedQuery":
+    from ..plan.lower import build_logical_plan
+    from ..plan.optimizer import Streamertail, annotate_needed
+    if db.neural_relations:
+        from ..ml.neural_relations import materialize_for_select
+        materialize_for_select(select, db, prefixes)
+    view = _build_view(select, db, prefixes)
+    stats = db.get_or_build_stats()
+    logical = build_logical_plan(select.where, db, prefixes)
+    physical = Streamertail(stats).find_best_plan(logical)
+    annotate_needed(physical, _top_needed(select))
+    return PreparedQuery(select, physical, view, db.store.version)
+
+
+def _run_prepared(pq: "PreparedQuery", db) -> List[List[str]]:
+    ctx = ExecutionContext(db, pq.view)
+    rows = ExecutionEngine(ctx).execute(pq.physical, Bindings.unit(db.device))
+    final = finalize_select_bindings(pq.select, rows, db)
+    return decode_rows(pq.select, final, db)
+
+
 def execute_query(sparql: str, db) -> List[List[str]]:
     """Full request entry (ref execute_query_rayon_parallel2_volcano,
-    execute_query.rs:52)."""
+    execute_query.rs:52).  Pure SELECT queries hit the prepared-plan
+    cache (keyed on query text + store version)."""
+    cache = getattr(db, "_plan_cache", None)
+    if cache is None:
+        cache = db._plan_cache = {}
+    hit = cache.get(sparql)
+    if hit is not None and hit.store_version == db.store.version:
+        return _run_prepared(hit, db)
     cq = parse_combined_query(sparql)
     prefixes = dict(db.prefixes)
     prefixes.update(cq.prefixes)
@@ -107,6 +151,12 @@ def execute_query(sparql: str, db) -> List[List[str]]:
         if cq.select is None:
             return []
     if cq.select is not None:
+        if not cq.updates and not cq.rules and not cq.train_decls \
+                and not cq.register:
+            pq = _prepare_select(cq.select, db, prefixes)
+            if len(cache) < 256:
+                cache[sparql] = pq
+            return _run_prepared(pq, db)
         return execute_select(cq.select, db, prefixes)
     return []
 
