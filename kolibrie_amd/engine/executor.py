@@ -263,16 +263,18 @@ class ExecutionEngine:
                 if cnt is not None:
                     parts.append(Bindings({}, cnt, dev))
                     continue
-            li, s, p, o = scan_probe(idx, consts, probes, need=scan_need)
-            exec_stats.bump("QUADS_EXAMINED", li.numel())
-            # projection pushdown: gather only incoming columns the scan
-            # does not itself supply (pattern vars re-emerge from the index)
+            # incoming columns the scan does not itself supply are carried
+            # through the fused emit (projection pushdown + fused gather)
             if needed is None:
-                base_src = sub
+                carry_src = sub
             else:
                 pattern_names = set(var_pos.values())
-                base_src = _prune(sub, set(needed) - pattern_names)
-            base = base_src.gather(li)
+                carry_src = _prune(sub, set(needed) - pattern_names)
+            from .scan import scan_probe_carry
+            li, s, p, o, carried = scan_probe_carry(
+                idx, consts, probes, scan_need, dict(carry_src.cols))
+            exec_stats.bump("QUADS_EXAMINED", li.numel())
+            base = Bindings(carried, li.numel(), dev)
             cand = self._build_candidate(s, p, o, var_pos, qt_pos, base,
                                          needed, li.numel())
             parts.append(cand)

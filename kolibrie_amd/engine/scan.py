@@ -306,3 +306,56 @@ def scan_unit_count(idx: GraphIndex, consts: Dict[int, int]) -> Optional[int]:
         lo = int(torch.searchsorted(key12, klo, side="left").item())
         hi = int(torch.searchsorted(key12, khi, side="right").item())
     return hi - lo
+
+
+def scan_probe_carry(idx: GraphIndex, consts: Dict[int, int],
+                     probes: Dict[int, torch.Tensor], need,
+                     carry: Dict[str, torch.Tensor]):
+    """K1 probe with carry columns: the emit pass gathers the incoming
+    rows' surviving columns directly (fused kernel — no separate pack2
+    pass, no separate per-column gathers).
+
+    Returns (li, s, p, o, carried: {name: tensor}).
+    """
+    dev = idx.device
+    need_set = ALL_POSITIONS if need is None else frozenset(need)
+    bound = set(consts.keys()) | set(probes.keys())
+    code, plen = choose_order(bound, set(consts.keys()))
+    pos = _ORDER_POS[code]
+    post_positions = bound - set(pos[:plen])
+    from ..ops import native_for
+    key12, z = idx.orders[code]
+    native = native_for(key12) if idx.n else None
+    n_rows = next(iter(probes.values())).numel()
+    if (native is None or plen != 2 or post_positions or len(carry) > 4
+            or n_rows == 0 or idx.n == 0):
+        li, s, p, o = scan_probe(idx, consts, probes, need)
+        carried = {k: v[li] for k, v in carry.items()}
+        return li, s, p, o, carried
+
+    def col_or_const(position):
+        if position in probes:
+            return probes[position].contiguous(), 0
+        return None, consts[position]
+
+    a_col, a_const = col_or_const(pos[0])
+    b_col, b_const = col_or_const(pos[1])
+    names = list(carry.keys())
+    emit_b = pos[1] in need_set
+    out = native.probe_fused(key12, z, a_col, a_const, b_col, b_const,
+                             [carry[k].contiguous() for k in names], emit_b)
+    li, b_out, z_out = out[0], out[1], out[2]
+    carried = {k: t for k, t in zip(names, out[3:])}
+    cols = [None, None, None]
+    if pos[0] in need_set:
+        src0 = probes[pos[0]] if pos[0] in probes else None
+        if src0 is not None:
+            cols[pos[0]] = src0[li]
+        else:
+            cols[pos[0]] = torch.full((li.numel(),), consts[pos[0]],
+                                      dtype=torch.int32, device=dev)
+    if emit_b:
+        cols[pos[1]] = b_out
+    if pos[2] in need_set:
+        cols[pos[2]] = z_out
+    return li, cols[0], cols[1], cols[2], carried

@@ -185,6 +185,133 @@ __global__ void probe_emit_balanced(const int64_t* __restrict__ key12,
   }
 }
 
+// ---- fused probe: inline key packing + carry-column emit ------------------
+// Builds probe keys (a<<32)|b inside the kernel (a/b each a per-row column
+// or a broadcast constant) and gathers up to 4 "carry" columns by probe-row
+// index during the emit pass — the engine's bind-join step then needs no
+// separate pack2 pass and no separate gathers.
+constexpr int kMaxCarry = 4;
+
+struct CarryCols {
+  const int32_t* in[kMaxCarry];
+  int32_t* out[kMaxCarry];
+  int k;
+};
+
+__device__ __forceinline__ int64_t fused_key(const int32_t* a_col, int64_t a_const,
+                                             const int32_t* b_col, int64_t b_const,
+                                             int64_t i) {
+  int64_t a = a_col ? static_cast<int64_t>(a_col[i]) : a_const;
+  int64_t b = b_col ? static_cast<int64_t>(b_col[i]) : b_const;
+  return (a << 32) | (b & 0xFFFFFFFFLL);
+}
+
+__global__ void probe_count_fused(const int64_t* __restrict__ key12, int64_t n,
+                                  const int32_t* __restrict__ a_col, int64_t a_const,
+                                  const int32_t* __restrict__ b_col, int64_t b_const,
+                                  int64_t m,
+                                  int64_t* __restrict__ lo_out,
+                                  int32_t* __restrict__ cnt_out) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < m;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t k = fused_key(a_col, a_const, b_col, b_const, i);
+    int64_t lo = lower_bound_i64(key12, n, k);
+    int64_t hi = upper_bound_i64(key12, n, k);
+    lo_out[i] = lo;
+    cnt_out[i] = static_cast<int32_t>(hi - lo);
+  }
+}
+
+__global__ void tile_bounds_fused(const int64_t* __restrict__ key12, int64_t n,
+                                  const int32_t* __restrict__ a_col, int64_t a_const,
+                                  const int32_t* __restrict__ b_col, int64_t b_const,
+                                  int64_t m, int64_t n_tiles,
+                                  int64_t* __restrict__ win_lo,
+                                  int64_t* __restrict__ win_hi) {
+  for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; t < n_tiles;
+       t += (int64_t)gridDim.x * blockDim.x) {
+    int64_t first = fused_key(a_col, a_const, b_col, b_const, t * kTile);
+    int64_t last = fused_key(a_col, a_const, b_col, b_const,
+                             min((t + 1) * kTile - 1, m - 1));
+    win_lo[t] = lower_bound_i64(key12, n, first);
+    win_hi[t] = upper_bound_i64(key12, n, last);
+  }
+}
+
+__global__ void probe_count_fused_sorted(const int64_t* __restrict__ key12,
+                                         const int32_t* __restrict__ a_col, int64_t a_const,
+                                         const int32_t* __restrict__ b_col, int64_t b_const,
+                                         int64_t m,
+                                         const int64_t* __restrict__ win_lo,
+                                         const int64_t* __restrict__ win_hi,
+                                         int64_t* __restrict__ lo_out,
+                                         int32_t* __restrict__ cnt_out) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < m;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t t = i / kTile;
+    int64_t wlo = win_lo[t];
+    int64_t wspan = win_hi[t] - wlo;
+    int64_t k = fused_key(a_col, a_const, b_col, b_const, i);
+    int64_t lo = wlo + lower_bound_i64(key12 + wlo, wspan, k);
+    int64_t hi = wlo + upper_bound_i64(key12 + wlo, wspan, k);
+    lo_out[i] = lo;
+    cnt_out[i] = static_cast<int32_t>(hi - lo);
+  }
+}
+
+__global__ void probe_emit_carry(const int64_t* __restrict__ key12,
+                                 const int32_t* __restrict__ z, int64_t m,
+                                 const int64_t* __restrict__ lo,
+                                 const int32_t* __restrict__ cnt,
+                                 const int64_t* __restrict__ offs,
+                                 CarryCols carry,
+                                 int64_t* __restrict__ li_out,
+                                 int32_t* __restrict__ b_out,
+                                 int32_t* __restrict__ z_out,
+                                 bool emit_b) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < m;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t base = offs[i];
+    int64_t l = lo[i];
+    int32_t c = cnt[i];
+    int32_t cv[kMaxCarry];
+    for (int j = 0; j < carry.k; ++j) cv[j] = carry.in[j][i];
+    for (int32_t q = 0; q < c; ++q) {
+      li_out[base + q] = i;
+      if (emit_b)
+        b_out[base + q] = static_cast<int32_t>(key12[l + q] & 0xFFFFFFFFLL);
+      z_out[base + q] = z[l + q];
+      for (int j = 0; j < carry.k; ++j) carry.out[j][base + q] = cv[j];
+    }
+  }
+}
+
+__global__ void probe_emit_carry_balanced(const int64_t* __restrict__ key12,
+                                          const int32_t* __restrict__ z, int64_t m,
+                                          const int64_t* __restrict__ lo,
+                                          const int64_t* __restrict__ offs,  // [m+1]
+                                          int64_t total, CarryCols carry,
+                                          int64_t* __restrict__ li_out,
+                                          int32_t* __restrict__ b_out,
+                                          int32_t* __restrict__ z_out,
+                                          bool emit_b) {
+  for (int64_t idx = blockIdx.x * (int64_t)blockDim.x + threadIdx.x;
+       idx < total; idx += (int64_t)gridDim.x * blockDim.x) {
+    int64_t a = 0, b = m;
+    while (a + 1 < b) {
+      int64_t mid = (a + b) >> 1;
+      if (offs[mid] <= idx) a = mid; else b = mid;
+    }
+    int64_t j = idx - offs[a];
+    int64_t src = lo[a] + j;
+    li_out[idx] = a;
+    if (emit_b)
+      b_out[idx] = static_cast<int32_t>(key12[src] & 0xFFFFFFFFLL);
+    z_out[idx] = z[src];
+    for (int c = 0; c < carry.k; ++c) carry.out[c][idx] = carry.in[c][a];
+  }
+}
+
 // ------------------------------------------------------------- K2: hash join
 constexpr int kMaxKeyCols = 4;
 
@@ -469,6 +596,105 @@ std::vector<at::Tensor> probe_range(at::Tensor key12, at::Tensor z,
   return emit_phase(key12, z, lo, cnt, m, stream);
 }
 
+// K1 fused probe: inline key packing, optional merge-path for sorted keys,
+// carry columns gathered during emit.  a/b: per-row column or constant.
+std::vector<at::Tensor> probe_fused(at::Tensor key12, at::Tensor z,
+                                    c10::optional<at::Tensor> a_col,
+                                    int64_t a_const,
+                                    c10::optional<at::Tensor> b_col,
+                                    int64_t b_const,
+                                    std::vector<at::Tensor> carry_cols,
+                                    bool emit_b) {
+  TORCH_CHECK(key12.is_cuda() && z.is_cuda());
+  TORCH_CHECK(carry_cols.size() <= static_cast<size_t>(kMaxCarry));
+  int64_t m = a_col.has_value() ? a_col->numel()
+                                : (b_col.has_value() ? b_col->numel() : 0);
+  TORCH_CHECK(m > 0, "probe_fused: need at least one probe column");
+  auto n = key12.numel();
+  auto opts_long = key12.options();
+  auto opts_int = z.options();
+  const int32_t* a_ptr = a_col.has_value() ? a_col->data_ptr<int32_t>() : nullptr;
+  const int32_t* b_ptr = b_col.has_value() ? b_col->data_ptr<int32_t>() : nullptr;
+  auto lo = at::empty({m}, opts_long);
+  auto cnt = at::empty({m}, opts_int);
+  auto stream = cur_stream();
+  // merge-path when the single varying column is sorted (signed order is
+  // the packed order when values share a sign; negatives skip the check)
+  bool sorted = false;
+  if (m >= 262144 && (a_ptr == nullptr) != (b_ptr == nullptr)) {
+    const auto& col = a_col.has_value() ? *a_col : *b_col;
+    bool need_nonneg = b_col.has_value();  // low 32 bits compare unsigned
+    auto mono = at::all(col.slice(0, 1, m) >= col.slice(0, 0, m - 1));
+    if (need_nonneg) {
+      sorted = (mono & (col[0] >= 0)).item<bool>();
+    } else {
+      sorted = mono.item<bool>();
+    }
+  }
+  if (sorted) {
+    int64_t n_tiles = (m + kTile - 1) / kTile;
+    auto wlo = at::empty({n_tiles}, opts_long);
+    auto whi = at::empty({n_tiles}, opts_long);
+    hipLaunchKernelGGL(tile_bounds_fused, dim3(grid_for(n_tiles)),
+                       dim3(kBlock), 0, stream, key12.data_ptr<int64_t>(), n,
+                       a_ptr, a_const, b_ptr, b_const, m, n_tiles,
+                       wlo.data_ptr<int64_t>(), whi.data_ptr<int64_t>());
+    HIP_OK(hipGetLastError());
+    hipLaunchKernelGGL(probe_count_fused_sorted, dim3(grid_for(m)),
+                       dim3(kBlock), 0, stream, key12.data_ptr<int64_t>(),
+                       a_ptr, a_const, b_ptr, b_const, m,
+                       wlo.data_ptr<int64_t>(), whi.data_ptr<int64_t>(),
+                       lo.data_ptr<int64_t>(), cnt.data_ptr<int32_t>());
+  } else {
+    hipLaunchKernelGGL(probe_count_fused, dim3(grid_for(m)), dim3(kBlock), 0,
+                       stream, key12.data_ptr<int64_t>(), n, a_ptr, a_const,
+                       b_ptr, b_const, m, lo.data_ptr<int64_t>(),
+                       cnt.data_ptr<int32_t>());
+  }
+  HIP_OK(hipGetLastError());
+  auto offs = at::cumsum(cnt, 0, at::kLong);
+  int64_t total = offs[-1].item<int64_t>();
+  auto li = at::empty({total}, opts_long);
+  auto b_out = at::empty({emit_b ? total : 0}, opts_int);
+  auto zz = at::empty({total}, opts_int);
+  CarryCols carry{};
+  carry.k = static_cast<int>(carry_cols.size());
+  std::vector<at::Tensor> carried;
+  for (size_t j = 0; j < carry_cols.size(); ++j) {
+    TORCH_CHECK(carry_cols[j].is_cuda()
+                && carry_cols[j].dtype() == at::kInt
+                && carry_cols[j].numel() == m);
+    carry.in[j] = carry_cols[j].data_ptr<int32_t>();
+    carried.push_back(at::empty({total}, opts_int));
+    carry.out[j] = carried.back().data_ptr<int32_t>();
+  }
+  if (total > 0) {
+    int32_t* b_ptr_out = emit_b ? b_out.data_ptr<int32_t>() : nullptr;
+    if (total >= 8 * m) {
+      auto offs_full = at::zeros({m + 1}, opts_long);
+      offs_full.narrow(0, 1, m).copy_(offs);
+      hipLaunchKernelGGL(probe_emit_carry_balanced, dim3(grid_for(total)),
+                         dim3(kBlock), 0, stream, key12.data_ptr<int64_t>(),
+                         z.data_ptr<int32_t>(), m, lo.data_ptr<int64_t>(),
+                         offs_full.data_ptr<int64_t>(), total, carry,
+                         li.data_ptr<int64_t>(), b_ptr_out,
+                         zz.data_ptr<int32_t>(), emit_b);
+    } else {
+      auto offs_excl = offs - cnt.to(at::kLong);
+      hipLaunchKernelGGL(probe_emit_carry, dim3(grid_for(m)), dim3(kBlock), 0,
+                         stream, key12.data_ptr<int64_t>(),
+                         z.data_ptr<int32_t>(), m, lo.data_ptr<int64_t>(),
+                         cnt.data_ptr<int32_t>(), offs_excl.data_ptr<int64_t>(),
+                         carry, li.data_ptr<int64_t>(), b_ptr_out,
+                         zz.data_ptr<int32_t>(), emit_b);
+    }
+    HIP_OK(hipGetLastError());
+  }
+  std::vector<at::Tensor> out = {li, b_out, zz};
+  for (auto& t : carried) out.push_back(t);
+  return out;
+}
+
 // K1 count-only variants: COUNT(*) queries skip the emit pass entirely.
 at::Tensor probe_exact_counts(at::Tensor key12, at::Tensor keys) {
   TORCH_CHECK(key12.is_cuda() && keys.is_cuda());
@@ -624,6 +850,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "K1 scan-probe, leading-component range -> (li, b, z)");
   m.def("hash_join", &hash_join,
         "K2 chained hash join over int32 key columns -> (li, ri)");
+  m.def("probe_fused", &probe_fused,
+        "K1 fused probe: inline key pack + merge-path + carry emit");
   m.def("probe_exact_counts", &probe_exact_counts,
         "K1 count-only exact probe -> per-row match counts");
   m.def("probe_range_counts", &probe_range_counts,
