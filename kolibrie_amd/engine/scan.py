@@ -173,8 +173,12 @@ def scan_probe(
     native = native_for(key12)
     if native is not None and plen > 0:
         if plen == 2:
-            keys = pack2(col_for(pos[0]).contiguous(), col_for(pos[1]).contiguous())
-            li, b_col, z_col = native.probe_exact(key12, z, keys.contiguous())
+            a_col = probes[pos[0]].contiguous() if pos[0] in probes else None
+            a_const = consts.get(pos[0], 0)
+            b_colp = probes[pos[1]].contiguous() if pos[1] in probes else None
+            b_const = consts.get(pos[1], 0)
+            li, b_col, z_col = native.probe_fused(
+                key12, z, a_col, a_const, b_colp, b_const, [], True)
         else:
             li, b_col, z_col = native.probe_range(key12, z,
                                                   col_for(pos[0]).contiguous())

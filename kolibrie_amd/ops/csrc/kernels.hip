@@ -615,6 +615,21 @@ std::vector<at::Tensor> probe_fused(at::Tensor key12, at::Tensor z,
   auto opts_int = z.options();
   const int32_t* a_ptr = a_col.has_value() ? a_col->data_ptr<int32_t>() : nullptr;
   const int32_t* b_ptr = b_col.has_value() ? b_col->data_ptr<int32_t>() : nullptr;
+  // constant leading component: narrow every search to that contiguous
+  // region ONCE (e.g. one predicate's slice) — the region usually stays
+  // cache-resident, cutting the dependent-load chain per probe
+  if (!a_ptr && n > 0) {
+    auto reg_keys = at::empty({2}, opts_long);
+    reg_keys[0] = (a_const << 32);
+    reg_keys[1] = (a_const << 32) | 0xFFFFFFFFLL;
+    auto lo_t = at::searchsorted(key12, reg_keys[0], false, false);
+    auto hi_t = at::searchsorted(key12, reg_keys[1], false, true);
+    int64_t roff = lo_t.item<int64_t>();
+    int64_t rend = hi_t.item<int64_t>();
+    key12 = key12.narrow(0, roff, rend - roff);
+    z = z.narrow(0, roff, rend - roff);
+    n = rend - roff;
+  }
   auto lo = at::empty({m}, opts_long);
   auto cnt = at::empty({m}, opts_int);
   auto stream = cur_stream();
