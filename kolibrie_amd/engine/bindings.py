@@ -15,12 +15,17 @@ from ..storage.terms import UNBOUND
 
 
 class Bindings:
-    __slots__ = ("cols", "n", "device")
+    __slots__ = ("cols", "n", "device", "maybe_unbound")
 
-    def __init__(self, cols: Dict[str, torch.Tensor], n: int, device):
+    def __init__(self, cols: Dict[str, torch.Tensor], n: int, device,
+                 maybe_unbound: bool = False):
         self.cols = cols
         self.n = n
         self.device = torch.device(device)
+        # conservative taint: True if any column MAY contain UNBOUND cells
+        # (VALUES UNDEF, UNION fills, SUBJECT/... of non-triples).  False
+        # lets hot paths skip per-row boundness scans entirely.
+        self.maybe_unbound = maybe_unbound
 
     # ------------------------------------------------------------- factories
     @staticmethod
@@ -46,11 +51,14 @@ class Bindings:
                     vars_.append(k)
         n = len(rows)
         cols = {}
+        any_unbound = False
         for v in vars_:
-            data = [(-1 if r.get(v) is None else (r[v] & 0xFFFFFFFF)) for r in rows]
+            raw = [r.get(v) for r in rows]
+            any_unbound |= any(x is None for x in raw)
+            data = [(-1 if x is None else (x & 0xFFFFFFFF)) for x in raw]
             data = [x - 0x1_0000_0000 if x >= 0x8000_0000 else x for x in data]
             cols[v] = torch.tensor(data, dtype=torch.int32, device=dev)
-        return Bindings(cols, n, dev)
+        return Bindings(cols, n, dev, maybe_unbound=any_unbound)
 
     # ------------------------------------------------------------- accessors
     @property
@@ -81,28 +89,35 @@ class Bindings:
     # ------------------------------------------------------------ operations
     def select(self, mask: torch.Tensor) -> "Bindings":
         n = int(mask.sum().item())
-        return Bindings({v: c[mask] for v, c in self.cols.items()}, n, self.device)
+        return Bindings({v: c[mask] for v, c in self.cols.items()}, n,
+                        self.device, self.maybe_unbound)
 
     def gather(self, idx: torch.Tensor) -> "Bindings":
-        return Bindings({v: c[idx] for v, c in self.cols.items()}, idx.numel(), self.device)
+        return Bindings({v: c[idx] for v, c in self.cols.items()},
+                        idx.numel(), self.device, self.maybe_unbound)
 
-    def with_col(self, var: str, col: torch.Tensor) -> "Bindings":
+    def with_col(self, var: str, col: torch.Tensor,
+                 col_maybe_unbound: bool = False) -> "Bindings":
         cols = dict(self.cols)
         cols[var] = col
-        return Bindings(cols, self.n, self.device)
+        return Bindings(cols, self.n, self.device,
+                        self.maybe_unbound or col_maybe_unbound)
 
     def project(self, vars_: Sequence[str]) -> "Bindings":
         cols = {}
+        filled = False
         for v in vars_:
             if v in self.cols:
                 cols[v] = self.cols[v]
             else:
+                filled = True
                 cols[v] = torch.full((self.n,), UNBOUND, dtype=torch.int32, device=self.device)
-        return Bindings(cols, self.n, self.device)
+        return Bindings(cols, self.n, self.device,
+                        self.maybe_unbound or (filled and self.n > 0))
 
     def drop_cols(self, vars_: Sequence[str]) -> "Bindings":
         cols = {v: c for v, c in self.cols.items() if v not in vars_}
-        return Bindings(cols, self.n, self.device)
+        return Bindings(cols, self.n, self.device, self.maybe_unbound)
 
     @staticmethod
     def concat(parts: List["Bindings"], device) -> "Bindings":
@@ -118,15 +133,19 @@ class Bindings:
         n = sum(p.n for p in parts)
         dev = torch.device(device)
         cols = {}
+        filled = False
         for v in vars_:
             pieces = []
             for p in parts:
                 if v in p.cols:
                     pieces.append(p.cols[v])
                 else:
+                    if p.n > 0:
+                        filled = True
                     pieces.append(torch.full((p.n,), UNBOUND, dtype=torch.int32, device=dev))
             cols[v] = torch.cat(pieces) if pieces else torch.empty(0, dtype=torch.int32, device=dev)
-        return Bindings(cols, n, dev)
+        tainted = filled or any(p.maybe_unbound for p in parts)
+        return Bindings(cols, n, dev, tainted)
 
     def repeat_rows(self, k: int) -> "Bindings":
         """Each row repeated k times consecutively."""
@@ -134,8 +153,10 @@ class Bindings:
             {v: torch.repeat_interleave(c, k) for v, c in self.cols.items()},
             self.n * k,
             self.device,
+            self.maybe_unbound,
         )
 
     def tile_rows(self, k: int) -> "Bindings":
         """Whole table repeated k times."""
-        return Bindings({v: c.repeat(k) for v, c in self.cols.items()}, self.n * k, self.device)
+        return Bindings({v: c.repeat(k) for v, c in self.cols.items()},
+                        self.n * k, self.device, self.maybe_unbound)

@@ -117,7 +117,8 @@ class ExecutionEngine:
             if rows.is_empty():
                 return rows.project(rows.variables + [op.var])
             col = op.expr.eval_ids(rows, self.db)
-            return _prune(rows.with_col(op.var, col), needed)
+            taint = getattr(op.expr, "may_produce_unbound", False)
+            return _prune(rows.with_col(op.var, col, taint), needed)
         if isinstance(op, PValues):
             rows = self.execute(op.input, incoming)
             vals = self._values_bindings(op.variables, op.rows)
@@ -146,10 +147,12 @@ class ExecutionEngine:
         dev = self.device
         n = len(rows)
         cols = {}
+        any_undef = False
         for j, v in enumerate(variables):
+            any_undef |= any(r[j] is None for r in rows)
             data = [(UNBOUND if r[j] is None else r[j]) for r in rows]
             cols[v] = torch.tensor(data, dtype=torch.int32, device=dev)
-        return Bindings(cols, n, dev)
+        return Bindings(cols, n, dev, maybe_unbound=any_undef)
 
     # ----------------------------------------------------------------- scan --
     def _exec_scan(self, pattern: TriplePattern, scope, incoming: Bindings,
@@ -235,7 +238,33 @@ class ExecutionEngine:
             exec_stats.bump("ROWS_EMITTED", out.n)
             return self._add_extra(out, extra)
 
-        # group rows by which probe vars are actually bound (UNDEF handling)
+        # group rows by which probe vars are actually bound (UNDEF handling);
+        # untainted inputs (no UNBOUND anywhere) skip the grouping scans
+        if not inc.maybe_unbound:
+            if needed is not None and not needed and not scan_need \
+                    and not qt_pos and extra is None:
+                from .scan import scan_probe_count
+                cnt = scan_probe_count(
+                    idx, consts, {i: inc.col(v)
+                                  for i, v in probe_vars.items()})
+                if cnt is not None:
+                    exec_stats.bump("ROWS_EMITTED", cnt)
+                    return Bindings({}, cnt, dev)
+            probes = {i: inc.col(v) for i, v in probe_vars.items()}
+            if needed is None:
+                carry_src = inc
+            else:
+                pattern_names = set(var_pos.values())
+                carry_src = _prune(inc, set(needed) - pattern_names)
+            from .scan import scan_probe_carry
+            li, s, p, o, carried = scan_probe_carry(
+                idx, consts, probes, scan_need, dict(carry_src.cols))
+            exec_stats.bump("QUADS_EXAMINED", li.numel())
+            base = Bindings(carried, li.numel(), dev)
+            out = self._build_candidate(s, p, o, var_pos, qt_pos, base,
+                                        needed, li.numel())
+            exec_stats.bump("ROWS_EMITTED", out.n)
+            return self._add_extra(out, extra)
         masks = {i: (inc.col(v) != UNBOUND) for i, v in probe_vars.items()}
         sig = torch.zeros(inc.n, dtype=torch.int64, device=dev)
         for k, i in enumerate(sorted(masks)):
@@ -409,15 +438,19 @@ def join_bindings(left: Bindings, right: Bindings, needed=None) -> Bindings:
         li = torch.arange(left.n, dtype=torch.long, device=dev).repeat_interleave(right.n)
         ri = torch.arange(right.n, dtype=torch.long, device=dev).repeat(left.n)
         return _merge_pairs(left, right, li, ri, shared, needed)
-    lb = torch.ones(left.n, dtype=torch.bool, device=dev)
-    for v in shared:
-        lb &= left.col(v) != UNBOUND
-    rb = torch.ones(right.n, dtype=torch.bool, device=dev)
-    for v in shared:
-        rb &= right.col(v) != UNBOUND
     parts: List[Bindings] = []
-    l_keyed, l_unkeyed = left.select(lb), left.select(~lb)
-    r_keyed, r_unkeyed = right.select(rb), right.select(~rb)
+    if not left.maybe_unbound and not right.maybe_unbound:
+        l_keyed, l_unkeyed = left, Bindings.empty(dev, left.variables)
+        r_keyed, r_unkeyed = right, Bindings.empty(dev, right.variables)
+    else:
+        lb = torch.ones(left.n, dtype=torch.bool, device=dev)
+        for v in shared:
+            lb &= left.col(v) != UNBOUND
+        rb = torch.ones(right.n, dtype=torch.bool, device=dev)
+        for v in shared:
+            rb &= right.col(v) != UNBOUND
+        l_keyed, l_unkeyed = left.select(lb), left.select(~lb)
+        r_keyed, r_unkeyed = right.select(rb), right.select(~rb)
     count_only = needed is not None and len(needed) == 0
     if count_only and l_keyed.n and r_keyed.n:
         # COUNT(*) pushdown: match counts without emitting pairs

@@ -222,6 +222,10 @@ class CompiledBind:
         self.ast = ast
         self.prefixes = prefixes or {}
         CompiledExpr(ast, db)  # prepares literal ids/values in-place
+        # SUBJECT/PREDICATE/OBJECT of a non-triple yield UNBOUND cells
+        self.may_produce_unbound = (
+            isinstance(ast, EFunc)
+            and ast.name in ("SUBJECT", "PREDICATE", "OBJECT"))
 
     def eval_ids(self, b: Bindings, db) -> torch.Tensor:
         e = self.ast
