@@ -79,6 +79,25 @@ def _cols_from_order(idx: GraphIndex, code: int, sel: Optional[torch.Tensor],
     return out[0], out[1], out[2]
 
 
+def _cols_from_order_slice(idx: GraphIndex, code: int, lo: int, hi: int,
+                           need: frozenset = ALL_POSITIONS):
+    """Contiguous-range variant: views + unpack, no index gather."""
+    key12, z = idx.orders[code]
+    pos = _ORDER_POS[code]
+    out = [None, None, None]
+    k = key12[lo:hi]
+    if pos[0] in need and pos[1] in need:
+        a, b = unpack2(k)
+        out[pos[0]], out[pos[1]] = a, b
+    elif pos[0] in need:
+        out[pos[0]] = (k >> 32).to(torch.int32)
+    elif pos[1] in need:
+        out[pos[1]] = (k & 0xFFFFFFFF).to(torch.int32)
+    if pos[2] in need:
+        out[pos[2]] = z[lo:hi]
+    return out[0], out[1], out[2]
+
+
 def scan_unit(idx: GraphIndex, consts: Dict[int, int],
               sort_hint: int = None, need=None):
     """Scan with only constant bounds: one contiguous range slice.
@@ -116,8 +135,7 @@ def scan_unit(idx: GraphIndex, consts: Dict[int, int],
                         torch.tensor([-1], dtype=torch.int32, device=dev))
             lo = int(torch.searchsorted(key12, klo, side="left").item())
             hi = int(torch.searchsorted(key12, khi, side="right").item())
-        sel = torch.arange(lo, hi, dtype=torch.long, device=dev)
-        s, p, o = _cols_from_order(idx, code, sel, mat_need)
+        s, p, o = _cols_from_order_slice(idx, code, lo, hi, mat_need)
         n_rows = hi - lo
     # post-filter constants not covered by the prefix
     if uncovered:
