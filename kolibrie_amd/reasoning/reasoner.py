@@ -175,6 +175,86 @@ class Reasoner:
         self._flush()
         return self.facts.index().contains(*[i & 0xFFFFFFFF for i in ids])  # type: ignore
 
+    # ------------------------------------------------ provenance fixpoints --
+    def infer_new_facts_with_provenance(self, semiring=None,
+                                        seeds: Optional[Dict] = None):
+        """Tagged materialisation (ref provenance_semi_naive.rs:210).
+
+        `seeds`: {(s,p,o) id triple: probability}; defaults to
+        `probability_seeds`; facts without a seed get one().  Scalar
+        semirings (minmax/addmult/expiration) run the device-tagged K6
+        path; structured semirings (TopK, DNF-WMC, SDD) run the host
+        oracle.  Returns {triple: tag}.
+        """
+        from .provenance import MinMaxProbability, Provenance
+        self._flush()
+        if semiring is None:
+            semiring = MinMaxProbability()
+        if isinstance(semiring, str):
+            from .device_tags import ScalarSemiring
+            try:
+                sr = ScalarSemiring(semiring)
+            except ValueError:
+                from .provenance import semiring_by_name
+                semiring = semiring_by_name(semiring)
+            else:
+                return self._infer_device_tags(sr, seeds)
+        if getattr(semiring, "name", "") in ("minmax", "addmult", "expiration"):
+            from .device_tags import ScalarSemiring
+            return self._infer_device_tags(ScalarSemiring(semiring.name), seeds)
+        seed_map = self._seed_tags(semiring, seeds)
+        from .provenance_fixpoint import infer_with_provenance
+        return infer_with_provenance(self.rules, seed_map, semiring, self.db)
+
+    def _seed_tags(self, semiring, seeds):
+        raw = seeds if seeds is not None else self.probability_seeds
+        seed_map = {}
+        # deterministic seed-id assignment: sorted by triple
+        # (ref provenance_semi_naive.rs:220-228)
+        for t in sorted(self.all_fact_tuples()):
+            p = raw.get(t)
+            if p is None:
+                seed_map[t] = semiring.one()
+            else:
+                seed_map[t] = semiring.tag_from_probability(p)
+        return seed_map
+
+    def _infer_device_tags(self, sr, seeds):
+        from .device_tags import infer_with_provenance_device
+        raw = seeds if seeds is not None else self.probability_seeds
+        seed_map = {t: float(raw.get(t, 1.0))
+                    for t in sorted(self.all_fact_tuples())}
+        return infer_with_provenance_device(
+            self.rules, seed_map, sr, device=str(self.device), db=self.db)
+
+    def infer_with_sdd_seeds(self, seeds: Optional[Dict] = None):
+        """Semi-naive with SddProvenance initial tags (ref
+        sdd_seed_materialise.rs): returns ({triple: sdd node}, manager)."""
+        from .sdd import SddProvenance
+        prov = SddProvenance()
+        seed_map = self._seed_tags(prov, seeds)
+        from .provenance_fixpoint import infer_with_provenance
+        tags = infer_with_provenance(self.rules, seed_map, prov, self.db)
+        return tags, prov
+
+    # ---------------------------------------------------------- IAR queries --
+    def query_with_repairs(self, s=None, p=None, o=None, max_size: int = 3):
+        """Inconsistency-tolerant (IAR) answers: facts derivable under EVERY
+        minimal repair (ref semi_naive_with_repairs.rs / repairs.rs)."""
+        repairs = self.compute_repairs(max_size=max_size)
+        results = None
+        base = sorted(self.all_fact_tuples())
+        for removed in repairs:
+            trial = Reasoner(device=str(self.device), dictionary=self.dictionary)
+            for f in base:
+                if f not in removed:
+                    trial.add_fact_ids(*f)
+            trial.rules = self.rules
+            trial.infer_new_facts_semi_naive()
+            answers = set(trial.query_abox(s, p, o))
+            results = answers if results is None else (results & answers)
+        return sorted(results or [])
+
     # -------------------------------------------------- backward chaining --
     def backward_chaining(self, goal, max_depth: int = 24):
         """SLD goal resolution (ref backward_chaining.rs:150).  `goal` is a

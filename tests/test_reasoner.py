@@ -133,3 +133,64 @@ def test_deep_taxonomy_scaling():
     n = r.infer_new_facts_semi_naive()
     assert n == depth
     assert r.contains_fact("i0", "type", f"C{depth}")
+
+
+def test_provenance_entry_on_reasoner():
+    r = Reasoner()
+    r.add_abox_triple("a", "p", "b")
+    r.add_abox_triple("b", "p", "c")
+    a = r.dictionary.encode("a")
+    p = r.dictionary.encode("p")
+    b = r.dictionary.encode("b")
+    c = r.dictionary.encode("c")
+    q = r.dictionary.encode("q")
+    r.probability_seeds[(a, p, b)] = 0.9
+    r.probability_seeds[(b, p, c)] = 0.6
+    r.add_rule(Rule(
+        premise=[_tp(r, "?x", "p", "?y"), _tp(r, "?y", "p", "?z")],
+        conclusion=[_tp(r, "?x", "q", "?z")],
+    ))
+    tags = r.infer_new_facts_with_provenance("minmax")
+    assert abs(tags[(a, q, c)] - 0.6) < 1e-6
+
+
+def test_sdd_seeded_materialisation():
+    r = Reasoner()
+    r.add_abox_triple("a", "p", "b")
+    a = r.dictionary.encode("a")
+    p = r.dictionary.encode("p")
+    b = r.dictionary.encode("b")
+    q = r.dictionary.encode("q")
+    r.probability_seeds[(a, p, b)] = 0.7
+    r.add_rule(Rule(premise=[_tp(r, "?x", "p", "?y")],
+                    conclusion=[_tp(r, "?x", "q", "?y")]))
+    tags, prov = r.infer_with_sdd_seeds()
+    assert abs(prov.recover(tags[(a, q, b)]) - 0.7) < 1e-6
+
+
+def test_query_with_repairs_iar():
+    r = Reasoner()
+    r.add_abox_triple("x", "status", "on")
+    r.add_abox_triple("x", "status", "off")
+    r.add_abox_triple("y", "status", "on")
+    r.add_constraint(Rule(
+        premise=[_tp(r, "?s", "status", "on"), _tp(r, "?s", "status", "off")],
+        conclusion=[],
+    ))
+    # y's status survives every repair; x's conflicting facts do not
+    answers = r.query_with_repairs(None, "status", None)
+    assert ("y", "status", "on") in answers
+    assert ("x", "status", "on") not in answers
+    assert ("x", "status", "off") not in answers
+
+
+def test_window_runner():
+    from kolibrie_amd.rsp.s2r import CSPARQLWindow, Report, ReportStrategy, Tick
+    from kolibrie_amd.rsp.window_runner import WindowRunner
+    rep = Report()
+    rep.add(ReportStrategy.ON_WINDOW_CLOSE)
+    runner = WindowRunner(CSPARQLWindow(5, 5, rep, Tick.TIME_DRIVEN, "w"))
+    for ts in range(0, 11):
+        runner.push(("e", ts), ts)
+    fired = runner.drain()
+    assert len(fired) == 2
