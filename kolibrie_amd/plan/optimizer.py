@@ -286,8 +286,7 @@ class Streamertail:
         # subject-sorted (PSO) output whenever the predicate is constant and
         # the subject free: costs nothing, enables downstream merge joins
         hint = 0 if (isinstance(pattern.p, Constant)
-                     and isinstance(pattern.s, Variable)
-                     and pattern.s.name not in bound) else None
+                     and isinstance(pattern.s, Variable)) else None
         if n_bound >= 1:
             return PIndexScan(pattern, graph, sort_hint=hint)
         return PTableScan(pattern, graph, sort_hint=hint)
