@@ -68,9 +68,11 @@ def main():
     n_local = ddb.db.triple_count()
     log(rank, f"[bench] shard loaded: {n_local:,} triples in {time.time()-t0:.1f}s")
 
+    # project only the join key: the engine's projection pushdown then
+    # keeps the whole local pipeline column-minimal
     local_star = f"""
         PREFIX ds: <{DS}>
-        SELECT ?e ?d ?sal WHERE {{
+        SELECT ?d WHERE {{
             ?e ds:worksFor ?d .
             ?e ds:annual_salary ?sal .
         }}"""

@@ -129,3 +129,26 @@ def _p2p_exchange(out: torch.Tensor, inp: torch.Tensor,
             reqs.append(dist.isend(inp[send_offs[peer]:send_offs[peer + 1]].contiguous(), dst=peer))
     for r in reqs:
         r.wait()
+
+
+def all_gather_rows(cols: List[torch.Tensor]) -> List[torch.Tensor]:
+    """Replicate a small row table on every rank (broadcast join build
+    side): variable-size all_gather per column."""
+    if not is_dist():
+        return list(cols)
+    world = dist.get_world_size()
+    device = cols[0].device
+    n = torch.tensor([cols[0].numel()], dtype=torch.int64,
+                     device=device if dist.get_backend() == "nccl" else "cpu")
+    sizes = [torch.zeros_like(n) for _ in range(world)]
+    dist.all_gather(sizes, n)
+    sizes = [int(s.item()) for s in sizes]
+    max_n = max(sizes) if sizes else 0
+    out: List[torch.Tensor] = []
+    for c in cols:
+        pad = torch.zeros(max_n, dtype=c.dtype, device=device)
+        pad[:c.numel()] = c
+        gathered = [torch.zeros_like(pad) for _ in range(world)]
+        dist.all_gather(gathered, pad)
+        out.append(torch.cat([g[:sizes[i]] for i, g in enumerate(gathered)]))
+    return out

@@ -38,29 +38,41 @@ class DistributedDatabase:
     def load_shard_columns(self, s, p, o):
         self.db.load_columns(s, p, o)
 
+    BROADCAST_MAX_ROWS = 4_000_000
+
     def count_query_with_shuffle(
         self,
         local_star_sparql: str,
         shuffle_var: str,
         probe_sparql: str,
     ) -> int:
-        """Execute: local subquery -> shuffle rows by `shuffle_var` ->
-        local join with `probe_sparql` results -> global COUNT.
+        """Execute: local subquery -> exchange on `shuffle_var` -> local
+        join with `probe_sparql` results -> global COUNT.
 
-        Both subqueries run through the full parse/optimize/execute
-        pipeline on the local shard.
+        Exchange strategy (cost-based): when the
+        probe side is small it is REPLICATED with an all-gather (broadcast
+        join — the xGMI traffic is the small table, not the big
+        intermediate); otherwise both sides hash-repartition with the
+        pairwise all-to-all row shuffle.
         """
         left = self._rows_for(local_star_sparql)
         right = self._rows_for(probe_sparql)
-        if self.world > 1:
-            key = left.col(shuffle_var).to(torch.int64) & 0xFFFFFFFF
-            dest = key % self.world
-            names = left.variables
-            cols = D.all_to_all_rows([left.col(v) for v in names], dest)
-            left = Bindings(dict(zip(names, cols)), cols[0].numel() if cols else 0,
-                            self.device)
         from ..engine.executor import join_bindings
-        joined = join_bindings(left, right)
+        if self.world > 1:
+            right_total = D.allreduce_sum_scalar(right.n, self.device)
+            if right_total <= self.BROADCAST_MAX_ROWS:
+                names = right.variables
+                cols = D.all_gather_rows([right.col(v) for v in names])
+                right = Bindings(dict(zip(names, cols)),
+                                 cols[0].numel() if cols else 0, self.device)
+            else:
+                key = left.col(shuffle_var).to(torch.int64) & 0xFFFFFFFF
+                dest = key % self.world
+                names = left.variables
+                cols = D.all_to_all_rows([left.col(v) for v in names], dest)
+                left = Bindings(dict(zip(names, cols)),
+                                cols[0].numel() if cols else 0, self.device)
+        joined = join_bindings(left, right, needed=set())
         local = joined.n
         return D.allreduce_sum_scalar(local, self.device)
 
