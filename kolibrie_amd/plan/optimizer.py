@@ -140,9 +140,15 @@ class Streamertail:
             return sb or pb or ob
 
         while remaining:
-            connected = [s for s in remaining
-                         if anchored(s) or any(v in cur_bound for v in _pattern_vars(s.pattern))]
-            pool = connected if connected else remaining
+            if ordered:
+                # after the seed: only join-connected patterns (cartesian
+                # products are a last resort — ref greedy_order_scans:175)
+                connected = [s for s in remaining
+                             if any(v in cur_bound
+                                    for v in _pattern_vars(s.pattern))]
+                pool = connected if connected else remaining
+            else:
+                pool = remaining
             pool_anchored = [s for s in pool if anchored(s)]
             pick_from = pool_anchored if pool_anchored else pool
             best = min(pick_from, key=est)
