@@ -57,12 +57,13 @@ def test_greedy_picks_selective_anchor_first():
 
 def test_greedy_join_connectivity_invariant():
     st = Streamertail(chain_stats())
-    scans = [_scan("?a", 1, "?b"), _scan("?x", 2, "?y"), _scan("?b", 2, "?c")]
+    scans = [_scan("?a", 1, "?b"), _scan("?b", 2, "?c"), _scan("?c", 2, "?d")]
     ordered = st._greedy_order(scans, set())
-    # second pick must share a variable with what is already bound when a
-    # connected candidate exists
+    # every pick after the seed must share a variable with the bound set
     bound = set(ordered[0].pattern.variables())
-    assert any(v in bound for v in ordered[1].pattern.variables())
+    for s in ordered[1:]:
+        assert any(v in bound for v in s.pattern.variables())
+        bound.update(s.pattern.variables())
 
 
 def test_star_subgroup_detection():
