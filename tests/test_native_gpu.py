@@ -140,3 +140,101 @@ def test_reasoner_fixpoint_gpu():
     n = r.infer_new_facts_semi_naive()
     assert n == 1000 * 1001 // 2
     assert r.contains_fact("n0", "reach", "n1000")
+
+
+@requires_gpu
+def test_device_tags_fixpoint_gpu():
+    from kolibrie_amd.reasoning.device_tags import (
+        ScalarSemiring, infer_with_provenance_device,
+    )
+    from kolibrie_amd.reasoning.provenance import MinMaxProbability
+    from kolibrie_amd.reasoning.provenance_fixpoint import infer_with_provenance
+    from kolibrie_amd.reasoning.rule import Rule
+    from kolibrie_amd.storage.terms import Constant, TriplePattern, Variable
+    P, Q = 100, 101
+    rule = Rule(
+        premise=[TriplePattern(Variable("x"), Constant(P), Variable("y")),
+                 TriplePattern(Variable("y"), Constant(P), Variable("z"))],
+        conclusion=[TriplePattern(Variable("x"), Constant(Q), Variable("z"))],
+    )
+    import random
+    rng = random.Random(3)
+    seeds = {(rng.randint(1, 20), P, rng.randint(1, 20)): round(rng.random(), 3)
+             for _ in range(60)}
+    host = infer_with_provenance([rule], dict(seeds), MinMaxProbability())
+    dev = infer_with_provenance_device([rule], dict(seeds),
+                                       ScalarSemiring("minmax"),
+                                       device="cuda:0")
+    assert set(host) == set(dev)
+    for k in host:
+        assert abs(host[k] - dev[k]) < 1e-5
+
+
+@requires_gpu
+def test_bulk_streaming_gpu():
+    from kolibrie_amd.rsp import RSPBuilder
+    EX = "http://example.org/"
+    q = f"""PREFIX ex: <{EX}>
+REGISTER RSTREAM <http://out> AS
+SELECT (COUNT(*) AS ?c)
+FROM NAMED WINDOW <http://w1> ON STREAM <http://s1> [RANGE 10 STEP 10]
+WHERE {{ WINDOW <http://w1> {{ ?m ex:temp ?v }} }}"""
+    outputs = []
+    eng = (RSPBuilder(device="cuda:0").add_rsp_ql_query(q)
+           .add_consumer(outputs.append).build())
+    db = eng.store.db
+    temp = db.encode_term(f"<{EX}temp>")
+    n = 100_000
+    s = (torch.arange(n, dtype=torch.int32) % 977 + 50_000).cuda()
+    p = torch.full((n,), temp, dtype=torch.int32).cuda()
+    o = (torch.arange(n, dtype=torch.int32) % 4099 + 90_000).cuda()
+    ts = ((torch.arange(n, dtype=torch.int64) * 25) // n).cuda()
+    eng.add_to_stream_bulk("http://s1", s, p, o, ts)
+    counts = [int(rows[0][0]) for rows in outputs if rows]
+    assert len(counts) == 2
+    # set semantics: distinct (s,temp,o) triples with ts < 20
+    host = set()
+    tl = ts.cpu().tolist()
+    sl = s.cpu().tolist()
+    ol = o.cpu().tolist()
+    for i in range(n):
+        if tl[i] < 20:
+            host.add((sl[i], ol[i]))
+    # windows [0,10) and [10,20): each counts distinct triples in range
+    w1 = {(sl[i], ol[i]) for i in range(n) if tl[i] < 10}
+    w2 = {(sl[i], ol[i]) for i in range(n) if 10 <= tl[i] < 20}
+    assert counts == [len(w1), len(w2)]
+
+
+@requires_gpu
+def test_group_by_aggregate_gpu_equals_cpu():
+    from kolibrie_amd import SparqlDatabase
+    EX = "http://e/"
+    expect = None
+    for device in ("cpu", "cuda:0"):
+        db = SparqlDatabase(device=device)
+        for i in range(5000):
+            db.add_triple(f"<{EX}e{i}>", f"<{EX}dept>", f"<{EX}d{i % 13}>")
+            db.add_triple(f"<{EX}e{i}>", f"<{EX}sal>", f'"{1000 + (i % 50)}"')
+        rows = db.query(f"""
+            SELECT ?d (COUNT(?e) AS ?c) (AVG(?s) AS ?a) WHERE {{
+                ?e <{EX}dept> ?d . ?e <{EX}sal> ?s
+            }} GROUP BY ?d ORDER BY ?d""")
+        if expect is None:
+            expect = rows
+        else:
+            assert rows == expect
+
+
+@requires_gpu
+def test_prepared_plan_cache_gpu():
+    from kolibrie_amd import SparqlDatabase
+    EX = "http://e/"
+    db = SparqlDatabase(device="cuda:0")
+    for i in range(1000):
+        db.add_triple(f"<{EX}e{i}>", f"<{EX}p>", f'"{i}"')
+    q = f"SELECT (COUNT(*) AS ?c) WHERE {{ ?s <{EX}p> ?o }}"
+    assert db.query(q) == [["1000"]]
+    assert db.query(q) == [["1000"]]          # cache hit
+    db.add_triple(f"<{EX}extra>", f"<{EX}p>", '"x"')
+    assert db.query(q) == [["1001"]]          # invalidated by version bump
