@@ -68,6 +68,12 @@ class ExecutionEngine:
         if isinstance(op, PUnit):
             return incoming
         needed = getattr(op, "needed", None)
+        if (needed is not None and len(needed) == 0
+                and isinstance(op, (PBindJoin, PHashJoin))
+                and incoming.n == 1 and not incoming.cols):
+            fused = self._try_chain_count(op)
+            if fused is not None:
+                return Bindings({}, fused, self.device)
         if isinstance(op, (PTableScan, PIndexScan)):
             return self._exec_scan(op.pattern, op.graph, incoming, needed,
                                    sort_hint=getattr(op, "sort_hint", None))
@@ -140,6 +146,100 @@ class ExecutionEngine:
             from ..ml.predict import execute_ml_predict
             return execute_ml_predict(op.info, rows, self.db)
         raise ValueError(f"cannot execute {type(op).__name__}")
+
+    # --------------------------------------------- fused chain-count (K1+K4)
+    def _try_chain_count(self, op) -> Optional[int]:
+        """COUNT(*) over a probe chain keyed ONLY off the seed scan fuses
+        into one kernel: count = sum_rows prod_hops |matches(key(row))|.
+
+        Shape requirements (bail -> None, generic path runs):
+          - left-deep PBindJoin/PHashJoin tree of plain default-graph scans,
+          - seed pattern: (?s  P  ?o) with P constant, ?s != ?o,
+          - every hop: (?v  P_h  ?free) or (?free  P_h  ?v) with P_h
+            constant, ?v in {seed ?s, seed ?o}, and ?free occurring nowhere
+            else (no cross-hop constraints).
+        """
+        from ..storage.dataset import POS, PSO
+        from .tensor_utils import pack2
+        scans: List = []
+
+        def flatten(x) -> bool:
+            if isinstance(x, (PTableScan, PIndexScan)):
+                scans.append(x)
+                return x.graph is None
+            if isinstance(x, (PBindJoin, PHashJoin)):
+                return flatten(x.left) and flatten(x.right)
+            return False
+
+        if not flatten(op) or len(scans) < 2:
+            return None
+        if self.ctx.view.default_graphs != [DEFAULT_GRAPH] \
+                or self.ctx.view.named_graphs not in (None, []):
+            pass  # merged views still resolve through default_index below
+        base = scans[0].pattern
+        if not (isinstance(base.p, Constant) and isinstance(base.s, Variable)
+                and isinstance(base.o, Variable)
+                and base.s.name != base.o.name):
+            return None
+        var_counts: Dict[str, int] = {}
+        for sc in scans:
+            for v in sc.pattern.variables():
+                var_counts[v] = var_counts.get(v, 0) + 1
+        hops = []
+        for sc in scans[1:]:
+            pat = sc.pattern
+            if not isinstance(pat.p, Constant):
+                return None
+            s_t, o_t = pat.s, pat.o
+            if not (isinstance(s_t, Variable) and isinstance(o_t, Variable)):
+                return None
+            seed_vars = (base.s.name, base.o.name)
+            if s_t.name in seed_vars and var_counts.get(o_t.name, 0) == 1:
+                probe_pos, probe_var = 0, s_t.name
+            elif o_t.name in seed_vars and var_counts.get(s_t.name, 0) == 1:
+                probe_pos, probe_var = 2, o_t.name
+            else:
+                return None
+            hops.append((pat.p.id, probe_pos,
+                         0 if probe_var == base.s.name else 1))
+        idx = self.ctx.default_index()
+        if idx.n == 0:
+            return 0
+        seed_key12, seed_z, lo, hi = _pso_region(idx, base.p.id)
+        if hi <= lo:
+            return 0
+        hop_regions = []
+        for (pid, probe_pos, src) in hops:
+            code = PSO if probe_pos == 0 else POS
+            key12, _z = idx.orders[code]
+            import torch as _t
+            klo = pack2(_t.tensor([pid], dtype=_t.int32),
+                        _t.tensor([0], dtype=_t.int32)).to(idx.device)
+            khi = pack2(_t.tensor([pid], dtype=_t.int32),
+                        _t.tensor([-1], dtype=_t.int32)).to(idx.device)
+            rlo = int(_t.searchsorted(key12, klo, side="left").item())
+            rhi = int(_t.searchsorted(key12, khi, side="right").item())
+            hop_regions.append((key12[rlo:rhi], pid, src))
+        from ..ops import native_for
+        native = native_for(seed_key12)
+        if native is not None:
+            return int(native.chain_count(
+                seed_key12.contiguous(), seed_z.contiguous(),
+                [r[0].contiguous() for r in hop_regions],
+                [r[1] for r in hop_regions],
+                [r[2] for r in hop_regions]))
+        # torch fallback (CPU oracle): vectorized per-hop count product
+        import torch as _t
+        b_comp = (seed_key12 & 0xFFFFFFFF).to(_t.int32)
+        z_comp = seed_z
+        prod = _t.ones(seed_key12.numel(), dtype=_t.int64, device=idx.device)
+        for (region, pid, src) in hop_regions:
+            comp = b_comp if src == 0 else z_comp
+            keys = pack2(_t.full_like(comp, pid), comp)
+            lo_t = _t.searchsorted(region, keys, side="left")
+            hi_t = _t.searchsorted(region, keys, side="right")
+            prod *= (hi_t - lo_t)
+        return int(prod.sum().item())
 
     # --------------------------------------------------------------- values --
     def _values_bindings(self, variables: List[str],
@@ -632,3 +732,17 @@ def anti_join(left: Bindings, right: Bindings) -> Bindings:
                 any_shared |= lv != UNBOUND
             keep &= ~(m & any_shared)
     return left.select(keep)
+
+
+def _pso_region(idx: GraphIndex, pid: int):
+    """(key12 slice, z slice, lo, hi) of one predicate's PSO region."""
+    from ..storage.dataset import PSO
+    from .tensor_utils import pack2
+    key12, z = idx.orders[PSO]
+    klo = pack2(torch.tensor([pid], dtype=torch.int32),
+                torch.tensor([0], dtype=torch.int32)).to(idx.device)
+    khi = pack2(torch.tensor([pid], dtype=torch.int32),
+                torch.tensor([-1], dtype=torch.int32)).to(idx.device)
+    lo = int(torch.searchsorted(key12, klo, side="left").item())
+    hi = int(torch.searchsorted(key12, khi, side="right").item())
+    return key12[lo:hi], z[lo:hi], lo, hi

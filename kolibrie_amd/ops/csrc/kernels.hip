@@ -496,6 +496,77 @@ inline hipStream_t cur_stream() {
 
 }  // namespace
 
+// --------------------------------------------- fused chain-count (K1+K2+K4)
+// COUNT(*) over a probe chain whose hop keys all come from the SEED scan:
+//   count = sum over seed rows of prod_h |matches_h(key_h(row))|
+// One kernel, no intermediate materialization.  Hop regions are
+// pre-narrowed to their predicate slice on the host.
+constexpr int kMaxHops = 4;
+
+struct ChainHops {
+  const int64_t* key12[kMaxHops];  // narrowed sorted regions
+  int64_t n[kMaxHops];
+  int64_t const_hi[kMaxHops];      // high 32 bits of the probe key
+  int32_t src[kMaxHops];           // 0: seed B component, 1: seed Z column
+  int k;
+};
+
+__global__ void chain_count_kernel(const int64_t* __restrict__ seed_key12,
+                                   const int32_t* __restrict__ seed_z,
+                                   int64_t m, ChainHops hops,
+                                   unsigned long long* __restrict__ total) {
+  unsigned long long acc = 0;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < m;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t b_comp = seed_key12[i] & 0xFFFFFFFFLL;
+    int64_t z_comp = static_cast<int64_t>(seed_z[i]) & 0xFFFFFFFFLL;
+    unsigned long long prod = 1;
+    for (int h = 0; h < hops.k && prod; ++h) {
+      int64_t key = (hops.const_hi[h] << 32)
+                    | (hops.src[h] == 0 ? b_comp : z_comp);
+      int64_t lo = lower_bound_i64(hops.key12[h], hops.n[h], key);
+      int64_t hi = upper_bound_i64(hops.key12[h], hops.n[h], key);
+      prod *= static_cast<unsigned long long>(hi - lo);
+    }
+    acc += prod;
+  }
+  // wave reduction then one device-scope atomic per wave (guide G12)
+  for (int off = 32; off > 0; off >>= 1)
+    acc += __shfl_down(acc, off);
+  if ((threadIdx.x & 63) == 0 && acc)
+    atomicAdd(total, acc);
+}
+
+int64_t chain_count(at::Tensor seed_key12, at::Tensor seed_z,
+                    std::vector<at::Tensor> hop_key12,
+                    std::vector<int64_t> hop_const_hi,
+                    std::vector<int64_t> hop_src) {
+  TORCH_CHECK(seed_key12.is_cuda() && seed_z.is_cuda());
+  TORCH_CHECK(hop_key12.size() <= static_cast<size_t>(kMaxHops));
+  TORCH_CHECK(hop_key12.size() == hop_const_hi.size()
+              && hop_key12.size() == hop_src.size());
+  int64_t m = seed_key12.numel();
+  ChainHops hops{};
+  hops.k = static_cast<int>(hop_key12.size());
+  for (size_t h = 0; h < hop_key12.size(); ++h) {
+    TORCH_CHECK(hop_key12[h].is_cuda() && hop_key12[h].dtype() == at::kLong);
+    hops.key12[h] = hop_key12[h].data_ptr<int64_t>();
+    hops.n[h] = hop_key12[h].numel();
+    hops.const_hi[h] = hop_const_hi[h];
+    hops.src[h] = static_cast<int32_t>(hop_src[h]);
+  }
+  auto total = at::zeros({1}, seed_key12.options());
+  if (m > 0) {
+    hipLaunchKernelGGL(chain_count_kernel, dim3(grid_for(m)), dim3(kBlock), 0,
+                       cur_stream(), seed_key12.data_ptr<int64_t>(),
+                       seed_z.data_ptr<int32_t>(), m, hops,
+                       reinterpret_cast<unsigned long long*>(
+                           total.data_ptr<int64_t>()));
+    HIP_OK(hipGetLastError());
+  }
+  return total.item<int64_t>();
+}
+
 // ------------------------------------------------------------ host launchers
 
 // shared emit phase for the K1 probes: per-probe-row emit for low fanout,
@@ -865,6 +936,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "K1 scan-probe, leading-component range -> (li, b, z)");
   m.def("hash_join", &hash_join,
         "K2 chained hash join over int32 key columns -> (li, ri)");
+  m.def("chain_count", &chain_count,
+        "fused COUNT(*) over a seed scan + probe-hop chain");
   m.def("probe_fused", &probe_fused,
         "K1 fused probe: inline key pack + merge-path + carry emit");
   m.def("probe_exact_counts", &probe_exact_counts,
