@@ -173,6 +173,17 @@ class ExecutionEngine:
 
         if not flatten(op) or len(scans) < 2:
             return None
+        # regions are store-version-stable: cache them on the plan node
+        cache = getattr(op, "_chain_cache", None)
+        if cache is not None and cache[0] == self.db.store.version:
+            seed_key12, seed_z, hop_regions, native = cache[1:]
+            if native is not None:
+                return int(native.chain_count(
+                    seed_key12, seed_z,
+                    [r[0] for r in hop_regions],
+                    [r[1] for r in hop_regions],
+                    [r[2] for r in hop_regions]))
+            return self._chain_count_torch(seed_key12, seed_z, hop_regions)
         if self.ctx.view.default_graphs != [DEFAULT_GRAPH] \
                 or self.ctx.view.named_graphs not in (None, []):
             pass  # merged views still resolve through default_index below
@@ -222,17 +233,27 @@ class ExecutionEngine:
             hop_regions.append((key12[rlo:rhi], pid, src))
         from ..ops import native_for
         native = native_for(seed_key12)
+        seed_key12 = seed_key12.contiguous()
+        seed_z = seed_z.contiguous()
+        hop_regions = [(r[0].contiguous(), r[1], r[2]) for r in hop_regions]
+        op._chain_cache = (self.db.store.version, seed_key12, seed_z,
+                           hop_regions, native)
         if native is not None:
             return int(native.chain_count(
-                seed_key12.contiguous(), seed_z.contiguous(),
-                [r[0].contiguous() for r in hop_regions],
+                seed_key12, seed_z,
+                [r[0] for r in hop_regions],
                 [r[1] for r in hop_regions],
                 [r[2] for r in hop_regions]))
+        return self._chain_count_torch(seed_key12, seed_z, hop_regions)
+
+    def _chain_count_torch(self, seed_key12, seed_z, hop_regions) -> int:
         # torch fallback (CPU oracle): vectorized per-hop count product
+        from .tensor_utils import pack2
         import torch as _t
         b_comp = (seed_key12 & 0xFFFFFFFF).to(_t.int32)
         z_comp = seed_z
-        prod = _t.ones(seed_key12.numel(), dtype=_t.int64, device=idx.device)
+        prod = _t.ones(seed_key12.numel(), dtype=_t.int64,
+                       device=seed_key12.device)
         for (region, pid, src) in hop_regions:
             comp = b_comp if src == 0 else z_comp
             keys = pack2(_t.full_like(comp, pid), comp)
