@@ -63,7 +63,11 @@ def main():
     log(rank, f"[bench] generating shard: ~{total//max(1,world):,} of "
               f"{total:,} triples on {device} (world={world})")
     t0 = time.time()
-    s, p, o = generate_partition(ds, rank, world, args.seed, device)
+    # broadcast-table layout: the tiny department relation is replicated so
+    # the 3-way join needs NO per-query shuffle (the planner's choice for
+    # small build sides); employee triples stay subject-hash-partitioned
+    s, p, o = generate_partition(ds, rank, world, args.seed, device,
+                                 replicate_dept=(world > 1))
     ddb.load_shard_columns(s, p, o)
     n_local = ddb.db.triple_count()
     log(rank, f"[bench] shard loaded: {n_local:,} triples in {time.time()-t0:.1f}s")
@@ -82,7 +86,8 @@ def main():
 
     def run_query() -> int:
         if world > 1:
-            return ddb.count_query_with_shuffle(local_star, "d", probe_q)
+            rows = ddb.db.query(FLAGSHIP_QUERY)
+            return D.allreduce_sum_scalar(int(rows[0][0]), device)
         rows = ddb.db.query(FLAGSHIP_QUERY)
         return int(rows[0][0])
 
@@ -141,9 +146,9 @@ def main():
                 "total_triples": total,
                 "result_count": c,
                 "p50_ms": p50,
-                "parallelism": f"subject-hash-partition dp{world}, "
-                               "all-to-all join shuffle" if world > 1
-                               else "single-GPU",
+                "parallelism": (f"subject-hash-partition dp{world}, "
+                                "replicated small relation, COUNT all-reduce"
+                                if world > 1 else "single-GPU"),
             },
         }
         print(json.dumps(result), flush=True)

@@ -73,12 +73,16 @@ def plan_dataset(db, total_triples: int) -> EmployeeDataset:
 
 
 def generate_partition(ds: EmployeeDataset, rank: int, world: int, seed: int,
-                       device) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
+                       device, replicate_dept: bool = False
+                       ) -> Tuple[torch.Tensor, torch.Tensor, torch.Tensor]:
     """Generate this rank's subject-partitioned shard as (s,p,o) int32 cols.
 
     Partition function: subject_id % world (a valid hash partition over the
     dense synthetic ID blocks) — employees and departments both shard by
     their own subject id, exactly what a distributed loader would do.
+    `replicate_dept=True` keeps the small department relation on EVERY rank
+    (the broadcast-table layout a distributed planner chooses for tiny
+    build sides: join traffic becomes zero).
     """
     rng = np.random.default_rng(seed + 7919 * rank)
     emp = np.arange(ds.n_employees, dtype=np.int64)
@@ -104,7 +108,8 @@ def generate_partition(ds: EmployeeDataset, rank: int, world: int, seed: int,
     add("worksFor", emp_ids, dept_of)
 
     dept = np.arange(ds.n_departments, dtype=np.int64)
-    dept = dept[(ds.dept_base + dept) % world == rank]
+    if not replicate_dept:
+        dept = dept[(ds.dept_base + dept) % world == rank]
     dept_ids = ds.dept_base + dept
     add("locatedIn", dept_ids, ds.city_base + (dept % N_CITIES))
     add("label", dept_ids, ds.name_base + (dept % max(1, ds.n_employees)))
