@@ -348,7 +348,16 @@ class ExecutionEngine:
                     return Bindings({}, cnt, dev)
             s, p, o = scan_unit(idx, consts, sort_hint=sort_hint,
                                 need=scan_need)
-            n_sc = next((c.numel() for c in (s, p, o) if c is not None), 0)
+            n_sc = next((c.numel() for c in (s, p, o) if c is not None), -1)
+            if n_sc < 0:
+                # no column materialized (all pattern vars pruned): the row
+                # count still matters for multiplicity / graph binding
+                from .scan import scan_unit_count
+                n_sc = scan_unit_count(idx, consts)
+                if n_sc is None:
+                    s, p, o = scan_unit(idx, consts, sort_hint=sort_hint)
+                    n_sc = next((c.numel() for c in (s, p, o)
+                                 if c is not None), 0)
             exec_stats.bump("QUADS_EXAMINED", n_sc)
             cand = self._build_candidate(s, p, o, var_pos, qt_pos, None,
                                          needed, n_sc)

@@ -1,0 +1,110 @@
+"""Dataset / update regressions (mirrors sparql_dataset_regressions_test.rs,
+named_graph_test.rs, hybrid_test.rs shapes) + exec-stats counters."""
+import pytest
+
+from kolibrie_amd import SparqlDatabase
+
+EX = "http://example.org/"
+
+
+def test_nquads_roundtrip_with_graphs():
+    db = SparqlDatabase()
+    db.query(f"""INSERT DATA {{
+        <{EX}a> <{EX}p> "v0" .
+        GRAPH <{EX}g1> {{ <{EX}b> <{EX}p> "v1" }}
+        GRAPH <{EX}g2> {{ <{EX}c> <{EX}p> "v2" }}
+    }}""")
+    text = db.generate_nquads()
+    db2 = SparqlDatabase()
+    db2.parse_nquads(text)
+    assert db2.generate_nquads() == text
+    rows = db2.query(f"SELECT ?g WHERE {{ GRAPH ?g {{ ?s <{EX}p> ?o }} }}")
+    assert sorted(r[0] for r in rows) == [f"{EX}g1", f"{EX}g2"]
+
+
+def test_illegal_update_positions():
+    db = SparqlDatabase()
+    with pytest.raises(ValueError):
+        db.query(f'INSERT DATA {{ "literal" <{EX}p> "v" }}')
+    with pytest.raises(ValueError):
+        db.query(f'INSERT DATA {{ <{EX}s> "litpred" "v" }}')
+    with pytest.raises(ValueError):
+        db.query(f'INSERT DATA {{ ?var <{EX}p> "v" }}')
+
+
+def test_create_existing_graph_errors_unless_silent():
+    db = SparqlDatabase()
+    db.query(f"CREATE GRAPH <{EX}g>")
+    with pytest.raises(ValueError):
+        db.query(f"CREATE GRAPH <{EX}g>")
+    db.query(f"CREATE SILENT GRAPH <{EX}g>")   # no error
+    db.query(f"DROP GRAPH <{EX}g>")
+    with pytest.raises(ValueError):
+        db.query(f"DROP GRAPH <{EX}g>")
+
+
+def test_union_multiplicity_preserved():
+    db = SparqlDatabase()
+    db.add_triple(f"<{EX}a>", f"<{EX}p>", '"v"')
+    rows = db.query(f"""SELECT ?s WHERE {{
+        {{ ?s <{EX}p> "v" }} UNION {{ ?s <{EX}p> "v" }} UNION {{ ?s <{EX}p> "v" }}
+    }}""")
+    assert len(rows) == 3
+
+
+def test_values_multiplicity_with_undef():
+    db = SparqlDatabase()
+    db.add_triple(f"<{EX}a>", f"<{EX}p>", '"1"')
+    rows = db.query(f"""SELECT ?x ?y WHERE {{
+        VALUES (?x ?y) {{ (<{EX}a> "tag1") (<{EX}a> UNDEF) }}
+        ?x <{EX}p> ?o .
+    }}""")
+    assert len(rows) == 2
+
+
+def test_exec_stats_counters():
+    from kolibrie_amd.engine import exec_stats
+    db = SparqlDatabase()
+    for i in range(10):
+        db.add_triple(f"<{EX}e{i}>", f"<{EX}p>", f'"{i}"')
+    exec_stats.reset()
+    db.query(f"SELECT ?s WHERE {{ ?s <{EX}p> ?o }}")
+    snap = exec_stats.snapshot()
+    assert snap["SCAN_PROBES"] >= 1
+    assert snap["ROWS_EMITTED"] >= 10
+    exec_stats.reset()
+    assert exec_stats.snapshot()["ROWS_EMITTED"] == 0
+
+
+def test_hybrid_recursive_rule_rejected():
+    # hybrid evaluation requires monotone rules; recursion through negation
+    # is rejected (ref hybrid_test.rs recursion rejection)
+    from kolibrie_amd.reasoning.hybrid import validate_monotone
+    from kolibrie_amd.reasoning.rule import Rule
+    from kolibrie_amd.storage.terms import Constant, TriplePattern, Variable
+    bad = Rule(
+        premise=[TriplePattern(Variable("x"), Constant(1), Variable("y"))],
+        negative_premise=[TriplePattern(Variable("x"), Constant(2), Variable("y"))],
+        conclusion=[TriplePattern(Variable("x"), Constant(2), Variable("y"))],
+    )
+    with pytest.raises(ValueError):
+        validate_monotone([bad])
+
+
+def test_typed_hybrid_status_annotations():
+    from kolibrie_amd.reasoning.hybrid import (
+        HybridConfig, LineageStore, evaluate_hybrid, encode_results_rdf_star,
+        HybridProbabilityResult,
+    )
+    db = SparqlDatabase()
+    s = db.dictionary.encode("s")
+    p = db.dictionary.encode("p")
+    o = db.dictionary.encode("o")
+    st = LineageStore()
+    res = evaluate_hybrid(st, st.leaf(1), {1: 0.93}, HybridConfig())
+    encode_results_rdf_star({(s, p, o): res}, db)
+    rows = db.query("""
+        SELECT ?st WHERE {
+            ?t <http://kolibrie.amd/hybrid#status> ?st . FILTER(isTRIPLE(?t))
+        }""")
+    assert rows and rows[0][0] in ("Decided", "DecidedExact")
