@@ -20,6 +20,9 @@
 #include <hip/hip_runtime.h>
 
 #include <cstdint>
+#include <unordered_map>
+
+namespace py = pybind11;
 #include <vector>
 
 #define HIP_OK(expr)                                                          \
@@ -928,7 +931,146 @@ at::Tensor filter_bytecode(at::Tensor ops, at::Tensor args, at::Tensor consts,
   return out;
 }
 
+// ------------------------------------------------- host bulk N-Triples parse
+// Native replacement for the reference's crossbeam parse pipeline
+// (sparql_database.rs:636-795): one pass tokenizes lines, interns terms in
+// a local dictionary and emits local-id triples; Python merges the (much
+// smaller) unique-string table into the main dictionary and remaps the id
+// columns vectorized.  RDF-star lines (`<<`) fall back to the Python path.
+static inline std::string unescape_nt(const char* s, size_t len) {
+  std::string out;
+  out.reserve(len);
+  for (size_t i = 0; i < len; ++i) {
+    char c = s[i];
+    if (c == '\\' && i + 1 < len) {
+      char n = s[i + 1];
+      switch (n) {
+        case 'n': out.push_back('\n'); ++i; continue;
+        case 't': out.push_back('\t'); ++i; continue;
+        case 'r': out.push_back('\r'); ++i; continue;
+        case '"': out.push_back('"'); ++i; continue;
+        case '\\': out.push_back('\\'); ++i; continue;
+        case 'u':
+          if (i + 5 < len) {
+            unsigned cp = std::stoul(std::string(s + i + 2, 4), nullptr, 16);
+            // minimal UTF-8 encode
+            if (cp < 0x80) out.push_back(static_cast<char>(cp));
+            else if (cp < 0x800) {
+              out.push_back(static_cast<char>(0xC0 | (cp >> 6)));
+              out.push_back(static_cast<char>(0x80 | (cp & 0x3F)));
+            } else {
+              out.push_back(static_cast<char>(0xE0 | (cp >> 12)));
+              out.push_back(static_cast<char>(0x80 | ((cp >> 6) & 0x3F)));
+              out.push_back(static_cast<char>(0x80 | (cp & 0x3F)));
+            }
+            i += 5;
+            continue;
+          }
+          break;
+        default: break;
+      }
+    }
+    out.push_back(c);
+  }
+  return out;
+}
+
+py::tuple parse_ntriples_host(const std::string& text) {
+  std::unordered_map<std::string, int64_t> interned;
+  std::vector<std::string> strings;
+  std::vector<int64_t> ids;           // n*3 local ids
+  std::vector<int64_t> fallback;      // line numbers for the Python path
+  ids.reserve(1 << 16);
+
+  auto intern = [&](std::string&& s) -> int64_t {
+    auto it = interned.find(s);
+    if (it != interned.end()) return it->second;
+    int64_t id = static_cast<int64_t>(strings.size());
+    interned.emplace(s, id);
+    strings.push_back(std::move(s));
+    return id;
+  };
+
+  size_t pos = 0, line_no = 0;
+  const size_t n = text.size();
+  while (pos < n) {
+    size_t eol = text.find('\n', pos);
+    if (eol == std::string::npos) eol = n;
+    const char* line = text.data() + pos;
+    size_t len = eol - pos;
+    size_t save_pos = pos;
+    pos = eol + 1;
+    ++line_no;
+    // trim
+    size_t b = 0, e = len;
+    while (b < e && isspace(static_cast<unsigned char>(line[b]))) ++b;
+    while (e > b && isspace(static_cast<unsigned char>(line[e - 1]))) --e;
+    if (b >= e || line[b] == '#') continue;
+    // RDF-star or exotic: Python fallback
+    bool star = false;
+    for (size_t i = b; i + 1 < e; ++i)
+      if (line[i] == '<' && line[i + 1] == '<') { star = true; break; }
+    if (star) { fallback.push_back(static_cast<int64_t>(line_no - 1)); continue; }
+    int64_t term_ids[3];
+    int nt = 0;
+    size_t i = b;
+    bool ok = true;
+    while (i < e && nt < 3) {
+      while (i < e && isspace(static_cast<unsigned char>(line[i]))) ++i;
+      if (i >= e) break;
+      char c = line[i];
+      if (c == '<') {
+        size_t j = i + 1;
+        while (j < e && line[j] != '>') ++j;
+        if (j >= e) { ok = false; break; }
+        term_ids[nt++] = intern(unescape_nt(line + i + 1, j - i - 1));
+        i = j + 1;
+      } else if (c == '"') {
+        size_t j = i + 1;
+        while (j < e && !(line[j] == '"' && line[j - 1] != '\\')) ++j;
+        if (j >= e) { ok = false; break; }
+        term_ids[nt++] = intern(unescape_nt(line + i + 1, j - i - 1));
+        i = j + 1;
+        // skip @lang / ^^<datatype>
+        while (i < e && !isspace(static_cast<unsigned char>(line[i]))
+               && line[i] != '.') {
+          if (line[i] == '<') { while (i < e && line[i] != '>') ++i; }
+          ++i;
+        }
+      } else if (c == '_' ) {
+        size_t j = i;
+        while (j < e && !isspace(static_cast<unsigned char>(line[j]))) ++j;
+        term_ids[nt++] = intern(std::string(line + i, j - i));
+        i = j;
+      } else if (c == '.') {
+        break;
+      } else {
+        ok = false;
+        break;
+      }
+    }
+    if (!ok || nt != 3) {
+      fallback.push_back(static_cast<int64_t>(line_no - 1));
+      continue;
+    }
+    ids.push_back(term_ids[0]);
+    ids.push_back(term_ids[1]);
+    ids.push_back(term_ids[2]);
+    (void)save_pos;
+  }
+  auto t = at::from_blob(ids.data(),
+                         {static_cast<int64_t>(ids.size() / 3), 3},
+                         at::kLong).clone();
+  py::list pystrings;
+  for (auto& s : strings) pystrings.append(py::bytes(s));
+  py::list pyfallback;
+  for (auto f : fallback) pyfallback.append(f);
+  return py::make_tuple(t, pystrings, pyfallback);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("parse_ntriples_host", &parse_ntriples_host,
+        "bulk N-Triples parse -> (local-id triples, strings, fallback lines)");
   m.doc() = "kolibrie_amd native CDNA4 kernels (gfx950)";
   m.def("probe_exact", &probe_exact,
         "K1 scan-probe, packed (a,b) exact keys -> (li, b, z)");

@@ -84,45 +84,44 @@ def generate_partition(ds: EmployeeDataset, rank: int, world: int, seed: int,
     (the broadcast-table layout a distributed planner chooses for tiny
     build sides: join traffic becomes zero).
     """
-    rng = np.random.default_rng(seed + 7919 * rank)
-    emp = np.arange(ds.n_employees, dtype=np.int64)
-    emp = emp[(ds.emp_base + emp) % world == rank]
-    ne = emp.size
+    dev = torch.device(device)
+    gen = torch.Generator(device=dev)
+    gen.manual_seed(seed + 7919 * rank)
+    emp_all = torch.arange(ds.n_employees, dtype=torch.int64, device=dev)
+    emp = emp_all[(ds.emp_base + emp_all) % world == rank]
+    ne = emp.numel()
     p = ds.pred_ids
 
     s_parts, p_parts, o_parts = [], [], []
 
-    def add(pred_key: str, subjects: np.ndarray, objects: np.ndarray):
-        s_parts.append(subjects.astype(np.int64))
-        p_parts.append(np.full(subjects.size, p[pred_key], dtype=np.int64))
-        o_parts.append(objects.astype(np.int64))
+    def add(pred_key: str, subjects: torch.Tensor, objects: torch.Tensor):
+        s_parts.append(subjects.to(torch.int32))
+        p_parts.append(torch.full((subjects.numel(),), p[pred_key],
+                                  dtype=torch.int32, device=dev))
+        o_parts.append(objects.to(torch.int32))
+
+    def rnd(hi: int, n: int) -> torch.Tensor:
+        return torch.randint(0, hi, (n,), generator=gen, dtype=torch.int64,
+                             device=dev)
 
     emp_ids = ds.emp_base + emp
     add("name", emp_ids, ds.name_base + emp)
     add("homepage", emp_ids, ds.name_base + emp)  # homepage shares name id pool
-    add("salary", emp_ids, ds.salary_base + rng.integers(0, ds.n_salary_values, ne))
-    add("position", emp_ids, 1 + rng.integers(0, N_POSITIONS, ne))
+    add("salary", emp_ids, ds.salary_base + rnd(ds.n_salary_values, ne))
+    add("position", emp_ids, 1 + rnd(N_POSITIONS, ne))
     add("email", emp_ids, ds.name_base + emp)
-    add("age", emp_ids, ds.salary_base + rng.integers(0, 50, ne))
+    add("age", emp_ids, ds.salary_base + rnd(50, ne))
     dept_of = ds.dept_base + (emp % ds.n_departments)
     add("worksFor", emp_ids, dept_of)
 
-    dept = np.arange(ds.n_departments, dtype=np.int64)
+    dept = torch.arange(ds.n_departments, dtype=torch.int64, device=dev)
     if not replicate_dept:
         dept = dept[(ds.dept_base + dept) % world == rank]
     dept_ids = ds.dept_base + dept
     add("locatedIn", dept_ids, ds.city_base + (dept % N_CITIES))
     add("label", dept_ids, ds.name_base + (dept % max(1, ds.n_employees)))
 
-    s = np.concatenate(s_parts)
-    pp = np.concatenate(p_parts)
-    o = np.concatenate(o_parts)
-    to = torch.device(device)
-    return (
-        torch.from_numpy(s.astype(np.int32)).to(to),
-        torch.from_numpy(pp.astype(np.int32)).to(to),
-        torch.from_numpy(o.astype(np.int32)).to(to),
-    )
+    return torch.cat(s_parts), torch.cat(p_parts), torch.cat(o_parts)
 
 
 FLAGSHIP_QUERY = f"""

@@ -90,8 +90,34 @@ def _read_term(ts: _TokenStream) -> str:
 
 # ------------------------------------------------------------------ N-Triples
 def parse_ntriples_into(db, text: str):
-    """Line-oriented N-Triples-star (ref sparql_database.rs:1345-1463)."""
-    for line in text.split("\n"):
+    """Line-oriented N-Triples-star (ref sparql_database.rs:1345-1463).
+
+    Bulk path: the native C++ tokenizer (ops._native.parse_ntriples_host,
+    the MI355X replacement for the reference's crossbeam parse pipeline)
+    interns terms locally; the unique-string table merges into the main
+    dictionary and the id columns remap vectorized.  RDF-star lines fall
+    back to the Python tokenizer."""
+    from ..ops import _native
+    if _native is not None and len(text) > 4096:
+        import numpy as np
+        local_ids, strings, fallback = _native.parse_ntriples_host(text)
+        if local_ids.numel():
+            remap = np.empty(len(strings), dtype=np.uint32)
+            enc = db.dictionary.encode
+            for i, raw in enumerate(strings):
+                remap[i] = enc(raw.decode("utf-8", "replace"))
+            arr = local_ids.numpy()
+            mapped = remap[arr]
+            db.store.insert_bulk(0, mapped[:, 0], mapped[:, 1], mapped[:, 2])
+        if fallback:
+            lines = text.split("\n")
+            _parse_ntriples_lines(db, (lines[i] for i in fallback))
+        return
+    _parse_ntriples_lines(db, text.split("\n"))
+
+
+def _parse_ntriples_lines(db, lines):
+    for line in lines:
         line = line.strip()
         if not line or line.startswith("#"):
             continue

@@ -143,3 +143,28 @@ def test_union_reencoding():
     db2.add_triple("<http://e/b>", "<http://e/p>", '"2"')
     u = db1.union(db2)
     assert u.triple_count() == 2
+
+
+def test_native_ntriples_bulk_parse_matches_python():
+    """The C++ bulk N-Triples parser must agree with the Python tokenizer
+    (differential)."""
+    from kolibrie_amd.ops import _native
+    if _native is None:
+        import pytest
+        pytest.skip("native extension not built")
+    lines = []
+    for i in range(500):
+        lines.append(f'<http://e/s{i}> <http://e/p{i % 7}> "value {i} with \\"quote\\"" .')
+        lines.append(f'<http://e/s{i}> <http://e/q> _:b{i} .')
+    lines.append('<< <http://e/a> <http://e/b> <http://e/c> >> <http://e/cert> "0.9" .')
+    lines.append('<http://e/lang> <http://e/label> "hello"@en .')
+    lines.append('<http://e/typed> <http://e/num> "42"^^<http://www.w3.org/2001/XMLSchema#int> .')
+    text = "\n".join(lines)
+    db_native = SparqlDatabase()
+    db_native.parse_ntriples(text)           # bulk path (text > 4096 bytes)
+    db_py = SparqlDatabase()
+    from kolibrie_amd.parsing.rdf_formats import _parse_ntriples_lines
+    _parse_ntriples_lines(db_py, text.split("\n"))
+    assert sorted(db_native.triples_as_strings()) == \
+        sorted(db_py.triples_as_strings())
+    assert db_native.triple_count() == db_py.triple_count()
