@@ -514,21 +514,50 @@ struct ChainHops {
   int k;
 };
 
+// per-(tile, src0-hop) windows: the seed region is subject-sorted, so the
+// B component is monotone — each 256-seed tile's src-0 hop keys fall in a
+// narrow window of that hop's region (merge-path; computed in parallel).
+__global__ void chain_tile_bounds(const int64_t* __restrict__ seed_key12,
+                                  int64_t m, int64_t n_tiles, ChainHops hops,
+                                  int64_t* __restrict__ win) {  // [n_tiles][k][2]
+  for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; t < n_tiles;
+       t += (int64_t)gridDim.x * blockDim.x) {
+    int64_t first_b = seed_key12[t * kTile] & 0xFFFFFFFFLL;
+    int64_t last_b = seed_key12[min((t + 1) * kTile - 1, m - 1)] & 0xFFFFFFFFLL;
+    for (int h = 0; h < hops.k; ++h) {
+      int64_t lo = 0, hi = hops.n[h];
+      if (hops.src[h] == 0) {
+        lo = lower_bound_i64(hops.key12[h], hops.n[h],
+                             (hops.const_hi[h] << 32) | first_b);
+        hi = upper_bound_i64(hops.key12[h], hops.n[h],
+                             (hops.const_hi[h] << 32) | last_b);
+      }
+      win[(t * hops.k + h) * 2] = lo;
+      win[(t * hops.k + h) * 2 + 1] = hi;
+    }
+  }
+}
+
 __global__ void chain_count_kernel(const int64_t* __restrict__ seed_key12,
                                    const int32_t* __restrict__ seed_z,
                                    int64_t m, ChainHops hops,
+                                   const int64_t* __restrict__ win,
                                    unsigned long long* __restrict__ total) {
   unsigned long long acc = 0;
   for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < m;
        i += (int64_t)gridDim.x * blockDim.x) {
+    int64_t t = i / kTile;
     int64_t b_comp = seed_key12[i] & 0xFFFFFFFFLL;
     int64_t z_comp = static_cast<int64_t>(seed_z[i]) & 0xFFFFFFFFLL;
     unsigned long long prod = 1;
     for (int h = 0; h < hops.k && prod; ++h) {
       int64_t key = (hops.const_hi[h] << 32)
                     | (hops.src[h] == 0 ? b_comp : z_comp);
-      int64_t lo = lower_bound_i64(hops.key12[h], hops.n[h], key);
-      int64_t hi = upper_bound_i64(hops.key12[h], hops.n[h], key);
+      int64_t wlo = win[(t * hops.k + h) * 2];
+      int64_t wspan = win[(t * hops.k + h) * 2 + 1] - wlo;
+      const int64_t* base = hops.key12[h] + wlo;
+      int64_t lo = lower_bound_i64(base, wspan, key);
+      int64_t hi = upper_bound_i64(base, wspan, key);
       prod *= static_cast<unsigned long long>(hi - lo);
     }
     acc += prod;
@@ -560,9 +589,18 @@ int64_t chain_count(at::Tensor seed_key12, at::Tensor seed_z,
   }
   auto total = at::zeros({1}, seed_key12.options());
   if (m > 0) {
+    auto stream = cur_stream();
+    int64_t n_tiles = (m + kTile - 1) / kTile;
+    auto win = at::empty({n_tiles * hops.k * 2}, seed_key12.options());
+    hipLaunchKernelGGL(chain_tile_bounds, dim3(grid_for(n_tiles)),
+                       dim3(kBlock), 0, stream,
+                       seed_key12.data_ptr<int64_t>(), m, n_tiles, hops,
+                       win.data_ptr<int64_t>());
+    HIP_OK(hipGetLastError());
     hipLaunchKernelGGL(chain_count_kernel, dim3(grid_for(m)), dim3(kBlock), 0,
-                       cur_stream(), seed_key12.data_ptr<int64_t>(),
+                       stream, seed_key12.data_ptr<int64_t>(),
                        seed_z.data_ptr<int32_t>(), m, hops,
+                       win.data_ptr<int64_t>(),
                        reinterpret_cast<unsigned long long*>(
                            total.data_ptr<int64_t>()));
     HIP_OK(hipGetLastError());
