@@ -173,3 +173,10 @@ def execute_sparql_query(sparql: str, db) -> List[List[str]]:
     if cq.select is None:
         return []
     return execute_select(cq.select, db, prefixes)
+
+
+# Reference public entry-point names (execute_query.rs:52; the "rayon
+# parallel volcano" path IS the default engine here — wavefront parallelism
+# replaces the Rayon pool).
+execute_query_rayon_parallel2_volcano = execute_query
+execute_query_volcano = execute_query
