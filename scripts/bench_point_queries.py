@@ -36,6 +36,18 @@ def main():
             SELECT (COUNT(*) AS ?c) WHERE {{
                 ?e ds:worksFor <#d> . ?e ds:annual_salary ?sal }}"""
             .replace("<#d>", f"<http://synthetic/{dept}>"),
+        "S5 star: 5 attributes of 1 employee":
+            f"""PREFIX ds: <{DS}> PREFIX foaf: <http://xmlns.com/foaf/0.1/>
+            SELECT ?n ?sal ?pos ?hp ?age WHERE {{
+                <#e> foaf:name ?n ; ds:annual_salary ?sal ;
+                     ds:position ?pos ; foaf:workplaceHomepage ?hp ;
+                     ds:age ?age }}""".replace("<#e>", f"<http://synthetic/{emp}>"),
+        "C2 complex: dept chain + filter":
+            f"""PREFIX ds: <{DS}>
+            SELECT (COUNT(*) AS ?c) WHERE {{
+                ?e ds:worksFor <#d> . ?e ds:annual_salary ?sal .
+                ?e ds:position ?pos . FILTER(?sal > 0) }}"""
+            .replace("<#d>", f"<http://synthetic/{dept}>"),
     }
     # synthetic entity ids are not interned as IRIs; bind them directly
     db.dictionary.str_to_id[f"http://synthetic/{emp}"] = emp
