@@ -550,7 +550,10 @@ __global__ void chain_count_kernel(const int64_t* __restrict__ seed_key12,
     int64_t b_comp = seed_key12[i] & 0xFFFFFFFFLL;
     int64_t z_comp = static_cast<int64_t>(seed_z[i]) & 0xFFFFFFFFLL;
     unsigned long long prod = 1;
-    for (int h = 0; h < hops.k && prod; ++h) {
+    // no early exit: the hop searches are independent dependent-load
+    // chains — letting them all issue gives the scheduler ILP to hide
+    // L2 latency
+    for (int h = 0; h < hops.k; ++h) {
       int64_t key = (hops.const_hi[h] << 32)
                     | (hops.src[h] == 0 ? b_comp : z_comp);
       int64_t wlo = win[(t * hops.k + h) * 2];
