@@ -182,7 +182,8 @@ class ExecutionEngine:
                     seed_key12, seed_z,
                     [r[0] for r in hop_regions],
                     [r[1] for r in hop_regions],
-                    [r[2] for r in hop_regions]))
+                    [r[2] for r in hop_regions],
+                    [r[3] for r in hop_regions]))
             return self._chain_count_torch(seed_key12, seed_z, hop_regions)
         if self.ctx.view.default_graphs != [DEFAULT_GRAPH] \
                 or self.ctx.view.named_graphs not in (None, []):
@@ -235,7 +236,10 @@ class ExecutionEngine:
         native = native_for(seed_key12)
         seed_key12 = seed_key12.contiguous()
         seed_z = seed_z.contiguous()
-        hop_regions = [(r[0].contiguous(), r[1], r[2]) for r in hop_regions]
+        hop_regions = [
+            (r[0].contiguous(), r[1], r[2],
+             self._chain_hop_table(native, r[0], seed_key12.numel()))
+            for r in hop_regions]
         op._chain_cache = (self.db.store.version, seed_key12, seed_z,
                            hop_regions, native)
         if native is not None:
@@ -243,8 +247,26 @@ class ExecutionEngine:
                 seed_key12, seed_z,
                 [r[0] for r in hop_regions],
                 [r[1] for r in hop_regions],
-                [r[2] for r in hop_regions]))
+                [r[2] for r in hop_regions],
+                [r[3] for r in hop_regions]))
         return self._chain_count_torch(seed_key12, seed_z, hop_regions)
+
+    # hop regions much smaller than the seed count pay log2(n) L2 lines per
+    # seed in the fused kernel; collapse them ONCE (per store version, the
+    # cache above) to an open-addressing (value -> count) table so the hop
+    # is a single 8-byte load.  Only worthwhile when the table stays
+    # L2-resident and the seeds dominate the build cost.
+    _HOP_TABLE_MAX_ROWS = 1_000_000
+
+    def _chain_hop_table(self, native, region, n_seeds):
+        import torch as _t
+        n = region.numel()
+        if (native is None or not region.is_cuda or n == 0
+                or n > self._HOP_TABLE_MAX_ROWS or n * 4 > n_seeds):
+            return _t.empty(0, dtype=_t.int64, device=region.device)
+        vals, counts = _t.unique_consecutive(region & 0xFFFFFFFF,
+                                             return_counts=True)
+        return native.build_count_table((vals << 32) | counts)
 
     def _chain_count_torch(self, seed_key12, seed_z, hop_regions) -> int:
         # torch fallback (CPU oracle): vectorized per-hop count product
@@ -254,7 +276,7 @@ class ExecutionEngine:
         z_comp = seed_z
         prod = _t.ones(seed_key12.numel(), dtype=_t.int64,
                        device=seed_key12.device)
-        for (region, pid, src) in hop_regions:
+        for (region, pid, src, *_tbl) in hop_regions:
             comp = b_comp if src == 0 else z_comp
             keys = pack2(_t.full_like(comp, pid), comp)
             lo_t = _t.searchsorted(region, keys, side="left")

@@ -511,8 +511,55 @@ struct ChainHops {
   int64_t n[kMaxHops];
   int64_t const_hi[kMaxHops];      // high 32 bits of the probe key
   int32_t src[kMaxHops];           // 0: seed B component, 1: seed Z column
+  const unsigned long long* table[kMaxHops];  // optional count table
+  int64_t tmask[kMaxHops];
   int k;
 };
+
+// ---- per-hop COUNT tables -------------------------------------------------
+// A small hop region (e.g. the 142k-row locatedIn slice under 14.2M seeds)
+// makes every seed pay a ~17-level binary search = ~17 distinct L2 lines.
+// Instead the host collapses the region to (value, count) pairs once per
+// store version and an open-addressing table turns the hop into ONE 8-byte
+// L2 load (entry packed (val<<32)|count, empty slot = all-ones).
+constexpr unsigned long long kTblEmpty = ~0ull;
+
+__device__ __forceinline__ uint32_t h32(uint32_t x) {
+  x *= 2654435761u;
+  x ^= x >> 16;
+  return x;
+}
+
+__global__ void count_table_insert(const int64_t* __restrict__ entries,
+                                   int64_t u,
+                                   unsigned long long* __restrict__ table,
+                                   uint32_t mask) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < u;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    unsigned long long e = static_cast<unsigned long long>(entries[i]);
+    uint32_t slot = h32(static_cast<uint32_t>(e >> 32)) & mask;
+    while (atomicCAS(&table[slot], kTblEmpty, e) != kTblEmpty)
+      slot = (slot + 1) & mask;
+  }
+}
+
+at::Tensor build_count_table(at::Tensor packed_entries) {
+  TORCH_CHECK(packed_entries.is_cuda()
+              && packed_entries.dtype() == at::kLong);
+  int64_t u = packed_entries.numel();
+  int64_t size = 64;
+  while (size < 2 * u) size <<= 1;
+  auto table = at::full({size}, -1, packed_entries.options());
+  if (u > 0) {
+    hipLaunchKernelGGL(count_table_insert, dim3(grid_for(u)), dim3(kBlock), 0,
+                       cur_stream(), packed_entries.data_ptr<int64_t>(), u,
+                       reinterpret_cast<unsigned long long*>(
+                           table.data_ptr<int64_t>()),
+                       static_cast<uint32_t>(size - 1));
+    HIP_OK(hipGetLastError());
+  }
+  return table;
+}
 
 // per-(tile, src0-hop) windows: the seed region is subject-sorted, so the
 // B component is monotone — each 256-seed tile's src-0 hop keys fall in a
@@ -526,7 +573,9 @@ __global__ void chain_tile_bounds(const int64_t* __restrict__ seed_key12,
     int64_t last_b = seed_key12[min((t + 1) * kTile - 1, m - 1)] & 0xFFFFFFFFLL;
     for (int h = 0; h < hops.k; ++h) {
       int64_t lo = 0, hi = hops.n[h];
-      if (hops.src[h] == 0) {
+      if (hops.table[h] != nullptr) {
+        lo = hi = 0;  // hashed hop: window unused
+      } else if (hops.src[h] == 0) {
         lo = lower_bound_i64(hops.key12[h], hops.n[h],
                              (hops.const_hi[h] << 32) | first_b);
         hi = upper_bound_i64(hops.key12[h], hops.n[h],
@@ -554,8 +603,26 @@ __global__ void chain_count_kernel(const int64_t* __restrict__ seed_key12,
     // chains — letting them all issue gives the scheduler ILP to hide
     // L2 latency
     for (int h = 0; h < hops.k; ++h) {
-      int64_t key = (hops.const_hi[h] << 32)
-                    | (hops.src[h] == 0 ? b_comp : z_comp);
+      int64_t comp = hops.src[h] == 0 ? b_comp : z_comp;
+      if (hops.table[h] != nullptr) {
+        // count-table hop: one 8-byte L2 load (vs log2(n) lines)
+        uint32_t v = static_cast<uint32_t>(comp);
+        uint32_t mask = static_cast<uint32_t>(hops.tmask[h]);
+        uint32_t slot = h32(v) & mask;
+        unsigned long long cnt = 0;
+        for (;;) {
+          unsigned long long e = hops.table[h][slot];
+          if (e == kTblEmpty) break;
+          if (static_cast<uint32_t>(e >> 32) == v) {
+            cnt = e & 0xFFFFFFFFull;
+            break;
+          }
+          slot = (slot + 1) & mask;
+        }
+        prod *= cnt;
+        continue;
+      }
+      int64_t key = (hops.const_hi[h] << 32) | comp;
       int64_t wlo = win[(t * hops.k + h) * 2];
       int64_t wspan = win[(t * hops.k + h) * 2 + 1] - wlo;
       const int64_t* base = hops.key12[h] + wlo;
@@ -575,11 +642,13 @@ __global__ void chain_count_kernel(const int64_t* __restrict__ seed_key12,
 int64_t chain_count(at::Tensor seed_key12, at::Tensor seed_z,
                     std::vector<at::Tensor> hop_key12,
                     std::vector<int64_t> hop_const_hi,
-                    std::vector<int64_t> hop_src) {
+                    std::vector<int64_t> hop_src,
+                    std::vector<at::Tensor> hop_table) {
   TORCH_CHECK(seed_key12.is_cuda() && seed_z.is_cuda());
   TORCH_CHECK(hop_key12.size() <= static_cast<size_t>(kMaxHops));
   TORCH_CHECK(hop_key12.size() == hop_const_hi.size()
-              && hop_key12.size() == hop_src.size());
+              && hop_key12.size() == hop_src.size()
+              && hop_key12.size() == hop_table.size());
   int64_t m = seed_key12.numel();
   ChainHops hops{};
   hops.k = static_cast<int>(hop_key12.size());
@@ -589,6 +658,13 @@ int64_t chain_count(at::Tensor seed_key12, at::Tensor seed_z,
     hops.n[h] = hop_key12[h].numel();
     hops.const_hi[h] = hop_const_hi[h];
     hops.src[h] = static_cast<int32_t>(hop_src[h]);
+    if (hop_table[h].numel() > 0) {
+      TORCH_CHECK(hop_table[h].is_cuda()
+                  && hop_table[h].dtype() == at::kLong);
+      hops.table[h] = reinterpret_cast<const unsigned long long*>(
+          hop_table[h].data_ptr<int64_t>());
+      hops.tmask[h] = hop_table[h].numel() - 1;
+    }
   }
   auto total = at::zeros({1}, seed_key12.options());
   if (m > 0) {
@@ -1121,6 +1197,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "K2 chained hash join over int32 key columns -> (li, ri)");
   m.def("chain_count", &chain_count,
         "fused COUNT(*) over a seed scan + probe-hop chain");
+  m.def("build_count_table", &build_count_table,
+        "open-addressing (value -> match count) table from packed "
+        "(val<<32)|count entries, for hashed chain-count hops");
   m.def("probe_fused", &probe_fused,
         "K1 fused probe: inline key pack + merge-path + carry emit");
   m.def("probe_exact_counts", &probe_exact_counts,
