@@ -238,3 +238,65 @@ def test_prepared_plan_cache_gpu():
     assert db.query(q) == [["1000"]]          # cache hit
     db.add_triple(f"<{EX}extra>", f"<{EX}p>", '"x"')
     assert db.query(q) == [["1001"]]          # invalidated by version bump
+
+
+@requires_gpu
+def test_chain_count_table_hop_matches_search():
+    """The hashed count-table hop (small region, many seeds) must agree
+    with both the binary-search hop and the torch oracle."""
+    from kolibrie_amd.ops import _native
+    g = torch.Generator().manual_seed(5)
+    n_seeds, n_region = 20_000, 600
+    pid = 7
+    # seed: sorted (a,b) pairs + z column
+    a = torch.randint(0, 500, (n_seeds,), generator=g, dtype=torch.int64)
+    b = torch.randint(0, 300, (n_seeds,), generator=g, dtype=torch.int64)
+    key = (a << 32) | b
+    seed_key12, _ = torch.sort(key)
+    seed_z = torch.randint(0, 300, (n_seeds,), generator=g, dtype=torch.int32)
+    # hop region: sorted (pid, v) packed keys with duplicates
+    v = torch.randint(0, 300, (n_region,), generator=g, dtype=torch.int64)
+    region, _ = torch.sort((torch.full_like(v, pid) << 32) | v)
+
+    def oracle(src):
+        comp = (seed_key12 & 0xFFFFFFFF) if src == 0 else seed_z.to(torch.int64)
+        keys = (torch.full_like(comp, pid).to(torch.int64) << 32) | comp
+        lo = torch.searchsorted(region, keys, side="left")
+        hi = torch.searchsorted(region, keys, side="right")
+        return int((hi - lo).sum())
+
+    dev = "cuda:0"
+    sk, sz, rg = seed_key12.to(dev), seed_z.to(dev), region.to(dev)
+    vals, counts = torch.unique_consecutive(rg & 0xFFFFFFFF,
+                                            return_counts=True)
+    table = _native.build_count_table((vals << 32) | counts)
+    assert table.numel() >= 2 * vals.numel()
+    for src in (0, 1):
+        want = oracle(src)
+        got_search = _native.chain_count(
+            sk, sz, [rg], [pid], [src], [torch.empty(0, dtype=torch.int64,
+                                                     device=dev)])
+        got_table = _native.chain_count(sk, sz, [rg], [pid], [src], [table])
+        assert got_search == want
+        assert got_table == want
+
+
+@requires_gpu
+def test_chain_count_e2e_flagship_shape():
+    """End-to-end flagship COUNT query on GPU (exercises the cached
+    count-table path through the executor) vs the same data on CPU."""
+    from kolibrie_amd import SparqlDatabase
+    from kolibrie_amd.engine.query import execute_query
+    from kolibrie_amd.parallel.synthetic import (FLAGSHIP_QUERY, plan_dataset,
+                                                 generate_partition)
+    counts = {}
+    for dev in ("cpu", "cuda:0"):
+        db = SparqlDatabase(device=dev)
+        ds = plan_dataset(db, 5_000_000)
+        s, p, o = generate_partition(ds, 0, 1, 99, dev)
+        db.store.insert_bulk(0, s, p, o)
+        r1 = execute_query(FLAGSHIP_QUERY, db)
+        r2 = execute_query(FLAGSHIP_QUERY, db)  # cached-plan + cached-region
+        assert r1 == r2
+        counts[dev] = r1[0][0]
+    assert counts["cpu"] == counts["cuda:0"]
