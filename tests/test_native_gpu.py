@@ -248,11 +248,12 @@ def test_chain_count_table_hop_matches_search():
     g = torch.Generator().manual_seed(5)
     n_seeds, n_region = 20_000, 600
     pid = 7
-    # seed: sorted (a,b) pairs + z column
-    a = torch.randint(0, 500, (n_seeds,), generator=g, dtype=torch.int64)
-    b = torch.randint(0, 300, (n_seeds,), generator=g, dtype=torch.int64)
-    key = (a << 32) | b
-    seed_key12, _ = torch.sort(key)
+    # seed region mirrors a PSO predicate slice: constant leading component,
+    # MONOTONE low (subject) component — the precondition chain_tile_bounds
+    # documents for src==0 window narrowing
+    b, _ = torch.sort(torch.randint(0, 300, (n_seeds,), generator=g,
+                                    dtype=torch.int64))
+    seed_key12 = (torch.full_like(b, 3) << 32) | b
     seed_z = torch.randint(0, 300, (n_seeds,), generator=g, dtype=torch.int32)
     # hop region: sorted (pid, v) packed keys with duplicates
     v = torch.randint(0, 300, (n_region,), generator=g, dtype=torch.int64)

@@ -194,3 +194,81 @@ def test_window_runner():
         runner.push(("e", ts), ts)
     fired = runner.drain()
     assert len(fired) == 2
+
+
+# ---- forward-chaining shape parity (ref datalog/tests/reasoning_tests.rs:
+# fc_variable_predicate_premise, fc_repeated_variable_premise,
+# fc_constant_subject/object_premise, fc_multi_conclusion,
+# fc_diamond_ancestor, fc_shared_predicate_variable, fc_idempotent) ----
+
+def _rule(r, concl, body):
+    r.add_rule_text(f"RULE :t :- CONSTRUCT {{ {concl} }} WHERE {{ {body} }}")
+
+
+def test_fc_variable_predicate_premise():
+    r = Reasoner()
+    r.add_abox_triple("a", "likes", "b")
+    r.add_abox_triple("b", "hates", "c")
+    _rule(r, "?x <related> ?y", "?x ?p ?y")
+    r.infer_new_facts_semi_naive()
+    assert sorted(r.query_abox(None, "related", None)) == [
+        ("a", "related", "b"), ("b", "related", "c")]
+
+
+def test_fc_repeated_variable_premise():
+    r = Reasoner()
+    r.add_abox_triple("a", "p", "a")
+    r.add_abox_triple("a", "p", "b")
+    _rule(r, "?x <selfloop> ?x", "?x <p> ?x")
+    r.infer_new_facts_semi_naive()
+    assert r.query_abox(None, "selfloop", None) == [("a", "selfloop", "a")]
+
+
+def test_fc_shared_predicate_variable_join():
+    r = Reasoner()
+    r.add_abox_triple("a", "q", "b")
+    r.add_abox_triple("b", "q", "c")
+    r.add_abox_triple("b", "r", "d")  # different predicate: must not join
+    _rule(r, "?x <two> ?z", "?x ?p ?y . ?y ?p ?z")
+    r.infer_new_facts_semi_naive()
+    assert r.query_abox(None, "two", None) == [("a", "two", "c")]
+
+
+def test_fc_constant_subject_premise_filters():
+    r = Reasoner()
+    r.add_abox_triple("admin", "grants", "u1")
+    r.add_abox_triple("other", "grants", "u2")
+    _rule(r, "?u <trusted> <yes>", "<admin> <grants> ?u")
+    r.infer_new_facts_semi_naive()
+    assert r.query_abox(None, "trusted", None) == [("u1", "trusted", "yes")]
+
+
+def test_fc_multi_conclusion():
+    r = Reasoner()
+    r.add_abox_triple("x", "parent", "y")
+    _rule(r, "?b <child> ?a . ?a <ancestor> ?b", "?a <parent> ?b")
+    r.infer_new_facts_semi_naive()
+    assert r.query_abox(None, "child", None) == [("y", "child", "x")]
+    assert r.query_abox(None, "ancestor", None) == [("x", "ancestor", "y")]
+
+
+def test_fc_diamond_ancestor_dedup():
+    r = Reasoner()
+    for s, p, o in [("top", "edge", "l"), ("top", "edge", "r"),
+                    ("l", "edge", "bot"), ("r", "edge", "bot")]:
+        r.add_abox_triple(s, p, o)
+    _rule(r, "?x <reach> ?z", "?x <edge> ?z")
+    _rule(r, "?x <reach> ?z", "?x <reach> ?y . ?y <edge> ?z")
+    r.infer_new_facts_semi_naive()
+    # both diamond paths derive top->bot exactly once
+    assert sorted(r.query_abox("top", "reach", None)) == [
+        ("top", "reach", "bot"), ("top", "reach", "l"), ("top", "reach", "r")]
+
+
+def test_fc_idempotent_reinfer():
+    r = Reasoner()
+    r.add_abox_triple("a", "edge", "b")
+    _rule(r, "?x <reach> ?y", "?x <edge> ?y")
+    n1 = r.infer_new_facts_semi_naive()
+    n2 = r.infer_new_facts_semi_naive()
+    assert n1 >= 1 and n2 == 0
