@@ -221,6 +221,7 @@ class WindowSpec:
 class SyncPolicy:
     kind: str = "Wait"                    # Steal | Wait | Timeout
     timeout_ms: Optional[int] = None
+    fallback: str = "Steal"               # Timeout expiry action: Steal | Drop
 
 
 @dataclass

@@ -743,14 +743,32 @@ class Parser:
         if self.at_kw("WITH"):
             self.next()
             self.expect_kw("POLICY")
-            kind = self.next().upper()
-            if kind == "TIMEOUT":
+            if self.at("("):
+                # reference form (parser.rs:2737 parse_sync_policy_timeout):
+                # WITH POLICY (timeout = 5s, fallback = steal|drop)
+                self.next()
+                self.expect_kw("TIMEOUT")
+                self.expect("=")
                 dur = self.parse_policy_duration()
-                policy = SyncPolicy("Timeout", dur)
-            elif kind in ("STEAL", "WAIT"):
-                policy = SyncPolicy(kind.capitalize())
+                self.expect(",")
+                self.expect_kw("FALLBACK")
+                self.expect("=")
+                fb = self.next().upper().capitalize()
+                if fb not in ("Steal", "Drop"):
+                    raise ParseError(f"bad timeout fallback {fb!r}",
+                                     self.text, self.peek().pos
+                                     if self.peek() else 0)
+                self.expect(")")
+                policy = SyncPolicy("Timeout", dur, fb)
             else:
-                policy = SyncPolicy("Wait")
+                kind = self.next().upper()
+                if kind == "TIMEOUT":
+                    dur = self.parse_policy_duration()
+                    policy = SyncPolicy("Timeout", dur)
+                elif kind in ("STEAL", "WAIT"):
+                    policy = SyncPolicy(kind.capitalize())
+                else:
+                    policy = SyncPolicy("Wait")
         return WindowClause(window_iri, stream_iri, spec, policy)
 
     def parse_window_spec(self) -> WindowSpec:
