@@ -96,6 +96,7 @@ class RSPEngine:
         self.consumers: List[Callable[[List[Tuple]], None]] = []
         self.projection: Optional[SelectQuery] = None
         self._pending_results: Dict[str, WindowResult] = {}
+        self._pending_since: Optional[float] = None
         self._raw_by_window: Dict[str, List[Tuple[Tuple[int, int, int], int]]] = {}
         self._window_triples: Dict[str, List[Tuple[int, int, int]]] = {}
         self._coord_lock = threading.Lock()
@@ -246,6 +247,7 @@ class RSPEngine:
             try:
                 content = entry.queue.get(timeout=0.1)
             except Empty:
+                self._check_timeout_expiry()
                 continue
             self._process_window(entry, content)
 
@@ -274,6 +276,8 @@ class RSPEngine:
                 result = WindowResult(entry.iri, rows, entry.plan_vars,
                                       content.last_timestamp_changed)
             if len(self.windows) > 1:
+                if not self._pending_results:
+                    self._pending_since = _time.time()
                 self._pending_results[entry.iri] = result
                 self._try_emit_joined(result.ts)
             else:
@@ -296,7 +300,29 @@ class RSPEngine:
         results = [self._pending_results[w] for w in sorted(have)]
         if policy in ("Wait", "Timeout"):
             self._pending_results = {}
+            self._pending_since = None
         self._emit(results, ts)
+
+    def _check_timeout_expiry(self):
+        """Timeout policy, MultiThread mode: after `duration` with a partial
+        result set, apply the fallback — Steal emits with the windows that
+        HAVE fired, Drop discards the partial set (ref rsp_engine.rs:584-636,
+        shared/query.rs:243 Timeout{duration, fallback})."""
+        if self.sync_policy.kind != "Timeout":
+            return
+        with self._coord_lock:
+            if not self._pending_results or self._pending_since is None:
+                return
+            elapsed_ms = (_time.time() - self._pending_since) * 1000.0
+            if elapsed_ms <= (self.sync_policy.timeout_ms or 0):
+                return
+            pending = self._pending_results
+            self._pending_results = {}
+            self._pending_since = None
+            if self.sync_policy.fallback == "Steal":
+                results = [pending[w] for w in sorted(pending)]
+                self._emit(results, max(r.ts for r in results))
+            # Drop: partial set discarded, nothing emitted
 
     def _drain_multithread(self, deadline_ms: int):
         t0 = _time.time()

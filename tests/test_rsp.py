@@ -265,3 +265,78 @@ WHERE {{ WINDOW <http://w1> {{ ?m ex:temp ?v }} }}"""
     counts_host = [int(rows[0][0]) for rows in out_host if rows]
     assert counts_bulk == counts_host
     assert sum(counts_bulk) == sum(1 for t in ts.tolist() if t < 20)
+
+
+def _timeout_engine(fallback):
+    """Two-window MultiThread engine with a short Timeout policy."""
+    import time
+    from kolibrie_amd.parsing.ast import SyncPolicy
+    from kolibrie_amd.rsp.engine import OperationMode, RSPEngine
+    eng = RSPEngine(operation_mode=OperationMode.MULTI_THREAD,
+                    sync_policy=SyncPolicy("Timeout", 200, fallback))
+    eng.add_window("<w1>", "<s1>", 4, 4)
+    eng.add_window("<w2>", "<s2>", 4, 4)
+    got = []
+    eng.add_consumer(lambda rows: got.append(rows))
+    return eng, got
+
+
+def _wait_for(pred, timeout_s=3.0):
+    import time
+    t0 = time.time()
+    while time.time() - t0 < timeout_s:
+        if pred():
+            return True
+        time.sleep(0.02)
+    return pred()
+
+
+def test_timeout_steal_policy_emits_partial():
+    """ref rsp_engine_test.rs test_timeout_steal_policy: only one of two
+    windows fires; after the timeout the Steal fallback emits with the
+    window that did."""
+    eng, got = _timeout_engine("Steal")
+    for ts in range(5):
+        eng.add_to_stream("<s1>", ("<a>", "<p>", f"<o{ts}>"), ts)
+    # w1 fired at ts=4; w2 never does.  Expiry check runs in the workers.
+    assert _wait_for(lambda: len(got) > 0)
+    assert any(r for r in got)
+    eng.stop()
+
+
+def test_timeout_drop_policy_discards_partial():
+    """ref test_timeout_drop_policy: Drop fallback discards the partial
+    result set — nothing is emitted."""
+    import time
+    eng, got = _timeout_engine("Drop")
+    for ts in range(5):
+        eng.add_to_stream("<s1>", ("<a>", "<p>", f"<o{ts}>"), ts)
+    time.sleep(1.0)  # > timeout; workers run the expiry check
+    assert got == []
+    assert eng._pending_results == {}  # partial set was dropped, not stuck
+    eng.stop()
+
+
+def test_timeout_wait_completes_before_expiry():
+    """Both windows fire promptly: Timeout behaves like Wait (joined emit)."""
+    eng, got = _timeout_engine("Drop")
+    for ts in range(5):
+        eng.add_to_stream("<s1>", ("<a>", "<p>", "<o>"), ts)
+        eng.add_to_stream("<s2>", ("<a>", "<q>", "<u>"), ts)
+    assert _wait_for(lambda: len(got) > 0)
+    eng.stop()
+
+
+def test_policy_timeout_fallback_grammar():
+    """WITH POLICY (timeout = 5s, fallback = drop) — ref parser.rs:2737."""
+    from kolibrie_amd.parsing.sparql import parse_combined_query
+    q = parse_combined_query("""
+        REGISTER RSTREAM <out> AS
+        SELECT ?s FROM NAMED WINDOW <w1> ON STREAM <s1> [RANGE 10 STEP 5]
+        WITH POLICY (timeout = 5s, fallback = drop)
+        WHERE { WINDOW <w1> { ?s <p> ?o } }
+    """)
+    pol = q.register.windows[0].policy
+    assert pol.kind == "Timeout"
+    assert pol.timeout_ms == 5000
+    assert pol.fallback == "Drop"
