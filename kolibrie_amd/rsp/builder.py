@@ -130,6 +130,22 @@ class RSPBuilder:
         for cr in cq.rules:
             from ..reasoning.rule import convert_combined_rule
             engine.store.add_rule(convert_combined_rule(cr, db, prefixes))
+        # PROB(provenance=hybrid, ...) on any rule turns on hybrid window
+        # evaluation (ref rsp_engine_test.rs hybrid_rsp_rule: the annotation,
+        # not an explicit config, selects the hybrid path)
+        if engine.store.hybrid_config is None:
+            for r in engine.store.rules:
+                pa = getattr(r, "prob", None)
+                if pa is not None and pa.provenance == "hybrid":
+                    from ..reasoning.hybrid import HybridConfig
+                    cfg = HybridConfig()
+                    if pa.threshold is not None:
+                        cfg.threshold = pa.threshold
+                    if pa.confidence is not None:
+                        cfg.confidence = pa.confidence
+                    cfg.validate()
+                    engine.store.hybrid_config = cfg
+                    break
         for cw in self._cross_window_rules:
             from ..reasoning.n3_rules import parse_n3_rules_for_sds
             engine.cross_window_rules.extend(

@@ -97,6 +97,8 @@ class RSPEngine:
         self.projection: Optional[SelectQuery] = None
         self._pending_results: Dict[str, WindowResult] = {}
         self._pending_since: Optional[float] = None
+        self._stable_seed_ids: Dict[str, int] = {}
+        self._seed_probs: Dict[str, float] = {}
         self._raw_by_window: Dict[str, List[Tuple[Tuple[int, int, int], int]]] = {}
         self._window_triples: Dict[str, List[Tuple[int, int, int]]] = {}
         self._coord_lock = threading.Lock()
@@ -169,11 +171,32 @@ class RSPEngine:
         stream_iri = normalize_iri(stream_iri)
         item = self.store._encode(triple)
         self.store.db.probability_seeds[item] = probability
+        self._seed_probs[seed_id] = probability
         occ = ProbabilisticOccurrence(
             item=item, event=EventKey(stream_iri, ts), seed_id=seed_id)
         for entry in self.windows.values():
             if entry.stream_iri == stream_iri or entry.stream_iri.startswith("?"):
                 entry.window.add_probabilistic_to_window(occ)
+
+    def _snapshot_from_content(self, content):
+        """Window content -> hybrid SeedSnapshot: each probabilistic
+        occurrence keeps a stable numeric seed id (shared across
+        overlapping windows — ref rsp_overlapping_windows_share_one_
+        occurrence_identity), and a deterministic copy of the same triple
+        dominates (the triple is then a certain fact, not a seed — ref
+        rsp_deterministic_fact_dominates_probabilistic_copy)."""
+        from ..reasoning.hybrid import SeedSnapshot
+        seeds = {}
+        for occ in content.probabilistic_occurrences:
+            if content.is_deterministic(occ.item):
+                continue
+            sid = self._stable_seed_ids.setdefault(
+                occ.seed_id, len(self._stable_seed_ids) + 1)
+            prob = self._seed_probs.get(
+                occ.seed_id,
+                self.store.db.probability_seeds.get(occ.item, 0.5))
+            seeds[tuple(x & 0xFFFFFFFF for x in occ.item)] = (sid, prob)
+        return SeedSnapshot(seeds)
 
     # ------------------------------------------------- bulk (K7) ingestion
     def add_to_stream_bulk(self, stream_iri: str, s, p, o, ts):
@@ -264,6 +287,8 @@ class RSPEngine:
             if self.cross_window_mode is not None:
                 self._emit_cross_window(content.last_timestamp_changed)
                 return
+            if getattr(self.store, "hybrid_config", None) is not None:
+                self.store.seed_snapshot = self._snapshot_from_content(content)
             self.store.materialize()
             self.latest_hybrid_results = self.store.latest_hybrid_results
             if entry.plan is None:

@@ -340,3 +340,95 @@ def test_policy_timeout_fallback_grammar():
     assert pol.kind == "Timeout"
     assert pol.timeout_ms == 5000
     assert pol.fallback == "Drop"
+
+
+# ---- probabilistic / hybrid RSP (ref rsp_engine_test.rs:1516-1650) ----
+
+_HYBRID_QUERY = """
+    REGISTER RSTREAM <out> AS
+    SELECT ?s FROM NAMED WINDOW <w> ON STREAM <s1> [RANGE 4 STEP 4]
+    WHERE { WINDOW <w> { ?s <http://test/result> <http://test/yes> } }
+"""
+_HYBRID_RULE = """
+    RULE :Hybrid PROB(provenance=hybrid, threshold=0.7) :-
+    CONSTRUCT { ?s <http://test/result> <http://test/yes> }
+    WHERE { ?s <http://test/input> <http://test/yes> }
+"""
+
+
+def _hybrid_engine():
+    from kolibrie_amd.rsp.builder import RSPBuilder
+    return (RSPBuilder()
+            .add_rsp_ql_query(_HYBRID_QUERY)
+            .add_sparql_rules(_HYBRID_RULE)
+            .build())
+
+
+def test_prob_annotation_selects_hybrid_path():
+    eng = _hybrid_engine()
+    assert eng.store.hybrid_config is not None
+    assert abs(eng.store.hybrid_config.threshold - 0.7) < 1e-9
+
+
+def test_rsp_deterministic_fact_dominates_probabilistic_copy():
+    """Certain copy of a probabilistic event -> probability 1.0
+    (ref rsp_engine_test.rs:1516)."""
+    eng = _hybrid_engine()
+    certain = ("<http://test/certain>", "<http://test/input>", "<http://test/yes>")
+    trigger = ("<http://test/trigger>", "<http://test/input>", "<http://test/yes>")
+    eng.add_to_stream("<s1>", certain, 1)
+    eng.add_probabilistic_to_stream("<s1>", certain, 1, "e1", 0.2)
+    eng.add_probabilistic_to_stream("<s1>", trigger, 2, "e2", 0.6)
+    for entry in eng.windows.values():
+        entry.window.flush()
+    results = eng.latest_hybrid_results
+    assert results, "hybrid window evaluation must publish results"
+    assert any(r.probability is not None and abs(r.probability - 1.0) < 1e-9
+               for r in results.values())
+
+
+def test_rsp_probabilistic_occurrences_keep_stable_distinct_seed_ids():
+    eng = _hybrid_engine()
+    a = ("<http://test/a>", "<http://test/input>", "<http://test/yes>")
+    b = ("<http://test/b>", "<http://test/input>", "<http://test/yes>")
+    eng.add_probabilistic_to_stream("<s1>", a, 1, "occ-a", 0.4)
+    eng.add_probabilistic_to_stream("<s1>", b, 2, "occ-b", 0.9)
+    for entry in eng.windows.values():
+        entry.window.flush()
+    ids = eng._stable_seed_ids
+    assert ids["occ-a"] != ids["occ-b"]
+    # re-flushing does not renumber
+    before = dict(ids)
+    for entry in eng.windows.values():
+        entry.window.flush()
+    assert eng._stable_seed_ids == before
+
+
+def test_rsp_probabilistic_derivation_probability():
+    """Derived triple inherits its single seed's probability exactly."""
+    eng = _hybrid_engine()
+    t = ("<http://test/x>", "<http://test/input>", "<http://test/yes>")
+    eng.add_probabilistic_to_stream("<s1>", t, 1, "seed-x", 0.8)
+    for entry in eng.windows.values():
+        entry.window.flush()
+    results = eng.latest_hybrid_results
+    assert results
+    assert any(r.probability is not None and abs(r.probability - 0.8) < 1e-6
+               for r in results.values())
+
+
+def test_rsp_hybrid_threshold_gates_derivation():
+    """Below-threshold derivations are not asserted into the window store
+    (threshold=0.7; a 0.4 seed's conclusion stays out of query results)."""
+    eng = _hybrid_engine()
+    got = []
+    eng.add_consumer(lambda rows: got.append(rows))
+    lo = ("<http://test/lo>", "<http://test/input>", "<http://test/yes>")
+    hi = ("<http://test/hi>", "<http://test/input>", "<http://test/yes>")
+    eng.add_probabilistic_to_stream("<s1>", lo, 1, "s-lo", 0.4)
+    eng.add_probabilistic_to_stream("<s1>", hi, 2, "s-hi", 0.95)
+    for entry in eng.windows.values():
+        entry.window.flush()
+    flat = [v for rows in got for r in rows for v in r]
+    assert any("hi" in v for v in flat)
+    assert not any("/lo" in v for v in flat)
