@@ -108,3 +108,25 @@ def test_typed_hybrid_status_annotations():
             ?t <http://kolibrie.amd/hybrid#status> ?st . FILTER(isTRIPLE(?t))
         }""")
     assert rows and rows[0][0] in ("Decided", "DecidedExact")
+
+
+def test_count_star_fast_path_matches_generic():
+    """The colless COUNT(*) fast path must agree with the generic
+    finalize path (cached vs uncached entry), incl. empty data and
+    LIMIT/OFFSET edges."""
+    from kolibrie_amd import SparqlDatabase
+    from kolibrie_amd.engine.query import execute_query, execute_select
+    from kolibrie_amd.parsing.sparql import parse_combined_query
+    db = SparqlDatabase()
+    for i in range(20):
+        db.add_triple(f"<http://e/s{i}>", "<http://e/p>", f"<http://e/o{i%3}>")
+        db.add_triple(f"<http://e/s{i}>", "<http://e/q>", "<http://e/z>")
+    q = ('SELECT (COUNT(*) AS ?c) WHERE { ?s <http://e/p> ?o . '
+         '?s <http://e/q> ?z }')
+    cached = execute_query(q, db)
+    cq = parse_combined_query(q)
+    plain = execute_select(cq.select, db, dict(db.prefixes))
+    assert cached == plain == [["20"]]
+    assert execute_query(q + " LIMIT 0", db) == []
+    assert execute_query(q + " OFFSET 1", db) == []
+    assert execute_query(q, SparqlDatabase()) == [["0"]]
