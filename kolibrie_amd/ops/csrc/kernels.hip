@@ -627,7 +627,13 @@ __global__ void chain_count_kernel(const int64_t* __restrict__ seed_key12,
       int64_t wspan = win[(t * hops.k + h) * 2 + 1] - wlo;
       const int64_t* base = hops.key12[h] + wlo;
       int64_t lo = lower_bound_i64(base, wspan, key);
-      int64_t hi = upper_bound_i64(base, wspan, key);
+      // match runs are tiny (one object per subject in typical star data):
+      // walk forward a few cache-hot slots instead of paying a second full
+      // log2(wspan) dependent-load chain; fall back for genuine skew
+      int64_t hi = lo;
+      while (hi < wspan && hi - lo < 4 && base[hi] == key) ++hi;
+      if (hi - lo == 4 && hi < wspan && base[hi] == key)
+        hi = lo + upper_bound_i64(base + lo, wspan - lo, key);
       prod *= static_cast<unsigned long long>(hi - lo);
     }
     acc += prod;
