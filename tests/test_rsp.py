@@ -432,3 +432,28 @@ def test_rsp_hybrid_threshold_gates_derivation():
     flat = [v for rows in got for r in rows for v in r]
     assert any("hi" in v for v in flat)
     assert not any("/lo" in v for v in flat)
+
+
+def test_rsp_overlapping_windows_share_one_occurrence_identity():
+    """Sliding windows (RANGE 4 STEP 2): the same probabilistic occurrence
+    lands in two overlapping windows but keeps ONE seed identity
+    (ref rsp_engine_test.rs:1602)."""
+    from kolibrie_amd.rsp.builder import RSPBuilder
+    q = """
+        REGISTER RSTREAM <out> AS
+        SELECT ?s FROM NAMED WINDOW <w> ON STREAM <s1> [RANGE 4 STEP 2]
+        WHERE { WINDOW <w> { ?s <http://test/result> <http://test/yes> } }
+    """
+    eng = (RSPBuilder().add_rsp_ql_query(q)
+           .add_sparql_rules(_HYBRID_RULE).build())
+    t = ("<http://test/x>", "<http://test/input>", "<http://test/yes>")
+    eng.add_probabilistic_to_stream("<s1>", t, 3, "occ-shared", 0.6)
+    # push time forward so both overlapping windows [0,4) and [2,6) fire
+    for ts in range(4, 9):
+        eng.add_to_stream("<s1>", ("<http://test/tick>", "<http://test/t>",
+                                   f"<http://test/{ts}>"), ts)
+    for entry in eng.windows.values():
+        entry.window.flush()
+    assert list(eng._stable_seed_ids.values()).count(
+        eng._stable_seed_ids.get("occ-shared")) == 1
+    assert len(eng._stable_seed_ids) == 1
