@@ -54,8 +54,6 @@ def materialize_neural_relation(name: str, entry: dict, db,
     if norm is not None:
         x = (x - norm[0]) / norm[1]
     proba = model.predict_proba(x)
-    if proba.dim() > 1:
-        proba = proba[:, -1]
     # anchor = the subject variable of the first input pattern
     anchor_term = patterns[0][0]
     anchor_var = anchor_term[1:] if anchor_term.startswith("?") else None
@@ -63,14 +61,40 @@ def materialize_neural_relation(name: str, entry: dict, db,
         return 0
     pred_id = db.dictionary.encode(db.resolve_lexical(decl.name, prefixes))
     labels = (model_entry or {}).get("labels") or ["true"]
-    positive = db.dictionary.encode(labels[0])
-    threshold = float(decl.options.get("threshold", DEFAULT_THRESHOLD))
+    # re-materialization first removes THIS relation's previous assertions
+    # — never user facts with the same predicate nor other conclusions
+    # (ref neural_relations.rs:488 remove_materialized_triples +
+    # neural_materialized_triples registry; tests rerun_cleans_stale_
+    # predictions / preserves_non_ml_conclusions)
+    tracked = db.neural_materialized_triples.setdefault(pred_id, [])
+    for (ts, tp, to) in tracked:
+        db.store.delete_quad(0, ts, tp, to)
+        db.probability_seeds.pop((ts, tp, to), None)
+    tracked.clear()
     anchors = (rows.col(anchor_var).to(torch.int64) & 0xFFFFFFFF).cpu().tolist()
     n_asserted = 0
+    if proba.dim() > 1 and proba.shape[1] > 1 and len(labels) > 1:
+        # exclusive (multiclass) output: assert the argmax label per anchor
+        # (ref NeuralOutputKind::Exclusive, neural_relations.rs:492-511)
+        best = proba.argmax(dim=1).cpu().tolist()
+        conf = proba.max(dim=1).values.detach().cpu().tolist()
+        label_ids = [db.dictionary.encode(l) for l in labels]
+        for a, bi, p in zip(anchors, best, conf):
+            obj = label_ids[bi] if bi < len(label_ids) else label_ids[-1]
+            db.store.insert_quad(0, a, pred_id, obj)
+            db.probability_seeds[(a, pred_id, obj)] = p
+            tracked.append((a, pred_id, obj))
+            n_asserted += 1
+        return n_asserted
+    if proba.dim() > 1:
+        proba = proba[:, -1]
+    positive = db.dictionary.encode(labels[0])
+    threshold = float(decl.options.get("threshold", DEFAULT_THRESHOLD))
     for a, p in zip(anchors, proba.detach().cpu().tolist()):
         if p >= threshold:
             db.store.insert_quad(0, a, pred_id, positive)
             db.probability_seeds[(a, pred_id, positive)] = p
+            tracked.append((a, pred_id, positive))
             n_asserted += 1
     return n_asserted
 

@@ -144,6 +144,10 @@ class SparqlDatabase:
         self.rules: List = []  # parsed shared Rule objects registered via queries
         self.neural_models: Dict[str, dict] = {}
         self.neural_relations: Dict[str, dict] = {}
+        # pred_id -> [(s,p,o)] asserted by materialize_neural_relation;
+        # cleared+re-asserted on each re-materialization (ref
+        # neural_relations.rs neural_materialized_triples)
+        self.neural_materialized_triples: Dict[int, List] = {}
         self.train_decls: List[dict] = []
         self.probability_seeds: Dict[Tuple[int, int, int], float] = {}
         self._stats = None

@@ -133,3 +133,41 @@ def test_ml_predict_in_query():
     assert out.has("p")
     val = db.dictionary.decode(int(out.col("p")[0].item()) & 0xFFFFFFFF)
     assert 0.0 <= float(val) <= 1.0
+
+
+def test_rerun_cleans_stale_predictions_preserves_other_facts():
+    """Re-materialization removes the relation's previous assertions but
+    never user facts (ref ml_predict_candle_runtime.rs
+    rerun_cleans_stale_predictions / preserves_non_ml_conclusions)."""
+    torch.manual_seed(3)
+    db = SparqlDatabase()
+    for i in range(30):
+        sal = 1000 + i * 200
+        label = "1" if sal > 3500 else "0"
+        db.add_triple(f"<{EX}e{i}>", f"<{EX}salary>", f'"{sal}"')
+        db.add_triple(f"<{EX}e{i}>", f"<{EX}lbl>", f'"{label}"')
+    # a USER-asserted fact with the SAME predicate as the neural relation
+    db.add_triple(f"<{EX}manual>", f"<{EX}risky>", '"flagged"')
+    decls = f"""
+        MODEL "riskm" {{ ARCH MLP {{ HIDDEN [8] }} OUTPUT BINARY {{"flagged"}} }}
+        NEURAL RELATION <{EX}risky> USING MODEL "riskm" {{
+            INPUT {{ ?x <{EX}salary> ?s }}
+            FEATURES {{ ?s }}
+        }}
+        TRAIN NEURAL RELATION <{EX}risky> USING MODEL "riskm" {{
+            DATA {{ ?x <{EX}salary> ?s . ?x <{EX}lbl> ?y }}
+            label = ?y
+            epochs = 150
+        }}
+        SELECT ?x WHERE {{ ?x <{EX}risky> "flagged" }}
+    """
+    first = {r[0] for r in db.query(decls)}
+    assert f"{EX}manual" in first          # user fact survives materialization
+    assert f"{EX}e29" in first
+    # drop the high earners' salary facts; re-query re-materializes
+    for i in range(15, 30):
+        db.query(f'DELETE DATA {{ <{EX}e{i}> <{EX}salary> "{1000 + i * 200}" }}')
+    second = {r[0] for r in
+              db.query(f'SELECT ?x WHERE {{ ?x <{EX}risky> "flagged" }}')}
+    assert f"{EX}manual" in second         # user fact still there
+    assert f"{EX}e29" not in second        # stale prediction cleaned
