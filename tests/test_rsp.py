@@ -457,3 +457,25 @@ def test_rsp_overlapping_windows_share_one_occurrence_identity():
     assert list(eng._stable_seed_ids.values()).count(
         eng._stable_seed_ids.get("occ-shared")) == 1
     assert len(eng._stable_seed_ids) == 1
+
+
+def test_static_data_not_visible_in_window_query():
+    """Static triples share the dictionary but never enter windows
+    (ref rsp_engine_test.rs test_static_data_not_visible_in_window_query,
+    rsp_engine.rs:321-326)."""
+    from kolibrie_amd.rsp.builder import RSPBuilder
+    q = """
+        REGISTER RSTREAM <out> AS
+        SELECT ?s FROM NAMED WINDOW <w> ON STREAM <s1> [RANGE 4 STEP 4]
+        WHERE { WINDOW <w> { ?s <http://t/p> <http://t/o> } }
+    """
+    got = []
+    eng = (RSPBuilder().add_rsp_ql_query(q)
+           .add_static_ntriples("<http://t/static> <http://t/p> <http://t/o> .")
+           .add_consumer(lambda rows: got.append(rows)).build())
+    eng.add_to_stream("<s1>", ("<http://t/ev>", "<http://t/p>", "<http://t/o>"), 1)
+    for entry in eng.windows.values():
+        entry.window.flush()
+    flat = [v for rows in got for r in rows for v in r]
+    assert any("ev" in v for v in flat)
+    assert not any("static" in v for v in flat)
