@@ -270,3 +270,51 @@ def test_retrieve_clause():
     cq = parse_combined_query("RETRIEVE SOME LATENT <http://e/s1> <http://e/s2>")
     assert cq.retrieve.mode == "SOME"
     assert len(cq.retrieve.streams) == 2
+
+
+# ---- grammar-surface parity probes (ref parser_test.rs) ----
+
+def _ok(q):
+    from kolibrie_amd.parsing.sparql import parse_combined_query
+    return parse_combined_query(q)
+
+
+def test_a_syntax_in_select_and_rules():
+    _ok('SELECT ?x WHERE { ?x a <http://e/Person> }')
+    cq = _ok('RULE :t :- CONSTRUCT { ?x a <http://e/B> } '
+             'WHERE { ?x a <http://e/A> } '
+             'SELECT ?x WHERE { ?x a <http://e/B> }')
+    assert cq.rules
+
+
+def test_case_insensitive_keywords_nested_graph_union():
+    cq = _ok('select ?x where { { ?x a <http://e/A> } union '
+             '{ graph <http://g> { ?x <http://e/p> ?y } } }')
+    assert cq.select is not None
+
+
+def test_comments_inside_group():
+    _ok('SELECT ?x WHERE { ?x <http://e/p> ?y . # a comment\n'
+        ' FILTER(?y > 3) }')
+
+
+def test_literal_escapes_and_language_tags():
+    _ok('SELECT ?x WHERE { ?x <http://e/label> "hi"@en }')
+    _ok('SELECT ?x WHERE { ?x <http://e/p> "line\\nbreak \\"q\\"" }')
+
+
+def test_prefixed_names_with_dots():
+    _ok('PREFIX ex: <http://e/> SELECT ?x WHERE { ?x ex:foo.bar ?y }')
+
+
+def test_prob_annotation_variants():
+    base = (' :- CONSTRUCT { ?x <http://e/q> ?y } '
+            'WHERE { ?x <http://e/p> ?y } SELECT ?x WHERE { ?x <http://e/q> ?y }')
+    for ann in ("PROB(combination=min)", "PROB(provenance=topk, k=4)",
+                "PROB(provenance=wmc)",
+                "PROB(provenance=hybrid, threshold=0.7, confidence=0.9)"):
+        cq = _ok(f'RULE :r {ann}{base}')
+        assert cq.rules[0].prob is not None
+    # no annotation still works
+    cq = _ok('RULE :r' + base)
+    assert cq.rules[0].prob is None
