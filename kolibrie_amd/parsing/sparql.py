@@ -878,18 +878,83 @@ class Parser:
             negated=negated, prob=prob, ml_predict=ml_predict,
         )
 
+    _HYBRID_PROB_KEYS = {"band_epsilon", "marginal_floor", "k_initial",
+                         "k_max", "k_growth", "topk_budget_ms",
+                         "sdd_budget_ms", "node_budget"}
+
+    def _parse_threshold_value(self, pos) -> tuple:
+        """float -> (v, "Explicit"); `auto:cost(fp=a,fn=b)` -> (fp/(fp+fn),
+        "CostRatio"); everything else (incl. auto:quantile) rejected
+        (ref parser.rs:2896 parse_hybrid_threshold)."""
+        t = self.next().text
+        if not t.lower().startswith("auto"):
+            import math
+            try:
+                v = float(t)
+            except ValueError:
+                raise ParseError(f"bad threshold {t!r}", self.text, pos)
+            if not math.isfinite(v):
+                raise ParseError(f"bad threshold {t!r}", self.text, pos)
+            return v, "Explicit"
+        spec = t
+        while self.at(":"):
+            spec += self.next().text + self.next().text
+        if not self.at("("):
+            raise ParseError(f"bad hybrid threshold {spec!r}", self.text, pos)
+        self.next()
+        parts = []
+        while not self.at(")"):
+            parts.append(self.next().text)
+        self.expect(")")
+        body = "".join(parts)
+        if not spec.lower().endswith("cost"):
+            raise ParseError(f"bad hybrid threshold policy {spec!r}",
+                             self.text, pos)
+        costs = {}
+        import math
+        for pair in body.split(","):
+            if "=" not in pair:
+                raise ParseError("bad auto:cost spec", self.text, pos)
+            k, v = pair.split("=", 1)
+            k = k.strip().lower()
+            if k not in ("fp", "fn") or k in costs:
+                raise ParseError(f"bad auto:cost key {k!r}", self.text, pos)
+            try:
+                fv = float(v)
+            except ValueError:
+                raise ParseError(f"bad auto:cost value {v!r}", self.text, pos)
+            if not math.isfinite(fv) or fv < 0.0:
+                raise ParseError(f"bad auto:cost value {v!r}", self.text, pos)
+            costs[k] = fv
+        if set(costs) != {"fp", "fn"} or costs["fp"] + costs["fn"] <= 0.0:
+            raise ParseError("auto:cost needs fp and fn with fp+fn > 0",
+                             self.text, pos)
+        return costs["fp"] / (costs["fp"] + costs["fn"]), "CostRatio"
+
     def parse_prob_annotation(self) -> ProbAnnotation:
-        self.expect_kw("PROB")
+        """PROB(key=value, ...).  Hybrid provenance is strictly validated
+        (ref parser.rs:2931-3090: allowed keys only, threshold required,
+        no duplicates, no confidence, auto:cost thresholds)."""
+        tok = self.expect_kw("PROB")
         self.expect("(")
         ann = ProbAnnotation()
+        seen = set()
+        duplicate = False
         while not self.at(")"):
             key = self.next().text.lower()
             self.expect("=")
+            if key in seen:
+                duplicate = True
+            seen.add(key)
+            if key == "threshold":
+                ann.threshold, policy = self._parse_threshold_value(tok.pos)
+                ann.extra["threshold_policy"] = policy
+                if self.at(","):
+                    self.next()
+                continue
             val = self.next().text
             if key in ("provenance", "combination"):
                 ann.provenance = val.lower()
-            elif key == "threshold":
-                ann.threshold = float(val)
             elif key == "confidence":
                 ann.confidence = float(val)
             else:
@@ -897,6 +962,19 @@ class Parser:
             if self.at(","):
                 self.next()
         self.expect(")")
+        if ann.provenance == "hybrid":
+            unknown = (set(ann.extra) - self._HYBRID_PROB_KEYS
+                       - {"threshold_policy"})
+            if (duplicate or unknown or ann.confidence is not None
+                    or ann.threshold is None):
+                raise ParseError(
+                    "invalid hybrid PROB annotation (threshold required; "
+                    f"unknown/duplicate keys: {sorted(unknown)})",
+                    self.text, tok.pos)
+            if (ann.extra.get("threshold_policy") == "Explicit"
+                    and not (0.0 <= ann.threshold <= 1.0)):
+                raise ParseError("hybrid threshold must be in [0,1]",
+                                 self.text, tok.pos)
         return ann
 
     def parse_ml_predict(self) -> dict:

@@ -141,8 +141,19 @@ class RSPBuilder:
                     cfg = HybridConfig()
                     if pa.threshold is not None:
                         cfg.threshold = pa.threshold
-                    if pa.confidence is not None:
-                        cfg.confidence = pa.confidence
+                    cfg.policy = pa.extra.get("threshold_policy", "Explicit")
+                    if "band_epsilon" in pa.extra:
+                        cfg.band_epsilon = float(pa.extra["band_epsilon"])
+                    for k, attr in (("k_initial", "k_initial"),
+                                    ("k_max", "k_max"),
+                                    ("k_growth", "k_growth"),
+                                    ("node_budget", "sdd_node_cap")):
+                        if k in pa.extra:
+                            setattr(cfg, attr, int(pa.extra[k]))
+                    if "topk_budget_ms" in pa.extra:
+                        cfg.topk_budget_ms = float(pa.extra["topk_budget_ms"])
+                    if "sdd_budget_ms" in pa.extra:
+                        cfg.sdd_budget_ms = float(pa.extra["sdd_budget_ms"])
                     cfg.validate()
                     engine.store.hybrid_config = cfg
                     break

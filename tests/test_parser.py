@@ -312,9 +312,46 @@ def test_prob_annotation_variants():
             'WHERE { ?x <http://e/p> ?y } SELECT ?x WHERE { ?x <http://e/q> ?y }')
     for ann in ("PROB(combination=min)", "PROB(provenance=topk, k=4)",
                 "PROB(provenance=wmc)",
-                "PROB(provenance=hybrid, threshold=0.7, confidence=0.9)"):
+                "PROB(provenance=hybrid, threshold=0.7, band_epsilon=0.05)"):
         cq = _ok(f'RULE :r {ann}{base}')
         assert cq.rules[0].prob is not None
     # no annotation still works
     cq = _ok('RULE :r' + base)
     assert cq.rules[0].prob is None
+
+
+def test_hybrid_prob_validation():
+    """ref parser_test.rs hybrid_probability_annotation_rejects_*: hybrid
+    requires a threshold, rejects unknown/duplicate keys + confidence,
+    rejects invalid auto policies; auto:cost(fp,fn) -> CostRatio."""
+    import pytest
+    from kolibrie_amd.parsing.sparql import ParseError
+    base = (' :- CONSTRUCT {{ ?x <http://e/q> ?y }} '
+            'WHERE {{ ?x <http://e/p> ?y }} '
+            'SELECT ?x WHERE {{ ?x <http://e/q> ?y }}')
+
+    def rule(ann):
+        return _ok(('RULE :r ' + ann + base).format())
+
+    cq = rule('PROB(provenance=hybrid, threshold=auto:cost(fp=1,fn=3))')
+    pa = cq.rules[0].prob
+    assert abs(pa.threshold - 0.25) < 1e-9
+    assert pa.extra["threshold_policy"] == "CostRatio"
+
+    invalid = [
+        'PROB(provenance=hybrid)',                                # no threshold
+        'PROB(provenance=hybrid, threshold=0.7, mystery=1)',      # unknown key
+        'PROB(provenance=hybrid, threshold=0.7, confidence=0.9)', # confidence
+        'PROB(provenance=hybrid, threshold=auto:quantile(0.9))',  # quantile
+        'PROB(provenance=hybrid, threshold=auto:cost(fp=1))',     # missing fn
+        'PROB(provenance=hybrid, threshold=auto:cost(fp=0,fn=0))',# zero total
+        'PROB(provenance=hybrid, threshold=auto:cost(fp=-1,fn=2))',
+        'PROB(provenance=hybrid, threshold=1.7)',                 # out of range
+        'PROB(provenance=hybrid, threshold=0.4, threshold=0.6)',  # duplicate
+    ]
+    for ann in invalid:
+        with pytest.raises(ParseError):
+            rule(ann)
+    # non-hybrid still tolerates extra keys
+    cq = rule('PROB(provenance=topk, k=4, mystery=2)')
+    assert cq.rules[0].prob.provenance == "topk"
