@@ -587,17 +587,29 @@ __global__ void chain_tile_bounds(const int64_t* __restrict__ seed_key12,
   }
 }
 
+// LDS staging: a tile's src-0 window (avg ~|region|/n_tiles rows) is probed
+// by all 256 threads of the block — stage it through LDS once and search
+// there instead of hammering the same L1/L2 lines (guide: LDS-staged
+// probe structures).  16 KB leaves occupancy at 10 blocks/CU.
+constexpr int kChainLds = 2048;
+
 __global__ void chain_count_kernel(const int64_t* __restrict__ seed_key12,
                                    const int32_t* __restrict__ seed_z,
                                    int64_t m, ChainHops hops,
                                    const int64_t* __restrict__ win,
                                    unsigned long long* __restrict__ total) {
+  __shared__ int64_t lds[kChainLds];
   unsigned long long acc = 0;
-  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < m;
-       i += (int64_t)gridDim.x * blockDim.x) {
-    int64_t t = i / kTile;
-    int64_t b_comp = seed_key12[i] & 0xFFFFFFFFLL;
-    int64_t z_comp = static_cast<int64_t>(seed_z[i]) & 0xFFFFFFFFLL;
+  // block-uniform tile iteration (every thread of the block is in the same
+  // tile) so the cooperative LDS loads can barrier safely
+  for (int64_t t = blockIdx.x; t * kTile < m; t += gridDim.x) {
+    int64_t i = t * kTile + threadIdx.x;
+    bool active = i < m;
+    int64_t b_comp = 0, z_comp = 0;
+    if (active) {
+      b_comp = seed_key12[i] & 0xFFFFFFFFLL;
+      z_comp = static_cast<int64_t>(seed_z[i]) & 0xFFFFFFFFLL;
+    }
     unsigned long long prod = 1;
     // no early exit: the hop searches are independent dependent-load
     // chains — letting them all issue gives the scheduler ILP to hide
@@ -605,6 +617,7 @@ __global__ void chain_count_kernel(const int64_t* __restrict__ seed_key12,
     for (int h = 0; h < hops.k; ++h) {
       int64_t comp = hops.src[h] == 0 ? b_comp : z_comp;
       if (hops.table[h] != nullptr) {
+        if (!active) continue;
         // count-table hop: one 8-byte L2 load (vs log2(n) lines)
         uint32_t v = static_cast<uint32_t>(comp);
         uint32_t mask = static_cast<uint32_t>(hops.tmask[h]);
@@ -626,6 +639,23 @@ __global__ void chain_count_kernel(const int64_t* __restrict__ seed_key12,
       int64_t wlo = win[(t * hops.k + h) * 2];
       int64_t wspan = win[(t * hops.k + h) * 2 + 1] - wlo;
       const int64_t* base = hops.key12[h] + wlo;
+      if (hops.src[h] == 0 && wspan <= kChainLds) {
+        // cooperative stage + barrier: search LDS, not HBM/L2
+        for (int64_t j = threadIdx.x; j < wspan; j += blockDim.x)
+          lds[j] = base[j];
+        __syncthreads();
+        if (active) {
+          int64_t lo = lower_bound_i64(lds, wspan, key);
+          int64_t hi = lo;
+          while (hi < wspan && hi - lo < 4 && lds[hi] == key) ++hi;
+          if (hi - lo == 4 && hi < wspan && lds[hi] == key)
+            hi = lo + upper_bound_i64(lds + lo, wspan - lo, key);
+          prod *= static_cast<unsigned long long>(hi - lo);
+        }
+        __syncthreads();  // before the next hop reuses the buffer
+        continue;
+      }
+      if (!active) continue;
       int64_t lo = lower_bound_i64(base, wspan, key);
       // match runs are tiny (one object per subject in typical star data):
       // walk forward a few cache-hot slots instead of paying a second full
@@ -636,7 +666,7 @@ __global__ void chain_count_kernel(const int64_t* __restrict__ seed_key12,
         hi = lo + upper_bound_i64(base + lo, wspan - lo, key);
       prod *= static_cast<unsigned long long>(hi - lo);
     }
-    acc += prod;
+    if (active) acc += prod;
   }
   // wave reduction then one device-scope atomic per wave (guide G12)
   for (int off = 32; off > 0; off >>= 1)
