@@ -41,6 +41,10 @@ class DeviceStreamWindow:
         self.app_time = 0
         self._next_close = slide  # first window [slide-width, slide)
         self.callback: Optional[Callable[[ColumnContent], None]] = None
+        # host-sourced batches upload on a side HIP stream so H2D copies
+        # overlap window compute on the default stream (ingest pipeline)
+        self._copy_stream = (torch.cuda.Stream(self.device)
+                             if self.device.type == "cuda" else None)
 
     def register_callback(self, fn: Callable[[ColumnContent], None]):
         self.callback = fn
@@ -61,10 +65,26 @@ class DeviceStreamWindow:
     def add_batch(self, s, p, o, ts):
         """Append a time-ordered event batch; fire every window whose close
         falls inside (app_time, max_ts]."""
-        s = s.to(self.device).to(torch.int32)
-        p = p.to(self.device).to(torch.int32)
-        o = o.to(self.device).to(torch.int32)
-        ts = ts.to(self.device).to(torch.int64)
+        if self._copy_stream is not None and s.device.type == "cpu":
+            with torch.cuda.stream(self._copy_stream):
+                s = s.to(torch.int32).pin_memory().to(self.device,
+                                                      non_blocking=True)
+                p = p.to(torch.int32).pin_memory().to(self.device,
+                                                      non_blocking=True)
+                o = o.to(torch.int32).pin_memory().to(self.device,
+                                                      non_blocking=True)
+                ts = ts.to(torch.int64).pin_memory().to(self.device,
+                                                        non_blocking=True)
+            ev = torch.cuda.Event()
+            ev.record(self._copy_stream)
+            # the compute stream consumes the batch only after the async
+            # copies land; no host-side sync
+            torch.cuda.current_stream(self.device).wait_event(ev)
+        else:
+            s = s.to(self.device).to(torch.int32)
+            p = p.to(self.device).to(torch.int32)
+            o = o.to(self.device).to(torch.int32)
+            ts = ts.to(self.device).to(torch.int64)
         self._bufs.append((s, p, o, ts))
         if ts.numel() == 0:
             return

@@ -301,3 +301,30 @@ def test_chain_count_e2e_flagship_shape():
         assert r1 == r2
         counts[dev] = r1[0][0]
     assert counts["cpu"] == counts["cuda:0"]
+
+
+@requires_gpu
+def test_ring_side_stream_host_ingest():
+    """Host-sourced event batches upload via the side copy stream and
+    produce the same firings as device-sourced batches."""
+    from kolibrie_amd.rsp.ring import DeviceStreamWindow
+    fired = {}
+
+    def mk(device):
+        w = DeviceStreamWindow(width=10, slide=10, device=device)
+        out = []
+        w.register_callback(lambda c: out.append((c.open, c.close, c.n)))
+        return w, out
+
+    w_host, out_host = mk("cuda:0")
+    w_dev, out_dev = mk("cuda:0")
+    for batch in range(4):
+        base = batch * 5
+        s = torch.arange(5, dtype=torch.int32) + base
+        p = torch.ones(5, dtype=torch.int32)
+        o = torch.ones(5, dtype=torch.int32)
+        ts = torch.arange(5, dtype=torch.int64) + base
+        w_host.add_batch(s, p, o, ts)                       # cpu -> side stream
+        w_dev.add_batch(s.cuda(), p.cuda(), o.cuda(), ts.cuda())
+    torch.cuda.synchronize()
+    assert out_host == out_dev and len(out_host) >= 1
