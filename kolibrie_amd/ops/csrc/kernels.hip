@@ -353,6 +353,63 @@ __global__ void hj_build(KeyCols build, int64_t r, uint32_t mask,
   }
 }
 
+// ---- LDS-staged variant for SMALL build sides -----------------------------
+// Build sides up to 8192 rows fit a per-block chained table entirely in LDS
+// (32 KB heads + 32 KB next = 64 KB -> 2 blocks/CU, 8 waves).  Every block
+// rebuilds the table from the global build columns (<= 8192 LDS atomics,
+// amortized over its share of probes) and the probe chain walk then costs
+// LDS latency instead of L2 round trips.  This is the selective-query /
+// broadcast-join shape: tiny build side, large probe side.
+constexpr int kHjLdsSlots = 8192;   // heads, power of 2
+constexpr int kHjLdsRows = 8192;    // max build rows
+
+__device__ __forceinline__ void hj_lds_build(KeyCols build, int64_t r,
+                                             int32_t* heads, int32_t* next) {
+  for (int j = threadIdx.x; j < kHjLdsSlots; j += blockDim.x) heads[j] = -1;
+  __syncthreads();
+  for (int64_t i = threadIdx.x; i < r; i += blockDim.x) {
+    uint32_t h = static_cast<uint32_t>(mix_hash(build, i)) & (kHjLdsSlots - 1);
+    next[i] = atomicExch(&heads[h], static_cast<int32_t>(i));
+  }
+  __syncthreads();
+}
+
+__global__ void hj_count_lds(KeyCols probe, int64_t l, KeyCols build,
+                             int64_t r, int32_t* __restrict__ cnt) {
+  __shared__ int32_t heads[kHjLdsSlots];
+  __shared__ int32_t next[kHjLdsRows];
+  hj_lds_build(build, r, heads, next);
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < l;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint32_t h = static_cast<uint32_t>(mix_hash(probe, i)) & (kHjLdsSlots - 1);
+    int32_t c = 0;
+    for (int32_t b = heads[h]; b >= 0; b = next[b])
+      if (keys_equal(probe, i, build, b)) ++c;
+    cnt[i] = c;
+  }
+}
+
+__global__ void hj_emit_lds(KeyCols probe, int64_t l, KeyCols build,
+                            int64_t r, const int64_t* __restrict__ offs,
+                            int64_t* __restrict__ li_out,
+                            int64_t* __restrict__ ri_out) {
+  __shared__ int32_t heads[kHjLdsSlots];
+  __shared__ int32_t next[kHjLdsRows];
+  hj_lds_build(build, r, heads, next);
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < l;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint32_t h = static_cast<uint32_t>(mix_hash(probe, i)) & (kHjLdsSlots - 1);
+    int64_t base = offs[i];
+    for (int32_t b = heads[h]; b >= 0; b = next[b]) {
+      if (keys_equal(probe, i, build, b)) {
+        li_out[base] = i;
+        ri_out[base] = b;
+        ++base;
+      }
+    }
+  }
+}
+
 __global__ void hj_count(KeyCols probe, int64_t l, KeyCols build,
                          const int32_t* __restrict__ heads,
                          const int32_t* __restrict__ next, uint32_t mask,
@@ -985,17 +1042,23 @@ at::Tensor hash_join_counts(std::vector<at::Tensor> left_cols,
     probe.c[j] = left_cols[j].data_ptr<int32_t>();
     build.c[j] = right_cols[j].data_ptr<int32_t>();
   }
+  auto stream = cur_stream();
+  auto cnt = at::empty({l}, opts_int);
+  if (r <= kHjLdsRows) {
+    hipLaunchKernelGGL(hj_count_lds, dim3(grid_for(l)), dim3(kBlock), 0,
+                       stream, probe, l, build, r, cnt.data_ptr<int32_t>());
+    HIP_OK(hipGetLastError());
+    return cnt;
+  }
   uint64_t h = 1;
   while (h < static_cast<uint64_t>(2 * r)) h <<= 1;
   uint32_t mask = static_cast<uint32_t>(h - 1);
   auto heads = at::full({static_cast<int64_t>(h)}, -1, opts_int);
   auto next = at::empty({r}, opts_int);
-  auto stream = cur_stream();
   hipLaunchKernelGGL(hj_build, dim3(grid_for(r)), dim3(kBlock), 0, stream,
                      build, r, mask, heads.data_ptr<int32_t>(),
                      next.data_ptr<int32_t>());
   HIP_OK(hipGetLastError());
-  auto cnt = at::empty({l}, opts_int);
   hipLaunchKernelGGL(hj_count, dim3(grid_for(l)), dim3(kBlock), 0, stream,
                      probe, l, build, heads.data_ptr<int32_t>(),
                      next.data_ptr<int32_t>(), mask, cnt.data_ptr<int32_t>());
@@ -1023,13 +1086,36 @@ std::vector<at::Tensor> hash_join(std::vector<at::Tensor> left_cols,
     probe.c[j] = left_cols[j].data_ptr<int32_t>();
     build.c[j] = right_cols[j].data_ptr<int32_t>();
   }
+  auto stream = cur_stream();
+  bool lds = r <= kHjLdsRows;
+  at::Tensor heads, next;
+  uint32_t mask = 0;
+  if (lds) {
+    // small build side: each block stages the chained table in LDS
+    auto cnt = at::empty({l}, opts_int);
+    hipLaunchKernelGGL(hj_count_lds, dim3(grid_for(l)), dim3(kBlock), 0,
+                       stream, probe, l, build, r, cnt.data_ptr<int32_t>());
+    HIP_OK(hipGetLastError());
+    auto offs = at::cumsum(cnt, 0, at::kLong);
+    int64_t total = offs[-1].item<int64_t>();
+    auto offs_excl = offs - cnt.to(at::kLong);
+    auto li = at::empty({total}, opts_long);
+    auto ri = at::empty({total}, opts_long);
+    if (total > 0) {
+      hipLaunchKernelGGL(hj_emit_lds, dim3(grid_for(l)), dim3(kBlock), 0,
+                         stream, probe, l, build, r,
+                         offs_excl.data_ptr<int64_t>(),
+                         li.data_ptr<int64_t>(), ri.data_ptr<int64_t>());
+      HIP_OK(hipGetLastError());
+    }
+    return {li, ri};
+  }
   // table size: next pow2 >= 2r
   uint64_t h = 1;
   while (h < static_cast<uint64_t>(2 * r)) h <<= 1;
-  uint32_t mask = static_cast<uint32_t>(h - 1);
-  auto heads = at::full({static_cast<int64_t>(h)}, -1, opts_int);
-  auto next = at::empty({r}, opts_int);
-  auto stream = cur_stream();
+  mask = static_cast<uint32_t>(h - 1);
+  heads = at::full({static_cast<int64_t>(h)}, -1, opts_int);
+  next = at::empty({r}, opts_int);
   hipLaunchKernelGGL(hj_build, dim3(grid_for(r)), dim3(kBlock), 0, stream,
                      build, r, mask, heads.data_ptr<int32_t>(),
                      next.data_ptr<int32_t>());

@@ -328,3 +328,29 @@ def test_ring_side_stream_host_ingest():
         w_dev.add_batch(s.cuda(), p.cuda(), o.cuda(), ts.cuda())
     torch.cuda.synchronize()
     assert out_host == out_dev and len(out_host) >= 1
+
+
+@requires_gpu
+def test_hash_join_lds_small_build_matches_cpu():
+    """The LDS-staged small-build hash join (build <= 8192 rows) must agree
+    with the CPU oracle, including duplicate keys and a probe side large
+    enough that every block restages the table."""
+    from kolibrie_amd.engine.bindings import Bindings
+    from kolibrie_amd.engine.executor import join_bindings
+    torch.manual_seed(11)
+    nl, nr = 300_000, 5_000           # large probe, LDS-sized build
+    lk = torch.randint(0, 2000, (nl,), dtype=torch.int32)
+    lv = torch.randint(0, 100, (nl,), dtype=torch.int32)
+    rk = torch.randint(0, 2000, (nr,), dtype=torch.int32)
+    rv = torch.randint(0, 100, (nr,), dtype=torch.int32)
+
+    def join_on(device):
+        l = Bindings({"k": lk.to(device), "a": lv.to(device)}, nl, device)
+        r = Bindings({"k": rk.to(device), "b": rv.to(device)}, nr, device)
+        out = join_bindings(l, r)
+        rows = torch.stack([out.col("k").to(torch.int64),
+                            out.col("a").to(torch.int64),
+                            out.col("b").to(torch.int64)])
+        return sorted(map(tuple, rows.t().cpu().tolist()))
+
+    assert join_on("cpu") == join_on("cuda:0")
