@@ -39,3 +39,9 @@ def test_gen_data_roundtrip(tmp_path):
         PREFIX ds: <https://data.cityofchicago.org/resource/xzkq-xp2w/>
         SELECT (COUNT(*) AS ?c) WHERE { ?e ds:annual_salary ?s }""")
     assert rows == [["50"]]
+
+
+def test_fraud_detection_example():
+    from examples.fraud_detection import main
+    verdicts = main()
+    assert len(verdicts) == 10
