@@ -176,14 +176,9 @@ class ExecutionEngine:
         # regions are store-version-stable: cache them on the plan node
         cache = getattr(op, "_chain_cache", None)
         if cache is not None and cache[0] == self.db.store.version:
-            seed_key12, seed_z, hop_regions, native = cache[1:]
+            seed_key12, seed_z, hop_regions, native = cache[1:5]
             if native is not None:
-                return int(native.chain_count(
-                    seed_key12, seed_z,
-                    [r[0] for r in hop_regions],
-                    [r[1] for r in hop_regions],
-                    [r[2] for r in hop_regions],
-                    [r[3] for r in hop_regions]))
+                return int(native.chain_count(seed_key12, seed_z, *cache[5]))
             return self._chain_count_torch(seed_key12, seed_z, hop_regions)
         if self.ctx.view.default_graphs != [DEFAULT_GRAPH] \
                 or self.ctx.view.named_graphs not in (None, []):
@@ -240,15 +235,14 @@ class ExecutionEngine:
             (r[0].contiguous(), r[1], r[2],
              self._chain_hop_table(native, r[0], seed_key12.numel()))
             for r in hop_regions]
+        hop_args = ([r[0] for r in hop_regions],
+                    [r[1] for r in hop_regions],
+                    [r[2] for r in hop_regions],
+                    [r[3] for r in hop_regions])
         op._chain_cache = (self.db.store.version, seed_key12, seed_z,
-                           hop_regions, native)
+                           hop_regions, native, hop_args)
         if native is not None:
-            return int(native.chain_count(
-                seed_key12, seed_z,
-                [r[0] for r in hop_regions],
-                [r[1] for r in hop_regions],
-                [r[2] for r in hop_regions],
-                [r[3] for r in hop_regions]))
+            return int(native.chain_count(seed_key12, seed_z, *hop_args))
         return self._chain_count_torch(seed_key12, seed_z, hop_regions)
 
     # hop regions much smaller than the seed count pay log2(n) L2 lines per
