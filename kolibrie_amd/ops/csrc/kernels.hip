@@ -657,108 +657,73 @@ __global__ void chain_count_kernel(const int64_t* __restrict__ seed_key12,
                                    unsigned long long* __restrict__ total) {
   __shared__ int64_t lds[kChainLds];
   unsigned long long acc = 0;
-  // block-uniform PAIR-of-tiles iteration: each thread carries TWO seed
-  // rows (tiles 2b and 2b+1) so every hop issues two independent
-  // dependent-load chains — the scheduler overlaps their L2/LDS latency —
-  // and the barrier cost per row halves.  Adjacent tiles' sorted windows
-  // are contiguous, so the staged hop unions them into one LDS span.
-  constexpr int64_t kPair = 2 * kTile;
-  for (int64_t t0 = blockIdx.x * 2; t0 * kTile < m; t0 += (int64_t)gridDim.x * 2) {
-    int64_t i0 = t0 * kTile + threadIdx.x;
-    int64_t i1 = i0 + kTile;
-    bool a0 = i0 < m, a1 = i1 < m;
-    int64_t b0 = 0, z0 = 0, b1 = 0, z1 = 0;
-    if (a0) {
-      b0 = seed_key12[i0] & 0xFFFFFFFFLL;
-      z0 = static_cast<int64_t>(seed_z[i0]) & 0xFFFFFFFFLL;
+  // block-uniform tile iteration (every thread of the block is in the same
+  // tile) so the cooperative LDS loads can barrier safely
+  for (int64_t t = blockIdx.x; t * kTile < m; t += gridDim.x) {
+    int64_t i = t * kTile + threadIdx.x;
+    bool active = i < m;
+    int64_t b_comp = 0, z_comp = 0;
+    if (active) {
+      b_comp = seed_key12[i] & 0xFFFFFFFFLL;
+      z_comp = static_cast<int64_t>(seed_z[i]) & 0xFFFFFFFFLL;
     }
-    if (a1) {
-      b1 = seed_key12[i1] & 0xFFFFFFFFLL;
-      z1 = static_cast<int64_t>(seed_z[i1]) & 0xFFFFFFFFLL;
-    }
-    unsigned long long p0 = 1, p1 = 1;
+    unsigned long long prod = 1;
+    // no early exit: the hop searches are independent dependent-load
+    // chains — letting them all issue gives the scheduler ILP to hide
+    // L2 latency
     for (int h = 0; h < hops.k; ++h) {
-      int64_t c0 = hops.src[h] == 0 ? b0 : z0;
-      int64_t c1 = hops.src[h] == 0 ? b1 : z1;
+      int64_t comp = hops.src[h] == 0 ? b_comp : z_comp;
       if (hops.table[h] != nullptr) {
-        // count-table hop: two independent 8-byte L2 probes in flight
+        if (!active) continue;
+        // count-table hop: one 8-byte L2 load (vs log2(n) lines)
+        uint32_t v = static_cast<uint32_t>(comp);
         uint32_t mask = static_cast<uint32_t>(hops.tmask[h]);
-        uint32_t v0 = static_cast<uint32_t>(c0);
-        uint32_t v1 = static_cast<uint32_t>(c1);
-        uint32_t s0 = h32(v0) & mask, s1 = h32(v1) & mask;
-        unsigned long long n0 = 0, n1 = 0;
-        bool d0 = !a0, d1 = !a1;
-        while (!d0 || !d1) {
-          unsigned long long e0 = d0 ? 0 : hops.table[h][s0];
-          unsigned long long e1 = d1 ? 0 : hops.table[h][s1];
-          if (!d0) {
-            if (e0 == kTblEmpty) d0 = true;
-            else if (static_cast<uint32_t>(e0 >> 32) == v0) {
-              n0 = e0 & 0xFFFFFFFFull; d0 = true;
-            } else s0 = (s0 + 1) & mask;
+        uint32_t slot = h32(v) & mask;
+        unsigned long long cnt = 0;
+        for (;;) {
+          unsigned long long e = hops.table[h][slot];
+          if (e == kTblEmpty) break;
+          if (static_cast<uint32_t>(e >> 32) == v) {
+            cnt = e & 0xFFFFFFFFull;
+            break;
           }
-          if (!d1) {
-            if (e1 == kTblEmpty) d1 = true;
-            else if (static_cast<uint32_t>(e1 >> 32) == v1) {
-              n1 = e1 & 0xFFFFFFFFull; d1 = true;
-            } else s1 = (s1 + 1) & mask;
-          }
+          slot = (slot + 1) & mask;
         }
-        if (a0) p0 *= n0;
-        if (a1) p1 *= n1;
+        prod *= cnt;
         continue;
       }
-      int64_t k0 = (hops.const_hi[h] << 32) | c0;
-      int64_t k1 = (hops.const_hi[h] << 32) | c1;
-      int64_t t1 = a1 ? t0 + 1 : t0;
-      int64_t wlo = win[(t0 * hops.k + h) * 2];
-      int64_t whi = win[(t1 * hops.k + h) * 2 + 1];
-      int64_t wspan = whi - wlo;
+      int64_t key = (hops.const_hi[h] << 32) | comp;
+      int64_t wlo = win[(t * hops.k + h) * 2];
+      int64_t wspan = win[(t * hops.k + h) * 2 + 1] - wlo;
       const int64_t* base = hops.key12[h] + wlo;
       if (hops.src[h] == 0 && wspan <= kChainLds) {
-        // cooperative stage of the UNION window + barrier; search LDS
+        // cooperative stage + barrier: search LDS, not HBM/L2
         for (int64_t j = threadIdx.x; j < wspan; j += blockDim.x)
           lds[j] = base[j];
         __syncthreads();
-        if (a0) {
-          int64_t lo = lower_bound_i64(lds, wspan, k0);
+        if (active) {
+          int64_t lo = lower_bound_i64(lds, wspan, key);
           int64_t hi = lo;
-          while (hi < wspan && hi - lo < 4 && lds[hi] == k0) ++hi;
-          if (hi - lo == 4 && hi < wspan && lds[hi] == k0)
-            hi = lo + upper_bound_i64(lds + lo, wspan - lo, k0);
-          p0 *= static_cast<unsigned long long>(hi - lo);
-        }
-        if (a1) {
-          int64_t lo = lower_bound_i64(lds, wspan, k1);
-          int64_t hi = lo;
-          while (hi < wspan && hi - lo < 4 && lds[hi] == k1) ++hi;
-          if (hi - lo == 4 && hi < wspan && lds[hi] == k1)
-            hi = lo + upper_bound_i64(lds + lo, wspan - lo, k1);
-          p1 *= static_cast<unsigned long long>(hi - lo);
+          while (hi < wspan && hi - lo < 4 && lds[hi] == key) ++hi;
+          if (hi - lo == 4 && hi < wspan && lds[hi] == key)
+            hi = lo + upper_bound_i64(lds + lo, wspan - lo, key);
+          prod *= static_cast<unsigned long long>(hi - lo);
         }
         __syncthreads();  // before the next hop reuses the buffer
         continue;
       }
-      if (a0) {
-        int64_t lo = lower_bound_i64(base, wspan, k0);
-        int64_t hi = lo;
-        while (hi < wspan && hi - lo < 4 && base[hi] == k0) ++hi;
-        if (hi - lo == 4 && hi < wspan && base[hi] == k0)
-          hi = lo + upper_bound_i64(base + lo, wspan - lo, k0);
-        p0 *= static_cast<unsigned long long>(hi - lo);
-      }
-      if (a1) {
-        int64_t lo = lower_bound_i64(base, wspan, k1);
-        int64_t hi = lo;
-        while (hi < wspan && hi - lo < 4 && base[hi] == k1) ++hi;
-        if (hi - lo == 4 && hi < wspan && base[hi] == k1)
-          hi = lo + upper_bound_i64(base + lo, wspan - lo, k1);
-        p1 *= static_cast<unsigned long long>(hi - lo);
-      }
+      if (!active) continue;
+      int64_t lo = lower_bound_i64(base, wspan, key);
+      // match runs are tiny (one object per subject in typical star data):
+      // walk forward a few cache-hot slots instead of paying a second full
+      // log2(wspan) dependent-load chain; fall back for genuine skew
+      int64_t hi = lo;
+      while (hi < wspan && hi - lo < 4 && base[hi] == key) ++hi;
+      if (hi - lo == 4 && hi < wspan && base[hi] == key)
+        hi = lo + upper_bound_i64(base + lo, wspan - lo, key);
+      prod *= static_cast<unsigned long long>(hi - lo);
     }
-    if (a0) acc += p0;
-    if (a1) acc += p1;
-    (void)kPair;
+    if (active) acc += prod;
   }
   // wave reduction then one device-scope atomic per wave (guide G12)
   for (int off = 32; off > 0; off >>= 1)
