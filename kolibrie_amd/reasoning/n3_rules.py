@@ -47,7 +47,8 @@ def _parse_graph(text: str, prefixes: Dict[str, str], db) -> List[TriplePattern]
 
 
 def _split_rules(text: str):
-    """Yield (premise_text, conclusion_text) for each `{..} => {..}`."""
+    """Yield (premise_text, conclusion_text, rule_start, rule_end) for each
+    `{..} => {..}` (rule_end is just past the conclusion's brace)."""
     i = 0
     n = len(text)
     while i < n:
@@ -80,22 +81,32 @@ def _split_rules(text: str):
                     break
             j2 += 1
         conclusion = text[start2 + 1:j2]
-        yield premise, conclusion
+        yield premise, conclusion, start, j2 + 1
         i = j2 + 1
 
 
 def parse_n3_rules(text: str, db) -> List[Rule]:
-    """Plain N3 rules (ref parser_n3_logic.rs `{ p } => { c }`)."""
+    """Plain N3 rules (ref parser_n3_logic.rs `{ p } => { c }`).  A missing
+    final `.` is tolerated; leftover non-whitespace that is not a rule is
+    rejected (ref cross_window_tests.rs parser tests)."""
     prefixes: Dict[str, str] = {}
     for m in _PREFIX_RE.finditer(text):
         prefixes[m.group(1)] = m.group(2)
     body = _PREFIX_RE.sub("", text)
     rules: List[Rule] = []
-    for premise_t, conclusion_t in _split_rules(body):
+    consumed_to = 0
+    for premise_t, conclusion_t, start, end in _split_rules(body):
+        gap = body[consumed_to:start].strip()
+        if gap and gap.strip(".").strip():
+            raise ValueError(f"n3 rules: unexpected input {gap[:40]!r}")
         rules.append(Rule(
             premise=_parse_graph(premise_t, prefixes, db),
             conclusion=_parse_graph(conclusion_t, prefixes, db),
         ))
+        consumed_to = end
+    tail = body[consumed_to:].strip()
+    if tail and tail.strip(".").strip():
+        raise ValueError(f"n3 rules: leftover input {tail[:40]!r}")
     return rules
 
 

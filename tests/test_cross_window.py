@@ -129,3 +129,22 @@ def test_expiry_chain_propagation():
     for ts in (5, 15, 50, 101):
         assert naive_sds_plus(sds, rules, db, ts) == \
             incremental_sds_plus(sds, rules, db, ts), ts
+
+
+def test_n3_parser_shared_prefixes_missing_dot_and_leftover():
+    """ref cross_window_tests.rs parser trio: shared prefixes apply to all
+    rules, a missing final `.` is tolerated, leftover input is rejected."""
+    import pytest
+    from kolibrie_amd.reasoning.n3_rules import parse_n3_rules
+    db = SparqlDatabase()
+    rules = parse_n3_rules("""
+        @prefix ex: <http://e/> .
+        { ?x ex:p ?y } => { ?x ex:q ?y } .
+        { ?x ex:q ?y . ?y ex:q ?z } => { ?x ex:r ?z }
+    """, db)
+    assert len(rules) == 2
+    assert rules[1].premise[0].p.id == db.dictionary.encode("http://e/q")
+    with pytest.raises(ValueError):
+        parse_n3_rules("{ ?x <p> ?y } => { ?x <q> ?y } . garbage here", db)
+    with pytest.raises(ValueError):
+        parse_n3_rules("junk { ?x <p> ?y } => { ?x <q> ?y } .", db)
