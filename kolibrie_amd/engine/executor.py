@@ -20,8 +20,8 @@ import torch
 
 from ..plan.physical import (
     PBind, PBindJoin, PFilter, PHashJoin, PIndexScan, PInMemoryBuffer,
-    PMLPredict, PMinus, PNestedLoopJoin, PProjection, PStarJoin, PSubquery,
-    PTableScan, PUnion, PUnit, PValues, PhysicalOp,
+    PLeftJoin, PMLPredict, PMinus, PNestedLoopJoin, PProjection, PStarJoin,
+    PSubquery, PTableScan, PUnion, PUnit, PValues, PhysicalOp,
 )
 from ..storage.dataset import DEFAULT_GRAPH, GraphIndex
 from ..storage.terms import (
@@ -133,6 +133,12 @@ class ExecutionEngine:
             rows = self.execute(op.input, incoming)
             sub = self._exec_subquery(op.select)
             return join_bindings(rows, sub)
+        if isinstance(op, PLeftJoin):
+            left = self.execute(op.left, incoming)
+            if left.is_empty():
+                return left
+            right = self.execute(op.right, Bindings.unit(self.device))
+            return left_outer_join(left, right, needed)
         if isinstance(op, PMinus):
             left = self.execute(op.left, incoming)
             if left.is_empty():
@@ -742,6 +748,20 @@ def _compat_nlj(left: Bindings, right: Bindings, shared: Sequence[str],
         mask &= (lv == rv) | (lv == UNBOUND) | (rv == UNBOUND)
     li, ri = li[mask], ri[mask]
     return _merge_pairs(left, right, li, ri, shared, needed)
+
+
+def left_outer_join(left: Bindings, right: Bindings,
+                    needed=None) -> Bindings:
+    """OPTIONAL: every left row survives; matched rows extend with right
+    columns, unmatched rows pad the right-only columns UNBOUND (SPARQL
+    left outer join; engine extension beyond the reference subset)."""
+    if right.is_empty():
+        return left
+    inner = join_bindings(left, right, needed)
+    unmatched = anti_join(left, right)
+    if unmatched.is_empty():
+        return inner
+    return Bindings.concat([inner, unmatched], left.device)
 
 
 def anti_join(left: Bindings, right: Bindings) -> Bindings:

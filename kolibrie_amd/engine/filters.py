@@ -199,6 +199,13 @@ def _eval_bool(e: Expr, b: Bindings, db) -> torch.Tensor:
                 return torch.zeros(b.n, dtype=torch.bool, device=dev)
             # quoted IDs have bit 31 set => negative int32 (and != UNBOUND)
             return (ids < 0) & (ids != UNBOUND)
+        if e.name == "BOUND" and e.args and isinstance(e.args[0], EVar):
+            # BOUND(?v) — matches the K5 bytecode OP_BOUND semantics
+            # (engine extension used with OPTIONAL; UNDEF-aware)
+            v = e.args[0].name
+            if not b.has(v):
+                return torch.zeros(b.n, dtype=torch.bool, device=dev)
+            return b.col(v) != UNBOUND
         # other functions are false in FILTER context (ref types.rs:444-455)
         return torch.zeros(b.n, dtype=torch.bool, device=dev)
     if isinstance(e, EVar):

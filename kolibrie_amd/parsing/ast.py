@@ -149,6 +149,14 @@ class GMinus(GGP):
     right: GGP
 
 
+@dataclass
+class GOptional(GGP):
+    """OPTIONAL { ... } — left outer join (engine extension beyond the
+    reference's SPARQL subset)."""
+    left: GGP
+    right: GGP
+
+
 # ------------------------------------------------------------------ select ---
 @dataclass
 class Projection:

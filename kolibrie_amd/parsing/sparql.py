@@ -17,7 +17,8 @@ from typing import Dict, List, Optional, Tuple
 
 from .ast import (
     CombinedQuery, CombinedRule, EAnd, EArith, ECmp, EFunc, ELit, ENot, EOr,
-    EVar, Expr, GBgp, GBind, GFilter, GGP, GGraph, GJoin, GMinus, GSubQuery,
+    EVar, Expr, GBgp, GBind, GFilter, GGP, GGraph, GJoin, GMinus, GOptional,
+    GSubQuery,
     GUnion, GUnit, GValues, GWindowBlock, ModelDecl, NeuralRelationDecl,
     OrderCondition, ProbAnnotation, Projection, QuadData, RegisterClause,
     RetrieveClause, SelectQuery, SyncPolicy, TrainNeuralRelationDecl,
@@ -76,7 +77,7 @@ _TOK = re.compile(
 KEYWORDS = {
     "SELECT", "WHERE", "DISTINCT", "FROM", "NAMED", "GROUP", "ORDER", "BY",
     "ASC", "DESC", "LIMIT", "OFFSET", "PREFIX", "FILTER", "BIND", "VALUES",
-    "UNION", "GRAPH", "AS", "UNDEF", "INSERT", "DELETE", "DATA", "CLEAR",
+    "UNION", "GRAPH", "AS", "UNDEF", "OPTIONAL", "INSERT", "DELETE", "DATA", "CLEAR",
     "CREATE", "DROP", "SILENT", "ALL", "DEFAULT", "WINDOW", "REGISTER",
     "RSTREAM", "ISTREAM", "DSTREAM", "RANGE", "TUMBLING", "SLIDING", "STEP",
     "REPORT", "TICK", "ON", "STREAM", "WITH", "POLICY", "RULE", "CONSTRUCT",
@@ -384,6 +385,10 @@ class Parser:
                 self.next()
                 inner = self.parse_group()
                 current = GMinus(current, inner)
+            elif self.at_kw("OPTIONAL"):
+                self.next()
+                inner = self.parse_group()
+                current = GOptional(current, inner)
             elif t.text == ".":
                 self.next()
             else:

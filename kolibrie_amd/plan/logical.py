@@ -89,6 +89,13 @@ class LMinus(LogicalOp):
     right: LogicalOp = field(default_factory=LUnit)
 
 
+@dataclass
+class LLeftJoin(LogicalOp):
+    """OPTIONAL left outer join."""
+    left: LogicalOp = field(default_factory=LUnit)
+    right: LogicalOp = field(default_factory=LUnit)
+
+
 def scans_of(op: LogicalOp) -> List[LScan]:
     out: List[LScan] = []
 

@@ -11,7 +11,8 @@ from typing import Dict, List, Optional
 
 from ..engine.filters import CompiledBind, CompiledExpr
 from ..parsing.ast import (
-    GBgp, GBind, GFilter, GGP, GGraph, GJoin, GMinus, GSubQuery, GUnion,
+    GBgp, GBind, GFilter, GGP, GGraph, GJoin, GMinus, GOptional, GSubQuery,
+    GUnion,
     GUnit, GValues, GWindowBlock, SelectQuery, TriplePatternAst,
 )
 from ..storage.terms import Constant, QuotedTriplePattern, TriplePattern, Variable
@@ -133,6 +134,12 @@ def build_logical_plan(
     if isinstance(g, GMinus):
         from .logical import LMinus
         return LMinus(
+            build_logical_plan(g.left, db, prefixes, scope),
+            build_logical_plan(g.right, db, prefixes, scope),
+        )
+    if isinstance(g, GOptional):
+        from .logical import LLeftJoin
+        return LLeftJoin(
             build_logical_plan(g.left, db, prefixes, scope),
             build_logical_plan(g.right, db, prefixes, scope),
         )

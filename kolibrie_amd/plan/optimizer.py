@@ -23,11 +23,13 @@ from typing import List, Optional, Set, Tuple
 from ..storage.terms import Constant, TriplePattern, Variable
 from .cost import CostEstimator
 from .logical import (
-    LBind, LJoin, LMLPredict, LMinus, LProjection, LScan, LSelection,
+    LBind, LJoin, LLeftJoin, LMLPredict, LMinus, LProjection, LScan,
+    LSelection,
     LSubquery, LUnion, LUnit, LValues, LogicalOp,
 )
 from .physical import (
-    PBind, PBindJoin, PFilter, PHashJoin, PIndexScan, PMLPredict, PMinus,
+    PBind, PBindJoin, PFilter, PHashJoin, PIndexScan, PLeftJoin,
+    PMLPredict, PMinus,
     PNestedLoopJoin, PProjection, PStarJoin, PSubquery, PTableScan, PUnion,
     PUnit, PValues, PhysicalOp,
 )
@@ -99,6 +101,10 @@ class Streamertail:
             lp, lr, lc = self._plan(op.left, bound)
             rp, rr, rc = self._plan(op.right, bound)
             return PMinus(lp, rp), lr, lc + rc
+        if isinstance(op, LLeftJoin):
+            lp, lr, lc = self._plan(op.left, bound)
+            rp, rr, rc = self._plan(op.right, bound)
+            return PLeftJoin(lp, rp), max(lr, lr * 1.0), lc + rc + lr + rr
         if isinstance(op, LProjection):
             ip, ir, ic = self._plan(op.input, bound)
             return PProjection(list(op.variables), ip), ir, ic
@@ -309,7 +315,7 @@ def _logical_out_vars(op: LogicalOp) -> Set[str]:
                 out.update(_pattern_vars(x.pattern))
                 if x.graph is not None and x.graph[0] == "var":
                     out.add(x.graph[1])
-            elif isinstance(x, (LJoin, LUnion, LMinus)):
+            elif isinstance(x, (LJoin, LUnion, LMinus, LLeftJoin)):
                 rec(x.left)
                 rec(x.right)
             elif isinstance(x, LBind):
@@ -345,7 +351,8 @@ def phys_out_vars(op: PhysicalOp) -> Set[str]:
         if op.graph is not None and op.graph[0] == "var":
             out.add(op.graph[1])
         return out
-    if isinstance(op, (PHashJoin, PBindJoin, PNestedLoopJoin, PUnion, PMinus)):
+    if isinstance(op, (PHashJoin, PBindJoin, PNestedLoopJoin, PUnion, PMinus,
+                       PLeftJoin)):
         return phys_out_vars(op.left) | phys_out_vars(op.right)
     if isinstance(op, PFilter):
         return phys_out_vars(op.input)
@@ -377,7 +384,7 @@ def annotate_needed(op: PhysicalOp, needed):
     (None = all).  Executors drop/skip unneeded columns — on 100M-row joins
     this eliminates whole gather passes over HBM."""
     from .physical import (
-        PBind, PBindJoin, PFilter, PHashJoin, PMLPredict, PMinus,
+        PBind, PBindJoin, PFilter, PHashJoin, PLeftJoin, PMLPredict, PMinus,
         PNestedLoopJoin, PStarJoin, PSubquery, PUnion, PValues,
     )
     op.needed = None if needed is None else frozenset(needed)
@@ -401,6 +408,15 @@ def annotate_needed(op: PhysicalOp, needed):
         shared = lv & rv
         annotate_needed(op.left, None if needed is None else set(needed) | shared)
         annotate_needed(op.right, None)
+        return
+    if isinstance(op, PLeftJoin):
+        lv = phys_out_vars(op.left)
+        rv = phys_out_vars(op.right)
+        shared = lv & rv
+        annotate_needed(op.left,
+                        None if needed is None else (set(needed) | shared) & lv)
+        annotate_needed(op.right,
+                        None if needed is None else (set(needed) | shared) & rv)
         return
     if isinstance(op, PUnion):
         annotate_needed(op.left, needed)

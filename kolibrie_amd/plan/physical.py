@@ -119,6 +119,13 @@ class PMinus(PhysicalOp):
 
 
 @dataclass
+class PLeftJoin(PhysicalOp):
+    """OPTIONAL: inner join plus unmatched left rows padded UNBOUND."""
+    left: PhysicalOp = field(default_factory=PUnit)
+    right: PhysicalOp = field(default_factory=PUnit)
+
+
+@dataclass
 class PInMemoryBuffer(PhysicalOp):
     """Materialized bindings injected by the RSP runtime."""
     bindings: object = None

@@ -281,3 +281,51 @@ def test_prefixed_query():
         PREFIX foaf: <http://xmlns.com/foaf/0.1/>
         SELECT ?n WHERE { ?x foaf:name ?n }""")
     assert rows == [["Alice"]]
+
+
+def test_optional_basic():
+    """OPTIONAL left outer join (engine extension beyond the reference's
+    SPARQL subset): unmatched left rows keep UNBOUND right vars."""
+    db = SparqlDatabase()
+    db.add_triple(f"<{EX}a>", f"<{EX}name>", '"Alice"')
+    db.add_triple(f"<{EX}b>", f"<{EX}name>", '"Bob"')
+    db.add_triple(f"<{EX}a>", f"<{EX}email>", '"a@x"')
+    rows = db.query(
+        f'SELECT ?n ?m WHERE {{ ?p <{EX}name> ?n . '
+        f'OPTIONAL {{ ?p <{EX}email> ?m }} }}')
+    assert sorted(rows) == [["Alice", "a@x"], ["Bob", ""]]
+
+
+def test_optional_filter_inside_group():
+    db = SparqlDatabase()
+    db.add_triple(f"<{EX}a>", f"<{EX}name>", '"Alice"')
+    db.add_triple(f"<{EX}b>", f"<{EX}name>", '"Bob"')
+    db.add_triple(f"<{EX}a>", f"<{EX}age>", '"17"')
+    db.add_triple(f"<{EX}b>", f"<{EX}age>", '"42"')
+    rows = db.query(
+        f'SELECT ?n ?a WHERE {{ ?p <{EX}name> ?n . '
+        f'OPTIONAL {{ ?p <{EX}age> ?a . FILTER(?a > 20) }} }}')
+    assert sorted(rows) == [["Alice", ""], ["Bob", "42"]]
+
+
+def test_optional_bound_filter_after():
+    """!BOUND over an OPTIONAL var selects the unmatched rows."""
+    db = SparqlDatabase()
+    db.add_triple(f"<{EX}a>", f"<{EX}name>", '"Alice"')
+    db.add_triple(f"<{EX}b>", f"<{EX}name>", '"Bob"')
+    db.add_triple(f"<{EX}a>", f"<{EX}email>", '"a@x"')
+    rows = db.query(
+        f'SELECT ?n WHERE {{ ?p <{EX}name> ?n . '
+        f'OPTIONAL {{ ?p <{EX}email> ?m }} FILTER(!BOUND(?m)) }}')
+    assert rows == [["Bob"]]
+
+
+def test_optional_multiple_matches_multiplicity():
+    db = SparqlDatabase()
+    db.add_triple(f"<{EX}a>", f"<{EX}name>", '"Alice"')
+    db.add_triple(f"<{EX}a>", f"<{EX}phone>", '"1"')
+    db.add_triple(f"<{EX}a>", f"<{EX}phone>", '"2"')
+    rows = db.query(
+        f'SELECT ?n ?ph WHERE {{ ?p <{EX}name> ?n . '
+        f'OPTIONAL {{ ?p <{EX}phone> ?ph }} }}')
+    assert sorted(rows) == [["Alice", "1"], ["Alice", "2"]]
