@@ -354,3 +354,27 @@ def test_hash_join_lds_small_build_matches_cpu():
         return sorted(map(tuple, rows.t().cpu().tolist()))
 
     assert join_on("cpu") == join_on("cuda:0")
+
+
+@requires_gpu
+def test_optional_bound_gpu_equals_cpu():
+    """OPTIONAL + BOUND filter: GPU bytecode path vs CPU oracle."""
+    from kolibrie_amd import SparqlDatabase
+    EX = "http://e/"
+    results = {}
+    for device in ["cpu", "cuda:0"]:
+        db = SparqlDatabase(device=device)
+        for i in range(2000):
+            db.add_triple(f"<{EX}s{i}>", f"<{EX}name>", f'"n{i}"')
+            if i % 3 == 0:
+                db.add_triple(f"<{EX}s{i}>", f"<{EX}email>", f'"e{i}"')
+        rows = db.query(
+            f'SELECT ?n ?m WHERE {{ ?s <{EX}name> ?n . '
+            f'OPTIONAL {{ ?s <{EX}email> ?m }} }}')
+        only_unbound = db.query(
+            f'SELECT ?n WHERE {{ ?s <{EX}name> ?n . '
+            f'OPTIONAL {{ ?s <{EX}email> ?m }} FILTER(!BOUND(?m)) }}')
+        results[device] = (sorted(map(tuple, rows)),
+                           sorted(map(tuple, only_unbound)))
+    assert results["cpu"] == results["cuda:0"]
+    assert len(results["cpu"][1]) == 2000 - len(range(0, 2000, 3))
