@@ -90,3 +90,56 @@ def test_reasoning_hierarchy():
     assert n >= 2
     ded = h.query_level(ReasoningLevel.DEDUCTIVE, "x", "needs", None)
     assert ded == [("x", "needs", "oxygen")]
+
+
+# ---- bc_* shape parity (ref reasoning_tests.rs bc_direct_fact ..
+# bc_no_spurious_negative) ----
+
+def _bc_reasoner():
+    r = Reasoner()
+    for s, p, o in [("a", "parent", "b"), ("b", "parent", "c"),
+                    ("c", "parent", "d"), ("x", "parent", "c")]:
+        r.add_abox_triple(s, p, o)
+    r.add_rule_text(
+        "RULE :anc :- CONSTRUCT { ?x <ancestor> ?y } WHERE { ?x <parent> ?y }")
+    r.add_rule_text(
+        "RULE :anc2 :- CONSTRUCT { ?x <ancestor> ?z } "
+        "WHERE { ?x <parent> ?y . ?y <ancestor> ?z }")
+    return r
+
+
+def test_bc_direct_fact():
+    r = _bc_reasoner()
+    assert r.backward_chaining(("a", "parent", "b")) != []
+
+
+def test_bc_3hop_transitive():
+    r = _bc_reasoner()
+    assert r.backward_chaining(("a", "ancestor", "d")) != []
+
+
+def test_bc_specific_target_and_no_result():
+    r = _bc_reasoner()
+    assert r.backward_chaining(("a", "ancestor", "c")) != []
+    assert r.backward_chaining(("d", "ancestor", "a")) == []
+    assert r.backward_chaining(("a", "unknownpred", "b")) == []
+
+
+def test_bc_full_scan_enumerates_all():
+    r = _bc_reasoner()
+    res = r.backward_chaining(("?s", "ancestor", "?o"))
+    dec = r.dictionary.decode
+    pairs = {(dec(b["s"]), dec(b["o"])) for b in res}
+    assert ("a", "d") in pairs and ("x", "d") in pairs
+    assert ("d", "a") not in pairs  # bc_no_spurious_negative
+
+
+def test_bc_sibling_join():
+    r = Reasoner()
+    r.add_abox_triple("p", "childOf", "f")
+    r.add_abox_triple("q", "childOf", "f")
+    r.add_rule_text(
+        "RULE :sib :- CONSTRUCT { ?a <sibling> ?b } "
+        "WHERE { ?a <childOf> ?f . ?b <childOf> ?f }")
+    res = r.backward_chaining(("p", "sibling", "?x"))
+    assert {r.dictionary.decode(b["x"]) for b in res} >= {"q"}
