@@ -132,6 +132,14 @@ class _Compiler:
                 self.push_id(e.args[0])
                 self.emit(OP_IS_TRIPLE)
                 return
+            if (e.name == "BOUND" and len(e.args) == 1
+                    and isinstance(e.args[0], EVar)):
+                # must match engine/filters.py _eval_bool BOUND semantics
+                if e.args[0].name not in self.var_cols:
+                    self.emit(OP_PUSH_FALSE)
+                    return
+                self.emit(OP_BOUND, self.var_cols[e.args[0].name])
+                return
             # other functions are false in FILTER context
             self.emit(OP_PUSH_FALSE)
             return
