@@ -1214,7 +1214,7 @@ static inline std::string unescape_nt(const char* s, size_t len) {
   return out;
 }
 
-py::tuple parse_ntriples_host(const std::string& text) {
+py::tuple parse_nlines_host(const std::string& text, bool quads) {
   std::unordered_map<std::string, int64_t> interned;
   std::vector<std::string> strings;
   std::vector<int64_t> ids;           // n*3 local ids
@@ -1250,11 +1250,12 @@ py::tuple parse_ntriples_host(const std::string& text) {
     for (size_t i = b; i + 1 < e; ++i)
       if (line[i] == '<' && line[i + 1] == '<') { star = true; break; }
     if (star) { fallback.push_back(static_cast<int64_t>(line_no - 1)); continue; }
-    int64_t term_ids[3];
+    int64_t term_ids[4];
+    const int max_terms = quads ? 4 : 3;
     int nt = 0;
     size_t i = b;
     bool ok = true;
-    while (i < e && nt < 3) {
+    while (i < e && nt < max_terms) {
       while (i < e && isspace(static_cast<unsigned char>(line[i]))) ++i;
       if (i >= e) break;
       char c = line[i];
@@ -1288,17 +1289,19 @@ py::tuple parse_ntriples_host(const std::string& text) {
         break;
       }
     }
-    if (!ok || nt != 3) {
+    if (!ok || nt < 3 || (!quads && nt != 3)) {
       fallback.push_back(static_cast<int64_t>(line_no - 1));
       continue;
     }
     ids.push_back(term_ids[0]);
     ids.push_back(term_ids[1]);
     ids.push_back(term_ids[2]);
+    if (quads) ids.push_back(nt == 4 ? term_ids[3] : -1);
     (void)save_pos;
   }
+  const int64_t width = quads ? 4 : 3;
   auto t = at::from_blob(ids.data(),
-                         {static_cast<int64_t>(ids.size() / 3), 3},
+                         {static_cast<int64_t>(ids.size()) / width, width},
                          at::kLong).clone();
   py::list pystrings;
   for (auto& s : strings) pystrings.append(py::bytes(s));
@@ -1307,9 +1310,21 @@ py::tuple parse_ntriples_host(const std::string& text) {
   return py::make_tuple(t, pystrings, pyfallback);
 }
 
+py::tuple parse_ntriples_host(const std::string& text) {
+  return parse_nlines_host(text, false);
+}
+
+// N-Quads: optional 4th graph term; absent -> -1 in the id row
+// (host bulk replacement for sparql_database.rs:1411 parse_nquads_and_add)
+py::tuple parse_nquads_host(const std::string& text) {
+  return parse_nlines_host(text, true);
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("parse_ntriples_host", &parse_ntriples_host,
         "bulk N-Triples parse -> (local-id triples, strings, fallback lines)");
+  m.def("parse_nquads_host", &parse_nquads_host,
+        "bulk N-Quads parse -> (local-id quads, strings, fallback lines)");
   m.doc() = "kolibrie_amd native CDNA4 kernels (gfx950)";
   m.def("probe_exact", &probe_exact,
         "K1 scan-probe, packed (a,b) exact keys -> (li, b, z)");

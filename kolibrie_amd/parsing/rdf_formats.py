@@ -136,8 +136,39 @@ def _parse_ntriples_lines(db, lines):
 
 
 def parse_nquads_into(db, text: str):
-    """N-Quads: optional 4th graph term (ref sparql_database.rs:1411-1463)."""
-    for line in text.split("\n"):
+    """N-Quads: optional 4th graph term (ref sparql_database.rs:1411-1463).
+
+    Bulk path mirrors parse_ntriples_into: the native tokenizer interns
+    terms locally (graph absent -> -1), the string table merges into the
+    dictionary and rows bulk-insert per graph."""
+    from ..ops import _native
+    if _native is not None and len(text) > 4096:
+        import numpy as np
+        local_ids, strings, fallback = _native.parse_nquads_host(text)
+        if local_ids.numel():
+            remap = np.empty(len(strings), dtype=np.int64)
+            enc = db.dictionary.encode
+            for i, raw in enumerate(strings):
+                remap[i] = enc(raw.decode("utf-8", "replace"))
+            arr = local_ids.numpy()
+            spo = remap[arr[:, :3]]
+            g = arr[:, 3]
+            gids = np.where(g < 0, 0, remap[np.maximum(g, 0)])
+            for gid in np.unique(gids):
+                m = gids == gid
+                db.store.insert_bulk(int(gid) & 0xFFFFFFFF,
+                                     spo[m, 0].astype(np.uint32),
+                                     spo[m, 1].astype(np.uint32),
+                                     spo[m, 2].astype(np.uint32))
+        if fallback:
+            lines = text.split("\n")
+            _parse_nquads_lines(db, (lines[i] for i in fallback))
+        return
+    _parse_nquads_lines(db, text.split("\n"))
+
+
+def _parse_nquads_lines(db, lines):
+    for line in lines:
         line = line.strip()
         if not line or line.startswith("#"):
             continue

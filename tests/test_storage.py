@@ -168,3 +168,25 @@ def test_native_ntriples_bulk_parse_matches_python():
     assert sorted(db_native.triples_as_strings()) == \
         sorted(db_py.triples_as_strings())
     assert db_native.triple_count() == db_py.triple_count()
+
+
+def test_native_nquads_bulk_parse_matches_python():
+    from kolibrie_amd.ops import _native
+    if _native is None:
+        pytest.skip("native extension not built")
+    lines = []
+    for i in range(400):
+        g = f"<http://e/g{i % 5}>" if i % 3 else ""
+        lines.append(f'<http://e/s{i}> <http://e/p{i % 7}> "v {i}" {g} .'.strip())
+        lines.append(f'<http://e/s{i}> <http://e/q> _:b{i} <http://e/g1> .')
+    lines.append('<< <http://e/a> <http://e/b> <http://e/c> >> <http://e/cert> "0.9" <http://e/g2> .')
+    text = "\n".join(lines)
+    db_native = SparqlDatabase()
+    db_native.parse_nquads(text)               # bulk path (> 4096 bytes)
+    db_py = SparqlDatabase()
+    from kolibrie_amd.parsing.rdf_formats import _parse_nquads_lines
+    _parse_nquads_lines(db_py, text.split("\n"))
+    assert db_native.generate_nquads().count("\n") == \
+        db_py.generate_nquads().count("\n")
+    assert sorted(db_native.generate_nquads().split("\n")) == \
+        sorted(db_py.generate_nquads().split("\n"))
