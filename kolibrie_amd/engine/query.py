@@ -54,6 +54,12 @@ def _top_needed(select: SelectQuery):
     return needed
 
 
+def _ask_result(select: SelectQuery, rows) -> Optional[List[List[str]]]:
+    if not getattr(select, "ask", False):
+        return None
+    return [["true" if rows.n > 0 else "false"]]
+
+
 def execute_select(select: SelectQuery, db, prefixes: Dict[str, str]
                    ) -> List[List[str]]:
     from ..plan.lower import build_logical_plan
@@ -71,6 +77,9 @@ def execute_select(select: SelectQuery, db, prefixes: Dict[str, str]
     annotate_needed(physical, _top_needed(select))
     ctx = ExecutionContext(db, view)
     rows = ExecutionEngine(ctx).execute(physical, Bindings.unit(db.device))
+    ask = _ask_result(select, rows)
+    if ask is not None:
+        return ask
     final = finalize_select_bindings(select, rows, db)
     return decode_rows(select, final, db)
 
@@ -142,6 +151,9 @@ def _count_star_fast(select: SelectQuery, rows: Bindings):
 def _run_prepared(pq: "PreparedQuery", db) -> List[List[str]]:
     ctx = ExecutionContext(db, pq.view)
     rows = ExecutionEngine(ctx).execute(pq.physical, Bindings.unit(db.device))
+    ask = _ask_result(pq.select, rows)
+    if ask is not None:
+        return ask
     fast = _count_star_fast(pq.select, rows)
     if fast is not None:
         return fast

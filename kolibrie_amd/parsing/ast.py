@@ -193,6 +193,7 @@ class SelectQuery:
     limit: Optional[int] = None
     offset: Optional[int] = None
     select_star: bool = False
+    ask: bool = False                  # ASK query (engine extension)
 
 
 # ------------------------------------------------------------------ update ---

@@ -83,7 +83,7 @@ KEYWORDS = {
     "REPORT", "TICK", "ON", "STREAM", "WITH", "POLICY", "RULE", "CONSTRUCT",
     "NOT", "MODEL", "NEURAL", "RELATION", "TRAIN", "USING", "RETRIEVE",
     "SOME", "EVERY", "LATENT", "ACTIVE", "PROB", "COUNT", "SUM", "AVG",
-    "MIN", "MAX", "MINUS",
+    "MIN", "MAX", "MINUS", "ASK",
 }
 
 AGGREGATES = {"COUNT", "SUM", "AVG", "MIN", "MAX"}
@@ -225,6 +225,12 @@ class Parser:
                 cq.retrieve = self.parse_retrieve()
             elif self.at_kw("SELECT"):
                 cq.select = self.parse_select_core()
+            elif self.at_kw("ASK"):
+                # ASK { pattern } — boolean query (engine extension)
+                self.next()
+                q = SelectQuery(select_star=True, ask=True, limit=1)
+                q.where = self.parse_group()
+                cq.select = q
             elif self.at_kw("INSERT", "DELETE", "CLEAR", "CREATE", "DROP"):
                 cq.updates.extend(self.parse_update_ops())
             elif self.at(";"):

@@ -329,3 +329,14 @@ def test_optional_multiple_matches_multiplicity():
         f'SELECT ?n ?ph WHERE {{ ?p <{EX}name> ?n . '
         f'OPTIONAL {{ ?p <{EX}phone> ?ph }} }}')
     assert sorted(rows) == [["Alice", "1"], ["Alice", "2"]]
+
+
+def test_ask_query():
+    """ASK { ... } boolean queries (engine extension)."""
+    db = SparqlDatabase()
+    db.add_triple(f"<{EX}a>", f"<{EX}p>", f"<{EX}b>")
+    assert db.query(f'ASK {{ <{EX}a> <{EX}p> ?x }}') == [["true"]]
+    assert db.query(f'ASK {{ <{EX}a> <{EX}q> ?x }}') == [["false"]]
+    assert db.query(f'ASK {{ ?s <{EX}p> ?o . FILTER(?s = ?o) }}') == [["false"]]
+    # cached second run
+    assert db.query(f'ASK {{ <{EX}a> <{EX}p> ?x }}') == [["true"]]
