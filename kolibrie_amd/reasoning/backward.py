@@ -7,7 +7,7 @@ that prove the goal.
 """
 from __future__ import annotations
 
-from typing import Dict, Iterator, List, Optional, Tuple, Union
+from typing import Dict, Iterator, List, Optional, Tuple
 
 from ..storage.terms import Constant, QuotedTriplePattern, TriplePattern, Variable
 from .rule import Rule

@@ -9,7 +9,7 @@ hierarchical_inference: per-level semi-naive then cross-level rules
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional
+from typing import Dict, List
 
 from .reasoner import Reasoner
 from .rule import Rule

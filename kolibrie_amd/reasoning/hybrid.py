@@ -20,7 +20,6 @@ HybridProbabilityResult per derived triple (:35).
 """
 from __future__ import annotations
 
-import itertools
 import time
 from dataclasses import dataclass, field
 from typing import Dict, List, Optional, Sequence, Set, Tuple

@@ -7,7 +7,7 @@ widths); plain `{ p } => { c }` rules for SimpleR2R.load_rules.
 from __future__ import annotations
 
 import re
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Optional
 
 from ..parsing.rdf_formats import tokenize
 from ..storage.terms import Constant, TriplePattern, Variable

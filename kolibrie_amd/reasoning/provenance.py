@@ -22,9 +22,8 @@ host-side.
 from __future__ import annotations
 
 import itertools
-import math
 from abc import ABC, abstractmethod
-from typing import Dict, FrozenSet, List, Optional, Sequence, Tuple
+from typing import Dict, FrozenSet, List, Optional, Tuple
 
 
 class Provenance(ABC):

@@ -5,7 +5,7 @@ WILDCARD keys, 8-case candidate dispatch)."""
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional, Set, Tuple
+from typing import Dict, List, Optional, Set
 
 from ..storage.terms import Constant, TriplePattern, Variable
 

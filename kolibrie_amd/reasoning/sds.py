@@ -16,7 +16,6 @@ unbounded base-store reasoning stays on the K6 columnar path.
 """
 from __future__ import annotations
 
-import math
 from dataclasses import dataclass
 from typing import Dict, List, Optional, Set, Tuple
 

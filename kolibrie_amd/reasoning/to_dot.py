@@ -2,7 +2,7 @@
 to_dot.rs, 115 LoC)."""
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 Triple = Tuple[int, int, int]
 

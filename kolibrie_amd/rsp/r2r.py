@@ -13,7 +13,7 @@ fixpoint directly over them.
 from __future__ import annotations
 
 from abc import ABC, abstractmethod
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Tuple
 
 from ..storage.database import SparqlDatabase
 from ..storage.dataset import DEFAULT_GRAPH

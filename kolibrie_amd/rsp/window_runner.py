@@ -2,7 +2,7 @@
 (ref: kolibrie/src/rsp/window_runner.rs:37-106)."""
 from __future__ import annotations
 
-from typing import Hashable, List, Optional
+from typing import Hashable, List
 
 from .s2r import CSPARQLWindow, ContentContainer
 
