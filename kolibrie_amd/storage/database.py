@@ -236,11 +236,15 @@ class SparqlDatabase:
             gid, self.encode_term_star(s), self.encode_term_star(p), self.encode_term_star(o)
         )
 
-    def delete_triple_parts(self, s: str, p: str, o: str):
+    def delete_triple_parts(self, s: str, p: str, o: str) -> bool:
+        """Returns True when the triple existed (ref sparql_database.rs
+        delete_triple_parts -> bool)."""
         ids = [self.dictionary.lookup(self.resolve_lexical(x)) for x in (s, p, o)]
         if any(i is None for i in ids):
-            return
+            return False
+        existed = self.store.graph_index(DEFAULT_GRAPH).contains(*ids)
         self.store.delete_quad(DEFAULT_GRAPH, *ids)  # type: ignore
+        return bool(existed)
 
     def triple_count(self) -> int:
         return self.store.triple_count()
@@ -325,6 +329,36 @@ class SparqlDatabase:
             self._stats = DatabaseStats.gather(self)
             self._stats_version = self.store.version
         return self._stats
+
+    def invalidate_stats_cache(self):
+        """Force stats re-gather on next use (ref sparql_database.rs
+        invalidate_stats_cache)."""
+        self._stats = None
+        self._stats_version = -1
+
+    def build_all_indexes(self):
+        """Reference-API parity (sparql_database.rs build_all_indexes):
+        here the four sorted permutations are maintained on every commit,
+        so this just flushes pending mutations."""
+        for g in list(self.store.graphs):
+            self.store.graph_index(g)
+
+    def query_builder(self):
+        """Fluent builder entry (the reference's `db.query()`; this class's
+        `query(sparql)` executes SPARQL directly instead)."""
+        from ..engine.query_builder import QueryBuilder
+        return QueryBuilder(self)
+
+    def decode_triple(self, triple) -> Optional[tuple]:
+        """Decode an (s,p,o) id triple to strings; None if any id unknown
+        (ref sparql_database.rs decode_triple)."""
+        out = []
+        for x in triple:
+            s = self.decode_term(int(x))
+            if s is None:
+                return None
+            out.append(s)
+        return tuple(out)
 
     # ----------------------------------------------------------- value column
     def value_column(self) -> torch.Tensor:

@@ -190,3 +190,24 @@ def test_native_nquads_bulk_parse_matches_python():
         db_py.generate_nquads().count("\n")
     assert sorted(db_native.generate_nquads().split("\n")) == \
         sorted(db_py.generate_nquads().split("\n"))
+
+
+def test_reference_api_surface():
+    """Reference README core-method parity (README.md:737-888)."""
+    db = SparqlDatabase()
+    db.add_triple_parts("<http://e/s>", "<http://e/p>", "<http://e/o>")
+    assert db.triple_count() == 1
+    assert db.delete_triple_parts("<http://e/s>", "<http://e/p>", "<http://e/o>")
+    assert db.triple_count() == 0
+    db.add_triple("<http://e/a>", "<http://e/p>", '"x"')
+    db.build_all_indexes()
+    st = db.get_or_build_stats()
+    db.invalidate_stats_cache()
+    assert db.get_or_build_stats() is not st
+    qb = db.query_builder()
+    assert hasattr(qb, "with_subject")
+    s = db.dictionary.lookup("http://e/a")
+    p = db.dictionary.lookup("http://e/p")
+    o = db.dictionary.lookup("x")
+    assert db.decode_triple((s, p, o)) == ("http://e/a", "http://e/p", "x")
+    assert db.decode_triple((s, p, 999999)) is None
