@@ -56,6 +56,27 @@ class Streamertail:
         self.est = CostEstimator(stats)
         self.memo = {}
 
+    @classmethod
+    def with_cached_stats(cls, stats: DatabaseStats) -> "Streamertail":
+        """Reference-API alias (README: Streamertail::with_cached_stats)."""
+        return cls(stats)
+
+    def execute_plan(self, plan: PhysicalOp, database) -> list:
+        """Run a physical plan against a database, returning decoded
+        variable->value rows (ref README execute_plan -> Vec<BTreeMap>)."""
+        from ..engine.bindings import Bindings
+        from ..engine.executor import (DatasetView, ExecutionContext,
+                                       ExecutionEngine)
+        ctx = ExecutionContext(database, DatasetView())
+        rows = ExecutionEngine(ctx).execute(
+            plan, Bindings.unit(database.device))
+        out = []
+        vars_ = rows.variables
+        cols = {v: rows.col(v).tolist() for v in vars_}
+        for i in range(rows.n):
+            out.append({v: database.decode_term(cols[v][i]) for v in vars_})
+        return out
+
     # ------------------------------------------------------------ entry ----
     def find_best_plan(self, op: LogicalOp) -> PhysicalOp:
         return self._plan(op, set())[0]

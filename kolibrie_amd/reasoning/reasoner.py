@@ -74,6 +74,11 @@ class Reasoner:
         x &= 0xFFFFFFFF
         return x - 0x1_0000_0000 if x >= 0x8000_0000 else x
 
+    def add_tbox_triple(self, s: str, p: str, o: str):
+        """Schema-level assertion — stored in the same fact store (the
+        reference keeps one store too; ref reasoning.rs add_tbox_triple)."""
+        self.add_abox_triple(s, p, o)
+
     def add_abox_triple(self, s: str, p: str, o: str):
         self._pending.append((
             self._i32(self.dictionary.encode(s)),
@@ -122,6 +127,19 @@ class Reasoner:
 
     def add_constraint(self, rule: Rule):
         self.constraints.append(rule)
+
+    def infer_new_facts_semi_naive_with_repairs(self) -> int:
+        """Semi-naive fixpoint, then drop facts participating in minimal
+        constraint repairs (ref reasoning.rs
+        infer_new_facts_semi_naive_with_repairs)."""
+        n = self.infer_new_facts_semi_naive()
+        repairs = self.compute_repairs()
+        removed = 0
+        for rep in repairs[:1]:   # apply one minimal repair, ref behavior
+            for (fs, fp, fo) in rep:
+                self.facts.remove(self._i32(fs), self._i32(fp), self._i32(fo))
+                removed += 1
+        return n - removed
 
     # --------------------------------------------------------- inference --
     def infer_new_facts(self) -> int:

@@ -83,6 +83,14 @@ class FactStore:
     def sorted_unique_rows(self):
         return unique_rows([self.s, self.p, self.o])
 
+    def remove(self, s: int, p: int, o: int) -> bool:
+        """Remove one fact (repair application); True if it was present."""
+        keep = ~((self.s == s) & (self.p == p) & (self.o == o))
+        if bool(keep.all()):
+            return False
+        self.set_columns(self.s[keep], self.p[keep], self.o[keep])
+        return True
+
 
 def membership_in_index(idx: GraphIndex, s, p, o) -> torch.Tensor:
     """Vectorized membership of (s,p,o) rows in a sorted index — exact

@@ -272,3 +272,55 @@ def test_fc_idempotent_reinfer():
     n1 = r.infer_new_facts_semi_naive()
     n2 = r.infer_new_facts_semi_naive()
     assert n1 >= 1 and n2 == 0
+
+
+def test_infer_with_repairs_restores_consistency():
+    """ref reasoning.rs infer_new_facts_semi_naive_with_repairs: after the
+    fixpoint, a minimal repair removes a violating base fact."""
+    r = Reasoner()
+    r.add_abox_triple("x", "status", "active")
+    r.add_abox_triple("x", "status", "banned")
+    r.add_abox_triple("y", "status", "active")
+    from kolibrie_amd.reasoning.rule import Rule
+    from kolibrie_amd.storage.terms import Constant, TriplePattern, Variable
+    status = r.dictionary.encode("status")
+    active = r.dictionary.encode("active")
+    banned = r.dictionary.encode("banned")
+    constraint = Rule(
+        premise=[TriplePattern(Variable("u"), Constant(status), Constant(active)),
+                 TriplePattern(Variable("u"), Constant(status), Constant(banned))],
+        conclusion=[],
+    )
+    r.add_constraint(constraint)
+    assert r.violates_constraints()
+    r.infer_new_facts_semi_naive_with_repairs()
+    assert not r.violates_constraints()
+    # y's fact untouched
+    assert ("y", "status", "active") in r.query_abox("y", None, None)
+
+
+def test_streamertail_execute_plan_api():
+    """ref README Streamertail::execute_plan returning var->value maps."""
+    from kolibrie_amd import SparqlDatabase
+    from kolibrie_amd.parsing.sparql import parse_combined_query
+    from kolibrie_amd.plan.lower import build_logical_plan
+    from kolibrie_amd.plan.optimizer import Streamertail
+    db = SparqlDatabase()
+    db.add_triple("<http://e/a>", "<http://e/p>", "<http://e/b>")
+    cq = parse_combined_query("SELECT ?s ?o WHERE { ?s <http://e/p> ?o }")
+    logical = build_logical_plan(cq.select.where, db, {})
+    st = Streamertail.with_cached_stats(db.get_or_build_stats())
+    plan = st.find_best_plan(logical)
+    rows = st.execute_plan(plan, db)
+    assert rows == [{"s": "http://e/a", "o": "http://e/b"}]
+
+
+def test_add_tbox_triple_participates_in_inference():
+    r = Reasoner()
+    r.add_tbox_triple("Dog", "subClassOf", "Animal")
+    r.add_abox_triple("rex", "type", "Dog")
+    r.add_rule_text(
+        "RULE :sc :- CONSTRUCT { ?x <type> ?super } "
+        "WHERE { ?x <type> ?c . ?c <subClassOf> ?super }")
+    r.infer_new_facts_semi_naive()
+    assert ("rex", "type", "Animal") in r.query_abox("rex", None, None)
