@@ -7,7 +7,7 @@ import time
 
 import torch
 
-sys.path.insert(0, ".")
+sys.path.insert(0, str(__import__("pathlib").Path(__file__).resolve().parent.parent))
 
 from kolibrie_amd.parallel.dist_engine import DistributedDatabase
 from kolibrie_amd.parallel.synthetic import DS, generate_partition, plan_dataset

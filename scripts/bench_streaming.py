@@ -11,7 +11,7 @@ import time
 
 import torch
 
-sys.path.insert(0, ".")
+sys.path.insert(0, str(__import__("pathlib").Path(__file__).resolve().parent.parent))
 
 from kolibrie_amd.rsp import RSPBuilder
 

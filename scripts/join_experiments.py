@@ -12,7 +12,7 @@ import time
 
 import torch
 
-sys.path.insert(0, ".")
+sys.path.insert(0, str(__import__("pathlib").Path(__file__).resolve().parent.parent))
 
 from kolibrie_amd.engine.scan import scan_probe, scan_unit
 from kolibrie_amd.engine.tensor_utils import pack2

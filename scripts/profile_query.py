@@ -9,7 +9,7 @@ import time
 
 import torch
 
-sys.path.insert(0, ".")
+sys.path.insert(0, str(__import__("pathlib").Path(__file__).resolve().parent.parent))
 
 from kolibrie_amd.engine.bindings import Bindings
 from kolibrie_amd.engine.executor import DatasetView, ExecutionContext, ExecutionEngine
