@@ -590,6 +590,20 @@ def join_bindings(left: Bindings, right: Bindings, needed=None) -> Bindings:
         li = torch.arange(left.n, dtype=torch.long, device=dev).repeat_interleave(right.n)
         ri = torch.arange(right.n, dtype=torch.long, device=dev).repeat(left.n)
         return _merge_pairs(left, right, li, ri, shared, needed)
+    if (not left.maybe_unbound and not right.maybe_unbound
+            and left.n * right.n <= 4096):
+        # tiny-tables fast path: selective point queries produce 1-row
+        # intermediates where the merge/hash machinery is pure per-op
+        # overhead — one broadcast compare replaces ~20 tensor ops
+        mask = left.col(shared[0]).unsqueeze(1) == \
+            right.col(shared[0]).unsqueeze(0)
+        for v in shared[1:]:
+            mask = mask & (left.col(v).unsqueeze(1) ==
+                           right.col(v).unsqueeze(0))
+        li, ri = mask.nonzero(as_tuple=True)
+        if needed is not None and len(needed) == 0:
+            return Bindings({}, int(li.numel()), dev)
+        return _merge_pairs(left, right, li, ri, shared, needed)
     parts: List[Bindings] = []
     if not left.maybe_unbound and not right.maybe_unbound:
         l_keyed, l_unkeyed = left, Bindings.empty(dev, left.variables)
