@@ -120,21 +120,23 @@ def scan_unit(idx: GraphIndex, consts: Dict[int, int],
         s, p, o = _cols_from_order(idx, SPO, None, mat_need)
         n_rows = idx.n
     else:
+        # packed bounds computed as PYTHON ints (pack2 convention:
+        # sign-extended hi << 32 | unsigned lo) and both resolved in ONE
+        # searchsorted + ONE device sync — point queries do 5+ of these
+        # per request, so the saved tensor creates/syncs are measurable
         if plen == 2:
-            k = pack2(
-                torch.tensor([consts[pos[0]]], dtype=torch.int32, device=dev),
-                torch.tensor([consts[pos[1]]], dtype=torch.int32, device=dev),
-            )
-            lo = int(torch.searchsorted(key12, k, side="left").item())
-            hi = int(torch.searchsorted(key12, k, side="right").item())
+            k = (consts[pos[0]] << 32) | (consts[pos[1]] & 0xFFFFFFFF)
+            k2 = k + 1  # side='right' of k == side='left' of k+1 (int keys)
         else:
-            v = consts[pos[0]]
-            klo = pack2(torch.tensor([v], dtype=torch.int32, device=dev),
-                        torch.tensor([0], dtype=torch.int32, device=dev))
-            khi = pack2(torch.tensor([v], dtype=torch.int32, device=dev),
-                        torch.tensor([-1], dtype=torch.int32, device=dev))
-            lo = int(torch.searchsorted(key12, klo, side="left").item())
-            hi = int(torch.searchsorted(key12, khi, side="right").item())
+            k = consts[pos[0]] << 32
+            k2 = k + 0x1_0000_0000
+        if k2 > 0x7FFF_FFFF_FFFF_FFFF:
+            probe = torch.tensor([k], dtype=torch.int64, device=dev)
+            lo = int(torch.searchsorted(key12, probe, side="left").item())
+            hi = idx.n
+        else:
+            probe = torch.tensor([k, k2], dtype=torch.int64, device=dev)
+            lo, hi = torch.searchsorted(key12, probe, side="left").tolist()
         s, p, o = _cols_from_order_slice(idx, code, lo, hi, mat_need)
         n_rows = hi - lo
     # post-filter constants not covered by the prefix
