@@ -130,3 +130,24 @@ def test_count_star_fast_path_matches_generic():
     assert execute_query(q + " LIMIT 0", db) == []
     assert execute_query(q + " OFFSET 1", db) == []
     assert execute_query(q, SparqlDatabase()) == [["0"]]
+
+
+def test_scan_unit_extreme_ids():
+    """Range-bound arithmetic edges: max-positive i32 leading constant
+    (k+1 overflow guard) and negative (quoted-style) ids."""
+    import torch
+    from kolibrie_amd.storage.dataset import GraphIndex
+    from kolibrie_amd.engine.scan import scan_unit
+    s = torch.tensor([0x7FFFFFFF, 0x7FFFFFFF, 5],
+                     dtype=torch.int64).to(torch.int32)
+    p = torch.tensor([1, 2, 1], dtype=torch.int32)
+    o = torch.tensor([10, 11, 12], dtype=torch.int32)
+    gi = GraphIndex.from_columns(s, p, o, device="cpu")
+    _, _, ro = scan_unit(gi, {0: 0x7FFFFFFF})
+    assert sorted(ro.tolist()) == [10, 11]
+    _, _, ro = scan_unit(gi, {0: 0x7FFFFFFF, 1: 2})
+    assert ro.tolist() == [11]
+    s2 = torch.tensor([-5, -5, 3], dtype=torch.int32)
+    gi2 = GraphIndex.from_columns(s2, p, o, device="cpu")
+    _, _, ro = scan_unit(gi2, {0: -5})
+    assert sorted(ro.tolist()) == [10, 11]
