@@ -183,8 +183,15 @@ class ExecutionEngine:
         cache = getattr(op, "_chain_cache", None)
         if cache is not None and cache[0] == self.db.store.version:
             seed_key12, seed_z, hop_regions, native = cache[1:5]
+            graph = getattr(op, "_chain_graph", None)
+            if isinstance(graph, tuple):
+                graph[0].replay()
+                return int(graph[1].item())
             if native is not None:
-                return int(native.chain_count(seed_key12, seed_z, *cache[5]))
+                cnt = int(native.chain_count(seed_key12, seed_z, *cache[5]))
+                self._maybe_capture_chain_graph(op, native, seed_key12,
+                                                seed_z, cache[5])
+                return cnt
             return self._chain_count_torch(seed_key12, seed_z, hop_regions)
         if self.ctx.view.default_graphs != [DEFAULT_GRAPH] \
                 or self.ctx.view.named_graphs not in (None, []):
@@ -247,9 +254,40 @@ class ExecutionEngine:
                     [r[3] for r in hop_regions])
         op._chain_cache = (self.db.store.version, seed_key12, seed_z,
                            hop_regions, native, hop_args)
+        if hasattr(op, "_chain_graph"):
+            del op._chain_graph  # stale capture from a previous store version
         if native is not None:
             return int(native.chain_count(seed_key12, seed_z, *hop_args))
         return self._chain_count_torch(seed_key12, seed_z, hop_regions)
+
+    def _maybe_capture_chain_graph(self, op, native, seed_key12, seed_z,
+                                   hop_args):
+        """Record the tile_bounds + chain_count launch pair into a hipGraph
+        (torch.cuda.CUDAGraph IS hipGraph on ROCm): the cached path becomes
+        one graph replay + one 8-byte read-back.  Capture failures fall
+        back permanently to plain launches."""
+        if not seed_key12.is_cuda or getattr(op, "_chain_graph", False) is None:
+            return
+        import torch as _t
+        try:
+            k = len(hop_args[0])
+            n_tiles = (seed_key12.numel() + 255) // 256
+            win = _t.empty(max(1, n_tiles * k * 2), dtype=_t.int64,
+                           device=seed_key12.device)
+            total = _t.zeros(1, dtype=_t.int64, device=seed_key12.device)
+            side = _t.cuda.Stream()
+            side.wait_stream(_t.cuda.current_stream())
+            with _t.cuda.stream(side):   # warmup outside capture
+                native.chain_count_into(seed_key12, seed_z, *hop_args,
+                                        win, total)
+            _t.cuda.current_stream().wait_stream(side)
+            g = _t.cuda.CUDAGraph()
+            with _t.cuda.graph(g):
+                native.chain_count_into(seed_key12, seed_z, *hop_args,
+                                        win, total)
+            op._chain_graph = (g, total, win)
+        except Exception:
+            op._chain_graph = None  # permanent fallback marker
 
     # hop regions much smaller than the seed count pay log2(n) L2 lines per
     # seed in the fused kernel; collapse them ONCE (per store version, the

@@ -732,6 +732,50 @@ __global__ void chain_count_kernel(const int64_t* __restrict__ seed_key12,
     atomicAdd(total, acc);
 }
 
+// Capture-friendly variant: caller owns the window + total buffers, no
+// allocation and no device->host sync inside — the launch pair can be
+// recorded into a hipGraph and replayed per query (launch-bound path).
+void chain_count_into(at::Tensor seed_key12, at::Tensor seed_z,
+                      std::vector<at::Tensor> hop_key12,
+                      std::vector<int64_t> hop_const_hi,
+                      std::vector<int64_t> hop_src,
+                      std::vector<at::Tensor> hop_table,
+                      at::Tensor win, at::Tensor total) {
+  TORCH_CHECK(seed_key12.is_cuda() && seed_z.is_cuda());
+  TORCH_CHECK(win.dtype() == at::kLong && total.dtype() == at::kLong);
+  int64_t m = seed_key12.numel();
+  ChainHops hops{};
+  hops.k = static_cast<int>(hop_key12.size());
+  for (size_t h = 0; h < hop_key12.size(); ++h) {
+    hops.key12[h] = hop_key12[h].data_ptr<int64_t>();
+    hops.n[h] = hop_key12[h].numel();
+    hops.const_hi[h] = hop_const_hi[h];
+    hops.src[h] = static_cast<int32_t>(hop_src[h]);
+    if (hop_table[h].numel() > 0) {
+      hops.table[h] = reinterpret_cast<const unsigned long long*>(
+          hop_table[h].data_ptr<int64_t>());
+      hops.tmask[h] = hop_table[h].numel() - 1;
+    }
+  }
+  auto stream = cur_stream();
+  total.zero_();
+  if (m == 0) return;
+  int64_t n_tiles = (m + kTile - 1) / kTile;
+  TORCH_CHECK(win.numel() >= n_tiles * hops.k * 2, "win buffer too small");
+  hipLaunchKernelGGL(chain_tile_bounds, dim3(grid_for(n_tiles)),
+                     dim3(kBlock), 0, stream,
+                     seed_key12.data_ptr<int64_t>(), m, n_tiles, hops,
+                     win.data_ptr<int64_t>());
+  HIP_OK(hipGetLastError());
+  hipLaunchKernelGGL(chain_count_kernel, dim3(grid_for(m)), dim3(kBlock), 0,
+                     stream, seed_key12.data_ptr<int64_t>(),
+                     seed_z.data_ptr<int32_t>(), m, hops,
+                     win.data_ptr<int64_t>(),
+                     reinterpret_cast<unsigned long long*>(
+                         total.data_ptr<int64_t>()));
+  HIP_OK(hipGetLastError());
+}
+
 int64_t chain_count(at::Tensor seed_key12, at::Tensor seed_z,
                     std::vector<at::Tensor> hop_key12,
                     std::vector<int64_t> hop_const_hi,
@@ -1334,6 +1378,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "K2 chained hash join over int32 key columns -> (li, ri)");
   m.def("chain_count", &chain_count,
         "fused COUNT(*) over a seed scan + probe-hop chain");
+  m.def("chain_count_into", &chain_count_into,
+        "allocation/sync-free chain count into caller buffers "
+        "(hipGraph-capturable)");
   m.def("build_count_table", &build_count_table,
         "open-addressing (value -> match count) table from packed "
         "(val<<32)|count entries, for hashed chain-count hops");
