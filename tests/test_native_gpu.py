@@ -378,3 +378,23 @@ def test_optional_bound_gpu_equals_cpu():
                            sorted(map(tuple, only_unbound)))
     assert results["cpu"] == results["cuda:0"]
     assert len(results["cpu"][1]) == 2000 - len(range(0, 2000, 3))
+
+
+@requires_gpu
+def test_chain_count_hipgraph_replay_stable():
+    """The hipGraph-captured chain-count path must return identical counts
+    across replays and after data changes (graph invalidation)."""
+    from kolibrie_amd import SparqlDatabase
+    from kolibrie_amd.engine.query import execute_query
+    from kolibrie_amd.parallel.synthetic import (FLAGSHIP_QUERY, plan_dataset,
+                                                 generate_partition)
+    db = SparqlDatabase(device="cuda:0")
+    ds = plan_dataset(db, 2_000_000)
+    s, p, o = generate_partition(ds, 0, 1, 42, "cuda:0")
+    db.store.insert_bulk(0, s, p, o)
+    counts = [execute_query(FLAGSHIP_QUERY, db)[0][0] for _ in range(5)]
+    assert len(set(counts)) == 1
+    # mutate the store: version bump must invalidate plan + graph
+    db.add_triple("<http://x/e>", "<http://x/p>", "<http://x/o>")
+    counts2 = [execute_query(FLAGSHIP_QUERY, db)[0][0] for _ in range(4)]
+    assert set(counts2) == set(counts)  # unrelated triple: same count
