@@ -167,19 +167,9 @@ class ExecutionEngine:
         """
         from ..storage.dataset import POS, PSO
         from .tensor_utils import pack2
-        scans: List = []
-
-        def flatten(x) -> bool:
-            if isinstance(x, (PTableScan, PIndexScan)):
-                scans.append(x)
-                return x.graph is None
-            if isinstance(x, (PBindJoin, PHashJoin)):
-                return flatten(x.left) and flatten(x.right)
-            return False
-
-        if not flatten(op) or len(scans) < 2:
-            return None
-        # regions are store-version-stable: cache them on the plan node
+        # regions are store-version-stable: cache them on the plan node.
+        # Checked BEFORE the plan walk — the cached path is the per-query
+        # hot loop and skips every isinstance traversal below.
         cache = getattr(op, "_chain_cache", None)
         if cache is not None and cache[0] == self.db.store.version:
             seed_key12, seed_z, hop_regions, native = cache[1:5]
@@ -193,6 +183,18 @@ class ExecutionEngine:
                                                 seed_z, cache[5])
                 return cnt
             return self._chain_count_torch(seed_key12, seed_z, hop_regions)
+        scans: List = []
+
+        def flatten(x) -> bool:
+            if isinstance(x, (PTableScan, PIndexScan)):
+                scans.append(x)
+                return x.graph is None
+            if isinstance(x, (PBindJoin, PHashJoin)):
+                return flatten(x.left) and flatten(x.right)
+            return False
+
+        if not flatten(op) or len(scans) < 2:
+            return None
         if self.ctx.view.default_graphs != [DEFAULT_GRAPH] \
                 or self.ctx.view.named_graphs not in (None, []):
             pass  # merged views still resolve through default_index below
