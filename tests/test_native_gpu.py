@@ -398,3 +398,27 @@ def test_chain_count_hipgraph_replay_stable():
     db.add_triple("<http://x/e>", "<http://x/p>", "<http://x/o>")
     counts2 = [execute_query(FLAGSHIP_QUERY, db)[0][0] for _ in range(4)]
     assert set(counts2) == set(counts)  # unrelated triple: same count
+
+
+@requires_gpu
+def test_binary_checkpoint_roundtrip_at_scale():
+    """Binary checkpoint/resume with a multi-million-triple GPU store:
+    counts and a query answer survive the round trip."""
+    import tempfile
+    from kolibrie_amd import SparqlDatabase
+    from kolibrie_amd.engine.query import execute_query
+    from kolibrie_amd.parallel.synthetic import (FLAGSHIP_QUERY, plan_dataset,
+                                                 generate_partition)
+    from kolibrie_amd.storage.checkpoint import load_binary, save_binary
+    db = SparqlDatabase(device="cuda:0")
+    ds = plan_dataset(db, 3_000_000)
+    s, p, o = generate_partition(ds, 0, 1, 11, "cuda:0")
+    db.store.insert_bulk(0, s, p, o)
+    want = execute_query(FLAGSHIP_QUERY, db)
+    with tempfile.TemporaryDirectory() as d:
+        path = d + "/ckpt"
+        save_binary(db, path)
+        db2 = SparqlDatabase(device="cuda:0")
+        load_binary(db2, path)
+        assert db2.triple_count() == db.triple_count()
+        assert execute_query(FLAGSHIP_QUERY, db2) == want
