@@ -73,11 +73,16 @@ def main():
     args = ap.parse_args()
     rng = random.Random(args.seed)
 
+    # generate ONCE on CPU: torch.Generator sequences differ per device,
+    # and the differential needs bit-identical stores
     dbs = {}
+    cols = None
     for dev in ["cpu"] + (["cuda:0"] if torch.cuda.is_available() else []):
         db = SparqlDatabase(device=dev)
         ds = plan_dataset(db, args.triples)
-        s, p, o = generate_partition(ds, 0, 1, 777, dev)
+        if cols is None:
+            cols = generate_partition(ds, 0, 1, 777, "cpu")
+        s, p, o = (c.to(dev) for c in cols)
         db.store.insert_bulk(0, s, p, o)
         dbs[dev] = db
     if "cuda:0" not in dbs:
