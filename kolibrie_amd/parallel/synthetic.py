@@ -40,6 +40,7 @@ N_CITIES = 1000
 @dataclass
 class EmployeeDataset:
     pred_ids: Dict[str, int]
+    position_ids: Tuple[int, ...]
     n_employees: int
     n_departments: int
     emp_base: int
@@ -56,8 +57,8 @@ def plan_dataset(db, total_triples: int) -> EmployeeDataset:
     n_emp = int(n_emp)
     n_dept = max(1, n_emp // DEPT_RATIO)
     pred_ids = {k: db.dictionary.encode(v) for k, v in PREDICATES.items()}
-    for i in range(N_POSITIONS):
-        db.dictionary.encode(["Manager", "Developer", "Salesperson"][i])
+    position_ids = tuple(
+        db.dictionary.encode(n) for n in ("Manager", "Developer", "Salesperson"))
     base = len(db.dictionary) + 64
     emp_base = base
     dept_base = emp_base + n_emp
@@ -66,7 +67,8 @@ def plan_dataset(db, total_triples: int) -> EmployeeDataset:
     n_salary = 120_000
     city_base = salary_base + n_salary
     return EmployeeDataset(
-        pred_ids=pred_ids, n_employees=n_emp, n_departments=n_dept,
+        pred_ids=pred_ids, position_ids=position_ids,
+        n_employees=n_emp, n_departments=n_dept,
         emp_base=emp_base, dept_base=dept_base, name_base=name_base,
         salary_base=salary_base, city_base=city_base, n_salary_values=n_salary,
     )
@@ -108,7 +110,8 @@ def generate_partition(ds: EmployeeDataset, rank: int, world: int, seed: int,
     add("name", emp_ids, ds.name_base + emp)
     add("homepage", emp_ids, ds.name_base + emp)  # homepage shares name id pool
     add("salary", emp_ids, ds.salary_base + rnd(ds.n_salary_values, ne))
-    add("position", emp_ids, 1 + rnd(N_POSITIONS, ne))
+    pos_tbl = torch.tensor(ds.position_ids, dtype=torch.int64, device=dev)
+    add("position", emp_ids, pos_tbl[rnd(N_POSITIONS, ne)])
     add("email", emp_ids, ds.name_base + emp)
     add("age", emp_ids, ds.salary_base + rnd(50, ne))
     dept_of = ds.dept_base + (emp % ds.n_departments)
