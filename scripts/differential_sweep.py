@@ -34,7 +34,21 @@ P = {
 def gen_query(rng: random.Random) -> str:
     """One random query over the employee schema."""
     kind = rng.choice(["star", "chain", "filter", "optional", "union",
-                      "agg", "distinct", "values"])
+                      "agg", "distinct", "values", "bind", "subquery",
+                      "minus", "ask"])
+    if kind == "bind":
+        return (f"SELECT (COUNT(*) AS ?c) WHERE {{ ?e {P['worksFor']} ?d . "
+                f"BIND(TRIPLE(?e, {P['worksFor']}, ?d) AS ?t) . "
+                f"FILTER(isTRIPLE(?t)) }}")
+    if kind == "subquery":
+        return (f"SELECT (COUNT(*) AS ?c) WHERE {{ "
+                f"{{ SELECT ?d WHERE {{ ?d {P['locatedIn']} ?city }} "
+                f"LIMIT 200 }} ?e {P['worksFor']} ?d }}")
+    if kind == "minus":
+        return (f"SELECT (COUNT(*) AS ?c) WHERE {{ ?e {P['position']} ?v . "
+                f"MINUS {{ ?e {P['age']} ?a }} }}")
+    if kind == "ask":
+        return f"ASK {{ ?e {P['worksFor']} ?d . ?d {P['locatedIn']} ?c }}"
     if kind == "star":
         preds = rng.sample(list(P.values()), rng.randint(2, 4))
         pats = " . ".join(f"?e {p} ?v{i}" for i, p in enumerate(preds))
