@@ -85,6 +85,8 @@ def main():
     ap = argparse.ArgumentParser()
     ap.add_argument("--device", default="cuda:0" if torch.cuda.is_available() else "cpu")
     ap.add_argument("--quick", action="store_true")
+    ap.add_argument("--huge", action="store_true",
+                    help="add a 100M-edge closure config (GPU capacity demo)")
     args = ap.parse_args()
 
     print("== deep taxonomy (type propagation, BASELINE.md item 2) ==",
@@ -104,6 +106,8 @@ def main():
 
     print("== 10M-edge transitive closure (BASELINE config 4) ==", flush=True)
     confs = [(100_000, 10), (1_000_000, 10)] if not args.quick else [(10_000, 10)]
+    if args.huge:
+        confs.append((10_000_000, 10))   # 100M edges -> 550M total facts
     for n_chains, depth in confs:
         res = run(n_chains, depth, args.device)
         print(f"chains {n_chains:,} depth {depth}: edges {res['edges']:,} "
