@@ -422,3 +422,27 @@ def test_binary_checkpoint_roundtrip_at_scale():
         load_binary(db2, path)
         assert db2.triple_count() == db.triple_count()
         assert execute_query(FLAGSHIP_QUERY, db2) == want
+
+
+@requires_gpu
+def test_serving_soak_memory_stable():
+    """3,000 cached queries: device memory must not grow (no per-query
+    allocation leaks on the replay path) and answers stay constant."""
+    from kolibrie_amd import SparqlDatabase
+    from kolibrie_amd.engine.query import execute_query
+    from kolibrie_amd.parallel.synthetic import (FLAGSHIP_QUERY, plan_dataset,
+                                                 generate_partition)
+    db = SparqlDatabase(device="cuda:0")
+    ds = plan_dataset(db, 5_000_000)
+    s, p, o = generate_partition(ds, 0, 1, 5, "cuda:0")
+    db.store.insert_bulk(0, s, p, o)
+    first = execute_query(FLAGSHIP_QUERY, db)
+    for _ in range(20):
+        execute_query(FLAGSHIP_QUERY, db)
+    torch.cuda.synchronize()
+    base = torch.cuda.memory_allocated()
+    for i in range(3000):
+        assert execute_query(FLAGSHIP_QUERY, db) == first
+    torch.cuda.synchronize()
+    growth = torch.cuda.memory_allocated() - base
+    assert growth < 16 * 1024 * 1024, f"leaked {growth} bytes over 3k queries"
