@@ -140,7 +140,18 @@ def _apply_templates(db, prefixes, templates: List[QuadData], rows: Bindings,
     unbound_mask = {v: (rows.col(v) == UNBOUND).cpu().tolist()
                     for v in rows.variables}
     for q in templates:
-        gid = _gid_of(db, q.g, prefixes)
+        # GRAPH ?g templates take the graph id from each solution's binding
+        # (ref sparql_graph_test.rs graph_variables_flow_from_where_into_
+        # delete_and_insert_templates)
+        g_ids = None
+        if q.g is not None and _is_var(q.g.strip()):
+            gname = q.g.strip()[1:]
+            if gname not in host_cols:
+                continue
+            g_ids = host_cols[gname]
+            gid = None
+        else:
+            gid = _gid_of(db, q.g, prefixes)
         bnode_map: Dict[str, List[int]] = {}
         term_ids = []
         ok_rows = [True] * n
@@ -175,11 +186,15 @@ def _apply_templates(db, prefixes, templates: List[QuadData], rows: Bindings,
         for i in range(n):
             if not ok_rows[i]:
                 continue
+            row_gid = g_ids[i] if g_ids is not None else gid
+            if g_ids is not None and unbound_mask.get(q.g.strip()[1:],
+                                                      [False] * n)[i]:
+                continue
             s, p, o = term_ids[0][i], term_ids[1][i], term_ids[2][i]
             if delete:
-                db.store.delete_quad(gid, s, p, o)
+                db.store.delete_quad(row_gid, s, p, o)
             else:
-                db.store.insert_quad(gid, s, p, o)
+                db.store.insert_quad(row_gid, s, p, o)
 
 
 def _graph_management(db, op: UpdateOperation, prefixes):

@@ -340,3 +340,20 @@ def test_ask_query():
     assert db.query(f'ASK {{ ?s <{EX}p> ?o . FILTER(?s = ?o) }}') == [["false"]]
     # cached second run
     assert db.query(f'ASK {{ <{EX}a> <{EX}p> ?x }}') == [["true"]]
+
+
+def test_graph_variable_flows_into_update_templates():
+    """ref sparql_graph_test.rs graph_variables_flow_from_where_into_
+    delete_and_insert_templates: GRAPH ?g templates substitute per
+    solution."""
+    db = SparqlDatabase()
+    db.query('INSERT DATA { GRAPH <http://g1> { <http://e/a> <http://e/p> <http://e/b> } }')
+    db.query('INSERT DATA { GRAPH <http://g2> { <http://e/c> <http://e/p> <http://e/d> } }')
+    db.query('''DELETE { GRAPH ?g { ?s <http://e/p> ?o } }
+                INSERT { GRAPH ?g { ?s <http://e/q> ?o } }
+                WHERE { GRAPH ?g { ?s <http://e/p> ?o } }''')
+    assert sorted(db.query(
+        'SELECT ?g ?s WHERE { GRAPH ?g { ?s <http://e/q> ?o } }')) == [
+        ["http://g1", "http://e/a"], ["http://g2", "http://e/c"]]
+    assert db.query(
+        'SELECT ?s WHERE { GRAPH <http://g1> { ?s <http://e/p> ?o } }') == []
