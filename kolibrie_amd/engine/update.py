@@ -171,8 +171,16 @@ def _apply_templates(db, prefixes, templates: List[QuadData], rows: Bindings,
                 # fresh blank node per solution (ref :603-629)
                 ids = []
                 for i in range(n):
-                    _BNODE_COUNTER[0] += 1
-                    ids.append(db.dictionary.encode(f"_:upd{_BNODE_COUNTER[0]}"))
+                    # fresh label: skip ids already interned (persisted
+                    # lexical collisions must not alias two bnodes — ref
+                    # update_blank_node_allocation_skips_persisted_
+                    # lexical_collisions)
+                    while True:
+                        _BNODE_COUNTER[0] += 1
+                        label = f"_:upd{_BNODE_COUNTER[0]}"
+                        if db.dictionary.lookup(label) is None:
+                            break
+                    ids.append(db.dictionary.encode(label))
                 term_ids.append(ids)
             else:
                 tid = db.encode_term_star(t, prefixes)

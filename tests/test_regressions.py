@@ -151,3 +151,31 @@ def test_scan_unit_extreme_ids():
     gi2 = GraphIndex.from_columns(s2, p, o, device="cpu")
     _, _, ro = scan_unit(gi2, {0: -5})
     assert sorted(ro.tolist()) == [10, 11]
+
+
+def test_update_bnode_allocation_skips_lexical_collisions():
+    """Pre-existing '_:updN' labels must not alias fresh update bnodes
+    (ref sparql_dataset_regressions_test.rs)."""
+    from kolibrie_amd import SparqlDatabase
+    db = SparqlDatabase()
+    db.add_triple("_:upd1", "<http://e/x>", '"old"')     # collide on purpose
+    db.add_triple("<http://e/s>", "<http://e/p>", '"v"')
+    db.query('INSERT { ?s <http://e/tag> _:b } WHERE { ?s <http://e/p> ?o }')
+    tags = db.query('SELECT ?b WHERE { <http://e/s> <http://e/tag> ?b }')
+    assert len(tags) == 1
+    # the fresh bnode must NOT be the pre-existing _:upd1
+    old = db.query('SELECT ?o WHERE { _:upd1 <http://e/x> ?o }')
+    assert old == [["old"]]
+    assert tags[0][0] != "_:upd1"
+
+
+def test_rebuild_indexes_keeps_named_quads_and_empty_graphs():
+    from kolibrie_amd import SparqlDatabase
+    db = SparqlDatabase()
+    db.query('INSERT DATA { GRAPH <http://g1> { <http://e/a> <http://e/p> <http://e/b> } }')
+    db.query('CREATE GRAPH <http://gEmpty>')
+    db.build_all_indexes()
+    gids = {db.decode_term(g) for g in db.store.named_graph_ids()}
+    assert "http://g1" in gids and "http://gEmpty" in gids
+    assert db.query('SELECT ?s WHERE { GRAPH <http://g1> { ?s ?p ?o } }') == \
+        [["http://e/a"]]
