@@ -78,7 +78,18 @@ def create_app(db: Optional[SparqlDatabase] = None, device: str = "cpu"):
                 return data["query"]
         except json.JSONDecodeError:
             pass
-        return body.decode("utf-8", "replace")
+        text = body.decode("utf-8", "replace")
+        ctype = request.headers.get("content-type", "")
+        if "application/x-www-form-urlencoded" in ctype:
+            # SPARQL-protocol form encoding: query=...&... (ref
+            # http_sparql_query_encodings_use_the_unified_query_executor)
+            from urllib.parse import parse_qs
+            form = parse_qs(text)
+            if "query" in form:
+                return form["query"][0]
+            if "update" in form:
+                return form["update"][0]
+        return text
 
     @app.get("/", response_class=HTMLResponse)
     async def index():

@@ -133,3 +133,16 @@ def test_query_engine_explain():
     assert qe.query(f"SELECT ?s WHERE {{ ?s <{EX}p> ?o }}") == [[f"{EX}a"]]
     plan = qe.explain(f"SELECT ?s WHERE {{ ?s <{EX}p> ?o . ?s <{EX}q> ?x }}")
     assert "Scan" in plan
+
+
+def test_http_query_form_encoding(client):
+    """SPARQL-protocol urlencoded form body (ref
+    http_sparql_query_encodings_use_the_unified_query_executor)."""
+    from urllib.parse import quote
+    q = f'SELECT ?n WHERE {{ <{EX}alice> <{EX}name> ?n }}'
+    r = client.post("/query", content=f"query={quote(q)}",
+                    headers={"content-type":
+                             "application/x-www-form-urlencoded"})
+    assert r.status_code == 200
+    vals = [b["n"]["value"] for b in r.json()["results"]["bindings"]]
+    assert vals == ["Alice"]
