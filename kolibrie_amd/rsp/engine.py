@@ -261,6 +261,8 @@ class RSPEngine:
     def flush_windows(self):
         for entry in self.windows.values():
             entry.window.flush()
+        for bw in getattr(self, "_bulk_windows", {}).values():
+            bw.flush()
         if self.operation_mode == OperationMode.MULTI_THREAD:
             self._drain_multithread(deadline_ms=2000)
 

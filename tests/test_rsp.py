@@ -479,3 +479,35 @@ def test_static_data_not_visible_in_window_query():
     flat = [v for rows in got for r in rows for v in r]
     assert any("ev" in v for v in flat)
     assert not any("static" in v for v in flat)
+
+
+def test_bulk_multi_window_join():
+    """Two windows fed by the COLUMNAR bulk path join like the host path."""
+    import torch
+    from kolibrie_amd.rsp.builder import RSPBuilder
+    q = """
+        REGISTER RSTREAM <out> AS
+        SELECT ?m ?t ?h
+        FROM NAMED WINDOW <wT> ON STREAM <sT> [RANGE 10 STEP 10]
+        FROM NAMED WINDOW <wH> ON STREAM <sH> [RANGE 10 STEP 10]
+        WHERE {
+            WINDOW <wT> { ?m <http://x/temp> ?t }
+            WINDOW <wH> { ?m <http://x/hum> ?h }
+        }
+    """
+    got = []
+    eng = (RSPBuilder().add_rsp_ql_query(q)
+           .add_consumer(lambda rows: got.append(rows)).build())
+    db = eng.store.db
+    temp = db.encode_term("<http://x/temp>")
+    hum = db.encode_term("<http://x/hum>")
+    m1 = db.encode_term("<http://x/m1>")
+    v1 = db.encode_term('"20"')
+    v2 = db.encode_term('"60"')
+    mk = lambda *xs: torch.tensor(xs, dtype=torch.int32)
+    ts = torch.tensor([3, 12], dtype=torch.int64)  # second event fires [0,10)
+    eng.add_to_stream_bulk("<sT>", mk(m1, m1), mk(temp, temp), mk(v1, v1), ts)
+    eng.add_to_stream_bulk("<sH>", mk(m1, m1), mk(hum, hum), mk(v2, v2), ts)
+    flat = [r for rows in got for r in rows]
+    assert any("m1" in str(r) and "20" in str(r) and "60" in str(r)
+               for r in flat), flat
