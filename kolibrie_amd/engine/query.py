@@ -161,6 +161,89 @@ def _run_prepared(pq: "PreparedQuery", db) -> List[List[str]]:
     return decode_rows(pq.select, final, db)
 
 
+def _execute_construct(cq: CombinedQuery, db, prefixes) -> List[List[str]]:
+    """CONSTRUCT { templates } WHERE {...}: instantiate templates per
+    solution, dedup, return decoded (s,p,o) rows (engine extension —
+    the reference only uses CONSTRUCT inside RULE definitions)."""
+    sel = cq.select
+    from ..plan.lower import build_logical_plan
+    from ..plan.optimizer import Streamertail, annotate_needed
+    view = _build_view(sel, db, prefixes)
+    logical = build_logical_plan(sel.where, db, prefixes)
+    physical = Streamertail(db.get_or_build_stats()).find_best_plan(logical)
+    annotate_needed(physical, None)
+    ctx = ExecutionContext(db, view)
+    rows = ExecutionEngine(ctx).execute(physical, Bindings.unit(db.device))
+    import torch
+    host = {v: (rows.col(v).to(torch.int64) & 0xFFFFFFFF).cpu().tolist()
+            for v in rows.variables}
+    unb = {v: (rows.col(v) == -1).cpu().tolist() for v in rows.variables}
+    out = []
+    seen = set()
+    for t in cq.construct:
+        ids = []
+        for term in (t.s, t.p, t.o):
+            term = term.strip()
+            if term.startswith("?") or term.startswith("$"):
+                name = term[1:]
+                ids.append(("v", name) if name in host else None)
+            else:
+                ids.append(("c", db.encode_term_star(term, prefixes)))
+        if any(x is None for x in ids):
+            continue
+        for i in range(rows.n):
+            trip = []
+            ok = True
+            for kind, val in ids:
+                if kind == "c":
+                    trip.append(val)
+                elif unb[val][i]:
+                    ok = False
+                    break
+                else:
+                    trip.append(host[val][i])
+            if not ok:
+                continue
+            key = tuple(trip)
+            if key in seen:
+                continue
+            seen.add(key)
+            out.append([db.decode_term(x) for x in trip])
+    if sel.offset:
+        out = out[sel.offset:]
+    if sel.limit is not None:
+        out = out[:sel.limit]
+    return out
+
+
+def _execute_describe(cq: CombinedQuery, db, prefixes) -> List[List[str]]:
+    """DESCRIBE <term>...: every triple whose subject or object is the
+    term (engine extension)."""
+    out = []
+    seen = set()
+    for term in cq.describe:
+        if term.startswith("?"):
+            continue  # variable DESCRIBE needs a WHERE; keep simple
+        tid = db.dictionary.lookup(db.resolve_lexical(term, prefixes))
+        if tid is None:
+            continue
+        idx = db.store.graph_index(DEFAULT_GRAPH)
+        for pos in (0, 2):
+            from .scan import scan_unit
+            s, p, o = scan_unit(idx, {pos: tid - 0x1_0000_0000
+                                      if tid >= 0x8000_0000 else tid})
+            for a, b, c in zip(
+                (s.to(__import__("torch").int64) & 0xFFFFFFFF).cpu().tolist(),
+                (p.to(__import__("torch").int64) & 0xFFFFFFFF).cpu().tolist(),
+                (o.to(__import__("torch").int64) & 0xFFFFFFFF).cpu().tolist(),
+            ):
+                if (a, b, c) not in seen:
+                    seen.add((a, b, c))
+                    out.append([db.decode_term(a), db.decode_term(b),
+                                db.decode_term(c)])
+    return out
+
+
 def execute_query(sparql: str, db) -> List[List[str]]:
     """Full request entry (ref execute_query_rayon_parallel2_volcano,
     execute_query.rs:52).  Pure SELECT queries hit the prepared-plan
@@ -180,6 +263,10 @@ def execute_query(sparql: str, db) -> List[List[str]]:
             execute_update(op, db, prefixes)
         if cq.select is None:
             return []
+    if cq.construct is not None:
+        return _execute_construct(cq, db, prefixes)
+    if cq.describe is not None:
+        return _execute_describe(cq, db, prefixes)
     if cq.select is not None:
         if not cq.updates and not cq.rules and not cq.train_decls \
                 and not cq.register:

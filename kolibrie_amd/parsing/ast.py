@@ -315,4 +315,6 @@ class CombinedQuery:
     register: Optional[RegisterClause] = None
     retrieve: Optional[RetrieveClause] = None
     select: Optional[SelectQuery] = None
+    construct: Optional[List[TriplePatternAst]] = None   # CONSTRUCT templates
+    describe: Optional[List[str]] = None                 # DESCRIBE terms
     updates: List[UpdateOperation] = field(default_factory=list)

@@ -225,6 +225,44 @@ class Parser:
                 cq.retrieve = self.parse_retrieve()
             elif self.at_kw("SELECT"):
                 cq.select = self.parse_select_core()
+            elif self.at_kw("CONSTRUCT"):
+                # standalone CONSTRUCT query (engine extension)
+                self.next()
+                self.expect("{")
+                pats = []
+                while not self.at("}"):
+                    if self.at("."):
+                        self.next()
+                        continue
+                    pats.extend(self.parse_triples_block().patterns)
+                self.expect("}")
+                self.expect_kw("WHERE")
+                q = SelectQuery(select_star=True)
+                q.where = self.parse_group()
+                while True:
+                    if self.at_kw("LIMIT"):
+                        self.next()
+                        q.limit = int(self.next().text)
+                    elif self.at_kw("OFFSET"):
+                        self.next()
+                        q.offset = int(self.next().text)
+                    else:
+                        break
+                cq.select = q
+                cq.construct = pats
+            elif self.at_kw("DESCRIBE"):
+                # DESCRIBE <iri>... [WHERE {...}] (engine extension)
+                self.next()
+                terms = []
+                while not self.eof() and not self.at_kw("WHERE") \
+                        and (self.peek().kind in ("iri", "pname", "var")):
+                    terms.append(self.next().text)
+                q = SelectQuery(select_star=True)
+                if self.at_kw("WHERE"):
+                    self.next()
+                    q.where = self.parse_group()
+                cq.select = q
+                cq.describe = terms
             elif self.at_kw("ASK"):
                 # ASK { pattern } — boolean query (engine extension)
                 self.next()

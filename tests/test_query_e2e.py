@@ -357,3 +357,30 @@ def test_graph_variable_flows_into_update_templates():
         ["http://g1", "http://e/a"], ["http://g2", "http://e/c"]]
     assert db.query(
         'SELECT ?s WHERE { GRAPH <http://g1> { ?s <http://e/p> ?o } }') == []
+
+
+def test_construct_query():
+    """Standalone CONSTRUCT (engine extension beyond the reference, which
+    only uses CONSTRUCT inside RULE bodies)."""
+    db = SparqlDatabase()
+    db.add_triple(f"<{EX}a>", f"<{EX}p>", f"<{EX}b>")
+    db.add_triple(f"<{EX}b>", f"<{EX}p>", f"<{EX}c>")
+    rows = db.query(
+        f'CONSTRUCT {{ ?o <{EX}invP> ?s }} WHERE {{ ?s <{EX}p> ?o }}')
+    assert sorted(rows) == [
+        [f"{EX}b", f"{EX}invP", f"{EX}a"],
+        [f"{EX}c", f"{EX}invP", f"{EX}b"]]
+    one = db.query(
+        f'CONSTRUCT {{ ?o <{EX}invP> ?s }} WHERE {{ ?s <{EX}p> ?o }} LIMIT 1')
+    assert len(one) == 1
+
+
+def test_describe_query():
+    db = SparqlDatabase()
+    db.add_triple(f"<{EX}a>", f"<{EX}p>", f"<{EX}b>")
+    db.add_triple(f"<{EX}b>", f"<{EX}p>", f"<{EX}c>")
+    rows = db.query(f'DESCRIBE <{EX}b>')
+    assert sorted(rows) == [
+        [f"{EX}a", f"{EX}p", f"{EX}b"],
+        [f"{EX}b", f"{EX}p", f"{EX}c"]]
+    assert db.query(f'DESCRIBE <{EX}missing>') == []
