@@ -64,7 +64,7 @@ _TOK = re.compile(
     | (?P<var>[?$][A-Za-z_][\w]*)
     | (?P<number>[+-]?(\d+\.\d*|\.\d+|\d+)([eE][+-]?\d+)?)
     | (?P<ruleop>:-)
-    | (?P<op><=|>=|!=|&&|\|\||[=<>!+\-*/])
+    | (?P<op><=|>=|!=|&&|\|\||[=<>!+\-*/^])
     | (?P<punct>[{}()\[\],;.])
     | (?P<bnode>_:[A-Za-z0-9_.-]+)
     | (?P<pname>[A-Za-z_][\w.-]*:[\w.-]*|:[\w.-]+)
@@ -462,14 +462,43 @@ class Parser:
             return GBgp(left.patterns + right.patterns)
         return GJoin(left, right)
 
+    _path_var_counter = [0]
+
+    def _parse_path_steps(self):
+        """Property-path subset (engine extension): sequences `p1/p2` and
+        inverse steps `^p`, desugared to fresh-variable join chains."""
+        steps = []
+        while True:
+            inv = False
+            if self.at("^"):
+                self.next()
+                inv = True
+            steps.append((inv, self.parse_term()))
+            if self.at("/"):
+                self.next()
+                continue
+            break
+        return steps
+
+    def _fresh_path_var(self) -> str:
+        Parser._path_var_counter[0] += 1
+        return f"?__pp{Parser._path_var_counter[0]}"
+
     def parse_triples_block(self) -> GBgp:
         pats: List[TriplePatternAst] = []
         s = self.parse_term()
         while True:
-            p = self.parse_term()
+            steps = self._parse_path_steps()
             while True:
                 o = self.parse_term()
-                pats.append(TriplePatternAst(s, p, o))
+                cur = s
+                for j, (inv, pred) in enumerate(steps):
+                    nxt = o if j == len(steps) - 1 else self._fresh_path_var()
+                    if inv:
+                        pats.append(TriplePatternAst(nxt, pred, cur))
+                    else:
+                        pats.append(TriplePatternAst(cur, pred, nxt))
+                    cur = nxt
                 if self.at(","):
                     self.next()
                     continue

@@ -384,3 +384,24 @@ def test_describe_query():
         [f"{EX}a", f"{EX}p", f"{EX}b"],
         [f"{EX}b", f"{EX}p", f"{EX}c"]]
     assert db.query(f'DESCRIBE <{EX}missing>') == []
+
+
+def test_property_path_sequence_and_inverse():
+    """Path subset (engine extension): `p1/p2` sequences and `^p` inverse
+    steps desugar to join chains."""
+    db = SparqlDatabase()
+    db.add_triple(f"<{EX}a>", f"<{EX}worksFor>", f"<{EX}d1>")
+    db.add_triple(f"<{EX}d1>", f"<{EX}locatedIn>", f"<{EX}c1>")
+    db.add_triple(f"<{EX}b>", f"<{EX}worksFor>", f"<{EX}d1>")
+    rows = db.query(
+        f'SELECT ?c WHERE {{ <{EX}a> <{EX}worksFor>/<{EX}locatedIn> ?c }}')
+    assert rows == [[f"{EX}c1"]]
+    rows = db.query(
+        f'SELECT ?e WHERE {{ ?e <{EX}worksFor>/<{EX}locatedIn> <{EX}c1> }}')
+    assert sorted(r[0] for r in rows) == [f"{EX}a", f"{EX}b"]
+    rows = db.query(f'SELECT ?d WHERE {{ ?d ^<{EX}worksFor> <{EX}a> }}')
+    assert rows == [[f"{EX}d1"]]
+    # 3-step with inverse in the middle: a worksFor d1 ^worksFor b
+    rows = db.query(
+        f'SELECT ?x WHERE {{ <{EX}a> <{EX}worksFor>/^<{EX}worksFor> ?x }}')
+    assert sorted(r[0] for r in rows) == [f"{EX}a", f"{EX}b"]
