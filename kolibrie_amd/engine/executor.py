@@ -348,6 +348,10 @@ class ExecutionEngine:
             idx = self.db.store.graph_index(scope[1] & 0xFFFFFFFF)
             return self._scan_index(idx, pattern, incoming, extra=None,
                                     needed=needed, sort_hint=sort_hint)
+        if scope[0] == "closure":
+            idx = self._closure_index(scope[1], scope[2])
+            return self._scan_index(idx, pattern, incoming, extra=None,
+                                    needed=needed, sort_hint=sort_hint)
         # GRAPH ?g — iterate named graphs, bind the graph variable
         gvar = scope[1]
         if needed is not None:
@@ -363,6 +367,62 @@ class ExecutionEngine:
             all_vars = pattern.variables() + [gvar]
             return Bindings.empty(self.device, all_vars)
         return Bindings.concat(parts, self.device)
+
+    def _closure_index(self, pid: int, reflexive: bool) -> GraphIndex:
+        """Transitive closure of predicate `pid` as an auxiliary
+        GraphIndex, materialized by log-doubling joins on device and
+        cached per store version (p+ / p* property paths).  `*` adds the
+        zero-length pairs over the predicate's node set."""
+        cache = getattr(self.db, "_closures", None)
+        if cache is None:
+            cache = self.db._closures = {}
+        key = (pid, reflexive)
+        hit = cache.get(key)
+        if hit is not None and hit[0] == self.db.store.version:
+            return hit[1]
+        from .scan import scan_unit
+        from .tensor_utils import pack2, unique_rows
+        pid_i32 = pid - 0x1_0000_0000 if pid >= 0x8000_0000 else pid
+        base = self.ctx.default_index()
+        s, _p, o = scan_unit(base, {1: pid_i32})
+        dev = self.device
+        cs, co = s, o
+        # squaring: C_{k+1} = C_k ∪ (C_k ∘ C_k) reaches length 2^k in k
+        # rounds (log-depth for chains; joins are the same sorted
+        # searchsorted machinery as the fixpoint kernels drive)
+        for _ in range(64):
+            cs, co = unique_rows([cs, co])
+            # join C.o == C.s
+            left_o = co.to(torch.int64) & 0xFFFFFFFF
+            right_s = (cs.to(torch.int64) & 0xFFFFFFFF)
+            order = torch.argsort(right_s)
+            rs_sorted = right_s[order]
+            lo = torch.searchsorted(rs_sorted, left_o, side="left")
+            hi = torch.searchsorted(rs_sorted, left_o, side="right")
+            cnt = hi - lo
+            li = torch.repeat_interleave(
+                torch.arange(cs.numel(), dtype=torch.long, device=dev), cnt)
+            starts = torch.cumsum(cnt, 0) - cnt
+            pos = torch.arange(int(cnt.sum().item()), dtype=torch.long,
+                               device=dev) - starts[li]
+            ri = order[lo[li] + pos]
+            new_s, new_o = cs[li], co[ri]
+            all_s = torch.cat([cs, new_s])
+            all_o = torch.cat([co, new_o])
+            u_s, u_o = unique_rows([all_s, all_o])
+            if u_s.numel() == cs.numel():
+                cs, co = u_s, u_o
+                break
+            cs, co = u_s, u_o
+        if reflexive:
+            nodes = torch.unique(torch.cat([s, o]))
+            cs = torch.cat([cs, nodes])
+            co = torch.cat([co, nodes])
+        pcol = torch.full((cs.numel(),), pid_i32, dtype=torch.int32,
+                          device=dev)
+        gi = GraphIndex.from_columns(cs, pcol, co, device=str(dev))
+        cache[key] = (self.db.store.version, gi)
+        return gi
 
     def _scan_index(self, idx: GraphIndex, pattern: TriplePattern,
                     incoming: Bindings, extra: Optional[Tuple[str, int]],

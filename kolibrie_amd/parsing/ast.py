@@ -73,6 +73,7 @@ class TriplePatternAst:
     s: str
     p: str
     o: str
+    path_mod: Optional[str] = None   # '+' | '*' transitive-closure steps
 
 
 @dataclass

@@ -473,7 +473,11 @@ class Parser:
             if self.at("^"):
                 self.next()
                 inv = True
-            steps.append((inv, self.parse_term()))
+            pred = self.parse_term()
+            mod = None
+            if self.at("+") or self.at("*"):
+                mod = self.next().text
+            steps.append((inv, pred, mod))
             if self.at("/"):
                 self.next()
                 continue
@@ -492,12 +496,12 @@ class Parser:
             while True:
                 o = self.parse_term()
                 cur = s
-                for j, (inv, pred) in enumerate(steps):
+                for j, (inv, pred, mod) in enumerate(steps):
                     nxt = o if j == len(steps) - 1 else self._fresh_path_var()
                     if inv:
-                        pats.append(TriplePatternAst(nxt, pred, cur))
+                        pats.append(TriplePatternAst(nxt, pred, cur, mod))
                     else:
-                        pats.append(TriplePatternAst(cur, pred, nxt))
+                        pats.append(TriplePatternAst(cur, pred, nxt, mod))
                     cur = nxt
                 if self.at(","):
                     self.next()

@@ -81,7 +81,21 @@ def build_logical_plan(
     if isinstance(g, GBgp):
         node: LogicalOp = LUnit()
         for pat in g.patterns:
-            scan = LScan(compile_triple_pattern(pat, prefixes, db), scope)
+            cp = compile_triple_pattern(pat, prefixes, db)
+            pat_scope = scope
+            if getattr(pat, "path_mod", None):
+                # p+ / p* closure steps scan a materialized closure index
+                # (built lazily by the executor from the device fixpoint)
+                from ..storage.terms import Constant
+                if not isinstance(cp.p, Constant):
+                    raise ValueError(
+                        "property-path closure requires a constant predicate")
+                if scope is not None:
+                    raise ValueError(
+                        "property-path closure inside GRAPH is unsupported")
+                pat_scope = ("closure", cp.p.id & 0xFFFFFFFF,
+                             pat.path_mod == "*")
+            scan = LScan(cp, pat_scope)
             node = scan if isinstance(node, LUnit) else LJoin(node, scan)
         return node
     if isinstance(g, GJoin):

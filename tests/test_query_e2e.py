@@ -405,3 +405,24 @@ def test_property_path_sequence_and_inverse():
     rows = db.query(
         f'SELECT ?x WHERE {{ <{EX}a> <{EX}worksFor>/^<{EX}worksFor> ?x }}')
     assert sorted(r[0] for r in rows) == [f"{EX}a", f"{EX}b"]
+
+
+def test_property_path_transitive_closure():
+    """p+ / p* closure paths (engine extension): materialized by the
+    log-doubling device join, cached per store version, cycle-safe."""
+    db = SparqlDatabase()
+    for i in range(6):
+        db.add_triple(f"<{EX}n{i}>", f"<{EX}next>", f"<{EX}n{i+1}>")
+    rows = db.query(f'SELECT ?o WHERE {{ <{EX}n0> <{EX}next>+ ?o }}')
+    assert sorted(r[0] for r in rows) == [f"{EX}n{i}" for i in range(1, 7)]
+    assert len(db.query(f'SELECT ?s ?o WHERE {{ ?s <{EX}next>+ ?o }}')) == 21
+    star = db.query(f'SELECT ?o WHERE {{ <{EX}n4> <{EX}next>* ?o }}')
+    assert sorted(r[0] for r in star) == [f"{EX}n4", f"{EX}n5", f"{EX}n6"]
+    # cycle: closure must terminate and be complete
+    db.add_triple(f"<{EX}n6>", f"<{EX}next>", f"<{EX}n0>")
+    assert len(db.query(f'SELECT ?s ?o WHERE {{ ?s <{EX}next>+ ?o }}')) == 49
+    # closure composes with further joins
+    db.add_triple(f"<{EX}n3>", f"<{EX}tag>", '"x"')
+    rows = db.query(
+        f'SELECT ?t WHERE {{ <{EX}n1> <{EX}next>+ ?m . ?m <{EX}tag> ?t }}')
+    assert rows == [["x"]]
