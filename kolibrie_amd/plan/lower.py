@@ -86,7 +86,6 @@ def build_logical_plan(
             if getattr(pat, "path_mod", None):
                 # p+ / p* closure steps scan a materialized closure index
                 # (built lazily by the executor from the device fixpoint)
-                from ..storage.terms import Constant
                 if not isinstance(cp.p, Constant):
                     raise ValueError(
                         "property-path closure requires a constant predicate")
