@@ -35,7 +35,10 @@ def gen_query(rng: random.Random) -> str:
     """One random query over the employee schema."""
     kind = rng.choice(["star", "chain", "filter", "optional", "union",
                       "agg", "distinct", "values", "bind", "subquery",
-                      "minus", "ask"])
+                      "minus", "ask", "path"])
+    if kind == "path":
+        return (f"SELECT (COUNT(*) AS ?c) WHERE {{ "
+                f"?e {P['worksFor']}/{P['locatedIn']} ?city }}")
     if kind == "bind":
         return (f"SELECT (COUNT(*) AS ?c) WHERE {{ ?e {P['worksFor']} ?d . "
                 f"BIND(TRIPLE(?e, {P['worksFor']}, ?d) AS ?t) . "
