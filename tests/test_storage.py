@@ -211,3 +211,14 @@ def test_reference_api_surface():
     o = db.dictionary.lookup("x")
     assert db.decode_triple((s, p, o)) == ("http://e/a", "http://e/p", "x")
     assert db.decode_triple((s, p, 999999)) is None
+
+
+def test_rdf_xml_roundtrip():
+    """generate_rdf_xml output re-parses to the same triples."""
+    db = SparqlDatabase()
+    db.add_triple("<http://e/alice>", "<http://e/knows>", "<http://e/bob>")
+    db.add_triple("<http://e/alice>", "<http://e/name>", '"Alice"')
+    xml = db.generate_rdf_xml()
+    db2 = SparqlDatabase()
+    db2.parse_rdf(xml)
+    assert sorted(db2.triples_as_strings()) == sorted(db.triples_as_strings())
