@@ -122,8 +122,16 @@ def _values_of(e: Expr, b: Bindings, db) -> torch.Tensor:
             return torch.zeros(b.n, dtype=torch.float64, device=b.device)
         ids = b.col(e.name).to(torch.int64) & 0xFFFFFFFF
         vc = db.value_column()
-        ids = torch.clamp(ids, max=vc.numel() - 1)
-        return vc[ids]
+        if vc.numel() == 0:
+            return torch.zeros(b.n, dtype=torch.float64, device=b.device)
+        # ids outside the interned vocabulary have no lexical form:
+        # value 0.0, exactly like the K5 kernel's `u < value_n` guard
+        # (clamping to the last entry borrowed ITS value — wrong)
+        in_range = ids < vc.numel()
+        vals = vc[torch.clamp(ids, max=vc.numel() - 1)]
+        return torch.where(in_range, vals,
+                           torch.zeros((), dtype=torch.float64,
+                                       device=b.device))
     if isinstance(e, ELit):
         return torch.full((b.n,), getattr(e, "num_value", 0.0),
                           dtype=torch.float64, device=b.device)

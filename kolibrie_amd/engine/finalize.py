@@ -93,7 +93,8 @@ def _order_perm(select: SelectQuery, rows: Bindings, db) -> torch.Tensor:
         all_numeric = all(_is_num(s) for s in strs) and len(strs) > 0
         if all_numeric:
             vc = db.value_column()
-            vals = vc[torch.clamp(ids_u, max=vc.numel() - 1)]
+            from .tensor_utils import values_for_ids
+            vals = values_for_ids(vc, ids_u)
             key = torch.argsort(vals, stable=True, descending=cond.descending)
         else:
             import numpy as np
@@ -167,7 +168,8 @@ def _aggregate(select: SelectQuery, rows: Bindings, db) -> Bindings:
                                               dtype=torch.int32, device=dev)
                 continue
             ids_u = col.to(torch.int64) & 0xFFFFFFFF
-            vals = vc[torch.clamp(ids_u, max=vc.numel() - 1)]
+            from .tensor_utils import values_for_ids
+            vals = values_for_ids(vc, ids_u)
             vals = vals[col != UNBOUND]
             if p.aggregate == "SUM":
                 r = float(vals.sum().item()) if vals.numel() else 0.0
@@ -224,7 +226,8 @@ def _aggregate(select: SelectQuery, rows: Bindings, db) -> Bindings:
                                         dtype=torch.int32, device=dev)
             continue
         ids_u = col.to(torch.int64) & 0xFFFFFFFF
-        vals = vc[torch.clamp(ids_u, max=vc.numel() - 1)]
+        from .tensor_utils import values_for_ids
+        vals = values_for_ids(vc, ids_u)
         bound = col != UNBOUND
         if p.aggregate in ("SUM", "AVG"):
             acc = torch.zeros(ng, dtype=torch.float64, device=dev)

@@ -183,3 +183,17 @@ def unpack2(key: torch.Tensor) -> Tuple[torch.Tensor, torch.Tensor]:
     hi = (key >> 32).to(torch.int32)
     lo = (key & 0xFFFFFFFF).to(torch.int32)
     return hi, lo
+
+
+def values_for_ids(vc, ids_u):
+    """f64 values for u32-view id tensor; ids outside the interned
+    vocabulary -> 0.0 (same guard as the K5 kernel's `u < value_n`)."""
+    import torch
+    if vc.numel() == 0:
+        return torch.zeros(ids_u.numel(), dtype=torch.float64,
+                           device=ids_u.device)
+    in_range = ids_u < vc.numel()
+    vals = vc[torch.clamp(ids_u, max=vc.numel() - 1)]
+    return torch.where(in_range, vals,
+                       torch.zeros((), dtype=torch.float64,
+                                   device=ids_u.device))

@@ -24,7 +24,8 @@ def features_from_bindings(rows: Bindings, feature_vars: Sequence[str], db
                                     device=rows.device))
             continue
         ids = rows.col(v).to(torch.int64) & 0xFFFFFFFF
-        cols.append(vc[torch.clamp(ids, max=max(0, vc.numel() - 1))])
+        from ..engine.tensor_utils import values_for_ids
+        cols.append(values_for_ids(vc, ids))
     if not cols:
         return torch.zeros((rows.n, 0), dtype=torch.float32)
     return torch.stack(cols, dim=-1).to(torch.float32)

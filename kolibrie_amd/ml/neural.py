@@ -55,7 +55,8 @@ def predict_rows(info: dict, rows, db):
     feats = []
     for v in feat_vars:
         ids = rows.col(v).to(torch.int64) & 0xFFFFFFFF
-        feats.append(vc[torch.clamp(ids, max=vc.numel() - 1)].to(torch.float32))
+        from ..engine.tensor_utils import values_for_ids
+        feats.append(values_for_ids(vc, ids).to(torch.float32))
     x = torch.stack(feats, dim=-1) if feats else torch.zeros(rows.n, 0)
     proba = model.predict_proba(x)
     out_var = info.get("output_var") or "prediction"

@@ -56,7 +56,8 @@ def _train_supervised(entry: dict, decl, db, prefixes,
     x = (x - mu) / sigma
     labels_ids = rows.col(label_var).to(torch.int64) & 0xFFFFFFFF
     vc = db.value_column()
-    y = vc[torch.clamp(labels_ids, max=vc.numel() - 1)].to(torch.float32)
+    from ..engine.tensor_utils import values_for_ids
+    y = values_for_ids(vc, labels_ids).to(torch.float32)
     hidden = [int(h) for h in
               entry.get("decl").options.get("hidden", "64,32").split(",")] \
         if entry.get("decl") is not None else [64, 32]

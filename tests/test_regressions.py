@@ -179,3 +179,26 @@ def test_rebuild_indexes_keeps_named_quads_and_empty_graphs():
     assert "http://g1" in gids and "http://gEmpty" in gids
     assert db.query('SELECT ?s WHERE { GRAPH <http://g1> { ?s ?p ?o } }') == \
         [["http://e/a"]]
+
+
+def test_out_of_vocabulary_ids_value_zero():
+    """Ids beyond the interned vocabulary (bulk synthetic loads) must
+    evaluate to 0.0 in FILTER/ORDER/aggregates — matching the K5 kernel's
+    bound check, NOT the last dictionary entry's value (regression found
+    by the CPU-vs-GPU differential sweep, seed 404)."""
+    import torch
+    from kolibrie_amd import SparqlDatabase
+    db = SparqlDatabase()
+    db.dictionary.encode("999999")  # a BIG numeric literal as last entry
+    p = db.dictionary.encode("http://e/v")
+    s = torch.tensor([5_000_000, 5_000_001], dtype=torch.int32)
+    db.store.insert_bulk(0, s, torch.full((2,), p, dtype=torch.int32),
+                         s + 10_000)  # objects also out-of-vocab
+    assert db.query(
+        'SELECT (COUNT(*) AS ?c) WHERE { ?s <http://e/v> ?o . '
+        'FILTER(?o > 1) }') == [["0"]]
+    assert db.query(
+        'SELECT (COUNT(*) AS ?c) WHERE { ?s <http://e/v> ?o . '
+        'FILTER(?o < 1) }') == [["2"]]
+    assert db.query(
+        'SELECT (SUM(?o) AS ?t) WHERE { ?s <http://e/v> ?o }') == [["0"]]
