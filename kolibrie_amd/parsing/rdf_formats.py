@@ -62,6 +62,8 @@ class _TokenStream:
         return self.toks[self.i] if self.i < len(self.toks) else None
 
     def next(self) -> str:
+        if self.i >= len(self.toks):
+            raise ValueError("unexpected end of RDF input")
         t = self.toks[self.i]
         self.i += 1
         return t

@@ -202,3 +202,19 @@ def test_out_of_vocabulary_ids_value_zero():
         'FILTER(?o < 1) }') == [["2"]]
     assert db.query(
         'SELECT (SUM(?o) AS ?t) WHERE { ?s <http://e/v> ?o }') == [["0"]]
+
+
+def test_rdf_parsers_raise_clean_errors_on_truncation():
+    """Truncated / mutated RDF inputs raise ValueError, never IndexError
+    (fuzz-found)."""
+    import pytest
+    from kolibrie_amd import SparqlDatabase
+    for text in ["<http://e/", "<< <http://e/a> <http://e/b> <http://e/",
+                 "<http://e/s> <http://e/p>"]:
+        for fn in ("parse_ntriples", "parse_turtle", "parse_nquads",
+                   "parse_n3"):
+            db = SparqlDatabase()
+            try:
+                getattr(db, fn)(text)
+            except ValueError:
+                pass  # clean parse error (or tolerated partial) is fine
