@@ -31,11 +31,43 @@ P = {
 }
 
 
+def gen_random_bgp(rng: random.Random) -> str:
+    """Fully random COUNT(*) BGP (joins/filters/optional/minus over random
+    pattern shapes incl. variable predicates) — the widest net."""
+    VARS = ["?a", "?b", "?c"]
+    preds = list(P.values()) + ["?pp"]
+
+    def pat():
+        s = rng.choice(VARS)
+        p = rng.choice(preds)
+        o = rng.choice(VARS)
+        return f"{s} {p} {o}"
+
+    # always join-connected (share ?a) — a blind cartesian at millions of
+    # rows would be a fixture bug, not an engine test
+    first = f"?a {rng.choice(preds)} ?b"
+    parts = [first]
+    if rng.random() < 0.5:
+        parts.append(f"?a {rng.choice(preds)} ?c")
+    pats = " . ".join(parts)
+    extra = ""
+    if rng.random() < 0.4:
+        extra += (f" FILTER({rng.choice(VARS)} "
+                  f"{rng.choice(['>', '<', '=', '!='])} {rng.randrange(9)})")
+    if rng.random() < 0.3:
+        extra += f" OPTIONAL {{ {pat()} }}"
+    if rng.random() < 0.25:
+        extra += f" MINUS {{ {pat()} }}"
+    return f"SELECT (COUNT(*) AS ?c) WHERE {{ {pats}{extra} }}"
+
+
 def gen_query(rng: random.Random) -> str:
     """One random query over the employee schema."""
     kind = rng.choice(["star", "chain", "filter", "optional", "union",
                       "agg", "distinct", "values", "bind", "subquery",
-                      "minus", "ask", "path"])
+                      "minus", "ask", "path", "rand", "rand"])
+    if kind == "rand":
+        return gen_random_bgp(rng)
     if kind == "path":
         return (f"SELECT (COUNT(*) AS ?c) WHERE {{ "
                 f"?e {P['worksFor']}/{P['locatedIn']} ?city }}")

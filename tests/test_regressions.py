@@ -218,3 +218,48 @@ def test_rdf_parsers_raise_clean_errors_on_truncation():
                 getattr(db, fn)(text)
             except ValueError:
                 pass  # clean parse error (or tolerated partial) is fine
+
+
+def test_execution_fuzz_deterministic():
+    """200 random (valid) query shapes execute without internal errors
+    (ValueError for user-level issues is acceptable)."""
+    import random
+    from kolibrie_amd import SparqlDatabase
+    rng = random.Random(7)
+    db = SparqlDatabase()
+    for i in range(120):
+        db.add_triple(f"<http://e/s{i%20}>", f"<http://e/p{i%4}>", f'"{i%9}"')
+        db.add_triple(f"<http://e/s{i%20}>", "<http://e/link>",
+                      f"<http://e/s{(i*7)%20}>")
+    VARS = ["?a", "?b", "?c"]
+
+    def pat():
+        s = rng.choice(VARS + ["<http://e/s3>"])
+        p = rng.choice([f"<http://e/p{rng.randrange(4)}>",
+                        "<http://e/link>", "?pp"])
+        o = rng.choice(VARS + ['"5"', "<http://e/s7>"])
+        return f"{s} {p} {o}"
+
+    for _ in range(200):
+        pats = " . ".join(pat() for _ in range(rng.randrange(1, 4)))
+        extra = ""
+        if rng.random() < 0.4:
+            extra += (f" FILTER({rng.choice(VARS)} "
+                      f"{rng.choice(['>', '<', '=', '!='])} "
+                      f"{rng.randrange(15)})")
+        if rng.random() < 0.3:
+            extra += f" OPTIONAL {{ {pat()} }}"
+        if rng.random() < 0.2:
+            extra += f" MINUS {{ {pat()} }}"
+        proj = rng.choice(["*", "?a", "?a ?b", "(COUNT(*) AS ?n)",
+                           "DISTINCT ?a"])
+        mods = ""
+        if rng.random() < 0.3:
+            mods += f" ORDER BY {rng.choice(VARS)}"
+        if rng.random() < 0.3:
+            mods += f" LIMIT {rng.randrange(5)}"
+        q = f"SELECT {proj} WHERE {{ {pats}{extra} }}{mods}"
+        try:
+            db.query(q)
+        except ValueError:
+            pass
