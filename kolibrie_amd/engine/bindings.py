@@ -7,7 +7,7 @@ implicitly deduplicated.  UNBOUND cells hold -1 (0xFFFFFFFF).
 """
 from __future__ import annotations
 
-from typing import Dict, Iterable, List, Optional, Sequence
+from typing import Dict, Iterable, List, Sequence
 
 import torch
 

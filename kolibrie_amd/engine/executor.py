@@ -14,13 +14,13 @@ matching follow engine.rs:946-1403.
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional, Sequence, Set, Tuple
+from typing import Dict, List, Optional, Sequence, Tuple
 
 import torch
 
 from ..plan.physical import (
     PBind, PBindJoin, PFilter, PHashJoin, PIndexScan, PInMemoryBuffer,
-    PLeftJoin, PMLPredict, PMinus, PNestedLoopJoin, PProjection, PStarJoin,
+    PLeftJoin, PMLPredict, PMinus, PNestedLoopJoin, PStarJoin,
     PSubquery, PTableScan, PUnion, PUnit, PValues, PhysicalOp,
 )
 from ..storage.dataset import DEFAULT_GRAPH, GraphIndex
@@ -28,7 +28,7 @@ from ..storage.terms import (
     Constant, QuotedTriplePattern, TriplePattern, UNBOUND, Variable,
 )
 from .bindings import Bindings
-from .scan import scan_probe, scan_unit
+from .scan import scan_unit
 from .tensor_utils import group_index, merge_join_indices
 from . import exec_stats
 

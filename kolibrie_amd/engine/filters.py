@@ -19,14 +19,14 @@ HIP K5 kernel consumes the same tree serialized to postfix bytecode
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Optional
 
 import torch
 
 from ..parsing.ast import (
     EAnd, EArith, ECmp, EFunc, ELit, ENot, EOr, EVar, Expr,
 )
-from ..storage.terms import UNBOUND, is_quoted_id
+from ..storage.terms import UNBOUND
 from .bindings import Bindings
 
 

@@ -16,7 +16,7 @@ dataset_index.rs:223-344 (8-way bound-pattern dispatch).
 """
 from __future__ import annotations
 
-from typing import Dict, Optional, Sequence, Tuple
+from typing import Dict, Optional, Tuple
 
 import torch
 

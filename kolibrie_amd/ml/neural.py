@@ -7,7 +7,7 @@ rows pulled from SPARQL results (ml_feature_loader.rs).
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import List, Optional
 
 import torch
 import torch.nn as nn

@@ -9,7 +9,7 @@ ML.PREDICT alias lowering (:553).
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional
+from typing import Dict
 
 import torch
 

@@ -16,7 +16,6 @@ holds the local shard in a SparqlDatabase and executes shuffle plans.
 """
 from __future__ import annotations
 
-from typing import Dict, List, Optional
 
 import torch
 

@@ -10,7 +10,7 @@ Terms inside the AST stay *surface strings* (`?x`, `<iri>`, `"lit"`,
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import Dict, List, Optional, Tuple, Union
+from typing import Dict, List, Optional
 
 
 # ------------------------------------------------------------ expressions ---

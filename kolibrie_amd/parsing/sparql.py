@@ -13,7 +13,7 @@ parse_sparql_query(text) -> SelectQuery.
 from __future__ import annotations
 
 import re
-from typing import Dict, List, Optional, Tuple
+from typing import Dict, List, Optional
 
 from .ast import (
     CombinedQuery, CombinedRule, EAnd, EArith, ECmp, EFunc, ELit, ENot, EOr,

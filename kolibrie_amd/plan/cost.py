@@ -8,7 +8,7 @@ Constants keep the reference's skeleton (scan=100/row, index=1/row with a
 distributed planner (per-link ~153 GB/s — SURVEY §2.10)."""
 from __future__ import annotations
 
-from typing import Dict, Set
+from typing import Set
 
 from ..storage.terms import Constant, TriplePattern, Variable
 from .stats import DatabaseStats

@@ -7,7 +7,7 @@ carried on each scan (reference lowers GRAPH onto scans, utils.rs:402-577).
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import List, Optional, Tuple, Union
+from typing import List, Optional, Tuple
 
 from ..storage.terms import Term, TriplePattern
 

@@ -17,7 +17,7 @@ from ..parsing.ast import (
 )
 from ..storage.terms import Constant, QuotedTriplePattern, TriplePattern, Variable
 from .logical import (
-    GraphScope, LBind, LJoin, LMLPredict, LProjection, LScan, LSelection,
+    GraphScope, LBind, LJoin, LScan, LSelection,
     LSubquery, LUnion, LUnit, LValues, LogicalOp,
 )
 

@@ -8,7 +8,7 @@ subject-star chain, Filter the K5 predicate kernel, etc.
 from __future__ import annotations
 
 from dataclasses import dataclass, field
-from typing import List, Optional, Tuple
+from typing import List, Optional
 
 from ..storage.terms import TriplePattern
 from .logical import GraphScope

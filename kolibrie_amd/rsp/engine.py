@@ -22,7 +22,7 @@ from __future__ import annotations
 
 import threading
 import time as _time
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from queue import Empty, Queue
 from typing import Callable, Dict, List, Optional, Tuple
 

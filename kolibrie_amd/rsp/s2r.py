@@ -14,7 +14,7 @@ the R2R store (the K7 device ring buffer handles the bulk-ingest path).
 from __future__ import annotations
 
 import math
-from dataclasses import dataclass, field
+from dataclasses import dataclass
 from queue import Queue
 from typing import Callable, Dict, Hashable, List, Optional, Tuple
 

@@ -8,7 +8,7 @@ literals as their lexical form, blank nodes as `_:name`
 """
 from __future__ import annotations
 
-from typing import Callable, Dict, List, Optional, Sequence, Tuple
+from typing import Callable, Dict, List, Optional, Tuple
 
 import torch
 

@@ -10,7 +10,7 @@ shared/src/quoted_triple_store.rs:17 (QUOTED_TRIPLE_ID_BIT).
 from __future__ import annotations
 
 from dataclasses import dataclass
-from typing import Optional, Tuple, Union
+from typing import Tuple, Union
 
 # ID space -------------------------------------------------------------------
 # Plain terms get IDs in [1, 2^31).  RDF-star quoted triples get IDs with bit
