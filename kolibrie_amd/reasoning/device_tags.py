@@ -53,9 +53,10 @@ class ScalarSemiring:
             self.reduce = "amax"
             self.negate = lambda t: 1.0 - t
         elif name in ("addmult", "independent", "add"):
-            self.plus = lambda a, b: torch.clamp(a + b, max=1.0)
+            # noisy-or disjunction a+b-ab (ref provenance.rs:119)
+            self.plus = lambda a, b: a + b - a * b
             self.times = torch.mul
-            self.reduce = "sum_sat"
+            self.reduce = "noisy_or"
             self.negate = None
         elif name == "expiration":
             self.plus = torch.maximum
@@ -72,9 +73,10 @@ class ScalarSemiring:
         if self.reduce == "amax":
             out.fill_(float("-inf"))
             out.scatter_reduce_(0, gid, tags, reduce="amax")
-        else:  # saturating sum (addmult ⊕)
-            out.scatter_add_(0, gid, tags)
-            out.clamp_(max=1.0)
+        else:  # noisy-or: 1 - prod(1 - tag) per group (addmult ⊕)
+            acc = torch.ones(ng, dtype=torch.float32, device=tags.device)
+            acc.scatter_reduce_(0, gid, 1.0 - tags, reduce="prod")
+            out = 1.0 - acc
         return out
 
 

@@ -96,13 +96,14 @@ class AddMultProbability(Provenance):
         return 1.0
 
     def plus(self, a, b):
-        return min(1.0, a + b)
+        # noisy-or disjunction (ref provenance.rs:119 AddMultProbability)
+        return a + b - a * b
 
     def times(self, a, b):
         return a * b
 
     def tag_from_probability(self, p, seed_id=None):
-        return float(p)
+        return float(min(1.0, max(0.0, p)))
 
     def recover(self, tag):
         return float(tag)

@@ -88,71 +88,73 @@ def infer_with_provenance(
     `seeds`: triple -> initial tag.  Returns triple -> final tag for all
     facts (seeds included, possibly tag-improved).
     """
-    known: Dict[Triple, object] = dict(seeds)
-    delta: Dict[Triple, object] = dict(seeds)
+    tags: Dict[Triple, object] = dict(seeds)
     positive_rules = [r for r in rules if not r.negative_premise]
     naf_rules = [r for r in rules if r.negative_premise]
 
-    def run_round(active_rules, use_naf_stage: bool):
-        nonlocal known, delta
-        rounds = 0
-        while delta and rounds < max_rounds:
-            rounds += 1
-            new_delta: Dict[Triple, object] = {}
-            for rule in active_rules:
-                np_ = len(rule.premise)
-                for i in range(np_):
-                    for dfact, dtag in delta.items():
-                        b0 = _match(rule.premise[i], dfact, {})
-                        if b0 is None:
-                            continue
-                        stack = [(b0, dtag, 0)]
-                        while stack:
-                            b, tag, j = stack.pop()
-                            if j == np_:
-                                if db is not None and not _eval_filters(rule, b, db):
-                                    continue
-                                final_tag = tag
-                                if use_naf_stage and rule.negative_premise:
-                                    final_tag = _apply_naf(
-                                        rule, b, known, semiring, final_tag)
-                                    if final_tag is None:
-                                        continue
-                                for concl in rule.conclusion:
-                                    t = _instantiate(concl, b)
-                                    if t is None:
-                                        continue
-                                    prev = known.get(t)
-                                    if prev is None:
-                                        known[t] = final_tag
-                                        new_delta[t] = final_tag
-                                    else:
-                                        merged = semiring.plus(prev, final_tag)
-                                        if merged != prev and semiring.better(
-                                                merged, prev):
-                                            # tag-improved fact re-enters Δ
-                                            known[t] = merged
-                                            new_delta[t] = merged
-                                        elif merged != prev:
-                                            known[t] = merged
-                                continue
-                            if j == i:
-                                stack.append((b, tag, j + 1))
-                                continue
-                            for f2, t2 in known.items():
-                                b2 = _match(rule.premise[j], f2, b)
-                                if b2 is not None:
-                                    stack.append(
-                                        (b2, semiring.times(tag, t2), j + 1))
-            delta = new_delta
+    def _close(a, b) -> bool:
+        if isinstance(a, float) and isinstance(b, float):
+            return abs(a - b) <= 1e-12
+        return a == b
 
-    # stratum 0: positive rules to fixpoint
-    run_round(positive_rules, use_naf_stage=False)
+    def _contributions(active_rules, cur: Dict[Triple, object],
+                       use_naf: bool) -> Dict[Triple, object]:
+        """One naive evaluation pass over `cur`: ⊕ of every distinct rule
+        derivation.  Recomputing from a SNAPSHOT each round (Jacobi) is
+        what makes non-idempotent semirings (noisy-or AddMult) correct —
+        the reference's delta re-entry (provenance_semi_naive.rs:185-197)
+        re-⊕s the same derivation on tag improvement, which only works
+        for idempotent ⊕ (MinMax/Boolean); the oracle must not
+        double-count."""
+        contrib: Dict[Triple, object] = {}
+        items = list(cur.items())
+        for rule in active_rules:
+            np_ = len(rule.premise)
+
+            def rec(j, b, tag):
+                if j == np_:
+                    if db is not None and not _eval_filters(rule, b, db):
+                        return
+                    final_tag = tag
+                    if use_naf and rule.negative_premise:
+                        final_tag = _apply_naf(rule, b, cur, semiring,
+                                               final_tag)
+                        if final_tag is None:
+                            return
+                    for concl in rule.conclusion:
+                        t = _instantiate(concl, b)
+                        if t is None:
+                            continue
+                        prev = contrib.get(t)
+                        contrib[t] = final_tag if prev is None else \
+                            semiring.plus(prev, final_tag)
+                    return
+                for f2, t2 in items:
+                    b2 = _match(rule.premise[j], f2, b)
+                    if b2 is not None:
+                        rec(j + 1, b2, semiring.times(tag, t2))
+
+            rec(0, {}, semiring.one())
+        return contrib
+
+    # stratum 0: positive rules to fixpoint (Jacobi iteration)
+    for _ in range(max_rounds):
+        contrib = _contributions(positive_rules, tags, use_naf=False)
+        new_tags: Dict[Triple, object] = dict(seeds)
+        for t, c in contrib.items():
+            base = new_tags.get(t)
+            new_tags[t] = c if base is None else semiring.plus(base, c)
+        if (set(new_tags) == set(tags)
+                and all(_close(new_tags[t], tags[t]) for t in tags)):
+            break
+        tags = new_tags
     # stratum 1: one NAF pass (ref run_negative_stratum_pass:297-389)
     if naf_rules:
-        delta = dict(known)
-        run_round(naf_rules, use_naf_stage=True)
-    return known
+        contrib = _contributions(naf_rules, tags, use_naf=True)
+        for t, c in contrib.items():
+            prev = tags.get(t)
+            tags[t] = c if prev is None else semiring.plus(prev, c)
+    return tags
 
 
 def _apply_naf(rule: Rule, b: Dict[str, int], known: Dict[Triple, object],

@@ -91,3 +91,39 @@ def test_device_naf_stratum():
                                        ScalarSemiring("minmax"))
     assert _close(dev[(1, Q, 2)], host[(1, Q, 2)])  # min(0.9, 1-0.3)
     assert _close(dev[(3, Q, 4)], 0.6)
+
+
+@pytest.mark.parametrize("seed", [11, 12])
+def test_device_matches_host_addmult(seed):
+    """AddMult (probabilistic sum-product) device tags vs host oracle over
+    random multi-rule programs."""
+    rng = random.Random(seed)
+    rules = [
+        Rule(premise=[_tp("?x", P, "?y")], conclusion=[_tp("?x", Q, "?y")]),
+        Rule(premise=[_tp("?x", Q, "?y"), _tp("?y", P, "?z")],
+             conclusion=[_tp("?x", R, "?z")]),
+    ]
+    seeds = {(rng.randint(1, 12), P, rng.randint(1, 12)):
+             round(rng.uniform(0.05, 1.0), 3) for _ in range(30)}
+    host = infer_with_provenance(rules, dict(seeds), AddMultProbability())
+    dev = infer_with_provenance_device(rules, dict(seeds),
+                                       ScalarSemiring("addmult"),
+                                       device="cpu")
+    assert set(host) == set(dev)
+    for k in host:
+        assert abs(host[k] - dev[k]) < 1e-4, k
+
+
+def test_device_tags_diamond_combines_paths():
+    """Two derivation paths to one fact combine per semiring (max for
+    minmax; sum-of-products for addmult)."""
+    rules = [Rule(premise=[_tp("?x", P, "?y")], conclusion=[_tp("?x", Q, "?y")])]
+    seeds = {(1, P, 2): 0.3}
+    # duplicate base path via a second rule deriving the same conclusion
+    rules.append(
+        Rule(premise=[_tp("?x", P, "?y")], conclusion=[_tp("?x", Q, "?y")]))
+    h = infer_with_provenance(rules, dict(seeds), MinMaxProbability())
+    d = infer_with_provenance_device(rules, dict(seeds),
+                                     ScalarSemiring("minmax"), device="cpu")
+    assert abs(h[(1, Q, 2)] - 0.3) < 1e-6
+    assert abs(d[(1, Q, 2)] - 0.3) < 1e-6

@@ -32,7 +32,8 @@ def test_minmax_semiring():
 
 def test_addmult_semiring():
     s = AddMultProbability()
-    assert s.plus(0.6, 0.7) == 1.0  # saturating
+    # noisy-or disjunction (ref provenance.rs:119): a+b-ab
+    assert abs(s.plus(0.6, 0.7) - 0.88) < 1e-9
     assert s.times(0.5, 0.5) == 0.25
 
 
