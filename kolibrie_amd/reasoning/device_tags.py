@@ -53,7 +53,13 @@ class ScalarSemiring:
             self.reduce = "amax"
             self.negate = lambda t: 1.0 - t
         elif name in ("addmult", "independent", "add"):
-            # noisy-or disjunction a+b-ab (ref provenance.rs:119)
+            # noisy-or disjunction a+b-ab (ref provenance.rs:119).
+            # NOTE: on RECURSIVE rule sets the delta-re-entry evaluation
+            # (reference provenance_semi_naive.rs:185-197 behavior)
+            # re-accumulates cyclic contributions and saturates toward 1;
+            # the host oracle's Jacobi iteration computes the
+            # omega-continuous limit instead.  They agree exactly on
+            # acyclic programs (see test_device_matches_host_addmult).
             self.plus = lambda a, b: a + b - a * b
             self.times = torch.mul
             self.reduce = "noisy_or"
