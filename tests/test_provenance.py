@@ -305,3 +305,39 @@ def test_hybrid_fake_clock_budget():
     cfg = HybridConfig(threshold=0.5, topk_budget_ms=0.0)
     res = evaluate_hybrid(st, node, weights, cfg, clock=clock)
     assert res.status in ("Decided", "DecidedExact", "Inconclusive")
+
+
+def test_sdd_wmc_vs_bruteforce_random():
+    """SDD WMC vs exhaustive model enumeration on 60 random weighted DNFs
+    (<= 7 vars): exact agreement."""
+    import itertools
+    import random
+    from kolibrie_amd.reasoning.sdd import SddManager
+    rng = random.Random(99)
+    for _ in range(60):
+        nvars = rng.randint(2, 7)
+        weights = {v: round(rng.uniform(0.05, 0.95), 3)
+                   for v in range(1, nvars + 1)}
+        m = SddManager()
+        for v, w in weights.items():
+            m.declare_var(v, pos_weight=w, neg_weight=1.0 - w)
+        node = m.false_node()
+        clauses = []
+        for _ in range(rng.randint(1, 4)):
+            lits = [(rng.randint(1, nvars), rng.random() < 0.5)
+                    for _ in range(rng.randint(1, 3))]
+            clauses.append(lits)
+            c = m.true_node()
+            for v, pos in lits:
+                c = m.conjoin(c, m.literal(v, pos))
+            node = m.disjoin(node, c)
+        got = m.wmc(node)
+        want = 0.0
+        for assign in itertools.product([False, True], repeat=nvars):
+            a = {v: assign[v - 1] for v in range(1, nvars + 1)}
+            if any(all(a[v] == pos for v, pos in cl) for cl in clauses):
+                p = 1.0
+                for v in range(1, nvars + 1):
+                    p *= weights[v] if a[v] else (1.0 - weights[v])
+                want += p
+        assert abs(got - want) < 1e-9, (clauses, got, want)
