@@ -379,3 +379,73 @@ def test_wmc_gradient_vs_finite_differences():
             mm, nm = build(wm)
             fd = (mp.wmc(np_) - mm.wmc(nm)) / (2 * eps)
             assert abs(grads.get(v, 0.0) - fd) < 1e-4, (trial, v)
+
+
+def test_hybrid_probability_vs_bruteforce_random():
+    """Hybrid escalation's exact probabilities vs brute-force enumeration
+    over random seed assignments (25 random 2-rule programs)."""
+    import itertools
+    import random
+    from kolibrie_amd.reasoning.hybrid import (HybridConfig,
+                                               evaluate_hybrid,
+                                               materialize_lineage)
+    from kolibrie_amd.reasoning.rule import Rule
+    from kolibrie_amd.storage.terms import Constant, TriplePattern, Variable
+
+    def _tp2(s, p, o):
+        t = lambda x: (Variable(x[1:]) if isinstance(x, str)
+                       and x.startswith("?") else Constant(x))
+        return TriplePattern(t(s), t(p), t(o))
+
+    P2, Q2, R2 = 100, 101, 102
+    rng = random.Random(8)
+    for trial in range(25):
+        rules = [
+            Rule(premise=[_tp2("?x", P2, "?y")],
+                 conclusion=[_tp2("?x", Q2, "?y")]),
+            Rule(premise=[_tp2("?x", Q2, "?y"), _tp2("?y", Q2, "?z")],
+                 conclusion=[_tp2("?x", R2, "?z")]),
+        ]
+        seeds = {(rng.randint(1, 5), P2, rng.randint(1, 5)):
+                 round(rng.uniform(0.1, 0.9), 3)
+                 for _ in range(rng.randint(2, 6))}
+        store, node_by_triple, weights = materialize_lineage(
+            rules, dict(seeds), set())
+        cfg = HybridConfig(threshold=0.5)
+        seed_list = sorted(weights)
+        for t, tag in node_by_triple.items():
+            if t in seeds or tag == -1:
+                continue
+            res = evaluate_hybrid(store, tag, weights, cfg)
+            if res.probability is None:
+                continue
+            want = 0.0
+            for assign in itertools.product([False, True],
+                                            repeat=len(seed_list)):
+                amap = dict(zip(seed_list, assign))
+                memo = {}
+
+                def ev(nid):
+                    if nid in memo:
+                        return memo[nid]
+                    kind, ch = store.nodes[nid]
+                    if kind == "true":
+                        r = True
+                    elif kind == "leaf":
+                        r = amap[ch[0]]
+                    elif kind == "and":
+                        r = all(ev(c) for c in ch)
+                    elif kind == "or":
+                        r = any(ev(c) for c in ch)
+                    else:
+                        r = not ev(ch[0])
+                    memo[nid] = r
+                    return r
+
+                if ev(tag):
+                    p = 1.0
+                    for sid, val in amap.items():
+                        w = weights[sid]
+                        p *= w if val else (1.0 - w)
+                    want += p
+            assert abs(res.probability - want) < 1e-6, (trial, t)
