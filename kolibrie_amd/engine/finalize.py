@@ -43,6 +43,15 @@ def finalize_select_bindings(select: SelectQuery, rows: Bindings, db
 
     if has_agg or select.group_by:
         rows = _aggregate(select, rows, db)
+        having = getattr(select, "having", None)
+        if having is not None and rows.n > 0:
+            # HAVING (engine extension): filter the aggregated rows; the
+            # aggregate columns are interned numerals, so the normal
+            # FILTER machinery (value column) evaluates them
+            from .filters import CompiledExpr
+            mask = CompiledExpr(having, db, dict(db.prefixes)).eval_mask(
+                rows, db)
+            rows = rows.select(mask)
     # ORDER BY first — it may sort on non-projected variables
     # (numeric-aware, ref execute_query.rs:477)
     if select.order_by and rows.n > 1:

@@ -40,6 +40,7 @@ def _top_needed(select: SelectQuery):
     if select.select_star or not select.variables:
         return None
     if (not select.distinct and not select.group_by and not select.order_by
+            and getattr(select, "having", None) is None
             and select.variables
             and all(p.aggregate == "COUNT" and p.agg_arg is None
                     and not p.distinct for p in select.variables)):
@@ -121,6 +122,7 @@ class PreparedQuery:
         # direct graph-replay shortcut in _run_prepared
         self.count_only = (
             not select.select_star and not select.group_by
+            and getattr(select, "having", None) is None
             and not select.order_by and not select.distinct
             and not getattr(select, "ask", False)
             and select.limit is None and select.offset in (None, 0)
@@ -150,6 +152,7 @@ def _count_star_fast(select: SelectQuery, rows: Bindings):
     tensor-based finalize/decode (saves ~40us/query of host time, which is
     what bounds strong scaling once the kernel shrinks with 1/N)."""
     if (rows.cols or select.group_by or select.order_by or select.distinct
+            or getattr(select, "having", None) is not None
             or select.select_star or not select.variables
             or not all(p.aggregate == "COUNT" and p.agg_arg is None
                        and not p.distinct for p in select.variables)):

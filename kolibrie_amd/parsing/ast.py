@@ -195,6 +195,7 @@ class SelectQuery:
     offset: Optional[int] = None
     select_star: bool = False
     ask: bool = False                  # ASK query (engine extension)
+    having: Optional["Expr"] = None    # HAVING expr (engine extension)
 
 
 # ------------------------------------------------------------------ update ---

@@ -81,7 +81,7 @@ KEYWORDS = {
     "CREATE", "DROP", "SILENT", "ALL", "DEFAULT", "WINDOW", "REGISTER",
     "RSTREAM", "ISTREAM", "DSTREAM", "RANGE", "TUMBLING", "SLIDING", "STEP",
     "REPORT", "TICK", "ON", "STREAM", "WITH", "POLICY", "RULE", "CONSTRUCT",
-    "NOT", "MODEL", "NEURAL", "RELATION", "TRAIN", "USING", "RETRIEVE",
+    "NOT", "HAVING", "MODEL", "NEURAL", "RELATION", "TRAIN", "USING", "RETRIEVE",
     "SOME", "EVERY", "LATENT", "ACTIVE", "PROB", "COUNT", "SUM", "AVG",
     "MIN", "MAX", "MINUS", "ASK",
 }
@@ -329,6 +329,9 @@ class Parser:
                 self.expect_kw("BY")
                 while self.peek() is not None and self.peek().kind == "var":
                     q.group_by.append(self.next().text[1:])
+            elif self.at_kw("HAVING"):
+                self.next()
+                q.having = self.parse_filter_expr()
             elif self.at_kw("ORDER"):
                 self.next()
                 self.expect_kw("BY")
@@ -804,6 +807,9 @@ class Parser:
                 self.expect_kw("BY")
                 while self.peek() is not None and self.peek().kind == "var":
                     sel.group_by.append(self.next().text[1:])
+            elif self.at_kw("HAVING"):
+                self.next()
+                sel.having = self.parse_filter_expr()
             elif self.at_kw("LIMIT"):
                 self.next()
                 sel.limit = int(self.next().text)

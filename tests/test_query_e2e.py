@@ -426,3 +426,21 @@ def test_property_path_transitive_closure():
     rows = db.query(
         f'SELECT ?t WHERE {{ <{EX}n1> <{EX}next>+ ?m . ?m <{EX}tag> ?t }}')
     assert rows == [["x"]]
+
+
+def test_having_clause():
+    """HAVING over aggregates (engine extension)."""
+    db = SparqlDatabase()
+    for i in range(10):
+        db.add_triple(f"<{EX}s{i}>", f"<{EX}grp>", f"<{EX}g{i % 3}>")
+    rows = db.query(
+        f'SELECT ?g (COUNT(*) AS ?c) WHERE {{ ?s <{EX}grp> ?g }} '
+        f'GROUP BY ?g HAVING(?c > 3) ORDER BY ?g')
+    assert rows == [[f"{EX}g0", "4"]]
+    # single-group HAVING gates the whole result
+    assert db.query(
+        f'SELECT (COUNT(*) AS ?c) WHERE {{ ?s <{EX}grp> ?g }} '
+        f'HAVING(?c > 100)') == []
+    assert db.query(
+        f'SELECT (COUNT(*) AS ?c) WHERE {{ ?s <{EX}grp> ?g }} '
+        f'HAVING(?c > 5)') == [["10"]]
