@@ -104,8 +104,10 @@ def gen_query(rng: random.Random) -> str:
         return (f"SELECT (COUNT(*) AS ?c) WHERE {{ "
                 f"{{ ?e {P[a]} ?v }} UNION {{ ?e {P[b]} ?v }} }}")
     if kind == "agg":
+        having = " HAVING(?c > 10)" if rng.random() < 0.4 else ""
         return (f"SELECT ?pos (COUNT(*) AS ?c) WHERE {{ "
-                f"?e {P['position']} ?pos }} GROUP BY ?pos ORDER BY ?pos")
+                f"?e {P['position']} ?pos }} GROUP BY ?pos{having} "
+                f"ORDER BY ?pos")
     if kind == "distinct":
         return (f"SELECT DISTINCT ?city WHERE {{ ?d {P['locatedIn']} ?city }} "
                 f"ORDER BY ?city LIMIT 50")
