@@ -511,3 +511,48 @@ def test_bulk_multi_window_join():
     flat = [r for rows in got for r in rows]
     assert any("m1" in str(r) and "20" in str(r) and "60" in str(r)
                for r in flat), flat
+
+
+def test_istream_range3_step1_sliding():
+    """ref rsp_ql_istream_range3_step1: RANGE 3 STEP 1 sliding window,
+    ISTREAM emits only the newly-arrived solutions per firing."""
+    from kolibrie_amd.rsp.builder import RSPBuilder
+    q = """
+        REGISTER ISTREAM <out> AS
+        SELECT ?s FROM NAMED WINDOW <w> ON STREAM <s1> [RANGE 3 STEP 1]
+        WHERE { WINDOW <w> { ?s <http://t/p> ?o } }
+    """
+    got = []
+    eng = (RSPBuilder().add_rsp_ql_query(q)
+           .add_consumer(lambda rows: got.append(list(rows))).build())
+    for ts in range(6):
+        eng.add_to_stream("<s1>", (f"<http://t/e{ts}>", "<http://t/p>",
+                                   "<http://t/o>"), ts)
+    flat = [v for rows in got for r in rows for v in r]
+    # each event appears as an ISTREAM addition at least once, no event
+    # is re-emitted as "new" twice in a row
+    for i in range(4):
+        assert any(f"e{i}" in v for v in flat), (i, flat)
+    for rows in got:
+        names = [v for r in rows for v in r]
+        assert len(names) == len(set(names))
+
+
+def test_istream_same_sp_diff_object_counts_both():
+    """ref rsp_ql_istream_same_sp_diff_object: two events sharing (s,p)
+    but different objects are distinct solutions."""
+    from kolibrie_amd.rsp.builder import RSPBuilder
+    q = """
+        REGISTER RSTREAM <out> AS
+        SELECT ?o FROM NAMED WINDOW <w> ON STREAM <s1> [RANGE 10 STEP 10]
+        WHERE { WINDOW <w> { <http://t/m> <http://t/p> ?o } }
+    """
+    got = []
+    eng = (RSPBuilder().add_rsp_ql_query(q)
+           .add_consumer(lambda rows: got.append(list(rows))).build())
+    eng.add_to_stream("<s1>", ("<http://t/m>", "<http://t/p>", '"a"'), 1)
+    eng.add_to_stream("<s1>", ("<http://t/m>", "<http://t/p>", '"b"'), 2)
+    for entry in eng.windows.values():
+        entry.window.flush()
+    flat = sorted(v for rows in got for r in rows for v in r)
+    assert flat == ["a", "b"]
