@@ -20,7 +20,7 @@ import torch
 
 from ..plan.physical import (
     PBind, PBindJoin, PFilter, PHashJoin, PIndexScan, PInMemoryBuffer,
-    PLeftJoin, PMLPredict, PMinus, PNestedLoopJoin, PStarJoin,
+    PConstStar, PLeftJoin, PMLPredict, PMinus, PNestedLoopJoin, PStarJoin,
     PSubquery, PTableScan, PUnion, PUnit, PValues, PhysicalOp,
 )
 from ..storage.dataset import DEFAULT_GRAPH, GraphIndex
@@ -77,6 +77,8 @@ class ExecutionEngine:
         if isinstance(op, (PTableScan, PIndexScan)):
             return self._exec_scan(op.pattern, op.graph, incoming, needed,
                                    sort_hint=getattr(op, "sort_hint", None))
+        if isinstance(op, PConstStar):
+            return self._exec_const_star(op, incoming, needed)
         if isinstance(op, PStarJoin):
             out = incoming
             for i, pat in enumerate(op.patterns):
@@ -152,6 +154,59 @@ class ExecutionEngine:
             from ..ml.predict import execute_ml_predict
             return execute_ml_predict(op.info, rows, self.db)
         raise ValueError(f"cannot execute {type(op).__name__}")
+
+    def _exec_const_star(self, op, incoming: Bindings, needed) -> Bindings:
+        """Bound-subject star: ONE SPO-region fetch + host-side pattern
+        evaluation (the region is the subject's handful of triples).  One
+        device sync total instead of two per pattern."""
+        idx = self.ctx.default_index()
+        sid = op.subject_id - 0x1_0000_0000 if op.subject_id >= 0x8000_0000 \
+            else op.subject_id
+        _s, p_col, o_col = scan_unit(idx, {0: sid}, need={1, 2})
+        n = p_col.numel()
+        vars_ = [v for _pid, v in op.items]
+        if n == 0 or n > 8192:
+            if n == 0:
+                star = Bindings.empty(self.device, vars_)
+                return star
+            # hub subject: fall back to per-pattern device masks
+            star = None
+            for pid, var in op.items:
+                pm = p_col == pid
+                b = Bindings({var: o_col[pm]}, int(pm.sum().item()),
+                             self.device)
+                star = b if star is None else join_bindings(star, b)
+            return join_bindings(incoming, _prune(star, needed), needed)
+        pl = p_col.cpu().tolist()
+        ol = o_col.cpu().tolist()
+        per = []
+        for pid, var in op.items:
+            vals = [o for pp, o in zip(pl, ol) if pp == pid]
+            if not vals:
+                return Bindings.empty(self.device, vars_)
+            per.append((var, vals))
+        # cartesian across patterns (typically 1 value each); repeated
+        # output vars constrain equality
+        rows = [{}]
+        for var, vals in per:
+            nxt = []
+            for r in rows:
+                for v in vals:
+                    if var in r:
+                        if r[var] == v:
+                            nxt.append(r)
+                    else:
+                        r2 = dict(r)
+                        r2[var] = v
+                        nxt.append(r2)
+            rows = nxt
+            if not rows:
+                return Bindings.empty(self.device, vars_)
+        uniq_vars = list(dict.fromkeys(vars_))
+        cols = {v: torch.tensor([r[v] for r in rows], dtype=torch.int32,
+                                device=self.device) for v in uniq_vars}
+        star = Bindings(cols, len(rows), self.device)
+        return join_bindings(incoming, _prune(star, needed), needed)
 
     # --------------------------------------------- fused chain-count (K1+K4)
     def _try_chain_count(self, op) -> Optional[int]:

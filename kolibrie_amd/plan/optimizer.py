@@ -28,8 +28,8 @@ from .logical import (
     LSubquery, LUnion, LUnit, LValues, LogicalOp,
 )
 from .physical import (
-    PBind, PBindJoin, PFilter, PHashJoin, PIndexScan, PLeftJoin,
-    PMLPredict, PMinus,
+    PBind, PBindJoin, PConstStar, PFilter, PHashJoin, PIndexScan,
+    PLeftJoin, PMLPredict, PMinus,
     PNestedLoopJoin, PProjection, PStarJoin, PSubquery, PTableScan, PUnion,
     PUnit, PValues, PhysicalOp,
 )
@@ -206,6 +206,22 @@ class Streamertail:
         greedy_order_scans:175) and star-first (subject-star merge chain on
         PSO slices — the MI355X-native StarJoin: sorted probes measured 2.5x
         faster than random).  Pick the cheaper chain."""
+        # const-subject stars: >=2 (Const s, Const p, Var o) patterns on
+        # the SAME subject fuse to one region fetch (PConstStar)
+        cstars, rest_scans = self._const_star_groups(scans)
+        if cstars:
+            cur_bound = set(bound)
+            for op_star, ovars in cstars:
+                cur_bound.update(ovars)
+            if rest_scans:
+                rest_plan, rrows, rcost = self._plan_scan_group(
+                    rest_scans, cur_bound)
+            else:
+                rest_plan, rrows, rcost = None, 1.0, 0.0
+            plan = rest_plan
+            for op_star, _ovars in reversed(cstars):
+                plan = op_star if plan is None else PBindJoin(op_star, plan)
+            return plan, max(1.0, rrows), rcost + 50.0 * len(cstars)
         candidates = [self._greedy_order(scans, bound)]
         star = self._star_subgroup(scans, bound)
         if star is not None and star != candidates[0]:
@@ -217,6 +233,30 @@ class Streamertail:
                 best = plan
         assert best is not None
         return best
+
+    def _const_star_groups(self, scans: List[LScan]):
+        """Extract (PConstStar, out_vars) fusions for default-graph
+        (Const s, Const p, Var o) patterns grouped by subject id."""
+        from collections import defaultdict
+        groups = defaultdict(list)
+        rest: List[LScan] = []
+        for s in scans:
+            pat = s.pattern
+            if (s.graph is None and isinstance(pat.s, Constant)
+                    and isinstance(pat.p, Constant)
+                    and isinstance(pat.o, Variable)):
+                groups[pat.s.id].append(s)
+            else:
+                rest.append(s)
+        out = []
+        for sid, ss in groups.items():
+            if len(ss) >= 2:
+                items = tuple((s.pattern.p.id, s.pattern.o.name) for s in ss)
+                out.append((PConstStar(sid, items),
+                            {s.pattern.o.name for s in ss}))
+            else:
+                rest.extend(ss)
+        return out, rest
 
     def _star_subgroup(self, scans: List[LScan], bound: Set[str]
                        ) -> Optional[List[LScan]]:
@@ -372,6 +412,8 @@ def phys_out_vars(op: PhysicalOp) -> Set[str]:
         if op.graph is not None and op.graph[0] == "var":
             out.add(op.graph[1])
         return out
+    if isinstance(op, PConstStar):
+        return {v for _pid, v in op.items}
     if isinstance(op, (PHashJoin, PBindJoin, PNestedLoopJoin, PUnion, PMinus,
                        PLeftJoin)):
         return phys_out_vars(op.left) | phys_out_vars(op.right)

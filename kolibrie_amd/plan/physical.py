@@ -119,6 +119,16 @@ class PMinus(PhysicalOp):
 
 
 @dataclass
+class PConstStar(PhysicalOp):
+    """Star of >=2 (CONST subject, CONST predicate, ?var) patterns: fetch
+    the subject's SPO region once and evaluate every pattern host-side in
+    one pass (1 device sync instead of 2 per pattern — the S4-class
+    point-query shape)."""
+    subject_id: int = 0
+    items: tuple = ()         # ((pid_i32, out_var), ...)
+
+
+@dataclass
 class PLeftJoin(PhysicalOp):
     """OPTIONAL: inner join plus unmatched left rows padded UNBOUND."""
     left: PhysicalOp = field(default_factory=PUnit)
