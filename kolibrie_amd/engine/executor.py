@@ -145,7 +145,14 @@ class ExecutionEngine:
             left = self.execute(op.left, incoming)
             if left.is_empty():
                 return left
-            right = self.execute(op.right, Bindings.unit(self.device))
+            if not left.maybe_unbound and left.n <= 5_000_000:
+                # semi-join pushdown: evaluate the right side PROBED by the
+                # left rows instead of scanning it in full — the probed
+                # result carries every left column, so anti-join
+                # compatibility reduces to "this left row found a match"
+                right = self.execute(op.right, left)
+            else:
+                right = self.execute(op.right, Bindings.unit(self.device))
             return anti_join(left, right)
         if isinstance(op, PInMemoryBuffer):
             return join_bindings(incoming, op.bindings)
