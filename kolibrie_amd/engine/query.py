@@ -104,6 +104,80 @@ def prepare_extensions(cq: CombinedQuery, db, prefixes: Dict[str, str]):
         db.rules.append(rule)
 
 
+def _detect_group_count(select: SelectQuery, physical):
+    """(?s P ?o) grouped by ?s or ?o with COUNT(*) only: the sorted
+    PSO/POS region yields groups+counts in ONE unique_consecutive pass
+    (no query-time sort).  Returns (pid, group_pos, names) or None."""
+    from ..plan.physical import PIndexScan, PTableScan
+    from ..storage.terms import Constant, Variable
+    if (select.select_star or select.distinct or select.limit is not None
+            or select.offset not in (None, 0)
+            or getattr(select, "having", None) is not None
+            or getattr(select, "ask", False)):
+        return None
+    if len(select.group_by) != 1 or not select.variables:
+        return None
+    if not isinstance(physical, (PIndexScan, PTableScan)) \
+            or physical.graph is not None:
+        return None
+    pat = physical.pattern
+    if not (isinstance(pat.s, Variable) and isinstance(pat.p, Constant)
+            and isinstance(pat.o, Variable) and pat.s.name != pat.o.name):
+        return None
+    g = select.group_by[0]
+    if g not in (pat.s.name, pat.o.name):
+        return None
+    names = []
+    for p in select.variables:
+        if p.aggregate == "COUNT" and p.agg_arg is None and not p.distinct:
+            names.append(("count", p.output_name()))
+        elif p.aggregate is None and p.var == g:
+            names.append(("group", p.output_name()))
+        else:
+            return None
+    if select.order_by and any(c.var not in [n for _, n in names]
+                               for c in select.order_by):
+        return None
+    return (pat.p.id & 0xFFFFFFFF, 0 if g == pat.s.name else 2, tuple(names))
+
+
+def _run_group_count(pq: "PreparedQuery", db) -> Optional[List[List[str]]]:
+    import torch
+    from ..storage.dataset import DEFAULT_GRAPH, POS, PSO
+    from ..parsing.ast import Projection
+    from dataclasses import replace
+    pid, gpos, names = pq.group_pushdown
+    idx = db.store.graph_index(DEFAULT_GRAPH)
+    if idx.n == 0:
+        return []
+    code = PSO if gpos == 0 else POS
+    key12, _z = idx.orders[code]
+    pid_i32 = pid - 0x1_0000_0000 if pid >= 0x8000_0000 else pid
+    k = pid_i32 << 32
+    probe = torch.tensor([k, k + 0x1_0000_0000], dtype=torch.int64,
+                         device=key12.device)
+    lo, hi = torch.searchsorted(key12, probe, side="left").tolist()
+    if hi <= lo:
+        return []
+    vals, counts = torch.unique_consecutive(
+        key12[lo:hi] & 0xFFFFFFFF, return_counts=True)
+    if vals.numel() > 100_000:
+        return None  # fall back to the generic path
+    cnt_ids = torch.tensor(
+        [db.dictionary.encode(str(int(c))) for c in counts.cpu().tolist()],
+        dtype=torch.int32, device=key12.device)
+    gcol = vals.to(torch.int32)
+    cols = {}
+    for kind, name in names:
+        cols[name] = gcol if kind == "group" else cnt_ids
+    rows = Bindings(cols, vals.numel(), db.device)
+    sel2 = replace(pq.select, variables=[Projection(var=n)
+                                         for _k, n in names],
+                   group_by=[], having=None)
+    final = finalize_select_bindings(sel2, rows, db)
+    return decode_rows(sel2, final, db)
+
+
 class PreparedQuery:
     """A parsed+planned SELECT, reusable across executions (plan cache).
 
@@ -111,7 +185,8 @@ class PreparedQuery:
     decisions from the stats snapshot, so entries key on the store version.
     """
 
-    __slots__ = ("select", "physical", "view", "store_version", "count_only")
+    __slots__ = ("select", "physical", "view", "store_version", "count_only",
+                 "group_pushdown")
 
     def __init__(self, select, physical, view, store_version):
         self.select = select
@@ -130,6 +205,7 @@ class PreparedQuery:
             and select.variables[0].aggregate == "COUNT"
             and select.variables[0].agg_arg is None
             and not select.variables[0].distinct)
+        self.group_pushdown = _detect_group_count(select, physical)
 
 
 def _prepare_select(select: SelectQuery, db, prefixes) -> "PreparedQuery":
@@ -163,6 +239,11 @@ def _count_star_fast(select: SelectQuery, rows: Bindings):
 
 
 def _run_prepared(pq: "PreparedQuery", db) -> List[List[str]]:
+    if pq.group_pushdown is not None \
+            and pq.view.default_graphs == [DEFAULT_GRAPH]:
+        fast = _run_group_count(pq, db)
+        if fast is not None:
+            return fast
     if pq.count_only:
         # serving hot loop: replay the captured hipGraph and read the
         # 8-byte total — no engine construction, no plan walk

@@ -486,3 +486,26 @@ def test_const_subject_star_fusion():
                 walk(getattr(x, attr))
     walk(plan)
     assert any(found)
+
+
+def test_group_count_pushdown_matches_generic():
+    """Sorted-region GROUP BY pushdown (cached path) vs the generic
+    aggregate path (uncached execute_select)."""
+    from kolibrie_amd.engine.query import execute_select
+    from kolibrie_amd.parsing.sparql import parse_combined_query
+    db = SparqlDatabase()
+    for i in range(40):
+        db.add_triple(f"<{EX}s{i}>", f"<{EX}grp>", f"<{EX}g{i % 7}>")
+        db.add_triple(f"<{EX}s{i}>", f"<{EX}other>", '"x"')
+    q = (f'SELECT ?g (COUNT(*) AS ?c) WHERE {{ ?s <{EX}grp> ?g }} '
+         f'GROUP BY ?g ORDER BY ?g')
+    cached = db.query(q)
+    cq = parse_combined_query(q)
+    generic = execute_select(cq.select, db, dict(db.prefixes))
+    assert cached == generic
+    assert len(cached) == 7 and cached[0][1] in ("5", "6")
+    # group by SUBJECT side too
+    q2 = (f'SELECT ?s (COUNT(*) AS ?c) WHERE {{ ?s <{EX}grp> ?o }} '
+          f'GROUP BY ?s ORDER BY ?s')
+    cq2 = parse_combined_query(q2)
+    assert db.query(q2) == execute_select(cq2.select, db, dict(db.prefixes))
