@@ -946,6 +946,16 @@ def anti_join(left: Bindings, right: Bindings) -> Bindings:
     shared = [v for v in left.variables if v in right.cols]
     if not shared or right.is_empty():
         return left
+    if (not left.maybe_unbound and not right.maybe_unbound
+            and left.n * right.n <= 4096):
+        # tiny-tables fast path (same shape as join_bindings'): one
+        # broadcast compare instead of unique+membership machinery
+        hit = left.col(shared[0]).unsqueeze(1) == \
+            right.col(shared[0]).unsqueeze(0)
+        for v in shared[1:]:
+            hit = hit & (left.col(v).unsqueeze(1) ==
+                         right.col(v).unsqueeze(0))
+        return left.select(~hit.any(dim=1))
     keep = torch.ones(left.n, dtype=torch.bool, device=dev)
     # vectorize the common fully-bound case; fall back to NLJ for unbound
     lb = torch.ones(left.n, dtype=torch.bool, device=dev)
