@@ -64,7 +64,7 @@ _TOK = re.compile(
     | (?P<var>[?$][A-Za-z_][\w]*)
     | (?P<number>[+-]?(\d+\.\d*|\.\d+|\d+)([eE][+-]?\d+)?)
     | (?P<ruleop>:-)
-    | (?P<op><=|>=|!=|&&|\|\||[=<>!+\-*/^])
+    | (?P<op><=|>=|!=|&&|\|\||[=<>!+\-*/^|])
     | (?P<punct>[{}()\[\],;.])
     | (?P<bnode>_:[A-Za-z0-9_.-]+)
     | (?P<pname>[A-Za-z_][\w.-]*:[\w.-]*|:[\w.-]+)
@@ -393,7 +393,9 @@ class Parser:
             t = self.peek()
             if t is None:
                 raise self.err("unterminated group")
-            if t.text == "{":
+            if False:
+                pass
+            elif t.text == "{":
                 # nested group or subquery, possibly UNION chain
                 node = self.parse_group_or_subquery()
                 while self.at_kw("UNION"):
@@ -440,7 +442,12 @@ class Parser:
                 self.next()
             else:
                 bgp = self.parse_triples_block()
-                current = self._join(current, bgp)
+                node: GGP = bgp
+                for var, alts in getattr(self, "_alt_values", []):
+                    node = GValues([var], [[a] for a in alts], node)
+                if getattr(self, "_alt_values", None):
+                    self._alt_values = []
+                current = self._join(current, node)
         self.expect("}")
         for f in filters:
             current = GFilter(f, current)
@@ -476,7 +483,21 @@ class Parser:
             if self.at("^"):
                 self.next()
                 inv = True
-            pred = self.parse_term()
+            if self.at("("):
+                # alternatives (p1|p2|...): desugar to a variable predicate
+                # constrained by VALUES (engine extension)
+                self.next()
+                alts = [self.parse_term()]
+                while self.at("|"):
+                    self.next()
+                    alts.append(self.parse_term())
+                self.expect(")")
+                pred = self._fresh_path_var()
+                if not hasattr(self, "_alt_values"):
+                    self._alt_values = []
+                self._alt_values.append((pred[1:], alts))
+            else:
+                pred = self.parse_term()
             mod = None
             if self.at("+") or self.at("*"):
                 mod = self.next().text

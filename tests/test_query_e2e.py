@@ -509,3 +509,19 @@ def test_group_count_pushdown_matches_generic():
           f'GROUP BY ?s ORDER BY ?s')
     cq2 = parse_combined_query(q2)
     assert db.query(q2) == execute_select(cq2.select, db, dict(db.prefixes))
+
+
+def test_property_path_alternatives():
+    """(p1|p2) alternatives (engine extension): desugared to a
+    VALUES-constrained variable predicate; composes with sequences."""
+    db = SparqlDatabase()
+    db.add_triple(f"<{EX}a>", f"<{EX}p1>", f"<{EX}x>")
+    db.add_triple(f"<{EX}a>", f"<{EX}p2>", f"<{EX}y>")
+    db.add_triple(f"<{EX}a>", f"<{EX}p3>", f"<{EX}z>")
+    db.add_triple(f"<{EX}x>", f"<{EX}q>", f"<{EX}deep>")
+    rows = db.query(
+        f'SELECT ?o WHERE {{ <{EX}a> (<{EX}p1>|<{EX}p2>) ?o }}')
+    assert sorted(r[0] for r in rows) == [f"{EX}x", f"{EX}y"]
+    rows = db.query(
+        f'SELECT ?o WHERE {{ <{EX}a> (<{EX}p1>|<{EX}p2>)/<{EX}q> ?o }}')
+    assert rows == [[f"{EX}deep"]]
