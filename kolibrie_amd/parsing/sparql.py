@@ -393,9 +393,7 @@ class Parser:
             t = self.peek()
             if t is None:
                 raise self.err("unterminated group")
-            if False:
-                pass
-            elif t.text == "{":
+            if t.text == "{":
                 # nested group or subquery, possibly UNION chain
                 node = self.parse_group_or_subquery()
                 while self.at_kw("UNION"):
