@@ -411,7 +411,7 @@ class ExecutionEngine:
             return self._scan_index(idx, pattern, incoming, extra=None,
                                     needed=needed, sort_hint=sort_hint)
         if scope[0] == "closure":
-            idx = self._closure_index(scope[1], scope[2])
+            idx = self._closure_index(scope[1], scope[2], scope[3])
             return self._scan_index(idx, pattern, incoming, extra=None,
                                     needed=needed, sort_hint=sort_hint)
         # GRAPH ?g — iterate named graphs, bind the graph variable
@@ -430,7 +430,8 @@ class ExecutionEngine:
             return Bindings.empty(self.device, all_vars)
         return Bindings.concat(parts, self.device)
 
-    def _closure_index(self, pid: int, reflexive: bool) -> GraphIndex:
+    def _closure_index(self, pid: int, reflexive: bool,
+                       gid: Optional[int] = None) -> GraphIndex:
         """Transitive closure of predicate `pid` as an auxiliary
         GraphIndex, materialized by log-doubling joins on device and
         cached per store version (p+ / p* property paths).  `*` adds the
@@ -438,14 +439,15 @@ class ExecutionEngine:
         cache = getattr(self.db, "_closures", None)
         if cache is None:
             cache = self.db._closures = {}
-        key = (pid, reflexive)
+        key = (pid, reflexive, gid)
         hit = cache.get(key)
         if hit is not None and hit[0] == self.db.store.version:
             return hit[1]
         from .scan import scan_unit
         from .tensor_utils import pack2, unique_rows
         pid_i32 = pid - 0x1_0000_0000 if pid >= 0x8000_0000 else pid
-        base = self.ctx.default_index()
+        base = self.ctx.default_index() if gid is None \
+            else self.db.store.graph_index(gid)
         s, _p, o = scan_unit(base, {1: pid_i32})
         dev = self.device
         cs, co = s, o

@@ -89,11 +89,13 @@ def build_logical_plan(
                 if not isinstance(cp.p, Constant):
                     raise ValueError(
                         "property-path closure requires a constant predicate")
-                if scope is not None:
+                if scope is not None and scope[0] != "const":
                     raise ValueError(
-                        "property-path closure inside GRAPH is unsupported")
+                        "property-path closure under GRAPH ?var is "
+                        "unsupported")
+                gid = None if scope is None else scope[1] & 0xFFFFFFFF
                 pat_scope = ("closure", cp.p.id & 0xFFFFFFFF,
-                             pat.path_mod == "*")
+                             pat.path_mod == "*", gid)
             scan = LScan(cp, pat_scope)
             node = scan if isinstance(node, LUnit) else LJoin(node, scan)
         return node

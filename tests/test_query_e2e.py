@@ -525,3 +525,14 @@ def test_property_path_alternatives():
     rows = db.query(
         f'SELECT ?o WHERE {{ <{EX}a> (<{EX}p1>|<{EX}p2>)/<{EX}q> ?o }}')
     assert rows == [[f"{EX}deep"]]
+
+
+def test_property_path_closure_in_named_graph():
+    """p+ closure scoped to GRAPH <iri>: only that graph's edges close."""
+    db = SparqlDatabase()
+    db.query(f'INSERT DATA {{ GRAPH <{EX}g> {{ <{EX}a> <{EX}n> <{EX}b> . '
+             f'<{EX}b> <{EX}n> <{EX}c> }} }}')
+    db.add_triple(f"<{EX}c>", f"<{EX}n>", f"<{EX}d>")  # default graph only
+    rows = db.query(
+        f'SELECT ?o WHERE {{ GRAPH <{EX}g> {{ <{EX}a> <{EX}n>+ ?o }} }}')
+    assert sorted(r[0] for r in rows) == [f"{EX}b", f"{EX}c"]
