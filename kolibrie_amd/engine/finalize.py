@@ -281,6 +281,10 @@ def decode_rows(select: SelectQuery, rows: Bindings, db) -> List[List[str]]:
             host[v] = (rows.col(v).to(torch.int64) & 0xFFFFFFFF).cpu().tolist()
         else:
             host[v] = [None] * rows.n
+    # plain (non-quoted, interned) ids decode by direct list index — the
+    # per-cell function call dominates row-heavy results otherwise
+    id2s = db.dictionary.id_to_str
+    n_plain = len(id2s)
     out: List[List[str]] = []
     for i in range(rows.n):
         row = []
@@ -288,6 +292,8 @@ def decode_rows(select: SelectQuery, rows: Bindings, db) -> List[List[str]]:
             x = host[v][i]
             if x is None or x == 0xFFFFFFFF:
                 row.append("")
+            elif x < n_plain and not (x & 0x8000_0000):
+                row.append(id2s[x] or "")
             else:
                 row.append(db.decode_term(x) or "")
         out.append(row)
