@@ -439,7 +439,7 @@ class Parser:
             elif t.text == ".":
                 self.next()
             else:
-                bgp = self.parse_triples_block()
+                bgp = self.parse_triples_block(allow_alts=True)
                 node: GGP = bgp
                 for var, alts in getattr(self, "_alt_values", []):
                     node = GValues([var], [[a] for a in alts], node)
@@ -510,7 +510,11 @@ class Parser:
         Parser._path_var_counter[0] += 1
         return f"?__pp{Parser._path_var_counter[0]}"
 
-    def parse_triples_block(self) -> GBgp:
+    def parse_triples_block(self, allow_alts: bool = False) -> GBgp:
+        """allow_alts: only parse_group consumes the (p1|p2) VALUES
+        desugaring; every other caller (rule conclusions, ML inputs,
+        CONSTRUCT/DESCRIBE templates) must reject alternatives instead of
+        leaking the pending constraint into a later group."""
         pats: List[TriplePatternAst] = []
         s = self.parse_term()
         while True:
@@ -538,6 +542,10 @@ class Parser:
             break
         if self.at("."):
             self.next()
+        if not allow_alts and getattr(self, "_alt_values", None):
+            self._alt_values = []
+            raise self.err("path alternatives (p|q) are only supported in "
+                           "WHERE groups")
         return GBgp(pats)
 
     def parse_values_body(self):
