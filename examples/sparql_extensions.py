@@ -32,7 +32,7 @@ def main():
           db.query(f'SELECT ?c WHERE {{ <{EX}alice> '
                    f'<{EX}worksFor>/<{EX}locatedIn> ?c }}'))
     print("inverse:",
-          db.query(f'SELECT ?e WHERE {{ ?e ^<{EX}locatedIn> <{EX}berlin> }}'))
+          db.query(f'SELECT ?e WHERE {{ <{EX}berlin> ^<{EX}locatedIn> ?e }}'))
     print("closure+:",
           sorted(db.query(f'SELECT ?x WHERE {{ <{EX}alice> <{EX}knows>+ ?x }}')))
     print("alts:",
