@@ -536,3 +536,19 @@ def test_property_path_closure_in_named_graph():
     rows = db.query(
         f'SELECT ?o WHERE {{ GRAPH <{EX}g> {{ <{EX}a> <{EX}n>+ ?o }} }}')
     assert sorted(r[0] for r in rows) == [f"{EX}b", f"{EX}c"]
+
+
+def test_group_pushdown_respects_from_dataset():
+    """GROUP BY pushdown must not bypass FROM dataset views."""
+    db = SparqlDatabase()
+    db.query(f'INSERT DATA {{ GRAPH <{EX}gX> {{ <{EX}s1> <{EX}grp> <{EX}g1> . '
+             f'<{EX}s2> <{EX}grp> <{EX}g1> }} }}')
+    db.add_triple(f"<{EX}s3>", f"<{EX}grp>", f"<{EX}g2>")
+    q = (f'SELECT ?g (COUNT(*) AS ?c) FROM <{EX}gX> WHERE '
+         f'{{ ?s <{EX}grp> ?g }} GROUP BY ?g ORDER BY ?g')
+    assert db.query(q) == [[f"{EX}g1", "2"]]
+    assert db.query(q) == [[f"{EX}g1", "2"]]  # cached run identical
+    # default-graph query unaffected
+    q2 = (f'SELECT ?g (COUNT(*) AS ?c) WHERE {{ ?s <{EX}grp> ?g }} '
+          f'GROUP BY ?g ORDER BY ?g')
+    assert db.query(q2) == [[f"{EX}g2", "1"]]
