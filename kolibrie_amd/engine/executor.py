@@ -145,11 +145,24 @@ class ExecutionEngine:
             left = self.execute(op.left, incoming)
             if left.is_empty():
                 return left
-            if not left.maybe_unbound and left.n <= 5_000_000:
+            from ..plan.physical import op_certain_vars, op_possible_vars
+            possible = op_possible_vars(op.right)
+            if possible is not None and not any(v in left.cols
+                                                for v in possible):
+                # SPARQL MINUS with statically disjoint domains removes
+                # nothing — skip evaluating the right side entirely
+                return left
+            certain = op_certain_vars(op.right)
+            if (not left.maybe_unbound and left.n <= 5_000_000
+                    and any(v in left.cols for v in certain)):
                 # semi-join pushdown: evaluate the right side PROBED by the
                 # left rows instead of scanning it in full — the probed
                 # result carries every left column, so anti-join
-                # compatibility reduces to "this left row found a match"
+                # compatibility reduces to "this left row found a match".
+                # Sound only because (a) every left shared var is bound
+                # (maybe_unbound is False) and (b) every right solution
+                # binds >=1 var shared with the left (certain-vars check),
+                # so the spec's dom-intersection requirement always holds.
                 right = self.execute(op.right, left)
             else:
                 right = self.execute(op.right, Bindings.unit(self.device))
@@ -928,18 +941,39 @@ def _compat_nlj(left: Bindings, right: Bindings, shared: Sequence[str],
     return _merge_pairs(left, right, li, ri, shared, needed)
 
 
+_LOJ_ROWVAR = "\x00loj_row"  # internal row-id column (never a user var)
+
+
 def left_outer_join(left: Bindings, right: Bindings,
                     needed=None) -> Bindings:
     """OPTIONAL: every left row survives; matched rows extend with right
     columns, unmatched rows pad the right-only columns UNBOUND (SPARQL
-    left outer join; engine extension beyond the reference subset)."""
+    left outer join; engine extension beyond the reference subset).
+
+    Matched-ness is tracked by a hidden per-row id column carried through
+    the inner join, so a left row is padded iff it produced NO joined row —
+    correct even when the shared-variable set is empty (cartesian: every
+    left row matches whenever right is non-empty) or left shared vars are
+    UNBOUND (compat semantics), where MINUS-style anti_join would
+    disagree."""
     if right.is_empty():
         return left
-    inner = join_bindings(left, right, needed)
-    unmatched = anti_join(left, right)
-    if unmatched.is_empty():
+    dev = left.device
+    rowid = torch.arange(left.n, dtype=torch.int32, device=dev)
+    inner = join_bindings(left.with_col(_LOJ_ROWVAR, rowid), right, None)
+    if inner.is_empty():
+        return left
+    matched = torch.zeros(left.n, dtype=torch.bool, device=dev)
+    matched[inner.col(_LOJ_ROWVAR).long()] = True
+    inner = inner.project([v for v in inner.variables if v != _LOJ_ROWVAR])
+    if needed is not None:
+        inner = _prune(inner, needed)
+    if bool(matched.all().item()):
         return inner
-    return Bindings.concat([inner, unmatched], left.device)
+    unmatched = left.select(~matched)
+    if needed is not None:
+        unmatched = _prune(unmatched, needed)
+    return Bindings.concat([inner, unmatched], dev)
 
 
 def anti_join(left: Bindings, right: Bindings) -> Bindings:
@@ -959,19 +993,37 @@ def anti_join(left: Bindings, right: Bindings) -> Bindings:
                          right.col(v).unsqueeze(0))
         return left.select(~hit.any(dim=1))
     keep = torch.ones(left.n, dtype=torch.bool, device=dev)
-    # vectorize the common fully-bound case; fall back to NLJ for unbound
-    lb = torch.ones(left.n, dtype=torch.bool, device=dev)
-    for v in shared:
-        lb &= left.col(v) != UNBOUND
+    # vectorize by left boundness signature; fall back to NLJ for
+    # right-side unbound rows
     rb = torch.ones(right.n, dtype=torch.bool, device=dev)
     for v in shared:
         rb &= right.col(v) != UNBOUND
     r_keyed = right.select(rb)
     if r_keyed.n:
         from .tensor_utils import membership_mask, unique_rows
-        rcols = unique_rows([r_keyed.col(v) for v in shared])
-        hit = membership_mask([left.col(v) for v in shared], rcols)
-        keep &= ~(hit & lb)
+        if not left.maybe_unbound:
+            # common case: every left shared var bound -> one membership
+            rcols = unique_rows([r_keyed.col(v) for v in shared])
+            hit = membership_mask([left.col(v) for v in shared], rcols)
+            keep &= ~hit
+        else:
+            # group left rows by which shared vars they bind: a row with a
+            # partially-unbound key is still removed when a keyed right
+            # row agrees on its BOUND shared vars (>=1 required — SPARQL
+            # dom-intersection rule); the membership key is restricted to
+            # that signature's bound vars
+            sig = torch.zeros(left.n, dtype=torch.long, device=dev)
+            for i, v in enumerate(shared):
+                sig |= (left.col(v) != UNBOUND).long() << i
+            for s in torch.unique(sig).tolist():
+                if s == 0:
+                    continue  # no bound shared var -> spec removes nothing
+                vars_b = [v for i, v in enumerate(shared) if s >> i & 1]
+                rows = (sig == s).nonzero(as_tuple=True)[0]
+                rcols = unique_rows([r_keyed.col(v) for v in vars_b])
+                hit = membership_mask(
+                    [left.col(v)[rows] for v in vars_b], rcols)
+                keep[rows[hit]] = False
     r_unkeyed = right.select(~rb)
     if r_unkeyed.n:
         for j in range(r_unkeyed.n):

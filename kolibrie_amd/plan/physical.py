@@ -141,6 +141,96 @@ class PInMemoryBuffer(PhysicalOp):
     bindings: object = None
 
 
+def _graph_var(graph: GraphScope):
+    if graph is not None and graph[0] == "var":
+        return graph[1]
+    return None
+
+
+def op_certain_vars(op: PhysicalOp) -> set:
+    """Variables bound in EVERY solution of the (sub)plan — the static
+    domain analysis MINUS needs (SPARQL spec: MINUS removes a left row only
+    when a compatible right row shares >=1 *bound* variable with it; a
+    disjoint-domain MINUS removes nothing).  Conservative: returns a subset
+    of the true certain set; unknown node kinds contribute nothing."""
+    if isinstance(op, (PTableScan, PIndexScan)):
+        vs = set(op.pattern.variables())
+        gv = _graph_var(op.graph)
+        if gv is not None:
+            vs.add(gv)
+        return vs
+    if isinstance(op, PStarJoin):
+        vs = {op.join_var}
+        for pat in op.patterns:
+            vs |= set(pat.variables())
+        gv = _graph_var(op.graph)
+        if gv is not None:
+            vs.add(gv)
+        return vs
+    if isinstance(op, PConstStar):
+        return {v for _p, v in op.items}
+    if isinstance(op, (PHashJoin, PBindJoin, PNestedLoopJoin)):
+        return op_certain_vars(op.left) | op_certain_vars(op.right)
+    if isinstance(op, PUnion):
+        return op_certain_vars(op.left) & op_certain_vars(op.right)
+    if isinstance(op, (PLeftJoin, PMinus)):
+        return op_certain_vars(op.left)
+    if isinstance(op, PFilter):
+        return op_certain_vars(op.input)
+    if isinstance(op, PBind):
+        base = op_certain_vars(op.input)
+        if not getattr(op.expr, "may_produce_unbound", False):
+            base = base | {op.var}
+        return base
+    if isinstance(op, PValues):
+        base = op_certain_vars(op.input)
+        if op.rows:
+            for j, v in enumerate(op.variables):
+                if all(r[j] is not None for r in op.rows):
+                    base = base | {v}
+        return base
+    if isinstance(op, PProjection):
+        return op_certain_vars(op.input) & set(op.variables)
+    if isinstance(op, PInMemoryBuffer) and op.bindings is not None:
+        b = op.bindings
+        if not getattr(b, "maybe_unbound", True):
+            return set(b.variables)
+        return set()
+    return set()
+
+
+def op_possible_vars(op: PhysicalOp):
+    """Superset of the variables the (sub)plan can ever bind, or None when
+    the node kind is opaque (subquery/ML) — callers must treat None as
+    "could bind anything"."""
+    if isinstance(op, (PTableScan, PIndexScan, PStarJoin, PConstStar)):
+        return op_certain_vars(op)
+    if isinstance(op, (PHashJoin, PBindJoin, PNestedLoopJoin, PUnion,
+                       PLeftJoin)):
+        l = op_possible_vars(op.left)
+        r = op_possible_vars(op.right)
+        if l is None or r is None:
+            return None
+        return l | r
+    if isinstance(op, PMinus):
+        return op_possible_vars(op.left)
+    if isinstance(op, PFilter):
+        return op_possible_vars(op.input)
+    if isinstance(op, PBind):
+        base = op_possible_vars(op.input)
+        return None if base is None else base | {op.var}
+    if isinstance(op, PValues):
+        base = op_possible_vars(op.input)
+        return None if base is None else base | set(op.variables)
+    if isinstance(op, PProjection):
+        return set(op.variables)
+    if isinstance(op, PInMemoryBuffer) and op.bindings is not None:
+        return set(op.bindings.variables)
+    if isinstance(op, PUnit):
+        return set()
+    return None
+
+
 def plan_key(op: PhysicalOp) -> str:
     """Serialize a plan to a memo key (ref optimizer.rs:751-848)."""
     if isinstance(op, PUnit):

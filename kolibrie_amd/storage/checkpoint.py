@@ -43,7 +43,7 @@ def save_binary(db, path: str, rank: int = 0):
         arrays[f"g{g}_o"] = o.cpu().numpy()
         graphs.append(g)
     meta = {
-        "version": 1,
+        "version": 2,
         "rank": rank,
         "graphs": graphs,
         "catalog": sorted(db.store.catalog),
@@ -53,11 +53,37 @@ def save_binary(db, path: str, rank: int = 0):
     with open(path + ".meta.json", "w", encoding="utf-8") as f:
         json.dump(meta, f)
     with open(path + ".dict", "w", encoding="utf-8") as f:
+        # one JSON string per line: robust against \n, \r, backslashes and
+        # any other control characters inside terms (v1's ad-hoc escaping
+        # corrupted strings containing '\r' or literal "\n")
         for s in db.dictionary.id_to_str:
-            f.write(s.replace("\\", "\\\\").replace("\n", "\\n") + "\n")
+            f.write(json.dumps(s, ensure_ascii=False) + "\n")
     if len(db.quoted_triples):
         qt = np.asarray(db.quoted_triples.id_to_triple, dtype=np.uint32)
         np.save(path + ".qt.npy", qt)
+
+
+def _unescape_v1(s: str) -> str:
+    """Single left-to-right pass over v1's escaping ('\\\\' and '\\n') —
+    v1's loader unescaped '\\n' before '\\\\', decoding literal
+    backslash-n wrong; this processes each escape exactly once."""
+    out = []
+    i = 0
+    while i < len(s):
+        c = s[i]
+        if c == "\\" and i + 1 < len(s):
+            n = s[i + 1]
+            if n == "n":
+                out.append("\n")
+                i += 2
+                continue
+            if n == "\\":
+                out.append("\\")
+                i += 2
+                continue
+        out.append(c)
+        i += 1
+    return "".join(out)
 
 
 def load_binary(db, path: str):
@@ -66,9 +92,21 @@ def load_binary(db, path: str):
     with open(path + ".meta.json", "r", encoding="utf-8") as f:
         meta = json.load(f)
     with open(path + ".dict", "r", encoding="utf-8") as f:
-        strings = [line[:-1].replace("\\n", "\n").replace("\\\\", "\\")
-                   for line in f]
+        if meta.get("version", 1) >= 2:
+            strings = [json.loads(line) for line in f if line.strip()]
+        else:
+            strings = [_unescape_v1(line[:-1]) for line in f]
     d = db.dictionary
+    if len(d) > 1:
+        # re-encoding into a non-empty dictionary would remap IDs and
+        # silently corrupt the restored columns; allow only an exact
+        # prefix match (e.g. re-loading into the same process)
+        for i, s in enumerate(strings[:len(d.id_to_str)]):
+            if d.id_to_str[i] != s:
+                raise ValueError(
+                    "load_binary into a non-empty dictionary whose entries "
+                    f"differ from the checkpoint at ID {i}: restored "
+                    "columns would be remapped. Load into a fresh database.")
     for s in strings:
         d.encode(s)
     qt_path = path + ".qt.npy"
