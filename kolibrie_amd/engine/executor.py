@@ -19,7 +19,8 @@ from typing import Dict, List, Optional, Sequence, Tuple
 import torch
 
 from ..plan.physical import (
-    PBind, PBindJoin, PFilter, PHashJoin, PIndexScan, PInMemoryBuffer,
+    PBind, PBindJoin, PExchange, PFilter, PHashJoin, PIndexScan,
+    PInMemoryBuffer,
     PConstStar, PLeftJoin, PMLPredict, PMinus, PNestedLoopJoin, PStarJoin,
     PSubquery, PTableScan, PUnion, PUnit, PValues, PhysicalOp,
 )
@@ -44,6 +45,10 @@ class DatasetView:
 class ExecutionContext:
     db: object
     view: DatasetView = field(default_factory=DatasetView)
+    # distributed execution (planner-emitted PExchange ops): world size and
+    # this process's rank; 1/0 = single-process (exchanges are identity)
+    world: int = 1
+    rank: int = 0
 
     def default_index(self) -> GraphIndex:
         gs = self.view.default_graphs
@@ -55,6 +60,28 @@ class ExecutionContext:
         if self.view.named_graphs is not None:
             return self.view.named_graphs
         return self.db.store.named_graph_ids()
+
+
+def _subtree_has_exchange(op: PhysicalOp) -> bool:
+    """True if the subtree contains a PExchange.  Distributed collectives
+    must run on EVERY rank, so empty-input short-circuits may not skip a
+    subtree containing one (a rank-local empty intermediate would
+    deadlock the other ranks).  Memoized on the plan node."""
+    cached = getattr(op, "_has_exch", None)
+    if cached is not None:
+        return cached
+    found = isinstance(op, PExchange)
+    if not found:
+        for name in ("input", "left", "right"):
+            child = getattr(op, name, None)
+            if isinstance(child, PhysicalOp) and _subtree_has_exchange(child):
+                found = True
+                break
+    try:
+        op._has_exch = found
+    except AttributeError:
+        pass
+    return found
 
 
 class ExecutionEngine:
@@ -95,21 +122,24 @@ class ExecutionEngine:
             return out
         if isinstance(op, PHashJoin):
             left = self.execute(op.left, incoming)
-            if left.is_empty():
+            if left.is_empty() and not _subtree_has_exchange(op.right):
                 return left
             right = self.execute(op.right, Bindings.unit(self.device))
             return join_bindings(left, right, needed)
         if isinstance(op, PBindJoin):
             left = self.execute(op.left, incoming)
-            if left.is_empty():
+            if left.is_empty() and not _subtree_has_exchange(op.right):
                 return left
             return self.execute(op.right, left)
         if isinstance(op, PNestedLoopJoin):
             left = self.execute(op.left, incoming)
-            if left.is_empty():
+            if left.is_empty() and not _subtree_has_exchange(op.right):
                 return left
             right = self.execute(op.right, Bindings.unit(self.device))
             return join_bindings(left, right, needed)
+        if isinstance(op, PExchange):
+            rows = self.execute(op.input, incoming)
+            return self._exec_exchange(op, rows)
         if isinstance(op, PUnion):
             l = self.execute(op.left, incoming)
             r = self.execute(op.right, incoming)
@@ -137,23 +167,27 @@ class ExecutionEngine:
             return join_bindings(rows, sub)
         if isinstance(op, PLeftJoin):
             left = self.execute(op.left, incoming)
-            if left.is_empty():
+            if left.is_empty() and not _subtree_has_exchange(op.right):
                 return left
             right = self.execute(op.right, Bindings.unit(self.device))
+            if left.is_empty():
+                return left
             return left_outer_join(left, right, needed)
         if isinstance(op, PMinus):
             left = self.execute(op.left, incoming)
-            if left.is_empty():
+            if left.is_empty() and not _subtree_has_exchange(op.right):
                 return left
             from ..plan.physical import op_certain_vars, op_possible_vars
             possible = op_possible_vars(op.right)
-            if possible is not None and not any(v in left.cols
-                                                for v in possible):
+            if (possible is not None
+                    and not _subtree_has_exchange(op.right)
+                    and not any(v in left.cols for v in possible)):
                 # SPARQL MINUS with statically disjoint domains removes
                 # nothing — skip evaluating the right side entirely
                 return left
             certain = op_certain_vars(op.right)
-            if (not left.maybe_unbound and left.n <= 5_000_000
+            if (self.ctx.world == 1 and not left.maybe_unbound
+                    and left.n <= 5_000_000
                     and any(v in left.cols for v in certain)):
                 # semi-join pushdown: evaluate the right side PROBED by the
                 # left rows instead of scanning it in full — the probed
@@ -174,6 +208,66 @@ class ExecutionEngine:
             from ..ml.predict import execute_ml_predict
             return execute_ml_predict(op.info, rows, self.db)
         raise ValueError(f"cannot execute {type(op).__name__}")
+
+    def _exec_exchange(self, op: PExchange, rows: Bindings) -> Bindings:
+        """Planner-emitted re-partition.
+
+        hash:      row i -> rank (row[var] & 0xFFFFFFFF) % world — the same
+                   partition function the subject-hash loader uses, so an
+                   exchanged intermediate is co-located with the shard rows
+                   of any pattern whose subject is `var`.
+        broadcast: replicate the (small) table on every rank (broadcast
+                   join build side / MINUS-OPTIONAL right sides).
+        rank0:     keep rows only on rank 0 (turns a replicated source into
+                   a valid partitioned one, e.g. a VALUES branch of UNION).
+        """
+        world = self.ctx.world
+        if world <= 1:
+            return rows
+        from ..parallel import dist as D
+        if not D.is_dist():
+            return rows
+        if op.mode == "rank0":
+            if self.ctx.rank == 0:
+                return rows
+            return Bindings.empty(self.device, rows.variables) \
+                if rows.cols else Bindings({}, 0, self.device)
+        # column-set synchronization: empty intermediates can carry fewer
+        # columns than non-empty ones on other ranks (pruned empty-join
+        # paths, probed scans on empty incoming, OPTIONAL early returns) —
+        # the per-column collectives below would then desynchronize and
+        # deadlock.  Exchange the column-name lists, use the stable union,
+        # and fill locally-missing columns with UNBOUND.
+        import torch.distributed as dist
+        meta = [None] * world
+        dist.all_gather_object(meta, (list(rows.variables),
+                                      bool(rows.maybe_unbound)))
+        names = list(dict.fromkeys(v for lst, _mu in meta for v in lst))
+        tainted = any(mu for _lst, mu in meta) or \
+            any(set(lst) != set(names) for lst, _mu in meta)
+        def _col(v):
+            if rows.has(v):
+                return rows.col(v)
+            return torch.full((rows.n,), UNBOUND, dtype=torch.int32,
+                              device=self.device)
+        if op.mode == "broadcast":
+            if not names:
+                n = D.allreduce_sum_scalar(rows.n, self.device)
+                return Bindings({}, n, self.device)
+            cols = D.all_gather_rows([_col(v) for v in names])
+            return Bindings(dict(zip(names, cols)), cols[0].numel(),
+                            self.device, maybe_unbound=tainted)
+        # hash re-partition on op.var
+        if op.var not in names:
+            raise ValueError(
+                f"PExchange(hash) key ?{op.var} missing from row table "
+                f"columns {names} — planner/runtime mismatch")
+        key = _col(op.var).to(torch.int64) & 0xFFFFFFFF
+        dest = key % world
+        cols = D.all_to_all_rows([_col(v) for v in names], dest)
+        return Bindings(dict(zip(names, cols)),
+                        cols[0].numel() if cols else 0, self.device,
+                        maybe_unbound=tainted)
 
     def _exec_const_star(self, op, incoming: Bindings, needed) -> Bindings:
         """Bound-subject star: ONE SPO-region fetch + host-side pattern
@@ -760,6 +854,11 @@ def join_bindings(left: Bindings, right: Bindings, needed=None) -> Bindings:
         return _prune(left, needed)
     if left.is_empty() or right.is_empty():
         vars_ = list(dict.fromkeys(left.variables + right.variables))
+        if needed is not None:
+            # match the non-empty path's pruning: a divergent column set on
+            # an empty intermediate would desynchronize distributed
+            # exchanges (per-column collectives) across ranks
+            vars_ = [v for v in vars_ if v in needed]
         return Bindings.empty(dev, vars_)
     shared = [v for v in left.variables if v in right.cols]
     if not shared:

@@ -419,6 +419,9 @@ def phys_out_vars(op: PhysicalOp) -> Set[str]:
         return phys_out_vars(op.left) | phys_out_vars(op.right)
     if isinstance(op, PFilter):
         return phys_out_vars(op.input)
+    from .physical import PExchange
+    if isinstance(op, PExchange):
+        return phys_out_vars(op.input)
     if isinstance(op, PBind):
         return phys_out_vars(op.input) | {op.var}
     if isinstance(op, PValues):
@@ -522,6 +525,13 @@ def annotate_needed(op: PhysicalOp, needed):
         return
     if isinstance(op, PMLPredict):
         annotate_needed(op.input, None)
+        return
+    from .physical import PExchange
+    if isinstance(op, PExchange):
+        # the exchange key column must survive projection pushdown
+        child = needed if (needed is None or op.mode != "hash") \
+            else set(needed) | {op.var}
+        annotate_needed(op.input, child)
         return
     if hasattr(op, "input"):
         annotate_needed(op.input, needed)

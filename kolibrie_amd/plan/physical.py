@@ -141,6 +141,21 @@ class PInMemoryBuffer(PhysicalOp):
     bindings: object = None
 
 
+@dataclass
+class PExchange(PhysicalOp):
+    """Distributed re-partition: the planner inserts
+    this when a join key is not the current partition key.
+
+    mode="hash":      rows move to rank hash(row[var]) % world via the
+                      pairwise xGMI all-to-all (all_to_all_rows).
+    mode="broadcast": the (small) input is replicated on every rank via
+                      all-gather — the broadcast-join build side.
+    Single-process execution is the identity."""
+    input: PhysicalOp = field(default_factory=PUnit)
+    var: str = ""            # partition key variable (hash mode)
+    mode: str = "hash"       # "hash" | "broadcast"
+
+
 def _graph_var(graph: GraphScope):
     if graph is not None and graph[0] == "var":
         return graph[1]
@@ -175,7 +190,7 @@ def op_certain_vars(op: PhysicalOp) -> set:
         return op_certain_vars(op.left) & op_certain_vars(op.right)
     if isinstance(op, (PLeftJoin, PMinus)):
         return op_certain_vars(op.left)
-    if isinstance(op, PFilter):
+    if isinstance(op, (PFilter, PExchange)):
         return op_certain_vars(op.input)
     if isinstance(op, PBind):
         base = op_certain_vars(op.input)
@@ -214,7 +229,7 @@ def op_possible_vars(op: PhysicalOp):
         return l | r
     if isinstance(op, PMinus):
         return op_possible_vars(op.left)
-    if isinstance(op, PFilter):
+    if isinstance(op, (PFilter, PExchange)):
         return op_possible_vars(op.input)
     if isinstance(op, PBind):
         base = op_possible_vars(op.input)
@@ -243,6 +258,8 @@ def plan_key(op: PhysicalOp) -> str:
     if isinstance(op, (PHashJoin, PBindJoin, PNestedLoopJoin, PUnion)):
         tag = type(op).__name__[1:3]
         return f"{tag}[{plan_key(op.left)},{plan_key(op.right)}]"
+    if isinstance(op, PExchange):
+        return f"X({op.mode},{op.var})[{plan_key(op.input)}]"
     if hasattr(op, "input"):
         return f"{type(op).__name__}[{plan_key(op.input)}]"
     return type(op).__name__
