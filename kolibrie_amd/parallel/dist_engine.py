@@ -76,6 +76,18 @@ class DistributedDatabase:
         self.device = torch.device(device)
         self.db = SparqlDatabase(device=str(device))
         self._global_stats = None
+        # predicate IDs whose relations the loader replicated on every
+        # rank (load-time broadcast-table layout); the planner treats
+        # their scans as part=REPLICATED
+        self.replicated_preds = set()
+
+    def declare_replicated(self, pred_iris):
+        """Record that the caller loaded these predicates' triples on
+        EVERY rank (small build-side relations)."""
+        for iri in pred_iris:
+            self.replicated_preds.add(
+                self.db.dictionary.encode(iri) & 0xFFFFFFFF)
+        self._global_stats = None
 
     # ------------------------------------------------------------- loading --
     def load_shard_columns(self, s, p, o):
@@ -111,7 +123,19 @@ class DistributedDatabase:
     def global_stats(self):
         if self._global_stats is None:
             local = self.db.get_or_build_stats()
-            self._global_stats = allreduce_stats(local, self.device)
+            g = allreduce_stats(local, self.device)
+            # replicated relations were summed world x by the allreduce;
+            # every rank holds the identical full copy, so the local
+            # counts ARE the global ones
+            for pid in self.replicated_preds:
+                for d_g, d_l in ((g.pred_count, local.pred_count),
+                                 (g.pred_distinct_subj,
+                                  local.pred_distinct_subj),
+                                 (g.pred_distinct_obj,
+                                  local.pred_distinct_obj)):
+                    if pid in d_l:
+                        d_g[pid] = d_l[pid]
+            self._global_stats = g
         return self._global_stats
 
     # -------------------------------------------------------------- query --
@@ -131,7 +155,8 @@ class DistributedDatabase:
         stats = self.global_stats()
         logical = build_logical_plan(sel.where, db, prefixes)
         physical = Streamertail(stats).find_best_plan(logical)
-        physical, part = distribute_plan(physical, stats, self.world)
+        physical, part = distribute_plan(physical, stats, self.world,
+                                         self.replicated_preds)
         annotate_needed(physical, _top_needed(sel))
         return sel, physical, part
 

@@ -61,6 +61,23 @@ Part = Union[str, Tuple[str, str]]  # "replicated" | "scattered" | ("hash",v)
 class DistPlanner:
     stats: object        # DatabaseStats (global, allreduced)
     world: int
+    # predicates whose relations the loader REPLICATED on every rank (the
+    # load-time broadcast-table layout for small build sides): scans of
+    # these are part=REPLICATED and never need an exchange
+    replicated_preds: frozenset = frozenset()
+
+    def _is_replicated_scan(self, op: PhysicalOp) -> bool:
+        if isinstance(op, (PTableScan, PIndexScan)):
+            return (isinstance(op.pattern.p, Constant)
+                    and (op.pattern.p.id & 0xFFFFFFFF)
+                    in self.replicated_preds)
+        if isinstance(op, PStarJoin):
+            return all(isinstance(p.p, Constant)
+                       and (p.p.id & 0xFFFFFFFF) in self.replicated_preds
+                       for p in op.patterns)
+        if isinstance(op, (PFilter, PBind, PProjection)):
+            return self._is_replicated_scan(op.input)
+        return False
 
     # ------------------------------------------------------------ helpers --
     def _est_rows(self, op: PhysicalOp) -> float:
@@ -109,10 +126,14 @@ class DistPlanner:
         if isinstance(op, PUnit):
             return op, REPLICATED
         if isinstance(op, (PTableScan, PIndexScan)):
+            if self._is_replicated_scan(op):
+                return op, REPLICATED
             if isinstance(op.pattern.s, Variable):
                 return op, ("hash", op.pattern.s.name)
             return op, SCATTERED  # const subject: all rows on one rank
         if isinstance(op, PStarJoin):
+            if self._is_replicated_scan(op):
+                return op, REPLICATED
             # all patterns share the subject var: co-located by definition
             return op, ("hash", op.join_var)
         if isinstance(op, PConstStar):
@@ -265,6 +286,10 @@ class DistPlanner:
         right = op.right
         probe_subject = self._probe_subject(right)
         lvars = phys_out_vars(l)
+        if self._leafish(right) and self._is_replicated_scan(right):
+            # probing a replicated relation is local for ANY left layout
+            op.left = l
+            return op, lp
         if not self._leafish(right):
             # compound right (e.g. const-star seeding a chain): its internal
             # joins need their own exchanges — distribute it, after placing
@@ -334,10 +359,11 @@ class DistPlanner:
         return SCATTERED
 
 
-def distribute_plan(physical: PhysicalOp, stats, world: int
-                    ) -> Tuple[PhysicalOp, Part]:
+def distribute_plan(physical: PhysicalOp, stats, world: int,
+                    replicated_preds=frozenset()) -> Tuple[PhysicalOp, Part]:
     """Entry point: rewrite a single-GPU physical plan for `world` ranks.
     Returns (plan, partition property of the root output)."""
     if world <= 1:
         return physical, REPLICATED
-    return DistPlanner(stats, world).distribute(physical)
+    return DistPlanner(stats, world,
+                       frozenset(replicated_preds)).distribute(physical)
