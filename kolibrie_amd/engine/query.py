@@ -161,11 +161,9 @@ def _run_group_count(pq: "PreparedQuery", db) -> Optional[List[List[str]]]:
         return []
     vals, counts = torch.unique_consecutive(
         key12[lo:hi] & 0xFFFFFFFF, return_counts=True)
-    if vals.numel() > 100_000:
-        return None  # fall back to the generic path
-    cnt_ids = torch.tensor(
-        [db.dictionary.encode(str(int(c))) for c in counts.cpu().tolist()],
-        dtype=torch.int32, device=key12.device)
+    from .finalize import _encode_numbers
+    cnt_ids = _encode_numbers(db, counts.to(torch.float64), key12.device,
+                              integral=True)
     gcol = vals.to(torch.int32)
     cols = {}
     for kind, name in names:

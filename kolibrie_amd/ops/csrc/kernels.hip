@@ -19,6 +19,7 @@
 #include <c10/hip/HIPStream.h>
 #include <hip/hip_runtime.h>
 
+#include <algorithm>
 #include <cstdint>
 #include <unordered_map>
 
@@ -1364,7 +1365,270 @@ py::tuple parse_nquads_host(const std::string& text) {
   return parse_nlines_host(text, true);
 }
 
+// ------------------------------------------------------------ K4: group agg
+// Hand-written GROUP BY hash aggregate (replaces reference
+// execute_query.rs:404-475 aggregate_rows).
+//
+// One pass over (key, value) rows into an open-addressing global table,
+// staged through a per-workgroup LDS table first: low-cardinality keys
+// (the contended case) are pre-aggregated at LDS-atomic speed and flushed
+// once per block, so global atomics see at most (#blocks x #groups)
+// traffic; high-cardinality keys miss the small LDS table and go straight
+// to the (uncontended) global table.  A compaction pass then emits the
+// occupied slots.  MIN/MAX on doubles use the monotone bit encoding so
+// unsigned 64-bit atomicMin/Max apply.
+//
+// Keys are caller-packed int64 (single var: (uint32)id; two vars:
+// (a<<32)|b).  `empty_key` is a caller-chosen value no row can produce.
+constexpr int kAggLdsSlots = 2048;   // 5 x 8 B x 2048 = 80 KiB LDS -> 2 WG/CU
+
+__device__ __forceinline__ uint32_t h64(unsigned long long k) {
+  k ^= k >> 33; k *= 0xff51afd7ed558ccdULL;
+  k ^= k >> 33; k *= 0xc4ceb9fe1a85ec53ULL;
+  k ^= k >> 33;
+  return static_cast<uint32_t>(k);
+}
+
+__device__ __forceinline__ unsigned long long enc_double(double x) {
+  unsigned long long b =
+      static_cast<unsigned long long>(__double_as_longlong(x));
+  return (b & 0x8000000000000000ULL) ? ~b : (b | 0x8000000000000000ULL);
+}
+
+__device__ __forceinline__ double dec_double(unsigned long long e) {
+  unsigned long long b =
+      (e & 0x8000000000000000ULL) ? (e & 0x7FFFFFFFFFFFFFFFULL) : ~e;
+  return __longlong_as_double(static_cast<long long>(b));
+}
+
+// probe-run cap: at the <=50% load factor the host wrapper guarantees on
+// the final attempt, expected linear-probe runs are ~2 slots; a run this
+// long means the table is too small — flag overflow and let the host
+// re-run bigger.  A full-table scan here (the naive failure mode) would
+// cost minutes at millions of threads.
+constexpr uint32_t kAggMaxProbe = 128;
+
+__device__ __forceinline__ int64_t agg_claim_slot(
+    unsigned long long* __restrict__ keys, uint32_t mask,
+    unsigned long long key, unsigned long long empty,
+    int32_t* __restrict__ overflow) {
+  uint32_t h = h64(key) & mask;
+  uint32_t cap = mask < kAggMaxProbe ? mask : kAggMaxProbe;
+  for (uint32_t probe = 0; probe <= cap; ++probe) {
+    unsigned long long cur = keys[h];
+    if (cur == key) return h;
+    if (cur == empty) {
+      unsigned long long prev = atomicCAS(&keys[h], empty, key);
+      if (prev == empty || prev == key) return h;
+      continue;  // lost the race to a different key: retry same slot read
+    }
+    h = (h + 1) & mask;
+  }
+  atomicExch(overflow, 1);  // table too loaded: caller re-runs bigger
+  return -1;
+}
+
+__global__ void __launch_bounds__(kBlock)
+group_agg_kernel(const int64_t* __restrict__ keys_in,
+                 const double* __restrict__ vals, int64_t n,
+                 unsigned long long empty,
+                 unsigned long long* __restrict__ gkeys,
+                 unsigned long long* __restrict__ gcnt,
+                 double* __restrict__ gsum,
+                 unsigned long long* __restrict__ gmn,
+                 unsigned long long* __restrict__ gmx,
+                 uint32_t gmask, int32_t* __restrict__ overflow) {
+  __shared__ unsigned long long lkeys[kAggLdsSlots];
+  __shared__ unsigned long long lcnt[kAggLdsSlots];
+  __shared__ double lsum[kAggLdsSlots];
+  __shared__ unsigned long long lmn[kAggLdsSlots];
+  __shared__ unsigned long long lmx[kAggLdsSlots];
+  for (int i = threadIdx.x; i < kAggLdsSlots; i += blockDim.x) {
+    lkeys[i] = empty;
+    lcnt[i] = 0ULL;
+    lsum[i] = 0.0;
+    lmn[i] = ~0ULL;
+    lmx[i] = 0ULL;
+  }
+  __syncthreads();
+  const bool want_val = vals != nullptr;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    if ((i & 0x3FF) == 0 && *overflow) return;  // abort a doomed attempt
+    unsigned long long key = static_cast<unsigned long long>(keys_in[i]);
+    double v = want_val ? vals[i] : 0.0;
+    unsigned long long ev = want_val ? enc_double(v) : 0ULL;
+    // LDS attempt: short probe run, then spill to the global table
+    uint32_t h = h64(key) & (kAggLdsSlots - 1);
+    int64_t slot = -1;
+    for (int p = 0; p < 8; ++p) {
+      unsigned long long cur = lkeys[h];
+      if (cur == key) { slot = h; break; }
+      if (cur == empty) {
+        unsigned long long prev = atomicCAS(&lkeys[h], empty, key);
+        if (prev == empty || prev == key) { slot = h; break; }
+        continue;  // re-read: someone else claimed it this cycle
+      }
+      h = (h + 1) & (kAggLdsSlots - 1);
+    }
+    if (slot >= 0) {
+      atomicAdd(&lcnt[slot], 1ULL);
+      if (want_val) {
+        if (gsum) atomicAdd(&lsum[slot], v);
+        if (gmn) atomicMin(&lmn[slot], ev);
+        if (gmx) atomicMax(&lmx[slot], ev);
+      }
+    } else {
+      int64_t gs = agg_claim_slot(gkeys, gmask, key, empty, overflow);
+      if (gs < 0) return;
+      atomicAdd(&gcnt[gs], 1ULL);
+      if (want_val) {
+        if (gsum) atomicAdd(&gsum[gs], v);
+        if (gmn) atomicMin(&gmn[gs], ev);
+        if (gmx) atomicMax(&gmx[gs], ev);
+      }
+    }
+  }
+  __syncthreads();
+  // flush the block's LDS pre-aggregates into the global table
+  for (int i = threadIdx.x; i < kAggLdsSlots; i += blockDim.x) {
+    if (lkeys[i] == empty) continue;
+    int64_t gs = agg_claim_slot(gkeys, gmask, lkeys[i], empty, overflow);
+    if (gs < 0) return;
+    atomicAdd(&gcnt[gs], lcnt[i]);
+    if (gsum) atomicAdd(&gsum[gs], lsum[i]);
+    if (gmn) atomicMin(&gmn[gs], lmn[i]);
+    if (gmx) atomicMax(&gmx[gs], lmx[i]);
+  }
+}
+
+__global__ void agg_compact(const unsigned long long* __restrict__ gkeys,
+                            const unsigned long long* __restrict__ gcnt,
+                            const double* __restrict__ gsum,
+                            const unsigned long long* __restrict__ gmn,
+                            const unsigned long long* __restrict__ gmx,
+                            int64_t cap, unsigned long long empty,
+                            int64_t* __restrict__ out_keys,
+                            int64_t* __restrict__ out_cnt,
+                            double* __restrict__ out_sum,
+                            double* __restrict__ out_mn,
+                            double* __restrict__ out_mx,
+                            int32_t* __restrict__ counter) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < cap;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    unsigned long long k = gkeys[i];
+    if (k == empty) continue;
+    int32_t idx = atomicAdd(counter, 1);
+    out_keys[idx] = static_cast<int64_t>(k);
+    out_cnt[idx] = static_cast<int64_t>(gcnt[i]);
+    if (out_sum) out_sum[idx] = gsum[i];
+    if (out_mn) out_mn[idx] = dec_double(gmn[i]);
+    if (out_mx) out_mx[idx] = dec_double(gmx[i]);
+  }
+}
+
+// host wrapper: (keys[n] int64, vals[n] double|None) ->
+//   (keys[g], cnt[g], sum[g]|None, min[g]|None, max[g]|None)
+py::tuple group_aggregate(at::Tensor keys, c10::optional<at::Tensor> vals,
+                          bool want_sum, bool want_min, bool want_max,
+                          int64_t empty_key) {
+  TORCH_CHECK(keys.is_cuda() && keys.dtype() == at::kLong && keys.dim() == 1);
+  const int64_t n = keys.numel();
+  const double* vptr = nullptr;
+  if (vals.has_value()) {
+    TORCH_CHECK(vals->is_cuda() && vals->dtype() == at::kDouble
+                && vals->numel() == n);
+    vptr = vals->data_ptr<double>();
+  }
+  const bool want_val = vptr != nullptr;
+  auto opts_i64 = keys.options();
+  auto opts_f64 = keys.options().dtype(at::kDouble);
+  auto opts_i32 = keys.options().dtype(at::kInt);
+  auto stream = cur_stream();
+  const unsigned long long empty =
+      static_cast<unsigned long long>(empty_key);
+  // start at ~n/4 slots (covers group counts up to ~n/8 at <=50% load) and
+  // retry x8 on overflow; the final rung 2*n can never overflow.  A failed
+  // attempt aborts early (probe cap + overflow poll), so a retry costs
+  // about one extra pass over the keys.
+  int64_t cap = 1 << 17;
+  while (cap < n / 4 && cap < (1LL << 27)) cap <<= 1;
+  for (;;) {
+    auto gkeys = at::full({cap}, empty_key, opts_i64);
+    auto gcnt = at::zeros({cap}, opts_i64);
+    at::Tensor gsum, gmn, gmx;
+    double* gsum_p = nullptr;
+    unsigned long long* gmn_p = nullptr;
+    unsigned long long* gmx_p = nullptr;
+    if (want_val && want_sum) {
+      gsum = at::zeros({cap}, opts_f64);
+      gsum_p = gsum.data_ptr<double>();
+    }
+    if (want_val && want_min) {
+      gmn = at::full({cap}, -1, opts_i64);  // ~0ULL = +inf in the encoding
+      gmn_p = reinterpret_cast<unsigned long long*>(gmn.data_ptr<int64_t>());
+    }
+    if (want_val && want_max) {
+      gmx = at::zeros({cap}, opts_i64);     // 0 = -inf in the encoding
+      gmx_p = reinterpret_cast<unsigned long long*>(gmx.data_ptr<int64_t>());
+    }
+    auto overflow = at::zeros({1}, opts_i32);
+    if (n > 0) {
+      hipLaunchKernelGGL(group_agg_kernel, dim3(grid_for(n)), dim3(kBlock), 0,
+                         stream, keys.data_ptr<int64_t>(), vptr, n, empty,
+                         reinterpret_cast<unsigned long long*>(
+                             gkeys.data_ptr<int64_t>()),
+                         reinterpret_cast<unsigned long long*>(
+                             gcnt.data_ptr<int64_t>()),
+                         gsum_p, gmn_p, gmx_p,
+                         static_cast<uint32_t>(cap - 1),
+                         overflow.data_ptr<int32_t>());
+      HIP_OK(hipGetLastError());
+    }
+    if (n > 0 && overflow.item<int32_t>() != 0) {
+      int64_t full = 1;
+      while (full < 2 * n) full <<= 1;
+      TORCH_CHECK(cap < full, "group_aggregate: overflow at 2*n capacity");
+      cap = std::min(cap << 3, full);
+      continue;
+    }
+    auto out_keys = at::empty({cap}, opts_i64);
+    auto out_cnt = at::empty({cap}, opts_i64);
+    at::Tensor out_sum, out_mn, out_mx;
+    double* os = nullptr; double* omn = nullptr; double* omx = nullptr;
+    if (gsum_p) { out_sum = at::empty({cap}, opts_f64);
+                  os = out_sum.data_ptr<double>(); }
+    if (gmn_p) { out_mn = at::empty({cap}, opts_f64);
+                 omn = out_mn.data_ptr<double>(); }
+    if (gmx_p) { out_mx = at::empty({cap}, opts_f64);
+                 omx = out_mx.data_ptr<double>(); }
+    auto counter = at::zeros({1}, opts_i32);
+    hipLaunchKernelGGL(agg_compact, dim3(grid_for(cap)), dim3(kBlock), 0,
+                       stream,
+                       reinterpret_cast<unsigned long long*>(
+                           gkeys.data_ptr<int64_t>()),
+                       reinterpret_cast<unsigned long long*>(
+                           gcnt.data_ptr<int64_t>()),
+                       gsum_p, gmn_p, gmx_p, cap, empty,
+                       out_keys.data_ptr<int64_t>(),
+                       out_cnt.data_ptr<int64_t>(), os, omn, omx,
+                       counter.data_ptr<int32_t>());
+    HIP_OK(hipGetLastError());
+    int64_t g = counter.item<int32_t>();
+    py::tuple t(5);
+    t[0] = out_keys.narrow(0, 0, g);
+    t[1] = out_cnt.narrow(0, 0, g);
+    t[2] = gsum_p ? py::cast(out_sum.narrow(0, 0, g)) : py::none();
+    t[3] = gmn_p ? py::cast(out_mn.narrow(0, 0, g)) : py::none();
+    t[4] = gmx_p ? py::cast(out_mx.narrow(0, 0, g)) : py::none();
+    return t;
+  }
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("group_aggregate", &group_aggregate,
+        "K4 LDS-staged hash GROUP BY: (keys, vals?, sum, min, max, empty) "
+        "-> (group keys, counts, sums?, mins?, maxs?)");
   m.def("parse_ntriples_host", &parse_ntriples_host,
         "bulk N-Triples parse -> (local-id triples, strings, fallback lines)");
   m.def("parse_nquads_host", &parse_nquads_host,
