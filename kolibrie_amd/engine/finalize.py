@@ -316,13 +316,32 @@ def _native_group_aggregate(select: SelectQuery, rows: Bindings, db,
         a = rows.col(group_vars[0]).to(torch.int64) & 0xFFFFFFFF
         b = rows.col(group_vars[1]).to(torch.int64) & 0xFFFFFFFF
         keys = (a << 32) | b
+    # sampled cardinality estimate: sizes the hash table in one shot and
+    # routes mostly-unique keys (where a sort beats hashing) to the torch
+    # path.  Uniform-groups model: E[u] = g(1 - exp(-s/g)) inverted by a
+    # few fixed-point steps.
+    n = keys.numel()
+    groups_hint = 0
+    if n > 300_000:
+        stride = max(1, n // 65536)
+        sample = keys[::stride]
+        s_n = sample.numel()
+        u = int(torch.unique(sample).numel())
+        r = u / max(1, s_n)
+        if r > 0.9:
+            return None  # ~all-unique: sort-based composite wins
+        import math
+        g = float(u)
+        for _ in range(20):
+            g = u / max(1e-9, 1.0 - math.exp(-s_n / max(g, 1.0)))
+        groups_hint = max(2048, min(n, int(2.0 * g)))
     vals = None
     if arg is not None:
         from .tensor_utils import values_for_ids
         ids_u = rows.col(arg).to(torch.int64) & 0xFFFFFFFF
         vals = values_for_ids(db.value_column(), ids_u).to(torch.float64)
     gkeys, cnt, gsum, gmn, gmx = native.group_aggregate(
-        keys, vals, want_sum, want_min, want_max, -1)
+        keys, vals, want_sum, want_min, want_max, -1, groups_hint)
     out_cols: Dict[str, torch.Tensor] = {}
     if len(group_vars) == 1:
         out_cols[group_vars[0]] = gkeys.to(torch.int32)

@@ -1437,7 +1437,8 @@ group_agg_kernel(const int64_t* __restrict__ keys_in,
                  double* __restrict__ gsum,
                  unsigned long long* __restrict__ gmn,
                  unsigned long long* __restrict__ gmx,
-                 uint32_t gmask, int32_t* __restrict__ overflow) {
+                 uint32_t gmask, int32_t* __restrict__ overflow,
+                 int use_lds) {
   __shared__ unsigned long long lkeys[kAggLdsSlots];
   __shared__ unsigned long long lcnt[kAggLdsSlots];
   __shared__ double lsum[kAggLdsSlots];
@@ -1458,10 +1459,12 @@ group_agg_kernel(const int64_t* __restrict__ keys_in,
     unsigned long long key = static_cast<unsigned long long>(keys_in[i]);
     double v = want_val ? vals[i] : 0.0;
     unsigned long long ev = want_val ? enc_double(v) : 0ULL;
-    // LDS attempt: short probe run, then spill to the global table
+    // LDS attempt: short probe run, then spill to the global table.
+    // use_lds=0 (high estimated cardinality: most keys would miss the
+    // small LDS table anyway) goes straight to the global table.
     uint32_t h = h64(key) & (kAggLdsSlots - 1);
     int64_t slot = -1;
-    for (int p = 0; p < 8; ++p) {
+    for (int p = 0; use_lds && p < 8; ++p) {
       unsigned long long cur = lkeys[h];
       if (cur == key) { slot = h; break; }
       if (cur == empty) {
@@ -1531,7 +1534,7 @@ __global__ void agg_compact(const unsigned long long* __restrict__ gkeys,
 //   (keys[g], cnt[g], sum[g]|None, min[g]|None, max[g]|None)
 py::tuple group_aggregate(at::Tensor keys, c10::optional<at::Tensor> vals,
                           bool want_sum, bool want_min, bool want_max,
-                          int64_t empty_key) {
+                          int64_t empty_key, int64_t groups_hint) {
   TORCH_CHECK(keys.is_cuda() && keys.dtype() == at::kLong && keys.dim() == 1);
   const int64_t n = keys.numel();
   const double* vptr = nullptr;
@@ -1547,12 +1550,18 @@ py::tuple group_aggregate(at::Tensor keys, c10::optional<at::Tensor> vals,
   auto stream = cur_stream();
   const unsigned long long empty =
       static_cast<unsigned long long>(empty_key);
-  // start at ~n/4 slots (covers group counts up to ~n/8 at <=50% load) and
-  // retry x8 on overflow; the final rung 2*n can never overflow.  A failed
-  // attempt aborts early (probe cap + overflow poll), so a retry costs
-  // about one extra pass over the keys.
-  int64_t cap = 1 << 17;
-  while (cap < n / 4 && cap < (1LL << 27)) cap <<= 1;
+  // size from the caller's sampled cardinality estimate (4x the upper
+  // bound keeps load <=25%) or default to ~n/4; retry x8 on overflow —
+  // the final rung 2*n can never overflow.  A failed attempt aborts
+  // early (probe cap + overflow poll), costing ~one extra key pass.
+  int64_t cap = 1 << 14;
+  if (groups_hint > 0) {
+    while (cap < 4 * groups_hint && cap < 2 * n) cap <<= 1;
+  } else {
+    while (cap < n / 4 && cap < (1LL << 27)) cap <<= 1;
+  }
+  const bool use_lds =
+      groups_hint > 0 ? (groups_hint <= 3000) : true;
   for (;;) {
     auto gkeys = at::full({cap}, empty_key, opts_i64);
     auto gcnt = at::zeros({cap}, opts_i64);
@@ -1582,7 +1591,8 @@ py::tuple group_aggregate(at::Tensor keys, c10::optional<at::Tensor> vals,
                              gcnt.data_ptr<int64_t>()),
                          gsum_p, gmn_p, gmx_p,
                          static_cast<uint32_t>(cap - 1),
-                         overflow.data_ptr<int32_t>());
+                         overflow.data_ptr<int32_t>(),
+                         use_lds ? 1 : 0);
       HIP_OK(hipGetLastError());
     }
     if (n > 0 && overflow.item<int32_t>() != 0) {
@@ -1627,8 +1637,8 @@ py::tuple group_aggregate(at::Tensor keys, c10::optional<at::Tensor> vals,
 
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("group_aggregate", &group_aggregate,
-        "K4 LDS-staged hash GROUP BY: (keys, vals?, sum, min, max, empty) "
-        "-> (group keys, counts, sums?, mins?, maxs?)");
+        "K4 LDS-staged hash GROUP BY: (keys, vals?, sum, min, max, empty, "
+        "groups_hint) -> (group keys, counts, sums?, mins?, maxs?)");
   m.def("parse_ntriples_host", &parse_ntriples_host,
         "bulk N-Triples parse -> (local-id triples, strings, fallback lines)");
   m.def("parse_nquads_host", &parse_nquads_host,

@@ -459,7 +459,7 @@ def test_group_aggregate_matches_torch_oracle():
         vals = torch.randn(n, dtype=torch.float64, device="cuda") * 100
         native = native_for(keys32.to(torch.int32))
         gk, cnt, gsum, gmn, gmx = native.group_aggregate(
-            keys32, vals, True, True, True, -1)
+            keys32, vals, True, True, True, -1, 0)
         # torch oracle
         uniq, inv = torch.unique(keys32, return_inverse=True)
         ng = uniq.numel()
