@@ -1416,12 +1416,13 @@ __device__ __forceinline__ int64_t agg_claim_slot(
   uint32_t cap = mask < kAggMaxProbe ? mask : kAggMaxProbe;
   for (uint32_t probe = 0; probe <= cap; ++probe) {
     unsigned long long cur = keys[h];
-    if (cur == key) return h;
     if (cur == empty) {
-      unsigned long long prev = atomicCAS(&keys[h], empty, key);
-      if (prev == empty || prev == key) return h;
-      continue;  // lost the race to a different key: retry same slot read
+      // resolve via CAS, not a re-read: atomics land in L2 and a stale
+      // L1 line could keep reporting `empty` forever
+      cur = atomicCAS(&keys[h], empty, key);
+      if (cur == empty) return h;
     }
+    if (cur == key) return h;
     h = (h + 1) & mask;
   }
   atomicExch(overflow, 1);  // table too loaded: caller re-runs bigger
@@ -1635,7 +1636,438 @@ py::tuple group_aggregate(at::Tensor keys, c10::optional<at::Tensor> vals,
   }
 }
 
+// ------------------------------------------------- K6: small-N fixpoint
+// Persistent SINGLE-WORKGROUP semi-naive fixpoint for small working sets
+// (the deep-taxonomy shape = 10 000 rounds of 1-fact
+// deltas, where per-round kernel-launch + host-sync overhead dominates
+// any columnar formulation).  One 1024-thread block runs the WHOLE
+// fixpoint: rounds are __syncthreads() boundaries (ns, not µs), and all
+// inter-thread data passes stay inside one CU so workgroup-scope fences
+// suffice (MI355X guide §Workgroup dispatch: no agent-scope traffic).
+//
+// Rule language (host driver rejects anything else and falls back):
+//   kind=1 (join):  (a P1 b), (c P2 d) -> conclusions, exactly ONE shared
+//                   var between the premises, predicates constant;
+//   kind=0 (copy):  (a P1 b) -> conclusions.
+// Conclusions (<=2 per rule): (src_s Pc src_o), srcs in {p1.s,p1.o,p2.s,
+// p2.o}.  Covers transitive closure, type propagation (deep taxonomy),
+// ancestor programs.
+//
+// State per predicate (global memory, host-allocated):
+//   - pair dedup set: open addressing on packed (s<<32)|o u64, CAS claim;
+//   - optional adjacency by-s / by-o: chained hash (heads + append-only
+//     {key,val,next} entries), only for directions some rule probes.
+// The fact log `out` doubles as the delta queue: [d_lo, d_hi) is the
+// current round's delta; appended facts form the next round's.
+//
+// Ref semantics: datalog semi_naive.rs:17-86 (one-shared-var case).
+constexpr int kFxBlock = 1024;
+constexpr unsigned long long kFxEmpty = ~0ULL;  // (UNBOUND,UNBOUND) pair
+
+struct FxRule {
+  int32_t kind;      // 0 copy, 1 join
+  int32_t p1, j1;    // premise1 pred idx; shared-var position (0=s,1=o)
+  int32_t p2, j2;    // premise2 (join only)
+  int32_t n_conc;
+  int32_t c_pred[2];
+  int32_t c_s_src[2];  // 0=p1.s 1=p1.o 2=p2.s 3=p2.o
+  int32_t c_o_src[2];
+};
+
+struct FxPred {
+  unsigned long long* set;   // pair dedup table
+  uint32_t set_mask;
+  int32_t* adj_s_heads; uint32_t adj_s_mask;   // by-subject chains
+  int32_t* adj_o_heads; uint32_t adj_o_mask;   // by-object chains
+};
+
+struct FxState {
+  // shared append-only pools (all predicates)
+  int64_t* adj_entries;      // packed (key<<32)|val
+  int32_t* adj_next;
+  int32_t* adj_pred_dir;     // unused slot kept for debug
+  int32_t* out_s; int32_t* out_p; int32_t* out_o;  // fact log (pred idx)
+  int32_t* counters;         // [0]=out_n [1]=adj_n [2]=overflow [3]=rounds
+  int64_t out_cap, adj_cap;
+};
+
+__device__ __forceinline__ bool fx_set_insert(const FxPred& pr,
+                                              uint32_t s, uint32_t o,
+                                              int32_t* overflow) {
+  unsigned long long key =
+      (static_cast<unsigned long long>(s) << 32) | o;
+  uint32_t h = h64(key) & pr.set_mask;
+  uint32_t cap = pr.set_mask < kAggMaxProbe ? pr.set_mask : kAggMaxProbe;
+  for (uint32_t p = 0; p <= cap; ++p) {
+    unsigned long long cur = pr.set[h];
+    if (cur == kFxEmpty) {
+      cur = atomicCAS(&pr.set[h], kFxEmpty, key);  // L2 truth, not L1
+      if (cur == kFxEmpty) return true;
+    }
+    if (cur == key) return false;
+    h = (h + 1) & pr.set_mask;
+  }
+  atomicExch(overflow, 1);
+  return false;
+}
+
+// L1-bypassing relaxed loads: atomic publishes (atomicExch heads,
+// atomicAdd counters) land in L2 and never refresh even this CU's own
+// L1, so every read of a mutable shared structure goes L2-served.
+__device__ __forceinline__ int32_t fx_ld(const int32_t* p) {
+  return __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+}
+__device__ __forceinline__ int64_t fx_ld64(const int64_t* p) {
+  return __hip_atomic_load(p, __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+}
+
+// Allocation counters / overflow flag / optional LDS delta ring.  In the
+// persistent single-WG kernel these live in LDS (per-emit bookkeeping at
+// LDS-atomic speed instead of ~4 L2 round trips); the multi-block seed
+// kernel passes the global counters instead.
+constexpr int kFxRing = 4096;
+
+struct FxCtx {
+  int32_t* out_n;
+  int32_t* adj_n;
+  int32_t* overflow;
+  int32_t* ring_s; int32_t* ring_p; int32_t* ring_o;  // LDS or null
+  int32_t ring_base;   // fact-log index of ring slot 0 (this round)
+};
+
+__device__ __forceinline__ void fx_adj_insert(int32_t* heads, uint32_t mask,
+                                              uint32_t key, uint32_t val,
+                                              FxCtx& cx, FxState& st) {
+  int32_t e = atomicAdd(cx.adj_n, 1);
+  if (e >= st.adj_cap) { atomicExch(cx.overflow, 1); return; }
+  __hip_atomic_store(
+      &st.adj_entries[e],
+      static_cast<int64_t>((static_cast<unsigned long long>(key) << 32) | val),
+      __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+  // lock-free push: next[e] must hold the correct predecessor BEFORE the
+  // head CAS makes e reachable, or a concurrent walker would read junk
+  uint32_t h = h32(key) & mask;
+  int32_t prev = fx_ld(&heads[h]);
+  for (;;) {
+    __hip_atomic_store(&st.adj_next[e], prev, __ATOMIC_RELAXED,
+                       __HIP_MEMORY_SCOPE_AGENT);
+    __threadfence_block();
+    int32_t seen = atomicCAS(&heads[h], prev, e);
+    if (seen == prev) break;
+    prev = seen;
+  }
+}
+
+// emit one derived fact: dedup -> fact log (= next delta) -> adjacencies
+__device__ __forceinline__ void fx_emit(int32_t pidx, uint32_t s, uint32_t o,
+                                        const FxPred* preds, FxCtx& cx,
+                                        FxState& st) {
+  const FxPred& pr = preds[pidx];
+  if (!fx_set_insert(pr, s, o, cx.overflow)) return;
+  int32_t idx = atomicAdd(cx.out_n, 1);
+  if (idx >= st.out_cap) { atomicExch(cx.overflow, 1); return; }
+  __hip_atomic_store(&st.out_s[idx], static_cast<int32_t>(s),
+                     __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+  __hip_atomic_store(&st.out_p[idx], pidx,
+                     __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+  __hip_atomic_store(&st.out_o[idx], static_cast<int32_t>(o),
+                     __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+  if (cx.ring_s) {
+    int32_t r = idx - cx.ring_base;
+    if (r >= 0 && r < kFxRing) {
+      cx.ring_s[r] = static_cast<int32_t>(s);
+      cx.ring_p[r] = pidx;
+      cx.ring_o[r] = static_cast<int32_t>(o);
+    }
+  }
+  if (pr.adj_s_heads)
+    fx_adj_insert(pr.adj_s_heads, pr.adj_s_mask, s, o, cx, st);
+  if (pr.adj_o_heads)
+    fx_adj_insert(pr.adj_o_heads, pr.adj_o_mask, o, s, cx, st);
+}
+
+__device__ __forceinline__ uint32_t fx_src(int src, uint32_t s1, uint32_t o1,
+                                           uint32_t s2, uint32_t o2) {
+  switch (src) {
+    case 0: return s1;
+    case 1: return o1;
+    case 2: return s2;
+    default: return o2;
+  }
+}
+
+// compile-time caps so ALL rule/dispatch metadata fits in LDS (a tiny
+// few KB): per-round reads of it then cost LDS cycles, not L2 latency —
+// at 10 000 one-fact rounds the metadata reads would otherwise dominate
+constexpr int kFxMaxRules = 64;
+constexpr int kFxMaxPreds = 16;
+constexpr int kFxMaxDisp = 128;
+
+__global__ void __launch_bounds__(kFxBlock)
+fx_kernel(const FxRule* __restrict__ g_rules, int n_rules,
+          const FxPred* __restrict__ g_preds, int n_preds,
+          const int32_t* __restrict__ g_disp_off,   // [n_preds+1]
+          const int32_t* __restrict__ g_disp_rule,  // rule index
+          const int32_t* __restrict__ g_disp_side,  // 0: fact is premise1
+          FxState st, int64_t n_init, int64_t adj_seeded,
+          int64_t max_rounds) {
+  __shared__ int32_t s_dlo, s_dhi, s_use_ring;
+  __shared__ int32_t sh_out_n, sh_adj_n, sh_overflow;
+  __shared__ FxRule rules[kFxMaxRules];
+  __shared__ FxPred preds[kFxMaxPreds];
+  __shared__ int32_t disp_off[kFxMaxPreds + 1];
+  __shared__ int32_t disp_rule[kFxMaxDisp];
+  __shared__ int32_t disp_side[kFxMaxDisp];
+  // double-buffered: this round's emits write parity buffer `round & 1`
+  // while readers consume the previous round's buffer
+  __shared__ int32_t ring_s[2][kFxRing], ring_p[2][kFxRing],
+      ring_o[2][kFxRing];
+  if (threadIdx.x == 0) {
+    s_dlo = 0;
+    s_dhi = static_cast<int32_t>(n_init);
+    s_use_ring = 0;
+    sh_out_n = static_cast<int32_t>(n_init);
+    sh_adj_n = static_cast<int32_t>(adj_seeded);
+    sh_overflow = 0;
+    for (int i = 0; i < n_rules; ++i) rules[i] = g_rules[i];
+    for (int i = 0; i < n_preds; ++i) preds[i] = g_preds[i];
+    int nd = g_disp_off[n_preds];
+    for (int i = 0; i <= n_preds; ++i) disp_off[i] = g_disp_off[i];
+    for (int i = 0; i < nd; ++i) {
+      disp_rule[i] = g_disp_rule[i];
+      disp_side[i] = g_disp_side[i];
+    }
+  }
+  __syncthreads();
+  int64_t rounds_done = 0;
+  for (int64_t round = 0; round < max_rounds; ++round) {
+    int32_t dlo = s_dlo, dhi = s_dhi;
+    if (dlo >= dhi) break;
+    bool ring_ok = s_use_ring != 0;
+    const int wb = static_cast<int>(round & 1), rb = wb ^ 1;
+    FxCtx cx{&sh_out_n, &sh_adj_n, &sh_overflow,
+             ring_s[wb], ring_p[wb], ring_o[wb], dhi};
+    for (int32_t i = dlo + threadIdx.x; i < dhi; i += blockDim.x) {
+      uint32_t fs, fo; int32_t fp;
+      if (ring_ok) {
+        fs = static_cast<uint32_t>(ring_s[rb][i - dlo]);
+        fp = ring_p[rb][i - dlo];
+        fo = static_cast<uint32_t>(ring_o[rb][i - dlo]);
+      } else {
+        fs = static_cast<uint32_t>(fx_ld(&st.out_s[i]));
+        fp = fx_ld(&st.out_p[i]);
+        fo = static_cast<uint32_t>(fx_ld(&st.out_o[i]));
+      }
+      for (int32_t d = disp_off[fp]; d < disp_off[fp + 1]; ++d) {
+        const FxRule& r = rules[disp_rule[d]];
+        int side = disp_side[d];
+        if (r.kind == 0) {
+          for (int c = 0; c < r.n_conc; ++c)
+            fx_emit(r.c_pred[c], fx_src(r.c_s_src[c], fs, fo, 0, 0),
+                    fx_src(r.c_o_src[c], fs, fo, 0, 0), preds, cx, st);
+          continue;
+        }
+        // join: this fact binds one premise; probe the other's adjacency
+        int other_p = side == 0 ? r.p2 : r.p1;
+        int my_j = side == 0 ? r.j1 : r.j2;
+        int ot_j = side == 0 ? r.j2 : r.j1;
+        uint32_t v = my_j == 0 ? fs : fo;   // shared-var value
+        const FxPred& op = preds[other_p];
+        int32_t* heads = ot_j == 0 ? op.adj_s_heads : op.adj_o_heads;
+        uint32_t mask = ot_j == 0 ? op.adj_s_mask : op.adj_o_mask;
+        if (!heads) continue;  // driver guarantees presence; safety
+        uint32_t h = h32(v) & mask;
+        for (int32_t e = fx_ld(&heads[h]); e >= 0;
+             e = fx_ld(&st.adj_next[e])) {
+          unsigned long long kv =
+              static_cast<unsigned long long>(fx_ld64(&st.adj_entries[e]));
+          if (static_cast<uint32_t>(kv >> 32) != v) continue;
+          uint32_t w = static_cast<uint32_t>(kv);
+          uint32_t s2, o2;
+          if (ot_j == 0) { s2 = v; o2 = w; } else { s2 = w; o2 = v; }
+          uint32_t s1, o1;
+          if (side == 0) { s1 = fs; o1 = fo; }
+          else { s1 = s2; o1 = o2; s2 = fs; o2 = fo; }
+          // (when side==1 the roles swap: this fact IS premise2)
+          for (int c = 0; c < r.n_conc; ++c)
+            fx_emit(r.c_pred[c], fx_src(r.c_s_src[c], s1, o1, s2, o2),
+                    fx_src(r.c_o_src[c], s1, o1, s2, o2), preds, cx, st);
+        }
+      }
+    }
+    __threadfence_block();
+    __syncthreads();
+    if (threadIdx.x == 0) {
+      s_dlo = dhi;
+      int32_t n = sh_out_n;
+      if (n > st.out_cap) n = static_cast<int32_t>(st.out_cap);
+      s_dhi = n;
+      // next round may read its delta from the LDS ring only if EVERY
+      // fact appended this round landed in it
+      s_use_ring = (n - dhi) <= kFxRing ? 1 : 0;
+    }
+    __syncthreads();
+    rounds_done = round + 1;
+    if (sh_overflow) break;  // overflow: host falls back
+  }
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    st.counters[0] = sh_out_n;
+    st.counters[1] = sh_adj_n;
+    st.counters[2] = sh_overflow;
+    st.counters[3] = static_cast<int32_t>(rounds_done);
+  }
+}
+
+// seed kernel: parallel insert of the initial facts (dedup + adjacency)
+__global__ void fx_seed(const int32_t* __restrict__ s,
+                        const int32_t* __restrict__ p,
+                        const int32_t* __restrict__ o, int64_t n,
+                        const FxPred* __restrict__ preds, FxState st) {
+  FxCtx cx{&st.counters[0], &st.counters[1], &st.counters[2],
+           nullptr, nullptr, nullptr, 0};
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    fx_emit(p[i], static_cast<uint32_t>(s[i]), static_cast<uint32_t>(o[i]),
+            preds, cx, st);
+  }
+}
+
+// host wrapper.  rules_flat: [n_rules,12] int32 rows =
+//   kind,p1,j1,p2,j2,n_conc,c0_pred,c0_s,c0_o,c1_pred,c1_s,c1_o
+// adj_need: [n_preds,2] bool-ish int32 (by_s, by_o)
+// facts: int32 columns with p already mapped to predicate indexes
+py::tuple small_fixpoint(at::Tensor rules_flat, at::Tensor adj_need,
+                         at::Tensor disp_off, at::Tensor disp_rule,
+                         at::Tensor disp_side,
+                         at::Tensor fs, at::Tensor fp, at::Tensor fo,
+                         int64_t budget, int64_t max_rounds) {
+  TORCH_CHECK(fs.is_cuda() && fs.dtype() == at::kInt);
+  const int64_t n_init = fs.numel();
+  const int n_rules = rules_flat.size(0);
+  const int n_preds = adj_need.size(0);
+  auto opts_i32 = fs.options();
+  auto opts_i64 = fs.options().dtype(at::kLong);
+  auto stream = cur_stream();
+
+  // ---- host-side structure assembly (tiny) ----
+  auto rules_h = rules_flat.to(at::kCPU).contiguous();
+  auto need_h = adj_need.to(at::kCPU).contiguous();
+  std::vector<FxRule> rules_v(n_rules);
+  const int32_t* rr = rules_h.data_ptr<int32_t>();
+  for (int i = 0; i < n_rules; ++i) {
+    const int32_t* q = rr + i * 12;
+    FxRule& r = rules_v[i];
+    r.kind = q[0]; r.p1 = q[1]; r.j1 = q[2]; r.p2 = q[3]; r.j2 = q[4];
+    r.n_conc = q[5];
+    r.c_pred[0] = q[6]; r.c_s_src[0] = q[7]; r.c_o_src[0] = q[8];
+    r.c_pred[1] = q[9]; r.c_s_src[1] = q[10]; r.c_o_src[1] = q[11];
+  }
+  int64_t set_cap = 64;
+  while (set_cap < 2 * (n_init + budget)) set_cap <<= 1;
+  int64_t head_cap = 64;
+  while (head_cap < n_init + budget) head_cap <<= 1;
+
+  std::vector<at::Tensor> keepalive;
+  std::vector<FxPred> preds_v(n_preds);
+  const int32_t* nd = need_h.data_ptr<int32_t>();
+  for (int i = 0; i < n_preds; ++i) {
+    auto set_t = at::full({set_cap}, -1, opts_i64);
+    keepalive.push_back(set_t);
+    preds_v[i].set = reinterpret_cast<unsigned long long*>(
+        set_t.data_ptr<int64_t>());
+    preds_v[i].set_mask = static_cast<uint32_t>(set_cap - 1);
+    preds_v[i].adj_s_heads = nullptr;
+    preds_v[i].adj_o_heads = nullptr;
+    preds_v[i].adj_s_mask = preds_v[i].adj_o_mask = 0;
+    if (nd[i * 2 + 0]) {
+      auto h = at::full({head_cap}, -1, opts_i32);
+      keepalive.push_back(h);
+      preds_v[i].adj_s_heads = h.data_ptr<int32_t>();
+      preds_v[i].adj_s_mask = static_cast<uint32_t>(head_cap - 1);
+    }
+    if (nd[i * 2 + 1]) {
+      auto h = at::full({head_cap}, -1, opts_i32);
+      keepalive.push_back(h);
+      preds_v[i].adj_o_heads = h.data_ptr<int32_t>();
+      preds_v[i].adj_o_mask = static_cast<uint32_t>(head_cap - 1);
+    }
+  }
+  const int64_t out_cap = n_init + budget;
+  const int64_t adj_cap = 2 * out_cap;
+  auto out_s = at::empty({out_cap}, opts_i32);
+  auto out_p = at::empty({out_cap}, opts_i32);
+  auto out_o = at::empty({out_cap}, opts_i32);
+  auto adj_e = at::empty({adj_cap}, opts_i64);
+  auto adj_n = at::empty({adj_cap}, opts_i32);
+  auto counters = at::zeros({8}, opts_i32);
+  FxState st;
+  st.adj_entries = adj_e.data_ptr<int64_t>();
+  st.adj_next = adj_n.data_ptr<int32_t>();
+  st.adj_pred_dir = nullptr;
+  st.out_s = out_s.data_ptr<int32_t>();
+  st.out_p = out_p.data_ptr<int32_t>();
+  st.out_o = out_o.data_ptr<int32_t>();
+  st.counters = counters.data_ptr<int32_t>();
+  st.out_cap = out_cap;
+  st.adj_cap = adj_cap;
+
+  // device copies of the rule/pred tables
+  auto rules_bytes = at::from_blob(rules_v.data(),
+      {static_cast<int64_t>(n_rules * sizeof(FxRule))},
+      at::TensorOptions().dtype(at::kByte)).clone().to(fs.device());
+  auto preds_bytes = at::from_blob(preds_v.data(),
+      {static_cast<int64_t>(n_preds * sizeof(FxPred))},
+      at::TensorOptions().dtype(at::kByte)).clone().to(fs.device());
+  keepalive.push_back(rules_bytes);
+  keepalive.push_back(preds_bytes);
+  auto d_off = disp_off.to(fs.device(), at::kInt).contiguous();
+  auto d_rule = disp_rule.to(fs.device(), at::kInt).contiguous();
+  auto d_side = disp_side.to(fs.device(), at::kInt).contiguous();
+
+  const FxPred* dp = reinterpret_cast<const FxPred*>(
+      preds_bytes.data_ptr<uint8_t>());
+  if (n_init > 0) {
+    hipLaunchKernelGGL(fx_seed, dim3(grid_for(n_init)), dim3(kBlock), 0,
+                       stream, fs.data_ptr<int32_t>(), fp.data_ptr<int32_t>(),
+                       fo.data_ptr<int32_t>(), n_init, dp, st);
+    HIP_OK(hipGetLastError());
+  }
+  // NOTE: the seed pass already appended the init facts to the log; the
+  // fixpoint's first delta is [0, counters[0]) == all seeded facts.
+  auto seed_cts = counters.to(at::kCPU);
+  int64_t seeded = seed_cts[0].item<int32_t>();
+  int64_t adj_seeded = seed_cts[1].item<int32_t>();
+  TORCH_CHECK(n_rules <= kFxMaxRules && n_preds <= kFxMaxPreds
+              && d_rule.numel() <= kFxMaxDisp,
+              "small_fixpoint: program exceeds LDS metadata caps");
+  hipLaunchKernelGGL(fx_kernel, dim3(1), dim3(kFxBlock), 0, stream,
+                     reinterpret_cast<const FxRule*>(
+                         rules_bytes.data_ptr<uint8_t>()),
+                     n_rules, dp, n_preds, d_off.data_ptr<int32_t>(),
+                     d_rule.data_ptr<int32_t>(), d_side.data_ptr<int32_t>(),
+                     st, seeded, adj_seeded, max_rounds);
+  HIP_OK(hipGetLastError());
+  auto cts = counters.to(at::kCPU);
+  int32_t n_out = cts[0].item<int32_t>();
+  int32_t overflow = cts[2].item<int32_t>();
+  int32_t rounds = cts[3].item<int32_t>();
+  if (n_out > out_cap) n_out = static_cast<int32_t>(out_cap);
+  py::tuple t(6);
+  t[0] = out_s.narrow(0, 0, n_out);
+  t[1] = out_p.narrow(0, 0, n_out);
+  t[2] = out_o.narrow(0, 0, n_out);
+  t[3] = py::int_(seeded);
+  t[4] = py::int_(overflow);
+  t[5] = py::int_(rounds);
+  return t;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("small_fixpoint", &small_fixpoint,
+        "K6 persistent single-workgroup semi-naive fixpoint for small "
+        "working sets (deep-taxonomy class) -> (s, p_idx, o, n_seeded, "
+        "overflow, rounds)");
   m.def("group_aggregate", &group_aggregate,
         "K4 LDS-staged hash GROUP BY: (keys, vals?, sum, min, max, empty, "
         "groups_hint) -> (group keys, counts, sums?, mins?, maxs?)");

@@ -372,10 +372,20 @@ def infer_fixpoint(rules: List[Rule], facts: FactStore, db,
     """Loop until no new facts (ref infer_generic.rs:27).  Returns the
     number of derived facts."""
     device = facts.device
-    # small working sets run the host hash path (deep-taxonomy-shaped
-    # workloads: thousands of tiny rounds where per-round device-op
-    # overhead dominates); large sets stay columnar on device
+    # Small working sets are launch-latency-bound on the columnar path
+    # (deep-taxonomy shape: thousands of tiny rounds).  Order of
+    # preference: (1) K6 persistent single-workgroup device kernel — the
+    # whole fixpoint in ONE launch, rounds are __syncthreads boundaries
+    # (eligible 1-2-premise const-predicate programs); (2) host hash
+    # fixpoint, kept as the fallback optimization for small programs the
+    # kernel's rule language doesn't cover; (3) columnar device rounds.
     from .host_fixpoint import HOST_PATH_MAX_FACTS, infer_fixpoint_host
+    if semi_naive and device.type == "cuda" \
+            and facts.n <= HOST_PATH_MAX_FACTS:
+        from .device_fixpoint import try_device_fixpoint
+        res = try_device_fixpoint(rules, facts, db)
+        if res is not None:
+            return res
     if semi_naive and facts.n <= HOST_PATH_MAX_FACTS:
         tuples = list(zip(facts.s.cpu().tolist(), facts.p.cpu().tolist(),
                           facts.o.cpu().tolist()))

@@ -499,3 +499,87 @@ def test_group_by_query_native_path():
         results[dev] = db.query(q)
     assert results["cpu"] == results["cuda:0"]
     assert len(results["cpu"]) == 13
+
+
+def _tc_reasoner(device, depth=60, chains=3):
+    from kolibrie_amd import Reasoner
+    from kolibrie_amd.reasoning.rule import Rule
+    from kolibrie_amd.storage.terms import Constant, TriplePattern, Variable
+    r = Reasoner(device=device)
+    sub = r._i32(r.dictionary.encode("sub"))
+    base = 1000
+    for c in range(chains):
+        for i in range(depth):
+            n = base + c * (depth + 1) + i
+            r.add_fact_ids(n, sub, n + 1)
+    r.add_rule(Rule(
+        premise=[TriplePattern(Variable("x"), Constant(sub), Variable("y")),
+                 TriplePattern(Variable("y"), Constant(sub), Variable("z"))],
+        conclusion=[TriplePattern(Variable("x"), Constant(sub), Variable("z"))],
+    ))
+    return r
+
+
+@requires_gpu
+def test_k6_small_fixpoint_closure_matches_cpu():
+    """Persistent single-WG fixpoint vs the CPU oracle on a closure."""
+    r_cpu = _tc_reasoner("cpu")
+    r_gpu = _tc_reasoner("cuda:0")
+    derived_gpu = r_gpu.infer_new_facts_semi_naive()
+    # the device kernel must actually run (not silently fall back)
+    assert getattr(r_gpu.facts, "k6_rounds", None) is not None, \
+        "K6 kernel did not engage"
+    derived_cpu = r_cpu.infer_new_facts_semi_naive()
+    assert derived_gpu == derived_cpu
+    a = sorted(zip(r_cpu.facts.s.tolist(), r_cpu.facts.p.tolist(),
+                   r_cpu.facts.o.tolist()))
+    b = sorted(zip(r_gpu.facts.s.cpu().tolist(), r_gpu.facts.p.cpu().tolist(),
+                   r_gpu.facts.o.cpu().tolist()))
+    assert a == b
+
+
+@requires_gpu
+def test_k6_deep_taxonomy_on_device():
+    import sys, os
+    sys.path.insert(0, os.path.dirname(os.path.dirname(
+        os.path.abspath(__file__))))
+    from scripts.bench_reasoning import build_deep_taxonomy
+    from kolibrie_amd.reasoning.device_fixpoint import try_device_fixpoint
+    r = build_deep_taxonomy(2000, "cuda:0")
+    derived = r.infer_new_facts_semi_naive()
+    assert derived == 2000
+    assert getattr(r.facts, "k6_rounds", None) is not None
+
+
+@requires_gpu
+def test_k6_copy_and_multi_conclusion_rules():
+    from kolibrie_amd import Reasoner
+    from kolibrie_amd.reasoning.rule import Rule
+    from kolibrie_amd.storage.terms import Constant, TriplePattern, Variable
+    results = {}
+    for device in ("cpu", "cuda:0"):
+        r = Reasoner(device=device)
+        e = r._i32(r.dictionary.encode("edge"))
+        k = r._i32(r.dictionary.encode("knows"))
+        f = r._i32(r.dictionary.encode("friend"))
+        for i in range(40):
+            r.add_fact_ids(100 + i, e, 100 + (i * 7 + 1) % 40)
+        r.add_rule(Rule(   # copy rule with two conclusions (one reversed)
+            premise=[TriplePattern(Variable("x"), Constant(e), Variable("y"))],
+            conclusion=[
+                TriplePattern(Variable("x"), Constant(k), Variable("y")),
+                TriplePattern(Variable("y"), Constant(f), Variable("x"))],
+        ))
+        r.add_rule(Rule(   # join over the derived relation
+            premise=[TriplePattern(Variable("a"), Constant(k), Variable("b")),
+                     TriplePattern(Variable("b"), Constant(f), Variable("c"))],
+            conclusion=[
+                TriplePattern(Variable("a"), Constant(f), Variable("c"))],
+        ))
+        derived = r.infer_new_facts_semi_naive()
+        if device != "cpu":
+            assert getattr(r.facts, "k6_rounds", None) is not None
+        results[device] = (derived, sorted(
+            zip(r.facts.s.cpu().tolist(), r.facts.p.cpu().tolist(),
+                r.facts.o.cpu().tolist())))
+    assert results["cpu"] == results["cuda:0"]
