@@ -204,8 +204,9 @@ def main():
                 "p50_ms": p50,
                 "plan_cache": "cold" if args.cold else "warm",
                 "timed_region": ("parse+plan+execute per step" if args.cold
-                                 else "prepared-plan serving (plan-cache "
-                                      "hit + captured-hipGraph replay)"),
+                                 else "prepared-plan serving (plan-cache hit "
+                                      "+ one C++ serve call: direct kernel "
+                                      "launches + pinned 8-byte readback)"),
                 "cold_ms_p50": cold_p50,
                 "shuffle_ms_per_step": shuffle_ms,
                 "bcast_ms_per_step": bcast_ms,

@@ -342,6 +342,11 @@ class ExecutionEngine:
         cache = getattr(op, "_chain_cache", None)
         if cache is not None and cache[0] == self.db.store.version:
             seed_key12, seed_z, hop_regions, native = cache[1:5]
+            serve = getattr(op, "_chain_serve", None)
+            if isinstance(serve, tuple):
+                # C++ serving path: one pybind call = launches + pinned
+                # readback (no torch dispatch, no graph-replay floor)
+                return serve[0].serve_chain_count(serve[1])
             graph = getattr(op, "_chain_graph", None)
             if isinstance(graph, tuple):
                 graph[0].replay()
@@ -440,6 +445,19 @@ class ExecutionEngine:
         if not seed_key12.is_cuda or getattr(op, "_chain_graph", False) is None:
             return
         import torch as _t
+        # preferred: C++ registry serving (direct launches beat a graph
+        # replay at this kernel count — MI355X guide §graph-replay-floor)
+        try:
+            from ..ops import native_for
+            native2 = native_for(seed_key12)
+            old = getattr(op, "_chain_serve", None)
+            if isinstance(old, tuple):
+                old[0].release_chain_serve(old[1])
+            sid = native2.register_chain_serve(seed_key12, seed_z, *hop_args)
+            op._chain_serve = (native2, sid)
+            return
+        except Exception:
+            op._chain_serve = None
         try:
             k = len(hop_args[0])
             n_tiles = (seed_key12.numel() + 255) // 256

@@ -243,11 +243,15 @@ def _run_prepared(pq: "PreparedQuery", db) -> List[List[str]]:
         if fast is not None:
             return fast
     if pq.count_only:
-        # serving hot loop: replay the captured hipGraph and read the
-        # 8-byte total — no engine construction, no plan walk
+        # serving hot loop: ONE C++ call (direct launches + pinned 8-byte
+        # readback), falling back to captured-hipGraph replay — no engine
+        # construction, no plan walk
         op = pq.physical
         cache = getattr(op, "_chain_cache", None)
         if cache is not None and cache[0] == db.store.version:
+            serve = getattr(op, "_chain_serve", None)
+            if isinstance(serve, tuple):
+                return [[str(serve[0].serve_chain_count(serve[1]))]]
             g = getattr(op, "_chain_graph", None)
             if isinstance(g, tuple):
                 g[0].replay()

@@ -624,7 +624,11 @@ at::Tensor build_count_table(at::Tensor packed_entries) {
 // narrow window of that hop's region (merge-path; computed in parallel).
 __global__ void chain_tile_bounds(const int64_t* __restrict__ seed_key12,
                                   int64_t m, int64_t n_tiles, ChainHops hops,
-                                  int64_t* __restrict__ win) {  // [n_tiles][k][2]
+                                  int64_t* __restrict__ win,  // [n_tiles][k][2]
+                                  unsigned long long* __restrict__ total_zero
+                                  ) {
+  if (total_zero && blockIdx.x == 0 && threadIdx.x == 0)
+    *total_zero = 0ULL;  // saves the serve path a hipMemsetAsync enqueue
   for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; t < n_tiles;
        t += (int64_t)gridDim.x * blockDim.x) {
     int64_t first_b = seed_key12[t * kTile] & 0xFFFFFFFFLL;
@@ -766,7 +770,8 @@ void chain_count_into(at::Tensor seed_key12, at::Tensor seed_z,
   hipLaunchKernelGGL(chain_tile_bounds, dim3(grid_for(n_tiles)),
                      dim3(kBlock), 0, stream,
                      seed_key12.data_ptr<int64_t>(), m, n_tiles, hops,
-                     win.data_ptr<int64_t>());
+                     win.data_ptr<int64_t>(),
+                     static_cast<unsigned long long*>(nullptr));
   HIP_OK(hipGetLastError());
   hipLaunchKernelGGL(chain_count_kernel, dim3(grid_for(m)), dim3(kBlock), 0,
                      stream, seed_key12.data_ptr<int64_t>(),
@@ -775,6 +780,101 @@ void chain_count_into(at::Tensor seed_key12, at::Tensor seed_z,
                      reinterpret_cast<unsigned long long*>(
                          total.data_ptr<int64_t>()));
   HIP_OK(hipGetLastError());
+}
+
+// ---- C++ serving path (crush the per-query host
+// floor).  All launch arguments for a cached COUNT chain live in a
+// C++-side registry; one pybind call then does: zero the accumulator,
+// two direct kernel launches, an 8-byte D2H into pinned memory and a
+// stream sync — no Python plan walk, no torch dispatcher, no graph
+// replay floor (direct launches are ~3-4 µs each vs 10-16 µs replay,
+// MI355X guide §graph-replay-floor).
+struct ChainServe {
+  at::Tensor seed_key12, seed_z, win, total;
+  std::vector<at::Tensor> keep;
+  ChainHops hops{};
+  int64_t m = 0, n_tiles = 0;
+  int64_t* pinned = nullptr;
+  ~ChainServe() {
+    if (pinned) (void)hipHostFree(pinned);
+  }
+};
+static std::vector<std::unique_ptr<ChainServe>> g_chain_serves;
+
+int64_t register_chain_serve(at::Tensor seed_key12, at::Tensor seed_z,
+                             std::vector<at::Tensor> hop_key12,
+                             std::vector<int64_t> hop_const_hi,
+                             std::vector<int64_t> hop_src,
+                             std::vector<at::Tensor> hop_table) {
+  TORCH_CHECK(seed_key12.is_cuda() && seed_z.is_cuda());
+  auto cs = std::make_unique<ChainServe>();
+  cs->seed_key12 = seed_key12;
+  cs->seed_z = seed_z;
+  cs->m = seed_key12.numel();
+  cs->hops.k = static_cast<int>(hop_key12.size());
+  for (size_t h = 0; h < hop_key12.size(); ++h) {
+    cs->keep.push_back(hop_key12[h]);
+    cs->keep.push_back(hop_table[h]);
+    cs->hops.key12[h] = hop_key12[h].data_ptr<int64_t>();
+    cs->hops.n[h] = hop_key12[h].numel();
+    cs->hops.const_hi[h] = hop_const_hi[h];
+    cs->hops.src[h] = static_cast<int32_t>(hop_src[h]);
+    if (hop_table[h].numel() > 0) {
+      cs->hops.table[h] = reinterpret_cast<const unsigned long long*>(
+          hop_table[h].data_ptr<int64_t>());
+      cs->hops.tmask[h] = hop_table[h].numel() - 1;
+    }
+  }
+  cs->n_tiles = (cs->m + kTile - 1) / kTile;
+  cs->win = at::empty({std::max<int64_t>(1, cs->n_tiles * cs->hops.k * 2)},
+                      seed_key12.options());
+  // total[0] = accumulator, total[1] = constant 1 (the completion flag's
+  // device source: same-stream copies are ordered, so once the flag
+  // lands in pinned memory the count before it is valid)
+  cs->total = at::zeros({2}, seed_key12.options());
+  cs->total[1] = 1;
+  HIP_OK(hipHostMalloc(reinterpret_cast<void**>(&cs->pinned), 16));
+  g_chain_serves.push_back(std::move(cs));
+  return static_cast<int64_t>(g_chain_serves.size() - 1);
+}
+
+int64_t serve_chain_count(int64_t id) {
+  auto& cs = *g_chain_serves.at(id);
+  if (cs.m == 0) return 0;
+  auto stream = cur_stream();
+  int64_t* total_ptr = cs.total.data_ptr<int64_t>();
+  volatile int64_t* flag = cs.pinned + 1;
+  *flag = 0;
+  hipLaunchKernelGGL(chain_tile_bounds, dim3(grid_for(cs.n_tiles)),
+                     dim3(kBlock), 0, stream,
+                     cs.seed_key12.data_ptr<int64_t>(), cs.m, cs.n_tiles,
+                     cs.hops, cs.win.data_ptr<int64_t>(),
+                     reinterpret_cast<unsigned long long*>(total_ptr));
+  hipLaunchKernelGGL(chain_count_kernel, dim3(grid_for(cs.m)), dim3(kBlock),
+                     0, stream, cs.seed_key12.data_ptr<int64_t>(),
+                     cs.seed_z.data_ptr<int32_t>(), cs.m, cs.hops,
+                     cs.win.data_ptr<int64_t>(),
+                     reinterpret_cast<unsigned long long*>(total_ptr));
+  HIP_OK(hipMemcpyAsync(cs.pinned, total_ptr, 16, hipMemcpyDeviceToHost,
+                        stream));
+  // spin on the pinned flag: hipStreamSynchronize pays tens of µs of
+  // scheduler yield latency, the DMA lands in ~µs.  Bounded: fall back
+  // to a real sync after ~2e9 spins (GPU hung elsewhere).
+  for (int64_t spins = 0; *flag == 0; ++spins) {
+#if defined(__x86_64__)
+    __builtin_ia32_pause();
+#endif
+    if (spins > 2000000000LL) {
+      HIP_OK(hipStreamSynchronize(stream));
+      break;
+    }
+  }
+  return cs.pinned[0];
+}
+
+void release_chain_serve(int64_t id) {
+  if (id >= 0 && id < static_cast<int64_t>(g_chain_serves.size()))
+    g_chain_serves[id].reset();
 }
 
 int64_t chain_count(at::Tensor seed_key12, at::Tensor seed_z,
@@ -812,7 +912,8 @@ int64_t chain_count(at::Tensor seed_key12, at::Tensor seed_z,
     hipLaunchKernelGGL(chain_tile_bounds, dim3(grid_for(n_tiles)),
                        dim3(kBlock), 0, stream,
                        seed_key12.data_ptr<int64_t>(), m, n_tiles, hops,
-                       win.data_ptr<int64_t>());
+                       win.data_ptr<int64_t>(),
+                       static_cast<unsigned long long*>(nullptr));
     HIP_OK(hipGetLastError());
     hipLaunchKernelGGL(chain_count_kernel, dim3(grid_for(m)), dim3(kBlock), 0,
                        stream, seed_key12.data_ptr<int64_t>(),
@@ -2087,6 +2188,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("chain_count_into", &chain_count_into,
         "allocation/sync-free chain count into caller buffers "
         "(hipGraph-capturable)");
+  m.def("register_chain_serve", &register_chain_serve,
+        "cache a COUNT chain's launch arguments C++-side; returns handle");
+  m.def("serve_chain_count", &serve_chain_count,
+        py::call_guard<py::gil_scoped_release>(),
+        "one-call serving: direct launches + pinned 8-byte readback");
+  m.def("release_chain_serve", &release_chain_serve,
+        "drop a cached serve handle (store changed)");
   m.def("build_count_table", &build_count_table,
         "open-addressing (value -> match count) table from packed "
         "(val<<32)|count entries, for hashed chain-count hops");
