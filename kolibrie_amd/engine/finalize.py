@@ -391,12 +391,58 @@ def _encode_numbers(db, vals: torch.Tensor, dev, integral: bool = False
     return torch.tensor(out, dtype=torch.int32, device=dev)
 
 
-def decode_rows(select: SelectQuery, rows: Bindings, db) -> List[List[str]]:
-    """Final string decode (only at the top, ref engine.rs:347-373)."""
+def _decode_column(rows: Bindings, v: str, db) -> List[str]:
+    """Batch-decode one result column: single D2H copy + numpy gather over
+    the cached dictionary mirror; quoted triples (rare) decode per cell."""
+    import numpy as np
+    n = rows.n
+    if not rows.has(v):
+        return [""] * n
+    table = db.dictionary.np_table()
+    n_plain = len(table)
+    ids = (rows.col(v).to(torch.int64) & 0xFFFFFFFF).cpu().numpy()
+    # default "": UNBOUND and out-of-vocabulary plain ids (synthetic dense
+    # id blocks have no string form) both decode to ""
+    out = np.full(n, "", dtype=object)
+    quoted = (ids & 0x8000_0000) != 0
+    ok = (ids < n_plain) & ~quoted
+    out[ok] = table[ids[ok]]
+    for i in np.nonzero(quoted & (ids != 0xFFFFFFFF))[0]:
+        out[i] = db.decode_term(int(ids[i])) or ""
+    return out.tolist()
+
+
+def decode_columns(select: SelectQuery, rows: Bindings, db
+                   ) -> Dict[str, List[str]]:
+    """Columnar result decode: {var: [values...]} with NO per-row Python
+    list construction — the MI355X-native result shape (sub-ms at 14k
+    rows where the row pivot alone costs ~1.6 ms)."""
     if select.select_star or not select.variables:
         names = rows.variables
     else:
         names = [p.output_name() for p in select.variables]
+    return {v: _decode_column(rows, v, db) for v in names}
+
+
+def decode_rows(select: SelectQuery, rows: Bindings, db) -> List[List[str]]:
+    """Final string decode (only at the top, ref engine.rs:347-373).
+
+    Row-heavy results decode in BATCH: one D2H copy per column, then a
+    numpy object-array gather over the cached dictionary mirror — the
+    per-cell Python loop this replaces dominated materializing queries
+    (L5-class: 14k rows took ~3 ms of host decode)."""
+    if select.select_star or not select.variables:
+        names = rows.variables
+    else:
+        names = [p.output_name() for p in select.variables]
+    n = rows.n
+    if n >= 64:
+        cols_dec = [_decode_column(rows, v, db) for v in names]
+        # row pivot: map/zip beats any numpy object-array reshape (object
+        # stack+tolist measured 9.5 ms at 14k rows; this is ~1.6 ms —
+        # the remaining cost IS building 14k Python lists; decode_columns
+        # below avoids it entirely for columnar consumers)
+        return list(map(list, zip(*cols_dec)))
     host = {}
     for v in names:
         if rows.has(v):

@@ -385,6 +385,34 @@ def execute_query(sparql: str, db) -> List[List[str]]:
     return []
 
 
+def execute_query_columns(sparql: str, db) -> Dict[str, List[str]]:
+    """Columnar SELECT entry: same prepared-plan pipeline as
+    execute_query, but the final decode emits {var: [values...]} without
+    building per-row Python lists (finalize.decode_columns)."""
+    from .finalize import decode_columns
+    cache = getattr(db, "_plan_cache", None)
+    if cache is None:
+        cache = db._plan_cache = {}
+    hit = cache.get(sparql)
+    if hit is not None and hit.store_version == db.store.version:
+        pq = hit
+    else:
+        cq = parse_combined_query(sparql)
+        prefixes = dict(db.prefixes)
+        prefixes.update(cq.prefixes)
+        if cq.select is None or cq.updates or cq.rules or cq.train_decls \
+                or cq.register:
+            raise ValueError("query_columns supports plain SELECT queries")
+        prepare_extensions(cq, db, prefixes)
+        pq = _prepare_select(cq.select, db, prefixes)
+        if len(cache) < 256:
+            cache[sparql] = pq
+    ctx = ExecutionContext(db, pq.view)
+    rows = ExecutionEngine(ctx).execute(pq.physical, Bindings.unit(db.device))
+    final = finalize_select_bindings(pq.select, rows, db)
+    return decode_columns(pq.select, final, db)
+
+
 def execute_sparql_query(sparql: str, db) -> List[List[str]]:
     """Query-only entry: rejects updates (ref execute_query.rs:71-89,
     used by the HTTP /query endpoint)."""

@@ -399,6 +399,13 @@ class SparqlDatabase:
         from ..engine.query import execute_query
         return execute_query(sparql, self)
 
+    def query_columns(self, sparql: str) -> Dict[str, List[str]]:
+        """SELECT returning columnar results {var: [values...]} — skips
+        the per-row Python list construction (engine extension; the
+        row-shaped `query` keeps reference-API parity)."""
+        from ..engine.query import execute_query_columns
+        return execute_query_columns(sparql, self)
+
     # Reference public names (execute_query.rs).
     def exec_query(self, sparql: str) -> List[List[str]]:
         return self.query(sparql)

@@ -35,7 +35,8 @@ class Dictionary:
     """Bidirectional string<->id map.  ID 0 is reserved for "" (and doubles
     as the default-graph ID)."""
 
-    __slots__ = ("str_to_id", "id_to_str", "values", "_values_dirty")
+    __slots__ = ("str_to_id", "id_to_str", "values", "_values_dirty",
+                 "_np_table", "_np_table_n")
 
     def __init__(self):
         self.str_to_id: Dict[str, int] = {"": 0}
@@ -44,6 +45,24 @@ class Dictionary:
         # non-numeric) — the device value column source.
         self.values: List[float] = [0.0]
         self._values_dirty = True
+        self._np_table = None     # numpy object array mirror of id_to_str
+        self._np_table_n = 0
+
+    def np_table(self):
+        """Numpy object-array mirror of id_to_str for BATCH decode
+        (decode_rows' per-cell Python loop was the materializing-query
+        bottleneck — VERDICT r1 item 6).  Grown incrementally: the
+        dictionary is append-only."""
+        import numpy as np
+        n = len(self.id_to_str)
+        if self._np_table is None:
+            self._np_table = np.array(self.id_to_str, dtype=object)
+            self._np_table_n = n
+        elif self._np_table_n < n:
+            tail = np.array(self.id_to_str[self._np_table_n:], dtype=object)
+            self._np_table = np.concatenate([self._np_table, tail])
+            self._np_table_n = n
+        return self._np_table
 
     def __len__(self):
         return len(self.id_to_str)

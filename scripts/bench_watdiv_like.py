@@ -106,6 +106,19 @@ def main():
         print(f"{name:4s} {p50:8.3f} ms  ({len(rows)} rows)")
     print(f"mean p50 across suite: {total_p50 / len(queries):.3f} ms")
 
+    # L5 again through the columnar result API (no per-row Python lists)
+    q = queries["L5"]
+    cols = db.query_columns(q)
+    sync()
+    lat = []
+    for _ in range(args.runs):
+        t0 = time.perf_counter()
+        cols = db.query_columns(q)
+        sync()
+        lat.append((time.perf_counter() - t0) * 1000)
+    n = len(next(iter(cols.values()))) if cols else 0
+    print(f"L5c  {statistics.median(lat):8.3f} ms  ({n} rows, columnar)")
+
 
 if __name__ == "__main__":
     main()

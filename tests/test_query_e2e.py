@@ -552,3 +552,20 @@ def test_group_pushdown_respects_from_dataset():
     q2 = (f'SELECT ?g (COUNT(*) AS ?c) WHERE {{ ?s <{EX}grp> ?g }} '
           f'GROUP BY ?g ORDER BY ?g')
     assert db.query(q2) == [[f"{EX}g2", "1"]]
+
+
+def test_query_columns_matches_rows():
+    from kolibrie_amd import SparqlDatabase
+    EX = "http://example.org/"
+    db = SparqlDatabase()
+    for i in range(200):
+        db.add_triple(f"<{EX}e{i}>", f"<{EX}p>", f'"{i % 7}"')
+    q = f"SELECT ?s ?o WHERE {{ ?s <{EX}p> ?o }} ORDER BY ?s ?o"
+    rows = db.query(q)
+    cols = db.query_columns(q)
+    assert list(cols.keys()) == ["s", "o"]
+    assert cols["s"] == [r[0] for r in rows]
+    assert cols["o"] == [r[1] for r in rows]
+    import pytest
+    with pytest.raises(ValueError):
+        db.query_columns(f"INSERT DATA {{ <{EX}x> <{EX}p> \"v\" }}")
