@@ -33,6 +33,10 @@ from .scan import scan_unit
 from .tensor_utils import group_index, merge_join_indices
 from . import exec_stats
 
+import logging
+
+_log = logging.getLogger("kolibrie_amd.engine")
+
 
 @dataclass
 class DatasetView:
@@ -457,6 +461,8 @@ class ExecutionEngine:
             op._chain_serve = (native2, sid)
             return
         except Exception:
+            _log.warning("chain-serve registration failed; falling back "
+                         "to hipGraph capture", exc_info=True)
             op._chain_serve = None
         try:
             k = len(hop_args[0])
@@ -476,6 +482,8 @@ class ExecutionEngine:
                                         win, total)
             op._chain_graph = (g, total, win)
         except Exception:
+            _log.warning("hipGraph capture failed; cached COUNT path will "
+                         "use plain launches", exc_info=True)
             op._chain_graph = None  # permanent fallback marker
 
     # hop regions much smaller than the seed count pay log2(n) L2 lines per

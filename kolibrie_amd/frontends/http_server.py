@@ -18,12 +18,15 @@ the compute path, not the socket loop).  CORS is open like the reference.
 
 import asyncio
 import json
+import logging
 import os
 import threading
 import uuid
 from typing import Dict, List, Optional
 
 from ..storage.database import SparqlDatabase
+
+_log = logging.getLogger("kolibrie_amd.http")
 
 MAX_BODY = 16 * 1024 * 1024  # request size cap (ref main.rs:432)
 
@@ -111,6 +114,7 @@ def create_app(db: Optional[SparqlDatabase] = None, device: str = "cpu"):
                 else [f"v{i}" for i in range(len(rows[0]))] if rows else []
             return JSONResponse(sparql_json_results(names, rows))
         except ValueError as e:
+            _log.warning("query rejected: %s", e)
             raise HTTPException(400, str(e))
 
     @app.post("/update")
@@ -121,6 +125,7 @@ def create_app(db: Optional[SparqlDatabase] = None, device: str = "cpu"):
                 state_db.query(sparql)
             return JSONResponse({"status": "ok"})
         except ValueError as e:
+            _log.warning("update rejected: %s", e)
             raise HTTPException(400, str(e))
 
     @app.post("/rsp-query")

@@ -148,14 +148,24 @@ class SimpleR2R(R2ROperator):
     def execute_query(self, plan) -> List[Tuple[str, ...]]:
         """Run a prepared physical plan over the store, returning sorted
         decoded rows (ref simple_r2r.rs:201-214)."""
+        from ..engine.finalize import decode_rows
+        select, final = self.execute_query_bindings(plan)
+        return sorted(tuple(r) for r in decode_rows(select, final, self.db))
+
+    def execute_query_bindings(self, plan):
+        """Columnar variant: returns (select, finalized Bindings) WITHOUT
+        decoding — the engine's single-window emission keeps results as
+        device columns through R2S (K10 rows_diff) and decodes only the
+        emitted Δ."""
         from ..engine.bindings import Bindings
-        from ..engine.executor import DatasetView, ExecutionContext, ExecutionEngine
-        from ..engine.finalize import decode_rows, finalize_select_bindings
+        from ..engine.executor import (DatasetView, ExecutionContext,
+                                       ExecutionEngine)
+        from ..engine.finalize import finalize_select_bindings
         select, physical = plan
         ctx = ExecutionContext(self.db, DatasetView())
-        rows = ExecutionEngine(ctx).execute(physical, Bindings.unit(self.db.device))
-        final = finalize_select_bindings(select, rows, self.db)
-        return sorted(tuple(r) for r in decode_rows(select, final, self.db))
+        rows = ExecutionEngine(ctx).execute(physical,
+                                            Bindings.unit(self.db.device))
+        return select, finalize_select_bindings(select, rows, self.db)
 
     def execute_sparql(self, sparql: str) -> List[List[str]]:
         return self.db.query(sparql)

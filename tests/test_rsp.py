@@ -556,3 +556,44 @@ def test_istream_same_sp_diff_object_counts_both():
         entry.window.flush()
     flat = sorted(v for rows in got for r in rows for v in r)
     assert flat == ["a", "b"]
+
+
+def test_single_window_emission_is_columnar():
+    """The single-window emission path must run R2S over device/CPU
+    COLUMNS (K10 rows_diff), not host tuple sets."""
+    from kolibrie_amd.rsp.builder import RSPBuilder
+    q = """
+        REGISTER ISTREAM <out> AS
+        SELECT ?s ?o FROM NAMED WINDOW <w> ON STREAM <s1> [RANGE 10 STEP 2]
+        WHERE { WINDOW <w> { ?s <http://t/p> ?o } }
+    """
+    got = []
+    eng = (RSPBuilder().add_rsp_ql_query(q)
+           .add_consumer(lambda rows: got.append(list(rows))).build())
+    for ts in range(8):
+        eng.add_to_stream("<s1>", (f"<http://t/e{ts}>", "<http://t/p>",
+                                   f"<http://t/o{ts % 3}>"), ts)
+    assert eng.r2s.previous_cols is not None, \
+        "columnar R2S path did not engage"
+    assert not eng.r2s.previous, "host-set path should be unused"
+    flat = [v for rows in got for r in rows for v in r]
+    assert any("e0" in v for v in flat)
+
+
+def test_columnar_istream_dstream_match_host_semantics():
+    """eval_columns must agree with the host-set eval on random firings."""
+    import torch
+    from kolibrie_amd.rsp.r2s import Relation2StreamOperator, StreamOperator
+    torch.manual_seed(3)
+    for mode in (StreamOperator.ISTREAM, StreamOperator.DSTREAM):
+        host = Relation2StreamOperator(mode)
+        dev = Relation2StreamOperator(mode)
+        for _ in range(6):
+            n = int(torch.randint(0, 50, (1,)))
+            a = torch.randint(0, 8, (n,), dtype=torch.int32)
+            b = torch.randint(0, 8, (n,), dtype=torch.int32)
+            host_out = host.eval([(int(x), int(y)) for x, y in zip(a, b)])
+            cols = dev.eval_columns([a, b])
+            dev_out = sorted(zip(cols[0].tolist(), cols[1].tolist())) \
+                if cols else []
+            assert sorted(set(host_out)) == dev_out, (mode, host_out, dev_out)
