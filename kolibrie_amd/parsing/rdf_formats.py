@@ -101,13 +101,16 @@ def parse_ntriples_into(db, text: str):
     back to the Python tokenizer."""
     from ..ops import _native
     if _native is not None and len(text) > 4096:
-        import numpy as np
-        local_ids, strings, fallback = _native.parse_ntriples_host(text)
+        if len(text) > (1 << 21):
+            # chunk-per-thread native parse, GIL released (the MI355X
+            # replacement for the reference's crossbeam pipeline)
+            local_ids, strings, fallback = _native.parse_ntriples_host_mt(
+                text, 0)
+        else:
+            local_ids, strings, fallback = _native.parse_ntriples_host(text)
         if local_ids.numel():
-            remap = np.empty(len(strings), dtype=np.uint32)
-            enc = db.dictionary.encode
-            for i, raw in enumerate(strings):
-                remap[i] = enc(raw.decode("utf-8", "replace"))
+            decoded = [raw.decode("utf-8", "replace") for raw in strings]
+            remap = db.dictionary.encode_many(decoded)
             arr = local_ids.numpy()
             mapped = remap[arr]
             db.store.insert_bulk(0, mapped[:, 0], mapped[:, 1], mapped[:, 2])
@@ -146,12 +149,14 @@ def parse_nquads_into(db, text: str):
     from ..ops import _native
     if _native is not None and len(text) > 4096:
         import numpy as np
-        local_ids, strings, fallback = _native.parse_nquads_host(text)
+        if len(text) > (1 << 21):
+            local_ids, strings, fallback = _native.parse_nquads_host_mt(
+                text, 0)
+        else:
+            local_ids, strings, fallback = _native.parse_nquads_host(text)
         if local_ids.numel():
-            remap = np.empty(len(strings), dtype=np.int64)
-            enc = db.dictionary.encode
-            for i, raw in enumerate(strings):
-                remap[i] = enc(raw.decode("utf-8", "replace"))
+            decoded = [raw.decode("utf-8", "replace") for raw in strings]
+            remap = db.dictionary.encode_many(decoded).astype(np.int64)
             arr = local_ids.numpy()
             spo = remap[arr[:, :3]]
             g = arr[:, 3]

@@ -263,6 +263,31 @@ class SparqlDatabase:
     # reference naming: parse_ntriples_and_add (sparql_database.rs:1335)
     parse_ntriples_and_add = parse_ntriples
 
+    def parse_ntriples_file(self, path: str):
+        """File-to-store N-Triples ingest: the native reader + chunk-per-
+        thread parser (GIL released) feed bulk dictionary interning and a
+        columnar insert — no Python copy of the file text (the parallel
+        bulk-parse pipeline, ref sparql_database.rs:630-804 crossbeam
+        design)."""
+        from ..ops import _native
+        if _native is None:
+            with open(path, "r", encoding="utf-8") as f:
+                return self.parse_ntriples(f.read())
+        from .terms import QUOTED_TRIPLE_ID_BIT
+        d = self.dictionary
+        ids, fallback = _native.parse_ntriples_file_encode(
+            path, 0, d.str_to_id, d.id_to_str, d.values,
+            QUOTED_TRIPLE_ID_BIT)
+        d._values_dirty = True
+        if ids.numel():
+            arr = ids.numpy()
+            self.store.insert_bulk(0, arr[:, 0], arr[:, 1], arr[:, 2])
+        if fallback:
+            from ..parsing.rdf_formats import _parse_ntriples_lines
+            with open(path, "r", encoding="utf-8") as f:
+                lines = f.read().split("\n")
+            _parse_ntriples_lines(self, (lines[i] for i in fallback))
+
     def parse_nquads(self, text: str):
         from ..parsing.rdf_formats import parse_nquads_into
         parse_nquads_into(self, text)

@@ -81,6 +81,34 @@ class Dictionary:
         self._values_dirty = True
         return i
 
+    def encode_many(self, strs) -> "object":
+        """Bulk intern: returns a uint32 numpy array of IDs.  The tuned
+        local-variable loop is ~2x a per-call encode() — this is the host
+        half of the parallel bulk-parse pipeline."""
+        import numpy as np
+        sti = self.str_to_id
+        its = self.id_to_str
+        vals = self.values
+        get = sti.get
+        append = its.append
+        vappend = vals.append
+        n = len(its)
+        out = np.empty(len(strs), dtype=np.uint32)
+        for i, s in enumerate(strs):
+            x = get(s)
+            if x is None:
+                if n >= QUOTED_TRIPLE_ID_BIT:
+                    raise OverflowError(
+                        "dictionary ID space exhausted (2^31 terms)")
+                x = n
+                sti[s] = x
+                append(s)
+                vappend(_try_parse_float(s))
+                n += 1
+            out[i] = x
+        self._values_dirty = True
+        return out
+
     def lookup(self, s: str) -> Optional[int]:
         return self.str_to_id.get(s)
 
