@@ -403,10 +403,17 @@ def evaluate_hybrid(lineage: LineageStore, node: int,
         if clock.now() > deadline:
             break
 
-    # stage 3: exact SDD WMC (ref compile_lineage_to_sdd:1248)
+    # stage 3: exact SDD WMC (ref compile_lineage_to_sdd:1248).  Larger
+    # seed sets compile on a BALANCED vtree (sdd_vtree.VtreeSddManager —
+    # the general-vtree engine); tiny ones keep the
+    # right-linear OBDD manager whose layout diff_sdd also understands.
     metrics.escalated = True
     t_sdd = clock.now()
-    manager = SddManager()
+    if len(weights) >= 8:
+        from .sdd_vtree import VtreeSddManager
+        manager = VtreeSddManager("balanced", sorted(weights))
+    else:
+        manager = SddManager()
     budget = SddOperationBudget(max_nodes=cfg.sdd_node_cap,
                                 deadline_s=cfg.sdd_budget_ms / 1000.0)
     sdd_node = lineage.to_sdd(node, manager, weights, budget)
