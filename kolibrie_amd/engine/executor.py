@@ -501,6 +501,12 @@ class ExecutionEngine:
             return _t.empty(0, dtype=_t.int64, device=region.device)
         vals, counts = _t.unique_consecutive(region & 0xFFFFFFFF,
                                              return_counts=True)
+        if vals.numel() and int(vals.max().item()) < 0x1FFFFFF \
+                and int(counts.max().item()) < 128:
+            # packed (val<<7)|count u32 table: half the per-probe bytes
+            # and table footprint (L2 residency beside the seed stream)
+            return native.build_count_table32(
+                ((vals << 7) | counts).to(_t.int32))
         return native.build_count_table((vals << 32) | counts)
 
     def _chain_count_torch(self, seed_key12, seed_z, hop_regions) -> int:

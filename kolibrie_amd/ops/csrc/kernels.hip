@@ -574,10 +574,15 @@ struct ChainHops {
   int64_t n[kMaxHops];
   int64_t const_hi[kMaxHops];      // high 32 bits of the probe key
   int32_t src[kMaxHops];           // 0: seed B component, 1: seed Z column
-  const unsigned long long* table[kMaxHops];  // optional count table
+  const unsigned long long* table[kMaxHops];  // optional count table (u64)
+  const uint32_t* table32[kMaxHops];  // packed (val<<7)|count u32 variant
   int64_t tmask[kMaxHops];
   int k;
 };
+
+// packed-u32 count table: halves the table footprint (better L2 residency
+// beside the seed stream) when every value < 2^25 and count < 2^7
+constexpr uint32_t kTbl32Empty = 0xFFFFFFFFu;
 
 // ---- per-hop COUNT tables -------------------------------------------------
 // A small hop region (e.g. the 142k-row locatedIn slice under 14.2M seeds)
@@ -604,6 +609,38 @@ __global__ void count_table_insert(const int64_t* __restrict__ entries,
     while (atomicCAS(&table[slot], kTblEmpty, e) != kTblEmpty)
       slot = (slot + 1) & mask;
   }
+}
+
+__global__ void count_table_insert32(const int32_t* __restrict__ entries,
+                                     int64_t u,
+                                     uint32_t* __restrict__ table,
+                                     uint32_t mask) {
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < u;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint32_t e = static_cast<uint32_t>(entries[i]);
+    uint32_t slot = h32(e >> 7) & mask;
+    while (atomicCAS(&table[slot], kTbl32Empty, e) != kTbl32Empty)
+      slot = (slot + 1) & mask;
+  }
+}
+
+at::Tensor build_count_table32(at::Tensor packed_entries) {
+  TORCH_CHECK(packed_entries.is_cuda()
+              && packed_entries.dtype() == at::kInt);
+  int64_t u = packed_entries.numel();
+  int64_t size = 64;
+  while (size < 2 * u) size <<= 1;
+  auto table = at::full({size}, -1, packed_entries.options());
+  if (u > 0) {
+    hipLaunchKernelGGL(count_table_insert32, dim3(grid_for(u)), dim3(kBlock),
+                       0, cur_stream(), packed_entries.data_ptr<int32_t>(),
+                       u,
+                       reinterpret_cast<uint32_t*>(
+                           table.data_ptr<int32_t>()),
+                       static_cast<uint32_t>(size - 1));
+    HIP_OK(hipGetLastError());
+  }
+  return table;
 }
 
 at::Tensor build_count_table(at::Tensor packed_entries) {
@@ -640,7 +677,7 @@ __global__ void chain_tile_bounds(const int64_t* __restrict__ seed_key12,
     int64_t last_b = seed_key12[min((t + 1) * kTile - 1, m - 1)] & 0xFFFFFFFFLL;
     for (int h = 0; h < hops.k; ++h) {
       int64_t lo = 0, hi = hops.n[h];
-      if (hops.table[h] != nullptr) {
+      if (hops.table[h] != nullptr || hops.table32[h] != nullptr) {
         lo = hi = 0;  // hashed hop: window unused
       } else if (hops.src[h] == 0) {
         lo = lower_bound_i64(hops.key12[h], hops.n[h],
@@ -683,6 +720,25 @@ __global__ void chain_count_kernel(const int64_t* __restrict__ seed_key12,
     // L2 latency
     for (int h = 0; h < hops.k; ++h) {
       int64_t comp = hops.src[h] == 0 ? b_comp : z_comp;
+      if (hops.table32[h] != nullptr) {
+        if (!active) continue;
+        // packed count-table hop: one 4-byte L2 load
+        uint32_t v = static_cast<uint32_t>(comp);
+        uint32_t mask = static_cast<uint32_t>(hops.tmask[h]);
+        uint32_t slot = h32(v) & mask;
+        unsigned long long cnt = 0;
+        for (;;) {
+          uint32_t e = hops.table32[h][slot];
+          if (e == kTbl32Empty) break;
+          if ((e >> 7) == v) {
+            cnt = e & 0x7Fu;
+            break;
+          }
+          slot = (slot + 1) & mask;
+        }
+        prod *= cnt;
+        continue;
+      }
       if (hops.table[h] != nullptr) {
         if (!active) continue;
         // count-table hop: one 8-byte L2 load (vs log2(n) lines)
@@ -762,8 +818,13 @@ void chain_count_into(at::Tensor seed_key12, at::Tensor seed_z,
     hops.const_hi[h] = hop_const_hi[h];
     hops.src[h] = static_cast<int32_t>(hop_src[h]);
     if (hop_table[h].numel() > 0) {
-      hops.table[h] = reinterpret_cast<const unsigned long long*>(
-          hop_table[h].data_ptr<int64_t>());
+      if (hop_table[h].dtype() == at::kInt) {
+        hops.table32[h] = reinterpret_cast<const uint32_t*>(
+            hop_table[h].data_ptr<int32_t>());
+      } else {
+        hops.table[h] = reinterpret_cast<const unsigned long long*>(
+            hop_table[h].data_ptr<int64_t>());
+      }
       hops.tmask[h] = hop_table[h].numel() - 1;
     }
   }
@@ -825,8 +886,13 @@ int64_t register_chain_serve(at::Tensor seed_key12, at::Tensor seed_z,
     cs->hops.const_hi[h] = hop_const_hi[h];
     cs->hops.src[h] = static_cast<int32_t>(hop_src[h]);
     if (hop_table[h].numel() > 0) {
-      cs->hops.table[h] = reinterpret_cast<const unsigned long long*>(
-          hop_table[h].data_ptr<int64_t>());
+      if (hop_table[h].dtype() == at::kInt) {
+        cs->hops.table32[h] = reinterpret_cast<const uint32_t*>(
+            hop_table[h].data_ptr<int32_t>());
+      } else {
+        cs->hops.table[h] = reinterpret_cast<const unsigned long long*>(
+            hop_table[h].data_ptr<int64_t>());
+      }
       cs->hops.tmask[h] = hop_table[h].numel() - 1;
     }
   }
@@ -903,9 +969,15 @@ int64_t chain_count(at::Tensor seed_key12, at::Tensor seed_z,
     hops.src[h] = static_cast<int32_t>(hop_src[h]);
     if (hop_table[h].numel() > 0) {
       TORCH_CHECK(hop_table[h].is_cuda()
-                  && hop_table[h].dtype() == at::kLong);
-      hops.table[h] = reinterpret_cast<const unsigned long long*>(
-          hop_table[h].data_ptr<int64_t>());
+                  && (hop_table[h].dtype() == at::kLong
+                      || hop_table[h].dtype() == at::kInt));
+      if (hop_table[h].dtype() == at::kInt) {
+        hops.table32[h] = reinterpret_cast<const uint32_t*>(
+            hop_table[h].data_ptr<int32_t>());
+      } else {
+        hops.table[h] = reinterpret_cast<const unsigned long long*>(
+            hop_table[h].data_ptr<int64_t>());
+      }
       hops.tmask[h] = hop_table[h].numel() - 1;
     }
   }
@@ -2627,6 +2699,9 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "one-call serving: direct launches + pinned 8-byte readback");
   m.def("release_chain_serve", &release_chain_serve,
         "drop a cached serve handle (store changed)");
+  m.def("build_count_table32", &build_count_table32,
+        "packed-u32 (val<<7)|count open-addressing table (half the L2 "
+        "footprint; requires val < 2^25, count < 128)");
   m.def("build_count_table", &build_count_table,
         "open-addressing (value -> match count) table from packed "
         "(val<<32)|count entries, for hashed chain-count hops");
