@@ -356,8 +356,8 @@ class ExecutionEngine:
                 graph[0].replay()
                 return int(graph[1].item())
             if native is not None:
-                cnt = int(native.chain_count(seed_key12, seed_z, *cache[5]))
-                self._maybe_capture_chain_graph(op, native, seed_key12,
+                cnt = int(native.chain_count(cache[6], seed_z, *cache[5]))
+                self._maybe_capture_chain_graph(op, native, cache[6],
                                                 seed_z, cache[5])
                 return cnt
             return self._chain_count_torch(seed_key12, seed_z, hop_regions)
@@ -421,6 +421,7 @@ class ExecutionEngine:
             rhi = int(_t.searchsorted(key12, khi, side="right").item())
             hop_regions.append((key12[rlo:rhi], pid, src))
         from ..ops import native_for
+        import torch as _t
         native = native_for(seed_key12)
         seed_key12 = seed_key12.contiguous()
         seed_z = seed_z.contiguous()
@@ -428,36 +429,43 @@ class ExecutionEngine:
             (r[0].contiguous(), r[1], r[2],
              self._chain_hop_table(native, r[0], seed_key12.numel()))
             for r in hop_regions]
-        hop_args = ([r[0] for r in hop_regions],
-                    [r[1] for r in hop_regions],
+        # native kernels stream 32-bit columns: the region high words are
+        # each hop's constant predicate and the seed key's low word is the
+        # probe component — dropping the high halves halves the HBM bytes
+        # and doubles the LDS window capacity (u32-order preserved)
+        seed_b = (seed_key12 & 0xFFFFFFFF).to(_t.int32).contiguous()
+        hop_args = ([(r[0] & 0xFFFFFFFF).to(_t.int32).contiguous()
+                     if r[3].numel() == 0 else
+                     _t.empty(0, dtype=_t.int32, device=seed_b.device)
+                     for r in hop_regions],
                     [r[2] for r in hop_regions],
                     [r[3] for r in hop_regions])
         op._chain_cache = (self.db.store.version, seed_key12, seed_z,
-                           hop_regions, native, hop_args)
+                           hop_regions, native, hop_args, seed_b)
         if hasattr(op, "_chain_graph"):
             del op._chain_graph  # stale capture from a previous store version
         if native is not None:
-            return int(native.chain_count(seed_key12, seed_z, *hop_args))
+            return int(native.chain_count(seed_b, seed_z, *hop_args))
         return self._chain_count_torch(seed_key12, seed_z, hop_regions)
 
-    def _maybe_capture_chain_graph(self, op, native, seed_key12, seed_z,
+    def _maybe_capture_chain_graph(self, op, native, seed_b, seed_z,
                                    hop_args):
         """Record the tile_bounds + chain_count launch pair into a hipGraph
         (torch.cuda.CUDAGraph IS hipGraph on ROCm): the cached path becomes
         one graph replay + one 8-byte read-back.  Capture failures fall
         back permanently to plain launches."""
-        if not seed_key12.is_cuda or getattr(op, "_chain_graph", False) is None:
+        if not seed_b.is_cuda or getattr(op, "_chain_graph", False) is None:
             return
         import torch as _t
         # preferred: C++ registry serving (direct launches beat a graph
         # replay at this kernel count — MI355X guide §graph-replay-floor)
         try:
             from ..ops import native_for
-            native2 = native_for(seed_key12)
+            native2 = native_for(seed_b)
             old = getattr(op, "_chain_serve", None)
             if isinstance(old, tuple):
                 old[0].release_chain_serve(old[1])
-            sid = native2.register_chain_serve(seed_key12, seed_z, *hop_args)
+            sid = native2.register_chain_serve(seed_b, seed_z, *hop_args)
             op._chain_serve = (native2, sid)
             return
         except Exception:
@@ -466,19 +474,19 @@ class ExecutionEngine:
             op._chain_serve = None
         try:
             k = len(hop_args[0])
-            n_tiles = (seed_key12.numel() + 255) // 256
+            n_tiles = (seed_b.numel() + 255) // 256
             win = _t.empty(max(1, n_tiles * k * 2), dtype=_t.int64,
-                           device=seed_key12.device)
-            total = _t.zeros(1, dtype=_t.int64, device=seed_key12.device)
+                           device=seed_b.device)
+            total = _t.zeros(1, dtype=_t.int64, device=seed_b.device)
             side = _t.cuda.Stream()
             side.wait_stream(_t.cuda.current_stream())
             with _t.cuda.stream(side):   # warmup outside capture
-                native.chain_count_into(seed_key12, seed_z, *hop_args,
+                native.chain_count_into(seed_b, seed_z, *hop_args,
                                         win, total)
             _t.cuda.current_stream().wait_stream(side)
             g = _t.cuda.CUDAGraph()
             with _t.cuda.graph(g):
-                native.chain_count_into(seed_key12, seed_z, *hop_args,
+                native.chain_count_into(seed_b, seed_z, *hop_args,
                                         win, total)
             op._chain_graph = (g, total, win)
         except Exception:

@@ -570,15 +570,38 @@ inline hipStream_t cur_stream() {
 constexpr int kMaxHops = 4;
 
 struct ChainHops {
-  const int64_t* key12[kMaxHops];  // narrowed sorted regions
+  // narrowed sorted regions, stored as the 32-bit LOW components only
+  // (the high word is the hop's constant predicate — dropping it halves
+  // the streamed bytes and doubles the LDS window capacity); ascending in
+  // UNSIGNED 32-bit order, so all searches compare as uint32
+  const int32_t* key32[kMaxHops];
   int64_t n[kMaxHops];
-  int64_t const_hi[kMaxHops];      // high 32 bits of the probe key
   int32_t src[kMaxHops];           // 0: seed B component, 1: seed Z column
   const unsigned long long* table[kMaxHops];  // optional count table (u64)
   const uint32_t* table32[kMaxHops];  // packed (val<<7)|count u32 variant
   int64_t tmask[kMaxHops];
   int k;
 };
+
+__device__ __forceinline__ int64_t lower_bound_u32(
+    const int32_t* __restrict__ a, int64_t n, uint32_t key) {
+  int64_t lo = 0, hi = n;
+  while (lo < hi) {
+    int64_t mid = (lo + hi) >> 1;
+    if (static_cast<uint32_t>(a[mid]) < key) lo = mid + 1; else hi = mid;
+  }
+  return lo;
+}
+
+__device__ __forceinline__ int64_t upper_bound_u32(
+    const int32_t* __restrict__ a, int64_t n, uint32_t key) {
+  int64_t lo = 0, hi = n;
+  while (lo < hi) {
+    int64_t mid = (lo + hi) >> 1;
+    if (static_cast<uint32_t>(a[mid]) <= key) lo = mid + 1; else hi = mid;
+  }
+  return lo;
+}
 
 // packed-u32 count table: halves the table footprint (better L2 residency
 // beside the seed stream) when every value < 2^25 and count < 2^7
@@ -664,7 +687,7 @@ at::Tensor build_count_table(at::Tensor packed_entries) {
 // per-(tile, src0-hop) windows: the seed region is subject-sorted, so the
 // B component is monotone — each 256-seed tile's src-0 hop keys fall in a
 // narrow window of that hop's region (merge-path; computed in parallel).
-__global__ void chain_tile_bounds(const int64_t* __restrict__ seed_key12,
+__global__ void chain_tile_bounds(const int32_t* __restrict__ seed_b,
                                   int64_t m, int64_t n_tiles, ChainHops hops,
                                   int64_t* __restrict__ win,  // [n_tiles][k][2]
                                   unsigned long long* __restrict__ total_zero
@@ -673,17 +696,16 @@ __global__ void chain_tile_bounds(const int64_t* __restrict__ seed_key12,
     *total_zero = 0ULL;  // saves the serve path a hipMemsetAsync enqueue
   for (int64_t t = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; t < n_tiles;
        t += (int64_t)gridDim.x * blockDim.x) {
-    int64_t first_b = seed_key12[t * kTile] & 0xFFFFFFFFLL;
-    int64_t last_b = seed_key12[min((t + 1) * kTile - 1, m - 1)] & 0xFFFFFFFFLL;
+    uint32_t first_b = static_cast<uint32_t>(seed_b[t * kTile]);
+    uint32_t last_b =
+        static_cast<uint32_t>(seed_b[min((t + 1) * kTile - 1, m - 1)]);
     for (int h = 0; h < hops.k; ++h) {
       int64_t lo = 0, hi = hops.n[h];
       if (hops.table[h] != nullptr || hops.table32[h] != nullptr) {
         lo = hi = 0;  // hashed hop: window unused
       } else if (hops.src[h] == 0) {
-        lo = lower_bound_i64(hops.key12[h], hops.n[h],
-                             (hops.const_hi[h] << 32) | first_b);
-        hi = upper_bound_i64(hops.key12[h], hops.n[h],
-                             (hops.const_hi[h] << 32) | last_b);
+        lo = lower_bound_u32(hops.key32[h], hops.n[h], first_b);
+        hi = upper_bound_u32(hops.key32[h], hops.n[h], last_b);
       }
       win[(t * hops.k + h) * 2] = lo;
       win[(t * hops.k + h) * 2 + 1] = hi;
@@ -695,31 +717,31 @@ __global__ void chain_tile_bounds(const int64_t* __restrict__ seed_key12,
 // by all 256 threads of the block — stage it through LDS once and search
 // there instead of hammering the same L1/L2 lines (guide: LDS-staged
 // probe structures).  16 KB leaves occupancy at 10 blocks/CU.
-constexpr int kChainLds = 2048;
+constexpr int kChainLds = 4096;   // int32 rows: 16 KB, 10 blocks/CU
 
-__global__ void chain_count_kernel(const int64_t* __restrict__ seed_key12,
+__global__ void chain_count_kernel(const int32_t* __restrict__ seed_b,
                                    const int32_t* __restrict__ seed_z,
                                    int64_t m, ChainHops hops,
                                    const int64_t* __restrict__ win,
                                    unsigned long long* __restrict__ total) {
-  __shared__ int64_t lds[kChainLds];
+  __shared__ int32_t lds[kChainLds];
   unsigned long long acc = 0;
   // block-uniform tile iteration (every thread of the block is in the same
   // tile) so the cooperative LDS loads can barrier safely
   for (int64_t t = blockIdx.x; t * kTile < m; t += gridDim.x) {
     int64_t i = t * kTile + threadIdx.x;
     bool active = i < m;
-    int64_t b_comp = 0, z_comp = 0;
+    uint32_t b_comp = 0, z_comp = 0;
     if (active) {
-      b_comp = seed_key12[i] & 0xFFFFFFFFLL;
-      z_comp = static_cast<int64_t>(seed_z[i]) & 0xFFFFFFFFLL;
+      b_comp = static_cast<uint32_t>(seed_b[i]);
+      z_comp = static_cast<uint32_t>(seed_z[i]);
     }
     unsigned long long prod = 1;
     // no early exit: the hop searches are independent dependent-load
     // chains — letting them all issue gives the scheduler ILP to hide
     // L2 latency
     for (int h = 0; h < hops.k; ++h) {
-      int64_t comp = hops.src[h] == 0 ? b_comp : z_comp;
+      uint32_t comp = hops.src[h] == 0 ? b_comp : z_comp;
       if (hops.table32[h] != nullptr) {
         if (!active) continue;
         // packed count-table hop: one 4-byte L2 load
@@ -758,35 +780,39 @@ __global__ void chain_count_kernel(const int64_t* __restrict__ seed_key12,
         prod *= cnt;
         continue;
       }
-      int64_t key = (hops.const_hi[h] << 32) | comp;
+      uint32_t key = comp;
       int64_t wlo = win[(t * hops.k + h) * 2];
       int64_t wspan = win[(t * hops.k + h) * 2 + 1] - wlo;
-      const int64_t* base = hops.key12[h] + wlo;
+      const int32_t* base = hops.key32[h] + wlo;
       if (hops.src[h] == 0 && wspan <= kChainLds) {
         // cooperative stage + barrier: search LDS, not HBM/L2
         for (int64_t j = threadIdx.x; j < wspan; j += blockDim.x)
           lds[j] = base[j];
         __syncthreads();
         if (active) {
-          int64_t lo = lower_bound_i64(lds, wspan, key);
+          int64_t lo = lower_bound_u32(lds, wspan, key);
           int64_t hi = lo;
-          while (hi < wspan && hi - lo < 4 && lds[hi] == key) ++hi;
-          if (hi - lo == 4 && hi < wspan && lds[hi] == key)
-            hi = lo + upper_bound_i64(lds + lo, wspan - lo, key);
+          while (hi < wspan && hi - lo < 4
+                 && static_cast<uint32_t>(lds[hi]) == key) ++hi;
+          if (hi - lo == 4 && hi < wspan
+              && static_cast<uint32_t>(lds[hi]) == key)
+            hi = lo + upper_bound_u32(lds + lo, wspan - lo, key);
           prod *= static_cast<unsigned long long>(hi - lo);
         }
         __syncthreads();  // before the next hop reuses the buffer
         continue;
       }
       if (!active) continue;
-      int64_t lo = lower_bound_i64(base, wspan, key);
+      int64_t lo = lower_bound_u32(base, wspan, key);
       // match runs are tiny (one object per subject in typical star data):
       // walk forward a few cache-hot slots instead of paying a second full
       // log2(wspan) dependent-load chain; fall back for genuine skew
       int64_t hi = lo;
-      while (hi < wspan && hi - lo < 4 && base[hi] == key) ++hi;
-      if (hi - lo == 4 && hi < wspan && base[hi] == key)
-        hi = lo + upper_bound_i64(base + lo, wspan - lo, key);
+      while (hi < wspan && hi - lo < 4
+             && static_cast<uint32_t>(base[hi]) == key) ++hi;
+      if (hi - lo == 4 && hi < wspan
+          && static_cast<uint32_t>(base[hi]) == key)
+        hi = lo + upper_bound_u32(base + lo, wspan - lo, key);
       prod *= static_cast<unsigned long long>(hi - lo);
     }
     if (active) acc += prod;
@@ -801,21 +827,21 @@ __global__ void chain_count_kernel(const int64_t* __restrict__ seed_key12,
 // Capture-friendly variant: caller owns the window + total buffers, no
 // allocation and no device->host sync inside — the launch pair can be
 // recorded into a hipGraph and replayed per query (launch-bound path).
-void chain_count_into(at::Tensor seed_key12, at::Tensor seed_z,
-                      std::vector<at::Tensor> hop_key12,
-                      std::vector<int64_t> hop_const_hi,
+void chain_count_into(at::Tensor seed_b, at::Tensor seed_z,
+                      std::vector<at::Tensor> hop_key32,
                       std::vector<int64_t> hop_src,
                       std::vector<at::Tensor> hop_table,
                       at::Tensor win, at::Tensor total) {
-  TORCH_CHECK(seed_key12.is_cuda() && seed_z.is_cuda());
+  TORCH_CHECK(seed_b.is_cuda() && seed_b.dtype() == at::kInt
+              && seed_z.is_cuda());
   TORCH_CHECK(win.dtype() == at::kLong && total.dtype() == at::kLong);
-  int64_t m = seed_key12.numel();
+  int64_t m = seed_b.numel();
   ChainHops hops{};
-  hops.k = static_cast<int>(hop_key12.size());
-  for (size_t h = 0; h < hop_key12.size(); ++h) {
-    hops.key12[h] = hop_key12[h].data_ptr<int64_t>();
-    hops.n[h] = hop_key12[h].numel();
-    hops.const_hi[h] = hop_const_hi[h];
+  hops.k = static_cast<int>(hop_key32.size());
+  for (size_t h = 0; h < hop_key32.size(); ++h) {
+    TORCH_CHECK(hop_key32[h].dtype() == at::kInt);
+    hops.key32[h] = hop_key32[h].data_ptr<int32_t>();
+    hops.n[h] = hop_key32[h].numel();
     hops.src[h] = static_cast<int32_t>(hop_src[h]);
     if (hop_table[h].numel() > 0) {
       if (hop_table[h].dtype() == at::kInt) {
@@ -835,12 +861,12 @@ void chain_count_into(at::Tensor seed_key12, at::Tensor seed_z,
   TORCH_CHECK(win.numel() >= n_tiles * hops.k * 2, "win buffer too small");
   hipLaunchKernelGGL(chain_tile_bounds, dim3(grid_for(n_tiles)),
                      dim3(kBlock), 0, stream,
-                     seed_key12.data_ptr<int64_t>(), m, n_tiles, hops,
+                     seed_b.data_ptr<int32_t>(), m, n_tiles, hops,
                      win.data_ptr<int64_t>(),
                      static_cast<unsigned long long*>(nullptr));
   HIP_OK(hipGetLastError());
   hipLaunchKernelGGL(chain_count_kernel, dim3(grid_for(m)), dim3(kBlock), 0,
-                     stream, seed_key12.data_ptr<int64_t>(),
+                     stream, seed_b.data_ptr<int32_t>(),
                      seed_z.data_ptr<int32_t>(), m, hops,
                      win.data_ptr<int64_t>(),
                      reinterpret_cast<unsigned long long*>(
@@ -856,7 +882,7 @@ void chain_count_into(at::Tensor seed_key12, at::Tensor seed_z,
 // replay floor (direct launches are ~3-4 µs each vs 10-16 µs replay,
 // MI355X guide §graph-replay-floor).
 struct ChainServe {
-  at::Tensor seed_key12, seed_z, win, total;
+  at::Tensor seed_b, seed_z, win, total;
   std::vector<at::Tensor> keep;
   ChainHops hops{};
   int64_t m = 0, n_tiles = 0;
@@ -867,23 +893,23 @@ struct ChainServe {
 };
 static std::vector<std::unique_ptr<ChainServe>> g_chain_serves;
 
-int64_t register_chain_serve(at::Tensor seed_key12, at::Tensor seed_z,
-                             std::vector<at::Tensor> hop_key12,
-                             std::vector<int64_t> hop_const_hi,
+int64_t register_chain_serve(at::Tensor seed_b, at::Tensor seed_z,
+                             std::vector<at::Tensor> hop_key32,
                              std::vector<int64_t> hop_src,
                              std::vector<at::Tensor> hop_table) {
-  TORCH_CHECK(seed_key12.is_cuda() && seed_z.is_cuda());
+  TORCH_CHECK(seed_b.is_cuda() && seed_b.dtype() == at::kInt
+              && seed_z.is_cuda());
   auto cs = std::make_unique<ChainServe>();
-  cs->seed_key12 = seed_key12;
+  cs->seed_b = seed_b;
   cs->seed_z = seed_z;
-  cs->m = seed_key12.numel();
-  cs->hops.k = static_cast<int>(hop_key12.size());
-  for (size_t h = 0; h < hop_key12.size(); ++h) {
-    cs->keep.push_back(hop_key12[h]);
+  cs->m = seed_b.numel();
+  cs->hops.k = static_cast<int>(hop_key32.size());
+  for (size_t h = 0; h < hop_key32.size(); ++h) {
+    TORCH_CHECK(hop_key32[h].dtype() == at::kInt);
+    cs->keep.push_back(hop_key32[h]);
     cs->keep.push_back(hop_table[h]);
-    cs->hops.key12[h] = hop_key12[h].data_ptr<int64_t>();
-    cs->hops.n[h] = hop_key12[h].numel();
-    cs->hops.const_hi[h] = hop_const_hi[h];
+    cs->hops.key32[h] = hop_key32[h].data_ptr<int32_t>();
+    cs->hops.n[h] = hop_key32[h].numel();
     cs->hops.src[h] = static_cast<int32_t>(hop_src[h]);
     if (hop_table[h].numel() > 0) {
       if (hop_table[h].dtype() == at::kInt) {
@@ -898,11 +924,11 @@ int64_t register_chain_serve(at::Tensor seed_key12, at::Tensor seed_z,
   }
   cs->n_tiles = (cs->m + kTile - 1) / kTile;
   cs->win = at::empty({std::max<int64_t>(1, cs->n_tiles * cs->hops.k * 2)},
-                      seed_key12.options());
+                      seed_b.options().dtype(at::kLong));
   // total[0] = accumulator, total[1] = constant 1 (the completion flag's
   // device source: same-stream copies are ordered, so once the flag
   // lands in pinned memory the count before it is valid)
-  cs->total = at::zeros({2}, seed_key12.options());
+  cs->total = at::zeros({2}, seed_b.options().dtype(at::kLong));
   cs->total[1] = 1;
   HIP_OK(hipHostMalloc(reinterpret_cast<void**>(&cs->pinned), 16));
   g_chain_serves.push_back(std::move(cs));
@@ -918,11 +944,11 @@ int64_t serve_chain_count(int64_t id) {
   *flag = 0;
   hipLaunchKernelGGL(chain_tile_bounds, dim3(grid_for(cs.n_tiles)),
                      dim3(kBlock), 0, stream,
-                     cs.seed_key12.data_ptr<int64_t>(), cs.m, cs.n_tiles,
+                     cs.seed_b.data_ptr<int32_t>(), cs.m, cs.n_tiles,
                      cs.hops, cs.win.data_ptr<int64_t>(),
                      reinterpret_cast<unsigned long long*>(total_ptr));
   hipLaunchKernelGGL(chain_count_kernel, dim3(grid_for(cs.m)), dim3(kBlock),
-                     0, stream, cs.seed_key12.data_ptr<int64_t>(),
+                     0, stream, cs.seed_b.data_ptr<int32_t>(),
                      cs.seed_z.data_ptr<int32_t>(), cs.m, cs.hops,
                      cs.win.data_ptr<int64_t>(),
                      reinterpret_cast<unsigned long long*>(total_ptr));
@@ -948,24 +974,22 @@ void release_chain_serve(int64_t id) {
     g_chain_serves[id].reset();
 }
 
-int64_t chain_count(at::Tensor seed_key12, at::Tensor seed_z,
-                    std::vector<at::Tensor> hop_key12,
-                    std::vector<int64_t> hop_const_hi,
+int64_t chain_count(at::Tensor seed_b, at::Tensor seed_z,
+                    std::vector<at::Tensor> hop_key32,
                     std::vector<int64_t> hop_src,
                     std::vector<at::Tensor> hop_table) {
-  TORCH_CHECK(seed_key12.is_cuda() && seed_z.is_cuda());
-  TORCH_CHECK(hop_key12.size() <= static_cast<size_t>(kMaxHops));
-  TORCH_CHECK(hop_key12.size() == hop_const_hi.size()
-              && hop_key12.size() == hop_src.size()
-              && hop_key12.size() == hop_table.size());
-  int64_t m = seed_key12.numel();
+  TORCH_CHECK(seed_b.is_cuda() && seed_b.dtype() == at::kInt
+              && seed_z.is_cuda());
+  TORCH_CHECK(hop_key32.size() <= static_cast<size_t>(kMaxHops));
+  TORCH_CHECK(hop_key32.size() == hop_src.size()
+              && hop_key32.size() == hop_table.size());
+  int64_t m = seed_b.numel();
   ChainHops hops{};
-  hops.k = static_cast<int>(hop_key12.size());
-  for (size_t h = 0; h < hop_key12.size(); ++h) {
-    TORCH_CHECK(hop_key12[h].is_cuda() && hop_key12[h].dtype() == at::kLong);
-    hops.key12[h] = hop_key12[h].data_ptr<int64_t>();
-    hops.n[h] = hop_key12[h].numel();
-    hops.const_hi[h] = hop_const_hi[h];
+  hops.k = static_cast<int>(hop_key32.size());
+  for (size_t h = 0; h < hop_key32.size(); ++h) {
+    TORCH_CHECK(hop_key32[h].is_cuda() && hop_key32[h].dtype() == at::kInt);
+    hops.key32[h] = hop_key32[h].data_ptr<int32_t>();
+    hops.n[h] = hop_key32[h].numel();
     hops.src[h] = static_cast<int32_t>(hop_src[h]);
     if (hop_table[h].numel() > 0) {
       TORCH_CHECK(hop_table[h].is_cuda()
@@ -981,19 +1005,20 @@ int64_t chain_count(at::Tensor seed_key12, at::Tensor seed_z,
       hops.tmask[h] = hop_table[h].numel() - 1;
     }
   }
-  auto total = at::zeros({1}, seed_key12.options());
+  auto total = at::zeros({1}, seed_b.options().dtype(at::kLong));
   if (m > 0) {
     auto stream = cur_stream();
     int64_t n_tiles = (m + kTile - 1) / kTile;
-    auto win = at::empty({n_tiles * hops.k * 2}, seed_key12.options());
+    auto win = at::empty({n_tiles * hops.k * 2},
+                         seed_b.options().dtype(at::kLong));
     hipLaunchKernelGGL(chain_tile_bounds, dim3(grid_for(n_tiles)),
                        dim3(kBlock), 0, stream,
-                       seed_key12.data_ptr<int64_t>(), m, n_tiles, hops,
+                       seed_b.data_ptr<int32_t>(), m, n_tiles, hops,
                        win.data_ptr<int64_t>(),
                        static_cast<unsigned long long*>(nullptr));
     HIP_OK(hipGetLastError());
     hipLaunchKernelGGL(chain_count_kernel, dim3(grid_for(m)), dim3(kBlock), 0,
-                       stream, seed_key12.data_ptr<int64_t>(),
+                       stream, seed_b.data_ptr<int32_t>(),
                        seed_z.data_ptr<int32_t>(), m, hops,
                        win.data_ptr<int64_t>(),
                        reinterpret_cast<unsigned long long*>(

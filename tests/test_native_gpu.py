@@ -267,19 +267,25 @@ def test_chain_count_table_hop_matches_search():
         return int((hi - lo).sum())
 
     dev = "cuda:0"
-    sk, sz, rg = seed_key12.to(dev), seed_z.to(dev), region.to(dev)
+    sb = (seed_key12 & 0xFFFFFFFF).to(torch.int32).to(dev)
+    sz, rg = seed_z.to(dev), region.to(dev)
+    rg32 = (rg & 0xFFFFFFFF).to(torch.int32)
     vals, counts = torch.unique_consecutive(rg & 0xFFFFFFFF,
                                             return_counts=True)
     table = _native.build_count_table((vals << 32) | counts)
+    table32 = _native.build_count_table32(
+        ((vals << 7) | counts).to(torch.int32))
     assert table.numel() >= 2 * vals.numel()
     for src in (0, 1):
         want = oracle(src)
         got_search = _native.chain_count(
-            sk, sz, [rg], [pid], [src], [torch.empty(0, dtype=torch.int64,
-                                                     device=dev)])
-        got_table = _native.chain_count(sk, sz, [rg], [pid], [src], [table])
+            sb, sz, [rg32], [src], [torch.empty(0, dtype=torch.int64,
+                                                device=dev)])
+        got_table = _native.chain_count(sb, sz, [rg32], [src], [table])
+        got_t32 = _native.chain_count(sb, sz, [rg32], [src], [table32])
         assert got_search == want
         assert got_table == want
+        assert got_t32 == want
 
 
 @requires_gpu
