@@ -502,10 +502,13 @@ class ExecutionEngine:
     _HOP_TABLE_MAX_ROWS = 1_000_000
 
     def _chain_hop_table(self, native, region, n_seeds):
+        import os
         import torch as _t
         n = region.numel()
+        cap = int(os.environ.get("KOLIBRIE_HOP_TABLE_MAX",
+                                 self._HOP_TABLE_MAX_ROWS))
         if (native is None or not region.is_cuda or n == 0
-                or n > self._HOP_TABLE_MAX_ROWS or n * 4 > n_seeds):
+                or n > cap or n * 4 > n_seeds):
             return _t.empty(0, dtype=_t.int64, device=region.device)
         vals, counts = _t.unique_consecutive(region & 0xFFFFFFFF,
                                              return_counts=True)

@@ -581,6 +581,10 @@ struct ChainHops {
   const uint32_t* table32[kMaxHops];  // packed (val<<7)|count u32 variant
   int64_t tmask[kMaxHops];
   int k;
+  // stage a window through LDS only when it exceeds this row count:
+  // small windows (few cache lines, shared by the whole block) search
+  // L1/L2 directly without the staging barrier
+  int lds_min = 0;
 };
 
 __device__ __forceinline__ int64_t lower_bound_u32(
@@ -784,7 +788,8 @@ __global__ void chain_count_kernel(const int32_t* __restrict__ seed_b,
       int64_t wlo = win[(t * hops.k + h) * 2];
       int64_t wspan = win[(t * hops.k + h) * 2 + 1] - wlo;
       const int32_t* base = hops.key32[h] + wlo;
-      if (hops.src[h] == 0 && wspan <= kChainLds) {
+      if (hops.src[h] == 0 && wspan <= kChainLds
+          && wspan > hops.lds_min) {
         // cooperative stage + barrier: search LDS, not HBM/L2
         for (int64_t j = threadIdx.x; j < wspan; j += blockDim.x)
           lds[j] = base[j];
@@ -837,6 +842,8 @@ void chain_count_into(at::Tensor seed_b, at::Tensor seed_z,
   TORCH_CHECK(win.dtype() == at::kLong && total.dtype() == at::kLong);
   int64_t m = seed_b.numel();
   ChainHops hops{};
+  if (const char* e = getenv("KOLIBRIE_CHAIN_LDS_MIN"))
+    hops.lds_min = atoi(e);
   hops.k = static_cast<int>(hop_key32.size());
   for (size_t h = 0; h < hop_key32.size(); ++h) {
     TORCH_CHECK(hop_key32[h].dtype() == at::kInt);
@@ -903,6 +910,8 @@ int64_t register_chain_serve(at::Tensor seed_b, at::Tensor seed_z,
   cs->seed_b = seed_b;
   cs->seed_z = seed_z;
   cs->m = seed_b.numel();
+  if (const char* e = getenv("KOLIBRIE_CHAIN_LDS_MIN"))
+    cs->hops.lds_min = atoi(e);
   cs->hops.k = static_cast<int>(hop_key32.size());
   for (size_t h = 0; h < hop_key32.size(); ++h) {
     TORCH_CHECK(hop_key32[h].dtype() == at::kInt);
@@ -985,6 +994,8 @@ int64_t chain_count(at::Tensor seed_b, at::Tensor seed_z,
               && hop_key32.size() == hop_table.size());
   int64_t m = seed_b.numel();
   ChainHops hops{};
+  if (const char* e = getenv("KOLIBRIE_CHAIN_LDS_MIN"))
+    hops.lds_min = atoi(e);
   hops.k = static_cast<int>(hop_key32.size());
   for (size_t h = 0; h < hop_key32.size(); ++h) {
     TORCH_CHECK(hop_key32[h].is_cuda() && hop_key32[h].dtype() == at::kInt);
