@@ -332,16 +332,11 @@ def parse_rdf_xml_into(db, xml_text: str):
     for el in nodes:
         walk_node(el)
 
-    # bulk encode -> columnar insert (GPU-side dedup/sort on commit)
-    import numpy as np
+    # bulk encode -> columnar insert (GPU-side dedup/sort on commit);
+    # encode_many's tuned intern loop replaces per-term encode() calls
     n = len(triples)
-    s_ids = np.empty(n, dtype=np.uint32)
-    p_ids = np.empty(n, dtype=np.uint32)
-    o_ids = np.empty(n, dtype=np.uint32)
-    enc = db.dictionary.encode
-    for i, (s, p, o) in enumerate(triples):
-        s_ids[i] = enc(s)
-        p_ids[i] = enc(p)
-        o_ids[i] = enc(o)
     if n:
-        db.store.insert_bulk(0, s_ids, p_ids, o_ids)
+        flat = [t for tr in triples for t in tr]
+        ids = db.dictionary.encode_many(flat).reshape(-1, 3)
+        db.store.insert_bulk(0, ids[:, 0].copy(), ids[:, 1].copy(),
+                             ids[:, 2].copy())
