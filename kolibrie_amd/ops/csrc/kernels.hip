@@ -1576,9 +1576,15 @@ py::tuple parse_nlines_host(const std::string& text, bool quads) {
 // main thread merges the (small) unique-string tables and the chunk
 // threads remap their id columns in parallel.  The GIL is released for
 // the whole native phase.
+#include <deque>
+
 struct ParseChunkOut {
-  std::unordered_map<std::string, int64_t> interned;
-  std::vector<std::string> strings;
+  // term views point into the shared input text (zero-copy for the
+  // overwhelmingly common escape-free terms) or into `owned` (deque:
+  // stable addresses) for unescaped materializations
+  std::unordered_map<std::string_view, int64_t> interned;
+  std::vector<std::string_view> views;
+  std::deque<std::string> owned;
   std::vector<int64_t> ids;
   std::vector<int64_t> fallback;   // chunk-local line numbers
   int64_t line_count = 0;
@@ -1586,12 +1592,25 @@ struct ParseChunkOut {
 
 static void parse_chunk_nt(const char* data, size_t begin, size_t end,
                            bool quads, ParseChunkOut& out) {
-  auto intern = [&](std::string&& s) -> int64_t {
-    auto it = out.interned.find(s);
+  auto intern_sv = [&](std::string_view sv) -> int64_t {
+    auto it = out.interned.find(sv);
     if (it != out.interned.end()) return it->second;
-    int64_t id = static_cast<int64_t>(out.strings.size());
-    out.interned.emplace(s, id);
-    out.strings.push_back(std::move(s));
+    int64_t id = static_cast<int64_t>(out.views.size());
+    out.interned.emplace(sv, id);
+    out.views.push_back(sv);
+    return id;
+  };
+  auto intern = [&](const char* ptr, size_t len) -> int64_t {
+    if (memchr(ptr, '\\', len) == nullptr)
+      return intern_sv(std::string_view(ptr, len));
+    std::string u = unescape_nt(ptr, len);
+    auto it = out.interned.find(std::string_view(u));
+    if (it != out.interned.end()) return it->second;
+    out.owned.push_back(std::move(u));
+    std::string_view sv(out.owned.back());
+    int64_t id = static_cast<int64_t>(out.views.size());
+    out.interned.emplace(sv, id);
+    out.views.push_back(sv);
     return id;
   };
   size_t pos = begin;
@@ -1625,13 +1644,13 @@ static void parse_chunk_nt(const char* data, size_t begin, size_t end,
         size_t j = i + 1;
         while (j < e && line[j] != '>') ++j;
         if (j >= e) { ok = false; break; }
-        term_ids[nt++] = intern(unescape_nt(line + i + 1, j - i - 1));
+        term_ids[nt++] = intern(line + i + 1, j - i - 1);
         i = j + 1;
       } else if (c == '"') {
         size_t j = i + 1;
         while (j < e && !(line[j] == '"' && line[j - 1] != '\\')) ++j;
         if (j >= e) { ok = false; break; }
-        term_ids[nt++] = intern(unescape_nt(line + i + 1, j - i - 1));
+        term_ids[nt++] = intern(line + i + 1, j - i - 1);
         i = j + 1;
         while (i < e && !isspace(static_cast<unsigned char>(line[i]))
                && line[i] != '.') {
@@ -1641,7 +1660,7 @@ static void parse_chunk_nt(const char* data, size_t begin, size_t end,
       } else if (c == '_') {
         size_t j = i;
         while (j < e && !isspace(static_cast<unsigned char>(line[j]))) ++j;
-        term_ids[nt++] = intern(std::string(line + i, j - i));
+        term_ids[nt++] = intern_sv(std::string_view(line + i, j - i));
         i = j;
       } else if (c == '.') {
         break;
@@ -1694,15 +1713,19 @@ py::tuple parse_nlines_host_mt(const std::string& text, bool quads,
                            starts[i + 1], quads, std::ref(outs[i]));
     for (auto& t : threads) t.join();
     // merge unique-string tables (small relative to the line count)
-    std::unordered_map<std::string, int64_t> global;
+    size_t uniq_upper0 = 0;
+    for (int i = 0; i < nt; ++i) uniq_upper0 += outs[i].views.size();
+    g_strings.reserve(uniq_upper0);
+    std::unordered_map<std::string_view, int64_t> global;
+    global.reserve(uniq_upper0 * 2);
     for (int i = 0; i < nt; ++i) {
-      remaps[i].resize(outs[i].strings.size());
-      for (size_t k = 0; k < outs[i].strings.size(); ++k) {
-        auto it = global.find(outs[i].strings[k]);
+      remaps[i].resize(outs[i].views.size());
+      for (size_t k = 0; k < outs[i].views.size(); ++k) {
+        auto it = global.find(outs[i].views[k]);
         if (it == global.end()) {
           int64_t id = static_cast<int64_t>(g_strings.size());
-          global.emplace(outs[i].strings[k], id);
-          g_strings.push_back(std::move(outs[i].strings[k]));
+          g_strings.emplace_back(outs[i].views[k]);
+          global.emplace(std::string_view(g_strings.back()), id);
           remaps[i][k] = id;
         } else {
           remaps[i][k] = it->second;
@@ -1815,17 +1838,17 @@ py::tuple parse_ntriples_file_encode(const std::string& path,
     // reserving capacity keeps SSO string objects (and thus the views)
     // stable under growth
     size_t uniq_upper = 0;
-    for (int i = 0; i < nt; ++i) uniq_upper += outs[i].strings.size();
+    for (int i = 0; i < nt; ++i) uniq_upper += outs[i].views.size();
     g_strings.reserve(uniq_upper);
     std::unordered_map<std::string_view, int64_t> global;
     global.reserve(uniq_upper * 2);
     for (int i = 0; i < nt; ++i) {
-      remaps[i].resize(outs[i].strings.size());
-      for (size_t k = 0; k < outs[i].strings.size(); ++k) {
-        auto it = global.find(std::string_view(outs[i].strings[k]));
+      remaps[i].resize(outs[i].views.size());
+      for (size_t k = 0; k < outs[i].views.size(); ++k) {
+        auto it = global.find(outs[i].views[k]);
         if (it == global.end()) {
           int64_t id = static_cast<int64_t>(g_strings.size());
-          g_strings.push_back(std::move(outs[i].strings[k]));
+          g_strings.emplace_back(outs[i].views[k]);
           global.emplace(std::string_view(g_strings.back()), id);
           remaps[i][k] = id;
         } else {
@@ -1838,11 +1861,14 @@ py::tuple parse_ntriples_file_encode(const std::string& path,
     if (dbg) { py::gil_scoped_acquire a; lap("merge"); }
   }
   // GIL pass: intern the unique strings into the Python dictionary
+  // (one SetDefault per string; a shared 0.0 float object for the
+  // non-numeric majority — the value column stores 0.0 for those)
   std::vector<int64_t> g_remap(g_strings.size());
   {
     PyObject* d = str_to_id.ptr();
     PyObject* lst = id_to_str.ptr();
     PyObject* vals = values.ptr();
+    PyObject* zero = PyFloat_FromDouble(0.0);
     int64_t next_id = static_cast<int64_t>(PyList_GET_SIZE(lst));
     for (size_t k = 0; k < g_strings.size(); ++k) {
       const std::string& s = g_strings[k];
@@ -1850,15 +1876,15 @@ py::tuple parse_ntriples_file_encode(const std::string& path,
                                            static_cast<Py_ssize_t>(s.size()),
                                            "replace");
       TORCH_CHECK(key != nullptr, "utf-8 decode failed");
-      PyObject* hit = PyDict_GetItem(d, key);  // borrowed
-      if (hit != nullptr) {
-        g_remap[k] = PyLong_AsLongLong(hit);
+      TORCH_CHECK(next_id < max_id, "dictionary ID space exhausted");
+      PyObject* idobj = PyLong_FromLongLong(next_id);
+      PyObject* prev = PyDict_SetDefault(d, key, idobj);  // borrowed
+      if (prev != idobj) {
+        g_remap[k] = PyLong_AsLongLong(prev);
+        Py_DECREF(idobj);
         Py_DECREF(key);
         continue;
       }
-      TORCH_CHECK(next_id < max_id, "dictionary ID space exhausted");
-      PyObject* idobj = PyLong_FromLongLong(next_id);
-      PyDict_SetItem(d, key, idobj);
       PyList_Append(lst, key);
       // numeric value column: full-consume strtod, non-finite -> 0.0
       // (fast-skip: strings not starting numeric cannot parse)
@@ -1875,13 +1901,18 @@ py::tuple parse_ntriples_file_encode(const std::string& path,
           if (*endp == '\0' && std::isfinite(parsed)) v = parsed;
         }
       }
-      PyObject* vobj = PyFloat_FromDouble(v);
-      PyList_Append(vals, vobj);
-      Py_DECREF(vobj);
+      if (v == 0.0) {
+        PyList_Append(vals, zero);
+      } else {
+        PyObject* vobj = PyFloat_FromDouble(v);
+        PyList_Append(vals, vobj);
+        Py_DECREF(vobj);
+      }
       Py_DECREF(idobj);
       Py_DECREF(key);
       g_remap[k] = next_id++;
     }
+    Py_DECREF(zero);
   }
   lap("dict intern");
   auto ids_t = at::empty({total_rows / 3, 3}, at::kInt);
