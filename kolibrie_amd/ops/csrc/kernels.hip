@@ -1822,7 +1822,7 @@ py::tuple parse_ntriples_file_encode(const std::string& path,
   }
   starts.push_back(n);
   std::vector<ParseChunkOut> outs(nt);
-  std::vector<std::string> g_strings;
+  std::vector<std::vector<std::string_view>> shard_views;
   std::vector<std::vector<int64_t>> remaps(nt);
   std::vector<int64_t> line_off(nt + 1, 0);
   int64_t total_rows = 0;
@@ -1834,83 +1834,105 @@ py::tuple parse_ntriples_file_encode(const std::string& path,
                            starts[i + 1], false, std::ref(outs[i]));
     for (auto& t : threads) t.join();
     if (dbg) { py::gil_scoped_acquire a; lap("parallel parse"); }
-    // string_view keys into g_strings avoid re-hash-copying every string;
-    // reserving capacity keeps SSO string objects (and thus the views)
-    // stable under growth
+    // SHARDED parallel merge: shard strings by hash; each thread owns one
+    // shard, scans every chunk's view table and dedups only its shard
+    // (remap entries encode (shard, local id); resolved after the GIL
+    // interning pass assigns Python ids per shard)
     size_t uniq_upper = 0;
-    for (int i = 0; i < nt; ++i) uniq_upper += outs[i].views.size();
-    g_strings.reserve(uniq_upper);
-    std::unordered_map<std::string_view, int64_t> global;
-    global.reserve(uniq_upper * 2);
     for (int i = 0; i < nt; ++i) {
       remaps[i].resize(outs[i].views.size());
-      for (size_t k = 0; k < outs[i].views.size(); ++k) {
-        auto it = global.find(outs[i].views[k]);
-        if (it == global.end()) {
-          int64_t id = static_cast<int64_t>(g_strings.size());
-          g_strings.emplace_back(outs[i].views[k]);
-          global.emplace(std::string_view(g_strings.back()), id);
-          remaps[i][k] = id;
-        } else {
-          remaps[i][k] = it->second;
-        }
-      }
+      uniq_upper += outs[i].views.size();
       line_off[i + 1] = line_off[i] + outs[i].line_count;
       total_rows += static_cast<int64_t>(outs[i].ids.size());
     }
+    shard_views.resize(nt);
+    {
+      std::vector<std::thread> mthreads;
+      for (int t = 0; t < nt; ++t)
+        mthreads.emplace_back([&, t]() {
+          std::hash<std::string_view> hasher;
+          std::unordered_map<std::string_view, int64_t> local;
+          local.reserve(uniq_upper / std::max(1, nt) * 2 + 64);
+          auto& sv = shard_views[t];
+          for (int i = 0; i < nt; ++i) {
+            auto& views = outs[i].views;
+            auto& rm = remaps[i];
+            for (size_t k = 0; k < views.size(); ++k) {
+              if (static_cast<int>(hasher(views[k]) % nt) != t) continue;
+              auto it = local.find(views[k]);
+              int64_t L;
+              if (it == local.end()) {
+                L = static_cast<int64_t>(sv.size());
+                local.emplace(views[k], L);
+                sv.push_back(views[k]);
+              } else {
+                L = it->second;
+              }
+              rm[k] = (static_cast<int64_t>(t) << 40) | L;
+            }
+          }
+        });
+      for (auto& t : mthreads) t.join();
+    }
     if (dbg) { py::gil_scoped_acquire a; lap("merge"); }
   }
-  // GIL pass: intern the unique strings into the Python dictionary
-  // (one SetDefault per string; a shared 0.0 float object for the
-  // non-numeric majority — the value column stores 0.0 for those)
-  std::vector<int64_t> g_remap(g_strings.size());
+  // GIL pass: intern each shard's unique strings into the Python
+  // dictionary (one SetDefault per string; a shared 0.0 float object for
+  // the non-numeric majority)
+  std::vector<std::vector<int64_t>> py_ids(shard_views.size());
   {
     PyObject* d = str_to_id.ptr();
     PyObject* lst = id_to_str.ptr();
     PyObject* vals = values.ptr();
     PyObject* zero = PyFloat_FromDouble(0.0);
     int64_t next_id = static_cast<int64_t>(PyList_GET_SIZE(lst));
-    for (size_t k = 0; k < g_strings.size(); ++k) {
-      const std::string& s = g_strings[k];
-      PyObject* key = PyUnicode_DecodeUTF8(s.data(),
-                                           static_cast<Py_ssize_t>(s.size()),
-                                           "replace");
-      TORCH_CHECK(key != nullptr, "utf-8 decode failed");
-      TORCH_CHECK(next_id < max_id, "dictionary ID space exhausted");
-      PyObject* idobj = PyLong_FromLongLong(next_id);
-      PyObject* prev = PyDict_SetDefault(d, key, idobj);  // borrowed
-      if (prev != idobj) {
-        g_remap[k] = PyLong_AsLongLong(prev);
+    char numbuf[64];
+    for (size_t t = 0; t < shard_views.size(); ++t) {
+      auto& sv = shard_views[t];
+      auto& pid = py_ids[t];
+      pid.resize(sv.size());
+      for (size_t k = 0; k < sv.size(); ++k) {
+        std::string_view s = sv[k];
+        PyObject* key = PyUnicode_DecodeUTF8(
+            s.data(), static_cast<Py_ssize_t>(s.size()), "replace");
+        TORCH_CHECK(key != nullptr, "utf-8 decode failed");
+        TORCH_CHECK(next_id < max_id, "dictionary ID space exhausted");
+        PyObject* idobj = PyLong_FromLongLong(next_id);
+        PyObject* prev = PyDict_SetDefault(d, key, idobj);  // borrowed
+        if (prev != idobj) {
+          pid[k] = PyLong_AsLongLong(prev);
+          Py_DECREF(idobj);
+          Py_DECREF(key);
+          continue;
+        }
+        PyList_Append(lst, key);
+        // numeric value column: full-consume strtod, non-finite -> 0.0
+        double v = 0.0;
+        char c0 = s.empty() ? 0 : s[0];
+        if (((c0 >= '0' && c0 <= '9') || c0 == '-' || c0 == '+'
+             || c0 == '.' || c0 == ' ' || c0 == '\t'
+             || c0 == 'i' || c0 == 'I' || c0 == 'n' || c0 == 'N')
+            && s.size() < sizeof(numbuf)) {
+          memcpy(numbuf, s.data(), s.size());
+          numbuf[s.size()] = '\0';
+          char* endp = nullptr;
+          double parsed = strtod(numbuf, &endp);
+          if (endp != numbuf && endp != nullptr) {
+            while (*endp == ' ' || *endp == '\t') ++endp;
+            if (*endp == '\0' && std::isfinite(parsed)) v = parsed;
+          }
+        }
+        if (v == 0.0) {
+          PyList_Append(vals, zero);
+        } else {
+          PyObject* vobj = PyFloat_FromDouble(v);
+          PyList_Append(vals, vobj);
+          Py_DECREF(vobj);
+        }
         Py_DECREF(idobj);
         Py_DECREF(key);
-        continue;
+        pid[k] = next_id++;
       }
-      PyList_Append(lst, key);
-      // numeric value column: full-consume strtod, non-finite -> 0.0
-      // (fast-skip: strings not starting numeric cannot parse)
-      double v = 0.0;
-      char c0 = s.empty() ? 0 : s[0];
-      if ((c0 >= '0' && c0 <= '9') || c0 == '-' || c0 == '+'
-          || c0 == '.' || c0 == ' ' || c0 == '\t'
-          || c0 == 'i' || c0 == 'I' || c0 == 'n' || c0 == 'N') {
-        const char* c = s.c_str();
-        char* endp = nullptr;
-        double parsed = strtod(c, &endp);
-        if (endp != c && endp != nullptr) {
-          while (*endp == ' ' || *endp == '\t') ++endp;
-          if (*endp == '\0' && std::isfinite(parsed)) v = parsed;
-        }
-      }
-      if (v == 0.0) {
-        PyList_Append(vals, zero);
-      } else {
-        PyObject* vobj = PyFloat_FromDouble(v);
-        PyList_Append(vals, vobj);
-        Py_DECREF(vobj);
-      }
-      Py_DECREF(idobj);
-      Py_DECREF(key);
-      g_remap[k] = next_id++;
     }
     Py_DECREF(zero);
   }
@@ -1927,9 +1949,11 @@ py::tuple parse_ntriples_file_encode(const std::string& path,
       threads.emplace_back([&, i]() {
         auto& rm = remaps[i];
         int32_t* d = dst + offs[i];
-        for (size_t k = 0; k < outs[i].ids.size(); ++k)
-          d[k] = static_cast<int32_t>(
-              static_cast<uint32_t>(g_remap[rm[outs[i].ids[k]]]));
+        for (size_t k = 0; k < outs[i].ids.size(); ++k) {
+          int64_t e = rm[outs[i].ids[k]];
+          d[k] = static_cast<int32_t>(static_cast<uint32_t>(
+              py_ids[e >> 40][e & ((1LL << 40) - 1)]));
+        }
       });
     for (auto& t : threads) t.join();
   }
