@@ -13,6 +13,8 @@ EXAMPLES = [
     "examples/streaming_rsp.py",
     "examples/hybrid_probability.py",
     "examples/mqtt_ingestion.py",
+    "examples/citybench_traffic.py",
+    "examples/distributed_query.py",
 ]
 
 
