@@ -910,7 +910,7 @@ def join_bindings(left: Bindings, right: Bindings, needed=None) -> Bindings:
         ri = torch.arange(right.n, dtype=torch.long, device=dev).repeat(left.n)
         return _merge_pairs(left, right, li, ri, shared, needed)
     if (not left.maybe_unbound and not right.maybe_unbound
-            and left.n * right.n <= 4096):
+            and left.n * right.n <= 262_144):
         # tiny-tables fast path: selective point queries produce 1-row
         # intermediates where the merge/hash machinery is pure per-op
         # overhead — one broadcast compare replaces ~20 tensor ops
@@ -1125,9 +1125,11 @@ def anti_join(left: Bindings, right: Bindings) -> Bindings:
     if not shared or right.is_empty():
         return left
     if (not left.maybe_unbound and not right.maybe_unbound
-            and left.n * right.n <= 4096):
+            and left.n * right.n <= 262_144):
         # tiny-tables fast path (same shape as join_bindings'): one
         # broadcast compare instead of unique+membership machinery
+        # (262k boolean pairs is noise on-device; the former 4096 cap
+        # pushed 100x100 MINUS shapes onto the multi-op membership path)
         hit = left.col(shared[0]).unsqueeze(1) == \
             right.col(shared[0]).unsqueeze(0)
         for v in shared[1:]:
