@@ -155,6 +155,21 @@ def main():
         _ms, cold_lat, _c = _time_steps(run_cold, 3, device, use_cuda)
         cold_p50 = statistics.median(cold_lat)
 
+    # row-returning flagship variant: same 3-way join
+    # shape but materializing ~n_dept/1000 x 100 result rows as strings
+    rows_ms = rows_n = None
+    if world == 1:
+        city = ds.city_base + 3
+        ddb.db.dictionary.str_to_id[f"http://synthetic/c{city}"] = city
+        q_rows = (f"PREFIX ds: <{DS}> SELECT ?e ?sal WHERE {{ "
+                  f"?e ds:worksFor ?d . ?e ds:annual_salary ?sal . "
+                  f"?d ds:locatedIn <http://synthetic/c{city}> }}")
+        out = ddb.db.query(q_rows)
+        k = max(2, min(10, args.steps))
+        rows_ms, _l, out = _time_steps(lambda: ddb.db.query(q_rows), k,
+                                       device, use_cuda)
+        rows_n = len(out)
+
     # secondary measurement: the REAL shuffle path on a fully-partitioned
     # copy (no replicated relations) — BASELINE config 3's all-to-all join
     shuffle_ms = bcast_ms = None
@@ -208,6 +223,8 @@ def main():
                                       "+ one C++ serve call: direct kernel "
                                       "launches + pinned 8-byte readback)"),
                 "cold_ms_p50": cold_p50,
+                "rows_variant_ms_per_step": rows_ms,
+                "rows_variant_result_rows": rows_n,
                 "shuffle_ms_per_step": shuffle_ms,
                 "bcast_ms_per_step": bcast_ms,
                 "parallelism": (f"subject-hash-partition dp{world}, "

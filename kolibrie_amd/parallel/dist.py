@@ -44,10 +44,18 @@ def barrier():
         dist.barrier()
 
 
+_AR_BUF = {}
+
+
 def allreduce_sum_scalar(x: int, device) -> int:
     if not is_dist():
         return x
-    t = torch.tensor([x], dtype=torch.int64, device=device)
+    # cached per-device buffer: the serving hot loop calls this per step
+    t = _AR_BUF.get(device)
+    if t is None:
+        t = _AR_BUF[device] = torch.zeros(1, dtype=torch.int64,
+                                          device=device)
+    t.fill_(x)
     dist.all_reduce(t, op=dist.ReduceOp.SUM)
     return int(t.item())
 

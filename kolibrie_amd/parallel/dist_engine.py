@@ -166,10 +166,12 @@ class DistributedDatabase:
         return self.execute_prepared(sel, physical, part)
 
     def execute_prepared(self, sel, physical, part) -> List[List[str]]:
-        ctx = ExecutionContext(self.db, DatasetView(),
-                               world=self.world, rank=self.rank)
-        rows = ExecutionEngine(ctx).execute(
-            physical, Bindings.unit(self.device))
+        eng = getattr(self, "_engine_cache", None)
+        if eng is None:
+            ctx = ExecutionContext(self.db, DatasetView(),
+                                   world=self.world, rank=self.rank)
+            eng = self._engine_cache = ExecutionEngine(ctx)
+        rows = eng.execute(physical, Bindings.unit(self.device))
         return self._finalize(sel, rows, part)
 
     # ----------------------------------------------------------- finalize --

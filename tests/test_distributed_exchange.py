@@ -76,3 +76,9 @@ def test_battery_is_nontrivial(single_rank):
     assert int(single_rank["obj_obj_count"][0][0]) > 0
     assert len(single_rank["select_rows"]) > 10
     assert len(single_rank["group_by"]) == 3
+
+
+def test_world8_forced_shuffle_matches_single_rank(single_rank, tmp_path):
+    """VERDICT r1 item 1 asked for world 2/4/8 equality."""
+    r8 = _run_worker(8, str(tmp_path / "w8.json"))
+    assert r8 == single_rank
