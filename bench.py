@@ -80,6 +80,8 @@ def _time_steps(fn, steps, device, use_cuda):
 
 
 def main():
+    import gc
+    gc.freeze()   # exclude torch's import-time object graph from gen scans
     ap = argparse.ArgumentParser()
     ap.add_argument("--gpus", type=int, default=1)
     ap.add_argument("--steps", type=int, default=20)
