@@ -3,10 +3,10 @@
 hash-join COUNT on 100M synthetic employee triples (BASELINE.json metric).
 
 What the timed region measures (truth-in-labeling):
-  - default (warm): prepared-statement serving — the plan cache is hit and
-    the fused chain-count kernel replays its captured hipGraph per step;
-    parse + Volcano planning ran once at warmup.  The kernel work (probe +
-    count over every seed row) is re-executed every step.
+  - default (warm): prepared-statement serving — the plan cache is hit
+    and ONE C++ serve call issues the fused chain-count kernels directly
+    (pinned readback); parse + Volcano planning ran once at warmup.  The
+    kernel work (probe + count over every seed row) re-executes per step.
   - --cold: the plan cache is cleared before every step, so each step pays
     parse -> Volcano plan -> distribute -> execute.
   A 3-step cold measurement is always reported in config.cold_ms_p50.
