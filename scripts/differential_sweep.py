@@ -64,8 +64,8 @@ def gen_random_bgp(rng: random.Random) -> str:
 def gen_query(rng: random.Random) -> str:
     """One random query over the employee schema."""
     kind = rng.choice(["star", "chain", "filter", "optional", "union",
-                      "agg", "distinct", "values", "bind", "subquery",
-                      "minus", "ask", "path", "rand", "rand"])
+                      "agg", "agg2", "rows", "distinct", "values", "bind",
+                      "subquery", "minus", "ask", "path", "rand", "rand"])
     if kind == "rand":
         return gen_random_bgp(rng)
     if kind == "path":
@@ -108,6 +108,20 @@ def gen_query(rng: random.Random) -> str:
         return (f"SELECT ?pos (COUNT(*) AS ?c) WHERE {{ "
                 f"?e {P['position']} ?pos }} GROUP BY ?pos{having} "
                 f"ORDER BY ?pos")
+    if kind == "agg2":
+        # multi-aggregate GROUP BY: exercises the native K4 hash-aggregate
+        # kernel on GPU vs the torch composite on CPU
+        aggs = rng.sample(["(SUM(?s) AS ?t)", "(AVG(?s) AS ?m)",
+                           "(MIN(?s) AS ?lo)", "(MAX(?s) AS ?hi)",
+                           "(COUNT(*) AS ?n)"], rng.randint(2, 4))
+        return (f"SELECT ?pos {' '.join(aggs)} WHERE {{ "
+                f"?e {P['position']} ?pos . ?e {P['salary']} ?s }} "
+                f"GROUP BY ?pos ORDER BY ?pos")
+    if kind == "rows":
+        lim = rng.randint(5, 400)
+        return (f"SELECT ?e ?city WHERE {{ ?e {P['worksFor']} ?d . "
+                f"?d {P['locatedIn']} ?city }} ORDER BY ?e ?city "
+                f"LIMIT {lim}")
     if kind == "distinct":
         return (f"SELECT DISTINCT ?city WHERE {{ ?d {P['locatedIn']} ?city }} "
                 f"ORDER BY ?city LIMIT 50")
