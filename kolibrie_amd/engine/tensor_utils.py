@@ -101,15 +101,6 @@ def rows_diff(a_cols: Rows, b_cols: Rows) -> List[torch.Tensor]:
     return [c[mask] for c in a_cols]
 
 
-def rows_intersect(a_cols: Rows, b_cols: Rows) -> List[torch.Tensor]:
-    a_cols = unique_rows(a_cols)
-    if a_cols[0].numel() == 0:
-        return list(a_cols)
-    b_cols = unique_rows(b_cols)
-    mask = membership_mask(a_cols, b_cols)
-    return [c[mask] for c in a_cols]
-
-
 def membership_mask(a_cols: Rows, b_sorted_unique: Rows) -> torch.Tensor:
     """For each row of A, True iff it occurs in B.
 
