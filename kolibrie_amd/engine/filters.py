@@ -64,7 +64,7 @@ class CompiledExpr:
             probe_col = next(iter(b.cols.values())) if b.cols else None
             native = native_for(probe_col) if probe_col is not None else None
             if native is not None:
-                prog = compile_filter(self.ast, b)
+                prog = compile_filter(self.ast, b, cache_holder=self)
                 if prog is not None:
                     ops_t, args_t, consts_t, col_list = prog
                     return native.filter_bytecode(

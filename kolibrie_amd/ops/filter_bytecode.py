@@ -159,29 +159,51 @@ def _vars_of(e: Expr) -> List[str]:
     return out
 
 
-def compile_filter(ast: Expr, bindings) -> Optional[Tuple]:
+def compile_filter(ast: Expr, bindings, cache_holder=None) -> Optional[Tuple]:
     """Compile to (ops, args, consts, col_list) device tensors.
 
     Returns None when the expression can't be expressed in bytecode (caller
-    falls back to the vectorized torch path)."""
+    falls back to the vectorized torch path).
+
+    The program depends only on the AST and WHICH vars are bound, so when
+    `cache_holder` is given (the CompiledExpr) the three device tensors are
+    cached on it — recompiling per evaluation cost ~3 small H2D uploads
+    (~12 µs) per query in the serving loop."""
     used = _vars_of(ast)
-    var_cols: Dict[str, int] = {}
-    col_list = []
-    for v in used:
-        if bindings.has(v):
-            var_cols[v] = len(col_list)
-            col_list.append(bindings.col(v).contiguous())
-    if len(col_list) > 16:
+    present = tuple(v for v in used if bindings.has(v))
+    dev = bindings.device
+    cache = None
+    if cache_holder is not None:
+        cache = getattr(cache_holder, "_fb_cache", None)
+        if cache is None:
+            cache = cache_holder._fb_cache = {}
+        hit = cache.get((present, dev))
+        if hit is False:
+            return None
+        if hit is not None:
+            ops_t, args_t, consts_t = hit
+            return ops_t, args_t, consts_t, \
+                [bindings.col(v).contiguous() for v in present]
+    if len(present) > 16:
+        if cache is not None:
+            cache[(present, dev)] = False
         return None
+    var_cols = {v: i for i, v in enumerate(present)}
     c = _Compiler(var_cols)
     try:
         c.push_bool(ast)
     except BytecodeError:
+        if cache is not None:
+            cache[(present, dev)] = False
         return None
     if len(c.ops) > 128:
+        if cache is not None:
+            cache[(present, dev)] = False
         return None
-    dev = bindings.device
     ops_t = torch.tensor(c.ops, dtype=torch.int32, device=dev)
     args_t = torch.tensor(c.args, dtype=torch.int32, device=dev)
     consts_t = torch.tensor(c.consts or [0.0], dtype=torch.float64, device=dev)
-    return ops_t, args_t, consts_t, col_list
+    if cache is not None:
+        cache[(present, dev)] = (ops_t, args_t, consts_t)
+    return ops_t, args_t, consts_t, \
+        [bindings.col(v).contiguous() for v in present]
