@@ -427,6 +427,18 @@ def _decode_column(rows: Bindings, v: str, db) -> List[str]:
     quoted = (ids & 0x8000_0000) != 0
     ok = (ids < n_plain) & ~quoted
     out[ok] = table[ids[ok]]
+    annex = db.dictionary.annex
+    if annex is not None:
+        # bulk-vocabulary annex ids batch-decode natively (only the
+        # REQUESTED ids materialize as Python strings)
+        amask = ~quoted & (ids >= n_plain) & (ids != 0xFFFFFFFF)
+        if amask.any():
+            idxs = np.nonzero(amask)[0]
+            lst = annex[0].vocab_decode_batch(
+                annex[1], torch.from_numpy(ids[idxs].astype(np.int64)))
+            arr = np.array(lst, dtype=object)
+            arr[np.equal(arr, None)] = ""
+            out[idxs] = arr
     for i in np.nonzero(quoted & (ids != 0xFFFFFFFF))[0]:
         out[i] = db.decode_term(int(ids[i])) or ""
     return out.tolist()

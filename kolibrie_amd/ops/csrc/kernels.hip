@@ -21,6 +21,7 @@
 
 #include <algorithm>
 #include <cstdint>
+#include <atomic>
 #include <chrono>
 #include <cmath>
 #include <string_view>
@@ -1964,6 +1965,272 @@ py::tuple parse_ntriples_file_encode(const std::string& path,
   return py::make_tuple(ids_t, pyfallback);
 }
 
+// ------------------------------------------------------------ vocab annex
+// Native bulk-vocabulary store (the C++-primary dictionary tail): once a
+// database bulk-loads, the Python dictionary's prefix [0, P0) freezes and
+// EVERY later term id lives here — sharded string->id maps with stable
+// deque arenas, a by-id pointer index, and the float64 value column.
+// Python's Dictionary consults it on lookup/encode/decode misses; result
+// decode uses the batch exporter so only requested ids materialize as
+// Python strings.
+struct VocabShard {
+  std::unordered_map<std::string_view, int64_t> map;
+  std::deque<std::string> arena;
+};
+
+struct Vocab {
+  int nshards = 32;
+  std::vector<VocabShard> shards;
+  std::vector<const std::string*> by_id;  // index = id - base
+  std::vector<double> values;
+  int64_t base = -1;
+  Vocab() : shards(nshards) {}
+};
+static std::vector<std::unique_ptr<Vocab>> g_vocabs;
+
+static double parse_value_full(const std::string& s) {
+  char c0 = s.empty() ? 0 : s[0];
+  if (!((c0 >= '0' && c0 <= '9') || c0 == '-' || c0 == '+' || c0 == '.'
+        || c0 == ' ' || c0 == '\t' || c0 == 'i' || c0 == 'I'
+        || c0 == 'n' || c0 == 'N'))
+    return 0.0;
+  if (s.size() >= 63) return 0.0;
+  char buf[64];
+  memcpy(buf, s.data(), s.size());
+  buf[s.size()] = '\0';
+  char* endp = nullptr;
+  double parsed = strtod(buf, &endp);
+  if (endp == buf || endp == nullptr) return 0.0;
+  while (*endp == ' ' || *endp == '\t') ++endp;
+  if (*endp != '\0' || !std::isfinite(parsed)) return 0.0;
+  return parsed;
+}
+
+int64_t vocab_create(int64_t base_id) {
+  auto v = std::make_unique<Vocab>();
+  v->base = base_id;
+  g_vocabs.push_back(std::move(v));
+  return static_cast<int64_t>(g_vocabs.size() - 1);
+}
+
+int64_t vocab_len(int64_t h) {
+  return static_cast<int64_t>(g_vocabs.at(h)->by_id.size());
+}
+
+int64_t vocab_lookup(int64_t h, const std::string& s) {
+  Vocab& v = *g_vocabs.at(h);
+  std::hash<std::string_view> hasher;
+  auto& sh = v.shards[hasher(std::string_view(s)) % v.nshards];
+  auto it = sh.map.find(std::string_view(s));
+  return it == sh.map.end() ? -1 : it->second;
+}
+
+int64_t vocab_insert(int64_t h, const std::string& s) {
+  Vocab& v = *g_vocabs.at(h);
+  std::hash<std::string_view> hasher;
+  auto& sh = v.shards[hasher(std::string_view(s)) % v.nshards];
+  auto it = sh.map.find(std::string_view(s));
+  if (it != sh.map.end()) return it->second;
+  sh.arena.push_back(s);
+  int64_t id = v.base + static_cast<int64_t>(v.by_id.size());
+  sh.map.emplace(std::string_view(sh.arena.back()), id);
+  v.by_id.push_back(&sh.arena.back());
+  v.values.push_back(parse_value_full(sh.arena.back()));
+  return id;
+}
+
+py::object vocab_get(int64_t h, int64_t id) {
+  Vocab& v = *g_vocabs.at(h);
+  int64_t k = id - v.base;
+  if (k < 0 || k >= static_cast<int64_t>(v.by_id.size()))
+    return py::none();
+  const std::string& s = *v.by_id[k];
+  PyObject* u = PyUnicode_DecodeUTF8(s.data(),
+                                     static_cast<Py_ssize_t>(s.size()),
+                                     "replace");
+  return py::reinterpret_steal<py::object>(u);
+}
+
+// batch decode for result materialization: CPU int64 id tensor -> Python
+// list of str (ids outside the annex range decode to None)
+py::list vocab_decode_batch(int64_t h, at::Tensor ids) {
+  Vocab& v = *g_vocabs.at(h);
+  TORCH_CHECK(!ids.is_cuda() && ids.dtype() == at::kLong);
+  const int64_t* r = ids.data_ptr<int64_t>();
+  int64_t n = ids.numel();
+  py::list out(n);
+  for (int64_t i = 0; i < n; ++i) {
+    int64_t k = r[i] - v.base;
+    PyObject* o;
+    if (k < 0 || k >= static_cast<int64_t>(v.by_id.size())) {
+      o = Py_None;
+      Py_INCREF(o);
+    } else {
+      const std::string& s = *v.by_id[k];
+      o = PyUnicode_DecodeUTF8(s.data(),
+                               static_cast<Py_ssize_t>(s.size()), "replace");
+      if (o == nullptr) { o = Py_None; Py_INCREF(o); }
+    }
+    PyList_SET_ITEM(out.ptr(), i, o);
+  }
+  return out;
+}
+
+double vocab_value(int64_t h, int64_t id) {
+  Vocab& v = *g_vocabs.at(h);
+  int64_t k = id - v.base;
+  if (k < 0 || k >= static_cast<int64_t>(v.values.size())) return 0.0;
+  return v.values[k];
+}
+
+at::Tensor vocab_values(int64_t h) {
+  Vocab& v = *g_vocabs.at(h);
+  return at::from_blob(v.values.data(),
+                       {static_cast<int64_t>(v.values.size())},
+                       at::kDouble).clone();
+}
+
+py::list vocab_export_strings(int64_t h, int64_t start, int64_t count) {
+  Vocab& v = *g_vocabs.at(h);
+  py::list out;
+  int64_t n = static_cast<int64_t>(v.by_id.size());
+  for (int64_t k = start; k < std::min(start + count, n); ++k) {
+    const std::string& s = *v.by_id[k];
+    PyObject* u = PyUnicode_DecodeUTF8(
+        s.data(), static_cast<Py_ssize_t>(s.size()), "replace");
+    out.append(py::reinterpret_steal<py::object>(u));
+  }
+  return out;
+}
+
+// file -> parse -> intern DIRECTLY into the vocab annex (no Python-object
+// pass at all); returns (int32 [n,3] global-id rows, fallback lines)
+py::tuple parse_ntriples_file_annex(const std::string& path,
+                                    int64_t n_threads, int64_t h,
+                                    int64_t max_id) {
+  Vocab& voc = *g_vocabs.at(h);
+  std::string text;
+  {
+    py::gil_scoped_release release;
+    FILE* f = fopen(path.c_str(), "rb");
+    TORCH_CHECK(f != nullptr, "cannot open ", path);
+    fseek(f, 0, SEEK_END);
+    long sz = ftell(f);
+    fseek(f, 0, SEEK_SET);
+    text.resize(static_cast<size_t>(sz));
+    size_t rd = fread(text.data(), 1, static_cast<size_t>(sz), f);
+    fclose(f);
+    TORCH_CHECK(rd == static_cast<size_t>(sz), "short read of ", path);
+  }
+  const size_t n = text.size();
+  int nt = static_cast<int>(n_threads);
+  if (nt <= 0) {
+    nt = static_cast<int>(std::thread::hardware_concurrency());
+    if (nt <= 0) nt = 8;
+  }
+  if (nt > 32) nt = 32;
+  if (n < (1 << 20)) nt = 1;
+  std::vector<size_t> starts(1, 0);
+  for (int i = 1; i < nt; ++i) {
+    size_t p = n * static_cast<size_t>(i) / nt;
+    const char* nl = static_cast<const char*>(
+        memchr(text.data() + p, '\n', n - p));
+    starts.push_back(nl ? static_cast<size_t>(nl - text.data()) + 1 : n);
+  }
+  starts.push_back(n);
+  std::vector<ParseChunkOut> outs(nt);
+  std::vector<std::vector<int64_t>> remaps(nt);
+  std::vector<int64_t> line_off(nt + 1, 0);
+  int64_t total_rows = 0;
+  at::Tensor ids_t;
+  {
+    py::gil_scoped_release release;
+    std::vector<std::thread> threads;
+    for (int i = 0; i < nt; ++i)
+      threads.emplace_back(parse_chunk_nt, text.data(), starts[i],
+                           starts[i + 1], false, std::ref(outs[i]));
+    for (auto& t : threads) t.join();
+    for (int i = 0; i < nt; ++i) {
+      remaps[i].resize(outs[i].views.size());
+      line_off[i + 1] = line_off[i] + outs[i].line_count;
+      total_rows += static_cast<int64_t>(outs[i].ids.size());
+    }
+    // sharded merge straight into the vocab: the vocab's shard count is
+    // the parallelism; thread t owns vocab shard t.  IDs come from one
+    // atomic counter (dense; assignment order is thread-interleaved, so
+    // annex ids are per-process — the distributed replicated-dictionary
+    // loaders use the Python-dict paths, not the annex).
+    const int S = voc.nshards;
+    std::hash<std::string_view> hasher;
+    std::vector<std::thread> mthreads;
+    std::atomic<int64_t> next{voc.base
+                              + static_cast<int64_t>(voc.by_id.size())};
+    std::vector<std::vector<std::pair<const std::string*, int64_t>>>
+        pending(S);
+    for (int t = 0; t < S; ++t)
+      mthreads.emplace_back([&, t]() {
+        auto& sh = voc.shards[t];
+        for (int i = 0; i < nt; ++i) {
+          auto& views = outs[i].views;
+          auto& rm = remaps[i];
+          for (size_t k = 0; k < views.size(); ++k) {
+            if (static_cast<int>(hasher(views[k]) % S) != t) continue;
+            auto it = sh.map.find(views[k]);
+            if (it != sh.map.end()) {
+              rm[k] = it->second;
+              continue;
+            }
+            sh.arena.emplace_back(views[k]);
+            int64_t id = next.fetch_add(1, std::memory_order_relaxed);
+            sh.map.emplace(std::string_view(sh.arena.back()), id);
+            pending[t].emplace_back(&sh.arena.back(), id);
+            rm[k] = id;
+          }
+        }
+      });
+    for (auto& t : mthreads) t.join();
+    // register new entries in by_id/values (ids are dense: base+old .. next)
+    int64_t old_n = static_cast<int64_t>(voc.by_id.size());
+    int64_t new_total = next.load() - voc.base;
+    TORCH_CHECK(voc.base + new_total <= max_id,
+                "dictionary ID space exhausted");
+    voc.by_id.resize(new_total, nullptr);
+    voc.values.resize(new_total, 0.0);
+    std::vector<std::thread> vthreads;
+    for (int t = 0; t < S; ++t)
+      vthreads.emplace_back([&, t]() {
+        for (auto& pr : pending[t]) {
+          int64_t k = pr.second - voc.base;
+          voc.by_id[k] = pr.first;
+          voc.values[k] = parse_value_full(*pr.first);
+        }
+      });
+    for (auto& t : vthreads) t.join();
+    for (int64_t k = old_n; k < new_total; ++k)
+      TORCH_CHECK(voc.by_id[k] != nullptr, "vocab id gap at ", k);
+    // remap rows to global ids, in parallel
+    ids_t = at::empty({total_rows / 3, 3}, at::kInt);
+    std::vector<int64_t> offs(nt + 1, 0);
+    for (int i = 0; i < nt; ++i)
+      offs[i + 1] = offs[i] + static_cast<int64_t>(outs[i].ids.size());
+    int32_t* dst = ids_t.data_ptr<int32_t>();
+    std::vector<std::thread> rthreads;
+    for (int i = 0; i < nt; ++i)
+      rthreads.emplace_back([&, i]() {
+        auto& rm = remaps[i];
+        int32_t* d = dst + offs[i];
+        for (size_t k = 0; k < outs[i].ids.size(); ++k)
+          d[k] = static_cast<int32_t>(
+              static_cast<uint32_t>(rm[outs[i].ids[k]]));
+      });
+    for (auto& t : rthreads) t.join();
+  }
+  py::list pyfallback;
+  for (int i = 0; i < nt; ++i)
+    for (auto f : outs[i].fallback) pyfallback.append(f + line_off[i]);
+  return py::make_tuple(ids_t, pyfallback);
+}
+
 py::tuple parse_ntriples_file_mt(const std::string& path,
                                  int64_t n_threads) {
   std::string text;
@@ -2771,6 +3038,17 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "global-id int32 rows");
   m.def("gen_employee_nt_file", &gen_employee_nt_file,
         "write a synthetic employee N-Triples file (bench tooling)");
+  m.def("vocab_create", &vocab_create, "native bulk-vocabulary annex");
+  m.def("vocab_len", &vocab_len);
+  m.def("vocab_lookup", &vocab_lookup);
+  m.def("vocab_insert", &vocab_insert);
+  m.def("vocab_get", &vocab_get);
+  m.def("vocab_decode_batch", &vocab_decode_batch);
+  m.def("vocab_value", &vocab_value);
+  m.def("vocab_values", &vocab_values);
+  m.def("vocab_export_strings", &vocab_export_strings);
+  m.def("parse_ntriples_file_annex", &parse_ntriples_file_annex,
+        "file -> parallel parse -> intern into the native vocab annex");
   m.doc() = "kolibrie_amd native CDNA4 kernels (gfx950)";
   m.def("probe_exact", &probe_exact,
         "K1 scan-probe, packed (a,b) exact keys -> (li, b, z)");

@@ -44,7 +44,7 @@ class _MiniDb:
         n = len(self.dictionary)
         if self._value_col_cache is None or self._value_col_len < n:
             import numpy as np
-            arr = np.asarray(self.dictionary.values, dtype=np.float64)
+            arr = self.dictionary.values_array()
             self._value_col_cache = torch.from_numpy(arr).to(self.device)
             self._value_col_len = n
         return self._value_col_cache

@@ -56,7 +56,7 @@ def save_binary(db, path: str, rank: int = 0):
         # one JSON string per line: robust against \n, \r, backslashes and
         # any other control characters inside terms (v1's ad-hoc escaping
         # corrupted strings containing '\r' or literal "\n")
-        for s in db.dictionary.id_to_str:
+        for s in db.dictionary.iter_strings():
             f.write(json.dumps(s, ensure_ascii=False) + "\n")
     if len(db.quoted_triples):
         qt = np.asarray(db.quoted_triples.id_to_triple, dtype=np.uint32)
@@ -101,8 +101,8 @@ def load_binary(db, path: str):
         # re-encoding into a non-empty dictionary would remap IDs and
         # silently corrupt the restored columns; allow only an exact
         # prefix match (e.g. re-loading into the same process)
-        for i, s in enumerate(strings[:len(d.id_to_str)]):
-            if d.id_to_str[i] != s:
+        for i, s in enumerate(strings[:len(d)]):
+            if d.decode(i) != s:
                 raise ValueError(
                     "load_binary into a non-empty dictionary whose entries "
                     f"differ from the checkpoint at ID {i}: restored "

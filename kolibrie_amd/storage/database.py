@@ -275,9 +275,9 @@ class SparqlDatabase:
                 return self.parse_ntriples(f.read())
         from .terms import QUOTED_TRIPLE_ID_BIT
         d = self.dictionary
-        ids, fallback = _native.parse_ntriples_file_encode(
-            path, 0, d.str_to_id, d.id_to_str, d.values,
-            QUOTED_TRIPLE_ID_BIT)
+        mod, h = d.attach_annex()
+        ids, fallback = mod.parse_ntriples_file_annex(
+            path, 0, h, QUOTED_TRIPLE_ID_BIT)
         d._values_dirty = True
         if ids.numel():
             arr = ids.numpy()
@@ -392,8 +392,7 @@ class SparqlDatabase:
         comparisons and aggregates so they never see strings."""
         n = len(self.dictionary)
         if self._value_col_cache is None or self._value_col_len < n:
-            import numpy as np
-            arr = np.asarray(self.dictionary.values, dtype=np.float64)
+            arr = self.dictionary.values_array()
             self._value_col_cache = torch.from_numpy(arr).to(self.device)
             self._value_col_len = n
         return self._value_col_cache

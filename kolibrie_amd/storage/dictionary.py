@@ -36,7 +36,7 @@ class Dictionary:
     as the default-graph ID)."""
 
     __slots__ = ("str_to_id", "id_to_str", "values", "_values_dirty",
-                 "_np_table", "_np_table_n")
+                 "_np_table", "_np_table_n", "annex")
 
     def __init__(self):
         self.str_to_id: Dict[str, int] = {"": 0}
@@ -47,6 +47,20 @@ class Dictionary:
         self._values_dirty = True
         self._np_table = None     # numpy object array mirror of id_to_str
         self._np_table_n = 0
+        # native bulk-vocabulary annex (C++-primary dictionary tail): once
+        # attached by a bulk file load, the Python prefix [0, P0) freezes
+        # and every later term id (bulk or incremental) lives natively;
+        # lookup/encode/decode consult it on prefix misses.  (module, handle)
+        self.annex = None
+
+    def attach_annex(self):
+        if self.annex is None:
+            from ..ops import _native
+            if _native is None:
+                raise RuntimeError("native extension required for the "
+                                   "bulk-vocabulary annex")
+            self.annex = (_native, _native.vocab_create(len(self.id_to_str)))
+        return self.annex
 
     def np_table(self):
         """Numpy object-array mirror of id_to_str for BATCH decode
@@ -65,12 +79,23 @@ class Dictionary:
         return self._np_table
 
     def __len__(self):
-        return len(self.id_to_str)
+        n = len(self.id_to_str)
+        if self.annex is not None:
+            n += self.annex[0].vocab_len(self.annex[1])
+        return n
 
     def encode(self, s: str) -> int:
         """Intern a string, returning its u32 ID (ref dictionary.rs:32)."""
         i = self.str_to_id.get(s)
         if i is not None:
+            return i
+        if self.annex is not None:
+            mod, h = self.annex
+            i = mod.vocab_insert(h, s)
+            if i >= QUOTED_TRIPLE_ID_BIT:
+                raise OverflowError(
+                    "dictionary ID space exhausted (2^31 terms)")
+            self._values_dirty = True
             return i
         i = len(self.id_to_str)
         if i >= QUOTED_TRIPLE_ID_BIT:
@@ -86,6 +111,12 @@ class Dictionary:
         local-variable loop is ~2x a per-call encode() — this is the host
         half of the parallel bulk-parse pipeline."""
         import numpy as np
+        if self.annex is not None:
+            out = np.empty(len(strs), dtype=np.uint32)
+            enc = self.encode
+            for i, s in enumerate(strs):
+                out[i] = enc(s)
+            return out
         sti = self.str_to_id
         its = self.id_to_str
         vals = self.values
@@ -110,29 +141,60 @@ class Dictionary:
         return out
 
     def lookup(self, s: str) -> Optional[int]:
-        return self.str_to_id.get(s)
+        i = self.str_to_id.get(s)
+        if i is None and self.annex is not None:
+            j = self.annex[0].vocab_lookup(self.annex[1], s)
+            return None if j < 0 else j
+        return i
 
     def decode(self, i: int) -> Optional[str]:
         """ID -> string (plain terms only; ref dictionary.rs:49)."""
         i &= 0xFFFFFFFF
         if i < len(self.id_to_str):
             return self.id_to_str[i]
+        if self.annex is not None:
+            return self.annex[0].vocab_get(self.annex[1], i)
         return None
 
     def contains(self, s: str) -> bool:
-        return s in self.str_to_id
+        if s in self.str_to_id:
+            return True
+        return (self.annex is not None
+                and self.annex[0].vocab_lookup(self.annex[1], s) >= 0)
 
     def numeric_value(self, i: int) -> float:
         i &= 0xFFFFFFFF
         if i < len(self.values):
             return self.values[i]
+        if self.annex is not None:
+            return self.annex[0].vocab_value(self.annex[1], i)
         return 0.0
+
+    def values_array(self):
+        """float64 numpy array over the WHOLE id space (prefix + annex) —
+        the device value-column source."""
+        import numpy as np
+        pre = np.asarray(self.values, dtype=np.float64)
+        if self.annex is None:
+            return pre
+        tail = self.annex[0].vocab_values(self.annex[1]).numpy()
+        return np.concatenate([pre, tail]) if tail.size else pre
+
+    def iter_strings(self):
+        """Iterate every interned string in id order (checkpoint/merge);
+        annex strings materialize in batches."""
+        yield from self.id_to_str
+        if self.annex is not None:
+            mod, h = self.annex
+            n = mod.vocab_len(h)
+            for start in range(0, n, 1 << 16):
+                yield from mod.vocab_export_strings(h, start, 1 << 16)
 
     def merge(self, other: "Dictionary") -> Dict[int, int]:
         """Merge `other` into self, returning an old-id -> new-id remap
         (ref dictionary.rs:82 — used by parallel parse shards)."""
         remap: Dict[int, int] = {}
-        for old_id, s in enumerate(other.id_to_str):
+        for old_id, s in enumerate(other.iter_strings()):
             remap[old_id] = self.encode(s)
         return remap
 

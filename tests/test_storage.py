@@ -247,3 +247,55 @@ def test_parse_ntriples_file_roundtrip(tmp_path):
     assert db1.query(q) == db2.query(q) == [["250"]]
     q2 = f'SELECT ?n WHERE {{ <{EX}e7> <{EX}name> ?n }}'
     assert db1.query(q2) == db2.query(q2)
+
+
+def test_vocab_annex_semantics(tmp_path):
+    """Bulk-vocabulary annex: ids from the native tail must behave exactly
+    like python-dict ids — encode/lookup/decode/len/value-column/
+    checkpoint round trip, plus post-annex incremental encodes."""
+    import numpy as np
+    import pytest
+    from kolibrie_amd import SparqlDatabase
+    from kolibrie_amd.ops import _native
+    from kolibrie_amd.storage import checkpoint as cp
+    if _native is None:
+        pytest.skip("native extension required")
+    EX = "http://example.org/"
+    lines = []
+    for i in range(300):
+        lines.append(f'<{EX}s{i}> <{EX}num> "{i * 3}" .')
+        lines.append(f'<{EX}s{i}> <{EX}lab> "name {i}" .')
+    p = tmp_path / "a.nt"
+    p.write_text("\n".join(lines) + "\n")
+    db = SparqlDatabase()
+    db.parse_ntriples_file(str(p))
+    d = db.dictionary
+    assert d.annex is not None
+    assert len(d) > 600
+    # lookup/contains/decode through the annex
+    sid = d.lookup(f"{EX}s7")
+    assert sid is not None and d.decode(sid) == f"{EX}s7"
+    assert d.contains("21") and not d.contains("nope-not-here")
+    # numeric value column covers annex ids (FILTER semantics)
+    rows = db.query(f"SELECT (COUNT(*) AS ?c) WHERE {{ ?s <{EX}num> ?v . "
+                    f"FILTER(?v >= 450) }}")
+    assert rows == [[str(sum(1 for i in range(300) if i * 3 >= 450))]]
+    # post-annex incremental encode allocates non-colliding ids
+    nid = d.encode("brand-new-term")
+    assert d.decode(nid) == "brand-new-term"
+    assert d.lookup("brand-new-term") == nid
+    # row decode materializes annex strings
+    out = db.query(f"SELECT ?s ?v WHERE {{ ?s <{EX}num> ?v }} "
+                   f"ORDER BY ?s LIMIT 3")
+    assert out[0][0].startswith(EX)
+    # values_array covers the whole id space
+    assert len(d.values_array()) == len(d)
+    # checkpoint round trip through iter_strings
+    path = str(tmp_path / "annexed.npz")
+    cp.save_binary(db, path, rank=0)
+    db2 = SparqlDatabase()
+    cp.load_binary(db2, path)
+    assert db2.query(f"SELECT (COUNT(*) AS ?c) WHERE {{ ?s ?p ?o }}") == \
+        db.query(f"SELECT (COUNT(*) AS ?c) WHERE {{ ?s ?p ?o }}")
+    assert db2.query(f'SELECT ?s WHERE {{ ?s <{EX}lab> "name 5" }}') == \
+        db.query(f'SELECT ?s WHERE {{ ?s <{EX}lab> "name 5" }}')
