@@ -1585,6 +1585,7 @@ struct ParseChunkOut {
   // stable addresses) for unescaped materializations
   std::unordered_map<std::string_view, int64_t> interned;
   std::vector<std::string_view> views;
+  std::vector<uint64_t> hashes;    // std::hash of each view, precomputed
   std::deque<std::string> owned;
   std::vector<int64_t> ids;
   std::vector<int64_t> fallback;   // chunk-local line numbers
@@ -1593,12 +1594,14 @@ struct ParseChunkOut {
 
 static void parse_chunk_nt(const char* data, size_t begin, size_t end,
                            bool quads, ParseChunkOut& out) {
+  std::hash<std::string_view> sv_hasher;
   auto intern_sv = [&](std::string_view sv) -> int64_t {
     auto it = out.interned.find(sv);
     if (it != out.interned.end()) return it->second;
     int64_t id = static_cast<int64_t>(out.views.size());
     out.interned.emplace(sv, id);
     out.views.push_back(sv);
+    out.hashes.push_back(sv_hasher(sv));
     return id;
   };
   auto intern = [&](const char* ptr, size_t len) -> int64_t {
@@ -1612,6 +1615,7 @@ static void parse_chunk_nt(const char* data, size_t begin, size_t end,
     int64_t id = static_cast<int64_t>(out.views.size());
     out.interned.emplace(sv, id);
     out.views.push_back(sv);
+    out.hashes.push_back(sv_hasher(sv));
     return id;
   };
   size_t pos = begin;
@@ -1851,15 +1855,15 @@ py::tuple parse_ntriples_file_encode(const std::string& path,
       std::vector<std::thread> mthreads;
       for (int t = 0; t < nt; ++t)
         mthreads.emplace_back([&, t]() {
-          std::hash<std::string_view> hasher;
           std::unordered_map<std::string_view, int64_t> local;
           local.reserve(uniq_upper / std::max(1, nt) * 2 + 64);
           auto& sv = shard_views[t];
           for (int i = 0; i < nt; ++i) {
             auto& views = outs[i].views;
+            auto& hs = outs[i].hashes;
             auto& rm = remaps[i];
             for (size_t k = 0; k < views.size(); ++k) {
-              if (static_cast<int>(hasher(views[k]) % nt) != t) continue;
+              if (static_cast<int>(hs[k] % nt) != t) continue;
               auto it = local.find(views[k]);
               int64_t L;
               if (it == local.end()) {
@@ -2108,6 +2112,15 @@ py::list vocab_export_strings(int64_t h, int64_t start, int64_t count) {
 py::tuple parse_ntriples_file_annex(const std::string& path,
                                     int64_t n_threads, int64_t h,
                                     int64_t max_id) {
+  const bool dbg = getenv("KOLIBRIE_PARSE_DEBUG") != nullptr;
+  auto tick = std::chrono::steady_clock::now();
+  auto lap = [&](const char* what) {
+    if (!dbg) return;
+    auto now = std::chrono::steady_clock::now();
+    fprintf(stderr, "[annex] %s: %.3fs\n", what,
+            std::chrono::duration<double>(now - tick).count());
+    tick = now;
+  };
   Vocab& voc = *g_vocabs.at(h);
   std::string text;
   {
@@ -2122,6 +2135,7 @@ py::tuple parse_ntriples_file_annex(const std::string& path,
     fclose(f);
     TORCH_CHECK(rd == static_cast<size_t>(sz), "short read of ", path);
   }
+  lap("read");
   const size_t n = text.size();
   int nt = static_cast<int>(n_threads);
   if (nt <= 0) {
@@ -2150,48 +2164,54 @@ py::tuple parse_ntriples_file_annex(const std::string& path,
       threads.emplace_back(parse_chunk_nt, text.data(), starts[i],
                            starts[i + 1], false, std::ref(outs[i]));
     for (auto& t : threads) t.join();
+    if (dbg) { py::gil_scoped_acquire a; lap("parallel parse"); }
     for (int i = 0; i < nt; ++i) {
       remaps[i].resize(outs[i].views.size());
       line_off[i + 1] = line_off[i] + outs[i].line_count;
       total_rows += static_cast<int64_t>(outs[i].ids.size());
     }
     // sharded merge straight into the vocab: the vocab's shard count is
-    // the parallelism; thread t owns vocab shard t.  IDs come from one
-    // atomic counter (dense; assignment order is thread-interleaved, so
-    // annex ids are per-process — the distributed replicated-dictionary
-    // loaders use the Python-dict paths, not the annex).
+    // the parallelism; thread t owns vocab shard t.  Two phases: dedup
+    // with PROVISIONAL (shard, local) ids, then per-shard contiguous
+    // real-id blocks from prefix sums — no contended atomic, and ids are
+    // deterministic for a given input + shard count.
     const int S = voc.nshards;
-    std::hash<std::string_view> hasher;
     std::vector<std::thread> mthreads;
-    std::atomic<int64_t> next{voc.base
-                              + static_cast<int64_t>(voc.by_id.size())};
-    std::vector<std::vector<std::pair<const std::string*, int64_t>>>
-        pending(S);
+    std::vector<std::vector<const std::string*>> pending(S);
+    const int64_t kProv = 1LL << 62;  // provisional marker
     for (int t = 0; t < S; ++t)
       mthreads.emplace_back([&, t]() {
         auto& sh = voc.shards[t];
+        auto& pend = pending[t];
         for (int i = 0; i < nt; ++i) {
           auto& views = outs[i].views;
+          auto& hs = outs[i].hashes;
           auto& rm = remaps[i];
           for (size_t k = 0; k < views.size(); ++k) {
-            if (static_cast<int>(hasher(views[k]) % S) != t) continue;
+            if (static_cast<int>(hs[k] % S) != t) continue;
             auto it = sh.map.find(views[k]);
             if (it != sh.map.end()) {
-              rm[k] = it->second;
+              rm[k] = it->second;   // may itself be provisional (this call)
               continue;
             }
             sh.arena.emplace_back(views[k]);
-            int64_t id = next.fetch_add(1, std::memory_order_relaxed);
-            sh.map.emplace(std::string_view(sh.arena.back()), id);
-            pending[t].emplace_back(&sh.arena.back(), id);
-            rm[k] = id;
+            int64_t prov = kProv | (static_cast<int64_t>(t) << 40)
+                           | static_cast<int64_t>(pend.size());
+            sh.map.emplace(std::string_view(sh.arena.back()), prov);
+            pend.push_back(&sh.arena.back());
+            rm[k] = prov;
           }
         }
       });
     for (auto& t : mthreads) t.join();
-    // register new entries in by_id/values (ids are dense: base+old .. next)
+    if (dbg) { py::gil_scoped_acquire a; lap("vocab merge"); }
+    // prefix sums -> real id blocks; fix the maps + fill by_id/values
     int64_t old_n = static_cast<int64_t>(voc.by_id.size());
-    int64_t new_total = next.load() - voc.base;
+    std::vector<int64_t> shard_base(S + 1, 0);
+    for (int t = 0; t < S; ++t)
+      shard_base[t + 1] = shard_base[t]
+                          + static_cast<int64_t>(pending[t].size());
+    int64_t new_total = old_n + shard_base[S];
     TORCH_CHECK(voc.base + new_total <= max_id,
                 "dictionary ID space exhausted");
     voc.by_id.resize(new_total, nullptr);
@@ -2199,15 +2219,31 @@ py::tuple parse_ntriples_file_annex(const std::string& path,
     std::vector<std::thread> vthreads;
     for (int t = 0; t < S; ++t)
       vthreads.emplace_back([&, t]() {
-        for (auto& pr : pending[t]) {
-          int64_t k = pr.second - voc.base;
-          voc.by_id[k] = pr.first;
-          voc.values[k] = parse_value_full(*pr.first);
+        auto& sh = voc.shards[t];
+        int64_t base = voc.base + old_n + shard_base[t];
+        for (size_t j = 0; j < pending[t].size(); ++j) {
+          const std::string* sp = pending[t][j];
+          int64_t id = base + static_cast<int64_t>(j);
+          sh.map[std::string_view(*sp)] = id;
+          int64_t k = id - voc.base;
+          voc.by_id[k] = sp;
+          voc.values[k] = parse_value_full(*sp);
         }
       });
     for (auto& t : vthreads) t.join();
-    for (int64_t k = old_n; k < new_total; ++k)
-      TORCH_CHECK(voc.by_id[k] != nullptr, "vocab id gap at ", k);
+    // resolve provisional remap entries to real ids
+    std::vector<std::thread> fthreads;
+    for (int i = 0; i < nt; ++i)
+      fthreads.emplace_back([&, i]() {
+        for (auto& e : remaps[i]) {
+          if (e & kProv) {
+            int64_t t = (e >> 40) & 0x3FFFFF;
+            int64_t j = e & ((1LL << 40) - 1);
+            e = voc.base + old_n + shard_base[t] + j;
+          }
+        }
+      });
+    for (auto& t : fthreads) t.join();
     // remap rows to global ids, in parallel
     ids_t = at::empty({total_rows / 3, 3}, at::kInt);
     std::vector<int64_t> offs(nt + 1, 0);
@@ -2224,6 +2260,7 @@ py::tuple parse_ntriples_file_annex(const std::string& path,
               static_cast<uint32_t>(rm[outs[i].ids[k]]));
       });
     for (auto& t : rthreads) t.join();
+    if (dbg) { py::gil_scoped_acquire a; lap("register+remap"); }
   }
   py::list pyfallback;
   for (int i = 0; i < nt; ++i)

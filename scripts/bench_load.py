@@ -78,6 +78,7 @@ def bench_from_nt(total: int, device: str):
     t0 = time.perf_counter()
     db.parse_ntriples_file(path)
     t_parse = time.perf_counter()
+    print(f"# parse+intern+insert-call: {t_parse - t0:.2f}s")
     n = db.triple_count()
     for code in range(4):
         _ = db.store.graph_index(0).orders[code]
