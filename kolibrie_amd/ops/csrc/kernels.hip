@@ -1579,6 +1579,8 @@ py::tuple parse_nlines_host(const std::string& text, bool quads) {
 // the whole native phase.
 #include <deque>
 
+constexpr int kVocabShards = 32;
+
 struct ParseChunkOut {
   // term views point into the shared input text (zero-copy for the
   // overwhelmingly common escape-free terms) or into `owned` (deque:
@@ -1586,6 +1588,10 @@ struct ParseChunkOut {
   std::unordered_map<std::string_view, int64_t> interned;
   std::vector<std::string_view> views;
   std::vector<uint64_t> hashes;    // std::hash of each view, precomputed
+  // view indices bucketed by hash % kVocabShards at parse time, so each
+  // vocab-merge shard thread walks only its own items instead of
+  // filter-scanning every chunk's full hash array
+  std::vector<uint32_t> by_shard[kVocabShards];
   std::deque<std::string> owned;
   std::vector<int64_t> ids;
   std::vector<int64_t> fallback;   // chunk-local line numbers
@@ -1601,7 +1607,9 @@ static void parse_chunk_nt(const char* data, size_t begin, size_t end,
     int64_t id = static_cast<int64_t>(out.views.size());
     out.interned.emplace(sv, id);
     out.views.push_back(sv);
-    out.hashes.push_back(sv_hasher(sv));
+    uint64_t hh = sv_hasher(sv);
+    out.hashes.push_back(hh);
+    out.by_shard[hh % kVocabShards].push_back(static_cast<uint32_t>(id));
     return id;
   };
   auto intern = [&](const char* ptr, size_t len) -> int64_t {
@@ -1615,7 +1623,9 @@ static void parse_chunk_nt(const char* data, size_t begin, size_t end,
     int64_t id = static_cast<int64_t>(out.views.size());
     out.interned.emplace(sv, id);
     out.views.push_back(sv);
-    out.hashes.push_back(sv_hasher(sv));
+    uint64_t hh = sv_hasher(sv);
+    out.hashes.push_back(hh);
+    out.by_shard[hh % kVocabShards].push_back(static_cast<uint32_t>(id));
     return id;
   };
   size_t pos = begin;
@@ -1977,22 +1987,42 @@ py::tuple parse_ntriples_file_encode(const std::string& path,
 // Python's Dictionary consults it on lookup/encode/decode misses; result
 // decode uses the batch exporter so only requested ids materialize as
 // Python strings.
+// append-only bump arena: one memcpy per interned string instead of a
+// per-string malloc (a deque<std::string> arena paid ~43M heap
+// allocations on a 100M-triple load — the dominant merge cost)
+struct BumpArena {
+  std::vector<std::unique_ptr<char[]>> blocks;
+  size_t used = 0, cap = 0;
+  std::string_view add(std::string_view sv) {
+    if (used + sv.size() > cap) {
+      size_t bs = std::max<size_t>(size_t(1) << 24, sv.size());
+      blocks.emplace_back(new char[bs]);
+      used = 0;
+      cap = bs;
+    }
+    char* dst = blocks.back().get() + used;
+    memcpy(dst, sv.data(), sv.size());
+    used += sv.size();
+    return std::string_view(dst, sv.size());
+  }
+};
+
 struct VocabShard {
   std::unordered_map<std::string_view, int64_t> map;
-  std::deque<std::string> arena;
+  BumpArena arena;
 };
 
 struct Vocab {
-  int nshards = 32;
+  int nshards = kVocabShards;
   std::vector<VocabShard> shards;
-  std::vector<const std::string*> by_id;  // index = id - base
+  std::vector<std::string_view> by_id;  // index = id - base
   std::vector<double> values;
   int64_t base = -1;
   Vocab() : shards(nshards) {}
 };
 static std::vector<std::unique_ptr<Vocab>> g_vocabs;
 
-static double parse_value_full(const std::string& s) {
+static double parse_value_full(std::string_view s) {
   char c0 = s.empty() ? 0 : s[0];
   if (!((c0 >= '0' && c0 <= '9') || c0 == '-' || c0 == '+' || c0 == '.'
         || c0 == ' ' || c0 == '\t' || c0 == 'i' || c0 == 'I'
@@ -2035,11 +2065,11 @@ int64_t vocab_insert(int64_t h, const std::string& s) {
   auto& sh = v.shards[hasher(std::string_view(s)) % v.nshards];
   auto it = sh.map.find(std::string_view(s));
   if (it != sh.map.end()) return it->second;
-  sh.arena.push_back(s);
+  std::string_view sv = sh.arena.add(std::string_view(s));
   int64_t id = v.base + static_cast<int64_t>(v.by_id.size());
-  sh.map.emplace(std::string_view(sh.arena.back()), id);
-  v.by_id.push_back(&sh.arena.back());
-  v.values.push_back(parse_value_full(sh.arena.back()));
+  sh.map.emplace(sv, id);
+  v.by_id.push_back(sv);
+  v.values.push_back(parse_value_full(sv));
   return id;
 }
 
@@ -2048,7 +2078,7 @@ py::object vocab_get(int64_t h, int64_t id) {
   int64_t k = id - v.base;
   if (k < 0 || k >= static_cast<int64_t>(v.by_id.size()))
     return py::none();
-  const std::string& s = *v.by_id[k];
+  std::string_view s = v.by_id[k];
   PyObject* u = PyUnicode_DecodeUTF8(s.data(),
                                      static_cast<Py_ssize_t>(s.size()),
                                      "replace");
@@ -2070,7 +2100,7 @@ py::list vocab_decode_batch(int64_t h, at::Tensor ids) {
       o = Py_None;
       Py_INCREF(o);
     } else {
-      const std::string& s = *v.by_id[k];
+      std::string_view s = v.by_id[k];
       o = PyUnicode_DecodeUTF8(s.data(),
                                static_cast<Py_ssize_t>(s.size()), "replace");
       if (o == nullptr) { o = Py_None; Py_INCREF(o); }
@@ -2099,7 +2129,7 @@ py::list vocab_export_strings(int64_t h, int64_t start, int64_t count) {
   py::list out;
   int64_t n = static_cast<int64_t>(v.by_id.size());
   for (int64_t k = start; k < std::min(start + count, n); ++k) {
-    const std::string& s = *v.by_id[k];
+    std::string_view s = v.by_id[k];
     PyObject* u = PyUnicode_DecodeUTF8(
         s.data(), static_cast<Py_ssize_t>(s.size()), "replace");
     out.append(py::reinterpret_steal<py::object>(u));
@@ -2122,7 +2152,17 @@ py::tuple parse_ntriples_file_annex(const std::string& path,
     tick = now;
   };
   Vocab& voc = *g_vocabs.at(h);
-  std::string text;
+  // the bulky scratch (10 GB file text + per-chunk intern maps with tens
+  // of millions of nodes) lives on the heap and is freed by a DETACHED
+  // reaper thread after return — its node-by-node teardown takes seconds
+  // and would otherwise serialize at scope exit, after all useful work
+  struct AnnexScratch {
+    std::string text;
+    std::vector<ParseChunkOut> outs;
+    std::vector<std::vector<int64_t>> remaps;
+  };
+  auto scratch = std::make_unique<AnnexScratch>();
+  std::string& text = scratch->text;
   {
     py::gil_scoped_release release;
     FILE* f = fopen(path.c_str(), "rb");
@@ -2152,8 +2192,10 @@ py::tuple parse_ntriples_file_annex(const std::string& path,
     starts.push_back(nl ? static_cast<size_t>(nl - text.data()) + 1 : n);
   }
   starts.push_back(n);
-  std::vector<ParseChunkOut> outs(nt);
-  std::vector<std::vector<int64_t>> remaps(nt);
+  scratch->outs.resize(nt);
+  scratch->remaps.resize(nt);
+  std::vector<ParseChunkOut>& outs = scratch->outs;
+  std::vector<std::vector<int64_t>>& remaps = scratch->remaps;
   std::vector<int64_t> line_off(nt + 1, 0);
   int64_t total_rows = 0;
   at::Tensor ids_t;
@@ -2177,7 +2219,7 @@ py::tuple parse_ntriples_file_annex(const std::string& path,
     // deterministic for a given input + shard count.
     const int S = voc.nshards;
     std::vector<std::thread> mthreads;
-    std::vector<std::vector<const std::string*>> pending(S);
+    std::vector<std::vector<std::string_view>> pending(S);
     const int64_t kProv = 1LL << 62;  // provisional marker
     for (int t = 0; t < S; ++t)
       mthreads.emplace_back([&, t]() {
@@ -2185,20 +2227,18 @@ py::tuple parse_ntriples_file_annex(const std::string& path,
         auto& pend = pending[t];
         for (int i = 0; i < nt; ++i) {
           auto& views = outs[i].views;
-          auto& hs = outs[i].hashes;
           auto& rm = remaps[i];
-          for (size_t k = 0; k < views.size(); ++k) {
-            if (static_cast<int>(hs[k] % S) != t) continue;
+          for (uint32_t k : outs[i].by_shard[t]) {
             auto it = sh.map.find(views[k]);
             if (it != sh.map.end()) {
               rm[k] = it->second;   // may itself be provisional (this call)
               continue;
             }
-            sh.arena.emplace_back(views[k]);
+            std::string_view sv = sh.arena.add(views[k]);
             int64_t prov = kProv | (static_cast<int64_t>(t) << 40)
                            | static_cast<int64_t>(pend.size());
-            sh.map.emplace(std::string_view(sh.arena.back()), prov);
-            pend.push_back(&sh.arena.back());
+            sh.map.emplace(sv, prov);
+            pend.push_back(sv);
             rm[k] = prov;
           }
         }
@@ -2214,7 +2254,7 @@ py::tuple parse_ntriples_file_annex(const std::string& path,
     int64_t new_total = old_n + shard_base[S];
     TORCH_CHECK(voc.base + new_total <= max_id,
                 "dictionary ID space exhausted");
-    voc.by_id.resize(new_total, nullptr);
+    voc.by_id.resize(new_total);
     voc.values.resize(new_total, 0.0);
     std::vector<std::thread> vthreads;
     for (int t = 0; t < S; ++t)
@@ -2222,12 +2262,12 @@ py::tuple parse_ntriples_file_annex(const std::string& path,
         auto& sh = voc.shards[t];
         int64_t base = voc.base + old_n + shard_base[t];
         for (size_t j = 0; j < pending[t].size(); ++j) {
-          const std::string* sp = pending[t][j];
+          std::string_view sv = pending[t][j];
           int64_t id = base + static_cast<int64_t>(j);
-          sh.map[std::string_view(*sp)] = id;
+          sh.map[sv] = id;
           int64_t k = id - voc.base;
-          voc.by_id[k] = sp;
-          voc.values[k] = parse_value_full(*sp);
+          voc.by_id[k] = sv;
+          voc.values[k] = parse_value_full(sv);
         }
       });
     for (auto& t : vthreads) t.join();
@@ -2265,6 +2305,7 @@ py::tuple parse_ntriples_file_annex(const std::string& path,
   py::list pyfallback;
   for (int i = 0; i < nt; ++i)
     for (auto f : outs[i].fallback) pyfallback.append(f + line_off[i]);
+  std::thread([sc = scratch.release()]() { delete sc; }).detach();
   return py::make_tuple(ids_t, pyfallback);
 }
 

@@ -280,8 +280,29 @@ class SparqlDatabase:
             path, 0, h, QUOTED_TRIPLE_ID_BIT)
         d._values_dirty = True
         if ids.numel():
-            arr = ids.numpy()
-            self.store.insert_bulk(0, arr[:, 0], arr[:, 1], arr[:, 2])
+            # one contiguous H2D of the whole [n,3] block, then device-side
+            # column extraction (strided numpy slices would each force a
+            # host-side contiguous copy before upload)
+            import os as _os
+            import time as _time
+            dbg = _os.environ.get("KOLIBRIE_PARSE_DEBUG")
+            if dbg:
+                _t0 = _time.perf_counter()
+            t = ids.to(self.store.device)
+            if dbg:
+                if str(self.store.device).startswith("cuda"):
+                    torch.cuda.synchronize()
+                print(f"[ingest] H2D: {_time.perf_counter()-_t0:.3f}s",
+                      flush=True)
+                _t0 = _time.perf_counter()
+            self.store.insert_bulk(0, t[:, 0].contiguous(),
+                                   t[:, 1].contiguous(),
+                                   t[:, 2].contiguous())
+            if dbg:
+                if str(self.store.device).startswith("cuda"):
+                    torch.cuda.synchronize()
+                print(f"[ingest] insert_bulk: "
+                      f"{_time.perf_counter()-_t0:.3f}s", flush=True)
         if fallback:
             from ..parsing.rdf_formats import _parse_ntriples_lines
             with open(path, "r", encoding="utf-8") as f:

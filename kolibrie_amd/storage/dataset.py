@@ -105,7 +105,15 @@ class GraphIndex:
                 continue
             a, b, c = _ORDER_COLS[code]
             k = pack2(cols[a], cols[b])
-            perm = lexsort([k, cols[c]])
+            if code == SPO:
+                # SPO's trailing o is binary-searched (full-triple lookups,
+                # seminaive contains-probe) — needs the full lexsort
+                perm = lexsort([k, cols[c]])
+            else:
+                # POS/OSP/PSO trailing columns are only ever SCANNED after
+                # a key12 range lookup, never searched — one radix argsort
+                # on the packed key instead of two stable passes
+                perm = torch.argsort(k)
             orders[code] = (k[perm], cols[c][perm])
         _ = start
         return GraphIndex(device, n, orders)
@@ -289,7 +297,12 @@ class QuadStore:
         b = self._buf(g)
         self._commit(g)
         new_idx = GraphIndex.from_columns(s, p, o, device=self.device)
-        b.index = GraphIndex.merge([b.index, new_idx], self.device)
+        if b.index.n == 0:
+            # from_columns already sorted + dedup'd — merging with an empty
+            # index would rebuild all four orders a second time
+            b.index = new_idx
+        else:
+            b.index = GraphIndex.merge([b.index, new_idx], self.device)
         self.version += 1
 
     def create_graph(self, g: int):
