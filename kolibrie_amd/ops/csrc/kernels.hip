@@ -1578,6 +1578,10 @@ py::tuple parse_nlines_host(const std::string& text, bool quads) {
 // threads remap their id columns in parallel.  The GIL is released for
 // the whole native phase.
 #include <deque>
+#include <fcntl.h>
+#include <sys/mman.h>
+#include <sys/stat.h>
+#include <unistd.h>
 
 constexpr int kVocabShards = 32;
 
@@ -2157,26 +2161,46 @@ py::tuple parse_ntriples_file_annex(const std::string& path,
   // reaper thread after return — its node-by-node teardown takes seconds
   // and would otherwise serialize at scope exit, after all useful work
   struct AnnexScratch {
-    std::string text;
+    const char* map_ptr = nullptr;   // mmap'd input (preferred: no copy)
+    size_t map_len = 0;
+    std::string text;                // fallback buffer if mmap fails
     std::vector<ParseChunkOut> outs;
     std::vector<std::vector<int64_t>> remaps;
+    ~AnnexScratch() {
+      if (map_ptr)
+        munmap(const_cast<char*>(map_ptr), map_len);
+    }
   };
   auto scratch = std::make_unique<AnnexScratch>();
-  std::string& text = scratch->text;
   {
     py::gil_scoped_release release;
-    FILE* f = fopen(path.c_str(), "rb");
-    TORCH_CHECK(f != nullptr, "cannot open ", path);
-    fseek(f, 0, SEEK_END);
-    long sz = ftell(f);
-    fseek(f, 0, SEEK_SET);
-    text.resize(static_cast<size_t>(sz));
-    size_t rd = fread(text.data(), 1, static_cast<size_t>(sz), f);
-    fclose(f);
-    TORCH_CHECK(rd == static_cast<size_t>(sz), "short read of ", path);
+    int fd = open(path.c_str(), O_RDONLY);
+    TORCH_CHECK(fd >= 0, "cannot open ", path);
+    struct stat st {};
+    TORCH_CHECK(fstat(fd, &st) == 0, "cannot stat ", path);
+    size_t sz = static_cast<size_t>(st.st_size);
+    void* m = sz ? mmap(nullptr, sz, PROT_READ, MAP_PRIVATE, fd, 0)
+                 : nullptr;
+    if (m != nullptr && m != MAP_FAILED) {
+      madvise(m, sz, MADV_WILLNEED);
+      scratch->map_ptr = static_cast<const char*>(m);
+      scratch->map_len = sz;
+    } else if (sz) {
+      scratch->text.resize(sz);
+      size_t off = 0;
+      while (off < sz) {
+        ssize_t rd = read(fd, scratch->text.data() + off, sz - off);
+        TORCH_CHECK(rd > 0, "short read of ", path);
+        off += static_cast<size_t>(rd);
+      }
+    }
+    close(fd);
   }
+  const char* data = scratch->map_ptr ? scratch->map_ptr
+                                      : scratch->text.data();
   lap("read");
-  const size_t n = text.size();
+  const size_t n = scratch->map_ptr ? scratch->map_len
+                                    : scratch->text.size();
   int nt = static_cast<int>(n_threads);
   if (nt <= 0) {
     nt = static_cast<int>(std::thread::hardware_concurrency());
@@ -2188,8 +2212,8 @@ py::tuple parse_ntriples_file_annex(const std::string& path,
   for (int i = 1; i < nt; ++i) {
     size_t p = n * static_cast<size_t>(i) / nt;
     const char* nl = static_cast<const char*>(
-        memchr(text.data() + p, '\n', n - p));
-    starts.push_back(nl ? static_cast<size_t>(nl - text.data()) + 1 : n);
+        memchr(data + p, '\n', n - p));
+    starts.push_back(nl ? static_cast<size_t>(nl - data) + 1 : n);
   }
   starts.push_back(n);
   scratch->outs.resize(nt);
@@ -2203,7 +2227,7 @@ py::tuple parse_ntriples_file_annex(const std::string& path,
     py::gil_scoped_release release;
     std::vector<std::thread> threads;
     for (int i = 0; i < nt; ++i)
-      threads.emplace_back(parse_chunk_nt, text.data(), starts[i],
+      threads.emplace_back(parse_chunk_nt, data, starts[i],
                            starts[i + 1], false, std::ref(outs[i]));
     for (auto& t : threads) t.join();
     if (dbg) { py::gil_scoped_acquire a; lap("parallel parse"); }
