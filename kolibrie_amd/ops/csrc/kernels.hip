@@ -1585,11 +1585,77 @@ py::tuple parse_nlines_host(const std::string& text, bool quads) {
 
 constexpr int kVocabShards = 32;
 
+// open-addressing string_view -> id map (linear probing, stored 64-bit
+// hash, power-of-two).  A node-based unordered_map pays one malloc plus
+// a pointer chase per entry — at the 100M-triple bulk-load's ~43M unique
+// terms (and ~130M chunk-local interns) that allocator traffic dominated
+// the parse and merge phases.
+struct FlatMap {
+  struct Slot {
+    const char* ptr = nullptr;   // nullptr => empty
+    uint64_t hash = 0;
+    int64_t id = 0;
+    uint32_t len = 0;
+  };
+  std::vector<Slot> slots;
+  size_t mask;
+  size_t count = 0;
+  FlatMap() : slots(1024), mask(1023) {}
+  void grow() {
+    std::vector<Slot> old = std::move(slots);
+    size_t nsz = (mask + 1) * 2;
+    slots.assign(nsz, Slot{});
+    mask = nsz - 1;
+    for (auto& s0 : old)
+      if (s0.ptr) {
+        size_t i = s0.hash & mask;
+        while (slots[i].ptr) i = (i + 1) & mask;
+        slots[i] = s0;
+      }
+  }
+  // returns the slot index for key (h, sv); found=false means the slot
+  // was claimed fresh — caller must fill ptr (e.g. repoint to an arena
+  // copy) and id.  Grows BEFORE probing, so a returned index is valid
+  // until the next find_or_insert.
+  size_t find_or_insert(uint64_t h, std::string_view sv, bool& found) {
+    if ((count + 1) * 10 >= (mask + 1) * 7) grow();
+    size_t i = h & mask;
+    for (;;) {
+      Slot& sl = slots[i];
+      if (sl.ptr == nullptr) {
+        sl.ptr = sv.data();
+        sl.hash = h;
+        sl.len = static_cast<uint32_t>(sv.size());
+        ++count;
+        found = false;
+        return i;
+      }
+      if (sl.hash == h && sl.len == sv.size()
+          && memcmp(sl.ptr, sv.data(), sl.len) == 0) {
+        found = true;
+        return i;
+      }
+      i = (i + 1) & mask;
+    }
+  }
+  // SIZE_MAX if absent
+  size_t find(uint64_t h, std::string_view sv) const {
+    size_t i = h & mask;
+    for (;;) {
+      const Slot& sl = slots[i];
+      if (sl.ptr == nullptr) return SIZE_MAX;
+      if (sl.hash == h && sl.len == sv.size()
+          && memcmp(sl.ptr, sv.data(), sl.len) == 0) return i;
+      i = (i + 1) & mask;
+    }
+  }
+};
+
 struct ParseChunkOut {
   // term views point into the shared input text (zero-copy for the
   // overwhelmingly common escape-free terms) or into `owned` (deque:
   // stable addresses) for unescaped materializations
-  std::unordered_map<std::string_view, int64_t> interned;
+  FlatMap interned;
   std::vector<std::string_view> views;
   std::vector<uint64_t> hashes;    // std::hash of each view, precomputed
   // view indices bucketed by hash % kVocabShards at parse time, so each
@@ -1606,12 +1672,13 @@ static void parse_chunk_nt(const char* data, size_t begin, size_t end,
                            bool quads, ParseChunkOut& out) {
   std::hash<std::string_view> sv_hasher;
   auto intern_sv = [&](std::string_view sv) -> int64_t {
-    auto it = out.interned.find(sv);
-    if (it != out.interned.end()) return it->second;
-    int64_t id = static_cast<int64_t>(out.views.size());
-    out.interned.emplace(sv, id);
-    out.views.push_back(sv);
     uint64_t hh = sv_hasher(sv);
+    bool found;
+    size_t si = out.interned.find_or_insert(hh, sv, found);
+    if (found) return out.interned.slots[si].id;
+    int64_t id = static_cast<int64_t>(out.views.size());
+    out.interned.slots[si].id = id;
+    out.views.push_back(sv);
     out.hashes.push_back(hh);
     out.by_shard[hh % kVocabShards].push_back(static_cast<uint32_t>(id));
     return id;
@@ -1620,14 +1687,17 @@ static void parse_chunk_nt(const char* data, size_t begin, size_t end,
     if (memchr(ptr, '\\', len) == nullptr)
       return intern_sv(std::string_view(ptr, len));
     std::string u = unescape_nt(ptr, len);
-    auto it = out.interned.find(std::string_view(u));
-    if (it != out.interned.end()) return it->second;
+    uint64_t hh = sv_hasher(std::string_view(u));
+    bool found;
+    size_t si = out.interned.find_or_insert(hh, std::string_view(u), found);
+    if (found) return out.interned.slots[si].id;
     out.owned.push_back(std::move(u));
     std::string_view sv(out.owned.back());
+    // the probe keyed on the TEMPORARY string — repoint to the stable copy
+    out.interned.slots[si].ptr = sv.data();
     int64_t id = static_cast<int64_t>(out.views.size());
-    out.interned.emplace(sv, id);
+    out.interned.slots[si].id = id;
     out.views.push_back(sv);
-    uint64_t hh = sv_hasher(sv);
     out.hashes.push_back(hh);
     out.by_shard[hh % kVocabShards].push_back(static_cast<uint32_t>(id));
     return id;
@@ -2012,7 +2082,7 @@ struct BumpArena {
 };
 
 struct VocabShard {
-  std::unordered_map<std::string_view, int64_t> map;
+  FlatMap map;
   BumpArena arena;
 };
 
@@ -2058,20 +2128,24 @@ int64_t vocab_len(int64_t h) {
 int64_t vocab_lookup(int64_t h, const std::string& s) {
   Vocab& v = *g_vocabs.at(h);
   std::hash<std::string_view> hasher;
-  auto& sh = v.shards[hasher(std::string_view(s)) % v.nshards];
-  auto it = sh.map.find(std::string_view(s));
-  return it == sh.map.end() ? -1 : it->second;
+  uint64_t hh = hasher(std::string_view(s));
+  auto& sh = v.shards[hh % v.nshards];
+  size_t si = sh.map.find(hh, std::string_view(s));
+  return si == SIZE_MAX ? -1 : sh.map.slots[si].id;
 }
 
 int64_t vocab_insert(int64_t h, const std::string& s) {
   Vocab& v = *g_vocabs.at(h);
   std::hash<std::string_view> hasher;
-  auto& sh = v.shards[hasher(std::string_view(s)) % v.nshards];
-  auto it = sh.map.find(std::string_view(s));
-  if (it != sh.map.end()) return it->second;
+  uint64_t hh = hasher(std::string_view(s));
+  auto& sh = v.shards[hh % v.nshards];
+  bool found;
+  size_t si = sh.map.find_or_insert(hh, std::string_view(s), found);
+  if (found) return sh.map.slots[si].id;
   std::string_view sv = sh.arena.add(std::string_view(s));
+  sh.map.slots[si].ptr = sv.data();  // repoint off the caller's buffer
   int64_t id = v.base + static_cast<int64_t>(v.by_id.size());
-  sh.map.emplace(sv, id);
+  sh.map.slots[si].id = id;
   v.by_id.push_back(sv);
   v.values.push_back(parse_value_full(sv));
   return id;
@@ -2252,16 +2326,19 @@ py::tuple parse_ntriples_file_annex(const std::string& path,
         for (int i = 0; i < nt; ++i) {
           auto& views = outs[i].views;
           auto& rm = remaps[i];
+          auto& hs = outs[i].hashes;
           for (uint32_t k : outs[i].by_shard[t]) {
-            auto it = sh.map.find(views[k]);
-            if (it != sh.map.end()) {
-              rm[k] = it->second;   // may itself be provisional (this call)
+            bool found;
+            size_t si = sh.map.find_or_insert(hs[k], views[k], found);
+            if (found) {
+              rm[k] = sh.map.slots[si].id;  // may itself be provisional
               continue;
             }
             std::string_view sv = sh.arena.add(views[k]);
+            sh.map.slots[si].ptr = sv.data();  // text/owned die with scratch
             int64_t prov = kProv | (static_cast<int64_t>(t) << 40)
                            | static_cast<int64_t>(pend.size());
-            sh.map.emplace(sv, prov);
+            sh.map.slots[si].id = prov;
             pend.push_back(sv);
             rm[k] = prov;
           }
@@ -2285,10 +2362,12 @@ py::tuple parse_ntriples_file_annex(const std::string& path,
       vthreads.emplace_back([&, t]() {
         auto& sh = voc.shards[t];
         int64_t base = voc.base + old_n + shard_base[t];
+        std::hash<std::string_view> hasher;
         for (size_t j = 0; j < pending[t].size(); ++j) {
           std::string_view sv = pending[t][j];
           int64_t id = base + static_cast<int64_t>(j);
-          sh.map[sv] = id;
+          size_t si = sh.map.find(hasher(sv), sv);
+          sh.map.slots[si].id = id;
           int64_t k = id - voc.base;
           voc.by_id[k] = sv;
           voc.values[k] = parse_value_full(sv);
