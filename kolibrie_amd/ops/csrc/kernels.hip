@@ -626,6 +626,156 @@ __device__ __forceinline__ uint32_t h32(uint32_t x) {
   return x;
 }
 
+// ---- K8: database statistics gather ---------------------------------------
+// One pass over the committed (s, p, o) columns computes everything the
+// cost model needs: per-predicate row counts, global distinct
+// subjects/objects, and per-predicate distinct subjects/objects (as
+// (p,x) pair claims).  Presence tables are open-addressing claims
+// (atomicCAS); row counts aggregate through a per-block LDS histogram
+// (predicate sets are tiny); distinct tallies use per-thread
+// accumulators folded by one wave reduction (guide G12).
+// Ref parity: streamertail_optimizer/stats/database_stats.rs:18-158.
+constexpr int kStatsPreds = 1024;  // open-addressing predicate slots
+constexpr uint32_t kStatsEmpty = 0xFFFFFFFFu;
+
+__device__ __forceinline__ int stats_pred_slot(uint32_t* pred_keys,
+                                               uint32_t pkey) {
+  uint32_t slot = h32(pkey) & (kStatsPreds - 1);
+  for (;;) {
+    uint32_t cur = __hip_atomic_load(&pred_keys[slot], __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_AGENT);
+    if (cur == pkey) return static_cast<int>(slot);
+    if (cur == kStatsEmpty) {
+      uint32_t prev = atomicCAS(&pred_keys[slot], kStatsEmpty, pkey);
+      if (prev == kStatsEmpty || prev == pkey)
+        return static_cast<int>(slot);
+      continue;  // lost the race to a different key: re-inspect the slot
+    }
+    slot = (slot + 1) & (kStatsPreds - 1);
+  }
+}
+
+__device__ __forceinline__ bool stats_claim_u32(uint32_t* tbl, uint32_t mask,
+                                                uint32_t v) {
+  uint32_t slot = h32(v) & mask;
+  for (;;) {
+    uint32_t cur = __hip_atomic_load(&tbl[slot], __ATOMIC_RELAXED,
+                                     __HIP_MEMORY_SCOPE_AGENT);
+    if (cur == v) return false;
+    if (cur == kStatsEmpty) {
+      uint32_t prev = atomicCAS(&tbl[slot], kStatsEmpty, v);
+      if (prev == kStatsEmpty) return true;
+      if (prev == v) return false;
+      continue;
+    }
+    slot = (slot + 1) & mask;
+  }
+}
+
+__device__ __forceinline__ bool stats_claim_u64(unsigned long long* tbl,
+                                                uint32_t mask,
+                                                unsigned long long v) {
+  uint32_t slot = h32(static_cast<uint32_t>(v)
+                      ^ static_cast<uint32_t>(v >> 32)) & mask;
+  for (;;) {
+    unsigned long long cur = __hip_atomic_load(
+        &tbl[slot], __ATOMIC_RELAXED, __HIP_MEMORY_SCOPE_AGENT);
+    if (cur == v) return false;
+    if (cur == ~0ull) {
+      unsigned long long prev = atomicCAS(&tbl[slot], ~0ull, v);
+      if (prev == ~0ull) return true;
+      if (prev == v) return false;
+      continue;
+    }
+    slot = (slot + 1) & mask;
+  }
+}
+
+__global__ void stats_gather_kernel(
+    const int32_t* __restrict__ s, const int32_t* __restrict__ p,
+    const int32_t* __restrict__ o, int64_t n, uint32_t* pred_keys,
+    unsigned long long* pred_rows, unsigned long long* pred_ds,
+    unsigned long long* pred_do, uint32_t* s_tbl, uint32_t* o_tbl,
+    uint32_t so_mask, unsigned long long* ps_tbl,
+    unsigned long long* po_tbl, uint32_t pair_mask,
+    unsigned long long* g_counts /* [0]=distinct_s [1]=distinct_o */) {
+  __shared__ uint32_t lds_rows[kStatsPreds];
+  for (int i = threadIdx.x; i < kStatsPreds; i += blockDim.x)
+    lds_rows[i] = 0;
+  __syncthreads();
+  unsigned long long acc_s = 0, acc_o = 0;
+  for (int64_t i = blockIdx.x * (int64_t)blockDim.x + threadIdx.x; i < n;
+       i += (int64_t)gridDim.x * blockDim.x) {
+    uint32_t pp = static_cast<uint32_t>(p[i]);
+    uint32_t ss = static_cast<uint32_t>(s[i]);
+    uint32_t oo = static_cast<uint32_t>(o[i]);
+    int slot = stats_pred_slot(pred_keys, pp);
+    atomicAdd(&lds_rows[slot], 1u);
+    if (stats_claim_u32(s_tbl, so_mask, ss)) ++acc_s;
+    if (stats_claim_u32(o_tbl, so_mask, oo)) ++acc_o;
+    unsigned long long base = static_cast<unsigned long long>(pp) << 32;
+    if (stats_claim_u64(ps_tbl, pair_mask, base | ss))
+      atomicAdd(&pred_ds[slot], 1ull);
+    if (stats_claim_u64(po_tbl, pair_mask, base | oo))
+      atomicAdd(&pred_do[slot], 1ull);
+  }
+  for (int off = 32; off > 0; off >>= 1) {
+    acc_s += __shfl_down(acc_s, off);
+    acc_o += __shfl_down(acc_o, off);
+  }
+  if ((threadIdx.x & 63) == 0) {
+    if (acc_s) atomicAdd(&g_counts[0], acc_s);
+    if (acc_o) atomicAdd(&g_counts[1], acc_o);
+  }
+  __syncthreads();
+  for (int i = threadIdx.x; i < kStatsPreds; i += blockDim.x)
+    if (lds_rows[i])
+      atomicAdd(&pred_rows[i], static_cast<unsigned long long>(lds_rows[i]));
+}
+
+std::vector<at::Tensor> stats_gather(at::Tensor s, at::Tensor p,
+                                     at::Tensor o) {
+  TORCH_CHECK(s.is_cuda() && p.is_cuda() && o.is_cuda(),
+              "stats_gather: device columns required");
+  TORCH_CHECK(s.dtype() == at::kInt && p.dtype() == at::kInt
+              && o.dtype() == at::kInt);
+  int64_t n = s.numel();
+  TORCH_CHECK(p.numel() == n && o.numel() == n);
+  int64_t so_size = 64;
+  while (so_size < 2 * n) so_size <<= 1;
+  auto opt_i = s.options();
+  auto opt_l = s.options().dtype(at::kLong);
+  auto pred_keys = at::full({kStatsPreds}, -1, opt_i);
+  auto pred_rows = at::zeros({kStatsPreds}, opt_l);
+  auto pred_ds = at::zeros({kStatsPreds}, opt_l);
+  auto pred_do = at::zeros({kStatsPreds}, opt_l);
+  auto s_tbl = at::full({so_size}, -1, opt_i);
+  auto o_tbl = at::full({so_size}, -1, opt_i);
+  auto ps_tbl = at::full({so_size}, -1, opt_l);
+  auto po_tbl = at::full({so_size}, -1, opt_l);
+  auto g_counts = at::zeros({2}, opt_l);
+  if (n > 0) {
+    hipLaunchKernelGGL(
+        stats_gather_kernel, dim3(grid_for(n)), dim3(kBlock), 0,
+        cur_stream(), s.data_ptr<int32_t>(), p.data_ptr<int32_t>(),
+        o.data_ptr<int32_t>(),
+        n, reinterpret_cast<uint32_t*>(pred_keys.data_ptr<int32_t>()),
+        reinterpret_cast<unsigned long long*>(pred_rows.data_ptr<int64_t>()),
+        reinterpret_cast<unsigned long long*>(pred_ds.data_ptr<int64_t>()),
+        reinterpret_cast<unsigned long long*>(pred_do.data_ptr<int64_t>()),
+        reinterpret_cast<uint32_t*>(s_tbl.data_ptr<int32_t>()),
+        reinterpret_cast<uint32_t*>(o_tbl.data_ptr<int32_t>()),
+        static_cast<uint32_t>(so_size - 1),
+        reinterpret_cast<unsigned long long*>(ps_tbl.data_ptr<int64_t>()),
+        reinterpret_cast<unsigned long long*>(po_tbl.data_ptr<int64_t>()),
+        static_cast<uint32_t>(so_size - 1),
+        reinterpret_cast<unsigned long long*>(g_counts.data_ptr<int64_t>()));
+    HIP_OK(hipGetLastError());
+  }
+  return {pred_keys, pred_rows, pred_ds, pred_do, g_counts};
+}
+
+
 __global__ void count_table_insert(const int64_t* __restrict__ entries,
                                    int64_t u,
                                    unsigned long long* __restrict__ table,
@@ -3228,6 +3378,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("vocab_value", &vocab_value);
   m.def("vocab_values", &vocab_values);
   m.def("vocab_export_strings", &vocab_export_strings);
+  m.def("stats_gather", &stats_gather, "K8 one-pass database statistics");
   m.def("parse_ntriples_file_annex", &parse_ntriples_file_annex,
         "file -> parallel parse -> intern into the native vocab annex");
   m.doc() = "kolibrie_amd native CDNA4 kernels (gfx950)";

@@ -48,25 +48,110 @@ class DatabaseStats:
             o_all.append(o)
         if not s_all:
             return st
+        st.quoted_count = len(db.quoted_triples)
+        live = [buf.index for buf in store.graphs.values() if buf.index.n]
+        if len(live) == 1:
+            # common case: every statistic is an order statistic of the
+            # sorted index permutations the store ALREADY maintains —
+            # transition counts + segment cumsums, pure bandwidth
+            st.total = live[0].n
+            st._gather_from_index(live[0])
+            return st
         s = torch.cat(s_all)
         p = torch.cat(p_all)
         o = torch.cat(o_all)
         st.total = int(s.numel())
-        st.quoted_count = len(db.quoted_triples)
+        if s.is_cuda:
+            from ..ops import _native
+            if _native is None:
+                raise RuntimeError(
+                    "native extension required for device stats gather")
+            st._gather_native(_native, s, p, o)
+            return st
+        st._gather_torch(s, p, o)
+        return st
+
+    def _gather_from_index(self, idx):
+        """Stats from the sorted orders (single-graph path): distinct
+        subjects/objects = leading-word transition counts of SPO/OSP;
+        per-predicate row counts and distinct subjects/objects = segment
+        boundaries + full-key transition cumsums over PSO/POS.  Measured
+        at 100M rows: ~2 ms vs 50 ms torch-unique composite vs 111 ms
+        one-pass hash-claim kernel (stats_gather, kept for the
+        multi-graph union case)."""
+        from ..storage.dataset import OSP, POS, PSO, SPO
+
+        def transitions(k):
+            return 1 + int((k[1:] != k[:-1]).sum().item())
+
+        k_spo, _ = idx.orders[SPO]
+        self.distinct_subjects = max(1, transitions(k_spo >> 32))
+        k_osp, _ = idx.orders[OSP]
+        self.distinct_objects = max(1, transitions(k_osp >> 32))
+        n = idx.n
+        for code, out in ((PSO, self.pred_distinct_subj),
+                          (POS, self.pred_distinct_obj)):
+            key, _ = idx.orders[code]
+            hi = key >> 32
+            change = torch.nonzero(hi[1:] != hi[:-1]).flatten() + 1
+            z = torch.zeros(1, dtype=change.dtype, device=change.device)
+            starts = torch.cat([z, change])
+            ends = torch.cat([change, torch.full((1,), n, dtype=change.dtype,
+                                                 device=change.device)])
+            preds = (hi[starts] & 0xFFFFFFFF).tolist()
+            # distinct full keys per segment: t[i]=1 iff key starts a new
+            # run; segment starts always do (p changed), so the count in
+            # [a,b) is C[b-1] - C[a-1] with C the inclusive cumsum
+            t = torch.ones(n, dtype=torch.int64, device=key.device)
+            t[1:] = (key[1:] != key[:-1]).to(torch.int64)
+            c = torch.cumsum(t, 0)
+            hi_c = c[ends - 1]
+            lo_c = torch.where(starts > 0, c[(starts - 1).clamp(min=0)],
+                               torch.zeros_like(hi_c))
+            seg = (hi_c - lo_c).tolist()
+            cnts = (ends - starts).tolist()
+            if code == PSO:
+                for pid, rows in zip(preds, cnts):
+                    self.pred_count[pid] = rows
+            for pid, d in zip(preds, seg):
+                out[pid] = d
+
+    def _gather_native(self, _native, s, p, o):
+        """K8 kernel: one pass over the columns computes the predicate
+        histogram, the global distinct subject/object counts and the
+        per-predicate distinct subjects/objects (ops/csrc/kernels.hip
+        stats_gather_kernel)."""
+        pred_keys, pred_rows, pred_ds, pred_do, gc = _native.stats_gather(
+            s.contiguous(), p.contiguous(), o.contiguous())
+        live = (pred_keys != -1).nonzero(as_tuple=True)[0]
+        keys = pred_keys[live].cpu().tolist()
+        rows = pred_rows[live].cpu().tolist()
+        ds = pred_ds[live].cpu().tolist()
+        do_ = pred_do[live].cpu().tolist()
+        for pid, cnt, a, b in zip(keys, rows, ds, do_):
+            pid &= 0xFFFFFFFF
+            self.pred_count[pid] = cnt
+            self.pred_distinct_subj[pid] = a
+            self.pred_distinct_obj[pid] = b
+        g = gc.cpu().tolist()
+        self.distinct_subjects = max(1, int(g[0]))
+        self.distinct_objects = max(1, int(g[1]))
+
+    def _gather_torch(self, s, p, o):
         # predicate histogram
         pv, pc = torch.unique(p, return_counts=True)
         for pid, cnt in zip(pv.tolist(), pc.tolist()):
-            st.pred_count[pid & 0xFFFFFFFF] = cnt
-        st.distinct_subjects = max(1, int(torch.unique(s).numel()))
-        st.distinct_objects = max(1, int(torch.unique(o).numel()))
+            self.pred_count[pid & 0xFFFFFFFF] = cnt
+        self.distinct_subjects = max(1, int(torch.unique(s).numel()))
+        self.distinct_objects = max(1, int(torch.unique(o).numel()))
         # per-predicate distinct subj/obj via (p,x) pair dedup then histogram
         from ..engine.tensor_utils import unique_rows
-        for col, out in ((s, st.pred_distinct_subj), (o, st.pred_distinct_obj)):
+        for col, out in ((s, self.pred_distinct_subj),
+                         (o, self.pred_distinct_obj)):
             up = unique_rows([p, col])[0]
             pv2, pc2 = torch.unique(up, return_counts=True)
             for pid, cnt in zip(pv2.tolist(), pc2.tolist()):
                 out[pid & 0xFFFFFFFF] = cnt
-        return st
 
     # ---------------------------------------------------------- estimation --
     def estimate_pattern(self, pattern, graph=None) -> float:

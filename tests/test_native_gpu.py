@@ -641,3 +641,47 @@ def test_hop_table_threshold_crossover():
             counts[dev] = db.query(q)
         assert counts["cpu"] == counts["cuda:0"], n_obj
     assert cap >= 1000  # sanity: the knob still exists
+
+
+@pytest.mark.gpu
+def test_k8_stats_gather_matches_torch():
+    """K8 one-pass stats kernel vs the torch unique/segment oracle."""
+    import torch
+    from kolibrie_amd.plan.stats import DatabaseStats
+
+    torch.manual_seed(5)
+    n = 2_000_000
+    s = torch.randint(0, 50_000, (n,), dtype=torch.int32, device="cuda")
+    p = torch.randint(0, 37, (n,), dtype=torch.int32, device="cuda")
+    o = torch.randint(0, 400_000, (n,), dtype=torch.int32, device="cuda")
+
+    from kolibrie_amd.ops import _native
+    assert _native is not None
+    native = DatabaseStats()
+    native._gather_native(_native, s, p, o)
+    oracle = DatabaseStats()
+    oracle._gather_torch(s.cpu(), p.cpu(), o.cpu())
+
+    assert native.pred_count == oracle.pred_count
+    assert native.pred_distinct_subj == oracle.pred_distinct_subj
+    assert native.pred_distinct_obj == oracle.pred_distinct_obj
+    assert native.distinct_subjects == oracle.distinct_subjects
+    assert native.distinct_objects == oracle.distinct_objects
+
+
+@pytest.mark.gpu
+def test_k8_stats_via_database_gather():
+    """DatabaseStats.gather on a cuda database runs the kernel path."""
+    import torch
+    from kolibrie_amd import SparqlDatabase
+    from kolibrie_amd.plan.stats import DatabaseStats
+
+    db = SparqlDatabase(device="cuda:0")
+    db.parse_ntriples("\n".join(
+        f"<http://s{i % 7}> <http://p{i % 3}> <http://o{i % 5}> ."
+        for i in range(200)))
+    st = DatabaseStats.gather(db)
+    assert st.total == len({(i % 7, i % 3, i % 5) for i in range(200)})
+    assert len(st.pred_count) == 3
+    assert st.distinct_subjects == 7
+    assert st.distinct_objects == 5

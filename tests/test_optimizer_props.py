@@ -124,3 +124,30 @@ def test_join_shapes_agree_on_results():
     os.environ.pop("KOLIBRIE_JOIN_MODE", None)
     assert results["auto"] == results["hash"] == results["bind"]
     assert len(results["auto"]) == 30
+
+
+def test_stats_index_path_matches_torch_oracle():
+    """Single-graph stats come from sorted-order transition counts; they
+    must equal the unique/segment oracle (and the multi-graph path)."""
+    import torch
+    from kolibrie_amd import SparqlDatabase
+    from kolibrie_amd.plan.stats import DatabaseStats
+
+    torch.manual_seed(11)
+    db = SparqlDatabase(device="cpu")
+    n = 5000
+    s = torch.randint(0, 200, (n,), dtype=torch.int32)
+    p = torch.randint(0, 9, (n,), dtype=torch.int32)
+    o = torch.randint(0, 400, (n,), dtype=torch.int32)
+    db.store.insert_bulk(0, s, p, o)
+
+    st = DatabaseStats.gather(db)
+    # oracle over the deduplicated committed columns
+    cs, cp, co = db.store.graph_index(0).columns()
+    oracle = DatabaseStats()
+    oracle._gather_torch(cs, cp, co)
+    assert st.pred_count == oracle.pred_count
+    assert st.pred_distinct_subj == oracle.pred_distinct_subj
+    assert st.pred_distinct_obj == oracle.pred_distinct_obj
+    assert st.distinct_subjects == oracle.distinct_subjects
+    assert st.distinct_objects == oracle.distinct_objects

@@ -597,3 +597,37 @@ def test_columnar_istream_dstream_match_host_semantics():
             dev_out = sorted(zip(cols[0].tolist(), cols[1].tolist())) \
                 if cols else []
             assert sorted(set(host_out)) == dev_out, (mode, host_out, dev_out)
+
+
+def test_device_window_sorted_views_match_mask_path():
+    """K7 zero-copy range scoping must equal the mask fallback, and
+    out-of-order batches must flip to the fallback automatically."""
+    import torch
+    from kolibrie_amd.rsp.ring import DeviceStreamWindow
+
+    def run(ts_batches, force_mask=False):
+        w = DeviceStreamWindow(width=10, slide=5, device="cpu")
+        fired = []
+        w.register_callback(
+            lambda c: fired.append((c.open, c.close,
+                                    c.ts.tolist(), c.s.tolist())))
+        if force_mask:
+            w._sorted = False
+        n = 0
+        for ts in ts_batches:
+            k = len(ts)
+            s = torch.arange(n, n + k, dtype=torch.int32)
+            n += k
+            w.add_batch(s, s.clone(), s.clone(),
+                        torch.tensor(ts, dtype=torch.int64))
+        return fired, w._sorted
+
+    batches = [[1, 2, 4], [5, 7, 9, 11], [12, 14, 21]]
+    a, sorted_a = run(batches)
+    b, _ = run(batches, force_mask=True)
+    assert sorted_a is True
+    assert a == b
+    # out-of-order batch: falls back, same results as the mask path
+    ooo = [[1, 4, 2], [5, 11, 7], [21, 12]]
+    c, sorted_c = run(ooo)
+    assert sorted_c is False
