@@ -220,10 +220,13 @@ def main():
                 "result_count": c,
                 "p50_ms": p50,
                 "plan_cache": "cold" if args.cold else "warm",
-                "timed_region": ("parse+plan+execute per step" if args.cold
-                                 else "prepared-plan serving (plan-cache hit "
-                                      "+ one C++ serve call: direct kernel "
-                                      "launches + pinned 8-byte readback)"),
+                "timed_region": (
+                    "parse+plan+execute per step" if args.cold
+                    else ("prepared distributed plan per step (local "
+                          "kernels + COUNT all-reduce)" if world > 1
+                          else "prepared-plan serving (plan-cache hit "
+                               "+ one C++ serve call: direct kernel "
+                               "launches + pinned 8-byte readback)")),
                 "cold_ms_p50": cold_p50,
                 "rows_variant_ms_per_step": rows_ms,
                 "rows_variant_result_rows": rows_n,
