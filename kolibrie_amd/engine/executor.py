@@ -479,7 +479,8 @@ class ExecutionEngine:
             op._chain_serve = None
         try:
             k = len(hop_args[0])
-            n_tiles = (seed_b.numel() + 255) // 256
+            tile = getattr(native, "chain_tile", lambda: 256)()
+            n_tiles = (seed_b.numel() + tile - 1) // tile
             win = _t.empty(max(1, n_tiles * k * 2), dtype=_t.int64,
                            device=seed_b.device)
             total = _t.zeros(1, dtype=_t.int64, device=seed_b.device)

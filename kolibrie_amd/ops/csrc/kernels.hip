@@ -109,7 +109,7 @@ __global__ void probe_count_range(const int64_t* __restrict__ key12, int64_t n,
 // a fully-parallel boundary pass computes each 256-probe tile's index
 // window; the count pass then searches only inside its tile window, which
 // is small and cache-hot — the probe becomes a merge join.
-constexpr int kTile = 256;
+constexpr int kTile = 512;
 
 __global__ void tile_bounds(const int64_t* __restrict__ key12, int64_t n,
                             const int64_t* __restrict__ keys, int64_t m,
@@ -874,6 +874,12 @@ __global__ void chain_tile_bounds(const int32_t* __restrict__ seed_b,
 // probe structures).  16 KB leaves occupancy at 10 blocks/CU.
 constexpr int kChainLds = 4096;   // int32 rows: 16 KB, 10 blocks/CU
 
+// each thread owns kSub seeds of its tile (kTile = kSub * kBlock): the
+// independent probe chains per thread double the memory-level
+// parallelism, and the per-tile costs (window bounds, LDS staging
+// barriers) amortize over twice the seeds
+constexpr int kSub = kTile / kBlock;
+
 __global__ void chain_count_kernel(const int32_t* __restrict__ seed_b,
                                    const int32_t* __restrict__ seed_z,
                                    int64_t m, ChainHops hops,
@@ -884,58 +890,62 @@ __global__ void chain_count_kernel(const int32_t* __restrict__ seed_b,
   // block-uniform tile iteration (every thread of the block is in the same
   // tile) so the cooperative LDS loads can barrier safely
   for (int64_t t = blockIdx.x; t * kTile < m; t += gridDim.x) {
-    int64_t i = t * kTile + threadIdx.x;
-    bool active = i < m;
-    uint32_t b_comp = 0, z_comp = 0;
-    if (active) {
-      b_comp = static_cast<uint32_t>(seed_b[i]);
-      z_comp = static_cast<uint32_t>(seed_z[i]);
+    uint32_t b_c[kSub], z_c[kSub];
+    bool act[kSub];
+    unsigned long long prod[kSub];
+    for (int u = 0; u < kSub; ++u) {
+      int64_t i = t * kTile + u * kBlock + threadIdx.x;
+      act[u] = i < m;
+      b_c[u] = act[u] ? static_cast<uint32_t>(seed_b[i]) : 0;
+      z_c[u] = act[u] ? static_cast<uint32_t>(seed_z[i]) : 0;
+      prod[u] = 1;
     }
-    unsigned long long prod = 1;
     // no early exit: the hop searches are independent dependent-load
     // chains — letting them all issue gives the scheduler ILP to hide
     // L2 latency
     for (int h = 0; h < hops.k; ++h) {
-      uint32_t comp = hops.src[h] == 0 ? b_comp : z_comp;
       if (hops.table32[h] != nullptr) {
-        if (!active) continue;
-        // packed count-table hop: one 4-byte L2 load
-        uint32_t v = static_cast<uint32_t>(comp);
+        // packed count-table hop: one 4-byte L2 load per seed
         uint32_t mask = static_cast<uint32_t>(hops.tmask[h]);
-        uint32_t slot = h32(v) & mask;
-        unsigned long long cnt = 0;
-        for (;;) {
-          uint32_t e = hops.table32[h][slot];
-          if (e == kTbl32Empty) break;
-          if ((e >> 7) == v) {
-            cnt = e & 0x7Fu;
-            break;
+        for (int u = 0; u < kSub; ++u) {
+          if (!act[u]) continue;
+          uint32_t v = hops.src[h] == 0 ? b_c[u] : z_c[u];
+          uint32_t slot = h32(v) & mask;
+          unsigned long long cnt = 0;
+          for (;;) {
+            uint32_t e = hops.table32[h][slot];
+            if (e == kTbl32Empty) break;
+            if ((e >> 7) == v) {
+              cnt = e & 0x7Fu;
+              break;
+            }
+            slot = (slot + 1) & mask;
           }
-          slot = (slot + 1) & mask;
+          prod[u] *= cnt;
         }
-        prod *= cnt;
         continue;
       }
       if (hops.table[h] != nullptr) {
-        if (!active) continue;
         // count-table hop: one 8-byte L2 load (vs log2(n) lines)
-        uint32_t v = static_cast<uint32_t>(comp);
         uint32_t mask = static_cast<uint32_t>(hops.tmask[h]);
-        uint32_t slot = h32(v) & mask;
-        unsigned long long cnt = 0;
-        for (;;) {
-          unsigned long long e = hops.table[h][slot];
-          if (e == kTblEmpty) break;
-          if (static_cast<uint32_t>(e >> 32) == v) {
-            cnt = e & 0xFFFFFFFFull;
-            break;
+        for (int u = 0; u < kSub; ++u) {
+          if (!act[u]) continue;
+          uint32_t v = hops.src[h] == 0 ? b_c[u] : z_c[u];
+          uint32_t slot = h32(v) & mask;
+          unsigned long long cnt = 0;
+          for (;;) {
+            unsigned long long e = hops.table[h][slot];
+            if (e == kTblEmpty) break;
+            if (static_cast<uint32_t>(e >> 32) == v) {
+              cnt = e & 0xFFFFFFFFull;
+              break;
+            }
+            slot = (slot + 1) & mask;
           }
-          slot = (slot + 1) & mask;
+          prod[u] *= cnt;
         }
-        prod *= cnt;
         continue;
       }
-      uint32_t key = comp;
       int64_t wlo = win[(t * hops.k + h) * 2];
       int64_t wspan = win[(t * hops.k + h) * 2 + 1] - wlo;
       const int32_t* base = hops.key32[h] + wlo;
@@ -945,7 +955,9 @@ __global__ void chain_count_kernel(const int32_t* __restrict__ seed_b,
         for (int64_t j = threadIdx.x; j < wspan; j += blockDim.x)
           lds[j] = base[j];
         __syncthreads();
-        if (active) {
+        for (int u = 0; u < kSub; ++u) {
+          if (!act[u]) continue;
+          uint32_t key = hops.src[h] == 0 ? b_c[u] : z_c[u];
           int64_t lo = lower_bound_u32(lds, wspan, key);
           int64_t hi = lo;
           while (hi < wspan && hi - lo < 4
@@ -953,25 +965,29 @@ __global__ void chain_count_kernel(const int32_t* __restrict__ seed_b,
           if (hi - lo == 4 && hi < wspan
               && static_cast<uint32_t>(lds[hi]) == key)
             hi = lo + upper_bound_u32(lds + lo, wspan - lo, key);
-          prod *= static_cast<unsigned long long>(hi - lo);
+          prod[u] *= static_cast<unsigned long long>(hi - lo);
         }
         __syncthreads();  // before the next hop reuses the buffer
         continue;
       }
-      if (!active) continue;
-      int64_t lo = lower_bound_u32(base, wspan, key);
-      // match runs are tiny (one object per subject in typical star data):
-      // walk forward a few cache-hot slots instead of paying a second full
-      // log2(wspan) dependent-load chain; fall back for genuine skew
-      int64_t hi = lo;
-      while (hi < wspan && hi - lo < 4
-             && static_cast<uint32_t>(base[hi]) == key) ++hi;
-      if (hi - lo == 4 && hi < wspan
-          && static_cast<uint32_t>(base[hi]) == key)
-        hi = lo + upper_bound_u32(base + lo, wspan - lo, key);
-      prod *= static_cast<unsigned long long>(hi - lo);
+      for (int u = 0; u < kSub; ++u) {
+        if (!act[u]) continue;
+        uint32_t key = hops.src[h] == 0 ? b_c[u] : z_c[u];
+        int64_t lo = lower_bound_u32(base, wspan, key);
+        // match runs are tiny (one object per subject in typical star
+        // data): walk forward a few cache-hot slots instead of paying a
+        // second full log2(wspan) dependent-load chain
+        int64_t hi = lo;
+        while (hi < wspan && hi - lo < 4
+               && static_cast<uint32_t>(base[hi]) == key) ++hi;
+        if (hi - lo == 4 && hi < wspan
+            && static_cast<uint32_t>(base[hi]) == key)
+          hi = lo + upper_bound_u32(base + lo, wspan - lo, key);
+        prod[u] *= static_cast<unsigned long long>(hi - lo);
+      }
     }
-    if (active) acc += prod;
+    for (int u = 0; u < kSub; ++u)
+      if (act[u]) acc += prod[u];
   }
   // wave reduction then one device-scope atomic per wave (guide G12)
   for (int off = 32; off > 0; off >>= 1)
@@ -3388,6 +3404,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         "K1 scan-probe, leading-component range -> (li, b, z)");
   m.def("hash_join", &hash_join,
         "K2 chained hash join over int32 key columns -> (li, ri)");
+  m.def("chain_tile", []() { return static_cast<int64_t>(kTile); },
+        "seed rows per chain tile (win buffer sizing)");
   m.def("chain_count", &chain_count,
         "fused COUNT(*) over a seed scan + probe-hop chain");
   m.def("chain_count_into", &chain_count_into,
