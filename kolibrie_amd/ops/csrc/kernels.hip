@@ -109,7 +109,7 @@ __global__ void probe_count_range(const int64_t* __restrict__ key12, int64_t n,
 // a fully-parallel boundary pass computes each 256-probe tile's index
 // window; the count pass then searches only inside its tile window, which
 // is small and cache-hot — the probe becomes a merge join.
-constexpr int kTile = 512;
+constexpr int kTile = 768;
 
 __global__ void tile_bounds(const int64_t* __restrict__ key12, int64_t n,
                             const int64_t* __restrict__ keys, int64_t m,
