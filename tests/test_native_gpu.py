@@ -404,6 +404,18 @@ def test_chain_count_hipgraph_replay_stable():
     db.add_triple("<http://x/e>", "<http://x/p>", "<http://x/o>")
     counts2 = [execute_query(FLAGSHIP_QUERY, db)[0][0] for _ in range(4)]
     assert set(counts2) == set(counts)  # unrelated triple: same count
+    # now a RELEVANT mutation: a fresh employee matching all three
+    # patterns must change the count — this catches a serve registration
+    # (cached seeds + precomputed hop windows) surviving a version bump
+    ds_pfx = "https://data.cityofchicago.org/resource/xzkq-xp2w/"
+    # the synthetic store is pre-encoded: map the dept IRI to the real id
+    db.dictionary.str_to_id[f"http://synthetic/d{ds.dept_base}"] = (
+        ds.dept_base)
+    dept = f"<http://synthetic/d{ds.dept_base}>"
+    db.add_triple("<http://x/new_e>", f"<{ds_pfx}worksFor>", dept)
+    db.add_triple("<http://x/new_e>", f"<{ds_pfx}annual_salary>", '"12345"')
+    counts3 = [execute_query(FLAGSHIP_QUERY, db)[0][0] for _ in range(3)]
+    assert {int(c) for c in counts3} == {int(counts[0]) + 1}
 
 
 @requires_gpu
