@@ -146,3 +146,45 @@ def test_http_query_form_encoding(client):
     assert r.status_code == 200
     vals = [b["n"]["value"] for b in r.json()["results"]["bindings"]]
     assert vals == ["Alice"]
+
+
+def test_query_builder_parity_surface():
+    """PyO3-surface parity: projections, group_by_* dicts, count,
+    asc/desc, streaming config introspection."""
+    from kolibrie_amd import SparqlDatabase
+    from kolibrie_amd.engine.query_builder import QueryBuilder
+
+    db = SparqlDatabase(device="cpu")
+    for i in range(6):
+        db.add_triple(f"<http://e/s{i % 3}>", f"<http://e/p{i % 2}>",
+                      f'"{i}"')
+    qb = QueryBuilder(db).with_predicate_starting("http://e/p")
+    assert qb.count() == 6
+    assert sorted(set(qb.get_subjects())) == [
+        "http://e/s0", "http://e/s1", "http://e/s2"]
+    assert set(QueryBuilder(db).distinct().get_predicates()) == {
+        "http://e/p0", "http://e/p1"}
+    g = QueryBuilder(db).group_by_subject()
+    assert len(g) == 3 and sum(len(v) for v in g.values()) == 6
+    gp = QueryBuilder(db).group_by_predicate()
+    assert len(gp) == 2
+    rows = QueryBuilder(db).with_subject("http://e/s0").desc().execute()
+    assert rows == sorted(rows, reverse=True)
+    # predicate_ending filter
+    assert QueryBuilder(db).with_predicate_ending("p1").count() == 3
+
+    # streaming config surface
+    qs = (QueryBuilder(db).window(10, 5).with_periodic_report(4)
+          .with_tick_strategy("TimeDriven").with_stream_operator("RSTREAM"))
+    assert qs.get_window_config() == (10, 5)
+    assert qs.get_report_strategies() == ["Periodic"]
+    assert qs.get_periodic_periods() == [4]
+    assert qs.get_stream_operator() == "RSTREAM"
+    assert not qs.is_streaming()
+    qs.as_stream()
+    assert qs.is_streaming()
+    qs.add_stream_triple(("<a>", "<b>", "<c>"), 1)
+    qs.stop_stream()
+    assert not qs.is_streaming()
+    qs.clear_stream_results()
+    assert qs.get_all_stream_results() == []
