@@ -440,7 +440,12 @@ class SparqlDatabase:
         return cache[1]
 
     # ------------------------------------------------------------------ query
-    def query(self, sparql: str) -> List[List[str]]:
+    def query(self, sparql: Optional[str] = None):
+        """With a SPARQL string: execute it (rows of decoded strings).
+        With no argument: return a fluent QueryBuilder — the reference's
+        PySparqlDatabase.query() surface (py_query_builder.rs:136)."""
+        if sparql is None:
+            return self.query_builder()
         from ..engine.query import execute_query
         return execute_query(sparql, self)
 

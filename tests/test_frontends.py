@@ -188,3 +188,18 @@ def test_query_builder_parity_surface():
     assert not qs.is_streaming()
     qs.clear_stream_results()
     assert qs.get_all_stream_results() == []
+
+
+def test_database_query_no_arg_returns_builder():
+    """Reference parity: db.query() with no argument is the fluent
+    builder entry point; db.query(sparql) executes."""
+    from kolibrie_amd import SparqlDatabase
+    from kolibrie_amd.engine.query_builder import QueryBuilder
+
+    db = SparqlDatabase(device="cpu")
+    db.add_triple("<http://e/a>", "<http://e/p>", "<http://e/b>")
+    qb = db.query()
+    assert isinstance(qb, QueryBuilder)
+    assert qb.count() == 1
+    rows = db.query("SELECT ?s WHERE { ?s <http://e/p> ?o }")
+    assert rows == [["http://e/a"]]
