@@ -109,6 +109,12 @@ class Reasoner:
         self._pending.clear()
 
     # ------------------------------------------------------------- rules --
+    def encode_term(self, term: str) -> int:
+        """Intern a raw term string, returning its u32 id (reference
+        PyKnowledgeGraph.encode_term, py_knowledge_graph.rs:232) — for
+        building Constant terms in hand-constructed rules."""
+        return self.dictionary.encode(term)
+
     def add_rule(self, rule: Rule):
         self.rules.append(rule)
         self.rule_index.add_rule(rule)

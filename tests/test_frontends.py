@@ -203,3 +203,42 @@ def test_database_query_no_arg_returns_builder():
     assert qb.count() == 1
     rows = db.query("SELECT ?s WHERE { ?s <http://e/p> ?o }")
     assert rows == [["http://e/a"]]
+
+
+def test_compat_knowledge_graph_rule_surface():
+    """Reference PyO3 KG parity: hand-built Rule(premise, filters,
+    conclusion) with Term.Variable/Constant and FilterCondition; the
+    infer_* entry points return decoded new triples."""
+    from kolibrie_amd.compat import (FilterCondition, PyKnowledgeGraph,
+                                     Rule_from_parts, Term, TriplePattern)
+    from kolibrie_amd.storage.terms import Constant
+
+    kg = PyKnowledgeGraph()
+    kg.add_abox_triple("<http://e/a>", "<http://e/salary>", "50")
+    kg.add_abox_triple("<http://e/b>", "<http://e/salary>", "10")
+    sal = Constant(kg.encode_term("<http://e/salary>"))
+    rich = Constant(kg.encode_term("<http://e/rich>"))
+    yes = Constant(kg.encode_term("yes"))
+    r = Rule_from_parts(
+        [TriplePattern(Term.Variable("x"), sal, Term.Variable("s"))],
+        [FilterCondition("s", ">", "20")],
+        [TriplePattern(Term.Variable("x"), rich, yes)])
+    kg.add_rule(r)
+    derived = kg.infer_new_facts_semi_naive()
+    assert derived == [("<http://e/a>", "<http://e/rich>", "yes")]
+    # second run: fixpoint reached, nothing new
+    assert kg.infer_new_facts_semi_naive() == []
+    assert kg.query_abox(p="<http://e/rich>") == [
+        ("<http://e/a>", "<http://e/rich>", "yes")]
+    # variable-vs-variable filter: != compares dictionary ids
+    same = Constant(kg.encode_term("<http://e/self_paid>"))
+    pay = Constant(kg.encode_term("<http://e/pays>"))
+    kg.add_abox_triple("<http://e/a>", "<http://e/pays>", "<http://e/a>")
+    kg.add_abox_triple("<http://e/a>", "<http://e/pays>", "<http://e/b>")
+    r2 = Rule_from_parts(
+        [TriplePattern(Term.Variable("x"), pay, Term.Variable("y"))],
+        [FilterCondition("x", "=", "y")],
+        [TriplePattern(Term.Variable("x"), same, yes)])
+    kg.add_rule(r2)
+    new2 = kg.infer_new_facts_semi_naive()
+    assert ("<http://e/a>", "<http://e/self_paid>", "yes") in new2
