@@ -136,3 +136,27 @@ def Rule_from_parts(premise, filters, conclusion):
     """Reference PyRule ctor shape: Rule(premise, filters, conclusion)."""
     return Rule(premise=list(premise), filters=list(filters),
                 conclusion=list(conclusion))
+
+# Enum/name exports matching the reference module registrations
+# (py_query_builder.rs:814-826): StreamingQuery is the builder itself
+# (as_stream returns the same fluent object), SortDirection maps to
+# asc()/desc(), and the RSP enums re-export the engine's constants.
+from .rsp.s2r import ReportStrategy, Tick  # noqa: F401,E402
+from .rsp.r2s import StreamOperator  # noqa: F401,E402
+
+StreamingQuery = QueryBuilder
+
+
+class SortDirection:
+    Ascending = "Ascending"
+    Descending = "Descending"
+
+
+class PeriodicReportStrategy:
+    """Periodic report with a period (reference PyPeriodicReportStrategy:
+    holds the period used by with_periodic_report)."""
+
+    __slots__ = ("period",)
+
+    def __init__(self, period: int):
+        self.period = period
