@@ -299,3 +299,48 @@ def test_vocab_annex_semantics(tmp_path):
         db.query(f"SELECT (COUNT(*) AS ?c) WHERE {{ ?s ?p ?o }}")
     assert db2.query(f'SELECT ?s WHERE {{ ?s <{EX}lab> "name 5" }}') == \
         db.query(f'SELECT ?s WHERE {{ ?s <{EX}lab> "name 5" }}')
+
+
+def test_parse_ntriples_file_multichunk_merge(tmp_path):
+    """Force the MULTI-chunk parallel path (>1 MiB file): cross-chunk
+    duplicate terms (incl. escaped literals materialized per chunk) must
+    dedup to single ids, and ids must be deterministic across loads of
+    the same file."""
+    from kolibrie_amd import SparqlDatabase
+    from kolibrie_amd.ops import _native
+    import pytest
+    if _native is None:
+        pytest.skip("native extension required")
+    EX = "http://example.org/padpadpadpadpadpadpadpadpadpad/"
+    lines = []
+    for i in range(12_000):
+        # shared terms recur across the whole file -> land in many chunks
+        lines.append(f'<{EX}e{i}> <{EX}type> <{EX}Employee> .')
+        lines.append(f'<{EX}e{i}> <{EX}note> "esc\\tx\\"q\\\\{i % 7}" .')
+    text = "\n".join(lines) + "\n"
+    p = tmp_path / "big.nt"
+    p.write_text(text)
+    assert p.stat().st_size > (1 << 20)  # multi-chunk threshold
+
+    db1 = SparqlDatabase()
+    db1.parse_ntriples_file(str(p))
+    assert db1.triple_count() == 24_000
+    # shared constants resolved to ONE id each
+    assert db1.query(
+        f"SELECT (COUNT(*) AS ?c) WHERE {{ ?e <{EX}type> <{EX}Employee> }}"
+    ) == [["12000"]]
+    # escaped literal dedup: 7 distinct notes
+    assert db1.query(
+        f"SELECT (COUNT(DISTINCT ?n) AS ?c) WHERE {{ ?e <{EX}note> ?n }}"
+    ) == [["7"]]
+    # escape decoding round-trips
+    out = db1.query(f'SELECT ?n WHERE {{ <{EX}e3> <{EX}note> ?n }}')
+    assert out == [['esc\tx"q\\3']]
+
+    # determinism: same file + shard count -> identical annex ids
+    db2 = SparqlDatabase()
+    db2.parse_ntriples_file(str(p))
+    assert db2.dictionary.lookup(f"{EX}e77") == \
+        db1.dictionary.lookup(f"{EX}e77")
+    assert db2.dictionary.lookup('esc\tx"q\\5') == \
+        db1.dictionary.lookup('esc\tx"q\\5')
