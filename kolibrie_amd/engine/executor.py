@@ -32,6 +32,7 @@ from .bindings import Bindings
 from .scan import scan_unit
 from .tensor_utils import group_index, merge_join_indices
 from . import exec_stats
+from . import tracer
 
 import logging
 
@@ -96,6 +97,15 @@ class ExecutionEngine:
 
     # ----------------------------------------------------------- dispatch --
     def execute(self, op: PhysicalOp, incoming: Bindings) -> Bindings:
+        if tracer.is_enabled():
+            sp = tracer.span(type(op).__name__, self.device)
+            try:
+                return self._execute(op, incoming)
+            finally:
+                sp.close()
+        return self._execute(op, incoming)
+
+    def _execute(self, op: PhysicalOp, incoming: Bindings) -> Bindings:
         if isinstance(op, PUnit):
             return incoming
         needed = getattr(op, "needed", None)

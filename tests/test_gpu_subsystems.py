@@ -108,3 +108,26 @@ def test_query_columns_device_matches_rows():
     cols = db.query_columns(q)
     assert cols["s"] == [r[0] for r in rows]
     assert cols["o"] == [r[1] for r in rows]
+
+
+@pytest.mark.gpu
+def test_tracer_per_op_device_timing():
+    """Opt-in tracer records hipEvent-timed per-operator spans on the
+    device path (SURVEY §5 tracing)."""
+    from kolibrie_amd import SparqlDatabase
+    from kolibrie_amd.engine import tracer
+
+    db = SparqlDatabase(device="cuda:0")
+    for i in range(200):
+        db.add_triple(f"<http://e/s{i}>", "<http://e/p>", f'"{i}"')
+    tracer.enable()
+    try:
+        tracer.reset()
+        db.query("SELECT (COUNT(*) AS ?c) WHERE "
+                 "{ ?s <http://e/p> ?o . FILTER(?o > 10) }")
+        snap = tracer.snapshot()
+    finally:
+        tracer.disable()
+        tracer.reset()
+    assert any("Scan" in k for k in snap)
+    assert all(calls >= 1 and ms >= 0.0 for calls, ms in snap.values())

@@ -569,3 +569,24 @@ def test_query_columns_matches_rows():
     import pytest
     with pytest.raises(ValueError):
         db.query_columns(f"INSERT DATA {{ <{EX}x> <{EX}p> \"v\" }}")
+
+
+def test_tracer_cpu_fallback_and_overhead_off():
+    """Tracer disabled: no records accumulate; enabled on CPU it
+    wall-clock-times each operator."""
+    from kolibrie_amd import SparqlDatabase
+    from kolibrie_amd.engine import tracer
+
+    db = SparqlDatabase(device="cpu")
+    db.add_triple("<http://e/a>", "<http://e/p>", '"1"')
+    tracer.reset()
+    db.query("SELECT ?s WHERE { ?s <http://e/p> ?o }")
+    assert tracer.snapshot() == {}
+    tracer.enable()
+    try:
+        db.query("SELECT ?s WHERE { ?s <http://e/p> ?o }")
+        snap = tracer.snapshot()
+    finally:
+        tracer.disable()
+        tracer.reset()
+    assert snap and all(ms >= 0.0 for _, ms in snap.values())
