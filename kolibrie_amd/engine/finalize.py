@@ -411,7 +411,23 @@ def _gc_paused():
             gc.enable()
 
 
-def _decode_column(rows: Bindings, v: str, db) -> List[str]:
+def _host_ids(rows: Bindings, names) -> Dict[str, "object"]:
+    """ONE device->host transfer for every requested column: the
+    per-column .cpu() each cost a full device sync (~15 us on ROCm), and
+    point queries decode 5-6 columns — batching them was worth ~30% of a
+    small SELECT's end-to-end time."""
+    present = [v for v in names if rows.has(v)]
+    if not present:
+        return {}
+    if len(present) == 1:
+        v = present[0]
+        return {v: (rows.col(v).to(torch.int64) & 0xFFFFFFFF).cpu().numpy()}
+    m = torch.stack([rows.col(v).to(torch.int64) & 0xFFFFFFFF
+                     for v in present]).cpu().numpy()
+    return {v: m[i] for i, v in enumerate(present)}
+
+
+def _decode_column(rows: Bindings, v: str, db, ids=None) -> List[str]:
     """Batch-decode one result column: single D2H copy + numpy gather over
     the cached dictionary mirror; quoted triples (rare) decode per cell."""
     import numpy as np
@@ -420,7 +436,8 @@ def _decode_column(rows: Bindings, v: str, db) -> List[str]:
         return [""] * n
     table = db.dictionary.np_table()
     n_plain = len(table)
-    ids = (rows.col(v).to(torch.int64) & 0xFFFFFFFF).cpu().numpy()
+    if ids is None:
+        ids = (rows.col(v).to(torch.int64) & 0xFFFFFFFF).cpu().numpy()
     # default "": UNBOUND and out-of-vocabulary plain ids (synthetic dense
     # id blocks have no string form) both decode to ""
     out = np.full(n, "", dtype=object)
@@ -453,7 +470,8 @@ def decode_columns(select: SelectQuery, rows: Bindings, db
         names = rows.variables
     else:
         names = [p.output_name() for p in select.variables]
-    return {v: _decode_column(rows, v, db) for v in names}
+    host = _host_ids(rows, names)
+    return {v: _decode_column(rows, v, db, host.get(v)) for v in names}
 
 
 def decode_rows(select: SelectQuery, rows: Bindings, db) -> List[List[str]]:
@@ -470,16 +488,19 @@ def decode_rows(select: SelectQuery, rows: Bindings, db) -> List[List[str]]:
     n = rows.n
     if n >= 64:
         with _gc_paused():
-            cols_dec = [_decode_column(rows, v, db) for v in names]
+            host_ids = _host_ids(rows, names)
+            cols_dec = [_decode_column(rows, v, db, host_ids.get(v))
+                        for v in names]
             # row pivot: map/zip beats any numpy object-array reshape
             # (object stack+tolist measured 9.5 ms at 14k rows); the
             # remaining cost IS building the row lists — decode_columns
             # below avoids it entirely for columnar consumers
             return list(map(list, zip(*cols_dec)))
+    host_np = _host_ids(rows, names)
     host = {}
     for v in names:
-        if rows.has(v):
-            host[v] = (rows.col(v).to(torch.int64) & 0xFFFFFFFF).cpu().tolist()
+        if v in host_np:
+            host[v] = host_np[v].tolist()
         else:
             host[v] = [None] * rows.n
     # plain (non-quoted, interned) ids decode by direct list index — the

@@ -130,13 +130,19 @@ def scan_unit(idx: GraphIndex, consts: Dict[int, int],
         else:
             k = consts[pos[0]] << 32
             k2 = k + 0x1_0000_0000
-        if k2 > 0x7FFF_FFFF_FFFF_FFFF:
-            probe = torch.tensor([k], dtype=torch.int64, device=dev)
-            lo = int(torch.searchsorted(key12, probe, side="left").item())
-            hi = idx.n
+        cached = idx.rcache.get((code, k))
+        if cached is not None:
+            lo, hi = cached
         else:
-            probe = torch.tensor([k, k2], dtype=torch.int64, device=dev)
-            lo, hi = torch.searchsorted(key12, probe, side="left").tolist()
+            if k2 > 0x7FFF_FFFF_FFFF_FFFF:
+                probe = torch.tensor([k], dtype=torch.int64, device=dev)
+                lo = int(torch.searchsorted(key12, probe, side="left").item())
+                hi = idx.n
+            else:
+                probe = torch.tensor([k, k2], dtype=torch.int64, device=dev)
+                lo, hi = torch.searchsorted(key12, probe, side="left").tolist()
+            if len(idx.rcache) < 65536:
+                idx.rcache[(code, k)] = (lo, hi)
         s, p, o = _cols_from_order_slice(idx, code, lo, hi, mat_need)
         n_rows = hi - lo
     # post-filter constants not covered by the prefix

@@ -60,13 +60,17 @@ def _u32_tensor_to_i64(c: torch.Tensor) -> torch.Tensor:
 class GraphIndex:
     """Immutable sorted index of one graph's triples (or a merged view)."""
 
-    __slots__ = ("device", "n", "orders")
+    __slots__ = ("device", "n", "orders", "rcache")
 
     def __init__(self, device: torch.device, n: int, orders):
         self.device = device
         self.n = n
         # orders[code] = (key12 int64 [n] sorted, z int32 [n])
         self.orders = orders
+        # (order, packed key) -> (lo, hi) range memo: the index is
+        # immutable, and each searchsorted + host readback costs a device
+        # sync — repeated plans re-scan identical constant ranges
+        self.rcache = {}
 
     @staticmethod
     def empty(device) -> "GraphIndex":
