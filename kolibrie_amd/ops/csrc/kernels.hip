@@ -1896,9 +1896,15 @@ static void parse_chunk_nt(const char* data, size_t begin, size_t end,
     while (b < e && isspace(static_cast<unsigned char>(line[b]))) ++b;
     while (e > b && isspace(static_cast<unsigned char>(line[e - 1]))) --e;
     if (b >= e || line[b] == '#') continue;
+    // RDF-star detection via memchr hops (the naive per-byte scan read
+    // every line twice; memchr skips to the handful of '<' per line)
     bool star = false;
-    for (size_t i = b; i + 1 < e; ++i)
-      if (line[i] == '<' && line[i + 1] == '<') { star = true; break; }
+    for (const char* q = line + b;
+         (q = static_cast<const char*>(
+              memchr(q, '<', static_cast<size_t>(line + e - q)))) != nullptr;
+         ++q) {
+      if (q + 1 < line + e && q[1] == '<') { star = true; break; }
+    }
     if (star) { out.fallback.push_back(line_no - 1); continue; }
     int64_t term_ids[4];
     const int max_terms = quads ? 4 : 3;
@@ -1910,14 +1916,22 @@ static void parse_chunk_nt(const char* data, size_t begin, size_t end,
       if (i >= e) break;
       char c = line[i];
       if (c == '<') {
-        size_t j = i + 1;
-        while (j < e && line[j] != '>') ++j;
-        if (j >= e) { ok = false; break; }
+        const char* gt = static_cast<const char*>(
+            memchr(line + i + 1, '>', e - i - 1));
+        if (gt == nullptr) { ok = false; break; }
+        size_t j = static_cast<size_t>(gt - line);
         term_ids[nt++] = intern(line + i + 1, j - i - 1);
         i = j + 1;
       } else if (c == '"') {
         size_t j = i + 1;
-        while (j < e && !(line[j] == '"' && line[j - 1] != '\\')) ++j;
+        for (;;) {
+          const char* qq = static_cast<const char*>(
+              memchr(line + j, '"', e - j));
+          if (qq == nullptr) { j = e; break; }
+          j = static_cast<size_t>(qq - line);
+          if (line[j - 1] != '\\') break;
+          ++j;
+        }
         if (j >= e) { ok = false; break; }
         term_ids[nt++] = intern(line + i + 1, j - i - 1);
         i = j + 1;
