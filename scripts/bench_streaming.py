@@ -102,6 +102,11 @@ def bench_r2s_columnar(device="cuda:0", n=100_000, runs=20):
         s[idx] = torch.randint(0, 1 << 20, (k,), dtype=torch.int32, device=dev)
         frames.append((s, o))
     for mode in (StreamOperator.ISTREAM, StreamOperator.DSTREAM):
+        # untimed warmup: the very first rows_diff pays allocator growth
+        # + first-kernel costs that otherwise land in ISTREAM's average
+        wop = Relation2StreamOperator(mode)
+        for s, o in frames[:2]:
+            wop.eval_columns([s, o])
         op = Relation2StreamOperator(mode)
         if dev.type == "cuda":
             torch.cuda.synchronize()
