@@ -442,7 +442,7 @@ class ExecutionEngine:
         seed_z = seed_z.contiguous()
         hop_regions = [
             (r[0].contiguous(), r[1], r[2],
-             self._chain_hop_table(native, r[0], seed_key12.numel()))
+             *self._chain_hop_table(native, r[0], seed_key12.numel()))
             for r in hop_regions]
         # native kernels stream 32-bit columns: the region high words are
         # each hop's constant predicate and the seed key's low word is the
@@ -454,7 +454,8 @@ class ExecutionEngine:
                      _t.empty(0, dtype=_t.int32, device=seed_b.device)
                      for r in hop_regions],
                     [r[2] for r in hop_regions],
-                    [r[3] for r in hop_regions])
+                    [r[3] for r in hop_regions],
+                    [r[4] for r in hop_regions])
         op._chain_cache = (self.db.store.version, seed_key12, seed_z,
                            hop_regions, native, hop_args, seed_b)
         if hasattr(op, "_chain_graph"):
@@ -518,6 +519,9 @@ class ExecutionEngine:
     _HOP_TABLE_MAX_ROWS = 1_000_000
 
     def _chain_hop_table(self, native, region, n_seeds):
+        """Returns (table, dmin): dmin >= 0 marks a DIRECT-indexed dense
+        table counts[v - dmin] (coalesced probes); dmin == -1 marks the
+        hashed open-addressing forms; empty table = binary-search hop."""
         import os
         import torch as _t
         n = region.numel()
@@ -525,16 +529,28 @@ class ExecutionEngine:
                                  self._HOP_TABLE_MAX_ROWS))
         if (native is None or not region.is_cuda or n == 0
                 or n > cap or n * 4 > n_seeds):
-            return _t.empty(0, dtype=_t.int64, device=region.device)
+            return _t.empty(0, dtype=_t.int64, device=region.device), -1
         vals, counts = _t.unique_consecutive(region & 0xFFFFFFFF,
                                              return_counts=True)
-        if vals.numel() and int(vals.max().item()) < 0x1FFFFFF \
+        vmin = int(vals.min().item()) if vals.numel() else 0
+        vmax = int(vals.max().item()) if vals.numel() else 0
+        span = vmax - vmin + 1
+        if (vals.numel() and span <= (1 << 22)
+                and span <= max(4 * vals.numel(), 1024)
+                and os.environ.get("KOLIBRIE_DIRECT_HOP", "1") != "0"):
+            # dense value space (e.g. a contiguous id block): direct
+            # index beats hashing — consecutive seed values read
+            # consecutive table slots, so the wave's loads coalesce
+            tbl = _t.zeros(span, dtype=_t.int32, device=region.device)
+            tbl[(vals - vmin).to(_t.long)] = counts.to(_t.int32)
+            return tbl, vmin
+        if vals.numel() and vmax < 0x1FFFFFF \
                 and int(counts.max().item()) < 128:
             # packed (val<<7)|count u32 table: half the per-probe bytes
             # and table footprint (L2 residency beside the seed stream)
             return native.build_count_table32(
-                ((vals << 7) | counts).to(_t.int32))
-        return native.build_count_table((vals << 32) | counts)
+                ((vals << 7) | counts).to(_t.int32)), -1
+        return native.build_count_table((vals << 32) | counts), -1
 
     def _chain_count_torch(self, seed_key12, seed_z, hop_regions) -> int:
         # torch fallback (CPU oracle): vectorized per-hop count product

@@ -580,6 +580,13 @@ struct ChainHops {
   int32_t src[kMaxHops];           // 0: seed B component, 1: seed Z column
   const unsigned long long* table[kMaxHops];  // optional count table (u64)
   const uint32_t* table32[kMaxHops];  // packed (val<<7)|count u32 variant
+  // DIRECT-indexed count table (dense value spaces, e.g. a contiguous
+  // id block): counts[v - dmin] — consecutive seed values hit
+  // consecutive slots, so the probe loads COALESCE instead of hashing
+  // to random lines
+  const uint32_t* direct[kMaxHops];
+  int64_t dmin[kMaxHops];
+  int64_t dlen[kMaxHops];
   int64_t tmask[kMaxHops];
   int k;
   // stage a window through LDS only when it exceeds this row count:
@@ -856,8 +863,9 @@ __global__ void chain_tile_bounds(const int32_t* __restrict__ seed_b,
         static_cast<uint32_t>(seed_b[min((t + 1) * kTile - 1, m - 1)]);
     for (int h = 0; h < hops.k; ++h) {
       int64_t lo = 0, hi = hops.n[h];
-      if (hops.table[h] != nullptr || hops.table32[h] != nullptr) {
-        lo = hi = 0;  // hashed hop: window unused
+      if (hops.table[h] != nullptr || hops.table32[h] != nullptr
+          || hops.direct[h] != nullptr) {
+        lo = hi = 0;  // hashed/direct hop: window unused
       } else if (hops.src[h] == 0) {
         lo = lower_bound_u32(hops.key32[h], hops.n[h], first_b);
         hi = upper_bound_u32(hops.key32[h], hops.n[h], last_b);
@@ -904,6 +912,18 @@ __global__ void chain_count_kernel(const int32_t* __restrict__ seed_b,
     // chains — letting them all issue gives the scheduler ILP to hide
     // L2 latency
     for (int h = 0; h < hops.k; ++h) {
+      if (hops.direct[h] != nullptr) {
+        // dense-value hop: counts[v - dmin], coalesced across the wave
+        for (int u = 0; u < kSub; ++u) {
+          if (!act[u]) continue;
+          uint32_t v = hops.src[h] == 0 ? b_c[u] : z_c[u];
+          uint64_t off = static_cast<uint64_t>(
+              static_cast<int64_t>(v) - hops.dmin[h]);
+          prod[u] *= off < static_cast<uint64_t>(hops.dlen[h])
+                         ? hops.direct[h][off] : 0u;
+        }
+        continue;
+      }
       if (hops.table32[h] != nullptr) {
         // packed count-table hop: one 4-byte L2 load per seed
         uint32_t mask = static_cast<uint32_t>(hops.tmask[h]);
@@ -1003,6 +1023,7 @@ void chain_count_into(at::Tensor seed_b, at::Tensor seed_z,
                       std::vector<at::Tensor> hop_key32,
                       std::vector<int64_t> hop_src,
                       std::vector<at::Tensor> hop_table,
+                      std::vector<int64_t> hop_dmin,
                       at::Tensor win, at::Tensor total) {
   TORCH_CHECK(seed_b.is_cuda() && seed_b.dtype() == at::kInt
               && seed_z.is_cuda());
@@ -1018,7 +1039,13 @@ void chain_count_into(at::Tensor seed_b, at::Tensor seed_z,
     hops.n[h] = hop_key32[h].numel();
     hops.src[h] = static_cast<int32_t>(hop_src[h]);
     if (hop_table[h].numel() > 0) {
-      if (hop_table[h].dtype() == at::kInt) {
+      bool is_direct = h < hop_dmin.size() && hop_dmin[h] >= 0;
+      if (is_direct) {
+        hops.direct[h] = reinterpret_cast<const uint32_t*>(
+            hop_table[h].data_ptr<int32_t>());
+        hops.dmin[h] = hop_dmin[h];
+        hops.dlen[h] = hop_table[h].numel();
+      } else if (hop_table[h].dtype() == at::kInt) {
         hops.table32[h] = reinterpret_cast<const uint32_t*>(
             hop_table[h].data_ptr<int32_t>());
       } else {
@@ -1070,7 +1097,8 @@ static std::vector<std::unique_ptr<ChainServe>> g_chain_serves;
 int64_t register_chain_serve(at::Tensor seed_b, at::Tensor seed_z,
                              std::vector<at::Tensor> hop_key32,
                              std::vector<int64_t> hop_src,
-                             std::vector<at::Tensor> hop_table) {
+                             std::vector<at::Tensor> hop_table,
+                             std::vector<int64_t> hop_dmin) {
   TORCH_CHECK(seed_b.is_cuda() && seed_b.dtype() == at::kInt
               && seed_z.is_cuda());
   auto cs = std::make_unique<ChainServe>();
@@ -1088,7 +1116,13 @@ int64_t register_chain_serve(at::Tensor seed_b, at::Tensor seed_z,
     cs->hops.n[h] = hop_key32[h].numel();
     cs->hops.src[h] = static_cast<int32_t>(hop_src[h]);
     if (hop_table[h].numel() > 0) {
-      if (hop_table[h].dtype() == at::kInt) {
+      bool is_direct = h < hop_dmin.size() && hop_dmin[h] >= 0;
+      if (is_direct) {
+        cs->hops.direct[h] = reinterpret_cast<const uint32_t*>(
+            hop_table[h].data_ptr<int32_t>());
+        cs->hops.dmin[h] = hop_dmin[h];
+        cs->hops.dlen[h] = hop_table[h].numel();
+      } else if (hop_table[h].dtype() == at::kInt) {
         cs->hops.table32[h] = reinterpret_cast<const uint32_t*>(
             hop_table[h].data_ptr<int32_t>());
       } else {
@@ -1167,7 +1201,8 @@ void release_chain_serve(int64_t id) {
 int64_t chain_count(at::Tensor seed_b, at::Tensor seed_z,
                     std::vector<at::Tensor> hop_key32,
                     std::vector<int64_t> hop_src,
-                    std::vector<at::Tensor> hop_table) {
+                    std::vector<at::Tensor> hop_table,
+                    std::vector<int64_t> hop_dmin) {
   TORCH_CHECK(seed_b.is_cuda() && seed_b.dtype() == at::kInt
               && seed_z.is_cuda());
   TORCH_CHECK(hop_key32.size() <= static_cast<size_t>(kMaxHops));
@@ -1187,7 +1222,13 @@ int64_t chain_count(at::Tensor seed_b, at::Tensor seed_z,
       TORCH_CHECK(hop_table[h].is_cuda()
                   && (hop_table[h].dtype() == at::kLong
                       || hop_table[h].dtype() == at::kInt));
-      if (hop_table[h].dtype() == at::kInt) {
+      bool is_direct = h < hop_dmin.size() && hop_dmin[h] >= 0;
+      if (is_direct) {
+        hops.direct[h] = reinterpret_cast<const uint32_t*>(
+            hop_table[h].data_ptr<int32_t>());
+        hops.dmin[h] = hop_dmin[h];
+        hops.dlen[h] = hop_table[h].numel();
+      } else if (hop_table[h].dtype() == at::kInt) {
         hops.table32[h] = reinterpret_cast<const uint32_t*>(
             hop_table[h].data_ptr<int32_t>());
       } else {
@@ -1196,6 +1237,7 @@ int64_t chain_count(at::Tensor seed_b, at::Tensor seed_z,
       }
       hops.tmask[h] = hop_table[h].numel() - 1;
     }
+
   }
   auto total = at::zeros({1}, seed_b.options().dtype(at::kLong));
   if (m > 0) {
@@ -3435,12 +3477,18 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("chain_tile", []() { return static_cast<int64_t>(kTile); },
         "seed rows per chain tile (win buffer sizing)");
   m.def("chain_count", &chain_count,
-        "fused COUNT(*) over a seed scan + probe-hop chain");
+        "fused COUNT(*) over a seed scan + probe-hop chain",
+        py::arg("seed_b"), py::arg("seed_z"), py::arg("hop_key32"),
+        py::arg("hop_src"), py::arg("hop_table"),
+        py::arg("hop_dmin") = std::vector<int64_t>{});
   m.def("chain_count_into", &chain_count_into,
         "allocation/sync-free chain count into caller buffers "
         "(hipGraph-capturable)");
   m.def("register_chain_serve", &register_chain_serve,
-        "cache a COUNT chain's launch arguments C++-side; returns handle");
+        "cache a COUNT chain's launch arguments C++-side; returns handle",
+        py::arg("seed_b"), py::arg("seed_z"), py::arg("hop_key32"),
+        py::arg("hop_src"), py::arg("hop_table"),
+        py::arg("hop_dmin") = std::vector<int64_t>{});
   m.def("serve_chain_count", &serve_chain_count,
         py::call_guard<py::gil_scoped_release>(),
         "one-call serving: direct launches + pinned 8-byte readback");

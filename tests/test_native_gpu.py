@@ -276,6 +276,11 @@ def test_chain_count_table_hop_matches_search():
     table32 = _native.build_count_table32(
         ((vals << 7) | counts).to(torch.int32))
     assert table.numel() >= 2 * vals.numel()
+    # DIRECT dense table: counts[v - vmin], coalesced probe path
+    vmin = int(vals.min())
+    span = int(vals.max()) - vmin + 1
+    direct = torch.zeros(span, dtype=torch.int32, device=dev)
+    direct[(vals - vmin).to(torch.long)] = counts.to(torch.int32)
     for src in (0, 1):
         want = oracle(src)
         got_search = _native.chain_count(
@@ -283,9 +288,12 @@ def test_chain_count_table_hop_matches_search():
                                                 device=dev)])
         got_table = _native.chain_count(sb, sz, [rg32], [src], [table])
         got_t32 = _native.chain_count(sb, sz, [rg32], [src], [table32])
+        got_direct = _native.chain_count(sb, sz, [rg32], [src], [direct],
+                                         [vmin])
         assert got_search == want
         assert got_table == want
         assert got_t32 == want
+        assert got_direct == want
 
 
 @requires_gpu
