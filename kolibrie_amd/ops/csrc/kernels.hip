@@ -2304,7 +2304,9 @@ struct BumpArena {
   std::vector<std::unique_ptr<char[]>> blocks;
   size_t used = 0, cap = 0;
   std::string_view add(std::string_view sv) {
-    if (used + sv.size() > cap) {
+    if (blocks.empty() || used + sv.size() > cap) {
+      // blocks.empty() guard: a zero-length first insert (e.g. the empty
+      // literal "") must still allocate before blocks.back()
       size_t bs = std::max<size_t>(size_t(1) << 24, sv.size());
       blocks.emplace_back(new char[bs]);
       used = 0;
@@ -2353,6 +2355,34 @@ static double parse_value_full(std::string_view s) {
 int64_t vocab_create(int64_t base_id) {
   auto v = std::make_unique<Vocab>();
   v->base = base_id;
+  g_vocabs.push_back(std::move(v));
+  return static_cast<int64_t>(g_vocabs.size() - 1);
+}
+
+// create seeded with the Python dictionary's frozen prefix: the bulk
+// parser dedups against these ids too (without this a prefix string —
+// including "" at id 0, or any term interned before the first bulk
+// load — would get a DUPLICATE annex id and constant-term queries
+// would miss the bulk-loaded rows)
+int64_t vocab_create_seeded(py::list strings) {
+  auto v = std::make_unique<Vocab>();
+  v->base = static_cast<int64_t>(strings.size());
+  std::hash<std::string_view> hasher;
+  int64_t i = 0;
+  for (auto h : strings) {
+    std::string str = py::cast<std::string>(h);
+    std::string_view key(str);
+    uint64_t hh = hasher(key);
+    auto& sh = v->shards[hh % v->nshards];
+    bool found;
+    size_t si = sh.map.find_or_insert(hh, key, found);
+    if (!found) {
+      std::string_view sv = sh.arena.add(key);
+      sh.map.slots[si].ptr = sv.data();
+      sh.map.slots[si].id = i;
+    }
+    ++i;
+  }
   g_vocabs.push_back(std::move(v));
   return static_cast<int64_t>(g_vocabs.size() - 1);
 }
@@ -3465,6 +3495,8 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("vocab_values", &vocab_values);
   m.def("vocab_export_strings", &vocab_export_strings);
   m.def("stats_gather", &stats_gather, "K8 one-pass database statistics");
+  m.def("vocab_create_seeded", &vocab_create_seeded,
+        "create a vocab annex pre-seeded with the frozen Python prefix");
   m.def("parse_ntriples_file_annex", &parse_ntriples_file_annex,
         "file -> parallel parse -> intern into the native vocab annex");
   m.doc() = "kolibrie_amd native CDNA4 kernels (gfx950)";

@@ -59,7 +59,11 @@ class Dictionary:
             if _native is None:
                 raise RuntimeError("native extension required for the "
                                    "bulk-vocabulary annex")
-            self.annex = (_native, _native.vocab_create(len(self.id_to_str)))
+            # seed with the frozen Python prefix so the native bulk
+            # parser dedups against EXISTING ids ("" at id 0, or any
+            # term interned before the first bulk load)
+            self.annex = (_native,
+                          _native.vocab_create_seeded(self.id_to_str))
         return self.annex
 
     def np_table(self):

@@ -344,3 +344,34 @@ def test_parse_ntriples_file_multichunk_merge(tmp_path):
         db1.dictionary.lookup(f"{EX}e77")
     assert db2.dictionary.lookup('esc\tx"q\\5') == \
         db1.dictionary.lookup('esc\tx"q\\5')
+
+
+def test_bulk_load_dedups_against_python_prefix(tmp_path):
+    """Terms already interned in the Python dictionary (including "" at
+    id 0) must resolve to their EXISTING ids during a native bulk load —
+    a duplicate annex id would make constant-term queries miss the
+    bulk-loaded rows.  (Found by property-based fuzzing: the empty
+    literal crashed the bump arena and then exposed the prefix-dedup
+    hole.)"""
+    import pytest
+    from kolibrie_amd import SparqlDatabase
+    from kolibrie_amd.ops import _native
+    if _native is None:
+        pytest.skip("native extension required")
+    p = tmp_path / "d.nt"
+    p.write_text('<a> <b> "" .\n<c> <b> "" .\n<a> <b> "x" .\n')
+    db = SparqlDatabase()
+    db.add_triple("<a>", "<pre>", '"seen-before"')  # pre-bulk interning
+    pre_id = db.dictionary.lookup("a")
+    db.parse_ntriples_file(str(p))
+    assert db.triple_count() == 4
+    # "" (Python prefix id 0) matches bulk-loaded rows
+    assert db.query('SELECT ?s WHERE { ?s ?p "" } ORDER BY ?s') == \
+        [["a"], ["c"]]
+    # the pre-interned subject unifies with its bulk occurrences
+    assert db.query('SELECT ?o WHERE { <a> <b> ?o } ORDER BY ?o') == \
+        [[""], ["x"]]
+    # no duplicate id was allocated
+    assert db.dictionary.lookup("a") == pre_id
+    mod, h = db.dictionary.annex
+    assert mod.vocab_lookup(h, "a") == pre_id
