@@ -132,3 +132,31 @@ def test_dictionary_encode_bijection(strings):
         assert d.encode(s) == i
     distinct = len(set(strings) | {""})
     assert len(d) == distinct
+
+
+@settings(max_examples=30, deadline=None)
+@given(st.lists(st.lists(st.tuples(st.integers(0, 20), st.integers(0, 20)),
+                         min_size=0, max_size=25),
+                min_size=1, max_size=6),
+       st.sampled_from(["RSTREAM", "ISTREAM", "DSTREAM"]))
+def test_r2s_columnar_equals_host_sets(frames, mode):
+    """K10 columnar R2S diff must emit the same SETS as the host
+    tuple-set operator for any frame sequence."""
+    import torch
+    from kolibrie_amd.rsp.r2s import Relation2StreamOperator
+
+    host = Relation2StreamOperator(mode)
+    dev = Relation2StreamOperator(mode)
+    for frame in frames:
+        got_host = set(map(tuple, host.eval([tuple(t) for t in frame])))
+        if frame:
+            s = torch.tensor([t[0] for t in frame], dtype=torch.int32)
+            o = torch.tensor([t[1] for t in frame], dtype=torch.int32)
+            cols = dev.eval_columns([s, o])
+        else:
+            cols = dev.eval_columns([
+                torch.empty(0, dtype=torch.int32),
+                torch.empty(0, dtype=torch.int32)])
+        got_dev = (set(zip(cols[0].tolist(), cols[1].tolist()))
+                   if cols else set())
+        assert got_dev == got_host, (mode, frame)
