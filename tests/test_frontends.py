@@ -242,3 +242,25 @@ def test_compat_knowledge_graph_rule_surface():
     kg.add_rule(r2)
     new2 = kg.infer_new_facts_semi_naive()
     assert ("<http://e/a>", "<http://e/self_paid>", "yes") in new2
+
+
+def test_cli_end_to_end(tmp_path, capsys):
+    """CLI: --file + --query loads (format-sniffed), executes, prints TSV
+    and JSON (ref cli/src/main.rs:15-41)."""
+    from kolibrie_amd.frontends.cli import main
+
+    nt = tmp_path / "data.nt"
+    nt.write_text('<http://e/a> <http://e/p> "1" .\n'
+                  '<http://e/b> <http://e/p> "2" .\n')
+    rc = main(["-f", str(nt), "-q",
+               "SELECT ?s ?o WHERE { ?s <http://e/p> ?o } ORDER BY ?o"])
+    out = capsys.readouterr().out.strip().splitlines()
+    assert rc == 0
+    assert out == ["http://e/a\t1", "http://e/b\t2"]
+
+    qf = tmp_path / "q.rq"
+    qf.write_text("SELECT (COUNT(*) AS ?c) WHERE { ?s ?p ?o }")
+    rc = main(["-f", str(nt), "--query-file", str(qf), "--format", "json"])
+    import json
+    assert rc == 0
+    assert json.loads(capsys.readouterr().out) == [["2"]]
