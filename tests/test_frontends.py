@@ -264,3 +264,23 @@ def test_cli_end_to_end(tmp_path, capsys):
     import json
     assert rc == 0
     assert json.loads(capsys.readouterr().out) == [["2"]]
+
+
+def test_query_engine_explain_renders_plan():
+    """explain() renders the optimized physical plan with scan/join
+    nodes and estimated orders (ref query_engine.rs explain)."""
+    from kolibrie_amd.engine.query_engine import QueryEngine
+
+    qe = QueryEngine()
+    for i in range(20):
+        qe.add_triple(f"<http://e/s{i}>", "<http://e/p>", f'"{i}"')
+        qe.add_triple(f"<http://e/s{i}>", "<http://e/q>", "<http://e/t>")
+    text = qe.explain(
+        "SELECT ?s ?o WHERE { ?s <http://e/p> ?o . ?s <http://e/q> ?t }")
+    assert "Scan" in text or "scan" in text
+    assert "Join" in text or "join" in text or "Star" in text
+    # the engine answers the same query it explained
+    rows = qe.query(
+        "SELECT (COUNT(*) AS ?c) WHERE { ?s <http://e/p> ?o . "
+        "?s <http://e/q> ?t }")
+    assert rows == [["20"]]
