@@ -160,3 +160,48 @@ def test_r2s_columnar_equals_host_sets(frames, mode):
         got_dev = (set(zip(cols[0].tolist(), cols[1].tolist()))
                    if cols else set())
         assert got_dev == got_host, (mode, frame)
+
+
+@settings(max_examples=15, deadline=None)
+@given(st.lists(st.tuples(st.integers(0, 6), st.integers(0, 6))
+                .filter(lambda e: e[0] < e[1]),   # DAG edges: SLD depth
+                min_size=1, max_size=12, unique=True))  # stays bounded
+def test_backward_chaining_agrees_with_forward_closure(edges):
+    """For the canonical transitive rule, backward goal resolution must
+    return exactly the forward-materialized answers (SLD vs semi-naive
+    oracle property)."""
+    from kolibrie_amd.reasoning.reasoner import Reasoner
+    from kolibrie_amd.reasoning.rule import Rule
+    from kolibrie_amd.storage.terms import Constant, TriplePattern, Variable
+
+    r = Reasoner()
+    for a, b in edges:
+        r.add_abox_triple(f"n{a}", "edge", f"n{b}")
+    pid = r._i32(r.dictionary.encode("path"))
+    eid = r._i32(r.dictionary.encode("edge"))
+    r.add_rule(Rule(
+        premise=[TriplePattern(Variable("x"), Constant(eid), Variable("y"))],
+        conclusion=[TriplePattern(Variable("x"), Constant(pid),
+                                  Variable("y"))]))
+    r.add_rule(Rule(
+        premise=[TriplePattern(Variable("x"), Constant(pid), Variable("y")),
+                 TriplePattern(Variable("y"), Constant(eid), Variable("z"))],
+        conclusion=[TriplePattern(Variable("x"), Constant(pid),
+                                  Variable("z"))]))
+
+    # forward oracle: python transitive closure over the edge set
+    import itertools
+    reach = set(edges)
+    changed = True
+    while changed:
+        changed = False
+        for (a, b), (c, d) in itertools.product(list(reach), list(reach)):
+            if b == c and (a, d) not in reach:
+                reach.add((a, d))
+                changed = True
+
+    start = edges[0][0]
+    want = sorted({d for (a, d) in reach if a == start})
+    got = r.backward_chaining((f"n{start}", "path", "?z"), max_depth=16)
+    got_ids = sorted({r.dictionary.decode(b["z"]) for b in got})
+    assert got_ids == [f"n{d}" for d in want], (edges, start)
