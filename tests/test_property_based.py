@@ -239,3 +239,30 @@ def test_group_aggregates_match_python_oracle(rows):
     got_n = [[r[0]] + [num(c) for c in r[1:]] for r in got]
     want_n = [[r[0]] + [num(c) for c in r[1:]] for r in want]
     assert got_n == want_n
+
+
+@settings(max_examples=25, deadline=None)
+@given(st.lists(st.integers(-500, 500), min_size=1, max_size=40,
+                unique=True),
+       st.integers(0, 10), st.integers(0, 10), st.booleans())
+def test_order_limit_offset_semantics(vals, limit, offset, desc):
+    """ORDER BY (numeric literals) + LIMIT/OFFSET must equal Python
+    sorted()[offset:offset+limit]; DISTINCT removes duplicates."""
+    from kolibrie_amd import SparqlDatabase
+
+    db = SparqlDatabase()
+    for i, v in enumerate(vals):
+        db.add_triple(f"<http://e/s{i}>", "<http://e/v>", f'"{v}"')
+        # duplicate object rows for the DISTINCT check
+        db.add_triple(f"<http://e/dup{i}>", "<http://e/v>", f'"{v}"')
+    direction = "DESC(?v)" if desc else "?v"
+    got = db.query(
+        f"SELECT ?v WHERE {{ ?s <http://e/v> ?v }} ORDER BY {direction} "
+        f"LIMIT {limit} OFFSET {offset}")
+    ordered = sorted(vals * 2, reverse=desc)
+    want = [[str(v)] for v in ordered[offset:offset + limit]]
+    assert [[r[0]] for r in got] == want
+
+    got_d = db.query(
+        "SELECT DISTINCT ?v WHERE { ?s <http://e/v> ?v }")
+    assert sorted(int(r[0]) for r in got_d) == sorted(vals)
