@@ -266,3 +266,38 @@ def test_order_limit_offset_semantics(vals, limit, offset, desc):
     got_d = db.query(
         "SELECT DISTINCT ?v WHERE { ?s <http://e/v> ?v }")
     assert sorted(int(r[0]) for r in got_d) == sorted(vals)
+
+
+# recursive RDF-star term strategy: IRIs at the leaves, quoted triples
+# nesting up to depth 3
+_star_leaf = st.text(alphabet=string.ascii_lowercase + string.digits,
+                     min_size=1, max_size=8).map(lambda x: f"<http://t/{x}>")
+_star_term = st.recursive(
+    _star_leaf,
+    lambda inner: st.tuples(inner, _star_leaf, inner).map(
+        lambda t: f"<< {t[0]} {t[1]} {t[2]} >>"),
+    max_leaves=6)
+
+
+@settings(max_examples=30, deadline=None)
+@given(_star_term)
+def test_rdf_star_encode_decode_roundtrip(term):
+    """encode_term_star -> decode_term -> encode_term_star must be the
+    identity for arbitrarily nested quoted triples, and interning is
+    idempotent (same id on re-encode of the original text)."""
+    from kolibrie_amd import SparqlDatabase
+
+    db = SparqlDatabase()
+    tid = db.encode_term_star(term)
+    assert db.encode_term_star(term) == tid  # idempotent interning
+    back = db.decode_term(tid)
+    assert back is not None
+    # decoded rendering re-encodes to the SAME id (round trip identity);
+    # plain IRIs decode bracket-less by design, so re-encode via the
+    # star-aware path only when the decoded form is a quoted triple
+    if back.startswith("<<"):
+        assert db.encode_term_star(back) == tid
+        for leaf in ("http://t/",):
+            assert leaf in back
+    else:
+        assert db.dictionary.lookup(back) == tid
